@@ -1,0 +1,39 @@
+# qrack_amd native build — hipcc, gfx950 only (MI355X / CDNA4).
+# Host-only translation units still compile with hipcc for flag consistency;
+# device code lives in csrc/hip/*.hip.
+
+HIPCC      ?= hipcc
+GPU_ARCH   ?= gfx950
+PYTHON     ?= python3
+EXT_SUFFIX := $(shell $(PYTHON) -c "import sysconfig; print(sysconfig.get_config_var('EXT_SUFFIX'))")
+PYBIND_INC := $(shell $(PYTHON) -m pybind11 --includes)
+
+BUILD      := build
+TARGET     := qrack_amd/_qrack$(EXT_SUFFIX)
+
+CXXFLAGS   := -O3 -std=c++17 -fPIC -Wno-unused-result -DQRACK_AMD_ENABLE_HIP \
+              --offload-arch=$(GPU_ARCH) -Icsrc $(PYBIND_INC)
+LDFLAGS    := -shared
+
+CPP_SRCS   := $(wildcard csrc/*.cpp) $(wildcard csrc/common/*.cpp)
+HIP_SRCS   := $(wildcard csrc/hip/*.hip)
+OBJS       := $(patsubst csrc/%.cpp,$(BUILD)/%.o,$(CPP_SRCS)) \
+              $(patsubst csrc/hip/%.hip,$(BUILD)/hip/%.o,$(HIP_SRCS))
+
+all: $(TARGET)
+
+$(BUILD)/%.o: csrc/%.cpp $(wildcard csrc/*.hpp) $(wildcard csrc/common/*.hpp) $(wildcard csrc/hip/*.hpp)
+	@mkdir -p $(dir $@)
+	$(HIPCC) $(CXXFLAGS) -c $< -o $@
+
+$(BUILD)/hip/%.o: csrc/hip/%.hip $(wildcard csrc/*.hpp) $(wildcard csrc/common/*.hpp) $(wildcard csrc/hip/*.hpp)
+	@mkdir -p $(dir $@)
+	$(HIPCC) $(CXXFLAGS) -x hip -c $< -o $@
+
+$(TARGET): $(OBJS)
+	$(HIPCC) $(LDFLAGS) $(OBJS) -o $@
+
+clean:
+	rm -rf $(BUILD) qrack_amd/_qrack*.so
+
+.PHONY: all clean

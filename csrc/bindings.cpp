@@ -1,0 +1,302 @@
+// pybind11 bindings for qrack_amd.
+//
+// The product API surface parallels the reference C ABI
+// (/root/reference/src/pinvoke_api.cpp) but is exposed as a proper Python
+// class per precision (QSimF / QSimD) instead of integer simulator handles;
+// the pure-Python package layer (qrack_amd/__init__.py) adds the stack
+// factory and torch interop.
+#include <pybind11/complex.h>
+#include <pybind11/numpy.h>
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include "qengine_cpu.hpp"
+#include "qfactory.hpp"
+
+namespace py = pybind11;
+using namespace qrack_amd;
+
+template <typename R> static const char* dtypeName();
+template <> const char* dtypeName<float>() { return "complex64"; }
+template <> const char* dtypeName<double>() { return "complex128"; }
+
+template <typename R> static void bindQInterface(py::module_& m, const char* name)
+{
+    using QI = QInterface<R>;
+    using Ptr = QInterfacePtr<R>;
+    using C = std::complex<R>;
+
+    auto cls = py::class_<QI, Ptr>(m, name);
+    cls.def_property_readonly("num_qubits", &QI::GetQubitCount)
+        .def_property_readonly("max_q_power", &QI::GetMaxQPower)
+        .def("set_random_seed", &QI::SetRandomSeed)
+        // ---- state access ----
+        .def("set_permutation",
+            [](QI& q, bitCapInt perm) { q.SetPermutation(perm); })
+        .def("get_amplitude", [](QI& q, bitCapInt i) { return to_std(q.GetAmplitude(i)); })
+        .def("set_amplitude", [](QI& q, bitCapInt i, C a) { q.SetAmplitude(i, from_std<R>(a)); })
+        .def("get_state_vector",
+            [](QI& q) {
+                py::array_t<C> out((py::ssize_t)q.GetMaxQPower());
+                q.GetQuantumState(reinterpret_cast<cplx<R>*>(out.mutable_data()));
+                return out;
+            })
+        .def("set_state_vector",
+            [](QI& q, py::array_t<C, py::array::c_style | py::array::forcecast> in) {
+                if ((bitCapInt)in.size() != q.GetMaxQPower())
+                    throw QrackError("state vector size mismatch");
+                q.SetQuantumState(reinterpret_cast<const cplx<R>*>(in.data()));
+            })
+        .def("get_probs",
+            [](QI& q) {
+                py::array_t<R> out((py::ssize_t)q.GetMaxQPower());
+                q.GetProbs(out.mutable_data());
+                return out;
+            })
+        // ---- single-qubit gates ----
+        .def("mtrx",
+            [](QI& q, std::vector<C> m, bitLenInt t) {
+                if (m.size() != 4) throw QrackError("mtrx: need 4 entries");
+                const cplx<R> mm[4] = { from_std<R>(m[0]), from_std<R>(m[1]), from_std<R>(m[2]),
+                    from_std<R>(m[3]) };
+                q.Mtrx(mm, t);
+            })
+        .def("phase",
+            [](QI& q, C tl, C br, bitLenInt t) { q.Phase(from_std<R>(tl), from_std<R>(br), t); })
+        .def("invert",
+            [](QI& q, C tr, C bl, bitLenInt t) { q.Invert(from_std<R>(tr), from_std<R>(bl), t); })
+        .def("x", [](QI& q, bitLenInt t) { q.X(t); })
+        .def("y", [](QI& q, bitLenInt t) { q.Y(t); })
+        .def("z", [](QI& q, bitLenInt t) { q.Z(t); })
+        .def("h", [](QI& q, bitLenInt t) { q.H(t); })
+        .def("s", [](QI& q, bitLenInt t) { q.S(t); })
+        .def("is_", [](QI& q, bitLenInt t) { q.IS(t); })
+        .def("t", [](QI& q, bitLenInt t) { q.T(t); })
+        .def("it", [](QI& q, bitLenInt t) { q.IT(t); })
+        .def("sqrt_x", [](QI& q, bitLenInt t) { q.SqrtX(t); })
+        .def("isqrt_x", [](QI& q, bitLenInt t) { q.ISqrtX(t); })
+        .def("rx", [](QI& q, R th, bitLenInt t) { q.RX(th, t); })
+        .def("ry", [](QI& q, R th, bitLenInt t) { q.RY(th, t); })
+        .def("rz", [](QI& q, R th, bitLenInt t) { q.RZ(th, t); })
+        .def("rt", [](QI& q, R th, bitLenInt t) { q.RT(th, t); })
+        .def("u", [](QI& q, bitLenInt t, R th, R ph, R lm) { q.U(t, th, ph, lm); })
+        .def("phase_root_n", [](QI& q, bitLenInt n, bitLenInt t) { q.PhaseRootN(n, t); })
+        // ---- controlled gates ----
+        .def("mcmtrx",
+            [](QI& q, std::vector<bitLenInt> c, std::vector<C> m, bitLenInt t) {
+                const cplx<R> mm[4] = { from_std<R>(m[0]), from_std<R>(m[1]), from_std<R>(m[2]),
+                    from_std<R>(m[3]) };
+                q.MCMtrx(c, mm, t);
+            })
+        .def("macmtrx",
+            [](QI& q, std::vector<bitLenInt> c, std::vector<C> m, bitLenInt t) {
+                const cplx<R> mm[4] = { from_std<R>(m[0]), from_std<R>(m[1]), from_std<R>(m[2]),
+                    from_std<R>(m[3]) };
+                q.MACMtrx(c, mm, t);
+            })
+        .def("ucmtrx",
+            [](QI& q, std::vector<bitLenInt> c, std::vector<C> m, bitLenInt t, bitCapInt perm) {
+                const cplx<R> mm[4] = { from_std<R>(m[0]), from_std<R>(m[1]), from_std<R>(m[2]),
+                    from_std<R>(m[3]) };
+                q.UCMtrx(c, mm, t, perm);
+            })
+        .def("uniformly_controlled_single_bit",
+            [](QI& q, std::vector<bitLenInt> c, bitLenInt t,
+                py::array_t<C, py::array::c_style | py::array::forcecast> mtrxs) {
+                if ((size_t)mtrxs.size() != (size_t)(4u * pow2((bitLenInt)c.size())))
+                    throw QrackError("multiplexer: need 4*2^len(controls) entries");
+                q.UniformlyControlledSingleBit(
+                    c, t, reinterpret_cast<const cplx<R>*>(mtrxs.data()));
+            })
+        .def("mcphase",
+            [](QI& q, std::vector<bitLenInt> c, C tl, C br, bitLenInt t) {
+                q.MCPhase(c, from_std<R>(tl), from_std<R>(br), t);
+            })
+        .def("mcinvert",
+            [](QI& q, std::vector<bitLenInt> c, C tr, C bl, bitLenInt t) {
+                q.MCInvert(c, from_std<R>(tr), from_std<R>(bl), t);
+            })
+        .def("macphase",
+            [](QI& q, std::vector<bitLenInt> c, C tl, C br, bitLenInt t) {
+                q.MACPhase(c, from_std<R>(tl), from_std<R>(br), t);
+            })
+        .def("macinvert",
+            [](QI& q, std::vector<bitLenInt> c, C tr, C bl, bitLenInt t) {
+                q.MACInvert(c, from_std<R>(tr), from_std<R>(bl), t);
+            })
+        .def("cnot", [](QI& q, bitLenInt c, bitLenInt t) { q.CNOT(c, t); })
+        .def("anti_cnot", [](QI& q, bitLenInt c, bitLenInt t) { q.AntiCNOT(c, t); })
+        .def("ccnot", [](QI& q, bitLenInt c1, bitLenInt c2, bitLenInt t) { q.CCNOT(c1, c2, t); })
+        .def("cy", [](QI& q, bitLenInt c, bitLenInt t) { q.CY(c, t); })
+        .def("cz", [](QI& q, bitLenInt c, bitLenInt t) { q.CZ(c, t); })
+        .def("ccz", [](QI& q, bitLenInt c1, bitLenInt c2, bitLenInt t) { q.CCZ(c1, c2, t); })
+        .def("ch", [](QI& q, bitLenInt c, bitLenInt t) { q.CH(c, t); })
+        .def("cs", [](QI& q, bitLenInt c, bitLenInt t) { q.CS(c, t); })
+        .def("cphase_root_n", [](QI& q, bitLenInt n, bitLenInt c, bitLenInt t) { q.CPhaseRootN(n, c, t); })
+        .def("crz", [](QI& q, R th, bitLenInt c, bitLenInt t) { q.CRZ(th, c, t); })
+        // ---- swaps ----
+        .def("swap", [](QI& q, bitLenInt a, bitLenInt b) { q.Swap(a, b); })
+        .def("iswap", [](QI& q, bitLenInt a, bitLenInt b) { q.ISwap(a, b); })
+        .def("iiswap", [](QI& q, bitLenInt a, bitLenInt b) { q.IISwap(a, b); })
+        .def("sqrt_swap", [](QI& q, bitLenInt a, bitLenInt b) { q.SqrtSwap(a, b); })
+        .def("isqrt_swap", [](QI& q, bitLenInt a, bitLenInt b) { q.ISqrtSwap(a, b); })
+        .def("fsim", [](QI& q, R th, R ph, bitLenInt a, bitLenInt b) { q.FSim(th, ph, a, b); })
+        .def("cswap",
+            [](QI& q, std::vector<bitLenInt> c, bitLenInt a, bitLenInt b) { q.CSwap(c, a, b); })
+        .def("csqrt_swap",
+            [](QI& q, std::vector<bitLenInt> c, bitLenInt a, bitLenInt b) { q.CSqrtSwap(c, a, b); })
+        // ---- mask gates ----
+        .def("x_mask", [](QI& q, bitCapInt m) { q.XMask(m); })
+        .def("y_mask", [](QI& q, bitCapInt m) { q.YMask(m); })
+        .def("z_mask", [](QI& q, bitCapInt m) { q.ZMask(m); })
+        .def("phase_parity", [](QI& q, R r, bitCapInt m) { q.PhaseParity(r, m); })
+        .def("uniform_parity_rz", [](QI& q, bitCapInt m, R a) { q.UniformParityRZ(m, a); })
+        .def("cuniform_parity_rz",
+            [](QI& q, std::vector<bitLenInt> c, bitCapInt m, R a) { q.CUniformParityRZ(c, m, a); })
+        // ---- QFT ----
+        .def("qft", [](QI& q, bitLenInt s, bitLenInt l) { q.QFT(s, l); }, py::arg("start") = 0,
+            py::arg("length") = 0)
+        .def("iqft", [](QI& q, bitLenInt s, bitLenInt l) { q.IQFT(s, l); }, py::arg("start") = 0,
+            py::arg("length") = 0)
+        // ---- structural ----
+        .def("compose", [](QI& q, Ptr other) { return q.Compose(other); })
+        .def("compose_at", [](QI& q, Ptr other, bitLenInt start) { return q.Compose(other, start); })
+        .def("decompose", [](QI& q, bitLenInt start, Ptr dest) { q.Decompose(start, dest); })
+        .def("dispose", [](QI& q, bitLenInt start, bitLenInt len) { q.Dispose(start, len); })
+        .def("dispose_perm",
+            [](QI& q, bitLenInt start, bitLenInt len, bitCapInt perm) { q.Dispose(start, len, perm); })
+        .def("allocate", [](QI& q, bitLenInt len) { return q.Allocate(len); })
+        .def("clone", [](QI& q) { return q.Clone(); })
+        .def("try_separate_1", [](QI& q, bitLenInt qb) { return q.TrySeparate(qb); })
+        .def("try_separate_2", [](QI& q, bitLenInt q1, bitLenInt q2) { return q.TrySeparate(q1, q2); })
+        .def("try_separate",
+            [](QI& q, std::vector<bitLenInt> qs, R tol) { return q.TrySeparate(qs, tol); })
+        // ---- probability / measurement ----
+        .def("prob", &QI::Prob)
+        .def("prob_all", &QI::ProbAll)
+        .def("prob_mask", &QI::ProbMask)
+        .def("prob_reg", &QI::ProbReg)
+        .def("prob_parity", &QI::ProbParity)
+        .def("force_m", &QI::ForceM, py::arg("qubit"), py::arg("result"), py::arg("do_force") = true,
+            py::arg("do_apply") = true)
+        .def("m", [](QI& q, bitLenInt t) { return q.M(t); })
+        .def("m_all", [](QI& q) { return q.MAll(); })
+        .def("m_reg", [](QI& q, bitLenInt s, bitLenInt l) { return q.MReg(s, l); })
+        .def("force_m_reg", &QI::ForceMReg, py::arg("start"), py::arg("length"), py::arg("result"),
+            py::arg("do_force") = true, py::arg("do_apply") = true)
+        .def("force_m_parity", &QI::ForceMParity, py::arg("mask"), py::arg("result"),
+            py::arg("do_force") = true)
+        .def("multi_shot_measure_mask",
+            [](QI& q, std::vector<bitCapInt> qPowers, unsigned shots) {
+                return q.MultiShotMeasureMask(qPowers, shots);
+            })
+        .def("expectation_bits_all",
+            [](QI& q, std::vector<bitLenInt> bits) { return q.ExpectationBitsAll(bits); })
+        .def("variance_bits_all",
+            [](QI& q, std::vector<bitLenInt> bits) { return q.VarianceBitsAll(bits); })
+        .def("pauli_expectation",
+            [](QI& q, std::vector<bitLenInt> bits, std::vector<int> paulis) {
+                std::vector<Pauli> ps;
+                for (int p : paulis) ps.push_back((Pauli)p);
+                return q.PauliExpectation(bits, ps);
+            })
+        // ---- ALU ----
+        .def("inc", &QI::INC)
+        .def("dec", &QI::DEC)
+        .def("cinc", &QI::CINC)
+        .def("incc", &QI::INCC)
+        .def("decc", &QI::DECC)
+        .def("incs", &QI::INCS)
+        .def("mul", &QI::MUL)
+        .def("div", &QI::DIV)
+        .def("mul_mod_n_out", &QI::MULModNOut)
+        .def("imul_mod_n_out", &QI::IMULModNOut)
+        .def("pow_mod_n_out", &QI::POWModNOut)
+        .def("cmul", &QI::CMUL)
+        .def("cdiv", &QI::CDIV)
+        .def("cmul_mod_n_out", &QI::CMULModNOut)
+        .def("cpow_mod_n_out", &QI::CPOWModNOut)
+        .def("indexed_lda",
+            [](QI& q, bitLenInt is, bitLenInt il, bitLenInt vs, bitLenInt vl,
+                py::bytes values, bool reset) {
+                std::string v = values;
+                return q.IndexedLDA(is, il, vs, vl, (const unsigned char*)v.data(), reset);
+            },
+            py::arg("index_start"), py::arg("index_length"), py::arg("value_start"),
+            py::arg("value_length"), py::arg("values"), py::arg("reset_value") = true)
+        .def("indexed_adc",
+            [](QI& q, bitLenInt is, bitLenInt il, bitLenInt vs, bitLenInt vl, bitLenInt c,
+                py::bytes values) {
+                std::string v = values;
+                return q.IndexedADC(is, il, vs, vl, c, (const unsigned char*)v.data());
+            })
+        .def("indexed_sbc",
+            [](QI& q, bitLenInt is, bitLenInt il, bitLenInt vs, bitLenInt vl, bitLenInt c,
+                py::bytes values) {
+                std::string v = values;
+                return q.IndexedSBC(is, il, vs, vl, c, (const unsigned char*)v.data());
+            })
+        .def("hash",
+            [](QI& q, bitLenInt s, bitLenInt l, py::bytes values) {
+                std::string v = values;
+                q.Hash(s, l, (const unsigned char*)v.data());
+            })
+        .def("full_add", &QI::FullAdd)
+        .def("ifull_add", &QI::IFullAdd)
+        .def("phase_flip_if_less", &QI::PhaseFlipIfLess)
+        .def("cphase_flip_if_less", &QI::CPhaseFlipIfLess)
+        .def("zero_phase_flip", &QI::ZeroPhaseFlip)
+        .def("phase_flip", &QI::PhaseFlip)
+        .def("rol", &QI::ROL)
+        .def("ror", &QI::ROR)
+        // ---- time evolution ----
+        .def("time_evolve",
+            [](QI& q, py::list ops, R t) {
+                std::vector<HamiltonianOp<R>> h;
+                for (auto item : ops) {
+                    py::dict d = item.cast<py::dict>();
+                    HamiltonianOp<R> op;
+                    op.target = d["target"].cast<bitLenInt>();
+                    if (d.contains("controls"))
+                        op.controls = d["controls"].cast<std::vector<bitLenInt>>();
+                    if (d.contains("anti")) op.anti = d["anti"].cast<bool>();
+                    if (d.contains("uniform")) op.uniform = d["uniform"].cast<bool>();
+                    auto mat = d["matrix"].cast<std::vector<C>>();
+                    for (C c : mat) op.matrix.push_back(from_std<R>(c));
+                    h.push_back(std::move(op));
+                }
+                q.TimeEvolve(h, t);
+            })
+        // ---- norm / compare ----
+        .def("update_running_norm", [](QI& q) { q.UpdateRunningNorm(); })
+        .def("normalize_state", [](QI& q) { q.NormalizeState(); })
+        .def("sum_sqr_diff", &QI::SumSqrDiff)
+        .def("approx_compare", &QI::ApproxCompare, py::arg("other"), py::arg("error_tol") = (R)1e-4)
+        .def("finish", &QI::Finish)
+        .def("is_finished", &QI::isFinished)
+        .def("get_unitary_fidelity", &QI::GetUnitaryFidelity)
+        .def("reset_unitary_fidelity", &QI::ResetUnitaryFidelity)
+        .def("set_device", &QI::SetDevice)
+        .def("get_device", &QI::GetDevice)
+        .def("is_clifford", [](QI& q) { return q.isClifford(); })
+        .def("depolarizing_channel_weak_1qb", &QI::DepolarizingChannelWeak1Qb);
+}
+
+PYBIND11_MODULE(_qrack, m)
+{
+    m.doc() = "qrack_amd native core (MI355X / HIP)";
+
+    bindQInterface<float>(m, "QInterfaceF");
+    bindQInterface<double>(m, "QInterfaceD");
+
+    m.def("create", &CreateStack<float>, py::arg("qubits"), py::arg("layers") = std::vector<std::string>{ "cpu" },
+        py::arg("init_perm") = (bitCapInt)0, py::arg("seed") = (int64_t)-1,
+        py::arg("device_id") = (int64_t)-1, py::arg("pages_per_device") = (bitLenInt)1);
+    m.def("create_d", &CreateStack<double>, py::arg("qubits"),
+        py::arg("layers") = std::vector<std::string>{ "cpu" }, py::arg("init_perm") = (bitCapInt)0,
+        py::arg("seed") = (int64_t)-1, py::arg("device_id") = (int64_t)-1,
+        py::arg("pages_per_device") = (bitLenInt)1);
+
+    m.def("hip_device_count", &HipDeviceCount);
+    m.attr("__version__") = "0.1.0";
+}

@@ -1,0 +1,136 @@
+// qrack_amd — CPU Schrödinger state-vector engine.
+//
+// Capability parity target: /root/reference/include/qengine_cpu.hpp +
+// /root/reference/src/qengine/{state,arithmetic,utility}.cpp. New design:
+// dense 64-byte-aligned amplitude array + the shared ThreadPool ParallelFor;
+// the full ALU family runs as out-of-place permutation copies.
+#pragma once
+
+#include "common/parallel_for.hpp"
+#include "qengine.hpp"
+
+namespace qrack_amd {
+
+template <typename R> class QEngineCPU;
+template <typename R> using QEngineCPUPtr = std::shared_ptr<QEngineCPU<R>>;
+
+template <typename R> class QEngineCPU : public QEngine<R>, public ParallelFor {
+protected:
+    using QInterface<R>::qubitCount;
+    using QInterface<R>::maxQPower;
+    using QInterface<R>::doNormalize;
+    using QInterface<R>::amplitudeFloor;
+    using QEngine<R>::runningNorm;
+
+    std::vector<cplx<R>> stateVec;
+
+public:
+    QEngineCPU(bitLenInt qBitCount, bitCapInt initState = 0u, RngPtr rgp = nullptr,
+        bool doNorm = true, R normThresh = eps<R>::value,
+        cplx<R> initPhase = cplx<R>((R)1, (R)0));
+
+    cplx<R>* Amplitudes() { return stateVec.data(); }
+
+    // ---- state access ----
+    void SetQuantumState(const cplx<R>* inputState) override;
+    void GetQuantumState(cplx<R>* outputState) override;
+    void GetProbs(R* outputProbs) override;
+    cplx<R> GetAmplitude(bitCapInt perm) override { return stateVec[perm]; }
+    void SetAmplitude(bitCapInt perm, cplx<R> amp) override { stateVec[perm] = amp; }
+    void SetPermutation(bitCapInt perm, cplx<R> phase = cplx<R>((R)1, (R)0)) override;
+
+    // ---- engine primitives ----
+    void Apply2x2(bitCapInt offset1, bitCapInt offset2, const cplx<R>* mtrx,
+        const std::vector<bitCapInt>& qPowersSorted) override;
+    void ApplyM(bitCapInt regMask, bitCapInt result, cplx<R> nrm) override;
+    void GetAmplitudePage(cplx<R>* pagePtr, bitCapInt offset, bitCapInt length) override;
+    void SetAmplitudePage(const cplx<R>* pagePtr, bitCapInt offset, bitCapInt length) override;
+    void SetAmplitudePage(
+        QEnginePtr<R> pageEnginePtr, bitCapInt srcOffset, bitCapInt dstOffset, bitCapInt length) override;
+    void ShuffleBuffers(QEnginePtr<R> engine) override;
+    void ZeroAmplitudes() override;
+    void CopyStateVec(QEnginePtr<R> src) override;
+    bool IsZeroAmplitude() override;
+
+    // ---- fast paths ----
+    void XMask(bitCapInt mask) override;
+    void ZMask(bitCapInt mask) override;
+    void PhaseParity(R radians, bitCapInt mask) override;
+    void UniformlyControlledSingleBit(
+        const std::vector<bitLenInt>& controls, bitLenInt target, const cplx<R>* mtrxs) override;
+    void ROL(bitLenInt shift, bitLenInt start, bitLenInt length) override;
+
+    // ---- probability / measurement ----
+    R Prob(bitLenInt q) override;
+    R ProbAll(bitCapInt perm) override { return norm(stateVec[perm]); }
+    R ProbMask(bitCapInt mask, bitCapInt permutation) override;
+    R ProbReg(bitLenInt start, bitLenInt length, bitCapInt permutation) override;
+    R ProbParity(bitCapInt mask) override;
+    bool ForceMParity(bitCapInt mask, bool result, bool doForce = true) override;
+    bitCapInt MAll() override;
+    std::map<bitCapInt, int> MultiShotMeasureMask(
+        const std::vector<bitCapInt>& qPowers, unsigned shots) override;
+    double ExpectationBitsFactorized(const std::vector<bitLenInt>& bits,
+        const std::vector<bitCapInt>& perms, bitCapInt offset = 0) override;
+    double VarianceBitsAll(const std::vector<bitLenInt>& bits, bitCapInt offset = 0) override;
+
+    // ---- structural ----
+    using QInterface<R>::Compose;
+    bitLenInt Compose(QInterfacePtr<R> toCopy, bitLenInt start) override;
+    void Decompose(bitLenInt start, QInterfacePtr<R> dest) override;
+    void Dispose(bitLenInt start, bitLenInt length) override;
+    void Dispose(bitLenInt start, bitLenInt length, bitCapInt disposedPerm) override;
+    bitLenInt Allocate(bitLenInt start, bitLenInt length) override;
+    QInterfacePtr<R> Clone() override;
+
+    // ---- norm ----
+    void UpdateRunningNorm(R norm_thresh = (R)-1) override;
+    void NormalizeState(R nrm = (R)-1, R norm_thresh = (R)-1, R phaseArg = 0) override;
+    double SumSqrDiff(QInterfacePtr<R> other) override;
+
+    // ---- ALU (parity: qheader_alu.cl kernel family) ----
+    void INC(bitCapInt toAdd, bitLenInt start, bitLenInt length) override;
+    void CINC(bitCapInt toAdd, bitLenInt start, bitLenInt length,
+        const std::vector<bitLenInt>& controls) override;
+    void INCC(bitCapInt toAdd, bitLenInt start, bitLenInt length, bitLenInt carryIndex) override;
+    void DECC(bitCapInt toSub, bitLenInt start, bitLenInt length, bitLenInt carryIndex) override;
+    void INCS(bitCapInt toAdd, bitLenInt start, bitLenInt length, bitLenInt overflowIndex) override;
+    void MUL(bitCapInt toMul, bitLenInt inOutStart, bitLenInt carryStart, bitLenInt length) override;
+    void DIV(bitCapInt toDiv, bitLenInt inOutStart, bitLenInt carryStart, bitLenInt length) override;
+    void MULModNOut(bitCapInt toMul, bitCapInt modN, bitLenInt inStart, bitLenInt outStart,
+        bitLenInt length) override;
+    void IMULModNOut(bitCapInt toMul, bitCapInt modN, bitLenInt inStart, bitLenInt outStart,
+        bitLenInt length) override;
+    void POWModNOut(bitCapInt base, bitCapInt modN, bitLenInt inStart, bitLenInt outStart,
+        bitLenInt length) override;
+    void CMUL(bitCapInt toMul, bitLenInt inOutStart, bitLenInt carryStart, bitLenInt length,
+        const std::vector<bitLenInt>& controls) override;
+    void CDIV(bitCapInt toDiv, bitLenInt inOutStart, bitLenInt carryStart, bitLenInt length,
+        const std::vector<bitLenInt>& controls) override;
+    void CMULModNOut(bitCapInt toMul, bitCapInt modN, bitLenInt inStart, bitLenInt outStart,
+        bitLenInt length, const std::vector<bitLenInt>& controls) override;
+    void CIMULModNOut(bitCapInt toMul, bitCapInt modN, bitLenInt inStart, bitLenInt outStart,
+        bitLenInt length, const std::vector<bitLenInt>& controls) override;
+    void CPOWModNOut(bitCapInt base, bitCapInt modN, bitLenInt inStart, bitLenInt outStart,
+        bitLenInt length, const std::vector<bitLenInt>& controls) override;
+    bitCapInt IndexedLDA(bitLenInt indexStart, bitLenInt indexLength, bitLenInt valueStart,
+        bitLenInt valueLength, const unsigned char* values, bool resetValue = true) override;
+    bitCapInt IndexedADC(bitLenInt indexStart, bitLenInt indexLength, bitLenInt valueStart,
+        bitLenInt valueLength, bitLenInt carryIndex, const unsigned char* values) override;
+    bitCapInt IndexedSBC(bitLenInt indexStart, bitLenInt indexLength, bitLenInt valueStart,
+        bitLenInt valueLength, bitLenInt carryIndex, const unsigned char* values) override;
+    void Hash(bitLenInt start, bitLenInt length, const unsigned char* values) override;
+    void PhaseFlipIfLess(bitCapInt greaterPerm, bitLenInt start, bitLenInt length) override;
+    void CPhaseFlipIfLess(
+        bitCapInt greaterPerm, bitLenInt start, bitLenInt length, bitLenInt flagIndex) override;
+
+protected:
+    // Run `perm` over all indices (optionally restricted by controlMask all-set),
+    // writing nState[f(i)] = state[i]; f must be a bijection on the iterated set.
+    void PermutationOp(const std::function<bitCapInt(bitCapInt)>& f);
+    void ControlledPermutationOp(
+        bitCapInt controlMask, const std::function<bitCapInt(bitCapInt)>& f);
+    bitCapInt SampleOnce();
+};
+
+} // namespace qrack_amd

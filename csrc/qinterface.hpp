@@ -1,0 +1,378 @@
+// qrack_amd — the QInterface public API.
+//
+// Capability parity target: /root/reference/include/qinterface.hpp (the ~300
+// virtual-method API every Qrack layer implements: gates, registers, ALU,
+// QFT, measurement, expectation, Compose/Decompose/Dispose, TimeEvolve).
+// This is a fresh design for the MI355X build: the primitive set engines must
+// implement is deliberately small (controlled 2x2 apply, multiplexer, mask
+// gates, probability reductions, compose/decompose, measurement collapse);
+// the rest of the API is default-lowered here, and wrapper layers (QUnit,
+// QPager, QStabilizerHybrid) intercept at the virtual gate level.
+#pragma once
+
+#include "common/rng.hpp"
+#include "common/types.hpp"
+
+#include <map>
+#include <memory>
+#include <vector>
+
+namespace qrack_amd {
+
+enum Pauli : uint8_t { PauliI = 0, PauliX = 1, PauliZ = 2, PauliY = 3 };
+
+template <typename R> class QInterface;
+template <typename R> using QInterfacePtr = std::shared_ptr<QInterface<R>>;
+
+// A term of a Trotterized Hamiltonian: (anti-)controlled 2x2 Hermitian op.
+// Parity target: /root/reference/include/hamiltonian.hpp:29-95.
+template <typename R> struct HamiltonianOp {
+    bitLenInt target;
+    std::vector<bitLenInt> controls;
+    bool anti = false;
+    std::vector<cplx<R>> matrix; // 4 entries; or 4*2^(nControls) when uniform
+    bool uniform = false;
+};
+
+template <typename R> class QInterface : public std::enable_shared_from_this<QInterface<R>> {
+protected:
+    bitLenInt qubitCount;
+    bitCapInt maxQPower;
+    RngPtr rand_generator;
+    bool doNormalize;
+    R amplitudeFloor;
+
+    void SetQubitCount(bitLenInt qb)
+    {
+        qubitCount = qb;
+        maxQPower = pow2(qb);
+    }
+
+public:
+    QInterface(bitLenInt nQubits, RngPtr rgp = nullptr, bool doNorm = true,
+        R normThresh = eps<R>::value)
+        : qubitCount(nQubits)
+        , maxQPower(pow2(nQubits))
+        , rand_generator(rgp ? rgp : std::make_shared<Rng>())
+        , doNormalize(doNorm)
+        , amplitudeFloor(normThresh)
+    {
+    }
+    virtual ~QInterface() = default;
+
+    bitLenInt GetQubitCount() const { return qubitCount; }
+    bitCapInt GetMaxQPower() const { return maxQPower; }
+    RngPtr GetRng() const { return rand_generator; }
+    double Rand() { return rand_generator->rand(); }
+    virtual void SetRandomSeed(uint64_t s) { rand_generator->seed(s); }
+    virtual bool isClifford() const { return false; }
+    virtual bool isClifford(bitLenInt q) const { return false; }
+
+    // ---- state access ------------------------------------------------------
+    virtual void SetQuantumState(const cplx<R>* inputState) = 0;
+    virtual void GetQuantumState(cplx<R>* outputState) = 0;
+    virtual void GetProbs(R* outputProbs);
+    virtual cplx<R> GetAmplitude(bitCapInt perm) = 0;
+    virtual void SetAmplitude(bitCapInt perm, cplx<R> amp) = 0;
+    virtual void SetPermutation(bitCapInt perm, cplx<R> phase = cplx<R>((R)1, (R)0)) = 0;
+
+    // ---- primitive gate API (wrapper-interceptable virtuals) ---------------
+    virtual void Mtrx(const cplx<R>* mtrx, bitLenInt target) = 0;
+    virtual void Phase(cplx<R> topLeft, cplx<R> bottomRight, bitLenInt target);
+    virtual void Invert(cplx<R> topRight, cplx<R> bottomLeft, bitLenInt target);
+    virtual void MCMtrx(const std::vector<bitLenInt>& controls, const cplx<R>* mtrx, bitLenInt target);
+    virtual void MACMtrx(const std::vector<bitLenInt>& controls, const cplx<R>* mtrx, bitLenInt target);
+    virtual void MCPhase(
+        const std::vector<bitLenInt>& controls, cplx<R> topLeft, cplx<R> bottomRight, bitLenInt target);
+    virtual void MCInvert(
+        const std::vector<bitLenInt>& controls, cplx<R> topRight, cplx<R> bottomLeft, bitLenInt target);
+    virtual void MACPhase(
+        const std::vector<bitLenInt>& controls, cplx<R> topLeft, cplx<R> bottomRight, bitLenInt target);
+    virtual void MACInvert(
+        const std::vector<bitLenInt>& controls, cplx<R> topRight, cplx<R> bottomLeft, bitLenInt target);
+    // Apply mtrx only for the control permutation controlPerm (bit i of
+    // controlPerm = required value of controls[i]); general mixed-polarity.
+    virtual void UCMtrx(const std::vector<bitLenInt>& controls, const cplx<R>* mtrx, bitLenInt target,
+        bitCapInt controlPerm) = 0;
+    // Multiplexer: selects mtrxs[perm(controls)] per basis state.
+    // Parity: qinterface.hpp UniformlyControlledSingleBit.
+    virtual void UniformlyControlledSingleBit(
+        const std::vector<bitLenInt>& controls, bitLenInt target, const cplx<R>* mtrxs) = 0;
+
+    virtual void Swap(bitLenInt q1, bitLenInt q2);
+    virtual void ISwap(bitLenInt q1, bitLenInt q2);
+    virtual void IISwap(bitLenInt q1, bitLenInt q2);
+    virtual void SqrtSwap(bitLenInt q1, bitLenInt q2);
+    virtual void ISqrtSwap(bitLenInt q1, bitLenInt q2);
+    virtual void FSim(R theta, R phi, bitLenInt q1, bitLenInt q2);
+    virtual void CSwap(const std::vector<bitLenInt>& controls, bitLenInt q1, bitLenInt q2);
+    virtual void AntiCSwap(const std::vector<bitLenInt>& controls, bitLenInt q1, bitLenInt q2);
+    virtual void CSqrtSwap(const std::vector<bitLenInt>& controls, bitLenInt q1, bitLenInt q2);
+    virtual void AntiCSqrtSwap(const std::vector<bitLenInt>& controls, bitLenInt q1, bitLenInt q2);
+
+    // multi-qubit mask gates (parity: qinterface.hpp XMask/ZMask/PhaseParity)
+    virtual void XMask(bitCapInt mask);
+    virtual void YMask(bitCapInt mask);
+    virtual void ZMask(bitCapInt mask);
+    virtual void PhaseParity(R radians, bitCapInt mask);
+
+    // ---- named single-qubit gates (convenience; non-virtual) ---------------
+    void X(bitLenInt q)
+    {
+        Invert(cplx<R>(1, 0), cplx<R>(1, 0), q);
+    }
+    void Y(bitLenInt q) { Invert(cplx<R>(0, -1), cplx<R>(0, 1), q); }
+    void Z(bitLenInt q) { Phase(cplx<R>(1, 0), cplx<R>(-1, 0), q); }
+    void H(bitLenInt q)
+    {
+        const R s = SQRT1_2_R<R>;
+        const cplx<R> m[4] = { { s, 0 }, { s, 0 }, { s, 0 }, { -s, 0 } };
+        Mtrx(m, q);
+    }
+    void S(bitLenInt q) { Phase(cplx<R>(1, 0), cplx<R>(0, 1), q); }
+    void IS(bitLenInt q) { Phase(cplx<R>(1, 0), cplx<R>(0, -1), q); }
+    void T(bitLenInt q) { Phase(cplx<R>(1, 0), polar<R>(1, PI_R<R> / 4), q); }
+    void IT(bitLenInt q) { Phase(cplx<R>(1, 0), polar<R>(1, -PI_R<R> / 4), q); }
+    void SqrtX(bitLenInt q)
+    {
+        const cplx<R> m[4] = { { (R)0.5, (R)0.5 }, { (R)0.5, (R)-0.5 }, { (R)0.5, (R)-0.5 },
+            { (R)0.5, (R)0.5 } };
+        Mtrx(m, q);
+    }
+    void ISqrtX(bitLenInt q)
+    {
+        const cplx<R> m[4] = { { (R)0.5, (R)-0.5 }, { (R)0.5, (R)0.5 }, { (R)0.5, (R)0.5 },
+            { (R)0.5, (R)-0.5 } };
+        Mtrx(m, q);
+    }
+    void SqrtY(bitLenInt q)
+    {
+        const cplx<R> m[4] = { { (R)0.5, (R)0.5 }, { (R)-0.5, (R)-0.5 }, { (R)0.5, (R)0.5 },
+            { (R)0.5, (R)0.5 } };
+        Mtrx(m, q);
+    }
+    void ISqrtY(bitLenInt q)
+    {
+        const cplx<R> m[4] = { { (R)0.5, (R)-0.5 }, { (R)0.5, (R)-0.5 }, { (R)-0.5, (R)0.5 },
+            { (R)0.5, (R)-0.5 } };
+        Mtrx(m, q);
+    }
+    void RX(R theta, bitLenInt q)
+    {
+        const R c = std::cos(theta / 2), s = std::sin(theta / 2);
+        const cplx<R> m[4] = { { c, 0 }, { 0, -s }, { 0, -s }, { c, 0 } };
+        Mtrx(m, q);
+    }
+    void RY(R theta, bitLenInt q)
+    {
+        const R c = std::cos(theta / 2), s = std::sin(theta / 2);
+        const cplx<R> m[4] = { { c, 0 }, { -s, 0 }, { s, 0 }, { c, 0 } };
+        Mtrx(m, q);
+    }
+    void RZ(R theta, bitLenInt q)
+    {
+        Phase(polar<R>(1, -theta / 2), polar<R>(1, theta / 2), q);
+    }
+    void RT(R theta, bitLenInt q) { Phase(cplx<R>(1, 0), polar<R>(1, theta), q); }
+    void U(bitLenInt q, R theta, R phi, R lambda)
+    {
+        const R c = std::cos(theta / 2), s = std::sin(theta / 2);
+        const cplx<R> m[4] = { { c, 0 }, (R)(-s) * polar<R>(1, lambda), s * polar<R>(1, phi),
+            c * polar<R>(1, phi + lambda) };
+        Mtrx(m, q);
+    }
+    void U2(bitLenInt q, R phi, R lambda) { U(q, PI_R<R> / 2, phi, lambda); }
+    // phase root N: diag(1, exp(2 pi i / 2^n)) — the QFT phase family
+    void PhaseRootN(bitLenInt n, bitLenInt q)
+    {
+        if (n == 0) return;
+        Phase(cplx<R>(1, 0), polar<R>(1, PI_R<R> / (R)pow2(n - 1)), q);
+    }
+    void IPhaseRootN(bitLenInt n, bitLenInt q)
+    {
+        if (n == 0) return;
+        Phase(cplx<R>(1, 0), polar<R>(1, -PI_R<R> / (R)pow2(n - 1)), q);
+    }
+
+    // named controlled gates
+    void CNOT(bitLenInt c, bitLenInt t) { MCInvert({ c }, cplx<R>(1, 0), cplx<R>(1, 0), t); }
+    void AntiCNOT(bitLenInt c, bitLenInt t) { MACInvert({ c }, cplx<R>(1, 0), cplx<R>(1, 0), t); }
+    void CCNOT(bitLenInt c1, bitLenInt c2, bitLenInt t)
+    {
+        MCInvert({ c1, c2 }, cplx<R>(1, 0), cplx<R>(1, 0), t);
+    }
+    void AntiCCNOT(bitLenInt c1, bitLenInt c2, bitLenInt t)
+    {
+        MACInvert({ c1, c2 }, cplx<R>(1, 0), cplx<R>(1, 0), t);
+    }
+    void CY(bitLenInt c, bitLenInt t) { MCInvert({ c }, cplx<R>(0, -1), cplx<R>(0, 1), t); }
+    void CZ(bitLenInt c, bitLenInt t) { MCPhase({ c }, cplx<R>(1, 0), cplx<R>(-1, 0), t); }
+    void AntiCZ(bitLenInt c, bitLenInt t) { MACPhase({ c }, cplx<R>(1, 0), cplx<R>(-1, 0), t); }
+    void CCZ(bitLenInt c1, bitLenInt c2, bitLenInt t)
+    {
+        MCPhase({ c1, c2 }, cplx<R>(1, 0), cplx<R>(-1, 0), t);
+    }
+    void CH(bitLenInt c, bitLenInt t)
+    {
+        const R s = SQRT1_2_R<R>;
+        const cplx<R> m[4] = { { s, 0 }, { s, 0 }, { s, 0 }, { -s, 0 } };
+        MCMtrx({ c }, m, t);
+    }
+    void CS(bitLenInt c, bitLenInt t) { MCPhase({ c }, cplx<R>(1, 0), cplx<R>(0, 1), t); }
+    void CIS(bitLenInt c, bitLenInt t) { MCPhase({ c }, cplx<R>(1, 0), cplx<R>(0, -1), t); }
+    void CPhaseRootN(bitLenInt n, bitLenInt c, bitLenInt t)
+    {
+        if (n == 0) return;
+        MCPhase({ c }, cplx<R>(1, 0), polar<R>(1, PI_R<R> / (R)pow2(n - 1)), t);
+    }
+    void CIPhaseRootN(bitLenInt n, bitLenInt c, bitLenInt t)
+    {
+        if (n == 0) return;
+        MCPhase({ c }, cplx<R>(1, 0), polar<R>(1, -PI_R<R> / (R)pow2(n - 1)), t);
+    }
+    void CRZ(R theta, bitLenInt c, bitLenInt t)
+    {
+        MCPhase({ c }, polar<R>(1, -theta / 2), polar<R>(1, theta / 2), t);
+    }
+
+    // ---- register-wide helpers ---------------------------------------------
+    void X(bitLenInt start, bitLenInt length)
+    {
+        XMask(pow2Mask(length) << start);
+    }
+    void H(bitLenInt start, bitLenInt length)
+    {
+        for (bitLenInt i = 0; i < length; ++i) H(start + i);
+    }
+    void Z(bitLenInt start, bitLenInt length) { ZMask(pow2Mask(length) << start); }
+
+    // ---- QFT (parity: src/qinterface/qinterface.cpp:114-194) ---------------
+    virtual void QFT(bitLenInt start, bitLenInt length, bool trySeparate = false);
+    virtual void IQFT(bitLenInt start, bitLenInt length, bool trySeparate = false);
+    virtual void QFTR(const std::vector<bitLenInt>& qubits, bool trySeparate = false);
+    virtual void IQFTR(const std::vector<bitLenInt>& qubits, bool trySeparate = false);
+
+    // ---- structural ops ----------------------------------------------------
+    virtual bitLenInt Compose(QInterfacePtr<R> toCopy);                   // append at top
+    virtual bitLenInt Compose(QInterfacePtr<R> toCopy, bitLenInt start) = 0;
+    virtual void Decompose(bitLenInt start, QInterfacePtr<R> dest) = 0;
+    virtual void Dispose(bitLenInt start, bitLenInt length) = 0;
+    virtual void Dispose(bitLenInt start, bitLenInt length, bitCapInt disposedPerm) = 0;
+    virtual bitLenInt Allocate(bitLenInt start, bitLenInt length) = 0;
+    bitLenInt Allocate(bitLenInt length) { return Allocate(qubitCount, length); }
+    virtual QInterfacePtr<R> Clone() = 0;
+    virtual bool TrySeparate(bitLenInt q) { return false; }
+    virtual bool TrySeparate(bitLenInt q1, bitLenInt q2) { return false; }
+    virtual bool TrySeparate(const std::vector<bitLenInt>& qubits, R error_tol) { return false; }
+
+    // ---- probability / measurement -----------------------------------------
+    virtual R Prob(bitLenInt q) = 0;
+    virtual R ProbAll(bitCapInt perm) { return norm(GetAmplitude(perm)); }
+    virtual R ProbMask(bitCapInt mask, bitCapInt permutation);
+    virtual R ProbReg(bitLenInt start, bitLenInt length, bitCapInt permutation)
+    {
+        return ProbMask(pow2Mask(length) << start, permutation << start);
+    }
+    virtual R ProbParity(bitCapInt mask);
+    virtual R CProb(bitLenInt control, bitLenInt target); // Prob(target | control=1)
+    virtual R ACProb(bitLenInt control, bitLenInt target);
+
+    virtual bool ForceM(bitLenInt q, bool result, bool doForce = true, bool doApply = true) = 0;
+    bool M(bitLenInt q) { return ForceM(q, false, false, true); }
+    virtual bool ForceMParity(bitCapInt mask, bool result, bool doForce = true);
+    virtual bitCapInt ForceMReg(
+        bitLenInt start, bitLenInt length, bitCapInt result, bool doForce = true, bool doApply = true);
+    bitCapInt MReg(bitLenInt start, bitLenInt length) { return ForceMReg(start, length, 0, false, true); }
+    virtual bitCapInt MAll() { return MReg(0, qubitCount); }
+    virtual std::map<bitCapInt, int> MultiShotMeasureMask(
+        const std::vector<bitCapInt>& qPowers, unsigned shots);
+    virtual void MultiShotMeasureMask(
+        const std::vector<bitCapInt>& qPowers, unsigned shots, unsigned long long* shotsArray);
+
+    // expectation / variance over bit-permutation values
+    // (parity: qinterface.hpp:2483-2798 family)
+    virtual double ExpectationBitsAll(const std::vector<bitLenInt>& bits, bitCapInt offset = 0);
+    virtual double ExpectationBitsFactorized(
+        const std::vector<bitLenInt>& bits, const std::vector<bitCapInt>& perms, bitCapInt offset = 0);
+    virtual double VarianceBitsAll(const std::vector<bitLenInt>& bits, bitCapInt offset = 0);
+    virtual double PauliExpectation(const std::vector<bitLenInt>& bits, const std::vector<Pauli>& paulis);
+
+    // parity rotation family (parity: include/qparity.hpp)
+    virtual void UniformParityRZ(bitCapInt mask, R angle);
+    virtual void CUniformParityRZ(const std::vector<bitLenInt>& controls, bitCapInt mask, R angle);
+
+    // ---- ALU (parity: include/qalu.hpp; implemented by engines) ------------
+    virtual void INC(bitCapInt toAdd, bitLenInt start, bitLenInt length);
+    virtual void DEC(bitCapInt toSub, bitLenInt start, bitLenInt length);
+    virtual void CINC(
+        bitCapInt toAdd, bitLenInt start, bitLenInt length, const std::vector<bitLenInt>& controls);
+    virtual void CDEC(
+        bitCapInt toSub, bitLenInt start, bitLenInt length, const std::vector<bitLenInt>& controls);
+    virtual void INCC(bitCapInt toAdd, bitLenInt start, bitLenInt length, bitLenInt carryIndex);
+    virtual void DECC(bitCapInt toSub, bitLenInt start, bitLenInt length, bitLenInt carryIndex);
+    virtual void INCS(bitCapInt toAdd, bitLenInt start, bitLenInt length, bitLenInt overflowIndex);
+    virtual void DECS(bitCapInt toSub, bitLenInt start, bitLenInt length, bitLenInt overflowIndex);
+    virtual void MUL(bitCapInt toMul, bitLenInt inOutStart, bitLenInt carryStart, bitLenInt length);
+    virtual void DIV(bitCapInt toDiv, bitLenInt inOutStart, bitLenInt carryStart, bitLenInt length);
+    virtual void MULModNOut(
+        bitCapInt toMul, bitCapInt modN, bitLenInt inStart, bitLenInt outStart, bitLenInt length);
+    virtual void IMULModNOut(
+        bitCapInt toMul, bitCapInt modN, bitLenInt inStart, bitLenInt outStart, bitLenInt length);
+    virtual void POWModNOut(
+        bitCapInt base, bitCapInt modN, bitLenInt inStart, bitLenInt outStart, bitLenInt length);
+    virtual void CMUL(bitCapInt toMul, bitLenInt inOutStart, bitLenInt carryStart, bitLenInt length,
+        const std::vector<bitLenInt>& controls);
+    virtual void CDIV(bitCapInt toDiv, bitLenInt inOutStart, bitLenInt carryStart, bitLenInt length,
+        const std::vector<bitLenInt>& controls);
+    virtual void CMULModNOut(bitCapInt toMul, bitCapInt modN, bitLenInt inStart, bitLenInt outStart,
+        bitLenInt length, const std::vector<bitLenInt>& controls);
+    virtual void CIMULModNOut(bitCapInt toMul, bitCapInt modN, bitLenInt inStart, bitLenInt outStart,
+        bitLenInt length, const std::vector<bitLenInt>& controls);
+    virtual void CPOWModNOut(bitCapInt base, bitCapInt modN, bitLenInt inStart, bitLenInt outStart,
+        bitLenInt length, const std::vector<bitLenInt>& controls);
+    virtual bitCapInt IndexedLDA(bitLenInt indexStart, bitLenInt indexLength, bitLenInt valueStart,
+        bitLenInt valueLength, const unsigned char* values, bool resetValue = true);
+    virtual bitCapInt IndexedADC(bitLenInt indexStart, bitLenInt indexLength, bitLenInt valueStart,
+        bitLenInt valueLength, bitLenInt carryIndex, const unsigned char* values);
+    virtual bitCapInt IndexedSBC(bitLenInt indexStart, bitLenInt indexLength, bitLenInt valueStart,
+        bitLenInt valueLength, bitLenInt carryIndex, const unsigned char* values);
+    virtual void Hash(bitLenInt start, bitLenInt length, const unsigned char* values);
+    virtual void FullAdd(bitLenInt inputBit1, bitLenInt inputBit2, bitLenInt carryInSumOut,
+        bitLenInt carryOut);
+    virtual void IFullAdd(bitLenInt inputBit1, bitLenInt inputBit2, bitLenInt carryInSumOut,
+        bitLenInt carryOut);
+    virtual void PhaseFlipIfLess(bitCapInt greaterPerm, bitLenInt start, bitLenInt length);
+    virtual void CPhaseFlipIfLess(
+        bitCapInt greaterPerm, bitLenInt start, bitLenInt length, bitLenInt flagIndex);
+    virtual void ZeroPhaseFlip(bitLenInt start, bitLenInt length);
+    virtual void PhaseFlip(); // global -1
+
+    // register shift/rotate (parity: src/qinterface/arithmetic.cpp ROL/ROR)
+    virtual void ROL(bitLenInt shift, bitLenInt start, bitLenInt length);
+    virtual void ROR(bitLenInt shift, bitLenInt start, bitLenInt length);
+
+    // ---- Hamiltonian evolution (parity: src/qinterface/gates.cpp:426) ------
+    virtual void TimeEvolve(const std::vector<HamiltonianOp<R>>& h, R timeDiff);
+
+    // ---- norm bookkeeping / fidelity ---------------------------------------
+    virtual void UpdateRunningNorm(R norm_thresh = (R)-1) = 0;
+    virtual void NormalizeState(
+        R nrm = (R)-1, R norm_thresh = (R)-1, R phaseArg = 0) = 0;
+    virtual double SumSqrDiff(QInterfacePtr<R> other) = 0;
+    virtual bool ApproxCompare(QInterfacePtr<R> other, R error_tol = (R)1e-4)
+    {
+        return SumSqrDiff(other) <= (double)error_tol;
+    }
+    virtual void Finish() {}
+    virtual bool isFinished() { return true; }
+    virtual void Dump() {}
+    virtual double GetUnitaryFidelity() { return 1.0; }
+    virtual void ResetUnitaryFidelity() {}
+    virtual void SetDevice(int64_t deviceId) {}
+    virtual int64_t GetDevice() const { return -1; }
+
+    // depolarizing noise channel (parity: qinterface.hpp:3104)
+    virtual void DepolarizingChannelWeak1Qb(bitLenInt q, R lambda);
+};
+
+} // namespace qrack_amd

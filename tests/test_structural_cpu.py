@@ -1,0 +1,205 @@
+"""Compose/Decompose/Dispose/Allocate + QFT + mirror circuits (CPU engine).
+
+Parity model: /root/reference/test/tests.cpp (test_compose, test_decompose,
+test_qft_h, [mirror] cases).
+"""
+
+import numpy as np
+import pytest
+
+import qrack_amd as qa
+from ref_sim import RefSim, assert_states_close
+
+
+def make(n, seed=7, precision="fp32"):
+    return qa.create_simulator(n, precision=precision, engine="cpu", seed=seed)
+
+
+def test_compose():
+    a = make(1, seed=1)
+    a.h(0)
+    b = make(1, seed=2)
+    b.x(0)
+    a.compose(b)  # b becomes qubit 1
+    assert a.num_qubits == 2
+    sv = a.get_state_vector()
+    # (|0>+|1>)/sqrt2 (x) |1> -> amplitudes at 2 and 3
+    assert abs(abs(sv[2]) - 1 / np.sqrt(2)) < 1e-5
+    assert abs(abs(sv[3]) - 1 / np.sqrt(2)) < 1e-5
+
+
+def test_compose_at():
+    a = make(2, seed=1)
+    a.x(1)  # |10>
+    b = make(1, seed=2)
+    b.x(0)
+    a.compose_at(b, 1)  # insert at position 1
+    assert a.num_qubits == 3
+    # old q1 becomes q2: state |1 1 0> = 6
+    assert a.m_all() == 0b110
+
+
+def test_decompose_product_state():
+    q = make(3, seed=3)
+    q.h(0)
+    q.x(2)
+    dest = make(1, seed=4)
+    q.decompose(2, dest)  # split off qubit 2 (|1>)
+    assert q.num_qubits == 2
+    assert dest.num_qubits == 1
+    assert abs(dest.prob(0) - 1.0) < 1e-5
+    assert abs(q.prob(0) - 0.5) < 1e-5
+
+
+def test_decompose_entangled_within_part():
+    # qubits (1,2) Bell-entangled with each other but separable from qubit 0
+    q = make(3, seed=5)
+    q.h(1)
+    q.cnot(1, 2)
+    q.h(0)
+    dest = make(2, seed=6)
+    q.decompose(1, dest)
+    assert abs(dest.prob_mask(0b11, 0b00) - 0.5) < 1e-5
+    assert abs(dest.prob_mask(0b11, 0b11) - 0.5) < 1e-5
+    assert abs(q.prob(0) - 0.5) < 1e-5
+
+
+def test_dispose():
+    q = make(3, seed=7)
+    q.h(0)
+    q.x(1)
+    q.dispose(1, 1)
+    assert q.num_qubits == 2
+    assert abs(q.prob(0) - 0.5) < 1e-5
+    assert abs(q.prob(1)) < 1e-5
+
+
+def test_dispose_perm():
+    q = make(3, seed=7)
+    q.h(0)
+    q.x(1)
+    q.dispose_perm(1, 1, 1)
+    assert q.num_qubits == 2
+    assert abs(q.prob(0) - 0.5) < 1e-5
+
+
+def test_allocate():
+    q = make(2, seed=8)
+    q.h(0)
+    q.allocate(2)
+    assert q.num_qubits == 4
+    assert abs(q.prob(2)) < 1e-6
+    assert abs(q.prob(0) - 0.5) < 1e-5
+
+
+def test_clone_independent():
+    q = make(2, seed=9)
+    q.h(0)
+    c = q.clone()
+    c.x(1)
+    assert abs(q.prob(1)) < 1e-6
+    assert abs(c.prob(1) - 1.0) < 1e-6
+
+
+def test_qft_matches_dft():
+    # QFT on computational basis state |x> gives DFT column (bit-reversed order)
+    n = 4
+    x = 5
+    q = make(n, seed=10)
+    for i in range(n):
+        if (x >> i) & 1:
+            q.x(i)
+    q.qft(0, n)
+    sv = np.asarray(q.get_state_vector()).astype(np.complex128)
+    N = 1 << n
+    k = np.arange(N)
+    expected_full = np.exp(2j * np.pi * x * k / N) / np.sqrt(N)
+    # output is bit-reversed: sv[rev(k)] == expected_full[k]
+    rev = np.array([int(format(i, f"0{n}b")[::-1], 2) for i in range(N)])
+    got = sv[rev]
+    err = np.max(np.abs(got - expected_full))
+    # allow global phase
+    inner = np.vdot(expected_full, got)
+    assert abs(abs(inner) - 1.0) < 1e-4, f"err={err} inner={inner}"
+
+
+def test_qft_roundtrip_random_state():
+    n = 6
+    rng = np.random.default_rng(12)
+    q = make(n, seed=11)
+    for i in range(n):
+        q.ry(rng.uniform(0, np.pi), i)
+    for i in range(n - 1):
+        q.cnot(i, i + 1)
+    before = q.get_state_vector()
+    q.qft(0, n)
+    q.iqft(0, n)
+    after = q.get_state_vector()
+    assert np.allclose(before, after, atol=1e-4)
+
+
+@pytest.mark.parametrize("seed", [1, 2, 3])
+def test_mirror_circuit(seed):
+    """Random circuit + its inverse returns to the initial basis state
+    (parity model: the reference's 37 [mirror] cases, tests.cpp:5462+)."""
+    n = 6
+    depth = 20
+    rng = np.random.default_rng(seed)
+    q = make(n, seed=seed)
+    init = int(rng.integers(1 << n))
+    for i in range(n):
+        if (init >> i) & 1:
+            q.x(i)
+    ops = []
+    for _ in range(depth):
+        kind = rng.integers(5)
+        if kind == 0:
+            t = int(rng.integers(n))
+            th = float(rng.uniform(0, 2 * np.pi))
+            q.ry(th, t)
+            ops.append(("ry", th, t))
+        elif kind == 1:
+            t = int(rng.integers(n))
+            th = float(rng.uniform(0, 2 * np.pi))
+            q.rz(th, t)
+            ops.append(("rz", th, t))
+        elif kind == 2:
+            a, b = rng.choice(n, 2, replace=False)
+            q.cnot(int(a), int(b))
+            ops.append(("cnot", int(a), int(b)))
+        elif kind == 3:
+            t = int(rng.integers(n))
+            q.h(t)
+            ops.append(("h", t))
+        else:
+            a, b = rng.choice(n, 2, replace=False)
+            q.swap(int(a), int(b))
+            ops.append(("swap", int(a), int(b)))
+    for op in reversed(ops):
+        if op[0] == "ry":
+            q.ry(-op[1], op[2])
+        elif op[0] == "rz":
+            q.rz(-op[1], op[2])
+        elif op[0] == "cnot":
+            q.cnot(op[1], op[2])
+        elif op[0] == "h":
+            q.h(op[1])
+        else:
+            q.swap(op[1], op[2])
+    assert q.m_all() == init
+
+
+def test_sum_sqr_diff_and_approx_compare():
+    a = make(3, seed=1)
+    b = make(3, seed=2)
+    a.h(0)
+    b.h(0)
+    assert a.approx_compare(b)
+    assert a.sum_sqr_diff(b) < 1e-8
+    b.x(2)
+    assert not a.approx_compare(b)
+    # global phase invariance
+    c = make(3, seed=3)
+    c.h(0)
+    c.phase_flip()
+    assert a.approx_compare(c)
