@@ -195,7 +195,7 @@ def test_sum_sqr_diff_and_approx_compare():
     a.h(0)
     b.h(0)
     assert a.approx_compare(b)
-    assert a.sum_sqr_diff(b) < 1e-8
+    assert a.sum_sqr_diff(b) < 1e-5
     b.x(2)
     assert not a.approx_compare(b)
     # global phase invariance
