@@ -12,10 +12,10 @@ BUILD      := build
 TARGET     := qrack_amd/_qrack$(EXT_SUFFIX)
 
 CXXFLAGS   := -O3 -std=c++17 -fPIC -Wno-unused-result -DQRACK_AMD_ENABLE_HIP \
-              --offload-arch=$(GPU_ARCH) -Icsrc $(PYBIND_INC)
+              -DQRACK_AMD_HIP_ENGINE --offload-arch=$(GPU_ARCH) -Icsrc $(PYBIND_INC)
 LDFLAGS    := -shared
 
-CPP_SRCS   := $(wildcard csrc/*.cpp) $(wildcard csrc/common/*.cpp)
+CPP_SRCS   := $(wildcard csrc/*.cpp) $(wildcard csrc/common/*.cpp) $(wildcard csrc/hip/*.cpp)
 HIP_SRCS   := $(wildcard csrc/hip/*.hip)
 OBJS       := $(patsubst csrc/%.cpp,$(BUILD)/%.o,$(CPP_SRCS)) \
               $(patsubst csrc/hip/%.hip,$(BUILD)/hip/%.o,$(HIP_SRCS))

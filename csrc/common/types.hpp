@@ -7,6 +7,12 @@
 // (float / double) instead of the reference's FPPOW macro scheme.
 #pragma once
 
+#if defined(__HIPCC__) || defined(__HIP__)
+#define QA_HD __host__ __device__
+#else
+#define QA_HD
+#endif
+
 #include <cstdint>
 #include <cmath>
 #include <complex>
@@ -36,25 +42,25 @@ template <typename R> struct cplx {
     R re;
     R im;
     cplx() = default;
-    constexpr cplx(R r, R i) : re(r), im(i) {}
-    constexpr cplx(R r) : re(r), im(0) {}
+    QA_HD constexpr cplx(R r, R i) : re(r), im(i) {}
+    QA_HD constexpr cplx(R r) : re(r), im(0) {}
 };
 
-template <typename R> inline cplx<R> operator+(cplx<R> a, cplx<R> b) { return { a.re + b.re, a.im + b.im }; }
-template <typename R> inline cplx<R> operator-(cplx<R> a, cplx<R> b) { return { a.re - b.re, a.im - b.im }; }
-template <typename R> inline cplx<R> operator*(cplx<R> a, cplx<R> b)
+template <typename R> QA_HD inline cplx<R> operator+(cplx<R> a, cplx<R> b) { return { a.re + b.re, a.im + b.im }; }
+template <typename R> QA_HD inline cplx<R> operator-(cplx<R> a, cplx<R> b) { return { a.re - b.re, a.im - b.im }; }
+template <typename R> QA_HD inline cplx<R> operator*(cplx<R> a, cplx<R> b)
 {
     return { a.re * b.re - a.im * b.im, a.re * b.im + a.im * b.re };
 }
-template <typename R> inline cplx<R> operator*(R s, cplx<R> a) { return { s * a.re, s * a.im }; }
-template <typename R> inline cplx<R> operator*(cplx<R> a, R s) { return { s * a.re, s * a.im }; }
+template <typename R> QA_HD inline cplx<R> operator*(R s, cplx<R> a) { return { s * a.re, s * a.im }; }
+template <typename R> QA_HD inline cplx<R> operator*(cplx<R> a, R s) { return { s * a.re, s * a.im }; }
 template <typename R> inline cplx<R> operator/(cplx<R> a, cplx<R> b)
 {
     R d = b.re * b.re + b.im * b.im;
     return { (a.re * b.re + a.im * b.im) / d, (a.im * b.re - a.re * b.im) / d };
 }
-template <typename R> inline cplx<R> conj(cplx<R> a) { return { a.re, -a.im }; }
-template <typename R> inline R norm(cplx<R> a) { return a.re * a.re + a.im * a.im; }
+template <typename R> QA_HD inline cplx<R> conj(cplx<R> a) { return { a.re, -a.im }; }
+template <typename R> QA_HD inline R norm(cplx<R> a) { return a.re * a.re + a.im * a.im; }
 template <typename R> inline R abs(cplx<R> a) { return std::sqrt(norm(a)); }
 template <typename R> inline R arg(cplx<R> a)
 {

@@ -1,0 +1,823 @@
+// qrack_amd — CDNA4 (gfx950) device kernels for the HIP state-vector engine.
+//
+// Parity target: the reference kernel inventory in
+// /root/reference/src/common/qengine.cl + qheader_alu.cl,
+// re-designed for MI355X rather than translated:
+//  - wave64 blocks (256 threads), grid-stride loops capped at ~8 blocks/CU;
+//  - fp32 amplitude streams vectorized as float4 (two complex amps per lane,
+//    16 B/lane — G13 of the CDNA HIP guide); fp64 is naturally 16 B/lane;
+//  - reductions: wave shuffle (64-wide) -> LDS across the block's 4 waves ->
+//    one partial per block, finished on host (no global atomics);
+//  - the entire ALU family is ONE uniform-opcode permutation kernel (the
+//    branch is wave-uniform) instead of 26 near-identical kernels.
+#include "kernels.hpp"
+
+namespace qrack_amd {
+
+constexpr int QA_BLOCK = 256;
+constexpr int QA_MAX_BLOCKS = 2048; // 256 CUs x 8 blocks
+
+static inline int gridFor(bitCapInt n)
+{
+    bitCapInt b = (n + QA_BLOCK - 1) / QA_BLOCK;
+    if (b > (bitCapInt)QA_MAX_BLOCKS) b = QA_MAX_BLOCKS;
+    if (b < 1) b = 1;
+    return (int)b;
+}
+
+int reduceGridSize(bitCapInt n) { return gridFor(n); }
+
+__device__ __forceinline__ bitCapInt expandBits(bitCapInt j, const bitCapInt* pows, int n)
+{
+    for (int k = 0; k < n; ++k) {
+        j = ((j & ~(pows[k] - 1u)) << 1u) | (j & (pows[k] - 1u));
+    }
+    return j;
+}
+
+// ---- gate apply -------------------------------------------------------------
+
+// KIND: 0 = generic 2x2, 1 = phase (diagonal), 2 = invert (antidiagonal)
+template <typename R, int KIND> __global__ void k_apply2x2(cplx<R>* sv, GateArgs<R> a)
+{
+    const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
+    for (bitCapInt j = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; j < a.maxI; j += stride) {
+        const bitCapInt i = expandBits(j, a.qPowers, a.nPowers);
+        const bitCapInt i1 = i | a.offset1;
+        const bitCapInt i2 = i | a.offset2;
+        if (KIND == 1) {
+            sv[i1] = a.m[0] * sv[i1];
+            sv[i2] = a.m[3] * sv[i2];
+        } else if (KIND == 2) {
+            const cplx<R> t = sv[i1];
+            sv[i1] = a.m[1] * sv[i2];
+            sv[i2] = a.m[2] * t;
+        } else {
+            const cplx<R> x = sv[i1];
+            const cplx<R> y = sv[i2];
+            sv[i1] = a.m[0] * x + a.m[1] * y;
+            sv[i2] = a.m[2] * x + a.m[3] * y;
+        }
+    }
+}
+
+// fp32 vectorized single-target path: each lane handles TWO adjacent pairs
+// through float4 (16 B) loads/stores. Requires nPowers == 1 and even maxI.
+template <int KIND> __global__ void k_apply2x2_1v(cplx<float>* sv, GateArgs<float> a)
+{
+    float4* sv4 = reinterpret_cast<float4*>(sv);
+    const bitCapInt p = a.qPowers[0];
+    const bitCapInt half = a.maxI >> 1u; // float4-pair iterations
+    const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
+    const cplx<float> m0 = a.m[0], m1 = a.m[1], m2 = a.m[2], m3 = a.m[3];
+    if (p == 1u) {
+        // target bit 0: each pair is adjacent = one float4
+        for (bitCapInt k = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; k < half; k += stride) {
+            // two pairs: float4s at 2k and 2k+1
+            for (int h = 0; h < 2; ++h) {
+                const bitCapInt idx = 2u * k + h;
+                float4 v = sv4[idx];
+                const cplx<float> x{ v.x, v.y }, y{ v.z, v.w };
+                cplx<float> nx, ny;
+                if (KIND == 1) {
+                    nx = m0 * x;
+                    ny = m3 * y;
+                } else if (KIND == 2) {
+                    nx = m1 * y;
+                    ny = m2 * x;
+                } else {
+                    nx = m0 * x + m1 * y;
+                    ny = m2 * x + m3 * y;
+                }
+                sv4[idx] = make_float4(nx.re, nx.im, ny.re, ny.im);
+            }
+        }
+    } else {
+        // target bit >= 1: pairs (i, i+p); two consecutive j share the block
+        for (bitCapInt k = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; k < half; k += stride) {
+            const bitCapInt j = 2u * k;
+            const bitCapInt i = ((j & ~(p - 1u)) << 1u) | (j & (p - 1u));
+            const bitCapInt lo4 = i >> 1u;       // float4 index of (i, i+1)
+            const bitCapInt hi4 = (i + p) >> 1u; // float4 index of (i+p, i+p+1)
+            float4 vlo = sv4[lo4];
+            float4 vhi = sv4[hi4];
+            const cplx<float> x0{ vlo.x, vlo.y }, x1{ vlo.z, vlo.w };
+            const cplx<float> y0{ vhi.x, vhi.y }, y1{ vhi.z, vhi.w };
+            cplx<float> a0, a1, b0, b1;
+            if (KIND == 1) {
+                a0 = m0 * x0; a1 = m0 * x1;
+                b0 = m3 * y0; b1 = m3 * y1;
+            } else if (KIND == 2) {
+                a0 = m1 * y0; a1 = m1 * y1;
+                b0 = m2 * x0; b1 = m2 * x1;
+            } else {
+                a0 = m0 * x0 + m1 * y0; a1 = m0 * x1 + m1 * y1;
+                b0 = m2 * x0 + m3 * y0; b1 = m2 * x1 + m3 * y1;
+            }
+            sv4[lo4] = make_float4(a0.re, a0.im, a1.re, a1.im);
+            sv4[hi4] = make_float4(b0.re, b0.im, b1.re, b1.im);
+        }
+    }
+}
+
+template <typename R> static int matrixKind(const cplx<R>* m)
+{
+    const bool isPhase = (norm(m[1]) <= 0) && (norm(m[2]) <= 0);
+    const bool isInvert = (norm(m[0]) <= 0) && (norm(m[3]) <= 0);
+    return isPhase ? 1 : (isInvert ? 2 : 0);
+}
+
+template <typename R>
+void launchApply2x2(cplx<R>* sv, const GateArgs<R>& a, hipStream_t stream)
+{
+    const int kind = matrixKind(a.m);
+    const int grid = gridFor(a.maxI);
+    if constexpr (std::is_same_v<R, float>) {
+        if (a.nPowers == 1 && (a.maxI & 1u) == 0u && a.maxI >= 2u && a.offset1 == 0u) {
+            const int gridv = gridFor(a.maxI >> 1u);
+            switch (kind) {
+            case 1:
+                hipLaunchKernelGGL((k_apply2x2_1v<1>), dim3(gridv), dim3(QA_BLOCK), 0, stream, sv, a);
+                return;
+            case 2:
+                hipLaunchKernelGGL((k_apply2x2_1v<2>), dim3(gridv), dim3(QA_BLOCK), 0, stream, sv, a);
+                return;
+            default:
+                hipLaunchKernelGGL((k_apply2x2_1v<0>), dim3(gridv), dim3(QA_BLOCK), 0, stream, sv, a);
+                return;
+            }
+        }
+    }
+    switch (kind) {
+    case 1:
+        hipLaunchKernelGGL((k_apply2x2<R, 1>), dim3(grid), dim3(QA_BLOCK), 0, stream, sv, a);
+        break;
+    case 2:
+        hipLaunchKernelGGL((k_apply2x2<R, 2>), dim3(grid), dim3(QA_BLOCK), 0, stream, sv, a);
+        break;
+    default:
+        hipLaunchKernelGGL((k_apply2x2<R, 0>), dim3(grid), dim3(QA_BLOCK), 0, stream, sv, a);
+        break;
+    }
+}
+
+// ---- multiplexer ------------------------------------------------------------
+
+template <typename R>
+__global__ void k_uniformly_ctrl(cplx<R>* sv, bitCapInt maxI, bitCapInt targetPower,
+    const bitCapInt* ctrlPowers, int nCtrls, const cplx<R>* mtrxs)
+{
+    const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
+    for (bitCapInt j = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; j < maxI; j += stride) {
+        const bitCapInt i = ((j & ~(targetPower - 1u)) << 1u) | (j & (targetPower - 1u));
+        bitCapInt sel = 0;
+        for (int b = 0; b < nCtrls; ++b) {
+            if (i & ctrlPowers[b]) sel |= (ONE_BCI << b);
+        }
+        const cplx<R>* m = mtrxs + 4u * sel;
+        const cplx<R> x = sv[i];
+        const cplx<R> y = sv[i | targetPower];
+        sv[i] = m[0] * x + m[1] * y;
+        sv[i | targetPower] = m[2] * x + m[3] * y;
+    }
+}
+
+template <typename R>
+void launchUniformlyControlled(cplx<R>* sv, bitCapInt maxI, bitCapInt targetPower,
+    const bitCapInt* ctrlPowersDev, int nCtrls, const cplx<R>* mtrxsDev, hipStream_t stream)
+{
+    hipLaunchKernelGGL((k_uniformly_ctrl<R>), dim3(gridFor(maxI)), dim3(QA_BLOCK), 0, stream, sv,
+        maxI, targetPower, ctrlPowersDev, nCtrls, mtrxsDev);
+}
+
+// ---- mask gates -------------------------------------------------------------
+
+template <typename R> __global__ void k_xmask(cplx<R>* sv, bitCapInt maxQPower, bitCapInt mask)
+{
+    const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
+    for (bitCapInt i = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; i < maxQPower; i += stride) {
+        const bitCapInt j = i ^ mask;
+        if (i < j) {
+            const cplx<R> t = sv[i];
+            sv[i] = sv[j];
+            sv[j] = t;
+        }
+    }
+}
+
+template <typename R>
+void launchXMask(cplx<R>* sv, bitCapInt maxQPower, bitCapInt mask, hipStream_t stream)
+{
+    hipLaunchKernelGGL(
+        (k_xmask<R>), dim3(gridFor(maxQPower)), dim3(QA_BLOCK), 0, stream, sv, maxQPower, mask);
+}
+
+template <typename R>
+__global__ void k_parityphase(cplx<R>* sv, bitCapInt maxQPower, bitCapInt mask, cplx<R> even, cplx<R> odd)
+{
+    const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
+    for (bitCapInt i = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; i < maxQPower; i += stride) {
+        sv[i] = (__popcll(i & mask) & 1 ? odd : even) * sv[i];
+    }
+}
+
+template <typename R>
+void launchParityPhase(
+    cplx<R>* sv, bitCapInt maxQPower, bitCapInt mask, cplx<R> even, cplx<R> odd, hipStream_t stream)
+{
+    hipLaunchKernelGGL((k_parityphase<R>), dim3(gridFor(maxQPower)), dim3(QA_BLOCK), 0, stream, sv,
+        maxQPower, mask, even, odd);
+}
+
+// ---- reductions --------------------------------------------------------------
+
+__device__ __forceinline__ double waveReduceSum(double v)
+{
+    for (int off = 32; off; off >>= 1) v += __shfl_down(v, off);
+    return v;
+}
+
+template <typename R> __global__ void k_reduce(const cplx<R>* sv, ReduceArgs a, int op, double* partials)
+{
+    double s = 0;
+    const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
+    for (bitCapInt i = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; i < a.maxI; i += stride) {
+        const double n = (double)norm(sv[i]);
+        switch ((ReduceOp)op) {
+        case ReduceOp::NORM_ALL:
+            s += n;
+            break;
+        case ReduceOp::PROB_BITSET:
+            if ((i & a.mask) == a.mask) s += n;
+            break;
+        case ReduceOp::PROB_MASK:
+            if ((i & a.mask) == a.perm) s += n;
+            break;
+        case ReduceOp::PROB_PARITY:
+            if (__popcll(i & a.mask) & 1) s += n;
+            break;
+        case ReduceOp::EXP_PERM:
+        case ReduceOp::EXP_PERM_SQ: {
+            double val = a.offset;
+            for (int b = 0; b < a.nBits; ++b) {
+                if ((i >> a.bits[b]) & 1u) val += (double)a.perms[b];
+            }
+            s += ((ReduceOp)op == ReduceOp::EXP_PERM) ? val * n : val * val * n;
+            break;
+        }
+        case ReduceOp::NORM_FLOOR:
+            if (n >= a.normThresh) s += n;
+            break;
+        }
+    }
+    s = waveReduceSum(s);
+    __shared__ double waveSums[QA_BLOCK / 64];
+    const int lane = threadIdx.x & 63;
+    const int wid = threadIdx.x >> 6;
+    if (lane == 0) waveSums[wid] = s;
+    __syncthreads();
+    if (wid == 0) {
+        double t = (lane < QA_BLOCK / 64) ? waveSums[lane] : 0.0;
+        t = waveReduceSum(t);
+        if (lane == 0) partials[blockIdx.x] = t;
+    }
+}
+
+template <typename R>
+int launchReduce(const cplx<R>* sv, const ReduceArgs& a, int op, double* partialsDev,
+    hipStream_t stream)
+{
+    const int grid = gridFor(a.maxI);
+    hipLaunchKernelGGL(
+        (k_reduce<R>), dim3(grid), dim3(QA_BLOCK), 0, stream, sv, a, op, partialsDev);
+    return grid;
+}
+
+template <typename R>
+__global__ void k_argmax(const cplx<R>* sv, bitCapInt maxI, double* vals, bitCapInt* idxs)
+{
+    double best = -1.0;
+    bitCapInt bestI = 0;
+    const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
+    for (bitCapInt i = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; i < maxI; i += stride) {
+        const double n = (double)norm(sv[i]);
+        if (n > best) {
+            best = n;
+            bestI = i;
+        }
+    }
+    for (int off = 32; off; off >>= 1) {
+        const double ov = __shfl_down(best, off);
+        const bitCapInt oi = (bitCapInt)__shfl_down((unsigned long long)bestI, off);
+        if (ov > best) {
+            best = ov;
+            bestI = oi;
+        }
+    }
+    __shared__ double wv[QA_BLOCK / 64];
+    __shared__ bitCapInt wi[QA_BLOCK / 64];
+    const int lane = threadIdx.x & 63;
+    const int wid = threadIdx.x >> 6;
+    if (lane == 0) {
+        wv[wid] = best;
+        wi[wid] = bestI;
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        for (int k = 1; k < QA_BLOCK / 64; ++k) {
+            if (wv[k] > wv[0]) {
+                wv[0] = wv[k];
+                wi[0] = wi[k];
+            }
+        }
+        vals[blockIdx.x] = wv[0];
+        idxs[blockIdx.x] = wi[0];
+    }
+}
+
+template <typename R>
+int launchArgMax(const cplx<R>* sv, bitCapInt maxI, double* valsDev, bitCapInt* idxDev,
+    hipStream_t stream)
+{
+    const int grid = gridFor(maxI);
+    hipLaunchKernelGGL(
+        (k_argmax<R>), dim3(grid), dim3(QA_BLOCK), 0, stream, sv, maxI, valsDev, idxDev);
+    return grid;
+}
+
+// ---- normalize / projection --------------------------------------------------
+
+template <typename R>
+__global__ void k_normalize(cplx<R>* sv, bitCapInt maxQPower, cplx<R> factor, R normThresh)
+{
+    const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
+    for (bitCapInt i = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; i < maxQPower; i += stride) {
+        const cplx<R> v = sv[i];
+        sv[i] = (norm(v) < normThresh) ? cplx<R>(0, 0) : factor * v;
+    }
+}
+
+template <typename R>
+void launchNormalize(cplx<R>* sv, bitCapInt maxQPower, cplx<R> factor, R normThresh, hipStream_t stream)
+{
+    hipLaunchKernelGGL((k_normalize<R>), dim3(gridFor(maxQPower)), dim3(QA_BLOCK), 0, stream, sv,
+        maxQPower, factor, normThresh);
+}
+
+template <typename R>
+__global__ void k_applym(cplx<R>* sv, bitCapInt maxQPower, bitCapInt mask, bitCapInt result, cplx<R> nrm)
+{
+    const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
+    for (bitCapInt i = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; i < maxQPower; i += stride) {
+        sv[i] = ((i & mask) == result) ? nrm * sv[i] : cplx<R>(0, 0);
+    }
+}
+
+template <typename R>
+void launchApplyM(cplx<R>* sv, bitCapInt maxQPower, bitCapInt mask, bitCapInt result, cplx<R> nrm,
+    hipStream_t stream)
+{
+    hipLaunchKernelGGL((k_applym<R>), dim3(gridFor(maxQPower)), dim3(QA_BLOCK), 0, stream, sv,
+        maxQPower, mask, result, nrm);
+}
+
+template <typename R>
+__global__ void k_applyparity(cplx<R>* sv, bitCapInt maxQPower, bitCapInt mask, int odd, cplx<R> nrm)
+{
+    const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
+    for (bitCapInt i = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; i < maxQPower; i += stride) {
+        sv[i] = ((__popcll(i & mask) & 1) == odd) ? nrm * sv[i] : cplx<R>(0, 0);
+    }
+}
+
+template <typename R>
+void launchApplyParity(cplx<R>* sv, bitCapInt maxQPower, bitCapInt mask, bool odd, cplx<R> nrm,
+    hipStream_t stream)
+{
+    hipLaunchKernelGGL((k_applyparity<R>), dim3(gridFor(maxQPower)), dim3(QA_BLOCK), 0, stream, sv,
+        maxQPower, mask, odd ? 1 : 0, nrm);
+}
+
+// ---- structural ---------------------------------------------------------------
+
+template <typename R>
+__global__ void k_compose(const cplx<R>* a, const cplx<R>* b, cplx<R>* out, bitCapInt nMaxQPower,
+    bitLenInt start, bitLenInt oQubits)
+{
+    const bitCapInt lowMask = (ONE_BCI << start) - 1u;
+    const bitCapInt midMask = (ONE_BCI << oQubits) - 1u;
+    const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
+    for (bitCapInt i = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; i < nMaxQPower; i += stride) {
+        const bitCapInt low = i & lowMask;
+        const bitCapInt mid = (i >> start) & midMask;
+        const bitCapInt high = i >> (start + oQubits);
+        out[i] = a[low | (high << start)] * b[mid];
+    }
+}
+
+template <typename R>
+void launchCompose(const cplx<R>* a, const cplx<R>* b, cplx<R>* out, bitCapInt nMaxQPower,
+    bitLenInt start, bitLenInt oQubits, hipStream_t stream)
+{
+    hipLaunchKernelGGL((k_compose<R>), dim3(gridFor(nMaxQPower)), dim3(QA_BLOCK), 0, stream, a, b,
+        out, nMaxQPower, start, oQubits);
+}
+
+template <typename R>
+__global__ void k_dispose_slice(const cplx<R>* in, cplx<R>* out, bitCapInt remPower, bitLenInt start,
+    bitLenInt length, bitCapInt slicePerm, cplx<R> scale)
+{
+    const bitCapInt lowMask = (ONE_BCI << start) - 1u;
+    const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
+    for (bitCapInt r = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; r < remPower; r += stride) {
+        const bitCapInt low = r & lowMask;
+        const bitCapInt high = (r >> start) << (start + length);
+        out[r] = scale * in[low | (slicePerm << start) | high];
+    }
+}
+
+template <typename R>
+void launchDisposeSlice(const cplx<R>* in, cplx<R>* out, bitCapInt remPower, bitLenInt start,
+    bitLenInt length, bitCapInt slicePerm, cplx<R> scale, hipStream_t stream)
+{
+    hipLaunchKernelGGL((k_dispose_slice<R>), dim3(gridFor(remPower)), dim3(QA_BLOCK), 0, stream, in,
+        out, remPower, start, length, slicePerm, scale);
+}
+
+template <typename R>
+__global__ void k_gather_part(const cplx<R>* in, cplx<R>* dest, bitCapInt partPower, bitLenInt start,
+    bitLenInt length, bitCapInt remIndex, cplx<R> scale)
+{
+    const bitCapInt lowMask = (ONE_BCI << start) - 1u;
+    const bitCapInt low = remIndex & lowMask;
+    const bitCapInt high = (remIndex >> start) << (start + length);
+    const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
+    for (bitCapInt p = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; p < partPower; p += stride) {
+        dest[p] = scale * in[low | (p << start) | high];
+    }
+}
+
+template <typename R>
+void launchGatherPart(const cplx<R>* in, cplx<R>* dest, bitCapInt partPower, bitLenInt start,
+    bitLenInt length, bitCapInt remIndex, cplx<R> scale, hipStream_t stream)
+{
+    hipLaunchKernelGGL((k_gather_part<R>), dim3(gridFor(partPower)), dim3(QA_BLOCK), 0, stream, in,
+        dest, partPower, start, length, remIndex, scale);
+}
+
+template <typename R>
+__global__ void k_allocate_expand(const cplx<R>* in, cplx<R>* out, bitCapInt nMaxQPower,
+    bitLenInt start, bitLenInt length)
+{
+    const bitCapInt lowMask = (ONE_BCI << start) - 1u;
+    const bitCapInt midMask = (ONE_BCI << length) - 1u;
+    const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
+    for (bitCapInt i = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; i < nMaxQPower; i += stride) {
+        const bitCapInt low = i & lowMask;
+        const bitCapInt mid = (i >> start) & midMask;
+        const bitCapInt high = i >> (start + length);
+        out[i] = mid ? cplx<R>(0, 0) : in[low | (high << start)];
+    }
+}
+
+template <typename R>
+void launchAllocateExpand(const cplx<R>* in, cplx<R>* out, bitCapInt nMaxQPower, bitLenInt start,
+    bitLenInt length, hipStream_t stream)
+{
+    hipLaunchKernelGGL((k_allocate_expand<R>), dim3(gridFor(nMaxQPower)), dim3(QA_BLOCK), 0, stream,
+        in, out, nMaxQPower, start, length);
+}
+
+template <typename R> __global__ void k_shuffle_swap(cplx<R>* aHigh, cplx<R>* bLow, bitCapInt half)
+{
+    const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
+    for (bitCapInt i = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; i < half; i += stride) {
+        const cplx<R> t = aHigh[i];
+        aHigh[i] = bLow[i];
+        bLow[i] = t;
+    }
+}
+
+template <typename R>
+void launchShuffleSwap(cplx<R>* aHigh, cplx<R>* bLow, bitCapInt half, hipStream_t stream)
+{
+    hipLaunchKernelGGL(
+        (k_shuffle_swap<R>), dim3(gridFor(half)), dim3(QA_BLOCK), 0, stream, aHigh, bLow, half);
+}
+
+// ---- ALU / permutation ---------------------------------------------------------
+
+__device__ __forceinline__ bitCapInt devModMul(bitCapInt a, bitCapInt b, bitCapInt m)
+{
+    return (bitCapInt)(((__uint128_t)a * b) % m);
+}
+
+__device__ __forceinline__ bitCapInt devModPow(bitCapInt base, bitCapInt e, bitCapInt m)
+{
+    bitCapInt result = 1u % m;
+    base %= m;
+    while (e) {
+        if (e & 1u) result = devModMul(result, base, m);
+        base = devModMul(base, base, m);
+        e >>= 1u;
+    }
+    return result;
+}
+
+__device__ __forceinline__ bitCapInt tableRead(
+    const unsigned char* table, bitCapInt entry, int bytes)
+{
+    bitCapInt v = 0;
+    for (int b = 0; b < bytes; ++b) v |= ((bitCapInt)table[entry * bytes + b]) << (8 * b);
+    return v;
+}
+
+template <typename R> __global__ void k_permute(const cplx<R>* sv, cplx<R>* nsv, PermArgs a)
+{
+    const bitCapInt lenMask = (ONE_BCI << a.length) - 1u;
+    const bitCapInt regMask = lenMask << a.start;
+    const bitCapInt lenPower = ONE_BCI << a.length;
+    const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
+    for (bitCapInt j = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; j < a.maxI; j += stride) {
+        bitCapInt i = expandBits(j, a.qPowers, a.nPowers) | a.controlMask;
+        const bitCapInt reg = (i & regMask) >> a.start;
+        switch ((PermOp)a.op) {
+        case PermOp::INC: {
+            nsv[(i & ~regMask) | (((reg + a.operand) & lenMask) << a.start)] = sv[i];
+            break;
+        }
+        case PermOp::INCDECC: {
+            const bitCapInt out = reg + a.operand;
+            const bitCapInt res = (out < lenPower)
+                ? ((i & ~regMask) | (out << a.start))
+                : ((i & ~regMask) | ((out - lenPower) << a.start) | a.carryMask);
+            nsv[res] = sv[i];
+            break;
+        }
+        case PermOp::INCS: {
+            const bitCapInt out = (reg + a.operand) & lenMask;
+            const bitCapInt signBit = ONE_BCI << (a.length - 1u);
+            const bool ovf = (~(reg ^ a.operand) & (reg ^ out) & signBit) != 0;
+            bitCapInt res = (i & ~regMask) | (out << a.start);
+            if (ovf) res ^= a.carryMask;
+            nsv[res] = sv[i];
+            break;
+        }
+        case PermOp::MUL:
+        case PermOp::DIV: {
+            const bitCapInt carryRegMask = lenMask << a.start2;
+            const bitCapInt out = reg * a.operand;
+            const bitCapInt mapped = (i & ~(regMask | carryRegMask)) |
+                ((out & lenMask) << a.start) | (((out >> a.length) & lenMask) << a.start2);
+            if ((PermOp)a.op == PermOp::MUL) {
+                nsv[mapped] = sv[i];
+            } else {
+                nsv[i] = sv[mapped];
+            }
+            break;
+        }
+        case PermOp::MULMODN:
+        case PermOp::IMULMODN: {
+            const bitCapInt outMask = ((ONE_BCI << a.length2) - 1u) << a.start2;
+            const bitCapInt out = devModMul(reg, a.operand, a.modN);
+            const bitCapInt mapped = (i & ~outMask) | (out << a.start2);
+            if ((PermOp)a.op == PermOp::MULMODN) {
+                nsv[mapped] = sv[i];
+            } else {
+                nsv[i] = sv[mapped];
+            }
+            break;
+        }
+        case PermOp::POWMODN: {
+            const bitCapInt outMask = ((ONE_BCI << a.length2) - 1u) << a.start2;
+            const bitCapInt out = devModPow(a.operand, reg, a.modN);
+            nsv[(i & ~outMask) | (out << a.start2)] = sv[i];
+            break;
+        }
+        case PermOp::HASH: {
+            const bitCapInt val = tableRead(a.table, reg, a.tableBytes) & lenMask;
+            nsv[(i & ~regMask) | (val << a.start)] = sv[i];
+            break;
+        }
+        case PermOp::LDA: {
+            // a.start/a.length = index register; a.start2/length2 = value register
+            const bitCapInt idx = reg;
+            const bitCapInt valMask = ((ONE_BCI << a.length2) - 1u) << a.start2;
+            const bitCapInt val =
+                tableRead(a.table, idx, a.tableBytes) & ((ONE_BCI << a.length2) - 1u);
+            nsv[(i & ~valMask) | (val << a.start2)] = sv[i];
+            break;
+        }
+        case PermOp::ADC:
+        case PermOp::SBC: {
+            const bitCapInt idx = reg;
+            const bitCapInt valPower = ONE_BCI << a.length2;
+            const bitCapInt valMask = (valPower - 1u) << a.start2;
+            const bitCapInt tval = tableRead(a.table, idx, a.tableBytes) & (valPower - 1u);
+            const bitCapInt val = (i & valMask) >> a.start2;
+            bitCapInt out;
+            if ((PermOp)a.op == PermOp::ADC) {
+                out = val + tval + a.extra;
+            } else {
+                out = val + valPower - tval + a.extra;
+            }
+            const bitCapInt wrapped = out & (valPower - 1u);
+            bitCapInt res = (i & ~valMask) | (wrapped << a.start2);
+            if (out >= valPower) res |= a.carryMask;
+            nsv[res] = sv[i];
+            break;
+        }
+        case PermOp::ROL: {
+            const bitLenInt shift = (bitLenInt)a.operand;
+            const bitCapInt nreg = ((reg << shift) | (reg >> (a.length - shift))) & lenMask;
+            nsv[(i & ~regMask) | (nreg << a.start)] = sv[i];
+            break;
+        }
+        }
+    }
+}
+
+template <typename R>
+void launchPermute(const cplx<R>* sv, cplx<R>* nsv, const PermArgs& a, hipStream_t stream)
+{
+    hipLaunchKernelGGL((k_permute<R>), dim3(gridFor(a.maxI)), dim3(QA_BLOCK), 0, stream, sv, nsv, a);
+}
+
+template <typename R>
+__global__ void k_phaseflipless(cplx<R>* sv, bitCapInt maxQPower, bitCapInt greaterPerm,
+    bitLenInt start, bitCapInt regMask, bitCapInt flagMask)
+{
+    const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
+    for (bitCapInt i = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; i < maxQPower; i += stride) {
+        if ((!flagMask || (i & flagMask)) && (((i & regMask) >> start) < greaterPerm)) {
+            sv[i] = cplx<R>(-1, 0) * sv[i];
+        }
+    }
+}
+
+template <typename R>
+void launchPhaseFlipIfLess(cplx<R>* sv, bitCapInt maxQPower, bitCapInt greaterPerm, bitLenInt start,
+    bitCapInt regMask, bitCapInt flagMask, hipStream_t stream)
+{
+    hipLaunchKernelGGL((k_phaseflipless<R>), dim3(gridFor(maxQPower)), dim3(QA_BLOCK), 0, stream, sv,
+        maxQPower, greaterPerm, start, regMask, flagMask);
+}
+
+// ---- sampling / inner product / marginals --------------------------------------
+
+template <typename R>
+__global__ void k_chunk_sums(const cplx<R>* sv, bitCapInt chunkLen, double* sums)
+{
+    // one block per contiguous chunk
+    const bitCapInt lo = (bitCapInt)blockIdx.x * chunkLen;
+    double s = 0;
+    for (bitCapInt i = lo + threadIdx.x; i < lo + chunkLen; i += blockDim.x) {
+        s += (double)norm(sv[i]);
+    }
+    s = waveReduceSum(s);
+    __shared__ double waveSums[QA_BLOCK / 64];
+    const int lane = threadIdx.x & 63;
+    const int wid = threadIdx.x >> 6;
+    if (lane == 0) waveSums[wid] = s;
+    __syncthreads();
+    if (wid == 0) {
+        double t = (lane < QA_BLOCK / 64) ? waveSums[lane] : 0.0;
+        t = waveReduceSum(t);
+        if (lane == 0) sums[blockIdx.x] = t;
+    }
+}
+
+template <typename R>
+void launchChunkSums(
+    const cplx<R>* sv, bitCapInt nChunks, bitCapInt chunkLen, double* sumsDev, hipStream_t stream)
+{
+    hipLaunchKernelGGL(
+        (k_chunk_sums<R>), dim3((uint32_t)nChunks), dim3(QA_BLOCK), 0, stream, sv, chunkLen, sumsDev);
+}
+
+template <typename R>
+__global__ void k_inner(
+    const cplx<R>* a, const cplx<R>* b, bitCapInt maxI, double* partialsRe, double* partialsIm)
+{
+    double re = 0, im = 0;
+    const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
+    for (bitCapInt i = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; i < maxI; i += stride) {
+        const cplx<R> x = a[i];
+        const cplx<R> y = b[i];
+        // conj(b) * a
+        re += (double)(y.re * x.re + y.im * x.im);
+        im += (double)(y.re * x.im - y.im * x.re);
+    }
+    re = waveReduceSum(re);
+    im = waveReduceSum(im);
+    __shared__ double wr[QA_BLOCK / 64];
+    __shared__ double wi2[QA_BLOCK / 64];
+    const int lane = threadIdx.x & 63;
+    const int wid = threadIdx.x >> 6;
+    if (lane == 0) {
+        wr[wid] = re;
+        wi2[wid] = im;
+    }
+    __syncthreads();
+    if (wid == 0) {
+        double tr = (lane < QA_BLOCK / 64) ? wr[lane] : 0.0;
+        double ti = (lane < QA_BLOCK / 64) ? wi2[lane] : 0.0;
+        tr = waveReduceSum(tr);
+        ti = waveReduceSum(ti);
+        if (lane == 0) {
+            partialsRe[blockIdx.x] = tr;
+            partialsIm[blockIdx.x] = ti;
+        }
+    }
+}
+
+template <typename R>
+int launchInner(const cplx<R>* a, const cplx<R>* b, bitCapInt maxI, double* partialsRe,
+    double* partialsIm, hipStream_t stream)
+{
+    const int grid = gridFor(maxI);
+    hipLaunchKernelGGL(
+        (k_inner<R>), dim3(grid), dim3(QA_BLOCK), 0, stream, a, b, maxI, partialsRe, partialsIm);
+    return grid;
+}
+
+template <typename R, bool USE_LDS>
+__global__ void k_part_probs(
+    const cplx<R>* sv, bitCapInt maxQPower, bitLenInt start, bitLenInt length, double* probs)
+{
+    const bitCapInt lenMask = (ONE_BCI << length) - 1u;
+    const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
+    if constexpr (USE_LDS) {
+        extern __shared__ __attribute__((aligned(16))) char smemRaw[];
+        double* hist = reinterpret_cast<double*>(smemRaw);
+        const bitCapInt nBins = ONE_BCI << length;
+        for (bitCapInt p = threadIdx.x; p < nBins; p += blockDim.x) hist[p] = 0;
+        __syncthreads();
+        for (bitCapInt i = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; i < maxQPower;
+             i += stride) {
+            const bitCapInt p = (i >> start) & lenMask;
+            atomicAdd(&hist[p], (double)norm(sv[i]));
+        }
+        __syncthreads();
+        for (bitCapInt p = threadIdx.x; p < nBins; p += blockDim.x) {
+            if (hist[p] != 0.0) atomicAdd(&probs[p], hist[p]);
+        }
+    } else {
+        for (bitCapInt i = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; i < maxQPower;
+             i += stride) {
+            const bitCapInt p = (i >> start) & lenMask;
+            atomicAdd(&probs[p], (double)norm(sv[i]));
+        }
+    }
+}
+
+template <typename R>
+void launchPartProbs(const cplx<R>* sv, bitCapInt maxQPower, bitLenInt start, bitLenInt length,
+    double* probsDev, hipStream_t stream)
+{
+    const int grid = gridFor(maxQPower);
+    const bitCapInt nBins = ONE_BCI << length;
+    if (nBins <= 2048u) {
+        hipLaunchKernelGGL((k_part_probs<R, true>), dim3(grid), dim3(QA_BLOCK),
+            (uint32_t)(nBins * sizeof(double)), stream, sv, maxQPower, start, length, probsDev);
+    } else {
+        hipLaunchKernelGGL((k_part_probs<R, false>), dim3(grid), dim3(QA_BLOCK), 0, stream, sv,
+            maxQPower, start, length, probsDev);
+    }
+}
+
+// ---- explicit instantiations ---------------------------------------------------
+
+#define QA_INSTANTIATE(R)                                                                           \
+    template void launchApply2x2<R>(cplx<R>*, const GateArgs<R>&, hipStream_t);                     \
+    template void launchUniformlyControlled<R>(                                                     \
+        cplx<R>*, bitCapInt, bitCapInt, const bitCapInt*, int, const cplx<R>*, hipStream_t);        \
+    template void launchXMask<R>(cplx<R>*, bitCapInt, bitCapInt, hipStream_t);                      \
+    template void launchParityPhase<R>(cplx<R>*, bitCapInt, bitCapInt, cplx<R>, cplx<R>, hipStream_t); \
+    template int launchReduce<R>(const cplx<R>*, const ReduceArgs&, int, double*, hipStream_t);     \
+    template int launchArgMax<R>(const cplx<R>*, bitCapInt, double*, bitCapInt*, hipStream_t);      \
+    template void launchNormalize<R>(cplx<R>*, bitCapInt, cplx<R>, R, hipStream_t);                 \
+    template void launchApplyM<R>(cplx<R>*, bitCapInt, bitCapInt, bitCapInt, cplx<R>, hipStream_t); \
+    template void launchApplyParity<R>(cplx<R>*, bitCapInt, bitCapInt, bool, cplx<R>, hipStream_t); \
+    template void launchCompose<R>(                                                                 \
+        const cplx<R>*, const cplx<R>*, cplx<R>*, bitCapInt, bitLenInt, bitLenInt, hipStream_t);    \
+    template void launchDisposeSlice<R>(const cplx<R>*, cplx<R>*, bitCapInt, bitLenInt, bitLenInt,  \
+        bitCapInt, cplx<R>, hipStream_t);                                                           \
+    template void launchGatherPart<R>(const cplx<R>*, cplx<R>*, bitCapInt, bitLenInt, bitLenInt,    \
+        bitCapInt, cplx<R>, hipStream_t);                                                           \
+    template void launchAllocateExpand<R>(                                                          \
+        const cplx<R>*, cplx<R>*, bitCapInt, bitLenInt, bitLenInt, hipStream_t);                    \
+    template void launchShuffleSwap<R>(cplx<R>*, cplx<R>*, bitCapInt, hipStream_t);                 \
+    template void launchPermute<R>(const cplx<R>*, cplx<R>*, const PermArgs&, hipStream_t);         \
+    template void launchPhaseFlipIfLess<R>(                                                         \
+        cplx<R>*, bitCapInt, bitCapInt, bitLenInt, bitCapInt, bitCapInt, hipStream_t);              \
+    template void launchChunkSums<R>(const cplx<R>*, bitCapInt, bitCapInt, double*, hipStream_t);   \
+    template int launchInner<R>(                                                                    \
+        const cplx<R>*, const cplx<R>*, bitCapInt, double*, double*, hipStream_t);                  \
+    template void launchPartProbs<R>(                                                               \
+        const cplx<R>*, bitCapInt, bitLenInt, bitLenInt, double*, hipStream_t);
+
+QA_INSTANTIATE(float)
+QA_INSTANTIATE(double)
+
+} // namespace qrack_amd

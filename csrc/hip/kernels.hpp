@@ -1,0 +1,173 @@
+// qrack_amd — HIP kernel launch API (host-visible declarations).
+//
+// Capability parity target: the reference GPU kernel inventory
+// (/root/reference/src/common/qengine.cl 74 kernels + qheader_alu.cl;
+// SURVEY.md §2.2). Fresh CDNA4 design: wave64 blocks of 256, grid-stride
+// loops capped near 8 blocks/CU (256 CUs), float4/double2-vectorized
+// amplitude streaming, single-pass two-stage reductions (wave shuffle +
+// LDS + per-block partials), and one uniform-opcode permutation kernel for
+// the whole ALU family instead of 26 near-identical kernels.
+#pragma once
+
+#include "../common/types.hpp"
+
+#include <hip/hip_runtime.h>
+
+namespace qrack_amd {
+
+constexpr int QA_MAX_SKIP_POWERS = 16;
+
+// by-value argument block for gate kernels
+template <typename R> struct GateArgs {
+    cplx<R> m[4];
+    bitCapInt offset1;
+    bitCapInt offset2;
+    bitCapInt qPowers[QA_MAX_SKIP_POWERS]; // sorted ascending
+    int nPowers;
+    bitCapInt maxI; // iterations (pairs)
+};
+
+enum class PermOp : int {
+    INC = 0,
+    INCDECC,
+    INCS,
+    MUL,
+    DIV,
+    MULMODN,
+    IMULMODN,
+    POWMODN,
+    HASH,
+    LDA,
+    ADC,
+    SBC,
+    ROL,
+};
+
+struct PermArgs {
+    int op;
+    bitCapInt maxI;                        // iterated indices (sub-space size)
+    bitCapInt qPowers[QA_MAX_SKIP_POWERS]; // skip powers for the iteration
+    int nPowers;
+    bitCapInt controlMask; // OR into every iterated index
+    bitLenInt start;       // in/out register start
+    bitLenInt length;      // register length
+    bitCapInt operand;     // toAdd / toMul / base
+    bitCapInt modN;
+    bitCapInt carryMask;    // carry / overflow qubit power
+    bitLenInt start2;       // out register start (modN ops) / value start (LDA)
+    bitLenInt length2;      // out register length
+    bitCapInt extra;        // carry-in addend etc.
+    const unsigned char* table; // device pointer for HASH/LDA/ADC/SBC
+    int tableBytes;             // bytes per table entry
+};
+
+enum class ReduceOp : int {
+    NORM_ALL = 0,   // sum |amp|^2
+    PROB_BITSET,    // sum |amp|^2 where (i & mask) == mask (single-bit use)
+    PROB_MASK,      // sum |amp|^2 where (i & mask) == perm
+    PROB_PARITY,    // sum |amp|^2 where parity(i & mask)
+    EXP_PERM,       // sum value(i) * |amp|^2 (factorized bit values)
+    EXP_PERM_SQ,    // sum value(i)^2 * |amp|^2
+    NORM_FLOOR,     // sum |amp|^2 with amplitude floor (UpdateRunningNorm)
+};
+
+struct ReduceArgs {
+    bitCapInt maxI;
+    bitCapInt mask;
+    bitCapInt perm;
+    double offset;       // EXP_PERM value offset
+    double normThresh;   // NORM_FLOOR
+    const bitLenInt* bits;   // device ptrs for EXP_PERM
+    const bitCapInt* perms;
+    int nBits;
+};
+
+// ---- launches (all asynchronous on `stream`) -------------------------------
+
+template <typename R>
+void launchApply2x2(cplx<R>* sv, const GateArgs<R>& a, hipStream_t stream);
+
+template <typename R>
+void launchUniformlyControlled(cplx<R>* sv, bitCapInt maxI, bitCapInt targetPower,
+    const bitCapInt* ctrlPowersDev, int nCtrls, const cplx<R>* mtrxsDev, hipStream_t stream);
+
+template <typename R>
+void launchXMask(cplx<R>* sv, bitCapInt maxQPower, bitCapInt mask, hipStream_t stream);
+
+template <typename R>
+void launchParityPhase(
+    cplx<R>* sv, bitCapInt maxQPower, bitCapInt mask, cplx<R> even, cplx<R> odd, hipStream_t stream);
+
+// two-stage reduction: stage 1 fills partials[gridSize]; returns grid size used
+template <typename R>
+int launchReduce(const cplx<R>* sv, const ReduceArgs& a, int op, double* partialsDev,
+    hipStream_t stream);
+
+// argmax of |amp|^2: fills (value,index) pairs per block
+template <typename R>
+int launchArgMax(const cplx<R>* sv, bitCapInt maxI, double* valsDev, bitCapInt* idxDev,
+    hipStream_t stream);
+
+template <typename R>
+void launchNormalize(
+    cplx<R>* sv, bitCapInt maxQPower, cplx<R> factor, R normThresh, hipStream_t stream);
+
+template <typename R>
+void launchApplyM(cplx<R>* sv, bitCapInt maxQPower, bitCapInt mask, bitCapInt result, cplx<R> nrm,
+    hipStream_t stream);
+
+template <typename R>
+void launchApplyParity(cplx<R>* sv, bitCapInt maxQPower, bitCapInt mask, bool odd, cplx<R> nrm,
+    hipStream_t stream);
+
+template <typename R>
+void launchCompose(const cplx<R>* a, const cplx<R>* b, cplx<R>* out, bitCapInt nMaxQPower,
+    bitLenInt start, bitLenInt oQubits, hipStream_t stream);
+
+// out[r] = in[low | (slicePerm << start) | high] * scale  (Dispose/Decompose)
+template <typename R>
+void launchDisposeSlice(const cplx<R>* in, cplx<R>* out, bitCapInt remPower, bitLenInt start,
+    bitLenInt length, bitCapInt slicePerm, cplx<R> scale, hipStream_t stream);
+
+// dest[p] = in[low(rem) | (p << start) | high(rem)] * scale  (Decompose part)
+template <typename R>
+void launchGatherPart(const cplx<R>* in, cplx<R>* dest, bitCapInt partPower, bitLenInt start,
+    bitLenInt length, bitCapInt remIndex, cplx<R> scale, hipStream_t stream);
+
+// out[i] = (mid(i)==0) ? in[collapse(i)] : 0   (Allocate)
+template <typename R>
+void launchAllocateExpand(const cplx<R>* in, cplx<R>* out, bitCapInt nMaxQPower, bitLenInt start,
+    bitLenInt length, hipStream_t stream);
+
+// swap upper half of a with lower half of b
+template <typename R>
+void launchShuffleSwap(cplx<R>* aHigh, cplx<R>* bLow, bitCapInt half, hipStream_t stream);
+
+template <typename R>
+void launchPermute(const cplx<R>* sv, cplx<R>* nsv, const PermArgs& a, hipStream_t stream);
+
+template <typename R>
+void launchPhaseFlipIfLess(cplx<R>* sv, bitCapInt maxQPower, bitCapInt greaterPerm, bitLenInt start,
+    bitCapInt regMask, bitCapInt flagMask, hipStream_t stream);
+
+int reduceGridSize(bitCapInt n);
+
+// contiguous per-chunk |amp|^2 sums (inverse-CDF sampling support):
+// sums[c] = sum over [c*chunkLen, (c+1)*chunkLen)
+template <typename R>
+void launchChunkSums(
+    const cplx<R>* sv, bitCapInt nChunks, bitCapInt chunkLen, double* sumsDev, hipStream_t stream);
+
+// complex inner product <b|a>: per-block partial (re, im) pairs
+template <typename R>
+int launchInner(const cplx<R>* a, const cplx<R>* b, bitCapInt maxI, double* partialsRe,
+    double* partialsIm, hipStream_t stream);
+
+// marginal probabilities of the [start, start+length) register:
+// probs[p] += |amp|^2 over all amplitudes with register == p  (probsDev zeroed
+// by caller). Single pass, LDS histogram when 2^length <= 2048.
+template <typename R>
+void launchPartProbs(const cplx<R>* sv, bitCapInt maxQPower, bitLenInt start, bitLenInt length,
+    double* probsDev, hipStream_t stream);
+
+} // namespace qrack_amd

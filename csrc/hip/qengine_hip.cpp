@@ -1,0 +1,1307 @@
+// qrack_amd — HIP engine host implementation (see qengine_hip.hpp).
+#include "qengine_hip.hpp"
+
+#include "../qfactory.hpp"
+
+#include <algorithm>
+#include <cstring>
+
+namespace qrack_amd {
+
+// ---- device tracker ---------------------------------------------------------
+
+HipDeviceTracker& HipDeviceTracker::instance()
+{
+    static HipDeviceTracker t;
+    return t;
+}
+
+HipDeviceTracker::HipDeviceTracker()
+{
+    if (hipGetDeviceCount(&count_) != hipSuccess) count_ = 0;
+    totals_.resize(count_);
+    caps_.resize(count_);
+    active_ = std::vector<std::atomic<size_t>>(count_);
+    size_t capMb = 0;
+    if (const char* env = std::getenv("QRACK_MAX_ALLOC_MB")) capMb = (size_t)std::atoll(env);
+    for (int d = 0; d < count_; ++d) {
+        hipDeviceProp_t props;
+        if (hipGetDeviceProperties(&props, d) == hipSuccess) {
+            totals_[d] = props.totalGlobalMem;
+        }
+        active_[d] = 0;
+        // default cap: 15/16 of VRAM (state + equal-size scratch both fit at
+        // the max paged state of 2^33 fp32 amps; reference used OclMemDenom=3)
+        caps_[d] = capMb ? capMb * (1ull << 20) : (totals_[d] / 16) * 15;
+    }
+}
+
+int HipDeviceTracker::deviceCount() { return count_; }
+size_t HipDeviceTracker::totalMem(int dev) { return totals_.at(dev); }
+size_t HipDeviceTracker::activeAlloc(int dev) { return active_.at(dev).load(); }
+int HipDeviceTracker::defaultDevice()
+{
+    if (const char* env = std::getenv("QRACK_HIP_DEFAULT_DEVICE")) return std::atoi(env);
+    return 0;
+}
+
+void HipDeviceTracker::add(int dev, size_t bytes)
+{
+    const size_t now = active_.at(dev).fetch_add(bytes) + bytes;
+    if (now > caps_.at(dev)) {
+        active_.at(dev).fetch_sub(bytes);
+        throw std::bad_alloc();
+    }
+}
+
+void HipDeviceTracker::sub(int dev, size_t bytes) { active_.at(dev).fetch_sub(bytes); }
+
+// ---- ctor / alloc -----------------------------------------------------------
+
+template <typename R>
+QEngineHIP<R>::QEngineHIP(bitLenInt qBitCount, bitCapInt initState, RngPtr rgp, bool doNorm,
+    R normThresh, int64_t devId, cplx<R> initPhase)
+    : QEngine<R>(qBitCount, rgp, doNorm, normThresh)
+{
+    auto& tracker = HipDeviceTracker::instance();
+    if (tracker.deviceCount() < 1) throw QrackError("no HIP device visible");
+    deviceId = (devId < 0) ? tracker.defaultDevice() : (int)devId;
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    QA_HIP_CHECK(hipStreamCreateWithFlags(&stream, hipStreamNonBlocking));
+    dState = allocDev(maxQPower);
+    const int maxGrid = 4096;
+    QA_HIP_CHECK(hipMalloc(&dPartials, maxGrid * 2 * sizeof(double)));
+    QA_HIP_CHECK(hipMalloc(&dIdx, maxGrid * sizeof(bitCapInt)));
+    hPartials.resize(maxGrid * 2);
+    SetPermutation(initState, initPhase);
+}
+
+template <typename R> QEngineHIP<R>::~QEngineHIP()
+{
+    hipStreamSynchronize(stream);
+    if (dState) freeDev(dState, maxQPower);
+    releaseScratch();
+    if (dPartials) hipFree(dPartials);
+    if (dIdx) hipFree(dIdx);
+    hipStreamDestroy(stream);
+}
+
+template <typename R> cplx<R>* QEngineHIP<R>::allocDev(bitCapInt nAmps)
+{
+    auto& tracker = HipDeviceTracker::instance();
+    const size_t bytes = sizeof(cplx<R>) * nAmps;
+    tracker.add(deviceId, bytes);
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    cplx<R>* p = nullptr;
+    const hipError_t err = hipMalloc(&p, bytes);
+    if (err != hipSuccess) {
+        tracker.sub(deviceId, bytes);
+        if (err == hipErrorOutOfMemory) throw std::bad_alloc();
+        QA_HIP_CHECK(err);
+    }
+    return p;
+}
+
+template <typename R> void QEngineHIP<R>::freeDev(cplx<R>* p, bitCapInt nAmps)
+{
+    if (!p) return;
+    hipFree(p);
+    HipDeviceTracker::instance().sub(deviceId, sizeof(cplx<R>) * nAmps);
+}
+
+template <typename R> void QEngineHIP<R>::ensureScratch()
+{
+    if (!dScratch) dScratch = allocDev(maxQPower);
+}
+
+template <typename R> void QEngineHIP<R>::releaseScratch()
+{
+    if (dScratch) {
+        freeDev(dScratch, maxQPower);
+        dScratch = nullptr;
+    }
+}
+
+template <typename R> void QEngineHIP<R>::swapScratch() { std::swap(dState, dScratch); }
+
+template <typename R> void QEngineHIP<R>::resizeState(bitCapInt nAmps, cplx<R>* newBuf)
+{
+    Finish();
+    releaseScratch();
+    freeDev(dState, maxQPower);
+    dState = newBuf;
+}
+
+template <typename R> void QEngineHIP<R>::SetDevice(int64_t devId)
+{
+    if (devId == deviceId || devId < 0) return;
+    // migrate the state buffer to another GPU
+    Finish();
+    std::vector<cplx<R>> host(maxQPower);
+    GetQuantumState(host.data());
+    releaseScratch();
+    freeDev(dState, maxQPower);
+    hipStreamDestroy(stream);
+    hipFree(dPartials);
+    hipFree(dIdx);
+    deviceId = (int)devId;
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    QA_HIP_CHECK(hipStreamCreateWithFlags(&stream, hipStreamNonBlocking));
+    dState = allocDev(maxQPower);
+    QA_HIP_CHECK(hipMalloc(&dPartials, 4096 * 2 * sizeof(double)));
+    QA_HIP_CHECK(hipMalloc(&dIdx, 4096 * sizeof(bitCapInt)));
+    SetQuantumState(host.data());
+}
+
+// ---- state access -----------------------------------------------------------
+
+template <typename R> void QEngineHIP<R>::SetQuantumState(const cplx<R>* inputState)
+{
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    QA_HIP_CHECK(hipMemcpyAsync(
+        dState, inputState, sizeof(cplx<R>) * maxQPower, hipMemcpyHostToDevice, stream));
+    Finish();
+    runningNorm = (R)-1;
+}
+
+template <typename R> void QEngineHIP<R>::GetQuantumState(cplx<R>* outputState)
+{
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    QA_HIP_CHECK(hipMemcpyAsync(
+        outputState, dState, sizeof(cplx<R>) * maxQPower, hipMemcpyDeviceToHost, stream));
+    Finish();
+}
+
+template <typename R> cplx<R> QEngineHIP<R>::GetAmplitude(bitCapInt perm)
+{
+    cplx<R> amp;
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    QA_HIP_CHECK(
+        hipMemcpyAsync(&amp, dState + perm, sizeof(cplx<R>), hipMemcpyDeviceToHost, stream));
+    Finish();
+    return amp;
+}
+
+template <typename R> void QEngineHIP<R>::SetAmplitude(bitCapInt perm, cplx<R> amp)
+{
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    QA_HIP_CHECK(
+        hipMemcpyAsync(dState + perm, &amp, sizeof(cplx<R>), hipMemcpyHostToDevice, stream));
+    Finish();
+    runningNorm = (R)-1;
+}
+
+template <typename R> void QEngineHIP<R>::SetPermutation(bitCapInt perm, cplx<R> phase)
+{
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    QA_HIP_CHECK(hipMemsetAsync(dState, 0, sizeof(cplx<R>) * maxQPower, stream));
+    if (norm(phase) <= 0) phase = cplx<R>(1, 0);
+    SetAmplitude(perm, phase);
+    runningNorm = (R)1;
+}
+
+template <typename R> void QEngineHIP<R>::ZeroAmplitudes()
+{
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    QA_HIP_CHECK(hipMemsetAsync(dState, 0, sizeof(cplx<R>) * maxQPower, stream));
+    runningNorm = 0;
+}
+
+template <typename R> void QEngineHIP<R>::CopyStateVec(QEnginePtr<R> src)
+{
+    QEngineHIP<R>* o = dynamic_cast<QEngineHIP<R>*>(src.get());
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    if (o && o->deviceId == deviceId) {
+        o->Finish();
+        QA_HIP_CHECK(hipMemcpyAsync(
+            dState, o->dState, sizeof(cplx<R>) * maxQPower, hipMemcpyDeviceToDevice, stream));
+        Finish();
+    } else {
+        std::vector<cplx<R>> host(maxQPower);
+        src->GetQuantumState(host.data());
+        SetQuantumState(host.data());
+    }
+    runningNorm = (R)-1;
+}
+
+template <typename R> bool QEngineHIP<R>::IsZeroAmplitude()
+{
+    ReduceArgs a{};
+    a.maxI = maxQPower;
+    return reduceSum((int)ReduceOp::NORM_ALL, a) <= 0.0;
+}
+
+template <typename R>
+void QEngineHIP<R>::GetAmplitudePage(cplx<R>* pagePtr, bitCapInt offset, bitCapInt length)
+{
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    QA_HIP_CHECK(hipMemcpyAsync(
+        pagePtr, dState + offset, sizeof(cplx<R>) * length, hipMemcpyDeviceToHost, stream));
+    Finish();
+}
+
+template <typename R>
+void QEngineHIP<R>::SetAmplitudePage(const cplx<R>* pagePtr, bitCapInt offset, bitCapInt length)
+{
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    QA_HIP_CHECK(hipMemcpyAsync(
+        dState + offset, pagePtr, sizeof(cplx<R>) * length, hipMemcpyHostToDevice, stream));
+    Finish();
+    runningNorm = (R)-1;
+}
+
+template <typename R>
+void QEngineHIP<R>::SetAmplitudePage(
+    QEnginePtr<R> pageEnginePtr, bitCapInt srcOffset, bitCapInt dstOffset, bitCapInt length)
+{
+    QEngineHIP<R>* o = dynamic_cast<QEngineHIP<R>*>(pageEnginePtr.get());
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    if (o) {
+        o->Finish();
+        if (o->deviceId == deviceId) {
+            QA_HIP_CHECK(hipMemcpyAsync(dState + dstOffset, o->dState + srcOffset,
+                sizeof(cplx<R>) * length, hipMemcpyDeviceToDevice, stream));
+        } else {
+            QA_HIP_CHECK(hipMemcpyPeerAsync(dState + dstOffset, deviceId, o->dState + srcOffset,
+                o->deviceId, sizeof(cplx<R>) * length, stream));
+        }
+        Finish();
+    } else {
+        std::vector<cplx<R>> host(length);
+        pageEnginePtr->GetAmplitudePage(host.data(), srcOffset, length);
+        SetAmplitudePage(host.data(), dstOffset, length);
+    }
+    runningNorm = (R)-1;
+}
+
+template <typename R> void QEngineHIP<R>::ShuffleBuffers(QEnginePtr<R> engine)
+{
+    QEngineHIP<R>* o = dynamic_cast<QEngineHIP<R>*>(engine.get());
+    const bitCapInt half = maxQPower >> 1u;
+    if (o && o->deviceId == deviceId) {
+        o->Finish();
+        QA_HIP_CHECK(hipSetDevice(deviceId));
+        launchShuffleSwap<R>(dState + half, o->dState, half, stream);
+        Finish();
+    } else if (o) {
+        // cross-device in-process: stage through a temp on this device
+        o->Finish();
+        QA_HIP_CHECK(hipSetDevice(deviceId));
+        cplx<R>* tmp = allocDev(half);
+        QA_HIP_CHECK(hipMemcpyAsync(
+            tmp, dState + half, sizeof(cplx<R>) * half, hipMemcpyDeviceToDevice, stream));
+        QA_HIP_CHECK(hipMemcpyPeerAsync(
+            dState + half, deviceId, o->dState, o->deviceId, sizeof(cplx<R>) * half, stream));
+        QA_HIP_CHECK(hipMemcpyPeerAsync(
+            o->dState, o->deviceId, tmp, deviceId, sizeof(cplx<R>) * half, stream));
+        Finish();
+        freeDev(tmp, half);
+    } else {
+        // CPU peer: stage through host
+        std::vector<cplx<R>> mine(half), theirs(half);
+        GetAmplitudePage(mine.data(), half, half);
+        engine->GetAmplitudePage(theirs.data(), 0, half);
+        SetAmplitudePage(theirs.data(), half, half);
+        engine->SetAmplitudePage(mine.data(), 0, half);
+    }
+    runningNorm = (R)-1;
+    if (o) o->runningNorm = (R)-1;
+}
+
+// ---- gates ------------------------------------------------------------------
+
+template <typename R>
+GateArgs<R> QEngineHIP<R>::makeGateArgs(bitCapInt offset1, bitCapInt offset2, const cplx<R>* mtrx,
+    const std::vector<bitCapInt>& qPowersSorted)
+{
+    if ((int)qPowersSorted.size() > QA_MAX_SKIP_POWERS) {
+        throw QrackError("too many control qubits for one gate (max 15 controls)");
+    }
+    GateArgs<R> a{};
+    for (int i = 0; i < 4; ++i) a.m[i] = mtrx[i];
+    a.offset1 = offset1;
+    a.offset2 = offset2;
+    a.nPowers = (int)qPowersSorted.size();
+    for (int i = 0; i < a.nPowers; ++i) a.qPowers[i] = qPowersSorted[i];
+    a.maxI = maxQPower >> (bitLenInt)qPowersSorted.size();
+    return a;
+}
+
+template <typename R>
+void QEngineHIP<R>::Apply2x2(bitCapInt offset1, bitCapInt offset2, const cplx<R>* mtrx,
+    const std::vector<bitCapInt>& qPowersSorted)
+{
+    GateArgs<R> a = makeGateArgs(offset1, offset2, mtrx, qPowersSorted);
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    launchApply2x2<R>(dState, a, stream);
+}
+
+template <typename R> void QEngineHIP<R>::XMask(bitCapInt mask)
+{
+    if (!mask) return;
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    launchXMask<R>(dState, maxQPower, mask, stream);
+}
+
+template <typename R> void QEngineHIP<R>::ZMask(bitCapInt mask)
+{
+    if (!mask) return;
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    launchParityPhase<R>(dState, maxQPower, mask, cplx<R>(1, 0), cplx<R>(-1, 0), stream);
+}
+
+template <typename R> void QEngineHIP<R>::PhaseParity(R radians, bitCapInt mask)
+{
+    if (!mask) return;
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    launchParityPhase<R>(
+        dState, maxQPower, mask, polar<R>(1, -radians / 2), polar<R>(1, radians / 2), stream);
+}
+
+template <typename R>
+void QEngineHIP<R>::UniformlyControlledSingleBit(
+    const std::vector<bitLenInt>& controls, bitLenInt target, const cplx<R>* mtrxs)
+{
+    if (controls.empty()) {
+        this->Mtrx(mtrxs, target);
+        return;
+    }
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    const bitCapInt nMtrx = pow2((bitLenInt)controls.size());
+    std::vector<bitCapInt> ctrlPowers(controls.size());
+    for (size_t i = 0; i < controls.size(); ++i) ctrlPowers[i] = pow2(controls[i]);
+    bitCapInt* dPowers = nullptr;
+    cplx<R>* dMtrxs = nullptr;
+    QA_HIP_CHECK(hipMallocAsync(&dPowers, sizeof(bitCapInt) * controls.size(), stream));
+    QA_HIP_CHECK(hipMallocAsync(&dMtrxs, sizeof(cplx<R>) * 4 * nMtrx, stream));
+    QA_HIP_CHECK(hipMemcpyAsync(dPowers, ctrlPowers.data(), sizeof(bitCapInt) * controls.size(),
+        hipMemcpyHostToDevice, stream));
+    QA_HIP_CHECK(
+        hipMemcpyAsync(dMtrxs, mtrxs, sizeof(cplx<R>) * 4 * nMtrx, hipMemcpyHostToDevice, stream));
+    launchUniformlyControlled<R>(dState, maxQPower >> 1u, pow2(target), dPowers,
+        (int)controls.size(), dMtrxs, stream);
+    QA_HIP_CHECK(hipFreeAsync(dPowers, stream));
+    QA_HIP_CHECK(hipFreeAsync(dMtrxs, stream));
+}
+
+// ---- reductions -------------------------------------------------------------
+
+template <typename R> double QEngineHIP<R>::reduceSum(int op, const ReduceArgs& a)
+{
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    const int grid = launchReduce<R>(dState, a, op, dPartials, stream);
+    QA_HIP_CHECK(hipMemcpyAsync(
+        hPartials.data(), dPartials, grid * sizeof(double), hipMemcpyDeviceToHost, stream));
+    Finish();
+    double s = 0;
+    for (int i = 0; i < grid; ++i) s += hPartials[i];
+    return s;
+}
+
+template <typename R> std::pair<double, bitCapInt> QEngineHIP<R>::argMax()
+{
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    const int grid = launchArgMax<R>(dState, maxQPower, dPartials, dIdx, stream);
+    std::vector<bitCapInt> hIdx(grid);
+    QA_HIP_CHECK(hipMemcpyAsync(
+        hPartials.data(), dPartials, grid * sizeof(double), hipMemcpyDeviceToHost, stream));
+    QA_HIP_CHECK(
+        hipMemcpyAsync(hIdx.data(), dIdx, grid * sizeof(bitCapInt), hipMemcpyDeviceToHost, stream));
+    Finish();
+    int best = 0;
+    for (int i = 1; i < grid; ++i) {
+        if (hPartials[i] > hPartials[best]) best = i;
+    }
+    return { hPartials[best], hIdx[best] };
+}
+
+template <typename R> R QEngineHIP<R>::Prob(bitLenInt q)
+{
+    ReduceArgs a{};
+    a.maxI = maxQPower;
+    a.mask = pow2(q);
+    const double p = reduceSum((int)ReduceOp::PROB_BITSET, a);
+    return (R)std::min(1.0, std::max(0.0, p));
+}
+
+template <typename R> R QEngineHIP<R>::ProbMask(bitCapInt mask, bitCapInt permutation)
+{
+    ReduceArgs a{};
+    a.maxI = maxQPower;
+    a.mask = mask;
+    a.perm = permutation;
+    const double p = reduceSum((int)ReduceOp::PROB_MASK, a);
+    return (R)std::min(1.0, std::max(0.0, p));
+}
+
+template <typename R> R QEngineHIP<R>::ProbReg(bitLenInt start, bitLenInt length, bitCapInt permutation)
+{
+    return ProbMask(pow2Mask(length) << start, permutation << start);
+}
+
+template <typename R> R QEngineHIP<R>::ProbParity(bitCapInt mask)
+{
+    if (!mask) return 0;
+    ReduceArgs a{};
+    a.maxI = maxQPower;
+    a.mask = mask;
+    const double p = reduceSum((int)ReduceOp::PROB_PARITY, a);
+    return (R)std::min(1.0, std::max(0.0, p));
+}
+
+template <typename R> bool QEngineHIP<R>::ForceMParity(bitCapInt mask, bool result, bool doForce)
+{
+    if (!mask) return false;
+    const R oddProb = ProbParity(mask);
+    if (!doForce) result = (this->Rand() < (double)oddProb);
+    const R prob = result ? oddProb : ((R)1 - oddProb);
+    if (prob <= 0) throw QrackError("ForceMParity: impossible outcome");
+    const R nrm = (R)1 / std::sqrt(prob);
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    launchApplyParity<R>(dState, maxQPower, mask, result, cplx<R>(nrm, 0), stream);
+    runningNorm = (R)1;
+    return result;
+}
+
+template <typename R> void QEngineHIP<R>::ApplyM(bitCapInt regMask, bitCapInt result, cplx<R> nrm)
+{
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    launchApplyM<R>(dState, maxQPower, regMask, result, nrm, stream);
+    runningNorm = (R)1;
+}
+
+template <typename R>
+double QEngineHIP<R>::ExpectationBitsFactorized(
+    const std::vector<bitLenInt>& bits, const std::vector<bitCapInt>& perms, bitCapInt offset)
+{
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    bitLenInt* dBits = nullptr;
+    bitCapInt* dPerms = nullptr;
+    QA_HIP_CHECK(hipMallocAsync(&dBits, sizeof(bitLenInt) * bits.size(), stream));
+    QA_HIP_CHECK(hipMallocAsync(&dPerms, sizeof(bitCapInt) * perms.size(), stream));
+    QA_HIP_CHECK(hipMemcpyAsync(
+        dBits, bits.data(), sizeof(bitLenInt) * bits.size(), hipMemcpyHostToDevice, stream));
+    QA_HIP_CHECK(hipMemcpyAsync(
+        dPerms, perms.data(), sizeof(bitCapInt) * perms.size(), hipMemcpyHostToDevice, stream));
+    ReduceArgs a{};
+    a.maxI = maxQPower;
+    a.offset = (double)offset;
+    a.bits = dBits;
+    a.perms = dPerms;
+    a.nBits = (int)bits.size();
+    const double e = reduceSum((int)ReduceOp::EXP_PERM, a);
+    QA_HIP_CHECK(hipFreeAsync(dBits, stream));
+    QA_HIP_CHECK(hipFreeAsync(dPerms, stream));
+    return e;
+}
+
+template <typename R>
+double QEngineHIP<R>::VarianceBitsAll(const std::vector<bitLenInt>& bits, bitCapInt offset)
+{
+    const double mean = this->ExpectationBitsAll(bits, offset);
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    std::vector<bitCapInt> perms;
+    for (size_t b = 0; b < bits.size(); ++b) perms.push_back(pow2((bitLenInt)b));
+    bitLenInt* dBits = nullptr;
+    bitCapInt* dPerms = nullptr;
+    QA_HIP_CHECK(hipMallocAsync(&dBits, sizeof(bitLenInt) * bits.size(), stream));
+    QA_HIP_CHECK(hipMallocAsync(&dPerms, sizeof(bitCapInt) * perms.size(), stream));
+    QA_HIP_CHECK(hipMemcpyAsync(
+        dBits, bits.data(), sizeof(bitLenInt) * bits.size(), hipMemcpyHostToDevice, stream));
+    QA_HIP_CHECK(hipMemcpyAsync(
+        dPerms, perms.data(), sizeof(bitCapInt) * perms.size(), hipMemcpyHostToDevice, stream));
+    ReduceArgs a{};
+    a.maxI = maxQPower;
+    a.offset = (double)offset;
+    a.bits = dBits;
+    a.perms = dPerms;
+    a.nBits = (int)bits.size();
+    const double e2 = reduceSum((int)ReduceOp::EXP_PERM_SQ, a);
+    QA_HIP_CHECK(hipFreeAsync(dBits, stream));
+    QA_HIP_CHECK(hipFreeAsync(dPerms, stream));
+    return e2 - mean * mean;
+}
+
+// ---- sampling ---------------------------------------------------------------
+
+template <typename R> std::vector<double> QEngineHIP<R>::chunkSums(bitCapInt& chunkLenOut)
+{
+    const bitCapInt nChunks = std::min<bitCapInt>(maxQPower, 2048u);
+    const bitCapInt chunkLen = maxQPower / nChunks;
+    chunkLenOut = chunkLen;
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    launchChunkSums<R>(dState, nChunks, chunkLen, dPartials, stream);
+    std::vector<double> sums(nChunks);
+    QA_HIP_CHECK(hipMemcpyAsync(
+        sums.data(), dPartials, nChunks * sizeof(double), hipMemcpyDeviceToHost, stream));
+    Finish();
+    return sums;
+}
+
+template <typename R>
+bitCapInt QEngineHIP<R>::sampleOnce(const std::vector<double>& sums, bitCapInt chunkLen, double r,
+    std::vector<cplx<R>>& hostChunk, bitCapInt& cachedChunk)
+{
+    // r in [0, total)
+    bitCapInt c = 0;
+    while (c + 1 < (bitCapInt)sums.size() && r > sums[c]) {
+        r -= sums[c];
+        ++c;
+    }
+    if (cachedChunk != c) {
+        GetAmplitudePage(hostChunk.data(), c * chunkLen, chunkLen);
+        cachedChunk = c;
+    }
+    bitCapInt i = 0;
+    for (; i < chunkLen - 1; ++i) {
+        r -= (double)norm(hostChunk[i]);
+        if (r <= 0) break;
+    }
+    return c * chunkLen + i;
+}
+
+template <typename R> bitCapInt QEngineHIP<R>::MAll()
+{
+    bitCapInt chunkLen = 0;
+    const std::vector<double> sums = chunkSums(chunkLen);
+    double total = 0;
+    for (double s : sums) total += s;
+    std::vector<cplx<R>> hostChunk(chunkLen);
+    bitCapInt cached = (bitCapInt)-1;
+    const bitCapInt result = sampleOnce(sums, chunkLen, this->Rand() * total, hostChunk, cached);
+    SetPermutation(result);
+    return result;
+}
+
+template <typename R>
+std::map<bitCapInt, int> QEngineHIP<R>::MultiShotMeasureMask(
+    const std::vector<bitCapInt>& qPowers, unsigned shots)
+{
+    if (!shots) return {};
+    bitCapInt chunkLen = 0;
+    const std::vector<double> sums = chunkSums(chunkLen);
+    double total = 0;
+    for (double s : sums) total += s;
+    std::vector<double> rs(shots);
+    for (unsigned s = 0; s < shots; ++s) rs[s] = this->Rand() * total;
+    std::sort(rs.begin(), rs.end());
+    std::vector<cplx<R>> hostChunk(chunkLen);
+    bitCapInt cached = (bitCapInt)-1;
+    std::map<bitCapInt, int> results;
+    for (unsigned s = 0; s < shots; ++s) {
+        const bitCapInt i = sampleOnce(sums, chunkLen, rs[s], hostChunk, cached);
+        bitCapInt val = 0;
+        for (size_t b = 0; b < qPowers.size(); ++b) {
+            if (i & qPowers[b]) val |= (ONE_BCI << b);
+        }
+        results[val]++;
+    }
+    return results;
+}
+
+// ---- norm -------------------------------------------------------------------
+
+template <typename R> void QEngineHIP<R>::UpdateRunningNorm(R norm_thresh)
+{
+    if (norm_thresh < 0) norm_thresh = amplitudeFloor;
+    ReduceArgs a{};
+    a.maxI = maxQPower;
+    a.normThresh = (double)norm_thresh;
+    runningNorm = (R)reduceSum((int)ReduceOp::NORM_FLOOR, a);
+}
+
+template <typename R> void QEngineHIP<R>::NormalizeState(R nrm, R norm_thresh, R phaseArg)
+{
+    if (nrm < 0) {
+        if (runningNorm < 0) UpdateRunningNorm(norm_thresh);
+        nrm = runningNorm;
+    }
+    if (nrm <= 0) return;
+    if (norm_thresh < 0) norm_thresh = amplitudeFloor;
+    const cplx<R> factor = polar<R>((R)(1.0 / std::sqrt((double)nrm)), phaseArg);
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    launchNormalize<R>(dState, maxQPower, factor, norm_thresh * nrm, stream);
+    runningNorm = (R)1;
+}
+
+template <typename R> double QEngineHIP<R>::SumSqrDiff(QInterfacePtr<R> other)
+{
+    if (other->GetQubitCount() != qubitCount) return 2.0;
+    QEngineHIP<R>* o = dynamic_cast<QEngineHIP<R>*>(other.get());
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    cplx<R>* otherBuf = nullptr;
+    cplx<R>* tmp = nullptr;
+    if (o && o->deviceId == deviceId) {
+        o->Finish();
+        otherBuf = o->dState;
+    } else {
+        std::vector<cplx<R>> host(maxQPower);
+        other->GetQuantumState(host.data());
+        tmp = allocDev(maxQPower);
+        QA_HIP_CHECK(hipMemcpyAsync(
+            tmp, host.data(), sizeof(cplx<R>) * maxQPower, hipMemcpyHostToDevice, stream));
+        otherBuf = tmp;
+    }
+    const int grid = launchInner<R>(dState, otherBuf, maxQPower, dPartials, dPartials + 4096, stream);
+    QA_HIP_CHECK(hipMemcpyAsync(
+        hPartials.data(), dPartials, grid * sizeof(double), hipMemcpyDeviceToHost, stream));
+    QA_HIP_CHECK(hipMemcpyAsync(hPartials.data() + 4096, dPartials + 4096, grid * sizeof(double),
+        hipMemcpyDeviceToHost, stream));
+    Finish();
+    double re = 0, im = 0;
+    for (int i = 0; i < grid; ++i) {
+        re += hPartials[i];
+        im += hPartials[4096 + i];
+    }
+    if (tmp) freeDev(tmp, maxQPower);
+    const double inner = std::sqrt(re * re + im * im);
+    return std::max(0.0, 2.0 - 2.0 * inner);
+}
+
+// ---- structural -------------------------------------------------------------
+
+template <typename R> bitLenInt QEngineHIP<R>::Compose(QInterfacePtr<R> toCopy, bitLenInt start)
+{
+    const bitLenInt oQubits = toCopy->GetQubitCount();
+    const bitCapInt nMaxQPower = maxQPower << oQubits;
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    QEngineHIP<R>* o = dynamic_cast<QEngineHIP<R>*>(toCopy.get());
+    cplx<R>* otherBuf = nullptr;
+    cplx<R>* tmp = nullptr;
+    if (o && o->deviceId == deviceId) {
+        o->Finish();
+        otherBuf = o->dState;
+    } else {
+        std::vector<cplx<R>> host(toCopy->GetMaxQPower());
+        toCopy->GetQuantumState(host.data());
+        tmp = allocDev(toCopy->GetMaxQPower());
+        QA_HIP_CHECK(hipMemcpyAsync(tmp, host.data(), sizeof(cplx<R>) * toCopy->GetMaxQPower(),
+            hipMemcpyHostToDevice, stream));
+        otherBuf = tmp;
+    }
+    cplx<R>* nBuf = allocDev(nMaxQPower);
+    launchCompose<R>(dState, otherBuf, nBuf, nMaxQPower, start, oQubits, stream);
+    Finish();
+    if (tmp) freeDev(tmp, toCopy->GetMaxQPower());
+    resizeState(nMaxQPower, nBuf);
+    this->SetQubitCount(qubitCount + oQubits);
+    runningNorm = (R)-1;
+    return start;
+}
+
+template <typename R> std::vector<double> QEngineHIP<R>::partProbs(bitLenInt start, bitLenInt length)
+{
+    const bitCapInt partPower = pow2(length);
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    double* dProbs = nullptr;
+    QA_HIP_CHECK(hipMallocAsync(&dProbs, partPower * sizeof(double), stream));
+    QA_HIP_CHECK(hipMemsetAsync(dProbs, 0, partPower * sizeof(double), stream));
+    launchPartProbs<R>(dState, maxQPower, start, length, dProbs, stream);
+    std::vector<double> probs(partPower);
+    QA_HIP_CHECK(hipMemcpyAsync(
+        probs.data(), dProbs, partPower * sizeof(double), hipMemcpyDeviceToHost, stream));
+    Finish();
+    QA_HIP_CHECK(hipFreeAsync(dProbs, stream));
+    return probs;
+}
+
+template <typename R> void QEngineHIP<R>::Decompose(bitLenInt start, QInterfacePtr<R> dest)
+{
+    // Schmidt-rank-1 split on-device (parity: decomposeprob/decomposeamp).
+    const bitLenInt len = dest->GetQubitCount();
+    const bitLenInt remLen = qubitCount - len;
+    const bitCapInt partPower = pow2(len);
+    const bitCapInt remPower = pow2(remLen);
+
+    const std::vector<double> pProbs = partProbs(start, len);
+    bitCapInt pStar = 0;
+    for (bitCapInt p = 1; p < partPower; ++p) {
+        if (pProbs[p] > pProbs[pStar]) pStar = p;
+    }
+    // global max amplitude gives the remainder row with max |a_r|
+    const auto mx = argMax();
+    const bitCapInt fullIdx = mx.second;
+    const bitCapInt lowMask = pow2Mask(start);
+    const bitCapInt rStar = (fullIdx & lowMask) | ((fullIdx >> (start + len)) << start);
+    // remProb[rStar] via masked reduction (remainder bits = all but the part register)
+    const bitCapInt partMask = pow2Mask(len) << start;
+    ReduceArgs ra{};
+    ra.maxI = maxQPower;
+    ra.mask = ~partMask & (maxQPower - 1u);
+    ra.perm = (fullIdx & lowMask) | (fullIdx & ~(lowMask | partMask));
+    const double remProbStar = reduceSum((int)ReduceOp::PROB_MASK, ra);
+    const cplx<R> ampRP = GetAmplitude(fullIdx & ~partMask | (pStar << start));
+
+    const double pNorm = 1.0 / std::sqrt(std::max(1e-300, pProbs[pStar]));
+    const double rNorm = 1.0 / std::sqrt(std::max(1e-300, remProbStar));
+
+    // dest amplitudes
+    QEngineHIP<R>* od = dynamic_cast<QEngineHIP<R>*>(dest.get());
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    cplx<R>* dDest = nullptr;
+    bool destLocal = od && od->deviceId == deviceId;
+    if (destLocal) {
+        od->Finish();
+        dDest = od->dState;
+    } else {
+        QA_HIP_CHECK(hipMallocAsync(&dDest, partPower * sizeof(cplx<R>), stream));
+    }
+    launchGatherPart<R>(dState, dDest, partPower, start, len, rStar, cplx<R>((R)rNorm, 0), stream);
+
+    // remainder amplitudes, with the double-counted-phase correction folded in
+    cplx<R>* nBuf = allocDev(remPower);
+    // corr = orig(rStar,pStar) / (remAmp[rStar] * partAmp[pStar]);
+    // remAmp[r] = pNorm * amp(r, pStar), partAmp[p] = rNorm * amp(rStar, p)
+    cplx<R> prod = cplx<R>((R)(pNorm * rNorm), 0) * ampRP * ampRP;
+    cplx<R> corrScale;
+    if (norm(prod) > 0) {
+        const std::complex<double> corr =
+            std::complex<double>(ampRP.re, ampRP.im) /
+            std::complex<double>(prod.re, prod.im);
+        corrScale = cplx<R>((R)(corr.real() * pNorm), (R)(corr.imag() * pNorm));
+    } else {
+        corrScale = cplx<R>((R)pNorm, 0);
+    }
+    launchDisposeSlice<R>(dState, nBuf, remPower, start, len, pStar, corrScale, stream);
+    Finish();
+
+    if (destLocal) {
+        od->runningNorm = (R)-1;
+    } else {
+        std::vector<cplx<R>> host(partPower);
+        QA_HIP_CHECK(hipMemcpyAsync(
+            host.data(), dDest, partPower * sizeof(cplx<R>), hipMemcpyDeviceToHost, stream));
+        Finish();
+        QA_HIP_CHECK(hipFreeAsync(dDest, stream));
+        dest->SetQuantumState(host.data());
+    }
+    resizeState(remPower, nBuf);
+    this->SetQubitCount(remLen);
+    runningNorm = (R)-1;
+    if (doNormalize) NormalizeState();
+}
+
+template <typename R> void QEngineHIP<R>::Dispose(bitLenInt start, bitLenInt length)
+{
+    const std::vector<double> pProbs = partProbs(start, length);
+    bitCapInt pStar = 0;
+    for (bitCapInt p = 1; p < (bitCapInt)pProbs.size(); ++p) {
+        if (pProbs[p] > pProbs[pStar]) pStar = p;
+    }
+    const double scale = 1.0 / std::sqrt(std::max(1e-300, pProbs[pStar]));
+    const bitCapInt remPower = maxQPower >> length;
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    cplx<R>* nBuf = allocDev(remPower);
+    launchDisposeSlice<R>(dState, nBuf, remPower, start, length, pStar, cplx<R>((R)scale, 0), stream);
+    Finish();
+    resizeState(remPower, nBuf);
+    this->SetQubitCount(qubitCount - length);
+    runningNorm = (R)-1;
+}
+
+template <typename R>
+void QEngineHIP<R>::Dispose(bitLenInt start, bitLenInt length, bitCapInt disposedPerm)
+{
+    const bitCapInt remPower = maxQPower >> length;
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    cplx<R>* nBuf = allocDev(remPower);
+    launchDisposeSlice<R>(dState, nBuf, remPower, start, length, disposedPerm, cplx<R>(1, 0), stream);
+    Finish();
+    resizeState(remPower, nBuf);
+    this->SetQubitCount(qubitCount - length);
+    runningNorm = (R)-1;
+    if (doNormalize) NormalizeState();
+}
+
+template <typename R> bitLenInt QEngineHIP<R>::Allocate(bitLenInt start, bitLenInt length)
+{
+    if (!length) return start;
+    const bitCapInt nMaxQPower = maxQPower << length;
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    cplx<R>* nBuf = allocDev(nMaxQPower);
+    launchAllocateExpand<R>(dState, nBuf, nMaxQPower, start, length, stream);
+    Finish();
+    resizeState(nMaxQPower, nBuf);
+    this->SetQubitCount(qubitCount + length);
+    return start;
+}
+
+template <typename R> QInterfacePtr<R> QEngineHIP<R>::Clone()
+{
+    auto clone = std::make_shared<QEngineHIP<R>>(
+        qubitCount, 0u, this->rand_generator, doNormalize, amplitudeFloor, deviceId);
+    Finish();
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    QA_HIP_CHECK(hipMemcpyAsync(
+        clone->dState, dState, sizeof(cplx<R>) * maxQPower, hipMemcpyDeviceToDevice, clone->stream));
+    clone->Finish();
+    clone->runningNorm = runningNorm;
+    return clone;
+}
+
+// ---- ALU --------------------------------------------------------------------
+
+template <typename R> void QEngineHIP<R>::permuteOp(PermArgs& a, bool partialSpace, bool copyFirst)
+{
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    ensureScratch();
+    if (copyFirst) {
+        QA_HIP_CHECK(hipMemcpyAsync(
+            dScratch, dState, sizeof(cplx<R>) * maxQPower, hipMemcpyDeviceToDevice, stream));
+    } else if (partialSpace) {
+        QA_HIP_CHECK(hipMemsetAsync(dScratch, 0, sizeof(cplx<R>) * maxQPower, stream));
+    }
+    launchPermute<R>(dState, dScratch, a, stream);
+    Finish();
+    swapScratch();
+}
+
+static void fillSkips(PermArgs& a, std::vector<bitCapInt> powers)
+{
+    std::sort(powers.begin(), powers.end());
+    if ((int)powers.size() > QA_MAX_SKIP_POWERS) throw QrackError("too many fixed qubits in ALU op");
+    a.nPowers = (int)powers.size();
+    for (int i = 0; i < a.nPowers; ++i) a.qPowers[i] = powers[i];
+}
+
+template <typename R> void QEngineHIP<R>::INC(bitCapInt toAdd, bitLenInt start, bitLenInt length)
+{
+    if (!length) return;
+    toAdd &= pow2Mask(length);
+    if (!toAdd) return;
+    PermArgs a{};
+    a.op = (int)PermOp::INC;
+    a.maxI = maxQPower;
+    a.start = start;
+    a.length = length;
+    a.operand = toAdd;
+    permuteOp(a, false, false);
+}
+
+template <typename R>
+void QEngineHIP<R>::CINC(
+    bitCapInt toAdd, bitLenInt start, bitLenInt length, const std::vector<bitLenInt>& controls)
+{
+    if (controls.empty()) {
+        INC(toAdd, start, length);
+        return;
+    }
+    if (!length) return;
+    toAdd &= pow2Mask(length);
+    if (!toAdd) return;
+    PermArgs a{};
+    a.op = (int)PermOp::INC;
+    a.start = start;
+    a.length = length;
+    a.operand = toAdd;
+    std::vector<bitCapInt> powers;
+    for (bitLenInt c : controls) {
+        powers.push_back(pow2(c));
+        a.controlMask |= pow2(c);
+    }
+    fillSkips(a, powers);
+    a.maxI = maxQPower >> (bitLenInt)controls.size();
+    permuteOp(a, false, true);
+}
+
+template <typename R>
+void QEngineHIP<R>::INCC(bitCapInt toAdd, bitLenInt start, bitLenInt length, bitLenInt carryIndex)
+{
+    const bool hasCarry = this->M(carryIndex);
+    if (hasCarry) {
+        this->X(carryIndex);
+        ++toAdd;
+    }
+    if (!length) return;
+    PermArgs a{};
+    a.op = (int)PermOp::INCDECC;
+    a.start = start;
+    a.length = length;
+    a.operand = toAdd & pow2Mask(length);
+    a.carryMask = pow2(carryIndex);
+    fillSkips(a, { a.carryMask });
+    a.maxI = maxQPower >> 1u;
+    permuteOp(a, true, false);
+}
+
+template <typename R>
+void QEngineHIP<R>::DECC(bitCapInt toSub, bitLenInt start, bitLenInt length, bitLenInt carryIndex)
+{
+    const bool hasCarry = this->M(carryIndex);
+    bitCapInt invToSub = (pow2(length) - toSub) & pow2Mask(length);
+    if (hasCarry) {
+        this->X(carryIndex);
+    } else {
+        invToSub = (invToSub - 1u) & pow2Mask(length);
+    }
+    if (!length) return;
+    PermArgs a{};
+    a.op = (int)PermOp::INCDECC;
+    a.start = start;
+    a.length = length;
+    a.operand = invToSub;
+    a.carryMask = pow2(carryIndex);
+    fillSkips(a, { a.carryMask });
+    a.maxI = maxQPower >> 1u;
+    permuteOp(a, true, false);
+}
+
+template <typename R>
+void QEngineHIP<R>::INCS(bitCapInt toAdd, bitLenInt start, bitLenInt length, bitLenInt overflowIndex)
+{
+    if (!length) return;
+    toAdd &= pow2Mask(length);
+    if (!toAdd) return;
+    PermArgs a{};
+    a.op = (int)PermOp::INCS;
+    a.maxI = maxQPower;
+    a.start = start;
+    a.length = length;
+    a.operand = toAdd;
+    a.carryMask = pow2(overflowIndex);
+    permuteOp(a, false, false);
+}
+
+template <typename R>
+void QEngineHIP<R>::MUL(bitCapInt toMul, bitLenInt inOutStart, bitLenInt carryStart, bitLenInt length)
+{
+    if (!toMul) throw QrackError("MUL by zero is not invertible");
+    if (toMul == 1u) return;
+    PermArgs a{};
+    a.op = (int)PermOp::MUL;
+    a.start = inOutStart;
+    a.length = length;
+    a.start2 = carryStart;
+    a.operand = toMul;
+    std::vector<bitCapInt> powers;
+    for (bitLenInt i = 0; i < length; ++i) powers.push_back(pow2(carryStart + i));
+    fillSkips(a, powers);
+    a.maxI = maxQPower >> length;
+    permuteOp(a, true, false);
+}
+
+template <typename R>
+void QEngineHIP<R>::DIV(bitCapInt toDiv, bitLenInt inOutStart, bitLenInt carryStart, bitLenInt length)
+{
+    if (!toDiv) throw QrackError("DIV by zero");
+    if (toDiv == 1u) return;
+    PermArgs a{};
+    a.op = (int)PermOp::DIV;
+    a.start = inOutStart;
+    a.length = length;
+    a.start2 = carryStart;
+    a.operand = toDiv;
+    std::vector<bitCapInt> powers;
+    for (bitLenInt i = 0; i < length; ++i) powers.push_back(pow2(carryStart + i));
+    fillSkips(a, powers);
+    a.maxI = maxQPower >> length;
+    permuteOp(a, true, false);
+}
+
+template <typename R>
+static void setupModArgs(PermArgs& a, PermOp op, bitCapInt operand, bitCapInt modN, bitLenInt inStart,
+    bitLenInt outStart, bitLenInt length, bitCapInt maxQPower, const std::vector<bitLenInt>& controls)
+{
+    a.op = (int)op;
+    a.start = inStart;
+    a.length = length;
+    a.start2 = outStart;
+    a.length2 = length;
+    a.operand = operand;
+    a.modN = modN;
+    std::vector<bitCapInt> powers;
+    for (bitLenInt i = 0; i < length; ++i) powers.push_back(pow2(outStart + i));
+    for (bitLenInt c : controls) {
+        powers.push_back(pow2(c));
+        a.controlMask |= pow2(c);
+    }
+    fillSkips(a, powers);
+    a.maxI = maxQPower >> (length + (bitLenInt)controls.size());
+}
+
+template <typename R>
+void QEngineHIP<R>::MULModNOut(
+    bitCapInt toMul, bitCapInt modN, bitLenInt inStart, bitLenInt outStart, bitLenInt length)
+{
+    PermArgs a{};
+    setupModArgs<R>(a, PermOp::MULMODN, toMul, modN, inStart, outStart, length, maxQPower, {});
+    permuteOp(a, true, false);
+}
+
+template <typename R>
+void QEngineHIP<R>::IMULModNOut(
+    bitCapInt toMul, bitCapInt modN, bitLenInt inStart, bitLenInt outStart, bitLenInt length)
+{
+    PermArgs a{};
+    setupModArgs<R>(a, PermOp::IMULMODN, toMul, modN, inStart, outStart, length, maxQPower, {});
+    permuteOp(a, true, false);
+}
+
+template <typename R>
+void QEngineHIP<R>::POWModNOut(
+    bitCapInt base, bitCapInt modN, bitLenInt inStart, bitLenInt outStart, bitLenInt length)
+{
+    PermArgs a{};
+    setupModArgs<R>(a, PermOp::POWMODN, base, modN, inStart, outStart, length, maxQPower, {});
+    permuteOp(a, true, false);
+}
+
+template <typename R>
+void QEngineHIP<R>::CMUL(bitCapInt toMul, bitLenInt inOutStart, bitLenInt carryStart, bitLenInt length,
+    const std::vector<bitLenInt>& controls)
+{
+    if (controls.empty()) {
+        MUL(toMul, inOutStart, carryStart, length);
+        return;
+    }
+    if (!toMul) throw QrackError("CMUL by zero is not invertible");
+    if (toMul == 1u) return;
+    PermArgs a{};
+    a.op = (int)PermOp::MUL;
+    a.start = inOutStart;
+    a.length = length;
+    a.start2 = carryStart;
+    a.operand = toMul;
+    std::vector<bitCapInt> powers;
+    for (bitLenInt i = 0; i < length; ++i) powers.push_back(pow2(carryStart + i));
+    for (bitLenInt c : controls) {
+        powers.push_back(pow2(c));
+        a.controlMask |= pow2(c);
+    }
+    fillSkips(a, powers);
+    a.maxI = maxQPower >> (length + (bitLenInt)controls.size());
+    permuteOp(a, true, true);
+}
+
+template <typename R>
+void QEngineHIP<R>::CDIV(bitCapInt toDiv, bitLenInt inOutStart, bitLenInt carryStart, bitLenInt length,
+    const std::vector<bitLenInt>& controls)
+{
+    if (controls.empty()) {
+        DIV(toDiv, inOutStart, carryStart, length);
+        return;
+    }
+    if (!toDiv) throw QrackError("CDIV by zero");
+    if (toDiv == 1u) return;
+    PermArgs a{};
+    a.op = (int)PermOp::DIV;
+    a.start = inOutStart;
+    a.length = length;
+    a.start2 = carryStart;
+    a.operand = toDiv;
+    std::vector<bitCapInt> powers;
+    for (bitLenInt i = 0; i < length; ++i) powers.push_back(pow2(carryStart + i));
+    for (bitLenInt c : controls) {
+        powers.push_back(pow2(c));
+        a.controlMask |= pow2(c);
+    }
+    fillSkips(a, powers);
+    a.maxI = maxQPower >> (length + (bitLenInt)controls.size());
+    permuteOp(a, true, true);
+}
+
+template <typename R>
+void QEngineHIP<R>::CMULModNOut(bitCapInt toMul, bitCapInt modN, bitLenInt inStart, bitLenInt outStart,
+    bitLenInt length, const std::vector<bitLenInt>& controls)
+{
+    if (controls.empty()) {
+        MULModNOut(toMul, modN, inStart, outStart, length);
+        return;
+    }
+    PermArgs a{};
+    setupModArgs<R>(a, PermOp::MULMODN, toMul, modN, inStart, outStart, length, maxQPower, controls);
+    permuteOp(a, true, true);
+}
+
+template <typename R>
+void QEngineHIP<R>::CIMULModNOut(bitCapInt toMul, bitCapInt modN, bitLenInt inStart, bitLenInt outStart,
+    bitLenInt length, const std::vector<bitLenInt>& controls)
+{
+    if (controls.empty()) {
+        IMULModNOut(toMul, modN, inStart, outStart, length);
+        return;
+    }
+    PermArgs a{};
+    setupModArgs<R>(a, PermOp::IMULMODN, toMul, modN, inStart, outStart, length, maxQPower, controls);
+    permuteOp(a, true, true);
+}
+
+template <typename R>
+void QEngineHIP<R>::CPOWModNOut(bitCapInt base, bitCapInt modN, bitLenInt inStart, bitLenInt outStart,
+    bitLenInt length, const std::vector<bitLenInt>& controls)
+{
+    if (controls.empty()) {
+        POWModNOut(base, modN, inStart, outStart, length);
+        return;
+    }
+    PermArgs a{};
+    setupModArgs<R>(a, PermOp::POWMODN, base, modN, inStart, outStart, length, maxQPower, controls);
+    permuteOp(a, true, true);
+}
+
+template <typename R>
+static unsigned char* uploadTable(
+    const unsigned char* values, size_t bytes, int deviceId, hipStream_t stream)
+{
+    unsigned char* dTable = nullptr;
+    QA_HIP_CHECK(hipMallocAsync(&dTable, bytes, stream));
+    QA_HIP_CHECK(hipMemcpyAsync(dTable, values, bytes, hipMemcpyHostToDevice, stream));
+    return dTable;
+}
+
+template <typename R>
+bitCapInt QEngineHIP<R>::IndexedLDA(bitLenInt indexStart, bitLenInt indexLength, bitLenInt valueStart,
+    bitLenInt valueLength, const unsigned char* values, bool resetValue)
+{
+    if (resetValue) {
+        const int bytes = (int)((valueLength + 7u) / 8u);
+        unsigned char* dTable =
+            uploadTable<R>(values, ((size_t)1 << indexLength) * bytes, deviceId, stream);
+        PermArgs a{};
+        a.op = (int)PermOp::LDA;
+        a.start = indexStart;
+        a.length = indexLength;
+        a.start2 = valueStart;
+        a.length2 = valueLength;
+        a.table = dTable;
+        a.tableBytes = bytes;
+        std::vector<bitCapInt> powers;
+        for (bitLenInt i = 0; i < valueLength; ++i) powers.push_back(pow2(valueStart + i));
+        fillSkips(a, powers);
+        a.maxI = maxQPower >> valueLength;
+        permuteOp(a, true, false);
+        QA_HIP_CHECK(hipFreeAsync(dTable, stream));
+    }
+    std::vector<bitLenInt> bits;
+    for (bitLenInt i = 0; i < valueLength; ++i) bits.push_back(valueStart + i);
+    return (bitCapInt)(this->ExpectationBitsAll(bits) + 0.5);
+}
+
+template <typename R>
+bitCapInt QEngineHIP<R>::IndexedADC(bitLenInt indexStart, bitLenInt indexLength, bitLenInt valueStart,
+    bitLenInt valueLength, bitLenInt carryIndex, const unsigned char* values)
+{
+    const bool hasCarry = this->M(carryIndex);
+    bitCapInt extra = 0;
+    if (hasCarry) {
+        this->X(carryIndex);
+        extra = 1;
+    }
+    const int bytes = (int)((valueLength + 7u) / 8u);
+    unsigned char* dTable =
+        uploadTable<R>(values, ((size_t)1 << indexLength) * bytes, deviceId, stream);
+    PermArgs a{};
+    a.op = (int)PermOp::ADC;
+    a.start = indexStart;
+    a.length = indexLength;
+    a.start2 = valueStart;
+    a.length2 = valueLength;
+    a.carryMask = pow2(carryIndex);
+    a.extra = extra;
+    a.table = dTable;
+    a.tableBytes = bytes;
+    fillSkips(a, { a.carryMask });
+    a.maxI = maxQPower >> 1u;
+    permuteOp(a, true, false);
+    QA_HIP_CHECK(hipFreeAsync(dTable, stream));
+    std::vector<bitLenInt> bits;
+    for (bitLenInt i = 0; i < valueLength; ++i) bits.push_back(valueStart + i);
+    return (bitCapInt)(this->ExpectationBitsAll(bits) + 0.5);
+}
+
+template <typename R>
+bitCapInt QEngineHIP<R>::IndexedSBC(bitLenInt indexStart, bitLenInt indexLength, bitLenInt valueStart,
+    bitLenInt valueLength, bitLenInt carryIndex, const unsigned char* values)
+{
+    const bool hasCarry = this->M(carryIndex);
+    bitCapInt extra = 0;
+    if (hasCarry) {
+        this->X(carryIndex);
+    } else {
+        extra = (bitCapInt)0 - 1u;
+    }
+    const int bytes = (int)((valueLength + 7u) / 8u);
+    unsigned char* dTable =
+        uploadTable<R>(values, ((size_t)1 << indexLength) * bytes, deviceId, stream);
+    PermArgs a{};
+    a.op = (int)PermOp::SBC;
+    a.start = indexStart;
+    a.length = indexLength;
+    a.start2 = valueStart;
+    a.length2 = valueLength;
+    a.carryMask = pow2(carryIndex);
+    a.extra = extra;
+    a.table = dTable;
+    a.tableBytes = bytes;
+    fillSkips(a, { a.carryMask });
+    a.maxI = maxQPower >> 1u;
+    permuteOp(a, true, false);
+    QA_HIP_CHECK(hipFreeAsync(dTable, stream));
+    std::vector<bitLenInt> bits;
+    for (bitLenInt i = 0; i < valueLength; ++i) bits.push_back(valueStart + i);
+    return (bitCapInt)(this->ExpectationBitsAll(bits) + 0.5);
+}
+
+template <typename R>
+void QEngineHIP<R>::Hash(bitLenInt start, bitLenInt length, const unsigned char* values)
+{
+    const int bytes = (int)((length + 7u) / 8u);
+    unsigned char* dTable = uploadTable<R>(values, ((size_t)1 << length) * bytes, deviceId, stream);
+    PermArgs a{};
+    a.op = (int)PermOp::HASH;
+    a.maxI = maxQPower;
+    a.start = start;
+    a.length = length;
+    a.table = dTable;
+    a.tableBytes = bytes;
+    permuteOp(a, false, false);
+    QA_HIP_CHECK(hipFreeAsync(dTable, stream));
+}
+
+template <typename R> void QEngineHIP<R>::ROL(bitLenInt shift, bitLenInt start, bitLenInt length)
+{
+    if (!length) return;
+    shift %= length;
+    if (!shift) return;
+    PermArgs a{};
+    a.op = (int)PermOp::ROL;
+    a.maxI = maxQPower;
+    a.start = start;
+    a.length = length;
+    a.operand = shift;
+    permuteOp(a, false, false);
+}
+
+template <typename R>
+void QEngineHIP<R>::PhaseFlipIfLess(bitCapInt greaterPerm, bitLenInt start, bitLenInt length)
+{
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    launchPhaseFlipIfLess<R>(
+        dState, maxQPower, greaterPerm, start, pow2Mask(length) << start, 0u, stream);
+}
+
+template <typename R>
+void QEngineHIP<R>::CPhaseFlipIfLess(
+    bitCapInt greaterPerm, bitLenInt start, bitLenInt length, bitLenInt flagIndex)
+{
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    launchPhaseFlipIfLess<R>(
+        dState, maxQPower, greaterPerm, start, pow2Mask(length) << start, pow2(flagIndex), stream);
+}
+
+// ---- factory hook -----------------------------------------------------------
+
+template <typename R>
+QInterfacePtr<R> MakeHipEngine(bitLenInt qubits, bitCapInt initPerm, RngPtr rng, int64_t deviceId)
+{
+    return std::make_shared<QEngineHIP<R>>(
+        qubits, initPerm, rng, true, eps<R>::value, deviceId);
+}
+
+template QInterfacePtr<float> MakeHipEngine<float>(bitLenInt, bitCapInt, RngPtr, int64_t);
+template QInterfacePtr<double> MakeHipEngine<double>(bitLenInt, bitCapInt, RngPtr, int64_t);
+
+template class QEngineHIP<float>;
+template class QEngineHIP<double>;
+
+} // namespace qrack_amd

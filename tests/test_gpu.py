@@ -1,0 +1,325 @@
+"""HIP engine (MI355X) tests — run on a GPU box via gpurun.
+
+Each numerics test compares the HIP engine against the plain numpy fp64
+reference simulator (ref_sim.RefSim) and/or the CPU engine with identical
+seeds. Parity model: the engine-matrix rerun strategy of
+/root/reference/test/test_main.cpp (same cases, every backend).
+"""
+
+import numpy as np
+import pytest
+
+import qrack_amd as qa
+from ref_sim import RefSim, assert_states_close
+
+pytestmark = pytest.mark.gpu
+
+requires_gpu = pytest.mark.skipif(qa.hip_device_count() == 0, reason="no HIP device")
+
+
+def make(n, seed=7, precision="fp32"):
+    assert qa.hip_device_count() > 0, "HIP engine must be present on GPU box (no silent fallback)"
+    return qa.create_simulator(n, precision=precision, engine="hip", seed=seed)
+
+
+def rand_unitary_2x2(rng):
+    m = rng.normal(size=(2, 2)) + 1j * rng.normal(size=(2, 2))
+    q, r = np.linalg.qr(m)
+    return q * (np.diag(r) / np.abs(np.diag(r)))
+
+
+@pytest.mark.parametrize("precision", ["fp32", "fp64"])
+def test_gate_numerics_vs_reference(precision):
+    n = 10
+    rng = np.random.default_rng(101)
+    q = make(n, precision=precision)
+    ref = RefSim(n)
+    for layer in range(8):
+        for i in range(n):
+            u = rand_unitary_2x2(rng)
+            q.mtrx(list(u.flatten()), i)
+            ref.mtrx(u, i)
+        for i in range(0, n - 1, 2):
+            q.cnot(i, i + 1)
+            ref.x(i + 1, controls=[i])
+        c, t = rng.choice(n, 2, replace=False)
+        q.cz(int(c), int(t))
+        ref.z(int(t), controls=[int(c)])
+    atol = 2e-4 if precision == "fp32" else 1e-9
+    assert_states_close(q.get_state_vector(), ref.state, atol)
+
+
+def test_phase_invert_fast_paths():
+    n = 8
+    rng = np.random.default_rng(103)
+    q = make(n)
+    ref = RefSim(n)
+    for i in range(n):
+        q.h(i)
+        ref.h(i)
+    for _ in range(20):
+        t = int(rng.integers(n))
+        which = rng.integers(4)
+        if which == 0:
+            q.z(t); ref.z(t)
+        elif which == 1:
+            q.s(t); ref.s(t)
+        elif which == 2:
+            q.x(t); ref.x(t)
+        else:
+            q.y(t); ref.y(t)
+    assert_states_close(q.get_state_vector(), ref.state, 1e-4)
+
+
+def test_controlled_and_swap():
+    n = 8
+    rng = np.random.default_rng(107)
+    q = make(n)
+    ref = RefSim(n)
+    for i in range(n):
+        th = rng.uniform(0, np.pi)
+        q.ry(th, i)
+        ref.ry(th, i)
+    q.ccnot(0, 1, 2); ref.x(2, controls=[0, 1])
+    q.swap(3, 6); ref.swap(3, 6)
+    q.mcmtrx([4, 5], list(rand_unitary_2x2(np.random.default_rng(1)).flatten()), 7)
+    ref.mtrx(rand_unitary_2x2(np.random.default_rng(1)), 7, controls=[4, 5])
+    q.sqrt_swap(0, 7)
+    q.sqrt_swap(0, 7)
+    ref.swap(0, 7)
+    assert_states_close(q.get_state_vector(), ref.state, 1e-4)
+
+
+def test_multiplexer_gpu():
+    n = 6
+    rng = np.random.default_rng(109)
+    q = make(n)
+    ref = RefSim(n)
+    for i in range(n):
+        q.h(i)
+        ref.h(i)
+    mtrxs = [rand_unitary_2x2(rng) for _ in range(4)]
+    flat = np.concatenate([m.flatten() for m in mtrxs]).astype(np.complex128)
+    q.uniformly_controlled_single_bit([1, 3], 0, flat)
+    ref.mtrx(mtrxs[0], 0, anti=[1, 3])
+    ref.mtrx(mtrxs[1], 0, controls=[1], anti=[3])
+    ref.mtrx(mtrxs[2], 0, controls=[3], anti=[1])
+    ref.mtrx(mtrxs[3], 0, controls=[1, 3])
+    assert_states_close(q.get_state_vector(), ref.state, 1e-4)
+
+
+def test_mask_gates_gpu():
+    n = 8
+    q = make(n)
+    ref = RefSim(n)
+    for i in range(n):
+        q.h(i)
+        ref.h(i)
+    q.x_mask(0b1010101)
+    for t in (0, 2, 4, 6):
+        ref.x(t)
+    q.z_mask(0b0110)
+    ref.z(1)
+    ref.z(2)
+    q.phase_parity(0.9, 0b11011)
+    idx = np.arange(1 << n)
+    par = np.zeros(1 << n, dtype=bool)
+    for b in (0, 1, 3, 4):
+        par ^= ((idx >> b) & 1).astype(bool)
+    ref.state[par] *= np.exp(1j * 0.45)
+    ref.state[~par] *= np.exp(-1j * 0.45)
+    assert_states_close(q.get_state_vector(), ref.state, 1e-4)
+
+
+def test_prob_and_measure_gpu():
+    q = make(12, seed=5)
+    q.h(0)
+    q.cnot(0, 1)
+    assert abs(q.prob(1) - 0.5) < 1e-5
+    assert abs(q.prob_mask(0b11, 0b11) - 0.5) < 1e-5
+    assert abs(q.prob_parity(0b11)) < 1e-5
+    r = q.force_m(0, True)
+    assert r
+    assert abs(q.prob(1) - 1.0) < 1e-5
+
+
+def test_m_all_and_multishot_gpu():
+    q = make(10, seed=6)
+    q.h(0)
+    q.cnot(0, 1)
+    res = q.multi_shot_measure_mask([1, 2], 1000)
+    assert sum(res.values()) == 1000
+    assert set(res.keys()) <= {0, 3}
+    assert 350 < res.get(0, 0) < 650
+    r = q.m_all()
+    assert r in (0, 3)
+
+
+def test_qft_roundtrip_gpu():
+    n = 12
+    rng = np.random.default_rng(11)
+    q = make(n, seed=11)
+    for i in range(n):
+        q.ry(float(rng.uniform(0, np.pi)), i)
+    for i in range(n - 1):
+        q.cnot(i, i + 1)
+    before = q.get_state_vector()
+    q.qft(0, n)
+    q.iqft(0, n)
+    after = q.get_state_vector()
+    assert np.allclose(before, after, atol=1e-3)
+
+
+def test_alu_gpu():
+    q = make(8)
+    for i in (0, 2):  # reg = 5
+        q.x(i)
+    q.inc(3, 0, 4)
+    assert q.m_reg(0, 4) == 8
+    q2 = make(8)
+    q2.x(0)
+    q2.x(1)  # 3
+    q2.mul(5, 0, 4, 4)
+    assert q2.m_reg(0, 8) == 15
+    q3 = make(8)
+    q3.x(1)
+    q3.x(2)  # in = 6
+    q3.mul_mod_n_out(7, 15, 0, 4, 4)
+    r = q3.m_all()
+    assert (r >> 4) == 12
+    q4 = make(8)
+    q4.x(0)
+    q4.x(1)  # 3
+    q4.pow_mod_n_out(2, 15, 0, 4, 4)
+    assert (q4.m_all() >> 4) == 8
+
+
+def test_alu_superposition_gpu():
+    q = make(6, seed=9)
+    q.h(0)
+    q.inc(1, 0, 3)
+    sv = q.get_state_vector()
+    assert abs(abs(sv[1]) - 1 / np.sqrt(2)) < 1e-4
+    assert abs(abs(sv[2]) - 1 / np.sqrt(2)) < 1e-4
+
+
+def test_indexed_lda_gpu():
+    table = bytes([10, 20, 30, 40])
+    q = make(8)
+    q.x(1)  # index 2
+    q.indexed_lda(0, 2, 2, 6, table)
+    assert (q.m_all() >> 2) == 30
+
+
+def test_hash_gpu():
+    table = bytes([2, 0, 3, 1])
+    q = make(2, seed=3)
+    q.h(0)
+    q.hash(0, 2, table)
+    sv = q.get_state_vector()
+    assert abs(abs(sv[2]) - 1 / np.sqrt(2)) < 1e-4
+    assert abs(abs(sv[0]) - 1 / np.sqrt(2)) < 1e-4
+
+
+def test_compose_decompose_gpu():
+    a = make(2, seed=1)
+    a.h(0)
+    b = make(1, seed=2)
+    b.x(0)
+    a.compose(b)
+    assert a.num_qubits == 3
+    assert abs(a.prob(2) - 1.0) < 1e-5
+    dest = make(1, seed=3)
+    a.decompose(2, dest)
+    assert a.num_qubits == 2
+    assert abs(dest.prob(0) - 1.0) < 1e-5
+    assert abs(a.prob(0) - 0.5) < 1e-5
+
+
+def test_decompose_entangled_part_gpu():
+    q = make(4, seed=5)
+    q.h(1)
+    q.cnot(1, 2)
+    q.h(0)
+    dest = make(2, seed=6)
+    q.decompose(1, dest)
+    assert abs(dest.prob_mask(0b11, 0b00) - 0.5) < 1e-4
+    assert abs(dest.prob_mask(0b11, 0b11) - 0.5) < 1e-4
+    assert abs(q.prob(0) - 0.5) < 1e-4
+
+
+def test_dispose_gpu():
+    q = make(4, seed=7)
+    q.h(0)
+    q.x(2)
+    q.dispose(2, 1)
+    assert q.num_qubits == 3
+    assert abs(q.prob(0) - 0.5) < 1e-4
+
+
+def test_allocate_clone_gpu():
+    q = make(3, seed=8)
+    q.h(0)
+    q.allocate(2)
+    assert q.num_qubits == 5
+    c = q.clone()
+    c.x(4)
+    assert abs(q.prob(4)) < 1e-5
+    assert abs(c.prob(4) - 1.0) < 1e-5
+    assert not q.approx_compare(c)
+
+
+def test_cpu_gpu_equivalence():
+    """Same circuit, same seed -> same measurement outcomes and states."""
+    n = 10
+    rng = np.random.default_rng(77)
+    qc = qa.create_simulator(n, engine="cpu", seed=55)
+    qg = make(n, seed=55)
+    for layer in range(5):
+        for i in range(n):
+            th = float(rng.uniform(0, 2 * np.pi))
+            qc.ry(th, i)
+            qg.ry(th, i)
+        for i in range(n - 1):
+            qc.cnot(i, i + 1)
+            qg.cnot(i, i + 1)
+    svc = np.asarray(qc.get_state_vector())
+    svg = np.asarray(qg.get_state_vector())
+    assert np.allclose(svc, svg, atol=2e-4)
+    assert qc.m_all() == qg.m_all()
+
+
+def test_expectation_gpu():
+    q = make(5, seed=3)
+    q.x(1)
+    q.h(0)
+    assert abs(q.expectation_bits_all([0, 1, 2]) - 2.5) < 1e-4
+    assert abs(q.variance_bits_all([0, 1, 2]) - 0.25) < 1e-4
+
+
+def test_phase_flip_if_less_gpu():
+    q = make(3, seed=4)
+    q.h(0)
+    q.h(1)
+    q.phase_flip_if_less(2, 0, 3)
+    sv = q.get_state_vector()
+    assert sv[0].real < 0 and sv[1].real < 0
+    assert sv[2].real > 0
+
+
+def test_sum_sqr_diff_gpu():
+    a = make(6, seed=1)
+    b = make(6, seed=2)
+    a.h(0)
+    b.h(0)
+    assert a.sum_sqr_diff(b) < 1e-5
+    b.x(3)
+    assert a.sum_sqr_diff(b) > 1.0
+
+
+def test_native_extension_loaded():
+    """Fail loudly if the native HIP path is absent on a GPU box."""
+    import qrack_amd._qrack as native
+
+    assert native.hip_device_count() > 0
+    assert "_qrack" in native.__file__
