@@ -10,8 +10,12 @@
 #include <pybind11/pybind11.h>
 #include <pybind11/stl.h>
 
+#include "common/dlpack_min.hpp"
 #include "qengine_cpu.hpp"
 #include "qfactory.hpp"
+#ifdef QRACK_AMD_HIP_ENGINE
+#include "hip/qengine_hip.hpp"
+#endif
 
 namespace py = pybind11;
 using namespace qrack_amd;
@@ -278,6 +282,90 @@ template <typename R> static void bindQInterface(py::module_& m, const char* nam
         .def("reset_unitary_fidelity", &QI::ResetUnitaryFidelity)
         .def("set_device", &QI::SetDevice)
         .def("get_device", &QI::GetDevice)
+        // ---- engine-level primitives (QPager / distributed pager support) ----
+        .def("apply_m",
+            [](Ptr q, bitCapInt mask, bitCapInt result, C nrm) {
+                auto eng = std::dynamic_pointer_cast<QEngine<R>>(q);
+                if (!eng) throw QrackError("apply_m requires a state-vector engine");
+                eng->ApplyM(mask, result, from_std<R>(nrm));
+            })
+        .def("zero_amplitudes",
+            [](Ptr q) {
+                auto eng = std::dynamic_pointer_cast<QEngine<R>>(q);
+                if (!eng) throw QrackError("zero_amplitudes requires a state-vector engine");
+                eng->ZeroAmplitudes();
+            })
+        .def("global_phase", [](QI& q, C f) { q.Phase(from_std<R>(f), from_std<R>(f), 0); })
+        .def("norm_total",
+            [](Ptr q) {
+                auto eng = std::dynamic_pointer_cast<QEngine<R>>(q);
+                if (!eng) throw QrackError("norm_total requires a state-vector engine");
+                eng->UpdateRunningNorm((R)0);
+                return (double)eng->GetRunningNorm();
+            })
+        .def("get_amplitude_page",
+            [](Ptr q, bitCapInt offset, bitCapInt length) {
+                auto eng = std::dynamic_pointer_cast<QEngine<R>>(q);
+                if (!eng) throw QrackError("get_amplitude_page requires an engine");
+                py::array_t<C> out((py::ssize_t)length);
+                eng->GetAmplitudePage(reinterpret_cast<cplx<R>*>(out.mutable_data()), offset, length);
+                return out;
+            })
+        .def("set_amplitude_page",
+            [](Ptr q, py::array_t<C, py::array::c_style | py::array::forcecast> in, bitCapInt offset) {
+                auto eng = std::dynamic_pointer_cast<QEngine<R>>(q);
+                if (!eng) throw QrackError("set_amplitude_page requires an engine");
+                eng->SetAmplitudePage(
+                    reinterpret_cast<const cplx<R>*>(in.data()), offset, (bitCapInt)in.size());
+            })
+        .def("dlpack_view",
+            [](Ptr q, bitCapInt offset, bitCapInt length) {
+                // zero-copy 1-D complex view of the amplitude buffer, consumed by
+                // torch.from_dlpack for RCCL exchanges. The engine must stay
+                // alive and un-resized while the view is in use; the holder
+                // keeps a shared_ptr so teardown order is safe.
+                struct Holder {
+                    DLManagedTensor mt;
+                    int64_t shape[1];
+                    QInterfacePtr<R> keep;
+                };
+                void* data = nullptr;
+                DLDevice dev{};
+#ifdef QRACK_AMD_HIP_ENGINE
+                if (auto hip = std::dynamic_pointer_cast<QEngineHIP<R>>(q)) {
+                    hip->Finish();
+                    data = (void*)(hip->DeviceBuffer() + offset);
+                    dev = { kDLROCM, hip->DeviceId() };
+                }
+#endif
+                if (!data) {
+                    if (auto cpu = std::dynamic_pointer_cast<QEngineCPU<R>>(q)) {
+                        data = (void*)(cpu->Amplitudes() + offset);
+                        dev = { kDLCPU, 0 };
+                    } else {
+                        throw QrackError("dlpack_view requires a state-vector engine");
+                    }
+                }
+                Holder* h = new Holder{};
+                h->keep = q;
+                h->shape[0] = (int64_t)length;
+                h->mt.dl_tensor.data = data;
+                h->mt.dl_tensor.device = dev;
+                h->mt.dl_tensor.ndim = 1;
+                h->mt.dl_tensor.dtype = { kDLComplex, (uint8_t)(sizeof(cplx<R>) * 8), 1 };
+                h->mt.dl_tensor.shape = h->shape;
+                h->mt.dl_tensor.strides = nullptr;
+                h->mt.dl_tensor.byte_offset = 0;
+                h->mt.manager_ctx = h;
+                h->mt.deleter = [](DLManagedTensor* t) { delete (Holder*)t->manager_ctx; };
+                return py::capsule(&h->mt, "dltensor", [](PyObject* cap) {
+                    if (PyCapsule_IsValid(cap, "dltensor")) {
+                        DLManagedTensor* mt =
+                            (DLManagedTensor*)PyCapsule_GetPointer(cap, "dltensor");
+                        if (mt && mt->deleter) mt->deleter(mt);
+                    }
+                });
+            })
         .def("is_clifford", [](QI& q) { return q.isClifford(); })
         .def("depolarizing_channel_weak_1qb", &QI::DepolarizingChannelWeak1Qb);
 }

@@ -120,6 +120,69 @@ template <int KIND> __global__ void k_apply2x2_1v(cplx<float>* sv, GateArgs<floa
     }
 }
 
+// one-sided diagonal scale: sv[i|offset] *= f for every expanded i.
+// The dominant QFT kernel: CPhase(topLeft=1) touches only the
+// control=1 & target=1 quarter of the state (the reference's phasesingle
+// touches half; this is the MI355X-native improvement).
+template <typename R> __global__ void k_scale_side(cplx<R>* sv, GateArgs<R> a, cplx<R> f)
+{
+    const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
+    for (bitCapInt j = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; j < a.maxI; j += stride) {
+        const bitCapInt i = expandBits(j, a.qPowers, a.nPowers) | a.offset1;
+        sv[i] = f * sv[i];
+    }
+}
+
+// fp32 vectorized one-sided scale: two adjacent expanded indices per lane
+// (requires qPowers[0] >= 2 so consecutive j stay adjacent, and even maxI)
+__global__ void k_scale_side_v(cplx<float>* sv, GateArgs<float> a, cplx<float> f)
+{
+    float4* sv4 = reinterpret_cast<float4*>(sv);
+    const bitCapInt half = a.maxI >> 1u;
+    const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
+    for (bitCapInt k = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; k < half; k += stride) {
+        const bitCapInt i = (expandBits(2u * k, a.qPowers, a.nPowers) | a.offset1) >> 1u;
+        float4 v = sv4[i];
+        const cplx<float> x{ v.x, v.y }, y{ v.z, v.w };
+        const cplx<float> nx = f * x, ny = f * y;
+        sv4[i] = make_float4(nx.re, nx.im, ny.re, ny.im);
+    }
+}
+
+// fp32 vectorized general/phase/invert pair kernel for ANY skip set with
+// qPowers[0] >= 2: two adjacent pairs per lane via float4
+template <int KIND> __global__ void k_apply2x2_v(cplx<float>* sv, GateArgs<float> a)
+{
+    float4* sv4 = reinterpret_cast<float4*>(sv);
+    const bitCapInt half = a.maxI >> 1u;
+    const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
+    const cplx<float> m0 = a.m[0], m1 = a.m[1], m2 = a.m[2], m3 = a.m[3];
+    for (bitCapInt k = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; k < half; k += stride) {
+        const bitCapInt i = expandBits(2u * k, a.qPowers, a.nPowers);
+        const bitCapInt lo4 = (i | a.offset1) >> 1u;
+        const bitCapInt hi4 = (i | a.offset2) >> 1u;
+        float4 vlo = sv4[lo4];
+        float4 vhi = sv4[hi4];
+        const cplx<float> x0{ vlo.x, vlo.y }, x1{ vlo.z, vlo.w };
+        const cplx<float> y0{ vhi.x, vhi.y }, y1{ vhi.z, vhi.w };
+        cplx<float> a0, a1, b0, b1;
+        if (KIND == 1) {
+            a0 = m0 * x0; a1 = m0 * x1;
+            b0 = m3 * y0; b1 = m3 * y1;
+        } else if (KIND == 2) {
+            a0 = m1 * y0; a1 = m1 * y1;
+            b0 = m2 * x0; b1 = m2 * x1;
+        } else {
+            a0 = m0 * x0 + m1 * y0; a1 = m0 * x1 + m1 * y1;
+            b0 = m2 * x0 + m3 * y0; b1 = m2 * x1 + m3 * y1;
+        }
+        sv4[lo4] = make_float4(a0.re, a0.im, a1.re, a1.im);
+        sv4[hi4] = make_float4(b0.re, b0.im, b1.re, b1.im);
+    }
+}
+
+template <typename R> static bool isOne(cplx<R> c) { return c.re == (R)1 && c.im == (R)0; }
+
 template <typename R> static int matrixKind(const cplx<R>* m)
 {
     const bool isPhase = (norm(m[1]) <= 0) && (norm(m[2]) <= 0);
@@ -132,8 +195,34 @@ void launchApply2x2(cplx<R>* sv, const GateArgs<R>& a, hipStream_t stream)
 {
     const int kind = matrixKind(a.m);
     const int grid = gridFor(a.maxI);
+    // fp32 pairs can vectorize as float4 whenever consecutive iteration
+    // indices stay memory-adjacent: lowest skip power >= 2
+    const bool vecOk = (a.nPowers > 0) && (a.qPowers[0] >= 2u) && ((a.maxI & 1u) == 0u);
+
+    if (kind == 1) {
+        // diagonal: skip untouched sides entirely
+        const bool tl1 = isOne(a.m[0]);
+        const bool br1 = isOne(a.m[3]);
+        if (tl1 && br1) return; // identity
+        if (tl1 || br1) {
+            GateArgs<R> s = a;
+            s.offset1 = tl1 ? a.offset2 : a.offset1;
+            const cplx<R> f = tl1 ? a.m[3] : a.m[0];
+            if constexpr (std::is_same_v<R, float>) {
+                if (vecOk) {
+                    hipLaunchKernelGGL((k_scale_side_v), dim3(gridFor(a.maxI >> 1u)),
+                        dim3(QA_BLOCK), 0, stream, sv, s, f);
+                    return;
+                }
+            }
+            hipLaunchKernelGGL(
+                (k_scale_side<R>), dim3(grid), dim3(QA_BLOCK), 0, stream, sv, s, f);
+            return;
+        }
+    }
     if constexpr (std::is_same_v<R, float>) {
         if (a.nPowers == 1 && (a.maxI & 1u) == 0u && a.maxI >= 2u && a.offset1 == 0u) {
+            // single-target fast path (handles target bit 0 too)
             const int gridv = gridFor(a.maxI >> 1u);
             switch (kind) {
             case 1:
@@ -144,6 +233,20 @@ void launchApply2x2(cplx<R>* sv, const GateArgs<R>& a, hipStream_t stream)
                 return;
             default:
                 hipLaunchKernelGGL((k_apply2x2_1v<0>), dim3(gridv), dim3(QA_BLOCK), 0, stream, sv, a);
+                return;
+            }
+        }
+        if (vecOk) {
+            const int gridv = gridFor(a.maxI >> 1u);
+            switch (kind) {
+            case 1:
+                hipLaunchKernelGGL((k_apply2x2_v<1>), dim3(gridv), dim3(QA_BLOCK), 0, stream, sv, a);
+                return;
+            case 2:
+                hipLaunchKernelGGL((k_apply2x2_v<2>), dim3(gridv), dim3(QA_BLOCK), 0, stream, sv, a);
+                return;
+            default:
+                hipLaunchKernelGGL((k_apply2x2_v<0>), dim3(gridv), dim3(QA_BLOCK), 0, stream, sv, a);
                 return;
             }
         }
