@@ -766,6 +766,82 @@ void launchPhaseFlipIfLess(cplx<R>* sv, bitCapInt maxQPower, bitCapInt greaterPe
         maxQPower, greaterPerm, start, regMask, flagMask);
 }
 
+// ---- fused QFT phase ramp ------------------------------------------------------
+
+template <typename R> __device__ __forceinline__ void devSinCos(R t, R* s, R* c);
+template <> __device__ __forceinline__ void devSinCos<float>(float t, float* s, float* c)
+{
+    __sincosf(t, s, c);
+}
+template <> __device__ __forceinline__ void devSinCos<double>(double t, double* s, double* c)
+{
+    sincos(t, s, c);
+}
+
+template <typename R>
+__global__ void k_qft_ramp(
+    cplx<R>* sv, bitCapInt maxI, bitCapInt colPow, bitLenInt start, R scale)
+{
+    // iterate indices with bit (start+col) set: maxI = maxQPower/2 pairs,
+    // expand with the column power as skip bit, then OR it in
+    const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
+    const bitCapInt lowMask = colPow - 1u; // bits below start+col
+    for (bitCapInt j = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; j < maxI; j += stride) {
+        const bitCapInt i = (((j & ~(colPow - 1u)) << 1u) | (j & (colPow - 1u))) | colPow;
+        const bitCapInt frac = (i & lowMask) >> start;
+        const R theta = scale * (R)frac;
+        R s, c;
+        devSinCos<R>(theta, &s, &c);
+        const cplx<R> f{ c, s };
+        sv[i] = f * sv[i];
+    }
+}
+
+// fp32 vectorized: two adjacent set-bit indices per lane (valid when
+// start >= 1 would change theta... adjacent i differ in bit 0: theta differs
+// unless start > 0; compute both thetas — still one float4 load/store)
+__global__ void k_qft_ramp_v(
+    cplx<float>* sv, bitCapInt maxI, bitCapInt colPow, bitLenInt start, float scale)
+{
+    float4* sv4 = reinterpret_cast<float4*>(sv);
+    const bitCapInt half = maxI >> 1u;
+    const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
+    const bitCapInt lowMask = colPow - 1u;
+    for (bitCapInt k = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; k < half; k += stride) {
+        const bitCapInt j = 2u * k;
+        const bitCapInt i = (((j & ~(colPow - 1u)) << 1u) | (j & (colPow - 1u))) | colPow;
+        const bitCapInt i4 = i >> 1u;
+        float4 v = sv4[i4];
+        const bitCapInt f0 = (i & lowMask) >> start;
+        const bitCapInt f1 = ((i + 1u) & lowMask) >> start;
+        float s0, c0, s1, c1;
+        __sincosf(scale * (float)f0, &s0, &c0);
+        __sincosf(scale * (float)f1, &s1, &c1);
+        const cplx<float> a{ v.x, v.y }, b{ v.z, v.w };
+        const cplx<float> na = cplx<float>{ c0, s0 } * a;
+        const cplx<float> nb = cplx<float>{ c1, s1 } * b;
+        sv4[i4] = make_float4(na.re, na.im, nb.re, nb.im);
+    }
+}
+
+template <typename R>
+void launchQftRamp(cplx<R>* sv, bitCapInt maxQPower, bitLenInt start, bitLenInt col, int sign,
+    hipStream_t stream)
+{
+    const bitCapInt colPow = ONE_BCI << (start + col);
+    const bitCapInt maxI = maxQPower >> 1u;
+    const R scale = (R)sign * PI_R<R> / (R)(ONE_BCI << col);
+    if constexpr (std::is_same_v<R, float>) {
+        if (colPow >= 2u && (maxI & 1u) == 0u) {
+            hipLaunchKernelGGL((k_qft_ramp_v), dim3(gridFor(maxI >> 1u)), dim3(QA_BLOCK), 0,
+                stream, sv, maxI, colPow, start, scale);
+            return;
+        }
+    }
+    hipLaunchKernelGGL((k_qft_ramp<R>), dim3(gridFor(maxI)), dim3(QA_BLOCK), 0, stream, sv, maxI,
+        colPow, start, scale);
+}
+
 // ---- sampling / inner product / marginals --------------------------------------
 
 template <typename R>
@@ -918,7 +994,8 @@ void launchPartProbs(const cplx<R>* sv, bitCapInt maxQPower, bitLenInt start, bi
     template int launchInner<R>(                                                                    \
         const cplx<R>*, const cplx<R>*, bitCapInt, double*, double*, hipStream_t);                  \
     template void launchPartProbs<R>(                                                               \
-        const cplx<R>*, bitCapInt, bitLenInt, bitLenInt, double*, hipStream_t);
+        const cplx<R>*, bitCapInt, bitLenInt, bitLenInt, double*, hipStream_t);                     \
+    template void launchQftRamp<R>(cplx<R>*, bitCapInt, bitLenInt, bitLenInt, int, hipStream_t);
 
 QA_INSTANTIATE(float)
 QA_INSTANTIATE(double)

@@ -152,6 +152,13 @@ void launchPhaseFlipIfLess(cplx<R>* sv, bitCapInt maxQPower, bitCapInt greaterPe
 
 int reduceGridSize(bitCapInt n);
 
+// fused QFT phase-ramp: for every index with bit (start+i) set,
+// amp *= exp(sign * i*pi * ((x >> start) mod 2^i) / 2^i)
+// — one pass replaces the i controlled-phase gates of a QFT column.
+template <typename R>
+void launchQftRamp(cplx<R>* sv, bitCapInt maxQPower, bitLenInt start, bitLenInt col, int sign,
+    hipStream_t stream);
+
 // contiguous per-chunk |amp|^2 sums (inverse-CDF sampling support):
 // sums[c] = sum over [c*chunkLen, (c+1)*chunkLen)
 template <typename R>

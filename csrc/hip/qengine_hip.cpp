@@ -384,6 +384,29 @@ void QEngineHIP<R>::UniformlyControlledSingleBit(
     QA_HIP_CHECK(hipFreeAsync(dMtrxs, stream));
 }
 
+template <typename R> void QEngineHIP<R>::QFT(bitLenInt start, bitLenInt length, bool)
+{
+    // H on the top column, then ONE fused ramp per column (the i controlled
+    // phases CPhaseRootN(i-j+1, j, i) collapse to exp(i*pi*(x mod 2^i)/2^i)
+    // on the bit-i-set half)
+    if (!length) return;
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    for (bitLenInt i = length; i-- > 0;) {
+        this->H(start + i);
+        if (i) launchQftRamp<R>(dState, maxQPower, start, i, +1, stream);
+    }
+}
+
+template <typename R> void QEngineHIP<R>::IQFT(bitLenInt start, bitLenInt length, bool)
+{
+    if (!length) return;
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    for (bitLenInt i = 0; i < length; ++i) {
+        if (i) launchQftRamp<R>(dState, maxQPower, start, i, -1, stream);
+        this->H(start + i);
+    }
+}
+
 // ---- reductions -------------------------------------------------------------
 
 template <typename R> double QEngineHIP<R>::reduceSum(int op, const ReduceArgs& a)

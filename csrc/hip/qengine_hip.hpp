@@ -112,6 +112,10 @@ public:
     void UniformlyControlledSingleBit(
         const std::vector<bitLenInt>& controls, bitLenInt target, const cplx<R>* mtrxs) override;
     void ROL(bitLenInt shift, bitLenInt start, bitLenInt length) override;
+    // fused QFT: per column, ONE phase-ramp kernel replaces the
+    // controlled-phase ladder (O(n) full-state passes instead of O(n^2/2))
+    void QFT(bitLenInt start, bitLenInt length, bool trySeparate = false) override;
+    void IQFT(bitLenInt start, bitLenInt length, bool trySeparate = false) override;
 
     // ---- probability / measurement ----
     R Prob(bitLenInt q) override;

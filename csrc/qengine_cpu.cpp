@@ -217,6 +217,40 @@ template <typename R> void QEngineCPU<R>::ROL(bitLenInt shift, bitLenInt start, 
     });
 }
 
+template <typename R> void QEngineCPU<R>::QftRamp(bitLenInt start, bitLenInt col, int sign)
+{
+    // fused QFT column: one pass applies all of the column's controlled
+    // phases: amp *= exp(sign*i*pi*((x>>start) mod 2^col)/2^col) on the
+    // bit-(start+col)-set half
+    const bitCapInt colPow = pow2(start + col);
+    const bitCapInt lowMask = colPow - 1u;
+    const R scale = (R)sign * PI_R<R> / (R)pow2(col);
+    cplx<R>* sv = stateVec.data();
+    this->par_for_skip(maxQPower >> 1u, colPow, [=](const bitCapInt& j, unsigned) {
+        const bitCapInt i = j | colPow;
+        const R theta = scale * (R)((i & lowMask) >> start);
+        sv[i] = polar<R>(1, theta) * sv[i];
+    });
+}
+
+template <typename R> void QEngineCPU<R>::QFT(bitLenInt start, bitLenInt length, bool)
+{
+    if (!length) return;
+    for (bitLenInt i = length; i-- > 0;) {
+        this->H(start + i);
+        if (i) QftRamp(start, i, +1);
+    }
+}
+
+template <typename R> void QEngineCPU<R>::IQFT(bitLenInt start, bitLenInt length, bool)
+{
+    if (!length) return;
+    for (bitLenInt i = 0; i < length; ++i) {
+        if (i) QftRamp(start, i, -1);
+        this->H(start + i);
+    }
+}
+
 // ---- probability -----------------------------------------------------------
 
 template <typename R> R QEngineCPU<R>::Prob(bitLenInt q)

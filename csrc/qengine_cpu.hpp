@@ -59,6 +59,8 @@ public:
     void UniformlyControlledSingleBit(
         const std::vector<bitLenInt>& controls, bitLenInt target, const cplx<R>* mtrxs) override;
     void ROL(bitLenInt shift, bitLenInt start, bitLenInt length) override;
+    void QFT(bitLenInt start, bitLenInt length, bool trySeparate = false) override;
+    void IQFT(bitLenInt start, bitLenInt length, bool trySeparate = false) override;
 
     // ---- probability / measurement ----
     R Prob(bitLenInt q) override;
@@ -131,6 +133,7 @@ protected:
     void ControlledPermutationOp(
         bitCapInt controlMask, const std::function<bitCapInt(bitCapInt)>& f);
     bitCapInt SampleOnce();
+    void QftRamp(bitLenInt start, bitLenInt col, int sign);
 };
 
 } // namespace qrack_amd

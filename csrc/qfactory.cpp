@@ -1,5 +1,7 @@
 #include "qfactory.hpp"
 
+#include "qstabilizer.hpp"
+
 #ifdef QRACK_AMD_ENABLE_HIP
 #include <hip/hip_runtime.h>
 #endif
@@ -33,6 +35,8 @@ QInterfacePtr<R> CreateStack(bitLenInt qubits, std::vector<std::string> layers, 
         engine = std::make_shared<QEngineCPU<R>>(qubits, initPerm, rng);
     } else if (inner == "hip") {
         engine = MakeHipEngine<R>(qubits, initPerm, rng, deviceId);
+    } else if (inner == "stabilizer") {
+        engine = std::make_shared<QStabilizer<R>>(qubits, initPerm, rng);
     } else {
         throw QrackError("unknown engine layer: " + inner);
     }
