@@ -1,6 +1,7 @@
 #include "qfactory.hpp"
 
 #include "qstabilizer.hpp"
+#include "qstabilizerhybrid.hpp"
 
 #ifdef QRACK_AMD_ENABLE_HIP
 #include <hip/hip_runtime.h>
@@ -19,33 +20,46 @@ int HipDeviceCount()
 #endif
 }
 
+// Recursive layer-stack factory: layers[0] is outermost.
+// (parity: qfactory.hpp CreateQuantumInterface / CreateArrangedLayersFull)
+template <typename R>
+EngineFactoryFn<R> LayerFactory(
+    std::vector<std::string> layers, RngPtr rng, int64_t deviceId, bitLenInt pagesPerDevice)
+{
+    if (layers.empty()) layers.push_back("cpu");
+    const std::string head = layers.front();
+    std::vector<std::string> tail(layers.begin() + 1, layers.end());
+
+    if (head == "cpu") {
+        return [rng](bitLenInt n, bitCapInt perm) -> QInterfacePtr<R> {
+            return std::make_shared<QEngineCPU<R>>(n, perm, rng);
+        };
+    }
+    if (head == "hip") {
+        return [rng, deviceId](bitLenInt n, bitCapInt perm) -> QInterfacePtr<R> {
+            return MakeHipEngine<R>(n, perm, rng, deviceId);
+        };
+    }
+    if (head == "stabilizer") {
+        return [rng](bitLenInt n, bitCapInt perm) -> QInterfacePtr<R> {
+            return std::make_shared<QStabilizer<R>>(n, perm, rng);
+        };
+    }
+    if (head == "stabilizer_hybrid") {
+        EngineFactoryFn<R> sub = LayerFactory<R>(tail, rng, deviceId, pagesPerDevice);
+        return [rng, sub](bitLenInt n, bitCapInt perm) -> QInterfacePtr<R> {
+            return std::make_shared<QStabilizerHybrid<R>>(n, perm, rng, sub);
+        };
+    }
+    throw QrackError("unknown layer: " + head);
+}
+
 template <typename R>
 QInterfacePtr<R> CreateStack(bitLenInt qubits, std::vector<std::string> layers, bitCapInt initPerm,
     int64_t seed, int64_t deviceId, bitLenInt pagesPerDevice)
 {
     RngPtr rng = (seed < 0) ? std::make_shared<Rng>() : std::make_shared<Rng>((uint64_t)seed);
-
-    // innermost engine first
-    if (layers.empty()) layers.push_back("cpu");
-
-    // Build from the inside out; wrapper layers are added as they land.
-    QInterfacePtr<R> engine;
-    const std::string& inner = layers.back();
-    if (inner == "cpu") {
-        engine = std::make_shared<QEngineCPU<R>>(qubits, initPerm, rng);
-    } else if (inner == "hip") {
-        engine = MakeHipEngine<R>(qubits, initPerm, rng, deviceId);
-    } else if (inner == "stabilizer") {
-        engine = std::make_shared<QStabilizer<R>>(qubits, initPerm, rng);
-    } else {
-        throw QrackError("unknown engine layer: " + inner);
-    }
-
-    for (size_t li = layers.size() - 1; li-- > 0;) {
-        const std::string& layer = layers[li];
-        throw QrackError("layer not yet available: " + layer);
-    }
-    return engine;
+    return LayerFactory<R>(layers, rng, deviceId, pagesPerDevice)(qubits, initPerm);
 }
 
 #ifndef QRACK_AMD_HIP_ENGINE

@@ -1,0 +1,517 @@
+// qrack_amd — QStabilizerHybrid implementation (see header).
+#include "qstabilizerhybrid.hpp"
+
+#include "qengine_cpu.hpp"
+
+namespace qrack_amd {
+
+template <typename R>
+QStabilizerHybrid<R>::QStabilizerHybrid(bitLenInt qBitCount, bitCapInt initState, RngPtr rgp,
+    EngineFactoryFn<R> factory, bool doNorm, R normThresh)
+    : QInterface<R>(qBitCount, rgp, doNorm, normThresh)
+    , stabilizer(std::make_shared<QStabilizer<R>>(qBitCount, initState, this->rand_generator))
+    , engine(nullptr)
+    , engineFactory(factory)
+    , shards(qBitCount)
+{
+    if (!engineFactory) {
+        RngPtr rng = this->rand_generator;
+        engineFactory = [rng](bitLenInt n, bitCapInt perm) -> QInterfacePtr<R> {
+            return std::make_shared<QEngineCPU<R>>(n, perm, rng);
+        };
+    }
+}
+
+// ---- shard machinery --------------------------------------------------------
+
+template <typename R> bool QStabilizerHybrid<R>::ShardIsIdentity(bitLenInt q) const
+{
+    return !shards[q];
+}
+
+template <typename R> bool QStabilizerHybrid<R>::ShardIsPhase(bitLenInt q) const
+{
+    if (!shards[q]) return true;
+    const auto& m = *shards[q];
+    return (norm(m[1]) <= (R)1e-12) && (norm(m[2]) <= (R)1e-12);
+}
+
+template <typename R> void QStabilizerHybrid<R>::ComposeShard(bitLenInt q, const cplx<R>* m)
+{
+    if (!shards[q]) {
+        shards[q] = std::make_unique<std::array<cplx<R>, 4>>();
+        (*shards[q])[0] = cplx<R>(1, 0);
+        (*shards[q])[1] = cplx<R>(0, 0);
+        (*shards[q])[2] = cplx<R>(0, 0);
+        (*shards[q])[3] = cplx<R>(1, 0);
+    }
+    cplx<R> out[4];
+    mul2x2(m, shards[q]->data(), out); // new gate LEFT-multiplies
+    for (int i = 0; i < 4; ++i) (*shards[q])[i] = out[i];
+}
+
+template <typename R> bool QStabilizerHybrid<R>::TryShardFlushClifford(bitLenInt q)
+{
+    if (!shards[q]) return true;
+    if (!stabilizer) return false;
+    try {
+        stabilizer->Mtrx(shards[q]->data(), q);
+        shards[q].reset();
+        return true;
+    } catch (const QrackError&) {
+        return false;
+    }
+}
+
+template <typename R> void QStabilizerHybrid<R>::FlushShard(bitLenInt q)
+{
+    if (!shards[q]) return;
+    if (engine) {
+        engine->Mtrx(shards[q]->data(), q);
+        shards[q].reset();
+        return;
+    }
+    if (!TryShardFlushClifford(q)) {
+        SwitchToEngine();
+        if (shards[q]) {
+            engine->Mtrx(shards[q]->data(), q);
+            shards[q].reset();
+        }
+    }
+}
+
+template <typename R> void QStabilizerHybrid<R>::DumpShardIfPhase(bitLenInt q)
+{
+    // a diagonal shard commutes with Z-basis probability queries
+    if (shards[q] && ShardIsPhase(q)) return;
+    FlushShard(q);
+}
+
+template <typename R> void QStabilizerHybrid<R>::SwitchToEngine()
+{
+    if (engine) return;
+    // materialize the tableau state into a fresh state-vector engine
+    // (parity: qstabilizerhybrid.cpp:435-511)
+    engine = engineFactory(qubitCount, 0u);
+    std::vector<cplx<R>> buf(maxQPower);
+    stabilizer->GetQuantumState(buf.data());
+    engine->SetQuantumState(buf.data());
+    stabilizer.reset();
+    // flush every pending shard into the engine
+    for (bitLenInt q = 0; q < qubitCount; ++q) {
+        if (shards[q]) {
+            engine->Mtrx(shards[q]->data(), q);
+            shards[q].reset();
+        }
+    }
+}
+
+// ---- state ------------------------------------------------------------------
+
+template <typename R> void QStabilizerHybrid<R>::SetPermutation(bitCapInt perm, cplx<R> phase)
+{
+    engine.reset();
+    for (auto& s : shards) s.reset();
+    stabilizer = std::make_shared<QStabilizer<R>>(qubitCount, perm, this->rand_generator);
+}
+
+template <typename R> void QStabilizerHybrid<R>::SetQuantumState(const cplx<R>* inputState)
+{
+    SwitchToEngine();
+    engine->SetQuantumState(inputState);
+}
+
+template <typename R> void QStabilizerHybrid<R>::GetQuantumState(cplx<R>* outputState)
+{
+    if (engine) {
+        for (bitLenInt q = 0; q < qubitCount; ++q) FlushShard(q);
+        engine->GetQuantumState(outputState);
+        return;
+    }
+    // clone so shard flushes don't disturb this instance's lazy state
+    QStabilizerHybridPtr<R> c = std::static_pointer_cast<QStabilizerHybrid<R>>(Clone());
+    c->SwitchToEngine();
+    c->engine->GetQuantumState(outputState);
+}
+
+template <typename R> cplx<R> QStabilizerHybrid<R>::GetAmplitude(bitCapInt perm)
+{
+    bool anyShard = false;
+    for (auto& s : shards) {
+        if (s) anyShard = true;
+    }
+    if (!engine && !anyShard) return stabilizer->GetAmplitude(perm);
+    if (engine && !anyShard) return engine->GetAmplitude(perm);
+    QStabilizerHybridPtr<R> c = std::static_pointer_cast<QStabilizerHybrid<R>>(Clone());
+    c->SwitchToEngine();
+    return c->engine->GetAmplitude(perm);
+}
+
+template <typename R> void QStabilizerHybrid<R>::SetAmplitude(bitCapInt perm, cplx<R> amp)
+{
+    SwitchToEngine();
+    engine->SetAmplitude(perm, amp);
+}
+
+// ---- gates ------------------------------------------------------------------
+
+template <typename R> void QStabilizerHybrid<R>::Mtrx(const cplx<R>* m, bitLenInt t)
+{
+    if (engine) {
+        engine->Mtrx(m, t);
+        return;
+    }
+    ComposeShard(t, m);
+    TryShardFlushClifford(t); // keep the tableau hot when the product is Clifford
+}
+
+template <typename R> void QStabilizerHybrid<R>::Phase(cplx<R> tl, cplx<R> br, bitLenInt t)
+{
+    const cplx<R> m[4] = { tl, cplx<R>(0, 0), cplx<R>(0, 0), br };
+    Mtrx(m, t);
+}
+
+template <typename R> void QStabilizerHybrid<R>::Invert(cplx<R> tr, cplx<R> bl, bitLenInt t)
+{
+    const cplx<R> m[4] = { cplx<R>(0, 0), tr, bl, cplx<R>(0, 0) };
+    Mtrx(m, t);
+}
+
+template <typename R>
+void QStabilizerHybrid<R>::MCMtrx(const std::vector<bitLenInt>& controls, const cplx<R>* m, bitLenInt t)
+{
+    if (controls.empty()) {
+        Mtrx(m, t);
+        return;
+    }
+    if (!engine) {
+        // flush shards touching the gate support, then try the tableau
+        for (bitLenInt c : controls) FlushShard(c);
+        if (!engine) FlushShard(t);
+        if (!engine) {
+            try {
+                stabilizer->MCMtrx(controls, m, t);
+                return;
+            } catch (const QrackError&) {
+                SwitchToEngine();
+            }
+        }
+    }
+    engine->MCMtrx(controls, m, t);
+}
+
+template <typename R>
+void QStabilizerHybrid<R>::MACMtrx(const std::vector<bitLenInt>& controls, const cplx<R>* m, bitLenInt t)
+{
+    if (controls.empty()) {
+        Mtrx(m, t);
+        return;
+    }
+    if (!engine) {
+        for (bitLenInt c : controls) FlushShard(c);
+        if (!engine) FlushShard(t);
+        if (!engine) {
+            try {
+                stabilizer->MACMtrx(controls, m, t);
+                return;
+            } catch (const QrackError&) {
+                SwitchToEngine();
+            }
+        }
+    }
+    engine->MACMtrx(controls, m, t);
+}
+
+template <typename R>
+void QStabilizerHybrid<R>::UCMtrx(
+    const std::vector<bitLenInt>& controls, const cplx<R>* m, bitLenInt t, bitCapInt perm)
+{
+    if (controls.empty()) {
+        Mtrx(m, t);
+        return;
+    }
+    SwitchToEngine();
+    engine->UCMtrx(controls, m, t, perm);
+}
+
+template <typename R>
+void QStabilizerHybrid<R>::UniformlyControlledSingleBit(
+    const std::vector<bitLenInt>& controls, bitLenInt t, const cplx<R>* mtrxs)
+{
+    if (controls.empty()) {
+        Mtrx(mtrxs, t);
+        return;
+    }
+    SwitchToEngine();
+    engine->UniformlyControlledSingleBit(controls, t, mtrxs);
+}
+
+template <typename R> void QStabilizerHybrid<R>::Swap(bitLenInt q1, bitLenInt q2)
+{
+    if (q1 == q2) return;
+    std::swap(shards[q1], shards[q2]);
+    if (engine) {
+        engine->Swap(q1, q2);
+    } else {
+        stabilizer->SwapGate(q1, q2);
+    }
+}
+
+template <typename R> void QStabilizerHybrid<R>::ISwap(bitLenInt q1, bitLenInt q2)
+{
+    FlushShard(q1);
+    if (!engine) FlushShard(q2);
+    if (engine) {
+        engine->ISwap(q1, q2);
+    } else {
+        stabilizer->ISwapGate(q1, q2);
+    }
+}
+
+template <typename R> void QStabilizerHybrid<R>::IISwap(bitLenInt q1, bitLenInt q2)
+{
+    FlushShard(q1);
+    if (!engine) FlushShard(q2);
+    if (engine) {
+        engine->IISwap(q1, q2);
+    } else {
+        stabilizer->IISwapGate(q1, q2);
+    }
+}
+
+// ---- measurement -------------------------------------------------------------
+
+template <typename R> R QStabilizerHybrid<R>::Prob(bitLenInt q)
+{
+    if (!engine && ShardIsPhase(q)) {
+        return stabilizer->Prob(q);
+    }
+    if (!engine && shards[q]) {
+        // general shard on a tableau qubit: apply the shard to the reduced
+        // single-qubit state when separable, else materialize
+        const uint8_t sep = stabilizer->IsSeparable(q);
+        if (sep) {
+            FlushShard(q);
+            if (!engine) return stabilizer->Prob(q);
+            return engine->Prob(q);
+        }
+        SwitchToEngine();
+    }
+    if (engine) {
+        FlushShard(q);
+        return engine->Prob(q);
+    }
+    return stabilizer->Prob(q);
+}
+
+template <typename R>
+bool QStabilizerHybrid<R>::ForceM(bitLenInt q, bool result, bool doForce, bool doApply)
+{
+    if (!engine && !ShardIsPhase(q)) {
+        FlushShard(q);
+    } else if (!engine && shards[q]) {
+        shards[q].reset(); // diagonal shard is global phase after collapse
+    }
+    if (engine) {
+        FlushShard(q);
+        return engine->ForceM(q, result, doForce, doApply);
+    }
+    return stabilizer->ForceM(q, result, doForce, doApply);
+}
+
+template <typename R> bitCapInt QStabilizerHybrid<R>::MAll()
+{
+    bitCapInt result = 0;
+    for (bitLenInt q = 0; q < qubitCount; ++q) {
+        if (ForceM(q, false, false, true)) result |= pow2(q);
+    }
+    if (!engine) {
+        stabilizer->SetPermutation(result);
+    } else {
+        engine->SetPermutation(result);
+    }
+    return result;
+}
+
+template <typename R>
+std::map<bitCapInt, int> QStabilizerHybrid<R>::MultiShotMeasureMask(
+    const std::vector<bitCapInt>& qPowers, unsigned shots)
+{
+    bool anyShard = false;
+    for (auto& s : shards) {
+        if (s && !ShardIsPhase((bitLenInt)(&s - &shards[0]))) anyShard = true;
+    }
+    if (engine && !anyShard) return engine->MultiShotMeasureMask(qPowers, shots);
+    if (!engine && !anyShard) {
+        // tableau sampling: clone + MAll per shot (tableau ops are cheap)
+        std::map<bitCapInt, int> results;
+        for (unsigned s = 0; s < shots; ++s) {
+            QStabilizerPtr<R> c = std::static_pointer_cast<QStabilizer<R>>(stabilizer->Clone());
+            const bitCapInt all = c->MAll();
+            bitCapInt val = 0;
+            for (size_t b = 0; b < qPowers.size(); ++b) {
+                if (all & qPowers[b]) val |= (ONE_BCI << b);
+            }
+            results[val]++;
+        }
+        return results;
+    }
+    QStabilizerHybridPtr<R> c = std::static_pointer_cast<QStabilizerHybrid<R>>(Clone());
+    c->SwitchToEngine();
+    return c->engine->MultiShotMeasureMask(qPowers, shots);
+}
+
+template <typename R> R QStabilizerHybrid<R>::ProbMask(bitCapInt mask, bitCapInt permutation)
+{
+    QStabilizerHybridPtr<R> c = std::static_pointer_cast<QStabilizerHybrid<R>>(Clone());
+    c->SwitchToEngine();
+    return c->engine->ProbMask(mask, permutation);
+}
+
+// ---- structural ---------------------------------------------------------------
+
+template <typename R> bitLenInt QStabilizerHybrid<R>::Compose(QInterfacePtr<R> toCopy, bitLenInt start)
+{
+    QStabilizerHybrid<R>* o = dynamic_cast<QStabilizerHybrid<R>*>(toCopy.get());
+    const bitLenInt oQubits = toCopy->GetQubitCount();
+    const bitLenInt nQubits = qubitCount + oQubits;
+    if (o && !engine && !o->engine) {
+        stabilizer->Compose(o->stabilizer, start);
+        for (bitLenInt q = 0; q < oQubits; ++q) shards.emplace(shards.begin() + start);
+        for (bitLenInt q = 0; q < oQubits; ++q) {
+            if (o->shards[q]) {
+                shards[start + q] = std::make_unique<std::array<cplx<R>, 4>>(*o->shards[q]);
+            }
+        }
+        this->SetQubitCount(nQubits);
+        return start;
+    }
+    SwitchToEngine();
+    if (o) {
+        QStabilizerHybridPtr<R> oc = std::static_pointer_cast<QStabilizerHybrid<R>>(o->Clone());
+        oc->SwitchToEngine();
+        engine->Compose(oc->engine, start);
+    } else {
+        engine->Compose(toCopy, start);
+    }
+    for (bitLenInt q = 0; q < oQubits; ++q) shards.emplace(shards.begin() + start);
+    this->SetQubitCount(nQubits);
+    return start;
+}
+
+template <typename R> void QStabilizerHybrid<R>::Decompose(bitLenInt start, QInterfacePtr<R> dest)
+{
+    const bitLenInt len = dest->GetQubitCount();
+    QStabilizerHybrid<R>* o = dynamic_cast<QStabilizerHybrid<R>*>(dest.get());
+    if (!engine && o && stabilizer->CanDecomposeDispose(start, len)) {
+        for (bitLenInt q = start; q < start + len; ++q) FlushShard(q);
+        if (!engine) {
+            o->engine.reset();
+            o->stabilizer = std::make_shared<QStabilizer<R>>(len, 0u, this->rand_generator);
+            stabilizer->Decompose(start, o->stabilizer);
+            for (auto& s : o->shards) s.reset();
+            shards.erase(shards.begin() + start, shards.begin() + start + len);
+            this->SetQubitCount(qubitCount - len);
+            return;
+        }
+    }
+    SwitchToEngine();
+    if (o) {
+        o->SwitchToEngine();
+        engine->Decompose(start, o->engine);
+    } else {
+        engine->Decompose(start, dest);
+    }
+    shards.erase(shards.begin() + start, shards.begin() + start + len);
+    this->SetQubitCount(qubitCount - len);
+}
+
+template <typename R> void QStabilizerHybrid<R>::Dispose(bitLenInt start, bitLenInt length)
+{
+    for (bitLenInt q = start; q < start + length; ++q) FlushShard(q);
+    if (!engine && stabilizer->CanDecomposeDispose(start, length)) {
+        stabilizer->Dispose(start, length);
+    } else {
+        SwitchToEngine();
+        engine->Dispose(start, length);
+    }
+    shards.erase(shards.begin() + start, shards.begin() + start + length);
+    this->SetQubitCount(qubitCount - length);
+}
+
+template <typename R>
+void QStabilizerHybrid<R>::Dispose(bitLenInt start, bitLenInt length, bitCapInt disposedPerm)
+{
+    for (bitLenInt q = start; q < start + length; ++q) FlushShard(q);
+    if (!engine) {
+        stabilizer->Dispose(start, length, disposedPerm);
+    } else {
+        engine->Dispose(start, length, disposedPerm);
+    }
+    shards.erase(shards.begin() + start, shards.begin() + start + length);
+    this->SetQubitCount(qubitCount - length);
+}
+
+template <typename R> bitLenInt QStabilizerHybrid<R>::Allocate(bitLenInt start, bitLenInt length)
+{
+    if (!length) return start;
+    if (!engine) {
+        stabilizer->Allocate(start, length);
+    } else {
+        engine->Allocate(start, length);
+    }
+    for (bitLenInt q = 0; q < length; ++q) shards.emplace(shards.begin() + start);
+    this->SetQubitCount(qubitCount + length);
+    return start;
+}
+
+template <typename R> QInterfacePtr<R> QStabilizerHybrid<R>::Clone()
+{
+    auto clone = std::make_shared<QStabilizerHybrid<R>>(
+        qubitCount, 0u, this->rand_generator, engineFactory);
+    if (engine) {
+        clone->stabilizer.reset();
+        clone->engine = engine->Clone();
+    } else {
+        clone->stabilizer = std::static_pointer_cast<QStabilizer<R>>(stabilizer->Clone());
+        clone->engine.reset();
+    }
+    for (bitLenInt q = 0; q < qubitCount; ++q) {
+        if (shards[q]) {
+            clone->shards[q] = std::make_unique<std::array<cplx<R>, 4>>(*shards[q]);
+        }
+    }
+    return clone;
+}
+
+// ---- norm ---------------------------------------------------------------------
+
+template <typename R> void QStabilizerHybrid<R>::UpdateRunningNorm(R norm_thresh)
+{
+    if (engine) engine->UpdateRunningNorm(norm_thresh);
+}
+
+template <typename R> void QStabilizerHybrid<R>::NormalizeState(R nrm, R norm_thresh, R phaseArg)
+{
+    if (engine) engine->NormalizeState(nrm, norm_thresh, phaseArg);
+}
+
+template <typename R> double QStabilizerHybrid<R>::SumSqrDiff(QInterfacePtr<R> other)
+{
+    if (other->GetQubitCount() != qubitCount) return 2.0;
+    if (qubitCount > 24u) throw QrackError("SumSqrDiff: too wide for dense compare");
+    std::vector<cplx<R>> a(maxQPower), b(maxQPower);
+    GetQuantumState(a.data());
+    other->GetQuantumState(b.data());
+    double re = 0, im = 0;
+    for (bitCapInt i = 0; i < maxQPower; ++i) {
+        re += (double)(b[i].re * a[i].re + b[i].im * a[i].im);
+        im += (double)(b[i].re * a[i].im - b[i].im * a[i].re);
+    }
+    return std::max(0.0, 2.0 - 2.0 * std::sqrt(re * re + im * im));
+}
+
+template class QStabilizerHybrid<float>;
+template class QStabilizerHybrid<double>;
+
+} // namespace qrack_amd

@@ -1,0 +1,152 @@
+// qrack_amd — hybrid stabilizer / state-vector layer.
+//
+// Capability parity target: /root/reference/include/qstabilizerhybrid.hpp +
+// src/qstabilizerhybrid.cpp: run on the CHP tableau until a non-Clifford
+// operation forces SwitchToEngine(); per-qubit MpsShard 2x2 buffers absorb
+// non-Clifford single-qubit gates so that sequences that multiply back to
+// Clifford never leave the tableau (reference: include/mpsshard.hpp).
+#pragma once
+
+#include "qstabilizer.hpp"
+
+#include <array>
+#include <functional>
+
+namespace qrack_amd {
+
+template <typename R>
+using EngineFactoryFn = std::function<QInterfacePtr<R>(bitLenInt qubits, bitCapInt perm)>;
+
+template <typename R> class QStabilizerHybrid;
+template <typename R> using QStabilizerHybridPtr = std::shared_ptr<QStabilizerHybrid<R>>;
+
+template <typename R> class QStabilizerHybrid : public QInterface<R> {
+protected:
+    using QInterface<R>::qubitCount;
+    using QInterface<R>::maxQPower;
+
+    QStabilizerPtr<R> stabilizer; // non-null while in Clifford mode
+    QInterfacePtr<R> engine;      // non-null after the switch
+    EngineFactoryFn<R> engineFactory;
+    // per-qubit buffered 2x2 (identity when absent)
+    std::vector<std::unique_ptr<std::array<cplx<R>, 4>>> shards;
+
+    bool InEngineMode() const { return (bool)engine; }
+    void FlushShard(bitLenInt q);        // apply buffered 2x2 to the active backend
+    void DumpShardIfPhase(bitLenInt q);  // drop diagonal shards (safe before Z ops)
+    bool ShardIsPhase(bitLenInt q) const;
+    bool ShardIsIdentity(bitLenInt q) const;
+    void ComposeShard(bitLenInt q, const cplx<R>* m);
+    bool TryShardFlushClifford(bitLenInt q); // flush if the shard became Clifford
+
+public:
+    QStabilizerHybrid(bitLenInt qBitCount, bitCapInt initState = 0u, RngPtr rgp = nullptr,
+        EngineFactoryFn<R> factory = nullptr, bool doNorm = true, R normThresh = eps<R>::value);
+
+    void SwitchToEngine();
+    bool isClifford() const override { return !InEngineMode(); }
+    bool isClifford(bitLenInt q) const override
+    {
+        return !InEngineMode() && ShardIsIdentity(q);
+    }
+    QInterfacePtr<R> ActiveBackend() { return engine ? engine : std::static_pointer_cast<QInterface<R>>(stabilizer); }
+
+    // ---- state ----
+    void SetPermutation(bitCapInt perm, cplx<R> phase = cplx<R>((R)1, (R)0)) override;
+    void SetQuantumState(const cplx<R>* inputState) override;
+    void GetQuantumState(cplx<R>* outputState) override;
+    cplx<R> GetAmplitude(bitCapInt perm) override;
+    void SetAmplitude(bitCapInt perm, cplx<R> amp) override;
+
+    // ---- gates ----
+    void Mtrx(const cplx<R>* mtrx, bitLenInt target) override;
+    void Phase(cplx<R> topLeft, cplx<R> bottomRight, bitLenInt target) override;
+    void Invert(cplx<R> topRight, cplx<R> bottomLeft, bitLenInt target) override;
+    void MCMtrx(const std::vector<bitLenInt>& controls, const cplx<R>* mtrx, bitLenInt target) override;
+    void MACMtrx(const std::vector<bitLenInt>& controls, const cplx<R>* mtrx, bitLenInt target) override;
+    void UCMtrx(const std::vector<bitLenInt>& controls, const cplx<R>* mtrx, bitLenInt target,
+        bitCapInt controlPerm) override;
+    void UniformlyControlledSingleBit(
+        const std::vector<bitLenInt>& controls, bitLenInt target, const cplx<R>* mtrxs) override;
+    void Swap(bitLenInt q1, bitLenInt q2) override;
+    void ISwap(bitLenInt q1, bitLenInt q2) override;
+    void IISwap(bitLenInt q1, bitLenInt q2) override;
+
+    // ---- measurement ----
+    R Prob(bitLenInt q) override;
+    bool ForceM(bitLenInt q, bool result, bool doForce = true, bool doApply = true) override;
+    bitCapInt MAll() override;
+    std::map<bitCapInt, int> MultiShotMeasureMask(
+        const std::vector<bitCapInt>& qPowers, unsigned shots) override;
+    R ProbMask(bitCapInt mask, bitCapInt permutation) override;
+
+    // ---- structural ----
+    using QInterface<R>::Compose;
+    bitLenInt Compose(QInterfacePtr<R> toCopy, bitLenInt start) override;
+    void Decompose(bitLenInt start, QInterfacePtr<R> dest) override;
+    void Dispose(bitLenInt start, bitLenInt length) override;
+    void Dispose(bitLenInt start, bitLenInt length, bitCapInt disposedPerm) override;
+    bitLenInt Allocate(bitLenInt start, bitLenInt length) override;
+    QInterfacePtr<R> Clone() override;
+
+    // ---- norm ----
+    void UpdateRunningNorm(R norm_thresh = (R)-1) override;
+    void NormalizeState(R nrm = (R)-1, R norm_thresh = (R)-1, R phaseArg = 0) override;
+    double SumSqrDiff(QInterfacePtr<R> other) override;
+    void Finish() override
+    {
+        if (engine) engine->Finish();
+    }
+
+    // ---- ALU: engine-only (switch on demand) ----
+    void INC(bitCapInt toAdd, bitLenInt start, bitLenInt length) override
+    {
+        SwitchToEngine();
+        engine->INC(toAdd, start, length);
+    }
+    void CINC(bitCapInt toAdd, bitLenInt start, bitLenInt length,
+        const std::vector<bitLenInt>& controls) override
+    {
+        SwitchToEngine();
+        engine->CINC(toAdd, start, length, controls);
+    }
+    void INCC(bitCapInt toAdd, bitLenInt start, bitLenInt length, bitLenInt carryIndex) override
+    {
+        SwitchToEngine();
+        engine->INCC(toAdd, start, length, carryIndex);
+    }
+    void MUL(bitCapInt toMul, bitLenInt inOutStart, bitLenInt carryStart, bitLenInt length) override
+    {
+        SwitchToEngine();
+        engine->MUL(toMul, inOutStart, carryStart, length);
+    }
+    void DIV(bitCapInt toDiv, bitLenInt inOutStart, bitLenInt carryStart, bitLenInt length) override
+    {
+        SwitchToEngine();
+        engine->DIV(toDiv, inOutStart, carryStart, length);
+    }
+    void MULModNOut(bitCapInt toMul, bitCapInt modN, bitLenInt inStart, bitLenInt outStart,
+        bitLenInt length) override
+    {
+        SwitchToEngine();
+        engine->MULModNOut(toMul, modN, inStart, outStart, length);
+    }
+    void POWModNOut(bitCapInt base, bitCapInt modN, bitLenInt inStart, bitLenInt outStart,
+        bitLenInt length) override
+    {
+        SwitchToEngine();
+        engine->POWModNOut(base, modN, inStart, outStart, length);
+    }
+    void PhaseFlipIfLess(bitCapInt greaterPerm, bitLenInt start, bitLenInt length) override
+    {
+        SwitchToEngine();
+        engine->PhaseFlipIfLess(greaterPerm, start, length);
+    }
+    void Hash(bitLenInt start, bitLenInt length, const unsigned char* values) override
+    {
+        SwitchToEngine();
+        engine->Hash(start, length, values);
+    }
+};
+
+} // namespace qrack_amd

@@ -71,20 +71,38 @@ class DistQPager:
 
     def _shuffle(self, partner_rank, i_am_low):
         """Swap my (upper if low page else lower) half with the partner's
-        opposite half — the reference's cross-device ShuffleBuffers, as an
-        RCCL sendrecv pair over xGMI."""
+        opposite half — the reference's cross-device ShuffleBuffers
+        (opencl.cpp:254-264, staged through HOST there), as an RCCL sendrecv
+        pair over xGMI on zero-copy HBM views here. Non-NCCL backends (gloo
+        CI without GPUs, or single-GPU validation) stage through host."""
         self._sync_engine()
-        view = self._half_view(low_half=not i_am_low)
-        tmp = torch.empty_like(view)
-        ops = [
-            dist.P2POp(dist.isend, view, partner_rank),
-            dist.P2POp(dist.irecv, tmp, partner_rank),
-        ]
-        reqs = dist.batch_isend_irecv(ops)
-        for r in reqs:
-            r.wait()
-        view.copy_(tmp)
-        self._sync_torch()
+        nccl = dist.get_backend() == "nccl"
+        if nccl or not self._is_hip():
+            view = self._half_view(low_half=not i_am_low)
+            tmp = torch.empty_like(view)
+            ops = [
+                dist.P2POp(dist.isend, view, partner_rank),
+                dist.P2POp(dist.irecv, tmp, partner_rank),
+            ]
+            reqs = dist.batch_isend_irecv(ops)
+            for r in reqs:
+                r.wait()
+            view.copy_(tmp)
+            self._sync_torch()
+        else:
+            off = self.page_len // 2 if i_am_low else 0
+            buf = self.q.get_amplitude_page(off, self.page_len // 2)
+            send = torch.from_numpy(buf)
+            tmp = torch.empty_like(send)
+            reqs = dist.batch_isend_irecv(
+                [
+                    dist.P2POp(dist.isend, send, partner_rank),
+                    dist.P2POp(dist.irecv, tmp, partner_rank),
+                ]
+            )
+            for r in reqs:
+                r.wait()
+            self.q.set_amplitude_page(tmp.numpy(), off)
 
     def _split_controls(self, controls):
         local = [c for c in controls if c < self.qpp]

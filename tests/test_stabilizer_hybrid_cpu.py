@@ -1,0 +1,185 @@
+"""QStabilizerHybrid tests: Clifford stays on the tableau, non-Clifford
+falls through to the state-vector engine; results always match dense.
+
+Parity model: /root/reference/src/qstabilizerhybrid.cpp behavior
+(SwitchToEngine, MpsShard gate fusion).
+"""
+
+import numpy as np
+import pytest
+
+import qrack_amd as qa
+from ref_sim import assert_states_close
+
+
+def make_h(n, seed=7):
+    return qa.create_simulator(n, layers=["stabilizer_hybrid", "cpu"], seed=seed)
+
+
+def make_cpu(n, seed=7):
+    return qa.create_simulator(n, engine="cpu", seed=seed)
+
+
+def test_stays_clifford():
+    q = make_h(4)
+    q.h(0)
+    q.cnot(0, 1)
+    q.s(1)
+    q.cz(1, 2)
+    assert q.is_clifford()
+
+
+def test_t_then_it_stays_clifford():
+    # shard fusion: T then T-dagger multiplies to identity
+    q = make_h(2)
+    q.h(0)
+    q.t(0)
+    assert q.is_clifford()  # T is buffered, not switched
+    q.it(0)
+    q.cnot(0, 1)  # forces shard flush; product was identity -> still Clifford
+    assert q.is_clifford()
+
+
+def test_t_t_makes_s():
+    q = make_h(1)
+    q.h(0)
+    q.t(0)
+    q.t(0)
+    # T*T = S: Clifford again; flush happens transparently
+    sv = q.get_state_vector()
+    cp = make_cpu(1)
+    cp.h(0)
+    cp.s(0)
+    assert_states_close(sv, cp.get_state_vector(), 1e-5)
+    assert q.is_clifford()
+
+
+def test_non_clifford_switches():
+    q = make_h(3, seed=5)
+    q.h(0)
+    q.cnot(0, 1)
+    q.t(1)
+    q.cnot(1, 2)  # flush forces the engine switch
+    assert not q.is_clifford()
+    cp = make_cpu(3)
+    cp.h(0)
+    cp.cnot(0, 1)
+    cp.t(1)
+    cp.cnot(1, 2)
+    assert_states_close(q.get_state_vector(), cp.get_state_vector(), 1e-5)
+
+
+@pytest.mark.parametrize("seed", [1, 2, 3])
+def test_random_mixed_circuit_vs_dense(seed):
+    n = 5
+    rng = np.random.default_rng(seed)
+    q = make_h(n, seed=seed)
+    cp = make_cpu(n, seed=seed)
+    gates1 = ["h", "s", "x", "z", "t", "it"]
+    for _ in range(30):
+        r = rng.random()
+        if r < 0.6:
+            g = gates1[rng.integers(len(gates1))]
+            t = int(rng.integers(n))
+            getattr(q, g)(t)
+            getattr(cp, g)(t)
+        elif r < 0.9:
+            a, b = rng.choice(n, 2, replace=False)
+            q.cnot(int(a), int(b))
+            cp.cnot(int(a), int(b))
+        else:
+            th = float(rng.uniform(0, 2 * np.pi))
+            t = int(rng.integers(n))
+            q.ry(th, t)
+            cp.ry(th, t)
+    assert_states_close(q.get_state_vector(), cp.get_state_vector(), 1e-4)
+
+
+def test_measurement_clifford_mode():
+    q = make_h(2, seed=4)
+    q.h(0)
+    q.cnot(0, 1)
+    r0 = q.m(0)
+    r1 = q.m(1)
+    assert r0 == r1
+    assert q.is_clifford()
+
+
+def test_measurement_after_switch():
+    q = make_h(2, seed=4)
+    q.h(0)
+    q.t(0)
+    q.h(0)  # H T H is non-Clifford; shard product non-Clifford
+    q.cnot(0, 1)
+    p = q.prob(1)
+    assert 0.0 <= p <= 1.0
+    cp = make_cpu(2)
+    cp.h(0)
+    cp.t(0)
+    cp.h(0)
+    cp.cnot(0, 1)
+    assert abs(p - cp.prob(1)) < 1e-5
+
+
+def test_multishot_clifford():
+    q = make_h(3, seed=9)
+    q.h(0)
+    q.cnot(0, 1)
+    res = q.multi_shot_measure_mask([1, 2], 300)
+    assert sum(res.values()) == 300
+    assert set(res.keys()) <= {0, 3}
+
+
+def test_mall():
+    q = make_h(4, seed=3)
+    q.h(0)
+    q.cnot(0, 3)
+    r = q.m_all()
+    assert r in (0, 9)
+
+
+def test_compose_clifford():
+    a = make_h(1, seed=1)
+    a.h(0)
+    b = make_h(1, seed=2)
+    b.x(0)
+    a.compose(b)
+    assert a.num_qubits == 2
+    assert a.is_clifford()
+    assert abs(a.prob(1) - 1.0) < 1e-6
+
+
+def test_qv_style_circuit():
+    """Quantum-volume-shaped workload (the reference QV benchmark protocol,
+    benchmarks.cpp quantum volume case): random SU(4)-ish layers."""
+    n = 5
+    rng = np.random.default_rng(17)
+    q = make_h(n, seed=17)
+    cp = make_cpu(n, seed=17)
+    for depth in range(n):
+        perm = rng.permutation(n)
+        for k in range(0, n - 1, 2):
+            a, b = int(perm[k]), int(perm[k + 1])
+            for t in (a, b):
+                th, ph, lm = rng.uniform(0, 2 * np.pi, 3)
+                q.u(t, th, ph, lm)
+                cp.u(t, th, ph, lm)
+            q.cnot(a, b)
+            cp.cnot(a, b)
+    assert_states_close(q.get_state_vector(), cp.get_state_vector(), 1e-4)
+    # terminal sampling agreement in distribution
+    res = q.multi_shot_measure_mask([1 << i for i in range(n)], 200)
+    assert sum(res.values()) == 200
+
+
+def test_clone_preserves_mode():
+    q = make_h(2, seed=6)
+    q.h(0)
+    c = q.clone()
+    assert c.is_clifford()
+    q.t(0)
+    q.ry(0.3, 0)
+    c2 = q.clone()
+    sv1 = q.get_state_vector()
+    sv2 = c2.get_state_vector()
+    assert np.allclose(sv1, sv2, atol=1e-6)
