@@ -2,6 +2,7 @@
 
 #include "qstabilizer.hpp"
 #include "qstabilizerhybrid.hpp"
+#include "qunit.hpp"
 
 #ifdef QRACK_AMD_ENABLE_HIP
 #include <hip/hip_runtime.h>
@@ -49,6 +50,12 @@ EngineFactoryFn<R> LayerFactory(
         EngineFactoryFn<R> sub = LayerFactory<R>(tail, rng, deviceId, pagesPerDevice);
         return [rng, sub](bitLenInt n, bitCapInt perm) -> QInterfacePtr<R> {
             return std::make_shared<QStabilizerHybrid<R>>(n, perm, rng, sub);
+        };
+    }
+    if (head == "qunit") {
+        EngineFactoryFn<R> sub = LayerFactory<R>(tail, rng, deviceId, pagesPerDevice);
+        return [rng, sub](bitLenInt n, bitCapInt perm) -> QInterfacePtr<R> {
+            return std::make_shared<QUnit<R>>(n, perm, rng, sub);
         };
     }
     throw QrackError("unknown layer: " + head);

@@ -1,0 +1,150 @@
+// qrack_amd — QUnit: Schmidt-decomposition layer.
+//
+// Capability parity target: /root/reference/include/qunit.hpp +
+// src/qunit.cpp (per-qubit shards, lazy entanglement, TrySeparate
+// Bloch tomography with post-selection rounding, fidelity bookkeeping).
+// Fresh design for round 1: every logical qubit points at a sub-unit
+// (QInterfacePtr, built by the sub-stack factory) plus its index inside it;
+// separable qubits live in width-1 units. Swap of qubits in different
+// units is a pure label swap (reference: qubitswapmap.hpp). Controlled
+// gates short-circuit on deterministic controls (probability 0/1 controls
+// are exactly identity/unconditioned — valid for any state).
+#pragma once
+
+#include "qstabilizerhybrid.hpp"
+
+namespace qrack_amd {
+
+template <typename R> class QUnit;
+template <typename R> using QUnitPtr = std::shared_ptr<QUnit<R>>;
+
+template <typename R> class QUnit : public QInterface<R> {
+protected:
+    using QInterface<R>::qubitCount;
+    using QInterface<R>::maxQPower;
+
+    struct Shard {
+        QInterfacePtr<R> unit;
+        bitLenInt mapped = 0;
+    };
+    std::vector<Shard> shards;
+    EngineFactoryFn<R> subFactory;
+    double logFidelity = 0.0;
+    R separabilityThreshold;
+
+    QInterfacePtr<R> MakeUnit(bitLenInt n, bitCapInt perm)
+    {
+        return subFactory(n, perm);
+    }
+
+    // merge all units containing `qs` into one; returns it
+    QInterfacePtr<R> EntangleAll(const std::vector<bitLenInt>& qs);
+    // EntangleAll + in-unit swaps so shard[qs[i]].mapped == i
+    QInterfacePtr<R> EntangleOrdered(const std::vector<bitLenInt>& qs);
+    // remove one measured qubit from its (multi-qubit) unit
+    void SeparateBit(bitLenInt q, bool value);
+    void FixMappedAfterRemoval(QInterfacePtr<R> unit, bitLenInt removedMapped);
+    std::vector<bitLenInt> UnitQubits(QInterfacePtr<R> unit) const; // logical qubits of a unit
+    bool ControlShortcut(bitLenInt control, bool anti, bool& alwaysOn);
+
+public:
+    QUnit(bitLenInt qBitCount, bitCapInt initState = 0u, RngPtr rgp = nullptr,
+        EngineFactoryFn<R> factory = nullptr, bool doNorm = true, R normThresh = eps<R>::value);
+
+    double GetUnitaryFidelity() override { return std::exp(logFidelity); }
+    void ResetUnitaryFidelity() override { logFidelity = 0.0; }
+
+    // ---- state ----
+    void SetPermutation(bitCapInt perm, cplx<R> phase = cplx<R>((R)1, (R)0)) override;
+    void SetQuantumState(const cplx<R>* inputState) override;
+    void GetQuantumState(cplx<R>* outputState) override;
+    cplx<R> GetAmplitude(bitCapInt perm) override;
+    void SetAmplitude(bitCapInt perm, cplx<R> amp) override;
+
+    // ---- gates ----
+    void Mtrx(const cplx<R>* mtrx, bitLenInt target) override;
+    void Phase(cplx<R> topLeft, cplx<R> bottomRight, bitLenInt target) override;
+    void Invert(cplx<R> topRight, cplx<R> bottomLeft, bitLenInt target) override;
+    void MCMtrx(const std::vector<bitLenInt>& controls, const cplx<R>* mtrx, bitLenInt target) override;
+    void MACMtrx(const std::vector<bitLenInt>& controls, const cplx<R>* mtrx, bitLenInt target) override;
+    void MCPhase(const std::vector<bitLenInt>& controls, cplx<R> topLeft, cplx<R> bottomRight,
+        bitLenInt target) override;
+    void MCInvert(const std::vector<bitLenInt>& controls, cplx<R> topRight, cplx<R> bottomLeft,
+        bitLenInt target) override;
+    void UCMtrx(const std::vector<bitLenInt>& controls, const cplx<R>* mtrx, bitLenInt target,
+        bitCapInt controlPerm) override;
+    void UniformlyControlledSingleBit(
+        const std::vector<bitLenInt>& controls, bitLenInt target, const cplx<R>* mtrxs) override;
+    void Swap(bitLenInt q1, bitLenInt q2) override;
+    void ISwap(bitLenInt q1, bitLenInt q2) override;
+    void IISwap(bitLenInt q1, bitLenInt q2) override;
+    void SqrtSwap(bitLenInt q1, bitLenInt q2) override;
+    void ISqrtSwap(bitLenInt q1, bitLenInt q2) override;
+    void FSim(R theta, R phi, bitLenInt q1, bitLenInt q2) override;
+
+    // ---- measurement ----
+    R Prob(bitLenInt q) override;
+    bool ForceM(bitLenInt q, bool result, bool doForce = true, bool doApply = true) override;
+    bitCapInt MAll() override;
+    std::map<bitCapInt, int> MultiShotMeasureMask(
+        const std::vector<bitCapInt>& qPowers, unsigned shots) override;
+    R ProbMask(bitCapInt mask, bitCapInt permutation) override;
+    R ProbParity(bitCapInt mask) override;
+    bool ForceMParity(bitCapInt mask, bool result, bool doForce = true) override;
+    double ExpectationBitsFactorized(const std::vector<bitLenInt>& bits,
+        const std::vector<bitCapInt>& perms, bitCapInt offset = 0) override;
+
+    // ---- separability ----
+    bool TrySeparate(bitLenInt q) override;
+    bool TrySeparate(bitLenInt q1, bitLenInt q2) override;
+    bool TrySeparate(const std::vector<bitLenInt>& qubits, R error_tol) override;
+
+    // ---- structural ----
+    using QInterface<R>::Compose;
+    bitLenInt Compose(QInterfacePtr<R> toCopy, bitLenInt start) override;
+    void Decompose(bitLenInt start, QInterfacePtr<R> dest) override;
+    void Dispose(bitLenInt start, bitLenInt length) override;
+    void Dispose(bitLenInt start, bitLenInt length, bitCapInt disposedPerm) override;
+    bitLenInt Allocate(bitLenInt start, bitLenInt length) override;
+    QInterfacePtr<R> Clone() override;
+
+    // ---- norm ----
+    void UpdateRunningNorm(R norm_thresh = (R)-1) override;
+    void NormalizeState(R nrm = (R)-1, R norm_thresh = (R)-1, R phaseArg = 0) override;
+    double SumSqrDiff(QInterfacePtr<R> other) override;
+    void Finish() override;
+    bool isFinished() override;
+
+    // ---- ALU: entangle the affected registers then forward ----
+    void INC(bitCapInt toAdd, bitLenInt start, bitLenInt length) override;
+    void CINC(bitCapInt toAdd, bitLenInt start, bitLenInt length,
+        const std::vector<bitLenInt>& controls) override;
+    void INCC(bitCapInt toAdd, bitLenInt start, bitLenInt length, bitLenInt carryIndex) override;
+    void DECC(bitCapInt toSub, bitLenInt start, bitLenInt length, bitLenInt carryIndex) override;
+    void INCS(bitCapInt toAdd, bitLenInt start, bitLenInt length, bitLenInt overflowIndex) override;
+    void MUL(bitCapInt toMul, bitLenInt inOutStart, bitLenInt carryStart, bitLenInt length) override;
+    void DIV(bitCapInt toDiv, bitLenInt inOutStart, bitLenInt carryStart, bitLenInt length) override;
+    void MULModNOut(bitCapInt toMul, bitCapInt modN, bitLenInt inStart, bitLenInt outStart,
+        bitLenInt length) override;
+    void IMULModNOut(bitCapInt toMul, bitCapInt modN, bitLenInt inStart, bitLenInt outStart,
+        bitLenInt length) override;
+    void POWModNOut(bitCapInt base, bitCapInt modN, bitLenInt inStart, bitLenInt outStart,
+        bitLenInt length) override;
+    void CMULModNOut(bitCapInt toMul, bitCapInt modN, bitLenInt inStart, bitLenInt outStart,
+        bitLenInt length, const std::vector<bitLenInt>& controls) override;
+    void CPOWModNOut(bitCapInt base, bitCapInt modN, bitLenInt inStart, bitLenInt outStart,
+        bitLenInt length, const std::vector<bitLenInt>& controls) override;
+    void PhaseFlipIfLess(bitCapInt greaterPerm, bitLenInt start, bitLenInt length) override;
+    void CPhaseFlipIfLess(
+        bitCapInt greaterPerm, bitLenInt start, bitLenInt length, bitLenInt flagIndex) override;
+    void Hash(bitLenInt start, bitLenInt length, const unsigned char* values) override;
+    bitCapInt IndexedLDA(bitLenInt indexStart, bitLenInt indexLength, bitLenInt valueStart,
+        bitLenInt valueLength, const unsigned char* values, bool resetValue = true) override;
+
+protected:
+    // helpers for ALU forwarding: entangle the given logical registers into
+    // one unit, ordered so each register is contiguous from position 0
+    QInterfacePtr<R> EntangleRegisters(const std::vector<std::pair<bitLenInt, bitLenInt>>& regs);
+};
+
+} // namespace qrack_amd

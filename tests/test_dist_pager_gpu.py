@@ -98,20 +98,41 @@ def test_measure_hip_world2():
     run_distributed("_body_measure_hip", world=2, qubits=5, port_off=3)
 
 
-def test_dlpack_view_roundtrip():
-    import torch
+_DLPACK_BODY = """
+import torch
+torch.cuda.init()   # torch first, engine second — bench.py ordering
+import sys
+sys.path.insert(0, {repo!r})
+import numpy as np
+import qrack_amd as qa
+q = qa.create_simulator(4, engine="hip", seed=3)
+q.h(0)
+q.cnot(0, 1)
+cap = q.dlpack_view(0, 16)
+t = torch.from_dlpack(cap)
+assert t.shape == (16,)
+assert t.is_cuda
+sv = np.asarray(q.get_state_vector())
+assert np.allclose(t.cpu().numpy(), sv, atol=1e-6)
+t[3] = 0.5 + 0.25j
+torch.cuda.synchronize()
+amp = q.get_amplitude(3)
+assert abs(amp - (0.5 + 0.25j)) < 1e-6
+print("DLPACK_OK")
+"""
 
-    q = qa.create_simulator(4, engine="hip", seed=3)
-    q.h(0)
-    q.cnot(0, 1)
-    cap = q.dlpack_view(0, 16)
-    t = torch.from_dlpack(cap)
-    assert t.shape == (16,)
-    assert t.is_cuda  # ROCm reports as cuda device in torch
-    sv = np.asarray(q.get_state_vector())
-    assert np.allclose(t.cpu().numpy(), sv, atol=1e-6)
-    # mutate through the view; engine must see it
-    t[3] = 0.5 + 0.25j
-    torch.cuda.synchronize()
-    amp = q.get_amplitude(3)
-    assert abs(amp - (0.5 + 0.25j)) < 1e-6
+
+def test_dlpack_view_roundtrip():
+    # fresh process: torch's HIP runtime must win the init race (same order
+    # bench.py uses: torch/process-group first, engines second)
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, "-c", _DLPACK_BODY.format(repo=repo)],
+        capture_output=True,
+        text=True,
+        timeout=240,
+    )
+    assert "DLPACK_OK" in out.stdout, f"stdout={out.stdout}\nstderr={out.stderr}"

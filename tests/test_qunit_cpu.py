@@ -1,0 +1,237 @@
+"""QUnit (Schmidt decomposition) tests.
+
+Parity model: /root/reference/src/qunit.cpp behavior — lazy entanglement,
+label-swap Swap, measurement separation, TrySeparate tomography — validated
+against the dense CPU engine and the full canonical stack
+["qunit", "stabilizer_hybrid", "cpu"].
+"""
+
+import numpy as np
+import pytest
+
+import qrack_amd as qa
+from ref_sim import assert_states_close
+
+STACKS = [["qunit", "cpu"], ["qunit", "stabilizer_hybrid", "cpu"]]
+
+
+def make(n, layers, seed=7):
+    return qa.create_simulator(n, layers=layers, seed=seed)
+
+
+def make_cpu(n, seed=7):
+    return qa.create_simulator(n, engine="cpu", seed=seed)
+
+
+@pytest.mark.parametrize("layers", STACKS)
+def test_separable_gates(layers):
+    q = make(4, layers)
+    cp = make_cpu(4)
+    for i in range(4):
+        q.h(i)
+        cp.h(i)
+    q.t(2)
+    cp.t(2)
+    q.rz(0.7, 1)
+    cp.rz(0.7, 1)
+    assert_states_close(q.get_state_vector(), cp.get_state_vector(), 1e-4)
+
+
+@pytest.mark.parametrize("layers", STACKS)
+def test_entangling(layers):
+    q = make(3, layers)
+    cp = make_cpu(3)
+    q.h(0)
+    cp.h(0)
+    q.cnot(0, 1)
+    cp.cnot(0, 1)
+    q.cnot(1, 2)
+    cp.cnot(1, 2)
+    assert_states_close(q.get_state_vector(), cp.get_state_vector(), 1e-4)
+    assert abs(q.prob(2) - 0.5) < 1e-5
+
+
+@pytest.mark.parametrize("layers", STACKS)
+def test_classical_control_shortcut(layers):
+    # control |0>: no entanglement should be created; still correct
+    q = make(3, layers)
+    cp = make_cpu(3)
+    q.cnot(0, 1)  # control is |0>: no-op
+    cp.cnot(0, 1)
+    q.x(0)
+    cp.x(0)
+    q.cnot(0, 2)  # control is |1>: unconditional X
+    cp.cnot(0, 2)
+    assert_states_close(q.get_state_vector(), cp.get_state_vector(), 1e-5)
+
+
+@pytest.mark.parametrize("layers", STACKS)
+def test_swap_label_only(layers):
+    q = make(4, layers)
+    cp = make_cpu(4)
+    q.h(0)
+    cp.h(0)
+    q.x(2)
+    cp.x(2)
+    q.swap(0, 3)
+    cp.swap(0, 3)
+    q.swap(2, 0)
+    cp.swap(2, 0)
+    assert_states_close(q.get_state_vector(), cp.get_state_vector(), 1e-5)
+
+
+@pytest.mark.parametrize("layers", STACKS)
+def test_measurement_separates(layers):
+    q = make(3, layers, seed=5)
+    q.h(0)
+    q.cnot(0, 1)
+    q.cnot(1, 2)
+    r0 = q.m(0)
+    assert q.prob(1) == pytest.approx(1.0 if r0 else 0.0, abs=1e-5)
+    assert q.prob(2) == pytest.approx(1.0 if r0 else 0.0, abs=1e-5)
+
+
+@pytest.mark.parametrize("layers", STACKS)
+@pytest.mark.parametrize("seed", [1, 2, 3])
+def test_random_circuit_vs_dense(layers, seed):
+    n = 6
+    rng = np.random.default_rng(seed)
+    q = make(n, layers, seed=seed)
+    cp = make_cpu(n, seed=seed)
+    for _ in range(25):
+        r = rng.random()
+        if r < 0.5:
+            t = int(rng.integers(n))
+            th = float(rng.uniform(0, 2 * np.pi))
+            q.ry(th, t)
+            cp.ry(th, t)
+        elif r < 0.8:
+            a, b = rng.choice(n, 2, replace=False)
+            q.cnot(int(a), int(b))
+            cp.cnot(int(a), int(b))
+        else:
+            a, b = rng.choice(n, 2, replace=False)
+            q.swap(int(a), int(b))
+            cp.swap(int(a), int(b))
+    assert_states_close(q.get_state_vector(), cp.get_state_vector(), 1e-4)
+
+
+def test_try_separate_after_uncompute():
+    q = make(2, ["qunit", "cpu"], seed=4)
+    q.h(0)
+    q.cnot(0, 1)
+    q.cnot(0, 1)  # uncompute: product state again, but still one unit
+    assert q.try_separate_1(1)
+    assert q.try_separate_1(0)
+    assert abs(q.prob(0) - 0.5) < 1e-5
+    assert abs(q.prob(1)) < 1e-5
+    assert q.get_unitary_fidelity() > 0.999
+
+
+def test_try_separate_entangled_fails():
+    q = make(2, ["qunit", "cpu"], seed=4)
+    q.h(0)
+    q.cnot(0, 1)
+    assert not q.try_separate_1(0)
+    # state unchanged by the failed attempt
+    assert abs(q.prob(0) - 0.5) < 1e-5
+    cp = make_cpu(2)
+    cp.h(0)
+    cp.cnot(0, 1)
+    assert_states_close(q.get_state_vector(), cp.get_state_vector(), 1e-4)
+
+
+@pytest.mark.parametrize("layers", STACKS)
+def test_multishot(layers):
+    q = make(3, layers, seed=9)
+    q.h(0)
+    q.cnot(0, 1)
+    res = q.multi_shot_measure_mask([1, 2], 400)
+    assert sum(res.values()) == 400
+    assert set(res.keys()) <= {0, 3}
+    assert 120 < res.get(0, 0) < 280
+
+
+@pytest.mark.parametrize("layers", STACKS)
+def test_prob_mask_factorized(layers):
+    q = make(4, layers, seed=2)
+    q.h(0)
+    q.x(2)
+    # qubit 0: p=0.5 each; qubit 2: |1>
+    assert abs(q.prob_mask(0b0101, 0b0100) - 0.5) < 1e-5
+    assert abs(q.prob_mask(0b0101, 0b0001) - 0.0) < 1e-5
+
+
+@pytest.mark.parametrize("layers", STACKS)
+def test_alu_on_qunit(layers):
+    q = make(6, layers)
+    q.x(0)
+    q.x(2)  # reg = 5
+    q.inc(3, 0, 4)
+    assert q.m_reg(0, 4) == 8
+    q2 = make(8, layers)
+    q2.x(0)
+    q2.x(1)
+    q2.mul(5, 0, 4, 4)
+    assert q2.m_reg(0, 8) == 15
+
+
+@pytest.mark.parametrize("layers", STACKS)
+def test_qft_small(layers):
+    n = 5
+    q = make(n, layers, seed=3)
+    cp = make_cpu(n, seed=3)
+    for i in range(n):
+        if i % 2:
+            q.x(i)
+            cp.x(i)
+    q.qft(0, n)
+    cp.qft(0, n)
+    assert_states_close(q.get_state_vector(), cp.get_state_vector(), 1e-4)
+
+
+def test_compose_decompose_qunit():
+    a = make(2, ["qunit", "cpu"], seed=1)
+    a.h(0)
+    b = make(2, ["qunit", "cpu"], seed=2)
+    b.x(0)
+    b.h(1)
+    a.compose(b)
+    assert a.num_qubits == 4
+    assert abs(a.prob(2) - 1.0) < 1e-5
+    assert abs(a.prob(3) - 0.5) < 1e-5
+    dest = make(2, ["qunit", "cpu"], seed=3)
+    a.decompose(2, dest)
+    assert a.num_qubits == 2
+    assert abs(dest.prob(0) - 1.0) < 1e-5
+    assert abs(dest.prob(1) - 0.5) < 1e-5
+
+
+def test_expectation_qunit():
+    q = make(3, ["qunit", "cpu"], seed=2)
+    q.x(1)
+    q.h(0)
+    assert abs(q.expectation_bits_all([0, 1, 2]) - 2.5) < 1e-5
+
+
+def test_sycamore_style_circuit():
+    """Random-circuit-sampling shaped workload on the canonical stack
+    (parity model: benchmarks.cpp test_quantum_supremacy)."""
+    n = 8
+    rng = np.random.default_rng(21)
+    q = make(n, ["qunit", "stabilizer_hybrid", "cpu"], seed=21)
+    cp = make_cpu(n, seed=21)
+    sq = ["sqrt_x", "s", "h"]
+    for d in range(6):
+        for i in range(n):
+            g = sq[rng.integers(3)]
+            getattr(q, g)(i)
+            getattr(cp, g)(i)
+        start = d % 2
+        for i in range(start, n - 1, 2):
+            th, ph = rng.uniform(0, 2 * np.pi, 2)
+            q.fsim(th, ph, i, i + 1)
+            cp.fsim(th, ph, i, i + 1)
+    assert_states_close(q.get_state_vector(), cp.get_state_vector(), 2e-4)
+    res = q.multi_shot_measure_mask([1 << i for i in range(n)], 100)
+    assert sum(res.values()) == 100
