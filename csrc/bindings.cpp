@@ -11,8 +11,11 @@
 #include <pybind11/stl.h>
 
 #include "common/dlpack_min.hpp"
+#include "qcircuit.hpp"
 #include "qengine_cpu.hpp"
 #include "qfactory.hpp"
+#include "qneuron.hpp"
+#include "serialize.hpp"
 #ifdef QRACK_AMD_HIP_ENGINE
 #include "hip/qengine_hip.hpp"
 #endif
@@ -370,12 +373,88 @@ template <typename R> static void bindQInterface(py::module_& m, const char* nam
         .def("depolarizing_channel_weak_1qb", &QI::DepolarizingChannelWeak1Qb);
 }
 
+template <typename R> static void bindExtras(py::module_& m, const char* suffix)
+{
+    using QI = QInterface<R>;
+    using Ptr = QInterfacePtr<R>;
+    using C = std::complex<R>;
+
+    // ---- QCircuit (parity: include/qcircuit.hpp) ----
+    py::class_<QCircuit<R>, QCircuitPtr<R>>(m, (std::string("QCircuit") + suffix).c_str())
+        .def(py::init<bitLenInt>())
+        .def_property_readonly("num_qubits", &QCircuit<R>::GetQubitCount)
+        .def_property_readonly("gate_count", &QCircuit<R>::GetGateCount)
+        .def("append_mtrx",
+            [](QCircuit<R>& c, std::vector<C> mv, bitLenInt t) {
+                cplx<R> m[4];
+                for (int i = 0; i < 4; ++i) m[i] = from_std<R>(mv[i]);
+                c.AppendMtrx(m, t);
+            })
+        .def("append_controlled",
+            [](QCircuit<R>& c, std::vector<C> mv, bitLenInt t, std::vector<bitLenInt> ctrls,
+                bitCapInt perm) {
+                cplx<R> m[4];
+                for (int i = 0; i < 4; ++i) m[i] = from_std<R>(mv[i]);
+                c.AppendControlled(m, t, ctrls, perm);
+            })
+        .def("swap", &QCircuit<R>::Swap)
+        .def("run", [](QCircuit<R>& c, Ptr q) { c.Run(q); })
+        .def("inverse", &QCircuit<R>::Inverse)
+        .def("past_light_cone",
+            [](QCircuit<R>& c, std::vector<bitLenInt> qs) {
+                return c.PastLightCone(std::set<bitLenInt>(qs.begin(), qs.end()));
+            })
+        .def("serialize", &QCircuit<R>::Serialize)
+        .def_static("deserialize", &QCircuit<R>::Deserialize);
+
+    // ---- QNeuron (parity: include/qneuron.hpp) ----
+    py::class_<QNeuron<R>, std::shared_ptr<QNeuron<R>>>(
+        m, (std::string("QNeuron") + suffix).c_str())
+        .def(py::init([](Ptr reg, std::vector<bitLenInt> inputs, bitLenInt output, int fn,
+                          R alpha) {
+            return std::make_shared<QNeuron<R>>(
+                reg, inputs, output, (QNeuronActivationFn)fn, alpha);
+        }),
+            py::arg("reg"), py::arg("inputs"), py::arg("output"), py::arg("activation_fn") = 0,
+            py::arg("alpha") = (R)1)
+        .def("predict", &QNeuron<R>::Predict, py::arg("expected") = true,
+            py::arg("reset_init") = true)
+        .def("unpredict", &QNeuron<R>::Unpredict, py::arg("expected") = true)
+        .def("learn", &QNeuron<R>::Learn, py::arg("eta"), py::arg("expected"),
+            py::arg("reset_init") = true)
+        .def("learn_permutation", &QNeuron<R>::LearnPermutation)
+        .def("set_angles", &QNeuron<R>::SetAngles)
+        .def("get_angles", &QNeuron<R>::GetAngles)
+        .def("set_activation_fn", [](QNeuron<R>& n, int fn) {
+            n.SetActivationFn((QNeuronActivationFn)fn);
+        });
+
+    // ---- serialization ----
+    m.def((std::string("save_stabilizer_") + suffix).c_str(),
+        [](Ptr q) { return SaveStabilizerText<R>(q); });
+    m.def((std::string("load_stabilizer_") + suffix).c_str(),
+        [](const std::string& s, int64_t seed) {
+            RngPtr rng = (seed < 0) ? std::make_shared<Rng>() : std::make_shared<Rng>((uint64_t)seed);
+            return LoadStabilizerText<R>(s, rng);
+        },
+        py::arg("text"), py::arg("seed") = (int64_t)-1);
+    m.def((std::string("lossy_save_") + suffix).c_str(),
+        [](Ptr q, const std::string& path, bitLenInt blockBits) {
+            LossySaveState<R>(q, path, blockBits);
+        },
+        py::arg("sim"), py::arg("path"), py::arg("block_bits") = (bitLenInt)12);
+    m.def((std::string("lossy_load_") + suffix).c_str(),
+        [](Ptr q, const std::string& path) { LossyLoadState<R>(q, path); });
+}
+
 PYBIND11_MODULE(_qrack, m)
 {
     m.doc() = "qrack_amd native core (MI355X / HIP)";
 
     bindQInterface<float>(m, "QInterfaceF");
     bindQInterface<double>(m, "QInterfaceD");
+    bindExtras<float>(m, "F");
+    bindExtras<double>(m, "D");
 
     m.def("create", &CreateStack<float>, py::arg("qubits"), py::arg("layers") = std::vector<std::string>{ "cpu" },
         py::arg("init_perm") = (bitCapInt)0, py::arg("seed") = (int64_t)-1,

@@ -2,6 +2,9 @@
 
 #include "qstabilizer.hpp"
 #include "qstabilizerhybrid.hpp"
+#include "qhybrid.hpp"
+#include "qinterface_noisy.hpp"
+#include "qtensornetwork.hpp"
 #include "qunit.hpp"
 
 #ifdef QRACK_AMD_ENABLE_HIP
@@ -50,6 +53,34 @@ EngineFactoryFn<R> LayerFactory(
         EngineFactoryFn<R> sub = LayerFactory<R>(tail, rng, deviceId, pagesPerDevice);
         return [rng, sub](bitLenInt n, bitCapInt perm) -> QInterfacePtr<R> {
             return std::make_shared<QStabilizerHybrid<R>>(n, perm, rng, sub);
+        };
+    }
+    if (head == "tensor_network") {
+        EngineFactoryFn<R> sub = LayerFactory<R>(tail, rng, deviceId, pagesPerDevice);
+        return [rng, sub](bitLenInt n, bitCapInt perm) -> QInterfacePtr<R> {
+            return std::make_shared<QTensorNetwork<R>>(n, perm, rng, sub);
+        };
+    }
+    if (head == "noisy") {
+        EngineFactoryFn<R> sub = LayerFactory<R>(tail, rng, deviceId, pagesPerDevice);
+        return [rng, sub](bitLenInt n, bitCapInt perm) -> QInterfacePtr<R> {
+            return std::make_shared<QInterfaceNoisy<R>>(n, sub(n, perm), (R)-1, rng);
+        };
+    }
+    if (head == "hybrid") {
+        EngineFactoryFn<R> cpuF = [rng](bitLenInt n, bitCapInt perm) -> QInterfacePtr<R> {
+            return std::make_shared<QEngineCPU<R>>(n, perm, rng);
+        };
+        EngineFactoryFn<R> gpuF;
+        if (HipDeviceCount() > 0) {
+            gpuF = [rng, deviceId](bitLenInt n, bitCapInt perm) -> QInterfacePtr<R> {
+                return MakeHipEngine<R>(n, perm, rng, deviceId);
+            };
+        } else {
+            gpuF = cpuF;
+        }
+        return [rng, cpuF, gpuF](bitLenInt n, bitCapInt perm) -> QInterfacePtr<R> {
+            return std::make_shared<QHybrid<R>>(n, perm, rng, cpuF, gpuF);
         };
     }
     if (head == "qunit") {

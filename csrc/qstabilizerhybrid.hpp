@@ -44,6 +44,16 @@ public:
         EngineFactoryFn<R> factory = nullptr, bool doNorm = true, R normThresh = eps<R>::value);
 
     void SwitchToEngine();
+    // try to fold every pending shard into the tableau; false if any shard
+    // is non-Clifford (the state is then NOT serializable as a tableau)
+    bool TryFlushAllShards()
+    {
+        if (engine) return false;
+        for (bitLenInt q = 0; q < qubitCount; ++q) {
+            if (shards[q] && !TryShardFlushClifford(q)) return false;
+        }
+        return true;
+    }
     bool isClifford() const override { return !InEngineMode(); }
     bool isClifford(bitLenInt q) const override
     {

@@ -14,9 +14,21 @@ QInterfaceEngine enum (/root/reference/include/qinterface.hpp:37-132).
 from qrack_amd._qrack import (  # noqa: F401
     QInterfaceF,
     QInterfaceD,
+    QCircuitF,
+    QCircuitD,
+    QNeuronF,
+    QNeuronD,
     create,
     create_d,
     hip_device_count,
+    save_stabilizer_F,
+    save_stabilizer_D,
+    load_stabilizer_F,
+    load_stabilizer_D,
+    lossy_save_F,
+    lossy_save_D,
+    lossy_load_F,
+    lossy_load_D,
     __version__,
 )
 
