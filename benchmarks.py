@@ -1,0 +1,139 @@
+#!/usr/bin/env python3
+"""qrack_amd benchmark harness — the reference's benchmarkLoopVariable
+protocol (/root/reference/test/benchmarks.cpp:98-290): sweep qubit widths
+min..max, N samples per width, wall-clock around workload + terminal
+measurement, CSV output (avg/std/min/max ms + fidelity estimate).
+
+Workloads (reference case names in parens):
+  qft            (test_qft_permutation_init)
+  ghz            (test_ghz)
+  random_circuit (test_random_circuit_sampling_nn)
+  supremacy      (test_quantum_supremacy, Sycamore-style fsim + sqrt gates)
+  qv             (quantum volume: depth == width, random SU(4)-ish layers)
+
+Usage:
+  python benchmarks.py --workload qft --min-qubits 4 --max-qubits 24 \
+      --samples 10 --layers hip --out profiles/qft_hip.csv
+"""
+
+import argparse
+import sys
+import time
+
+import numpy as np
+
+sys.path.insert(0, ".")
+import qrack_amd as qa
+
+
+def run_qft(q, n, rng, depth):
+    q.set_permutation(int(rng.integers(1 << min(n, 62))))
+    q.qft(0, n)
+
+
+def run_ghz(q, n, rng, depth):
+    q.set_permutation(0)
+    q.h(0)
+    for i in range(n - 1):
+        q.cnot(i, i + 1)
+
+
+def run_random_circuit(q, n, rng, depth):
+    q.set_permutation(0)
+    d = depth or n
+    for _ in range(d):
+        for i in range(n):
+            g = rng.integers(3)
+            if g == 0:
+                q.h(i)
+            elif g == 1:
+                q.x(i)
+            else:
+                q.t(i)
+        for i in range(0, n - 1, 2):
+            q.cnot(i, i + 1)
+
+
+def run_supremacy(q, n, rng, depth):
+    q.set_permutation(0)
+    d = depth or n
+    sq = ["sqrt_x", "s", "h"]
+    for layer in range(d):
+        for i in range(n):
+            getattr(q, sq[rng.integers(3)])(i)
+        start = layer % 2
+        for i in range(start, n - 1, 2):
+            th, ph = rng.uniform(0, 2 * np.pi, 2)
+            q.fsim(float(th), float(ph), i, i + 1)
+
+
+def run_qv(q, n, rng, depth):
+    q.set_permutation(0)
+    for _ in range(depth or n):
+        perm = rng.permutation(n)
+        for k in range(0, n - 1, 2):
+            a, b = int(perm[k]), int(perm[k + 1])
+            for t in (a, b):
+                th, ph, lm = rng.uniform(0, 2 * np.pi, 3)
+                q.u(t, float(th), float(ph), float(lm))
+            q.cnot(a, b)
+
+
+WORKLOADS = {
+    "qft": run_qft,
+    "ghz": run_ghz,
+    "random_circuit": run_random_circuit,
+    "supremacy": run_supremacy,
+    "qv": run_qv,
+}
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--workload", default="qft", choices=sorted(WORKLOADS))
+    p.add_argument("--min-qubits", type=int, default=4)
+    p.add_argument("--max-qubits", type=int, default=24)
+    p.add_argument("--samples", type=int, default=10)
+    p.add_argument("--benchmark-depth", type=int, default=0)
+    p.add_argument("--benchmark-shots", type=int, default=1)
+    p.add_argument("--layers", default="auto",
+        help="comma layer list (e.g. qunit,stabilizer_hybrid,hip) or auto")
+    p.add_argument("--precision", default="fp32", choices=["fp32", "fp64"])
+    p.add_argument("--seed", type=int, default=0, help="0 = time-seeded (printed for repro)")
+    p.add_argument("--out", default="")
+    args = p.parse_args()
+
+    seed = args.seed or int(time.time())
+    print(f"# workload={args.workload} layers={args.layers} precision={args.precision} "
+          f"seed={seed}", flush=True)
+
+    rows = ["width,samples,avg_ms,std_ms,min_ms,max_ms,fidelity"]
+    fn = WORKLOADS[args.workload]
+    for n in range(args.min_qubits, args.max_qubits + 1):
+        if args.layers == "auto":
+            q = qa.create_simulator(n, precision=args.precision, seed=seed + n)
+        else:
+            q = qa.create_simulator(
+                n, precision=args.precision, layers=args.layers.split(","), seed=seed + n)
+        rng = np.random.default_rng(seed + n)
+        times = []
+        for s in range(args.samples):
+            t0 = time.perf_counter()
+            fn(q, n, rng, args.benchmark_depth)
+            q.multi_shot_measure_mask(
+                [1 << i for i in range(min(n, 32))], args.benchmark_shots)
+            if hasattr(q, "finish"):
+                q.finish()
+            times.append(1000.0 * (time.perf_counter() - t0))
+        fid = q.get_unitary_fidelity() if hasattr(q, "get_unitary_fidelity") else 1.0
+        row = (f"{n},{args.samples},{np.mean(times):.3f},{np.std(times):.3f},"
+               f"{np.min(times):.3f},{np.max(times):.3f},{fid:.6f}")
+        rows.append(row)
+        print(row, flush=True)
+    if args.out:
+        with open(args.out, "w") as f:
+            f.write("\n".join(rows) + "\n")
+
+
+if __name__ == "__main__":
+    main()
