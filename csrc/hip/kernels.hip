@@ -779,41 +779,38 @@ template <> __device__ __forceinline__ void devSinCos<double>(double t, double* 
 }
 
 template <typename R>
-__global__ void k_qft_ramp(
-    cplx<R>* sv, bitCapInt maxI, bitCapInt colPow, bitLenInt start, R scale)
+__global__ void k_phase_ramp(
+    cplx<R>* sv, bitCapInt maxI, bitCapInt condPow, bitLenInt rampStart, bitCapInt rampMask, R scale)
 {
-    // iterate indices with bit (start+col) set: maxI = maxQPower/2 pairs,
-    // expand with the column power as skip bit, then OR it in
+    // condPow != 0: iterate the condPow-set half (maxI = maxQPower/2 with the
+    // bit inserted); condPow == 0: iterate everything (maxI = maxQPower)
     const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
-    const bitCapInt lowMask = colPow - 1u; // bits below start+col
     for (bitCapInt j = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; j < maxI; j += stride) {
-        const bitCapInt i = (((j & ~(colPow - 1u)) << 1u) | (j & (colPow - 1u))) | colPow;
-        const bitCapInt frac = (i & lowMask) >> start;
+        bitCapInt i = j;
+        if (condPow) i = (((j & ~(condPow - 1u)) << 1u) | (j & (condPow - 1u))) | condPow;
+        const bitCapInt frac = (i >> rampStart) & rampMask;
         const R theta = scale * (R)frac;
         R s, c;
         devSinCos<R>(theta, &s, &c);
-        const cplx<R> f{ c, s };
-        sv[i] = f * sv[i];
+        sv[i] = cplx<R>{ c, s } * sv[i];
     }
 }
 
-// fp32 vectorized: two adjacent set-bit indices per lane (valid when
-// start >= 1 would change theta... adjacent i differ in bit 0: theta differs
-// unless start > 0; compute both thetas — still one float4 load/store)
-__global__ void k_qft_ramp_v(
-    cplx<float>* sv, bitCapInt maxI, bitCapInt colPow, bitLenInt start, float scale)
+// fp32 vectorized: two adjacent indices per lane via float4
+__global__ void k_phase_ramp_v(cplx<float>* sv, bitCapInt maxI, bitCapInt condPow,
+    bitLenInt rampStart, bitCapInt rampMask, float scale)
 {
     float4* sv4 = reinterpret_cast<float4*>(sv);
     const bitCapInt half = maxI >> 1u;
     const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
-    const bitCapInt lowMask = colPow - 1u;
     for (bitCapInt k = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; k < half; k += stride) {
         const bitCapInt j = 2u * k;
-        const bitCapInt i = (((j & ~(colPow - 1u)) << 1u) | (j & (colPow - 1u))) | colPow;
+        bitCapInt i = j;
+        if (condPow) i = (((j & ~(condPow - 1u)) << 1u) | (j & (condPow - 1u))) | condPow;
         const bitCapInt i4 = i >> 1u;
         float4 v = sv4[i4];
-        const bitCapInt f0 = (i & lowMask) >> start;
-        const bitCapInt f1 = ((i + 1u) & lowMask) >> start;
+        const bitCapInt f0 = (i >> rampStart) & rampMask;
+        const bitCapInt f1 = ((i + 1u) >> rampStart) & rampMask;
         float s0, c0, s1, c1;
         __sincosf(scale * (float)f0, &s0, &c0);
         __sincosf(scale * (float)f1, &s1, &c1);
@@ -825,21 +822,20 @@ __global__ void k_qft_ramp_v(
 }
 
 template <typename R>
-void launchQftRamp(cplx<R>* sv, bitCapInt maxQPower, bitLenInt start, bitLenInt col, int sign,
-    hipStream_t stream)
+void launchPhaseRamp(cplx<R>* sv, bitCapInt maxQPower, bitLenInt rampStart, bitLenInt rampBits,
+    bitCapInt condPower, double scale, hipStream_t stream)
 {
-    const bitCapInt colPow = ONE_BCI << (start + col);
-    const bitCapInt maxI = maxQPower >> 1u;
-    const R scale = (R)sign * PI_R<R> / (R)(ONE_BCI << col);
+    const bitCapInt maxI = condPower ? (maxQPower >> 1u) : maxQPower;
+    const bitCapInt rampMask = (ONE_BCI << rampBits) - 1u;
     if constexpr (std::is_same_v<R, float>) {
-        if (colPow >= 2u && (maxI & 1u) == 0u) {
-            hipLaunchKernelGGL((k_qft_ramp_v), dim3(gridFor(maxI >> 1u)), dim3(QA_BLOCK), 0,
-                stream, sv, maxI, colPow, start, scale);
+        if ((condPower == 0u || condPower >= 2u) && (maxI & 1u) == 0u && maxI >= 2u) {
+            hipLaunchKernelGGL((k_phase_ramp_v), dim3(gridFor(maxI >> 1u)), dim3(QA_BLOCK), 0,
+                stream, sv, maxI, condPower, rampStart, rampMask, (float)scale);
             return;
         }
     }
-    hipLaunchKernelGGL((k_qft_ramp<R>), dim3(gridFor(maxI)), dim3(QA_BLOCK), 0, stream, sv, maxI,
-        colPow, start, scale);
+    hipLaunchKernelGGL((k_phase_ramp<R>), dim3(gridFor(maxI)), dim3(QA_BLOCK), 0, stream, sv,
+        maxI, condPower, rampStart, rampMask, (R)scale);
 }
 
 // ---- sampling / inner product / marginals --------------------------------------
@@ -995,7 +991,7 @@ void launchPartProbs(const cplx<R>* sv, bitCapInt maxQPower, bitLenInt start, bi
         const cplx<R>*, const cplx<R>*, bitCapInt, double*, double*, hipStream_t);                  \
     template void launchPartProbs<R>(                                                               \
         const cplx<R>*, bitCapInt, bitLenInt, bitLenInt, double*, hipStream_t);                     \
-    template void launchQftRamp<R>(cplx<R>*, bitCapInt, bitLenInt, bitLenInt, int, hipStream_t);
+    template void launchPhaseRamp<R>(cplx<R>*, bitCapInt, bitLenInt, bitLenInt, bitCapInt, double, hipStream_t);
 
 QA_INSTANTIATE(float)
 QA_INSTANTIATE(double)

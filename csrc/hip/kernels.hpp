@@ -152,12 +152,12 @@ void launchPhaseFlipIfLess(cplx<R>* sv, bitCapInt maxQPower, bitCapInt greaterPe
 
 int reduceGridSize(bitCapInt n);
 
-// fused QFT phase-ramp: for every index with bit (start+i) set,
-// amp *= exp(sign * i*pi * ((x >> start) mod 2^i) / 2^i)
-// — one pass replaces the i controlled-phase gates of a QFT column.
+// fused diagonal phase ramp: amp *= exp(i*scale*((x>>rampStart) mod
+// 2^rampBits)) where (condPower==0) || (x & condPower) — one pass replaces
+// rampBits (controlled-)phase gates (the QFT column ladder).
 template <typename R>
-void launchQftRamp(cplx<R>* sv, bitCapInt maxQPower, bitLenInt start, bitLenInt col, int sign,
-    hipStream_t stream);
+void launchPhaseRamp(cplx<R>* sv, bitCapInt maxQPower, bitLenInt rampStart, bitLenInt rampBits,
+    bitCapInt condPower, double scale, hipStream_t stream);
 
 // contiguous per-chunk |amp|^2 sums (inverse-CDF sampling support):
 // sums[c] = sum over [c*chunkLen, (c+1)*chunkLen)

@@ -384,25 +384,30 @@ void QEngineHIP<R>::UniformlyControlledSingleBit(
     QA_HIP_CHECK(hipFreeAsync(dMtrxs, stream));
 }
 
+template <typename R>
+void QEngineHIP<R>::PhaseRamp(R scale, bitLenInt rampStart, bitLenInt rampBits, bitCapInt condPower)
+{
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    launchPhaseRamp<R>(dState, maxQPower, rampStart, rampBits, condPower, (double)scale, stream);
+}
+
 template <typename R> void QEngineHIP<R>::QFT(bitLenInt start, bitLenInt length, bool)
 {
     // H on the top column, then ONE fused ramp per column (the i controlled
     // phases CPhaseRootN(i-j+1, j, i) collapse to exp(i*pi*(x mod 2^i)/2^i)
     // on the bit-i-set half)
     if (!length) return;
-    QA_HIP_CHECK(hipSetDevice(deviceId));
     for (bitLenInt i = length; i-- > 0;) {
         this->H(start + i);
-        if (i) launchQftRamp<R>(dState, maxQPower, start, i, +1, stream);
+        if (i) PhaseRamp(PI_R<R> / (R)pow2(i), start, i, pow2(start + i));
     }
 }
 
 template <typename R> void QEngineHIP<R>::IQFT(bitLenInt start, bitLenInt length, bool)
 {
     if (!length) return;
-    QA_HIP_CHECK(hipSetDevice(deviceId));
     for (bitLenInt i = 0; i < length; ++i) {
-        if (i) launchQftRamp<R>(dState, maxQPower, start, i, -1, stream);
+        if (i) PhaseRamp(-PI_R<R> / (R)pow2(i), start, i, pow2(start + i));
         this->H(start + i);
     }
 }

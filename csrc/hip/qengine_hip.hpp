@@ -116,6 +116,7 @@ public:
     // controlled-phase ladder (O(n) full-state passes instead of O(n^2/2))
     void QFT(bitLenInt start, bitLenInt length, bool trySeparate = false) override;
     void IQFT(bitLenInt start, bitLenInt length, bool trySeparate = false) override;
+    void PhaseRamp(R scale, bitLenInt rampStart, bitLenInt rampBits, bitCapInt condPower) override;
 
     // ---- probability / measurement ----
     R Prob(bitLenInt q) override;

@@ -169,6 +169,22 @@ public:
         ControlledSwapBlock(controls, q1, q2, false, true);
     }
 
+    // additive diagonal phase ramp: amp *= exp(i*scale*((x >> rampStart) mod
+    // 2^rampBits)) applied where (condPower == 0) || (x & condPower).
+    // One fused pass replaces rampBits (controlled-)phase gates; the default
+    // lowering is the gate product (engines override with a single kernel).
+    virtual void PhaseRamp(R scale, bitLenInt rampStart, bitLenInt rampBits, bitCapInt condPower)
+    {
+        for (bitLenInt b = 0; b < rampBits; ++b) {
+            const cplx<R> f = polar<R>((R)1, scale * (R)pow2(b));
+            if (condPower) {
+                this->MCPhase({ log2Ocl(condPower) }, cplx<R>(1, 0), f, rampStart + b);
+            } else {
+                this->Phase(cplx<R>(1, 0), f, rampStart + b);
+            }
+        }
+    }
+
     // ---- measurement --------------------------------------------------------
     bool ForceM(bitLenInt qubit, bool result, bool doForce = true, bool doApply = true) override;
 

@@ -217,20 +217,25 @@ template <typename R> void QEngineCPU<R>::ROL(bitLenInt shift, bitLenInt start, 
     });
 }
 
-template <typename R> void QEngineCPU<R>::QftRamp(bitLenInt start, bitLenInt col, int sign)
+template <typename R>
+void QEngineCPU<R>::PhaseRamp(R scale, bitLenInt rampStart, bitLenInt rampBits, bitCapInt condPower)
 {
-    // fused QFT column: one pass applies all of the column's controlled
-    // phases: amp *= exp(sign*i*pi*((x>>start) mod 2^col)/2^col) on the
-    // bit-(start+col)-set half
-    const bitCapInt colPow = pow2(start + col);
-    const bitCapInt lowMask = colPow - 1u;
-    const R scale = (R)sign * PI_R<R> / (R)pow2(col);
+    // fused diagonal ramp: amp *= exp(i*scale*((x>>rampStart) mod 2^rampBits))
+    // on the condPower-set half (or everywhere when condPower == 0)
+    const bitCapInt rampMask = pow2Mask(rampBits);
     cplx<R>* sv = stateVec.data();
-    this->par_for_skip(maxQPower >> 1u, colPow, [=](const bitCapInt& j, unsigned) {
-        const bitCapInt i = j | colPow;
-        const R theta = scale * (R)((i & lowMask) >> start);
-        sv[i] = polar<R>(1, theta) * sv[i];
-    });
+    if (condPower) {
+        this->par_for_skip(maxQPower >> 1u, condPower, [=](const bitCapInt& j, unsigned) {
+            const bitCapInt i = j | condPower;
+            const R theta = scale * (R)((i >> rampStart) & rampMask);
+            sv[i] = polar<R>(1, theta) * sv[i];
+        });
+    } else {
+        this->par_for(0, maxQPower, [=](const bitCapInt& i, unsigned) {
+            const R theta = scale * (R)((i >> rampStart) & rampMask);
+            sv[i] = polar<R>(1, theta) * sv[i];
+        });
+    }
 }
 
 template <typename R> void QEngineCPU<R>::QFT(bitLenInt start, bitLenInt length, bool)
@@ -238,7 +243,7 @@ template <typename R> void QEngineCPU<R>::QFT(bitLenInt start, bitLenInt length,
     if (!length) return;
     for (bitLenInt i = length; i-- > 0;) {
         this->H(start + i);
-        if (i) QftRamp(start, i, +1);
+        if (i) PhaseRamp(PI_R<R> / (R)pow2(i), start, i, pow2(start + i));
     }
 }
 
@@ -246,7 +251,7 @@ template <typename R> void QEngineCPU<R>::IQFT(bitLenInt start, bitLenInt length
 {
     if (!length) return;
     for (bitLenInt i = 0; i < length; ++i) {
-        if (i) QftRamp(start, i, -1);
+        if (i) PhaseRamp(-PI_R<R> / (R)pow2(i), start, i, pow2(start + i));
         this->H(start + i);
     }
 }

@@ -61,6 +61,7 @@ public:
     void ROL(bitLenInt shift, bitLenInt start, bitLenInt length) override;
     void QFT(bitLenInt start, bitLenInt length, bool trySeparate = false) override;
     void IQFT(bitLenInt start, bitLenInt length, bool trySeparate = false) override;
+    void PhaseRamp(R scale, bitLenInt rampStart, bitLenInt rampBits, bitCapInt condPower) override;
 
     // ---- probability / measurement ----
     R Prob(bitLenInt q) override;
@@ -133,7 +134,7 @@ protected:
     void ControlledPermutationOp(
         bitCapInt controlMask, const std::function<bitCapInt(bitCapInt)>& f);
     bitCapInt SampleOnce();
-    void QftRamp(bitLenInt start, bitLenInt col, int sign);
+
 };
 
 } // namespace qrack_amd

@@ -4,6 +4,7 @@
 #include "qstabilizerhybrid.hpp"
 #include "qhybrid.hpp"
 #include "qinterface_noisy.hpp"
+#include "qpager.hpp"
 #include "qtensornetwork.hpp"
 #include "qunit.hpp"
 
@@ -81,6 +82,38 @@ EngineFactoryFn<R> LayerFactory(
         }
         return [rng, cpuF, gpuF](bitLenInt n, bitCapInt perm) -> QInterfacePtr<R> {
             return std::make_shared<QHybrid<R>>(n, perm, rng, cpuF, gpuF);
+        };
+    }
+    if (head == "pager") {
+        EngineFactoryFn<R> sub = LayerFactory<R>(tail, rng, deviceId, pagesPerDevice);
+        // device list: QRACK_QPAGER_DEVICES "N.id,..." (reference syntax) or
+        // the single requested device
+        std::vector<int64_t> devs;
+        if (const char* env = std::getenv("QRACK_QPAGER_DEVICES")) {
+            std::string spec(env);
+            size_t pos = 0;
+            while (pos < spec.size()) {
+                size_t comma = spec.find(',', pos);
+                std::string tok = spec.substr(pos, comma == std::string::npos ? comma : comma - pos);
+                size_t dot = tok.find('.');
+                if (dot == std::string::npos) {
+                    devs.push_back(std::atoll(tok.c_str()));
+                } else {
+                    const int count = std::atoi(tok.substr(0, dot).c_str());
+                    const int64_t id = std::atoll(tok.substr(dot + 1).c_str());
+                    for (int k = 0; k < count; ++k) devs.push_back(id);
+                }
+                if (comma == std::string::npos) break;
+                pos = comma + 1;
+            }
+        }
+        const bitLenInt ppd = pagesPerDevice ? pagesPerDevice : 1u;
+        return [sub, rng, devs, ppd](bitLenInt n, bitCapInt perm) -> QInterfacePtr<R> {
+            const size_t nd = devs.empty() ? 1u : devs.size();
+            bitLenInt metaBits = log2Ocl((bitCapInt)(ppd * nd));
+            if (!isPowerOfTwo(ppd * nd)) metaBits++;
+            bitLenInt pq = (n > metaBits) ? (bitLenInt)(n - metaBits) : 1u;
+            return std::make_shared<QPager<R>>(n, perm, rng, sub, pq, devs);
         };
     }
     if (head == "qunit") {
