@@ -892,8 +892,17 @@ static void fillSkips(PermArgs& a, std::vector<bitCapInt> powers)
     for (int i = 0; i < a.nPowers; ++i) a.qPowers[i] = powers[i];
 }
 
+template <typename R>
+static void checkAluRangeHip(bitLenInt start, bitLenInt length, bitLenInt qubitCount, const char* op)
+{
+    if ((bitCapInt)start + length > qubitCount) {
+        throw QrackError(std::string(op) + ": register is out of the qubit range");
+    }
+}
+
 template <typename R> void QEngineHIP<R>::INC(bitCapInt toAdd, bitLenInt start, bitLenInt length)
 {
+    checkAluRangeHip<R>(start, length, qubitCount, "INC");
     if (!length) return;
     toAdd &= pow2Mask(length);
     if (!toAdd) return;
@@ -1051,6 +1060,8 @@ template <typename R>
 void QEngineHIP<R>::MULModNOut(
     bitCapInt toMul, bitCapInt modN, bitLenInt inStart, bitLenInt outStart, bitLenInt length)
 {
+    checkAluRangeHip<R>(inStart, length, qubitCount, "MULModNOut");
+    checkAluRangeHip<R>(outStart, length, qubitCount, "MULModNOut");
     PermArgs a{};
     setupModArgs<R>(a, PermOp::MULMODN, toMul, modN, inStart, outStart, length, maxQPower, {});
     permuteOp(a, true, false);
@@ -1069,6 +1080,8 @@ template <typename R>
 void QEngineHIP<R>::POWModNOut(
     bitCapInt base, bitCapInt modN, bitLenInt inStart, bitLenInt outStart, bitLenInt length)
 {
+    checkAluRangeHip<R>(inStart, length, qubitCount, "POWModNOut");
+    checkAluRangeHip<R>(outStart, length, qubitCount, "POWModNOut");
     PermArgs a{};
     setupModArgs<R>(a, PermOp::POWMODN, base, modN, inStart, outStart, length, maxQPower, {});
     permuteOp(a, true, false);
@@ -1315,6 +1328,13 @@ void QEngineHIP<R>::CPhaseFlipIfLess(
     QA_HIP_CHECK(hipSetDevice(deviceId));
     launchPhaseFlipIfLess<R>(
         dState, maxQPower, greaterPerm, start, pow2Mask(length) << start, pow2(flagIndex), stream);
+}
+
+size_t HipActiveAllocImpl(int device)
+{
+    auto& t = HipDeviceTracker::instance();
+    if (device < 0 || device >= t.deviceCount()) return 0;
+    return t.activeAlloc(device);
 }
 
 // ---- factory hook -----------------------------------------------------------

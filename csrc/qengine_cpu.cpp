@@ -656,8 +656,17 @@ void QEngineCPU<R>::ControlledPermutationOp(
 
 // ---- ALU -------------------------------------------------------------------
 
+template <typename R>
+static void checkAluRange(bitLenInt start, bitLenInt length, bitLenInt qubitCount, const char* op)
+{
+    if ((bitCapInt)start + length > qubitCount) {
+        throw QrackError(std::string(op) + ": register is out of the qubit range");
+    }
+}
+
 template <typename R> void QEngineCPU<R>::INC(bitCapInt toAdd, bitLenInt start, bitLenInt length)
 {
+    checkAluRange<R>(start, length, qubitCount, "INC");
     if (!length) return;
     const bitCapInt lenMask = pow2Mask(length);
     toAdd &= lenMask;
@@ -777,6 +786,8 @@ void QEngineCPU<R>::INCS(bitCapInt toAdd, bitLenInt start, bitLenInt length, bit
 template <typename R>
 void QEngineCPU<R>::MUL(bitCapInt toMul, bitLenInt inOutStart, bitLenInt carryStart, bitLenInt length)
 {
+    checkAluRange<R>(inOutStart, length, qubitCount, "MUL");
+    checkAluRange<R>(carryStart, length, qubitCount, "MUL");
     if (!toMul) throw QrackError("MUL by zero is not invertible");
     if (toMul == 1u) return;
     const bitCapInt lenMask = pow2Mask(length);
@@ -831,6 +842,8 @@ template <typename R>
 void QEngineCPU<R>::MULModNOut(
     bitCapInt toMul, bitCapInt modN, bitLenInt inStart, bitLenInt outStart, bitLenInt length)
 {
+    checkAluRange<R>(inStart, length, qubitCount, "MULModNOut");
+    checkAluRange<R>(outStart, length, qubitCount, "MULModNOut");
     const bitCapInt lenMask = pow2Mask(length);
     const bitCapInt inMask = lenMask << inStart;
     const bitCapInt outMask = lenMask << outStart;
@@ -885,6 +898,8 @@ template <typename R>
 void QEngineCPU<R>::POWModNOut(
     bitCapInt base, bitCapInt modN, bitLenInt inStart, bitLenInt outStart, bitLenInt length)
 {
+    checkAluRange<R>(inStart, length, qubitCount, "POWModNOut");
+    checkAluRange<R>(outStart, length, qubitCount, "POWModNOut");
     const bitCapInt lenMask = pow2Mask(length);
     const bitCapInt inMask = lenMask << inStart;
     const bitCapInt outMask = lenMask << outStart;

@@ -7,12 +7,25 @@
 #include "qpager.hpp"
 #include "qtensornetwork.hpp"
 #include "qunit.hpp"
+#include "qunitmulti.hpp"
 
 #ifdef QRACK_AMD_ENABLE_HIP
 #include <hip/hip_runtime.h>
 #endif
 
 namespace qrack_amd {
+
+size_t HipActiveAlloc(int device)
+{
+#ifdef QRACK_AMD_HIP_ENGINE
+    extern size_t HipActiveAllocImpl(int);
+    return HipActiveAllocImpl(device);
+#else
+    return 0;
+#endif
+}
+
+int HipVisibleDevices() { return HipDeviceCount(); }
 
 int HipDeviceCount()
 {
@@ -120,6 +133,12 @@ EngineFactoryFn<R> LayerFactory(
         EngineFactoryFn<R> sub = LayerFactory<R>(tail, rng, deviceId, pagesPerDevice);
         return [rng, sub](bitLenInt n, bitCapInt perm) -> QInterfacePtr<R> {
             return std::make_shared<QUnit<R>>(n, perm, rng, sub);
+        };
+    }
+    if (head == "qunit_multi") {
+        EngineFactoryFn<R> sub = LayerFactory<R>(tail, rng, deviceId, pagesPerDevice);
+        return [rng, sub](bitLenInt n, bitCapInt perm) -> QInterfacePtr<R> {
+            return std::make_shared<QUnitMulti<R>>(n, perm, rng, sub);
         };
     }
     throw QrackError("unknown layer: " + head);

@@ -1,0 +1,13 @@
+"""QFT period finding demo (parity: /root/reference/examples/qft.cpp)."""
+import sys
+sys.path.insert(0, "..")
+import qrack_amd as qa
+
+if __name__ == "__main__":
+    n = 10
+    q = qa.create_simulator(n, seed=3)
+    q.set_permutation(5)
+    q.qft(0, n)
+    q.iqft(0, n)
+    assert q.m_all() == 5
+    print("QFT round trip OK at", n, "qubits")
