@@ -12,17 +12,58 @@
 //    branch is wave-uniform) instead of 26 near-identical kernels.
 #include "kernels.hpp"
 
+#include <cstdlib>
+
 namespace qrack_amd {
 
 constexpr int QA_BLOCK = 256;
 constexpr int QA_MAX_BLOCKS = 2048; // 256 CUs x 8 blocks
 
+static inline int maxBlocks()
+{
+    static int v = [] {
+        if (const char* env = std::getenv("QRACK_GPU_BLOCKS")) return std::atoi(env);
+        return QA_MAX_BLOCKS;
+    }();
+    return v;
+}
+
+static inline bool useNontemporal()
+{
+    static bool v = [] {
+        if (const char* env = std::getenv("QRACK_GPU_NT")) return std::atoi(env) != 0;
+        return false;
+    }();
+    return v;
+}
+
 static inline int gridFor(bitCapInt n)
 {
     bitCapInt b = (n + QA_BLOCK - 1) / QA_BLOCK;
-    if (b > (bitCapInt)QA_MAX_BLOCKS) b = QA_MAX_BLOCKS;
+    if (b > (bitCapInt)maxBlocks()) b = maxBlocks();
     if (b < 1) b = 1;
     return (int)b;
+}
+
+// nontemporal (streaming) load/store option for the pure-stream kernels:
+// state-vector gate traffic has zero reuse, so bypassing L2/LLC can help
+typedef float nat_float4 __attribute__((ext_vector_type(4)));
+template <bool NT> __device__ __forceinline__ float4 ld4(const float4* p)
+{
+    if constexpr (NT) {
+        const nat_float4 v = __builtin_nontemporal_load(reinterpret_cast<const nat_float4*>(p));
+        return make_float4(v.x, v.y, v.z, v.w);
+    }
+    return *p;
+}
+template <bool NT> __device__ __forceinline__ void st4(float4* p, float4 v)
+{
+    if constexpr (NT) {
+        nat_float4 nv = { v.x, v.y, v.z, v.w };
+        __builtin_nontemporal_store(nv, reinterpret_cast<nat_float4*>(p));
+    } else {
+        *p = v;
+    }
 }
 
 int reduceGridSize(bitCapInt n) { return gridFor(n); }
@@ -63,7 +104,7 @@ template <typename R, int KIND> __global__ void k_apply2x2(cplx<R>* sv, GateArgs
 
 // fp32 vectorized single-target path: each lane handles TWO adjacent pairs
 // through float4 (16 B) loads/stores. Requires nPowers == 1 and even maxI.
-template <int KIND> __global__ void k_apply2x2_1v(cplx<float>* sv, GateArgs<float> a)
+template <int KIND, bool NT> __global__ void k_apply2x2_1v(cplx<float>* sv, GateArgs<float> a)
 {
     float4* sv4 = reinterpret_cast<float4*>(sv);
     const bitCapInt p = a.qPowers[0];
@@ -76,7 +117,7 @@ template <int KIND> __global__ void k_apply2x2_1v(cplx<float>* sv, GateArgs<floa
             // two pairs: float4s at 2k and 2k+1
             for (int h = 0; h < 2; ++h) {
                 const bitCapInt idx = 2u * k + h;
-                float4 v = sv4[idx];
+                float4 v = ld4<NT>(&sv4[idx]);
                 const cplx<float> x{ v.x, v.y }, y{ v.z, v.w };
                 cplx<float> nx, ny;
                 if (KIND == 1) {
@@ -89,7 +130,7 @@ template <int KIND> __global__ void k_apply2x2_1v(cplx<float>* sv, GateArgs<floa
                     nx = m0 * x + m1 * y;
                     ny = m2 * x + m3 * y;
                 }
-                sv4[idx] = make_float4(nx.re, nx.im, ny.re, ny.im);
+                st4<NT>(&sv4[idx], make_float4(nx.re, nx.im, ny.re, ny.im));
             }
         }
     } else {
@@ -99,8 +140,8 @@ template <int KIND> __global__ void k_apply2x2_1v(cplx<float>* sv, GateArgs<floa
             const bitCapInt i = ((j & ~(p - 1u)) << 1u) | (j & (p - 1u));
             const bitCapInt lo4 = i >> 1u;       // float4 index of (i, i+1)
             const bitCapInt hi4 = (i + p) >> 1u; // float4 index of (i+p, i+p+1)
-            float4 vlo = sv4[lo4];
-            float4 vhi = sv4[hi4];
+            float4 vlo = ld4<NT>(&sv4[lo4]);
+            float4 vhi = ld4<NT>(&sv4[hi4]);
             const cplx<float> x0{ vlo.x, vlo.y }, x1{ vlo.z, vlo.w };
             const cplx<float> y0{ vhi.x, vhi.y }, y1{ vhi.z, vhi.w };
             cplx<float> a0, a1, b0, b1;
@@ -114,8 +155,8 @@ template <int KIND> __global__ void k_apply2x2_1v(cplx<float>* sv, GateArgs<floa
                 a0 = m0 * x0 + m1 * y0; a1 = m0 * x1 + m1 * y1;
                 b0 = m2 * x0 + m3 * y0; b1 = m2 * x1 + m3 * y1;
             }
-            sv4[lo4] = make_float4(a0.re, a0.im, a1.re, a1.im);
-            sv4[hi4] = make_float4(b0.re, b0.im, b1.re, b1.im);
+            st4<NT>(&sv4[lo4], make_float4(a0.re, a0.im, a1.re, a1.im));
+            st4<NT>(&sv4[hi4], make_float4(b0.re, b0.im, b1.re, b1.im));
         }
     }
 }
@@ -224,15 +265,28 @@ void launchApply2x2(cplx<R>* sv, const GateArgs<R>& a, hipStream_t stream)
         if (a.nPowers == 1 && (a.maxI & 1u) == 0u && a.maxI >= 2u && a.offset1 == 0u) {
             // single-target fast path (handles target bit 0 too)
             const int gridv = gridFor(a.maxI >> 1u);
+            if (useNontemporal()) {
+                switch (kind) {
+                case 1:
+                    hipLaunchKernelGGL((k_apply2x2_1v<1, true>), dim3(gridv), dim3(QA_BLOCK), 0, stream, sv, a);
+                    return;
+                case 2:
+                    hipLaunchKernelGGL((k_apply2x2_1v<2, true>), dim3(gridv), dim3(QA_BLOCK), 0, stream, sv, a);
+                    return;
+                default:
+                    hipLaunchKernelGGL((k_apply2x2_1v<0, true>), dim3(gridv), dim3(QA_BLOCK), 0, stream, sv, a);
+                    return;
+                }
+            }
             switch (kind) {
             case 1:
-                hipLaunchKernelGGL((k_apply2x2_1v<1>), dim3(gridv), dim3(QA_BLOCK), 0, stream, sv, a);
+                hipLaunchKernelGGL((k_apply2x2_1v<1, false>), dim3(gridv), dim3(QA_BLOCK), 0, stream, sv, a);
                 return;
             case 2:
-                hipLaunchKernelGGL((k_apply2x2_1v<2>), dim3(gridv), dim3(QA_BLOCK), 0, stream, sv, a);
+                hipLaunchKernelGGL((k_apply2x2_1v<2, false>), dim3(gridv), dim3(QA_BLOCK), 0, stream, sv, a);
                 return;
             default:
-                hipLaunchKernelGGL((k_apply2x2_1v<0>), dim3(gridv), dim3(QA_BLOCK), 0, stream, sv, a);
+                hipLaunchKernelGGL((k_apply2x2_1v<0, false>), dim3(gridv), dim3(QA_BLOCK), 0, stream, sv, a);
                 return;
             }
         }
