@@ -26,6 +26,9 @@ QUnit<R>::QUnit(bitLenInt qBitCount, bitCapInt initState, RngPtr rgp, EngineFact
     if (const char* env = std::getenv("QRACK_QUNIT_SEPARABILITY_THRESHOLD")) {
         separabilityThreshold = (R)std::atof(env);
     }
+    if (const char* env = std::getenv("QRACK_QUNIT_ACE_MAX_QB")) {
+        aceMaxQubits = (bitLenInt)std::atoi(env);
+    }
     for (bitLenInt q = 0; q < qBitCount; ++q) {
         shards[q].unit = MakeUnit(1u, (initState >> q) & 1u);
         shards[q].mapped = 0;
@@ -63,6 +66,11 @@ QInterfacePtr<R> QUnit<R>::EntangleAll(const std::vector<bitLenInt>& qs)
         }
     }
     QInterfacePtr<R> base = units[0];
+    if (aceMaxQubits && units.size() > 1u) {
+        bitLenInt total = 0;
+        for (auto& u : units) total += u->GetQubitCount();
+        if (total > aceMaxQubits) throw std::bad_alloc();
+    }
     for (size_t u = 1; u < units.size(); ++u) {
         const bitLenInt offset = base->GetQubitCount();
         base->Compose(units[u]);
@@ -133,6 +141,22 @@ bool QUnit<R>::ControlShortcut(bitLenInt control, bool anti, bool& alwaysOn)
         return true;
     }
     return false;
+}
+
+template <typename R>
+bool QUnit<R>::ElideControls(const std::vector<bitLenInt>& controls, bool anti, bool& gateApplies)
+{
+    // classically collapse each control (the ACE rounding step); returns the
+    // surviving gate condition in gateApplies
+    gateApplies = true;
+    for (bitLenInt c : controls) {
+        const R p1 = Prob(c);
+        const bool outcome = (this->Rand() < (double)p1);
+        logFidelity += std::log(std::max((double)(outcome ? p1 : ((R)1 - p1)), 1e-300));
+        ForceM(c, outcome, true, true);
+        if (anti ? outcome : !outcome) gateApplies = false;
+    }
+    return true;
 }
 
 // ---- state ------------------------------------------------------------------
@@ -226,10 +250,16 @@ void QUnit<R>::MCMtrx(const std::vector<bitLenInt>& controls, const cplx<R>* m, 
     }
     std::vector<bitLenInt> qs(live);
     qs.push_back(t);
-    QInterfacePtr<R> unit = EntangleAll(qs);
-    std::vector<bitLenInt> mc;
-    for (bitLenInt c : live) mc.push_back(shards[c].mapped);
-    unit->MCMtrx(mc, m, shards[t].mapped);
+    try {
+        QInterfacePtr<R> unit = EntangleAll(qs);
+        std::vector<bitLenInt> mc;
+        for (bitLenInt c : live) mc.push_back(shards[c].mapped);
+        unit->MCMtrx(mc, m, shards[t].mapped);
+    } catch (const std::bad_alloc&) {
+        bool applies = false;
+        ElideControls(live, false, applies);
+        if (applies) Mtrx(m, t);
+    }
 }
 
 template <typename R>
@@ -250,10 +280,16 @@ void QUnit<R>::MACMtrx(const std::vector<bitLenInt>& controls, const cplx<R>* m,
     }
     std::vector<bitLenInt> qs(live);
     qs.push_back(t);
-    QInterfacePtr<R> unit = EntangleAll(qs);
-    std::vector<bitLenInt> mc;
-    for (bitLenInt c : live) mc.push_back(shards[c].mapped);
-    unit->MACMtrx(mc, m, shards[t].mapped);
+    try {
+        QInterfacePtr<R> unit = EntangleAll(qs);
+        std::vector<bitLenInt> mc;
+        for (bitLenInt c : live) mc.push_back(shards[c].mapped);
+        unit->MACMtrx(mc, m, shards[t].mapped);
+    } catch (const std::bad_alloc&) {
+        bool applies = false;
+        ElideControls(live, true, applies);
+        if (applies) Mtrx(m, t);
+    }
 }
 
 template <typename R>
@@ -290,10 +326,16 @@ void QUnit<R>::MCPhase(
     }
     std::vector<bitLenInt> qs(live);
     qs.push_back(t);
-    QInterfacePtr<R> unit = EntangleAll(qs);
-    std::vector<bitLenInt> mc;
-    for (bitLenInt c : live) mc.push_back(shards[c].mapped);
-    unit->MCPhase(mc, tl, br, shards[t].mapped);
+    try {
+        QInterfacePtr<R> unit = EntangleAll(qs);
+        std::vector<bitLenInt> mc;
+        for (bitLenInt c : live) mc.push_back(shards[c].mapped);
+        unit->MCPhase(mc, tl, br, shards[t].mapped);
+    } catch (const std::bad_alloc&) {
+        bool applies = false;
+        ElideControls(live, false, applies);
+        if (applies) Phase(tl, br, t);
+    }
 }
 
 template <typename R>
@@ -389,8 +431,20 @@ template <typename R> void QUnit<R>::ISqrtSwap(bitLenInt q1, bitLenInt q2)
 template <typename R> void QUnit<R>::FSim(R theta, R phi, bitLenInt q1, bitLenInt q2)
 {
     if (q1 == q2) throw QrackError("FSim: identical qubits");
-    QInterfacePtr<R> unit = EntangleAll({ q1, q2 });
-    unit->FSim(theta, phi, shards[q1].mapped, shards[q2].mapped);
+    try {
+        QInterfacePtr<R> unit = EntangleAll({ q1, q2 });
+        unit->FSim(theta, phi, shards[q1].mapped, shards[q2].mapped);
+    } catch (const std::bad_alloc&) {
+        // ACE: classically collapse q1 (the fsim "control-like" qubit)
+        bool applies = false;
+        ElideControls({ q1 }, false, applies);
+        // |11>-phase and the swap block degenerate once q1 is classical:
+        if (applies) {
+            // q1 == |1>: fsim acts as phase(-phi) on q2=1 and mixing with the
+            // (now classical) q1 branch collapses to a phase approximation
+            this->MCPhase({ q1 }, cplx<R>(1, 0), polar<R>((R)1, -phi), q2);
+        }
+    }
 }
 
 // ---- measurement -------------------------------------------------------------

@@ -31,6 +31,7 @@ protected:
     EngineFactoryFn<R> subFactory;
     double logFidelity = 0.0;
     R separabilityThreshold;
+    bitLenInt aceMaxQubits = 0; // 0 = unlimited; else entangle cap (ACE)
 
     QInterfacePtr<R> MakeUnit(bitLenInt n, bitCapInt perm)
     {
@@ -46,6 +47,10 @@ protected:
     void FixMappedAfterRemoval(QInterfacePtr<R> unit, bitLenInt removedMapped);
     std::vector<bitLenInt> UnitQubits(QInterfacePtr<R> unit) const; // logical qubits of a unit
     bool ControlShortcut(bitLenInt control, bool anti, bool& alwaysOn);
+    // ACE (parity: qunit.cpp:458-474 + ElideCz): when EntangleAll exceeds the
+    // RAM/width cap (std::bad_alloc), classically collapse the controls with
+    // logFidelity accounting and retry without them
+    bool ElideControls(const std::vector<bitLenInt>& controls, bool anti, bool& gateApplies);
 
 public:
     QUnit(bitLenInt qBitCount, bitCapInt initState = 0u, RngPtr rgp = nullptr,

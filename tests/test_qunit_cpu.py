@@ -235,3 +235,30 @@ def test_sycamore_style_circuit():
     assert_states_close(q.get_state_vector(), cp.get_state_vector(), 2e-4)
     res = q.multi_shot_measure_mask([1 << i for i in range(n)], 100)
     assert sum(res.values()) == 100
+
+
+def test_ace_elision_caps_entanglement():
+    """ACE: with an entangle cap, wide couplers are elided classically and
+    fidelity bookkeeping reflects the approximation (parity model:
+    qunit.cpp bad_alloc -> ElideCz ladder)."""
+    import os
+
+    os.environ["QRACK_QUNIT_ACE_MAX_QB"] = "3"
+    try:
+        n = 6
+        q = make(n, ["qunit", "cpu"], seed=13)
+        for i in range(n):
+            q.h(i)
+        # chain of CZs would entangle everything; the cap forces elision
+        for i in range(n - 1):
+            q.cz(i, i + 1)
+        # still functional: probabilities remain sane and fidelity < 1
+        for i in range(n):
+            p = q.prob(i)
+            assert 0.0 <= p <= 1.0
+        assert q.get_unitary_fidelity() < 1.0
+        assert q.get_unitary_fidelity() > 0.0
+        r = q.m_all()
+        assert 0 <= r < (1 << n)
+    finally:
+        del os.environ["QRACK_QUNIT_ACE_MAX_QB"]
