@@ -1,0 +1,518 @@
+// qrack_amd — QBdt implementation (see qbdt.hpp).
+// All tree operations are persistent (clone-on-write along touched paths),
+// so Clone() shares structure and subtree sharing survives gates.
+#include "qbdt.hpp"
+
+#include "qengine_cpu.hpp"
+
+#include <cstring>
+#include <functional>
+
+#include <cstdlib>
+#include <set>
+
+namespace qrack_amd {
+
+template <typename R> static bool isZero(cplx<R> w) { return norm(w) <= (R)1e-24; }
+
+template <typename R>
+QBdt<R>::QBdt(bitLenInt qBitCount, bitCapInt initState, RngPtr rgp)
+    : QInterface<R>(qBitCount, rgp)
+    , rootWeight(cplx<R>(1, 0))
+{
+    maxNodes = (size_t)1 << 22;
+    if (const char* env = std::getenv("QRACK_QBDT_MAX_NODES")) {
+        maxNodes = (size_t)std::atoll(env);
+    }
+    root = MakeBasis(0, qBitCount, initState);
+}
+
+template <typename R>
+QBdtNodePtr<R> QBdt<R>::MakeBasis(bitLenInt depth, bitLenInt nQubits, bitCapInt perm)
+{
+    if (depth >= nQubits) return nullptr;
+    auto n = std::make_shared<QBdtNode<R>>();
+    const int b = (int)((perm >> depth) & 1u);
+    n->w[b] = cplx<R>(1, 0);
+    n->c[b] = MakeBasis(depth + 1u, nQubits, perm);
+    return n;
+}
+
+template <typename R>
+std::pair<cplx<R>, QBdtNodePtr<R>> QBdt<R>::Add(
+    cplx<R> wa, QBdtNodePtr<R> a, cplx<R> wb, QBdtNodePtr<R> b, bitLenInt depth, bitLenInt nQubits)
+{
+    const bool za = isZero(wa) || (!a && depth < nQubits);
+    const bool zb = isZero(wb) || (!b && depth < nQubits);
+    if (depth >= nQubits) {
+        // terminal scalars
+        const cplx<R> s = (isZero(wa) ? cplx<R>(0, 0) : wa) + (isZero(wb) ? cplx<R>(0, 0) : wb);
+        return { isZero(s) ? cplx<R>(0, 0) : s, nullptr };
+    }
+    if (za && zb) return { cplx<R>(0, 0), nullptr };
+    if (za) return { wb, b };
+    if (zb) return { wa, a };
+    if (a == b) return { wa + wb, a };
+    auto n = std::make_shared<QBdtNode<R>>();
+    bool any = false;
+    for (int k = 0; k < 2; ++k) {
+        auto r = Add(wa * a->w[k], a->c[k], wb * b->w[k], b->c[k], depth + 1u, nQubits);
+        n->w[k] = r.first;
+        n->c[k] = r.second;
+        if (!isZero(r.first)) any = true;
+    }
+    if (!any) return { cplx<R>(0, 0), nullptr };
+    return { cplx<R>(1, 0), n };
+}
+
+// elementwise mix of two target-branch subtrees with remaining deep controls;
+// deepControls entries are (position, requiredBit) sorted ascending
+template <typename R>
+struct DeepCtrl {
+    bitLenInt pos;
+    bool required;
+};
+
+template <typename R>
+void QBdt<R>::PairMix(cplx<R>& wa, QBdtNodePtr<R>& a, cplx<R>& wb, QBdtNodePtr<R>& b,
+    bitLenInt depth, const cplx<R>* m, const std::vector<bitLenInt>& deepPos, size_t ctrlIdx,
+    bitCapInt reqBits, size_t permIdx, bitLenInt nQubits)
+{
+    if (ctrlIdx >= deepPos.size()) {
+        auto r0 = Add(m[0] * wa, a, m[1] * wb, b, depth, nQubits);
+        auto r1 = Add(m[2] * wa, a, m[3] * wb, b, depth, nQubits);
+        wa = r0.first;
+        a = r0.second;
+        wb = r1.first;
+        b = r1.second;
+        return;
+    }
+    if (isZero(wa) && isZero(wb)) return;
+    auto childW = [&](QBdtNodePtr<R>& n, int k) -> cplx<R> { return n ? n->w[k] : cplx<R>(0, 0); };
+    auto childC = [&](QBdtNodePtr<R>& n, int k) -> QBdtNodePtr<R> {
+        return n ? n->c[k] : nullptr;
+    };
+    if (depth == deepPos[ctrlIdx]) {
+        const int r = (int)((reqBits >> permIdx) & 1u);
+        cplx<R> cwa = wa * childW(a, r);
+        QBdtNodePtr<R> cca = childC(a, r);
+        cplx<R> cwb = wb * childW(b, r);
+        QBdtNodePtr<R> ccb = childC(b, r);
+        PairMix(cwa, cca, cwb, ccb, depth + 1u, m, deepPos, ctrlIdx + 1u, reqBits, permIdx + 1u,
+            nQubits);
+        auto na = std::make_shared<QBdtNode<R>>();
+        na->w[r] = cwa;
+        na->c[r] = cca;
+        na->w[1 - r] = wa * childW(a, 1 - r);
+        na->c[1 - r] = childC(a, 1 - r);
+        auto nb = std::make_shared<QBdtNode<R>>();
+        nb->w[r] = cwb;
+        nb->c[r] = ccb;
+        nb->w[1 - r] = wb * childW(b, 1 - r);
+        nb->c[1 - r] = childC(b, 1 - r);
+        wa = (isZero(na->w[0]) && isZero(na->w[1])) ? cplx<R>(0, 0) : cplx<R>(1, 0);
+        a = isZero(wa) ? nullptr : na;
+        wb = (isZero(nb->w[0]) && isZero(nb->w[1])) ? cplx<R>(0, 0) : cplx<R>(1, 0);
+        b = isZero(wb) ? nullptr : nb;
+        return;
+    }
+    // spectator depth below target but above the next deep control
+    auto na = std::make_shared<QBdtNode<R>>();
+    auto nb = std::make_shared<QBdtNode<R>>();
+    for (int k = 0; k < 2; ++k) {
+        cplx<R> cwa = wa * childW(a, k);
+        QBdtNodePtr<R> cca = childC(a, k);
+        cplx<R> cwb = wb * childW(b, k);
+        QBdtNodePtr<R> ccb = childC(b, k);
+        PairMix(cwa, cca, cwb, ccb, depth + 1u, m, deepPos, ctrlIdx, reqBits, permIdx, nQubits);
+        na->w[k] = cwa;
+        na->c[k] = cca;
+        nb->w[k] = cwb;
+        nb->c[k] = ccb;
+    }
+    wa = (isZero(na->w[0]) && isZero(na->w[1])) ? cplx<R>(0, 0) : cplx<R>(1, 0);
+    a = isZero(wa) ? nullptr : na;
+    wb = (isZero(nb->w[0]) && isZero(nb->w[1])) ? cplx<R>(0, 0) : cplx<R>(1, 0);
+    b = isZero(wb) ? nullptr : nb;
+}
+
+template <typename R>
+QBdtNodePtr<R> QBdt<R>::Apply(QBdtNodePtr<R> node, bitLenInt depth, const cplx<R>* m,
+    bitLenInt target, const std::vector<bitLenInt>& controls, bitCapInt controlPerm,
+    cplx<R>& weightInOut)
+{
+    if (isZero(weightInOut) || (!node && depth < qubitCount)) return node;
+    // find whether `depth` is a (shallow) control
+    size_t ci = 0;
+    while (ci < controls.size() && controls[ci] < depth) ++ci;
+    const bool isCtrl = (ci < controls.size()) && (controls[ci] == depth) && (depth < target);
+
+    if (depth == target) {
+        // split remaining deep controls ( > target )
+        std::vector<bitLenInt> deepPos;
+        bitCapInt reqBits = 0;
+        size_t nDeep = 0;
+        for (size_t k = 0; k < controls.size(); ++k) {
+            if (controls[k] > target) {
+                deepPos.push_back(controls[k]);
+                if ((controlPerm >> k) & 1u) reqBits |= (ONE_BCI << nDeep);
+                ++nDeep;
+            }
+        }
+        cplx<R> wa = node->w[0], wb = node->w[1];
+        QBdtNodePtr<R> ca = node->c[0], cb = node->c[1];
+        PairMix(wa, ca, wb, cb, depth + 1u, m, deepPos, 0u, reqBits, 0u, qubitCount);
+        auto n = std::make_shared<QBdtNode<R>>();
+        n->w[0] = wa;
+        n->c[0] = ca;
+        n->w[1] = wb;
+        n->c[1] = cb;
+        if (isZero(n->w[0]) && isZero(n->w[1])) {
+            weightInOut = cplx<R>(0, 0);
+            return nullptr;
+        }
+        return n;
+    }
+    auto n = std::make_shared<QBdtNode<R>>(*node); // shallow clone
+    if (isCtrl) {
+        const int r = (int)((controlPerm >> ci) & 1u);
+        cplx<R> cw = n->w[r];
+        n->c[r] = Apply(n->c[r], depth + 1u, m, target, controls, controlPerm, cw);
+        n->w[r] = cw;
+    } else {
+        for (int k = 0; k < 2; ++k) {
+            cplx<R> cw = n->w[k];
+            n->c[k] = Apply(n->c[k], depth + 1u, m, target, controls, controlPerm, cw);
+            n->w[k] = cw;
+        }
+    }
+    if (isZero(n->w[0]) && isZero(n->w[1])) {
+        weightInOut = cplx<R>(0, 0);
+        return nullptr;
+    }
+    return n;
+}
+
+template <typename R> void QBdt<R>::Mtrx(const cplx<R>* m, bitLenInt target)
+{
+    UCMtrx({}, m, target, 0u);
+}
+
+template <typename R>
+void QBdt<R>::UCMtrx(
+    const std::vector<bitLenInt>& controlsIn, const cplx<R>* m, bitLenInt target, bitCapInt perm)
+{
+    // sort controls ascending, permuting the required-bit mask with them
+    std::vector<std::pair<bitLenInt, bool>> cs;
+    for (size_t i = 0; i < controlsIn.size(); ++i) {
+        cs.push_back({ controlsIn[i], (bool)((perm >> i) & 1u) });
+    }
+    std::sort(cs.begin(), cs.end());
+    std::vector<bitLenInt> controls;
+    bitCapInt sortedPerm = 0;
+    for (size_t i = 0; i < cs.size(); ++i) {
+        controls.push_back(cs[i].first);
+        if (cs[i].second) sortedPerm |= (ONE_BCI << i);
+    }
+    cplx<R> w = rootWeight;
+    root = Apply(root, 0u, m, target, controls, sortedPerm, w);
+    rootWeight = w;
+    CheckGuard();
+}
+
+template <typename R>
+void QBdt<R>::UniformlyControlledSingleBit(
+    const std::vector<bitLenInt>& controls, bitLenInt target, const cplx<R>* mtrxs)
+{
+    const bitCapInt nPerms = pow2((bitLenInt)controls.size());
+    for (bitCapInt p = 0; p < nPerms; ++p) {
+        UCMtrx(controls, mtrxs + 4u * p, target, p);
+    }
+}
+
+// ---- norms / measurement -----------------------------------------------------
+
+template <typename R> double QBdt<R>::SubNorm(QBdtNodePtr<R> node, bitLenInt depth, bitLenInt nQubits)
+{
+    if (depth >= nQubits || !node) return 1.0;
+    double s = 0;
+    for (int k = 0; k < 2; ++k) {
+        if (!isZero(node->w[k])) {
+            s += (double)norm(node->w[k]) * SubNorm(node->c[k], depth + 1u, nQubits);
+        }
+    }
+    return s;
+}
+
+template <typename R> R QBdt<R>::Prob(bitLenInt q)
+{
+    // recursive (total, bit1) accumulation
+    std::function<std::pair<double, double>(QBdtNodePtr<R>, bitLenInt)> rec =
+        [&](QBdtNodePtr<R> node, bitLenInt depth) -> std::pair<double, double> {
+        if (depth >= qubitCount || !node) return { 1.0, 0.0 };
+        double tot = 0, one = 0;
+        for (int k = 0; k < 2; ++k) {
+            if (isZero(node->w[k])) continue;
+            auto sub = rec(node->c[k], depth + 1u);
+            const double wn = (double)norm(node->w[k]);
+            tot += wn * sub.first;
+            if (depth == q) {
+                if (k == 1) one += wn * sub.first;
+            } else {
+                one += wn * sub.second;
+            }
+        }
+        return { tot, one };
+    };
+    auto r = rec(root, 0u);
+    if (r.first <= 0) return 0;
+    return (R)std::min(1.0, std::max(0.0, r.second / r.first));
+}
+
+template <typename R> bool QBdt<R>::ForceM(bitLenInt q, bool result, bool doForce, bool doApply)
+{
+    const R p1 = Prob(q);
+    bool outcome = doForce ? result : (this->Rand() < (double)p1);
+    if (!doApply) return outcome;
+    const R prob = outcome ? p1 : ((R)1 - p1);
+    if (prob <= 0) throw QrackError("QBdt::ForceM: impossible outcome");
+    const int keep = outcome ? 1 : 0;
+    std::function<QBdtNodePtr<R>(QBdtNodePtr<R>, bitLenInt)> collapse =
+        [&](QBdtNodePtr<R> node, bitLenInt depth) -> QBdtNodePtr<R> {
+        if (depth >= qubitCount || !node) return node;
+        auto n = std::make_shared<QBdtNode<R>>(*node);
+        if (depth == q) {
+            n->w[1 - keep] = cplx<R>(0, 0);
+            n->c[1 - keep] = nullptr;
+        } else {
+            for (int k = 0; k < 2; ++k) {
+                if (!isZero(n->w[k])) n->c[k] = collapse(n->c[k], depth + 1u);
+            }
+        }
+        return n;
+    };
+    root = collapse(root, 0u);
+    NormalizeState();
+    return outcome;
+}
+
+template <typename R> void QBdt<R>::NormalizeState(R, R, R)
+{
+    const double total = (double)norm(rootWeight) * SubNorm(root, 0u, qubitCount);
+    if (total <= 0) return;
+    rootWeight = rootWeight * (R)(1.0 / std::sqrt(total));
+}
+
+// ---- state access -------------------------------------------------------------
+
+template <typename R> void QBdt<R>::SetPermutation(bitCapInt perm, cplx<R> phase)
+{
+    root = MakeBasis(0, qubitCount, perm);
+    rootWeight = (norm(phase) > 0) ? phase : cplx<R>(1, 0);
+}
+
+template <typename R>
+void QBdt<R>::Expand(
+    QBdtNodePtr<R> node, bitLenInt depth, cplx<R> weight, bitCapInt prefix, cplx<R>* out) const
+{
+    if (isZero(weight)) return;
+    if (depth >= qubitCount) {
+        out[prefix] = weight;
+        return;
+    }
+    if (!node) return;
+    for (int k = 0; k < 2; ++k) {
+        if (!isZero(node->w[k])) {
+            Expand(node->c[k], depth + 1u, weight * node->w[k],
+                prefix | ((bitCapInt)k << depth), out);
+        }
+    }
+}
+
+template <typename R> void QBdt<R>::GetQuantumState(cplx<R>* outputState)
+{
+    std::memset(outputState, 0, sizeof(cplx<R>) * maxQPower);
+    Expand(root, 0u, rootWeight, 0u, outputState);
+}
+
+template <typename R> cplx<R> QBdt<R>::GetAmplitude(bitCapInt perm)
+{
+    cplx<R> w = rootWeight;
+    QBdtNodePtr<R> n = root;
+    for (bitLenInt d = 0; d < qubitCount; ++d) {
+        if (!n) return cplx<R>(0, 0);
+        const int b = (int)((perm >> d) & 1u);
+        w = w * n->w[b];
+        if (isZero(w)) return cplx<R>(0, 0);
+        n = n->c[b];
+    }
+    return w;
+}
+
+template <typename R>
+QBdtNodePtr<R> QBdt<R>::FromDense(
+    const cplx<R>* amps, bitLenInt depth, bitLenInt nQubits, bitCapInt prefix, cplx<R>& wOut)
+{
+    if (depth >= nQubits) {
+        wOut = amps[prefix];
+        return nullptr;
+    }
+    auto n = std::make_shared<QBdtNode<R>>();
+    bool any = false;
+    for (int k = 0; k < 2; ++k) {
+        cplx<R> w;
+        n->c[k] = FromDense(amps, depth + 1u, nQubits, prefix | ((bitCapInt)k << depth), w);
+        n->w[k] = w;
+        if (!isZero(w)) any = true;
+    }
+    if (!any) {
+        wOut = cplx<R>(0, 0);
+        return nullptr;
+    }
+    wOut = cplx<R>(1, 0);
+    return n;
+}
+
+template <typename R> void QBdt<R>::SetQuantumState(const cplx<R>* inputState)
+{
+    cplx<R> w;
+    root = FromDense(inputState, 0u, qubitCount, 0u, w);
+    rootWeight = w;
+}
+
+// ---- structural ----------------------------------------------------------------
+
+template <typename R> bitLenInt QBdt<R>::Compose(QInterfacePtr<R> toCopy, bitLenInt start)
+{
+    if (start != qubitCount) {
+        throw QrackError("QBdt::Compose supports appending at the top only");
+    }
+    QBdt<R>* o = dynamic_cast<QBdt<R>*>(toCopy.get());
+    QBdtNodePtr<R> otherRoot;
+    cplx<R> otherW;
+    const bitLenInt oQubits = toCopy->GetQubitCount();
+    if (o) {
+        otherRoot = o->root;
+        otherW = o->rootWeight;
+    } else {
+        std::vector<cplx<R>> buf(toCopy->GetMaxQPower());
+        toCopy->GetQuantumState(buf.data());
+        otherRoot = FromDense(buf.data(), 0u, oQubits, 0u, otherW);
+    }
+    // graft the (shared) other tree at every depth-n terminal
+    std::function<QBdtNodePtr<R>(QBdtNodePtr<R>, bitLenInt)> graft =
+        [&](QBdtNodePtr<R> node, bitLenInt depth) -> QBdtNodePtr<R> {
+        if (!node) return node;
+        auto n = std::make_shared<QBdtNode<R>>(*node);
+        for (int k = 0; k < 2; ++k) {
+            if (isZero(n->w[k])) continue;
+            if (depth + 1u >= qubitCount) {
+                n->c[k] = otherRoot;
+            } else {
+                n->c[k] = graft(n->c[k], depth + 1u);
+            }
+        }
+        return n;
+    };
+    root = graft(root, 0u);
+    rootWeight = rootWeight * otherW;
+    this->SetQubitCount(qubitCount + oQubits);
+    return start;
+}
+
+template <typename R> void QBdt<R>::Decompose(bitLenInt start, QInterfacePtr<R> dest)
+{
+    if (qubitCount > 26u) throw QrackError("QBdt::Decompose: dense fallback width cap");
+    const bitLenInt len = dest->GetQubitCount();
+    std::vector<cplx<R>> buf(maxQPower);
+    GetQuantumState(buf.data());
+    auto tmp = std::make_shared<QEngineCPU<R>>(qubitCount, 0u, this->rand_generator);
+    tmp->SetQuantumState(buf.data());
+    tmp->Decompose(start, dest);
+    std::vector<cplx<R>> rem(tmp->GetMaxQPower());
+    tmp->GetQuantumState(rem.data());
+    this->SetQubitCount(qubitCount - len);
+    SetQuantumState(rem.data());
+}
+
+template <typename R> void QBdt<R>::Dispose(bitLenInt start, bitLenInt length)
+{
+    if (qubitCount > 26u) throw QrackError("QBdt::Dispose: dense fallback width cap");
+    std::vector<cplx<R>> buf(maxQPower);
+    GetQuantumState(buf.data());
+    auto tmp = std::make_shared<QEngineCPU<R>>(qubitCount, 0u, this->rand_generator);
+    tmp->SetQuantumState(buf.data());
+    tmp->Dispose(start, length);
+    std::vector<cplx<R>> rem(tmp->GetMaxQPower());
+    tmp->GetQuantumState(rem.data());
+    this->SetQubitCount(qubitCount - length);
+    SetQuantumState(rem.data());
+}
+
+template <typename R>
+void QBdt<R>::Dispose(bitLenInt start, bitLenInt length, bitCapInt disposedPerm)
+{
+    for (bitLenInt i = 0; i < length; ++i) {
+        ForceM(start + i, (disposedPerm >> i) & 1u, true, true);
+    }
+    Dispose(start, length);
+}
+
+template <typename R> bitLenInt QBdt<R>::Allocate(bitLenInt start, bitLenInt length)
+{
+    if (start != qubitCount) throw QrackError("QBdt::Allocate: top append only");
+    auto fresh = std::make_shared<QBdt<R>>(length, 0u, this->rand_generator);
+    Compose(fresh, start);
+    return start;
+}
+
+template <typename R> QInterfacePtr<R> QBdt<R>::Clone()
+{
+    auto c = std::make_shared<QBdt<R>>(qubitCount, 0u, this->rand_generator);
+    c->root = root; // persistent ops: structural sharing is safe
+    c->rootWeight = rootWeight;
+    return c;
+}
+
+template <typename R> double QBdt<R>::SumSqrDiff(QInterfacePtr<R> other)
+{
+    if (other->GetQubitCount() != qubitCount) return 2.0;
+    if (qubitCount > 24u) throw QrackError("QBdt::SumSqrDiff: dense compare cap");
+    std::vector<cplx<R>> a(maxQPower), b(maxQPower);
+    GetQuantumState(a.data());
+    other->GetQuantumState(b.data());
+    double re = 0, im = 0;
+    for (bitCapInt i = 0; i < maxQPower; ++i) {
+        re += (double)(b[i].re * a[i].re + b[i].im * a[i].im);
+        im += (double)(b[i].re * a[i].im - b[i].im * a[i].re);
+    }
+    return std::max(0.0, 2.0 - 2.0 * std::sqrt(re * re + im * im));
+}
+
+// ---- node accounting ------------------------------------------------------------
+
+template <typename R>
+void QBdt<R>::CountNodesRec(QBdtNodePtr<R> n, std::set<const QBdtNode<R>*>& seen)
+{
+    if (!n || seen.count(n.get())) return;
+    seen.insert(n.get());
+    CountNodesRec(n->c[0], seen);
+    CountNodesRec(n->c[1], seen);
+}
+
+template <typename R> size_t QBdt<R>::CountNodes() const
+{
+    std::set<const QBdtNode<R>*> seen;
+    CountNodesRec(root, seen);
+    return seen.size();
+}
+
+template <typename R> void QBdt<R>::CheckGuard() const
+{
+    if (CountNodes() > maxNodes) throw std::bad_alloc();
+}
+
+template class QBdt<float>;
+template class QBdt<double>;
+
+} // namespace qrack_amd

@@ -4,6 +4,8 @@
 #include "qstabilizerhybrid.hpp"
 #include "qhybrid.hpp"
 #include "qinterface_noisy.hpp"
+#include "qbdt.hpp"
+#include "qbdthybrid.hpp"
 #include "qpager.hpp"
 #include "qtensornetwork.hpp"
 #include "qunit.hpp"
@@ -95,6 +97,17 @@ EngineFactoryFn<R> LayerFactory(
         }
         return [rng, cpuF, gpuF](bitLenInt n, bitCapInt perm) -> QInterfacePtr<R> {
             return std::make_shared<QHybrid<R>>(n, perm, rng, cpuF, gpuF);
+        };
+    }
+    if (head == "bdt") {
+        return [rng](bitLenInt n, bitCapInt perm) -> QInterfacePtr<R> {
+            return std::make_shared<QBdt<R>>(n, perm, rng);
+        };
+    }
+    if (head == "bdt_hybrid") {
+        EngineFactoryFn<R> sub = LayerFactory<R>(tail, rng, deviceId, pagesPerDevice);
+        return [rng, sub](bitLenInt n, bitCapInt perm) -> QInterfacePtr<R> {
+            return std::make_shared<QBdtHybridImpl<R>>(n, perm, rng, sub);
         };
     }
     if (head == "pager") {
