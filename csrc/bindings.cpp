@@ -214,6 +214,8 @@ template <typename R> static void bindQInterface(py::module_& m, const char* nam
         .def("incc", &QI::INCC)
         .def("decc", &QI::DECC)
         .def("incs", &QI::INCS)
+        .def("incbcd", &QI::INCBCD)
+        .def("decbcd", &QI::DECBCD)
         .def("mul", &QI::MUL)
         .def("div", &QI::DIV)
         .def("mul_mod_n_out", &QI::MULModNOut)

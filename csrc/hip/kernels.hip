@@ -786,6 +786,30 @@ template <typename R> __global__ void k_permute(const cplx<R>* sv, cplx<R>* nsv,
             nsv[res] = sv[i];
             break;
         }
+        case PermOp::INCBCD: {
+            const int digits = (int)(a.length / 4u);
+            bitCapInt value = 0, mul = 1, tenPow = 1;
+            bool valid = true;
+            for (int d = 0; d < digits; ++d) {
+                const bitCapInt digit = (reg >> (4 * d)) & 0xFu;
+                if (digit > 9u) valid = false;
+                value += digit * mul;
+                mul *= 10u;
+                tenPow *= 10u;
+            }
+            if (!valid) {
+                nsv[i] = sv[i];
+                break;
+            }
+            bitCapInt out = (value + a.operand) % tenPow;
+            bitCapInt enc = 0;
+            for (int d = 0; d < digits; ++d) {
+                enc |= (out % 10u) << (4 * d);
+                out /= 10u;
+            }
+            nsv[(i & ~regMask) | (enc << a.start)] = sv[i];
+            break;
+        }
         case PermOp::ROL: {
             const bitLenInt shift = (bitLenInt)a.operand;
             const bitCapInt nreg = ((reg << shift) | (reg >> (a.length - shift))) & lenMask;

@@ -41,6 +41,7 @@ enum class PermOp : int {
     ADC,
     SBC,
     ROL,
+    INCBCD,
 };
 
 struct PermArgs {

@@ -1000,6 +1000,25 @@ void QEngineHIP<R>::INCS(bitCapInt toAdd, bitLenInt start, bitLenInt length, bit
 }
 
 template <typename R>
+void QEngineHIP<R>::INCBCD(bitCapInt toAdd, bitLenInt start, bitLenInt length)
+{
+    if (length % 4u) throw QrackError("INCBCD: length must be a multiple of 4");
+    checkAluRangeHip<R>(start, length, qubitCount, "INCBCD");
+    const bitLenInt digits = length / 4u;
+    bitCapInt tenPow = 1;
+    for (bitLenInt i = 0; i < digits; ++i) tenPow *= 10u;
+    toAdd %= tenPow;
+    if (!toAdd) return;
+    PermArgs a{};
+    a.op = (int)PermOp::INCBCD;
+    a.maxI = maxQPower;
+    a.start = start;
+    a.length = length;
+    a.operand = toAdd;
+    permuteOp(a, false, false);
+}
+
+template <typename R>
 void QEngineHIP<R>::MUL(bitCapInt toMul, bitLenInt inOutStart, bitLenInt carryStart, bitLenInt length)
 {
     if (!toMul) throw QrackError("MUL by zero is not invertible");

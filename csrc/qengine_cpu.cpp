@@ -784,6 +784,39 @@ void QEngineCPU<R>::INCS(bitCapInt toAdd, bitLenInt start, bitLenInt length, bit
 }
 
 template <typename R>
+void QEngineCPU<R>::INCBCD(bitCapInt toAdd, bitLenInt start, bitLenInt length)
+{
+    // decimal add on 4-bit digits; invalid (non-BCD) codes map to themselves
+    // so the permutation stays a bijection (parity: qheader_bcd.cl incbcd)
+    if (length % 4u) throw QrackError("INCBCD: length must be a multiple of 4");
+    const bitLenInt digits = length / 4u;
+    bitCapInt tenPow = 1;
+    for (bitLenInt i = 0; i < digits; ++i) tenPow *= 10u;
+    toAdd %= tenPow;
+    if (!toAdd) return;
+    const bitCapInt regMask = pow2Mask(length) << start;
+    const bitCapInt addVal = toAdd;
+    PermutationOp([=](bitCapInt i) {
+        const bitCapInt reg = (i & regMask) >> start;
+        // decode BCD; invalid digit -> identity
+        bitCapInt value = 0, mul = 1;
+        for (bitLenInt d = 0; d < digits; ++d) {
+            const bitCapInt digit = (reg >> (4u * d)) & 0xFu;
+            if (digit > 9u) return i;
+            value += digit * mul;
+            mul *= 10u;
+        }
+        bitCapInt out = (value + addVal) % tenPow;
+        bitCapInt enc = 0;
+        for (bitLenInt d = 0; d < digits; ++d) {
+            enc |= (out % 10u) << (4u * d);
+            out /= 10u;
+        }
+        return (i & ~regMask) | (enc << start);
+    });
+}
+
+template <typename R>
 void QEngineCPU<R>::MUL(bitCapInt toMul, bitLenInt inOutStart, bitLenInt carryStart, bitLenInt length)
 {
     checkAluRange<R>(inOutStart, length, qubitCount, "MUL");

@@ -98,6 +98,7 @@ public:
     void INCC(bitCapInt toAdd, bitLenInt start, bitLenInt length, bitLenInt carryIndex) override;
     void DECC(bitCapInt toSub, bitLenInt start, bitLenInt length, bitLenInt carryIndex) override;
     void INCS(bitCapInt toAdd, bitLenInt start, bitLenInt length, bitLenInt overflowIndex) override;
+    void INCBCD(bitCapInt toAdd, bitLenInt start, bitLenInt length) override;
     void MUL(bitCapInt toMul, bitLenInt inOutStart, bitLenInt carryStart, bitLenInt length) override;
     void DIV(bitCapInt toDiv, bitLenInt inOutStart, bitLenInt carryStart, bitLenInt length) override;
     void MULModNOut(bitCapInt toMul, bitCapInt modN, bitLenInt inStart, bitLenInt outStart,

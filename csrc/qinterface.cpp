@@ -531,6 +531,16 @@ template <typename R> void QInterface<R>::DECS(bitCapInt toSub, bitLenInt start,
 {
     INCS((pow2(length) - toSub) & pow2Mask(length), start, length, overflowIndex);
 }
+template <typename R> void QInterface<R>::INCBCD(bitCapInt, bitLenInt, bitLenInt) { aluThrow<R>(); }
+template <typename R> void QInterface<R>::DECBCD(bitCapInt toSub, bitLenInt start, bitLenInt length)
+{
+    // subtract d == add (10^k - d) in BCD
+    const bitLenInt digits = length / 4u;
+    bitCapInt tenPow = 1;
+    for (bitLenInt i = 0; i < digits; ++i) tenPow *= 10u;
+    INCBCD((tenPow - toSub) % tenPow, start, length);
+}
+
 template <typename R> void QInterface<R>::MUL(bitCapInt, bitLenInt, bitLenInt, bitLenInt)
 {
     aluThrow<R>();

@@ -227,3 +227,29 @@ def test_zero_phase_flip():
     sv = q.get_state_vector()
     assert sv[0].real < 0
     assert sv[1].real > 0
+
+
+def test_incbcd():
+    q = make(8)
+    # BCD "27" = 0x27
+    set_reg(q, 0, 8, 0x27)
+    q.incbcd(15, 0, 8)
+    assert q.m_reg(0, 8) == 0x42  # 27 + 15 = 42 in BCD
+    q2 = make(8)
+    set_reg(q2, 0, 8, 0x99)
+    q2.incbcd(1, 0, 8)
+    assert q2.m_reg(0, 8) == 0x00  # wraps mod 100
+
+
+def test_decbcd():
+    q = make(8)
+    set_reg(q, 0, 8, 0x42)
+    q.decbcd(15, 0, 8)
+    assert q.m_reg(0, 8) == 0x27
+
+
+def test_incbcd_superposition():
+    q = make(5, seed=4)
+    q.h(4)  # superpose a spectator qubit
+    q.incbcd(3, 0, 4)
+    assert q.m_reg(0, 4) == 0x3
