@@ -5,6 +5,7 @@
 #include "qhybrid.hpp"
 #include "qinterface_noisy.hpp"
 #include "qbdt.hpp"
+#include "qengine_sparse.hpp"
 #include "qbdthybrid.hpp"
 #include "qpager.hpp"
 #include "qtensornetwork.hpp"
@@ -58,6 +59,11 @@ EngineFactoryFn<R> LayerFactory(
     if (head == "hip") {
         return [rng, deviceId](bitLenInt n, bitCapInt perm) -> QInterfacePtr<R> {
             return MakeHipEngine<R>(n, perm, rng, deviceId);
+        };
+    }
+    if (head == "sparse") {
+        return [rng](bitLenInt n, bitCapInt perm) -> QInterfacePtr<R> {
+            return std::make_shared<QEngineSparse<R>>(n, perm, rng);
         };
     }
     if (head == "stabilizer") {

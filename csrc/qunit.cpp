@@ -485,11 +485,45 @@ template <typename R>
 std::map<bitCapInt, int> QUnit<R>::MultiShotMeasureMask(
     const std::vector<bitCapInt>& qPowers, unsigned shots)
 {
-    QUnitPtr<R> clone = std::static_pointer_cast<QUnit<R>>(Clone());
-    std::vector<bitLenInt> all(qubitCount);
-    for (bitLenInt q = 0; q < qubitCount; ++q) all[q] = q;
-    QInterfacePtr<R> unit = clone->EntangleOrdered(all);
-    return unit->MultiShotMeasureMask(qPowers, shots);
+    // units are independent subsystems: sample each separately and combine
+    // shots elementwise (no entanglement, no width blow-up)
+    if (!shots) return {};
+    std::vector<bitCapInt> joint(shots, 0u);
+    std::set<QInterfacePtr<R>> seen;
+    for (size_t b = 0; b < qPowers.size(); ++b) {
+        const bitLenInt q = log2Ocl(qPowers[b]);
+        QInterfacePtr<R> u = shards[q].unit;
+        if (seen.count(u)) continue;
+        seen.insert(u);
+        // output-bit <-> local-power mapping for every masked qubit of u
+        std::vector<size_t> outBits;
+        std::vector<bitCapInt> localPowers;
+        for (size_t k = 0; k < qPowers.size(); ++k) {
+            const bitLenInt lq = log2Ocl(qPowers[k]);
+            if (shards[lq].unit == u) {
+                outBits.push_back(k);
+                localPowers.push_back(pow2(shards[lq].mapped));
+            }
+        }
+        auto res = u->MultiShotMeasureMask(localPowers, shots);
+        std::vector<bitCapInt> draws;
+        draws.reserve(shots);
+        for (auto& kv : res) {
+            for (int c = 0; c < kv.second; ++c) draws.push_back(kv.first);
+        }
+        // shuffle so the elementwise pairing across units is i.i.d.
+        for (size_t i = draws.size(); i > 1; --i) {
+            std::swap(draws[i - 1], draws[(size_t)(this->Rand() * i) % i]);
+        }
+        for (unsigned s = 0; s < shots; ++s) {
+            for (size_t k = 0; k < outBits.size(); ++k) {
+                if ((draws[s] >> k) & 1u) joint[s] |= (ONE_BCI << outBits[k]);
+            }
+        }
+    }
+    std::map<bitCapInt, int> results;
+    for (unsigned s = 0; s < shots; ++s) results[joint[s]]++;
+    return results;
 }
 
 template <typename R> R QUnit<R>::ProbMask(bitCapInt mask, bitCapInt permutation)

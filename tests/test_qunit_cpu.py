@@ -262,3 +262,22 @@ def test_ace_elision_caps_entanglement():
         assert 0 <= r < (1 << n)
     finally:
         del os.environ["QRACK_QUNIT_ACE_MAX_QB"]
+
+
+def test_multishot_no_entangle_blowup():
+    """MultiShot must sample per independent unit (no full entangle):
+    with an ACE cap, the clone-free path keeps units small."""
+    import os
+
+    os.environ["QRACK_QUNIT_ACE_MAX_QB"] = "4"
+    try:
+        n = 26
+        q = make(n, ["qunit", "cpu"], seed=15)
+        for i in range(n):
+            q.h(i)
+        for i in range(n - 1):
+            q.cz(i, i + 1)
+        res = q.multi_shot_measure_mask([1 << i for i in range(n)], 50)
+        assert sum(res.values()) == 50
+    finally:
+        del os.environ["QRACK_QUNIT_ACE_MAX_QB"]
