@@ -163,6 +163,52 @@ template <int KIND, bool NT> __global__ void k_apply2x2_1v(cplx<float>* sv, Gate
     }
 }
 
+// wide fp32 single-target variant: FOUR adjacent pairs (two float4 per side,
+// 32 B/lane/side) — A/B candidate for the HBM-bound H stream
+template <int KIND> __global__ void k_apply2x2_1w(cplx<float>* sv, GateArgs<float> a)
+{
+    float4* sv4 = reinterpret_cast<float4*>(sv);
+    const bitCapInt p = a.qPowers[0];
+    const bitCapInt quarter = a.maxI >> 2u;
+    const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
+    const cplx<float> m0 = a.m[0], m1 = a.m[1], m2 = a.m[2], m3 = a.m[3];
+    for (bitCapInt k = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; k < quarter; k += stride) {
+        const bitCapInt j = 4u * k;
+        const bitCapInt i = ((j & ~(p - 1u)) << 1u) | (j & (p - 1u));
+        const bitCapInt lo4 = i >> 1u;
+        const bitCapInt hi4 = (i + p) >> 1u;
+#pragma unroll
+        for (int h = 0; h < 2; ++h) {
+            float4 vlo = sv4[lo4 + h];
+            float4 vhi = sv4[hi4 + h];
+            const cplx<float> x0{ vlo.x, vlo.y }, x1{ vlo.z, vlo.w };
+            const cplx<float> y0{ vhi.x, vhi.y }, y1{ vhi.z, vhi.w };
+            cplx<float> a0, a1, b0, b1;
+            if (KIND == 1) {
+                a0 = m0 * x0; a1 = m0 * x1;
+                b0 = m3 * y0; b1 = m3 * y1;
+            } else if (KIND == 2) {
+                a0 = m1 * y0; a1 = m1 * y1;
+                b0 = m2 * x0; b1 = m2 * x1;
+            } else {
+                a0 = m0 * x0 + m1 * y0; a1 = m0 * x1 + m1 * y1;
+                b0 = m2 * x0 + m3 * y0; b1 = m2 * x1 + m3 * y1;
+            }
+            sv4[lo4 + h] = make_float4(a0.re, a0.im, a1.re, a1.im);
+            sv4[hi4 + h] = make_float4(b0.re, b0.im, b1.re, b1.im);
+        }
+    }
+}
+
+static inline int wideMode()
+{
+    static int v = [] {
+        if (const char* env = std::getenv("QRACK_GPU_WIDE")) return std::atoi(env);
+        return 0;
+    }();
+    return v;
+}
+
 // one-sided diagonal scale: sv[i|offset] *= f for every expanded i.
 // The dominant QFT kernel: CPhase(topLeft=1) touches only the
 // control=1 & target=1 quarter of the state (the reference's phasesingle
@@ -266,6 +312,20 @@ void launchApply2x2(cplx<R>* sv, const GateArgs<R>& a, hipStream_t stream)
     if constexpr (std::is_same_v<R, float>) {
         if (a.nPowers == 1 && (a.maxI & 1u) == 0u && a.maxI >= 2u && a.offset1 == 0u) {
             // single-target fast path (handles target bit 0 too)
+            if (wideMode() && a.qPowers[0] >= 4u && (a.maxI & 3u) == 0u) {
+                const int gridw = gridFor(a.maxI >> 2u);
+                switch (kind) {
+                case 1:
+                    hipLaunchKernelGGL((k_apply2x2_1w<1>), dim3(gridw), dim3(QA_BLOCK), 0, stream, sv, a);
+                    return;
+                case 2:
+                    hipLaunchKernelGGL((k_apply2x2_1w<2>), dim3(gridw), dim3(QA_BLOCK), 0, stream, sv, a);
+                    return;
+                default:
+                    hipLaunchKernelGGL((k_apply2x2_1w<0>), dim3(gridw), dim3(QA_BLOCK), 0, stream, sv, a);
+                    return;
+                }
+            }
             const int gridv = gridFor(a.maxI >> 1u);
             if (useNontemporal()) {
                 switch (kind) {
