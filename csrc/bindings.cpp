@@ -301,6 +301,12 @@ template <typename R> static void bindQInterface(py::module_& m, const char* nam
                 eng->ZeroAmplitudes();
             })
         .def("global_phase", [](QI& q, C f) { q.Phase(from_std<R>(f), from_std<R>(f), 0); })
+        .def("phase_ramp",
+            [](Ptr q, R scale, bitLenInt rampStart, bitLenInt rampBits, bitCapInt condPower) {
+                auto eng = std::dynamic_pointer_cast<QEngine<R>>(q);
+                if (!eng) throw QrackError("phase_ramp requires a state-vector engine");
+                eng->PhaseRamp(scale, rampStart, rampBits, condPower);
+            })
         .def("norm_total",
             [](Ptr q) {
                 auto eng = std::dynamic_pointer_cast<QEngine<R>>(q);

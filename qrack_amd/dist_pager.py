@@ -328,19 +328,40 @@ class DistQPager:
         return v
 
     # ---- QFT ------------------------------------------------------------------
+    # column-fused: the controlled-phase ladder of column i collapses to ONE
+    # per-rank PhaseRamp kernel (+ a per-page scalar for meta bits); the only
+    # communication is the H exchange on meta columns (mirrors QPager::QFT in
+    # csrc/qpager.cpp)
+
+    def _qft_column_ramp(self, start, i, sign):
+        t = start + i
+        scale = sign * np.pi / (1 << i)
+        if t < self.qpp:
+            self.q.phase_ramp(scale, start, i, 1 << t)
+            return
+        tb = t - self.qpp
+        if not ((self.my_page >> tb) & 1):
+            return
+        intra_bits = (self.qpp - start) if start < self.qpp else 0
+        meta_start = 0 if start < self.qpp else start - self.qpp
+        if intra_bits:
+            self.q.phase_ramp(scale, start, intra_bits, 0)
+        nbits = tb - meta_start
+        meta_val = (self.my_page >> meta_start) & ((1 << nbits) - 1) if nbits > 0 else 0
+        if meta_val:
+            theta = scale * (meta_val << intra_bits)
+            self.q.global_phase(complex(np.exp(1j * theta)))
 
     def qft(self, start, length):
         for i in range(length - 1, -1, -1):
             self.h(start + i)
-            for j in range(i):
-                self.cphase_root_n(i - j + 1, start + j, start + i)
+            if i:
+                self._qft_column_ramp(start, i, +1)
 
     def iqft(self, start, length):
         for i in range(length):
-            for j in range(i - 1, -1, -1):
-                self.mcphase(
-                    [start + j], 1, np.exp(-1j * np.pi / (1 << (i - j))), start + i
-                )
+            if i:
+                self._qft_column_ramp(start, i, -1)
             self.h(start + i)
 
     # ---- measurement ----------------------------------------------------------
