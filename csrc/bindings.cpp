@@ -473,5 +473,18 @@ PYBIND11_MODULE(_qrack, m)
         py::arg("pages_per_device") = (bitLenInt)1);
 
     m.def("hip_device_count", &HipDeviceCount);
+#ifdef QRACK_AMD_HIP_ENGINE
+    m.def("profile_report", []() {
+        py::dict d;
+        for (auto& kv : HipProfiler::Report()) {
+            d[py::str(kv.first)] = py::make_tuple(kv.second.first, kv.second.second);
+        }
+        return d;
+    });
+    m.def("profile_reset", []() { HipProfiler::Reset(); });
+#else
+    m.def("profile_report", []() { return py::dict(); });
+    m.def("profile_reset", []() {});
+#endif
     m.attr("__version__") = "0.1.0";
 }

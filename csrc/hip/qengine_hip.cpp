@@ -3,10 +3,71 @@
 
 #include "../qfactory.hpp"
 
+#include <chrono>
+#include <mutex>
+
 #include <algorithm>
 #include <cstring>
 
 namespace qrack_amd {
+
+// ---- profiler ---------------------------------------------------------------
+
+namespace {
+std::mutex g_profMtx;
+std::map<std::string, std::pair<uint64_t, double>> g_prof;
+double nowMs()
+{
+    return std::chrono::duration<double, std::milli>(
+        std::chrono::steady_clock::now().time_since_epoch())
+        .count();
+}
+} // namespace
+
+bool HipProfiler::Enabled()
+{
+    static bool v = [] {
+        const char* env = std::getenv("QRACK_PROFILE");
+        return env && std::atoi(env) != 0;
+    }();
+    return v;
+}
+
+void HipProfiler::Add(const char* op, double ms)
+{
+    std::lock_guard<std::mutex> lk(g_profMtx);
+    auto& e = g_prof[op];
+    e.first++;
+    e.second += ms;
+}
+
+std::map<std::string, std::pair<uint64_t, double>> HipProfiler::Report()
+{
+    std::lock_guard<std::mutex> lk(g_profMtx);
+    return g_prof;
+}
+
+void HipProfiler::Reset()
+{
+    std::lock_guard<std::mutex> lk(g_profMtx);
+    g_prof.clear();
+}
+
+HipProfScope::HipProfScope(const char* op, hipStream_t stream)
+    : op_(op)
+    , stream_(stream)
+{
+    if (!HipProfiler::Enabled()) return;
+    hipStreamSynchronize(stream_);
+    t0_ = nowMs();
+}
+
+HipProfScope::~HipProfScope()
+{
+    if (!HipProfiler::Enabled()) return;
+    hipStreamSynchronize(stream_);
+    HipProfiler::Add(op_, nowMs() - t0_);
+}
 
 // ---- device tracker ---------------------------------------------------------
 
@@ -333,6 +394,7 @@ void QEngineHIP<R>::Apply2x2(bitCapInt offset1, bitCapInt offset2, const cplx<R>
 {
     GateArgs<R> a = makeGateArgs(offset1, offset2, mtrx, qPowersSorted);
     QA_HIP_CHECK(hipSetDevice(deviceId));
+    HipProfScope prof("apply2x2", stream);
     launchApply2x2<R>(dState, a, stream);
 }
 
@@ -388,6 +450,7 @@ template <typename R>
 void QEngineHIP<R>::PhaseRamp(R scale, bitLenInt rampStart, bitLenInt rampBits, bitCapInt condPower)
 {
     QA_HIP_CHECK(hipSetDevice(deviceId));
+    HipProfScope prof("phase_ramp", stream);
     launchPhaseRamp<R>(dState, maxQPower, rampStart, rampBits, condPower, (double)scale, stream);
 }
 
@@ -417,6 +480,7 @@ template <typename R> void QEngineHIP<R>::IQFT(bitLenInt start, bitLenInt length
 template <typename R> double QEngineHIP<R>::reduceSum(int op, const ReduceArgs& a)
 {
     QA_HIP_CHECK(hipSetDevice(deviceId));
+    HipProfScope prof("reduce", stream);
     const int grid = launchReduce<R>(dState, a, op, dPartials, stream);
     QA_HIP_CHECK(hipMemcpyAsync(
         hPartials.data(), dPartials, grid * sizeof(double), hipMemcpyDeviceToHost, stream));
@@ -872,6 +936,7 @@ template <typename R> QInterfacePtr<R> QEngineHIP<R>::Clone()
 template <typename R> void QEngineHIP<R>::permuteOp(PermArgs& a, bool partialSpace, bool copyFirst)
 {
     QA_HIP_CHECK(hipSetDevice(deviceId));
+    HipProfScope prof("alu_permute", stream);
     ensureScratch();
     if (copyFirst) {
         QA_HIP_CHECK(hipMemcpyAsync(

@@ -50,6 +50,27 @@ private:
     int count_ = 0;
 };
 
+// opt-in per-op wall-clock profiler (QRACK_PROFILE=1): synchronous timing
+// around each launch category, reported via qrack_amd.profile_report()
+// (aux-subsystem parity: SURVEY.md §5 tracing — HIP-event/QueueItem timing)
+struct HipProfiler {
+    static bool Enabled();
+    static void Add(const char* op, double ms);
+    static std::map<std::string, std::pair<uint64_t, double>> Report();
+    static void Reset();
+};
+
+class HipProfScope {
+public:
+    HipProfScope(const char* op, hipStream_t stream);
+    ~HipProfScope();
+
+private:
+    const char* op_;
+    hipStream_t stream_;
+    double t0_ = 0;
+};
+
 template <typename R> class QEngineHIP;
 template <typename R> using QEngineHIPPtr = std::shared_ptr<QEngineHIP<R>>;
 

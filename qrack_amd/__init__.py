@@ -21,6 +21,8 @@ from qrack_amd._qrack import (  # noqa: F401
     create,
     create_d,
     hip_device_count,
+    profile_report,
+    profile_reset,
     save_stabilizer_F,
     save_stabilizer_D,
     load_stabilizer_F,
