@@ -27,8 +27,13 @@ typedef uint64_t bitCapInt;  // amplitude index / bit mask
 
 constexpr bitCapInt ONE_BCI = 1u;
 
-inline bitCapInt pow2(bitLenInt p) { return ONE_BCI << p; }
-inline bitCapInt pow2Mask(bitLenInt p) { return (ONE_BCI << p) - 1u; }
+// NOTE: indices are 64-bit; layers that hold factorized/compressed states
+// (QUnit, QStabilizer, sparse) support MORE than 63 qubits as long as any
+// single dense mask stays under 64 bits. pow2 saturates (all-ones) past
+// that so the overflow is defined; packed >63-bit measurement results are
+// out of range by design (use per-qubit reads there).
+inline bitCapInt pow2(bitLenInt p) { return (p >= 64u) ? ~(bitCapInt)0u : (ONE_BCI << p); }
+inline bitCapInt pow2Mask(bitLenInt p) { return (p >= 64u) ? ~(bitCapInt)0u : ((ONE_BCI << p) - 1u); }
 inline bitLenInt log2Ocl(bitCapInt n) {
     bitLenInt r = 0;
     while (n >>= 1) r++;

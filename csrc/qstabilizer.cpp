@@ -597,6 +597,36 @@ template <typename R> cplx<R> QStabilizer<R>::ampPhase(int phase) const
     }
 }
 
+template <typename R>
+void QStabilizer<R>::ForEachNonzeroAmplitude(const std::function<void(bitCapInt, cplx<R>)>& fn)
+{
+    const bitLenInt n = qubitCount;
+    const bitLenInt g = gaussian();
+    const size_t scratch = 2u * (size_t)n;
+    bitCapInt basis;
+    int ph;
+    seed(g, basis, ph);
+    const R nrm = (R)(1.0 / std::sqrt((double)pow2(g)));
+    auto emit = [&]() {
+        int e = rPhase[scratch];
+        bitCapInt b = 0;
+        for (bitLenInt j = 0; j < n; ++j) {
+            const bool x = getX(scratch, j);
+            if (x && getZ(scratch, j)) e = (e + 1) % 4;
+            if (x) b |= (ONE_BCI << j);
+        }
+        fn(b, phaseOffset * (ampPhase(e) * nrm));
+    };
+    emit();
+    for (bitCapInt t = 0; t < pow2(g) - 1u; ++t) {
+        const bitCapInt t2 = t ^ (t + 1u);
+        for (bitLenInt i = 0; i < g; ++i) {
+            if ((t2 >> i) & 1u) rowsum(scratch, (size_t)n + i);
+        }
+        emit();
+    }
+}
+
 template <typename R> void QStabilizer<R>::GetQuantumState(cplx<R>* outputState)
 {
     const bitLenInt n = qubitCount;

@@ -124,6 +124,8 @@ public:
     // ---- state access ----
     void SetQuantumState(const cplx<R>* inputState) override;
     void GetQuantumState(cplx<R>* outputState) override;
+    // enumerate the 2^g nonzero amplitudes without materializing 2^n
+    void ForEachNonzeroAmplitude(const std::function<void(bitCapInt, cplx<R>)>& fn);
     cplx<R> GetAmplitude(bitCapInt perm) override;
     void SetAmplitude(bitCapInt, cplx<R>) override
     {

@@ -91,11 +91,22 @@ template <typename R> void QStabilizerHybrid<R>::SwitchToEngine()
 {
     if (engine) return;
     // materialize the tableau state into a fresh state-vector engine
-    // (parity: qstabilizerhybrid.cpp:435-511)
+    // (parity: qstabilizerhybrid.cpp:435-511). Wide states skip the dense
+    // 2^n buffer: the tableau's 2^g nonzero amplitudes stream directly into
+    // the engine (the sparse engine handles thousands of qubits this way).
     engine = engineFactory(qubitCount, 0u);
-    std::vector<cplx<R>> buf(maxQPower);
-    stabilizer->GetQuantumState(buf.data());
-    engine->SetQuantumState(buf.data());
+    if (qubitCount <= 26u) {
+        std::vector<cplx<R>> buf(maxQPower);
+        stabilizer->GetQuantumState(buf.data());
+        engine->SetQuantumState(buf.data());
+    } else {
+        auto eng = std::dynamic_pointer_cast<QEngine<R>>(engine);
+        if (!eng) throw QrackError("wide stabilizer switch needs an engine backend");
+        eng->ZeroAmplitudes();
+        QInterfacePtr<R> e = engine;
+        stabilizer->ForEachNonzeroAmplitude(
+            [&](bitCapInt idx, cplx<R> amp) { e->SetAmplitude(idx, amp); });
+    }
     stabilizer.reset();
     // flush every pending shard into the engine
     for (bitLenInt q = 0; q < qubitCount; ++q) {
