@@ -37,3 +37,12 @@ clean:
 	rm -rf $(BUILD) qrack_amd/_qrack*.so
 
 .PHONY: all clean
+
+# sanitizer lane: CPU-only battery under ASAN+UBSAN (no GPU needed)
+asan:
+	g++ -O1 -g -std=c++17 -fsanitize=address,undefined -fno-omit-frame-pointer \
+	    -Icsrc csrc/qinterface.cpp csrc/qengine_cpu.cpp csrc/qengine_sparse.cpp \
+	    csrc/qstabilizer.cpp csrc/qstabilizerhybrid.cpp csrc/qunit.cpp csrc/qbdt.cpp \
+	    csrc/qpager.cpp csrc/qfactory.cpp csrc/common/parallel_for.cpp \
+	    tools/asan_smoke.cpp -o build/asan_smoke -lpthread
+	./build/asan_smoke

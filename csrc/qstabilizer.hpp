@@ -12,6 +12,7 @@
 #include "qinterface.hpp"
 
 #include <cstring>
+#include <functional>
 
 namespace qrack_amd {
 
