@@ -13,7 +13,8 @@ import os
 import numpy as np
 import pytest
 
-PORT = 29541
+import os as _os
+PORT = 20000 + (_os.getpid() % 8000)
 
 
 def _run_worker(rank, world, fn_name, q, seed, port):

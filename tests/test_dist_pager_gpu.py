@@ -17,7 +17,8 @@ import qrack_amd as qa
 
 pytestmark = pytest.mark.gpu
 
-PORT = 29700
+import os as _os
+PORT = 28000 + (_os.getpid() % 8000)
 
 
 def _worker(rank, world, qubits, seed, port, fn_name):
