@@ -307,6 +307,14 @@ template <typename R> static void bindQInterface(py::module_& m, const char* nam
                 if (!eng) throw QrackError("phase_ramp requires a state-vector engine");
                 eng->PhaseRamp(scale, rampStart, rampBits, condPower);
             })
+        .def("phase_ramp_general",
+            [](Ptr q, R scale, bitLenInt rampStart, bitCapInt inPlaceRelMask,
+                std::vector<bitCapInt> sPows, std::vector<uint64_t> sWeights,
+                bitCapInt condPower) {
+                auto eng = std::dynamic_pointer_cast<QEngine<R>>(q);
+                if (!eng) throw QrackError("phase_ramp_general requires an engine");
+                eng->PhaseRampGeneral(scale, rampStart, inPlaceRelMask, sPows, sWeights, condPower);
+            })
         .def("norm_total",
             [](Ptr q) {
                 auto eng = std::dynamic_pointer_cast<QEngine<R>>(q);

@@ -978,6 +978,33 @@ void launchPhaseRamp(cplx<R>* sv, bitCapInt maxQPower, bitLenInt rampStart, bitL
         maxI, condPower, rampStart, rampMask, (R)scale);
 }
 
+template <typename R> __global__ void k_phase_ramp_gen(cplx<R>* sv, bitCapInt maxI, RampArgs a)
+{
+    const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
+    for (bitCapInt j = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; j < maxI; j += stride) {
+        bitCapInt i = j;
+        if (a.condPow) {
+            i = (((j & ~(a.condPow - 1u)) << 1u) | (j & (a.condPow - 1u))) | a.condPow;
+        }
+        uint64_t frac = (uint64_t)((i >> a.rampStart) & a.inPlaceRelMask);
+        for (int k = 0; k < a.nScattered; ++k) {
+            if (i & a.sPow[k]) frac += a.sWeight[k];
+        }
+        const R theta = (R)a.scale * (R)frac;
+        R s, c;
+        devSinCos<R>(theta, &s, &c);
+        sv[i] = cplx<R>{ c, s } * sv[i];
+    }
+}
+
+template <typename R>
+void launchPhaseRampGeneral(cplx<R>* sv, bitCapInt maxQPower, const RampArgs& a, hipStream_t stream)
+{
+    const bitCapInt maxI = a.condPow ? (maxQPower >> 1u) : maxQPower;
+    hipLaunchKernelGGL(
+        (k_phase_ramp_gen<R>), dim3(gridFor(maxI)), dim3(QA_BLOCK), 0, stream, sv, maxI, a);
+}
+
 // ---- sampling / inner product / marginals --------------------------------------
 
 template <typename R>
@@ -1131,7 +1158,8 @@ void launchPartProbs(const cplx<R>* sv, bitCapInt maxQPower, bitLenInt start, bi
         const cplx<R>*, const cplx<R>*, bitCapInt, double*, double*, hipStream_t);                  \
     template void launchPartProbs<R>(                                                               \
         const cplx<R>*, bitCapInt, bitLenInt, bitLenInt, double*, hipStream_t);                     \
-    template void launchPhaseRamp<R>(cplx<R>*, bitCapInt, bitLenInt, bitLenInt, bitCapInt, double, hipStream_t);
+    template void launchPhaseRamp<R>(cplx<R>*, bitCapInt, bitLenInt, bitLenInt, bitCapInt, double, hipStream_t);\
+    template void launchPhaseRampGeneral<R>(cplx<R>*, bitCapInt, const RampArgs&, hipStream_t);
 
 QA_INSTANTIATE(float)
 QA_INSTANTIATE(double)

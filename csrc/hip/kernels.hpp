@@ -160,6 +160,23 @@ template <typename R>
 void launchPhaseRamp(cplx<R>* sv, bitCapInt maxQPower, bitLenInt rampStart, bitLenInt rampBits,
     bitCapInt condPower, double scale, hipStream_t stream);
 
+// generalized diagonal ramp with relocated bits: frac(i) =
+// ((i >> rampStart) & inPlaceRelMask) + sum_k (i & sPow[k] ? sWeight[k] : 0);
+// amp *= exp(i*scale*frac) where (condPow==0)||(i & condPow). Supports the
+// distributed pager's lazily-permuted qubit maps (<= 8 relocated bits).
+struct RampArgs {
+    bitLenInt rampStart;
+    bitCapInt inPlaceRelMask;
+    int nScattered;
+    bitCapInt sPow[8];
+    uint64_t sWeight[8];
+    bitCapInt condPow;
+    double scale;
+};
+
+template <typename R>
+void launchPhaseRampGeneral(cplx<R>* sv, bitCapInt maxQPower, const RampArgs& a, hipStream_t stream);
+
 // contiguous per-chunk |amp|^2 sums (inverse-CDF sampling support):
 // sums[c] = sum over [c*chunkLen, (c+1)*chunkLen)
 template <typename R>

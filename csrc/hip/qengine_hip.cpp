@@ -454,6 +454,29 @@ void QEngineHIP<R>::PhaseRamp(R scale, bitLenInt rampStart, bitLenInt rampBits, 
     launchPhaseRamp<R>(dState, maxQPower, rampStart, rampBits, condPower, (double)scale, stream);
 }
 
+template <typename R>
+void QEngineHIP<R>::PhaseRampGeneral(R scale, bitLenInt rampStart, bitCapInt inPlaceRelMask,
+    const std::vector<bitCapInt>& sPows, const std::vector<uint64_t>& sWeights, bitCapInt condPower)
+{
+    if (sPows.size() > 8u) {
+        QEngine<R>::PhaseRampGeneral(scale, rampStart, inPlaceRelMask, sPows, sWeights, condPower);
+        return;
+    }
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    RampArgs a{};
+    a.rampStart = rampStart;
+    a.inPlaceRelMask = inPlaceRelMask;
+    a.nScattered = (int)sPows.size();
+    for (size_t k = 0; k < sPows.size(); ++k) {
+        a.sPow[k] = sPows[k];
+        a.sWeight[k] = sWeights[k];
+    }
+    a.condPow = condPower;
+    a.scale = (double)scale;
+    HipProfScope prof("phase_ramp", stream);
+    launchPhaseRampGeneral<R>(dState, maxQPower, a, stream);
+}
+
 template <typename R> void QEngineHIP<R>::QFT(bitLenInt start, bitLenInt length, bool)
 {
     // H on the top column, then ONE fused ramp per column (the i controlled

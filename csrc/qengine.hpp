@@ -185,6 +185,35 @@ public:
         }
     }
 
+    // generalized ramp with relocated bits: frac(i) =
+    // ((i >> rampStart) & inPlaceRelMask) + sum_k bit(i, sPows[k]) * sWeights[k];
+    // amp *= exp(i*scale*frac) on the condPower-set half (everywhere if 0)
+    virtual void PhaseRampGeneral(R scale, bitLenInt rampStart, bitCapInt inPlaceRelMask,
+        const std::vector<bitCapInt>& sPows, const std::vector<uint64_t>& sWeights,
+        bitCapInt condPower)
+    {
+        // default lowering: product of per-bit (controlled) phases
+        bitCapInt m = inPlaceRelMask;
+        while (m) {
+            const bitLenInt rel = log2Ocl(m & (~m + 1u));
+            const cplx<R> f = polar<R>((R)1, scale * (R)pow2(rel));
+            if (condPower) {
+                this->MCPhase({ log2Ocl(condPower) }, cplx<R>(1, 0), f, rampStart + rel);
+            } else {
+                this->Phase(cplx<R>(1, 0), f, rampStart + rel);
+            }
+            m &= m - 1u;
+        }
+        for (size_t k = 0; k < sPows.size(); ++k) {
+            const cplx<R> f = polar<R>((R)1, scale * (R)sWeights[k]);
+            if (condPower) {
+                this->MCPhase({ log2Ocl(condPower) }, cplx<R>(1, 0), f, log2Ocl(sPows[k]));
+            } else {
+                this->Phase(cplx<R>(1, 0), f, log2Ocl(sPows[k]));
+            }
+        }
+    }
+
     // ---- measurement --------------------------------------------------------
     bool ForceM(bitLenInt qubit, bool result, bool doForce = true, bool doApply = true) override;
 

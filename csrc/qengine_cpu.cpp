@@ -238,6 +238,31 @@ void QEngineCPU<R>::PhaseRamp(R scale, bitLenInt rampStart, bitLenInt rampBits, 
     }
 }
 
+template <typename R>
+void QEngineCPU<R>::PhaseRampGeneral(R scale, bitLenInt rampStart, bitCapInt inPlaceRelMask,
+    const std::vector<bitCapInt>& sPows, const std::vector<uint64_t>& sWeights, bitCapInt condPower)
+{
+    cplx<R>* sv = stateVec.data();
+    const size_t ns = sPows.size();
+    auto fracOf = [=, &sPows, &sWeights](bitCapInt i) {
+        uint64_t frac = (uint64_t)((i >> rampStart) & inPlaceRelMask);
+        for (size_t k = 0; k < ns; ++k) {
+            if (i & sPows[k]) frac += sWeights[k];
+        }
+        return frac;
+    };
+    if (condPower) {
+        this->par_for_skip(maxQPower >> 1u, condPower, [=](const bitCapInt& j, unsigned) {
+            const bitCapInt i = j | condPower;
+            sv[i] = polar<R>(1, scale * (R)fracOf(i)) * sv[i];
+        });
+    } else {
+        this->par_for(0, maxQPower, [=](const bitCapInt& i, unsigned) {
+            sv[i] = polar<R>(1, scale * (R)fracOf(i)) * sv[i];
+        });
+    }
+}
+
 template <typename R> void QEngineCPU<R>::QFT(bitLenInt start, bitLenInt length, bool)
 {
     if (!length) return;

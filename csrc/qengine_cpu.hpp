@@ -62,6 +62,9 @@ public:
     void QFT(bitLenInt start, bitLenInt length, bool trySeparate = false) override;
     void IQFT(bitLenInt start, bitLenInt length, bool trySeparate = false) override;
     void PhaseRamp(R scale, bitLenInt rampStart, bitLenInt rampBits, bitCapInt condPower) override;
+    void PhaseRampGeneral(R scale, bitLenInt rampStart, bitCapInt inPlaceRelMask,
+        const std::vector<bitCapInt>& sPows, const std::vector<uint64_t>& sWeights,
+        bitCapInt condPower) override;
 
     // ---- probability / measurement ----
     R Prob(bitLenInt q) override;

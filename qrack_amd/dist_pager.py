@@ -3,15 +3,20 @@
 MI355X-native replacement for the reference's QPager multi-device layer
 (/root/reference/src/qpager.cpp): one process per GPU,
 RCCL over xGMI via torch.distributed for the cross-page half-exchanges
-(`ShuffleBuffers`, opencl.cpp:254-264 in the reference staged through HOST
-memory — here it is a direct GPU-to-GPU sendrecv on dlpack views of HBM).
+(`ShuffleBuffers` — staged through HOST memory in the reference,
+opencl.cpp:254-264; here a direct GPU-to-GPU sendrecv on DLPack views of
+HBM).
 
-Page-index tricks are preserved: X/Y/phase gates on meta qubits move no
-data (page relabeling + per-page scalars, qpager.cpp:509-525); controlled
-gates split into intra / semi-meta / meta classes (qpager.cpp:1011-1058).
+Beyond the reference: a **lazy qubit map**. Logical qubits map to physical
+slots (local bits 0..qpp-1 + page-index "meta" bits). A general gate on a
+meta qubit costs ONE half-page exchange (the exchange itself realizes the
+swap meta<->local-top, recorded in the map) instead of the reference's
+shuffle/gate/shuffle sandwich — half the xGMI traffic of a multi-GPU QFT.
+Swap of ANY two logical qubits is a pure map update (zero traffic). Phase /
+X on meta qubits remain zero-traffic page tricks (qpager.cpp:509-525).
 
-All ranks hold the same `page_of_rank` table and the same decision RNG, so
-control flow is replicated deterministically; only amplitude data moves.
+All ranks hold the same map, page table and decision RNG, so control flow
+is replicated deterministically; only amplitude data moves.
 """
 
 import numpy as np
@@ -38,10 +43,10 @@ class DistQPager:
         self.q = qa.create_simulator(
             self.qpp, precision=precision, engine=engine, seed=seed, device_id=device_id
         )
-        # page_of_rank[r] = logical page index rank r holds (identity at reset)
         self.page_of_rank = list(range(self.world))
+        self.slot_of = list(range(qubits))  # logical qubit -> physical slot
+        self.logical_at = list(range(qubits))  # slot -> logical qubit
         self.rng = np.random.default_rng(seed)  # replicated decision RNG
-        self.torch_dtype = torch.complex64 if precision == "fp32" else torch.complex128
         self.page_len = 1 << self.qpp
 
     # ---- helpers ------------------------------------------------------------
@@ -56,39 +61,36 @@ class DistQPager:
     def _is_hip(self):
         return self.engine_kind == "hip"
 
-    def _sync_engine(self):
-        self.q.finish()
-
-    def _sync_torch(self):
-        if self._is_hip():
-            torch.cuda.synchronize(self.device_id)
+    def _swap_slots(self, s1, s2):
+        a, b = self.logical_at[s1], self.logical_at[s2]
+        self.logical_at[s1], self.logical_at[s2] = b, a
+        self.slot_of[a], self.slot_of[b] = s2, s1
 
     def _half_view(self, low_half):
-        """torch view (zero-copy on HIP) of one half of the local page."""
         off = 0 if low_half else self.page_len // 2
         cap = self.q.dlpack_view(off, self.page_len // 2)
         return torch.from_dlpack(cap)
 
     def _shuffle(self, partner_rank, i_am_low):
-        """Swap my (upper if low page else lower) half with the partner's
-        opposite half — the reference's cross-device ShuffleBuffers
-        (opencl.cpp:254-264, staged through HOST there), as an RCCL sendrecv
-        pair over xGMI on zero-copy HBM views here. Non-NCCL backends (gloo
-        CI without GPUs, or single-GPU validation) stage through host."""
-        self._sync_engine()
+        """One half-page exchange: low page's upper half <-> high page's
+        lower half. Afterwards the page-index bit of the pair holds what the
+        local top bit held (and vice versa) — callers record the slot swap."""
+        self.q.finish()
         nccl = dist.get_backend() == "nccl"
         if nccl or not self._is_hip():
             view = self._half_view(low_half=not i_am_low)
             tmp = torch.empty_like(view)
-            ops = [
-                dist.P2POp(dist.isend, view, partner_rank),
-                dist.P2POp(dist.irecv, tmp, partner_rank),
-            ]
-            reqs = dist.batch_isend_irecv(ops)
+            reqs = dist.batch_isend_irecv(
+                [
+                    dist.P2POp(dist.isend, view, partner_rank),
+                    dist.P2POp(dist.irecv, tmp, partner_rank),
+                ]
+            )
             for r in reqs:
                 r.wait()
             view.copy_(tmp)
-            self._sync_torch()
+            if self._is_hip():
+                torch.cuda.synchronize(self.device_id)
         else:
             off = self.page_len // 2 if i_am_low else 0
             buf = self.q.get_amplitude_page(off, self.page_len // 2)
@@ -104,35 +106,12 @@ class DistQPager:
                 r.wait()
             self.q.set_amplitude_page(tmp.numpy(), off)
 
-    def _split_controls(self, controls):
-        local = [c for c in controls if c < self.qpp]
-        meta = [c - self.qpp for c in controls if c >= self.qpp]
-        return local, meta
-
-    def _meta_controls_satisfied(self, meta, anti=False):
-        page = self.my_page
-        for b in meta:
-            bit = (page >> b) & 1
-            if anti and bit:
-                return False
-            if not anti and not bit:
-                return False
-        return True
-
-    def _pair_meta_controls_satisfied(self, meta, target_bit):
-        # controls other than the target bit are shared by both pages of a pair
-        page = self.my_page
-        for b in meta:
-            if b == target_bit:
-                continue
-            if not ((page >> b) & 1):
-                return False
-        return True
-
     # ---- state management ----------------------------------------------------
 
     def set_permutation(self, perm):
         self.page_of_rank = list(range(self.world))
+        self.slot_of = list(range(self.num_qubits))
+        self.logical_at = list(range(self.num_qubits))
         page = perm >> self.qpp
         if self.my_page == page:
             self.q.set_permutation(perm & (self.page_len - 1))
@@ -140,59 +119,178 @@ class DistQPager:
             self.q.zero_amplitudes()
 
     def finish(self):
-        self._sync_engine()
+        self.q.finish()
+
+    def _logical_index(self, phys):
+        out = 0
+        for q in range(self.num_qubits):
+            s = self.slot_of[q]
+            if (phys >> s) & 1:
+                out |= 1 << q
+        return out
 
     def get_state_vector(self):
-        """Gather the full state on every rank (test helper; small widths)."""
+        """Gather the full LOGICAL state on every rank (test helper)."""
         local = np.asarray(self.q.get_state_vector())
         out = [None] * self.world
         dist.all_gather_object(out, (self.my_page, local))
-        full = np.zeros(1 << self.num_qubits, dtype=local.dtype)
+        phys = np.zeros(1 << self.num_qubits, dtype=local.dtype)
         for page, arr in out:
-            full[page * self.page_len : (page + 1) * self.page_len] = arr
+            phys[page * self.page_len : (page + 1) * self.page_len] = arr
+        # un-permute physical -> logical indices
+        idx = np.arange(1 << self.num_qubits)
+        logical = np.zeros_like(idx)
+        for q in range(self.num_qubits):
+            logical |= ((idx >> self.slot_of[q]) & 1) << q
+        full = np.zeros_like(phys)
+        full[logical] = phys[idx]
         return full
 
-    # ---- gates ---------------------------------------------------------------
+    # ---- slot-level gate dispatch ---------------------------------------------
 
-    def mtrx(self, m, target):
-        self.mcmtrx([], m, target)
-
-    def phase(self, tl, br, target):
-        self.mcphase([], tl, br, target)
-
-    def invert(self, tr, bl, target):
-        self.mcmtrx([], [0, tr, bl, 0], target)
-
-    def mcmtrx(self, controls, m, target):
+    def _dispatch(self, m, target_slot, ctrl_slots, perm):
+        """Apply the 2x2 `m` at a physical slot with per-slot controls
+        (perm bit k = required value of ctrl_slots[k])."""
         m = [complex(x) for x in m]
         is_phase = m[1] == 0 and m[2] == 0
         is_invert = m[0] == 0 and m[3] == 0
-        local_c, meta_c = self._split_controls(controls)
-        if target < self.qpp:
-            # intra-page target; meta controls filter whole pages (semi-meta,
-            # zero communication)
-            if self._meta_controls_satisfied(meta_c):
-                if local_c:
-                    self.q.mcmtrx(local_c, m, target)
+        local_c, local_perm = [], 0
+        meta_on, meta_off = 0, 0
+        for k, c in enumerate(ctrl_slots):
+            want = (perm >> k) & 1
+            if c < self.qpp:
+                if want:
+                    local_perm |= 1 << len(local_c)
+                local_c.append(c)
+            else:
+                b = 1 << (c - self.qpp)
+                if want:
+                    meta_on |= b
                 else:
-                    self.q.mtrx(m, target)
+                    meta_off |= b
+
+        def participates(p):
+            return (p & meta_on) == meta_on and (p & meta_off) == 0
+
+        if target_slot < self.qpp:
+            if participates(self.my_page):
+                if local_c:
+                    self.q.ucmtrx(local_c, m, target_slot, local_perm)
+                else:
+                    self.q.mtrx(m, target_slot)
             return
-        tb = target - self.qpp
-        has_tb_control = tb in meta_c
+
+        tb = target_slot - self.qpp
+        tb_bit = 1 << tb
+        if (meta_on | meta_off) & tb_bit:
+            raise ValueError("gate controlled on its own target qubit")
+
         if is_phase:
-            if self._pair_meta_controls_satisfied(meta_c, tb):
-                self._meta_phase(local_c, m[0], m[3], tb, has_tb_control)
-        elif is_invert:
-            if has_tb_control:
-                raise ValueError("gate controlled on its own target qubit")
-            # NOTE: must run on EVERY rank — partial page relabeling mutates
-            # the replicated page_of_rank table
-            self._meta_invert(local_c, meta_c, m[1], m[2], tb)
+            if participates(self.my_page):
+                bit = (self.my_page >> tb) & 1
+                self._scale_local(local_c, local_perm, m[3] if bit else m[0])
+            return
+        if is_invert and not local_c and not (meta_on | meta_off):
+            # pure page relabel + per-page scalar: zero traffic
+            self.page_of_rank = [p ^ tb_bit for p in self.page_of_rank]
+            bit = (self.my_page >> tb) & 1
+            f = m[1] if bit == 0 else m[2]
+            if f != 1:
+                self.q.global_phase(f)
+            return
+        if is_invert and not local_c:
+            # partial relabel under meta controls (replicated on every rank)
+            def sat(p):
+                return (p & meta_on) == meta_on and (p & meta_off) == 0
+
+            self.page_of_rank = [p ^ tb_bit if sat(p) else p for p in self.page_of_rank]
+            if participates(self.my_page):
+                bit = (self.my_page >> tb) & 1
+                f = m[1] if bit == 0 else m[2]
+                if f != 1:
+                    self.q.global_phase(f)
+            return
+
+        if meta_on or meta_off:
+            # meta-controlled general gate: classic sandwich on participating
+            # pairs only (no relabel — layout must stay globally uniform)
+            self._meta_sandwich(m, tb, local_c, local_perm, participates)
+            return
+
+        # un-controlled-at-meta general gate: ONE exchange realizes the swap
+        # (meta tb <-> local top), recorded lazily in the qubit map
+        partner_rank = self._rank_of_page(self.my_page ^ tb_bit)
+        i_am_low = ((self.my_page >> tb) & 1) == 0
+        self._shuffle(partner_rank, i_am_low)
+        self._swap_slots(self.qpp - 1, target_slot)
+        # the gate target now lives at local top; re-dispatch (controls were
+        # all local and keep their slots — unless one sat at local top, which
+        # has just moved to the meta slot)
+        new_ctrls = [(self.qpp - 1 if c == target_slot else (target_slot if c == self.qpp - 1 else c)) for c in ctrl_slots]
+        self._dispatch(m, self.qpp - 1, new_ctrls, perm)
+
+    def _meta_sandwich(self, m, tb, local_c, local_perm, participates):
+        if not participates(self.my_page):
+            return
+        partner_rank = self._rank_of_page(self.my_page ^ (1 << tb))
+        i_am_low = ((self.my_page >> tb) & 1) == 0
+        top = self.qpp - 1
+        ctrls = list(local_c)
+        perm = local_perm
+        top_required = None
+        if top in ctrls:
+            k = ctrls.index(top)
+            top_required = (perm >> k) & 1
+            ctrls.pop(k)
+            perm = (perm & ((1 << k) - 1)) | ((perm >> (k + 1)) << k)
+        self._shuffle(partner_rank, i_am_low)
+        apply_here = True
+        if top_required is not None:
+            # original local-top bit is page-constant after the exchange:
+            # 0 on the low page, 1 on the high page
+            apply_here = (0 if i_am_low else 1) == top_required
+        if apply_here:
+            if ctrls:
+                self.q.ucmtrx(ctrls, m, top, perm)
+            else:
+                self.q.mtrx(m, top)
+        self._shuffle(partner_rank, i_am_low)
+
+    def _scale_local(self, local_c, local_perm, factor):
+        if factor == 1:
+            return
+        if not local_c:
+            self.q.global_phase(factor)
+            return
+        diag = [1, 0, 0, factor]
+        if len(local_c) == 1:
+            if local_perm & 1:
+                self.q.phase(1, factor, local_c[0])
+            else:
+                self.q.phase(factor, 1, local_c[0])
         else:
-            if has_tb_control:
-                raise ValueError("gate controlled on its own target qubit")
-            if self._pair_meta_controls_satisfied(meta_c, tb):
-                self._meta_mtrx(local_c, m, tb)
+            c, rest = local_c[-1], local_c[:-1]
+            want = (local_perm >> (len(local_c) - 1)) & 1
+            mm = [1, 0, 0, factor] if want else [factor, 0, 0, 1]
+            self.q.ucmtrx(rest, mm, c, local_perm & ((1 << (len(local_c) - 1)) - 1))
+
+    # ---- logical gate API ------------------------------------------------------
+
+    def mcmtrx(self, controls, m, target):
+        self._dispatch(m, self.slot_of[target], [self.slot_of[c] for c in controls],
+                       (1 << len(controls)) - 1)
+
+    def ucmtrx(self, controls, m, target, perm):
+        self._dispatch(m, self.slot_of[target], [self.slot_of[c] for c in controls], perm)
+
+    def mtrx(self, m, target):
+        self._dispatch(m, self.slot_of[target], [], 0)
+
+    def phase(self, tl, br, target):
+        self._dispatch([tl, 0, 0, br], self.slot_of[target], [], 0)
+
+    def invert(self, tr, bl, target):
+        self._dispatch([0, tr, bl, 0], self.slot_of[target], [], 0)
 
     def mcphase(self, controls, tl, br, target):
         self.mcmtrx(controls, [tl, 0, 0, br], target)
@@ -200,84 +298,16 @@ class DistQPager:
     def mcinvert(self, controls, tr, bl, target):
         self.mcmtrx(controls, [0, tr, bl, 0], target)
 
-    def _scale_local(self, local_c, factor):
-        """Multiply amplitudes with all local controls set by `factor`."""
-        if factor == 1:
-            return
-        if not local_c:
-            self.q.global_phase(factor)
-        elif len(local_c) == 1:
-            self.q.phase(1, factor, local_c[0])
-        else:
-            self.q.mcphase(local_c[:-1], 1, factor, local_c[-1])
-
-    def _meta_phase(self, local_c, tl, br, tb, has_tb_control):
-        bit = (self.my_page >> tb) & 1
-        if has_tb_control:
-            # control on the target bit itself: only bit=1 pages scale (by br)
-            if bit:
-                self._scale_local(local_c, br)
-            return
-        self._scale_local(local_c, br if bit else tl)
-
-    def _meta_invert(self, local_c, meta_c, tr, bl, tb):
-        participates = self._pair_meta_controls_satisfied(meta_c, tb)
-        if local_c:
-            # invert with local controls mixes controlled and uncontrolled
-            # amplitudes ACROSS pages: needs the exchange path
-            if participates:
-                self._meta_mtrx(local_c, [0, tr, bl, 0], tb)
-            return
-        # partial page relabel (+ per-page scalar) — zero data motion.
-        # Pages whose meta controls are satisfied swap labels with their
-        # tb-partner; the permutation is computed identically on every rank.
-        mask = 1 << tb
-        ctrl_bits = [b for b in meta_c if b != tb]
-
-        def sat(p):
-            return all((p >> b) & 1 for b in ctrl_bits)
-
-        self.page_of_rank = [p ^ mask if sat(p) else p for p in self.page_of_rank]
-        if participates:
-            new_bit = (self.my_page >> tb) & 1
-            factor = tr if new_bit == 0 else bl
-            if factor != 1:
-                self.q.global_phase(factor)
-
-    def _meta_mtrx(self, local_c, m, tb):
-        # ShuffleBuffers sandwich (reference: qpager.cpp:369-448): after the
-        # half-swap, the local top qubit (qpp-1) indexes the meta target bit
-        # on BOTH pages of the pair, so the same 2x2 applies per page; the
-        # second swap restores the layout.
-        partner_page = self.my_page ^ (1 << tb)
-        partner_rank = self._rank_of_page(partner_page)
-        i_am_low = ((self.my_page >> tb) & 1) == 0
-        top = self.qpp - 1
-        ctrls = list(local_c)
-        top_controlled = top in ctrls
-        if top_controlled:
-            # in the shuffled layout the ORIGINAL local bit (qpp-1) is
-            # page-constant: 0 on the low page, 1 on the high page
-            ctrls.remove(top)
-        self._shuffle(partner_rank, i_am_low)
-        if not (top_controlled and i_am_low):
-            if ctrls:
-                self.q.mcmtrx(ctrls, m, top)
-            else:
-                self.q.mtrx(m, top)
-        self._shuffle(partner_rank, i_am_low)
-
-    # ---- named gates ---------------------------------------------------------
-
+    # named gates
     def h(self, t):
         s = 1 / np.sqrt(2)
         self.mtrx([s, s, s, -s], t)
 
     def x(self, t):
-        self.mcmtrx([], [0, 1, 1, 0], t)
+        self.invert(1, 1, t)
 
     def y(self, t):
-        self.mcmtrx([], [0, -1j, 1j, 0], t)
+        self.invert(-1j, 1j, t)
 
     def z(self, t):
         self.phase(1, -1, t)
@@ -305,66 +335,81 @@ class DistQPager:
     def swap(self, a, b):
         if a == b:
             return
-        if a < self.qpp and b < self.qpp:
-            self.q.swap(a, b)
-            return
-        if a >= self.qpp and b >= self.qpp:
-            # meta-meta swap: pure page relabel
-            ba, bb = a - self.qpp, b - self.qpp
-            self.page_of_rank = [
-                self._swap_bits(p, ba, bb) for p in self.page_of_rank
-            ]
-            return
-        # local<->meta swap via 3 CNOTs (one comm round for the meta-target one)
-        self.cnot(a, b)
-        self.cnot(b, a)
-        self.cnot(a, b)
+        # logical swap is ALWAYS a pure map update
+        sa, sb = self.slot_of[a], self.slot_of[b]
+        self._swap_slots(sa, sb)
 
-    @staticmethod
-    def _swap_bits(v, i, j):
-        bi, bj = (v >> i) & 1, (v >> j) & 1
-        if bi != bj:
-            v ^= (1 << i) | (1 << j)
-        return v
+    # ---- QFT -------------------------------------------------------------------
+    # column-fused with scattered-bit ramps: the lazy map may scatter ramp
+    # bits across slots; contributions split into the engines' in-place mask,
+    # <=8 scattered local bits, and per-page meta scalars. The only
+    # communication is ONE exchange per meta-H column.
 
-    # ---- QFT ------------------------------------------------------------------
-    # column-fused: the controlled-phase ladder of column i collapses to ONE
-    # per-rank PhaseRamp kernel (+ a per-page scalar for meta bits); the only
-    # communication is the H exchange on meta columns (mirrors QPager::QFT in
-    # csrc/qpager.cpp)
-
-    def _qft_column_ramp(self, start, i, sign):
-        t = start + i
+    def _column_ramp(self, start, i, sign):
+        t_logical = start + i
         scale = sign * np.pi / (1 << i)
-        if t < self.qpp:
-            self.q.phase_ramp(scale, start, i, 1 << t)
-            return
-        tb = t - self.qpp
-        if not ((self.my_page >> tb) & 1):
-            return
-        intra_bits = (self.qpp - start) if start < self.qpp else 0
-        meta_start = 0 if start < self.qpp else start - self.qpp
-        if intra_bits:
-            self.q.phase_ramp(scale, start, intra_bits, 0)
-        nbits = tb - meta_start
-        meta_val = (self.my_page >> meta_start) & ((1 << nbits) - 1) if nbits > 0 else 0
-        if meta_val:
-            theta = scale * (meta_val << intra_bits)
-            self.q.global_phase(complex(np.exp(1j * theta)))
+        t_slot = self.slot_of[t_logical]
+        in_place = 0
+        scattered = []  # (pow, weight)
+        meta_weight = 0
+        for j in range(start, start + i):
+            s = self.slot_of[j]
+            w = 1 << (j - start)
+            if s < self.qpp:
+                if s == j and j < self.qpp:
+                    in_place |= 1 << (j - start)
+                else:
+                    scattered.append((1 << s, w))
+            else:
+                if (self.my_page >> (s - self.qpp)) & 1:
+                    meta_weight += w
+        # in_place uses rampStart = start only if start < qpp; else fold into scattered
+        ramp_start = start
+        if start >= self.qpp or in_place == 0:
+            ramp_start = 0
+            # fold any in_place bits into scattered (rare: start >= qpp)
+            m = in_place
+            j = 0
+            while m:
+                if m & 1:
+                    scattered.append((1 << (start + j), 1 << j))
+                m >>= 1
+                j += 1
+            in_place = 0
+        if t_slot < self.qpp:
+            cond = 1 << t_slot
+            self._ramp_apply(scale, ramp_start, in_place, scattered, meta_weight, cond)
+        else:
+            if (self.my_page >> (t_slot - self.qpp)) & 1:
+                self._ramp_apply(scale, ramp_start, in_place, scattered, meta_weight, 0)
+
+    def _ramp_apply(self, scale, ramp_start, in_place, scattered, meta_weight, cond):
+        pows = [p for p, _ in scattered]
+        weights = [w for _, w in scattered]
+        if in_place or pows:
+            self.q.phase_ramp_general(scale, ramp_start, in_place, pows, weights, cond)
+        if meta_weight:
+            theta = scale * meta_weight
+            f = complex(np.exp(1j * theta))
+            if cond:
+                # scalar on the cond-bit-set half
+                self.q.phase(1, f, int(np.log2(cond)))
+            else:
+                self.q.global_phase(f)
 
     def qft(self, start, length):
         for i in range(length - 1, -1, -1):
             self.h(start + i)
             if i:
-                self._qft_column_ramp(start, i, +1)
+                self._column_ramp(start, i, +1)
 
     def iqft(self, start, length):
         for i in range(length):
             if i:
-                self._qft_column_ramp(start, i, -1)
+                self._column_ramp(start, i, -1)
             self.h(start + i)
 
-    # ---- measurement ----------------------------------------------------------
+    # ---- measurement ------------------------------------------------------------
 
     def _local_norm(self):
         return float(self.q.norm_total())
@@ -375,28 +420,28 @@ class DistQPager:
         return float(t.item())
 
     def prob(self, q):
-        if q < self.qpp:
-            # engine prob is computed on the unnormalized page; it IS this
-            # page's contribution to the global probability
-            local = float(self.q.prob(q))
+        s = self.slot_of[q]
+        if s < self.qpp:
+            local = float(self.q.prob(s))
             return min(1.0, self._allreduce_scalar(local))
-        bit = (self.my_page >> (q - self.qpp)) & 1
+        bit = (self.my_page >> (s - self.qpp)) & 1
         local = self._local_norm() if bit else 0.0
         return min(1.0, self._allreduce_scalar(local))
 
     def force_m(self, q, result, do_force=True, do_apply=True):
+        s = self.slot_of[q]
         p1 = self.prob(q)
         if not do_force:
-            result = bool(self.rng.random() < p1)  # replicated RNG: same draw
+            result = bool(self.rng.random() < p1)  # replicated draw
         if do_apply:
-            prob = p1 if result else 1.0 - p1
-            if prob <= 0:
+            pr = p1 if result else 1.0 - p1
+            if pr <= 0:
                 raise RuntimeError("impossible measurement outcome")
-            nrm = 1.0 / np.sqrt(prob)
-            if q < self.qpp:
-                self.q.apply_m(1 << q, (1 << q) if result else 0, complex(nrm))
+            nrm = 1.0 / np.sqrt(pr)
+            if s < self.qpp:
+                self.q.apply_m(1 << s, (1 << s) if result else 0, complex(nrm))
             else:
-                bit = (self.my_page >> (q - self.qpp)) & 1
+                bit = (self.my_page >> (s - self.qpp)) & 1
                 if bool(bit) == bool(result):
                     self.q.global_phase(complex(nrm))
                 else:
@@ -407,32 +452,35 @@ class DistQPager:
         return self.force_m(q, False, do_force=False)
 
     def m_all(self):
-        norms = [0.0] * self.world
         t = torch.zeros(self.world, dtype=torch.float64)
         t[self.rank] = self._local_norm()
         dist.all_reduce(t)
         norms = t.tolist()
         total = sum(norms)
-        r = self.rng.random() * total  # replicated draw
-        owner = 0
+        r = self.rng.random() * total
+        owner = self.world - 1
         for rank in range(self.world):
-            if r <= norms[rank] or rank == self.world - 1:
+            if r <= norms[rank]:
                 owner = rank
                 break
             r -= norms[rank]
         res_t = torch.zeros(1, dtype=torch.int64)
         if self.rank == owner:
-            res = self.q.multi_shot_measure_mask(
-                [1 << i for i in range(self.qpp)], 1
-            )
+            res = self.q.multi_shot_measure_mask([1 << i for i in range(self.qpp)], 1)
             local_idx = next(iter(res.keys()))
             res_t[0] = (self.page_of_rank[owner] << self.qpp) | local_idx
         dist.broadcast(res_t, src=owner)
-        result = int(res_t.item())
-        self.set_permutation(result)
-        return result
+        phys = int(res_t.item())
+        logical = self._logical_index(phys)
+        self.set_permutation(logical)
+        return logical
 
     def multi_shot_measure_mask(self, q_powers, shots):
+        # translate caller bit order to physical slot powers
+        phys_powers = []
+        for p in q_powers:
+            q = int(p).bit_length() - 1
+            phys_powers.append(1 << self.slot_of[q])
         t = torch.zeros(self.world, dtype=torch.float64)
         t[self.rank] = self._local_norm()
         dist.all_reduce(t)
@@ -442,14 +490,13 @@ class DistQPager:
         my_count = int(counts[self.rank])
         local_results = {}
         if my_count > 0:
-            # sample local indices, then map through the global page index
             local_powers = [1 << i for i in range(self.qpp)]
             res = self.q.multi_shot_measure_mask(local_powers, my_count)
             page_high = self.my_page << self.qpp
             for local_idx, c in res.items():
                 g = page_high | local_idx
                 val = 0
-                for b, p in enumerate(q_powers):
+                for b, p in enumerate(phys_powers):
                     if g & p:
                         val |= 1 << b
                 local_results[val] = local_results.get(val, 0) + c
