@@ -386,7 +386,15 @@ template <typename R> static void bindQInterface(py::module_& m, const char* nam
                 });
             })
         .def("is_clifford", [](QI& q) { return q.isClifford(); })
-        .def("depolarizing_channel_weak_1qb", &QI::DepolarizingChannelWeak1Qb);
+        .def("depolarizing_channel_weak_1qb", &QI::DepolarizingChannelWeak1Qb)
+        .def("reduced_density_matrix", [](QI& q, bitLenInt qb) {
+            cplx<R> rho[4];
+            q.GetReducedDensityMatrix(qb, rho);
+            py::array_t<C> out({ 2, 2 });
+            auto* p = reinterpret_cast<cplx<R>*>(out.mutable_data());
+            for (int i = 0; i < 4; ++i) p[i] = rho[i];
+            return out;
+        });
 }
 
 template <typename R> static void bindExtras(py::module_& m, const char* suffix)

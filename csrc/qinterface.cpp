@@ -751,6 +751,30 @@ template <typename R> void QInterface<R>::DepolarizingChannelWeak1Qb(bitLenInt q
     }
 }
 
+template <typename R> void QInterface<R>::GetReducedDensityMatrix(bitLenInt q, cplx<R>* out)
+{
+    // dense fallback: rho_ab = sum_rest amp(rest, a) * conj(amp(rest, b))
+    if (qubitCount > 26u) throw QrackError("GetReducedDensityMatrix: dense fallback width cap");
+    std::vector<cplx<R>> sv(maxQPower);
+    GetQuantumState(sv.data());
+    const bitCapInt qPow = pow2(q);
+    double r00 = 0, r11 = 0, reC = 0, imC = 0;
+    for (bitCapInt i = 0; i < maxQPower; ++i) {
+        if (i & qPow) continue;
+        const cplx<R> a0 = sv[i];
+        const cplx<R> a1 = sv[i | qPow];
+        r00 += (double)norm(a0);
+        r11 += (double)norm(a1);
+        // <0|rho|1> = sum a0 * conj(a1)
+        reC += (double)(a0.re * a1.re + a0.im * a1.im);
+        imC += (double)(a0.im * a1.re - a0.re * a1.im);
+    }
+    out[0] = cplx<R>((R)r00, 0);
+    out[1] = cplx<R>((R)reC, (R)imC);
+    out[2] = cplx<R>((R)reC, (R)-imC);
+    out[3] = cplx<R>((R)r11, 0);
+}
+
 template class QInterface<float>;
 template class QInterface<double>;
 

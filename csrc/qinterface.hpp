@@ -376,6 +376,10 @@ public:
 
     // depolarizing noise channel (parity: qinterface.hpp:3104)
     virtual void DepolarizingChannelWeak1Qb(bitLenInt q, R lambda);
+
+    // single-qubit reduced density matrix (parity: qinterface.cpp:885
+    // GetReducedDensityMatrix); out = row-major 2x2
+    virtual void GetReducedDensityMatrix(bitLenInt q, cplx<R>* out);
 };
 
 } // namespace qrack_amd

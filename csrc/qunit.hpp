@@ -98,6 +98,10 @@ public:
     bool ForceMParity(bitCapInt mask, bool result, bool doForce = true) override;
     double ExpectationBitsFactorized(const std::vector<bitLenInt>& bits,
         const std::vector<bitCapInt>& perms, bitCapInt offset = 0) override;
+    void GetReducedDensityMatrix(bitLenInt q, cplx<R>* out) override
+    {
+        shards[q].unit->GetReducedDensityMatrix(shards[q].mapped, out);
+    }
 
     // ---- separability ----
     bool TrySeparate(bitLenInt q) override;

@@ -323,3 +323,46 @@ def test_native_extension_loaded():
 
     assert native.hip_device_count() > 0
     assert "_qrack" in native.__file__
+
+
+def test_mirror_circuit_24q_gpu():
+    """Deep mirror circuit at 24 qubits on the HIP engine: random layer +
+    inverse returns to the initial basis state (the reference's [mirror]
+    self-validation strategy at GPU scale)."""
+    n = 24
+    rng = np.random.default_rng(77)
+    q = make(n, seed=77)
+    init = int(rng.integers(1 << n))
+    for i in range(n):
+        if (init >> i) & 1:
+            q.x(i)
+    ops = []
+    for _ in range(60):
+        kind = rng.integers(4)
+        if kind == 0:
+            t = int(rng.integers(n))
+            th = float(rng.uniform(0, 2 * np.pi))
+            q.ry(th, t)
+            ops.append(("ry", th, t))
+        elif kind == 1:
+            a, b = rng.choice(n, 2, replace=False)
+            q.cnot(int(a), int(b))
+            ops.append(("cnot", int(a), int(b)))
+        elif kind == 2:
+            t = int(rng.integers(n))
+            q.t(t)
+            ops.append(("t", t))
+        else:
+            a, b = rng.choice(n, 2, replace=False)
+            q.swap(int(a), int(b))
+            ops.append(("swap", int(a), int(b)))
+    for op in reversed(ops):
+        if op[0] == "ry":
+            q.ry(-op[1], op[2])
+        elif op[0] == "cnot":
+            q.cnot(op[1], op[2])
+        elif op[0] == "t":
+            q.it(op[1])
+        else:
+            q.swap(op[1], op[2])
+    assert q.m_all() == init

@@ -122,3 +122,22 @@ def test_seeded_determinism():
             q.h(i)
         r1.append(q.m_all())
     assert r1[0] == r1[1]
+
+
+def test_reduced_density_matrix():
+    import numpy as np
+
+    q = make(2, seed=3)
+    q.h(0)
+    rho = np.asarray(q.reduced_density_matrix(0))
+    assert abs(rho[0, 0] - 0.5) < 1e-5
+    assert abs(rho[0, 1] - 0.5) < 1e-5  # coherent |+>
+    q.cnot(0, 1)  # now maximally mixed RDM
+    rho = np.asarray(q.reduced_density_matrix(0))
+    assert abs(rho[0, 0] - 0.5) < 1e-5
+    assert abs(rho[0, 1]) < 1e-5
+    # QUnit fast path
+    qu = qa.create_simulator(3, layers=["qunit", "cpu"], seed=4)
+    qu.h(1)
+    rho = np.asarray(qu.reduced_density_matrix(1))
+    assert abs(rho[0, 1] - 0.5) < 1e-5
