@@ -640,9 +640,40 @@ template <typename R> bool QUnit<R>::TrySeparate(bitLenInt q)
 
 template <typename R> bool QUnit<R>::TrySeparate(bitLenInt q1, bitLenInt q2)
 {
+    // single-qubit tomography first (also covers different-unit cases)
     const bool a = TrySeparate(q1);
     const bool b = TrySeparate(q2);
-    return a && b;
+    if (a && b) return true;
+    if (shards[q1].unit != shards[q2].unit) return a && b;
+    QInterfacePtr<R> unit = shards[q1].unit;
+    if (unit->GetQubitCount() == 2u) return true; // already the pair's own unit
+    if (unit->GetQubitCount() > 16u) return false;
+    // decompose-verify: extract the PAIR as a unit (catches entangled pairs
+    // embedded in a larger unit, e.g. Bell pairs — reference qunit.cpp pair
+    // tomography); verified on a clone before committing
+    EntangleOrdered({ q1, q2 });
+    unit = shards[q1].unit;
+    QInterfacePtr<R> probe = unit->Clone();
+    QInterfacePtr<R> dest = MakeUnit(2u, 0u);
+    try {
+        probe->Decompose(0, dest);
+    } catch (const QrackError&) {
+        return false;
+    }
+    probe->Compose(dest, 0);
+    const double diff = probe->SumSqrDiff(unit);
+    if (diff > (double)separabilityThreshold) return false;
+    // commit on the real unit
+    QInterfacePtr<R> pairUnit = MakeUnit(2u, 0u);
+    unit->Decompose(0, pairUnit);
+    for (bitLenInt q = 0; q < qubitCount; ++q) {
+        if (shards[q].unit == unit && shards[q].mapped >= 2u) shards[q].mapped -= 2u;
+    }
+    shards[q1].unit = pairUnit;
+    shards[q1].mapped = 0;
+    shards[q2].unit = pairUnit;
+    shards[q2].mapped = 1;
+    return true;
 }
 
 template <typename R> bool QUnit<R>::TrySeparate(const std::vector<bitLenInt>& qubits, R)

@@ -281,3 +281,25 @@ def test_multishot_no_entangle_blowup():
         assert sum(res.values()) == 50
     finally:
         del os.environ["QRACK_QUNIT_ACE_MAX_QB"]
+
+
+def test_try_separate_pair_embedded_bell():
+    """A Bell pair embedded in a 4-qubit unit separates as a PAIR."""
+    q = make(4, ["qunit", "cpu"], seed=17)
+    # entangle everything into one unit: q0-q1 Bell, q2-q3 Bell, then a
+    # couple of cross gates that cancel
+    q.h(0)
+    q.cnot(0, 1)
+    q.h(2)
+    q.cnot(2, 3)
+    q.cz(1, 2)
+    q.cz(1, 2)  # uncompute: (0,1) and (2,3) pairs are separable again
+    assert not q.try_separate_1(0)  # each single qubit is still mixed
+    assert q.try_separate_2(0, 1)
+    # states intact
+    cp = make_cpu(4)
+    cp.h(0)
+    cp.cnot(0, 1)
+    cp.h(2)
+    cp.cnot(2, 3)
+    assert_states_close(q.get_state_vector(), cp.get_state_vector(), 1e-4)
