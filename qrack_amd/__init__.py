@@ -53,6 +53,11 @@ def create_simulator(
     layers, when given, is the explicit outer-to-inner layer list
     (e.g. ["pager", "hip"]).
     """
+    if layers == "optimal":
+        # the canonical arranged stack (reference: CreateArrangedLayersFull,
+        # qfactory.hpp:266-314): Schmidt decomposition over tableau-hybrid
+        # over the width-switched CPU/GPU engine
+        layers = ["qunit", "stabilizer_hybrid", "hybrid"]
     if layers is None:
         if engine == "auto":
             engine = "hip" if hip_device_count() > 0 else "cpu"

@@ -303,3 +303,17 @@ def test_try_separate_pair_embedded_bell():
     cp.h(2)
     cp.cnot(2, 3)
     assert_states_close(q.get_state_vector(), cp.get_state_vector(), 1e-4)
+
+
+def test_optimal_stack_alias():
+    q = qa.create_simulator(6, layers="optimal", seed=3)
+    q.h(0)
+    q.cnot(0, 1)
+    q.t(1)
+    q.cnot(1, 2)
+    cp = make_cpu(6, seed=3)
+    cp.h(0)
+    cp.cnot(0, 1)
+    cp.t(1)
+    cp.cnot(1, 2)
+    assert_states_close(q.get_state_vector(), cp.get_state_vector(), 1e-4)
