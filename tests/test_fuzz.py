@@ -1,0 +1,86 @@
+"""Differential fuzzing: hypothesis-generated random circuits must produce
+identical states on every layer stack (vs the dense CPU engine).
+
+Extends the reference's randomized-circuit testing (rngSeed-printed random
+benchmark circuits, test_main.cpp:143-154) with property-based shrinking.
+"""
+
+import numpy as np
+import pytest
+from hypothesis import given, settings, strategies as st
+
+import qrack_amd as qa
+from ref_sim import assert_states_close
+
+N_QUBITS = 4
+
+GATE_1Q = ["h", "x", "y", "z", "s", "t", "sqrt_x"]
+
+
+@st.composite
+def circuits(draw):
+    ops = []
+    for _ in range(draw(st.integers(3, 14))):
+        kind = draw(st.integers(0, 3))
+        if kind == 0:
+            ops.append(("g1", draw(st.sampled_from(GATE_1Q)), draw(st.integers(0, N_QUBITS - 1))))
+        elif kind == 1:
+            t = draw(st.integers(0, N_QUBITS - 1))
+            ops.append(("ry", draw(st.floats(0.1, 6.2)), t))
+        elif kind == 2:
+            a = draw(st.integers(0, N_QUBITS - 1))
+            b = draw(st.integers(0, N_QUBITS - 1).filter(lambda x: x != a))
+            ops.append(("cnot", a, b))
+        else:
+            a = draw(st.integers(0, N_QUBITS - 1))
+            b = draw(st.integers(0, N_QUBITS - 1).filter(lambda x: x != a))
+            ops.append(("cz", a, b))
+    return ops
+
+
+def apply(q, ops):
+    for op in ops:
+        if op[0] == "g1":
+            getattr(q, op[1])(op[2])
+        elif op[0] == "ry":
+            q.ry(op[1], op[2])
+        elif op[0] == "cnot":
+            q.cnot(op[1], op[2])
+        else:
+            q.cz(op[1], op[2])
+
+
+STACKS = [
+    ["sparse"],
+    ["bdt"],
+    ["stabilizer_hybrid", "cpu"],
+    ["qunit", "cpu"],
+    ["qunit", "stabilizer_hybrid", "cpu"],
+    ["pager", "cpu"],
+]
+
+
+@pytest.mark.parametrize("layers", STACKS, ids=["-".join(s) for s in STACKS])
+@settings(max_examples=25, deadline=None)
+@given(ops=circuits())
+def test_stack_matches_dense(layers, ops):
+    q = qa.create_simulator(N_QUBITS, layers=layers, seed=3, pages_per_device=2)
+    cp = qa.create_simulator(N_QUBITS, engine="cpu", seed=3)
+    apply(q, ops)
+    apply(cp, ops)
+    assert_states_close(q.get_state_vector(), cp.get_state_vector(), 1e-4)
+
+
+@settings(max_examples=15, deadline=None)
+@given(ops=circuits(), ops2=circuits())
+def test_compose_invariant(ops, ops2):
+    """state(A) (x) state(B) == Compose(A, B) state."""
+    a = qa.create_simulator(N_QUBITS, engine="cpu", seed=1)
+    b = qa.create_simulator(N_QUBITS, engine="cpu", seed=2)
+    apply(a, ops)
+    apply(b, ops2)
+    sva = np.asarray(a.get_state_vector()).astype(np.complex128)
+    svb = np.asarray(b.get_state_vector()).astype(np.complex128)
+    a.compose(b)
+    expect = np.kron(svb, sva)  # qubit 0 is the LSB: b occupies high bits
+    assert_states_close(a.get_state_vector(), expect, 1e-4)
