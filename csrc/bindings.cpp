@@ -165,6 +165,8 @@ template <typename R> static void bindQInterface(py::module_& m, const char* nam
             py::arg("length") = 0)
         .def("iqft", [](QI& q, bitLenInt s, bitLenInt l) { q.IQFT(s, l); }, py::arg("start") = 0,
             py::arg("length") = 0)
+        .def("qftr", [](QI& q, std::vector<bitLenInt> qs) { q.QFTR(qs); })
+        .def("iqftr", [](QI& q, std::vector<bitLenInt> qs) { q.IQFTR(qs); })
         // ---- structural ----
         .def("compose", [](QI& q, Ptr other) { return q.Compose(other); })
         .def("compose_at", [](QI& q, Ptr other, bitLenInt start) { return q.Compose(other, start); })
