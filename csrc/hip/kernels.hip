@@ -978,6 +978,99 @@ void launchPhaseRamp(cplx<R>* sv, bitCapInt maxQPower, bitLenInt rampStart, bitL
         maxI, condPower, rampStart, rampMask, (R)scale);
 }
 
+// fully fused QFT column: H on bit t AND the column's phase ramp in ONE
+// pass. POST (QFT): out1 *= e^{i*scale*frac} after H; PRE (IQFT): in1 *=
+// e^{i*scale*frac} before H. frac depends only on the pair base's low bits,
+// so both elements of a float4 share their own frac values.
+template <typename R, bool PRE>
+__global__ void k_qft_col(
+    cplx<R>* sv, bitCapInt maxI, bitCapInt tPow, bitLenInt rampStart, bitCapInt rampMask, R scale)
+{
+    const R s = (R)0.70710678118654752440;
+    const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
+    for (bitCapInt j = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; j < maxI; j += stride) {
+        const bitCapInt i = ((j & ~(tPow - 1u)) << 1u) | (j & (tPow - 1u));
+        const bitCapInt frac = (i >> rampStart) & rampMask;
+        R sn, cs;
+        devSinCos<R>(scale * (R)frac, &sn, &cs);
+        const cplx<R> f{ cs, sn };
+        cplx<R> x = sv[i];
+        cplx<R> y = sv[i | tPow];
+        if (PRE) y = f * y;
+        cplx<R> o0 = s * (x + y);
+        cplx<R> o1 = s * (x - y);
+        if (!PRE) o1 = f * o1;
+        sv[i] = o0;
+        sv[i | tPow] = o1;
+    }
+}
+
+// fp32 vectorized: two adjacent pairs per lane (requires tPow >= 2)
+template <bool PRE>
+__global__ void k_qft_col_v(
+    cplx<float>* sv, bitCapInt maxI, bitCapInt tPow, bitLenInt rampStart, bitCapInt rampMask,
+    float scale)
+{
+    float4* sv4 = reinterpret_cast<float4*>(sv);
+    const float s = 0.70710678f;
+    const bitCapInt half = maxI >> 1u;
+    const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
+    for (bitCapInt k = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; k < half; k += stride) {
+        const bitCapInt j = 2u * k;
+        const bitCapInt i = ((j & ~(tPow - 1u)) << 1u) | (j & (tPow - 1u));
+        const bitCapInt lo4 = i >> 1u;
+        const bitCapInt hi4 = (i | tPow) >> 1u;
+        float4 vlo = sv4[lo4];
+        float4 vhi = sv4[hi4];
+        float s0, c0, s1, c1;
+        __sincosf(scale * (float)((i >> rampStart) & rampMask), &s0, &c0);
+        __sincosf(scale * (float)(((i + 1u) >> rampStart) & rampMask), &s1, &c1);
+        const cplx<float> f0{ c0, s0 }, f1{ c1, s1 };
+        cplx<float> x0{ vlo.x, vlo.y }, x1{ vlo.z, vlo.w };
+        cplx<float> y0{ vhi.x, vhi.y }, y1{ vhi.z, vhi.w };
+        if (PRE) {
+            y0 = f0 * y0;
+            y1 = f1 * y1;
+        }
+        cplx<float> a0 = s * (x0 + y0), a1 = s * (x1 + y1);
+        cplx<float> b0 = s * (x0 - y0), b1 = s * (x1 - y1);
+        if (!PRE) {
+            b0 = f0 * b0;
+            b1 = f1 * b1;
+        }
+        sv4[lo4] = make_float4(a0.re, a0.im, a1.re, a1.im);
+        sv4[hi4] = make_float4(b0.re, b0.im, b1.re, b1.im);
+    }
+}
+
+template <typename R>
+void launchQftColumn(cplx<R>* sv, bitCapInt maxQPower, bitLenInt rampStart, bitLenInt col,
+    bitCapInt tPow, int sign, bool pre, hipStream_t stream)
+{
+    const bitCapInt maxI = maxQPower >> 1u;
+    const bitCapInt rampMask = (ONE_BCI << col) - 1u;
+    const R scale = (R)sign * (R)3.14159265358979323846 / (R)(ONE_BCI << col);
+    if constexpr (std::is_same_v<R, float>) {
+        if (tPow >= 2u && (maxI & 1u) == 0u) {
+            if (pre) {
+                hipLaunchKernelGGL((k_qft_col_v<true>), dim3(gridFor(maxI >> 1u)), dim3(QA_BLOCK),
+                    0, stream, sv, maxI, tPow, rampStart, rampMask, (float)scale);
+            } else {
+                hipLaunchKernelGGL((k_qft_col_v<false>), dim3(gridFor(maxI >> 1u)), dim3(QA_BLOCK),
+                    0, stream, sv, maxI, tPow, rampStart, rampMask, (float)scale);
+            }
+            return;
+        }
+    }
+    if (pre) {
+        hipLaunchKernelGGL((k_qft_col<R, true>), dim3(gridFor(maxI)), dim3(QA_BLOCK), 0, stream,
+            sv, maxI, tPow, rampStart, rampMask, scale);
+    } else {
+        hipLaunchKernelGGL((k_qft_col<R, false>), dim3(gridFor(maxI)), dim3(QA_BLOCK), 0, stream,
+            sv, maxI, tPow, rampStart, rampMask, scale);
+    }
+}
+
 template <typename R> __global__ void k_phase_ramp_gen(cplx<R>* sv, bitCapInt maxI, RampArgs a)
 {
     const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
@@ -1159,7 +1252,9 @@ void launchPartProbs(const cplx<R>* sv, bitCapInt maxQPower, bitLenInt start, bi
     template void launchPartProbs<R>(                                                               \
         const cplx<R>*, bitCapInt, bitLenInt, bitLenInt, double*, hipStream_t);                     \
     template void launchPhaseRamp<R>(cplx<R>*, bitCapInt, bitLenInt, bitLenInt, bitCapInt, double, hipStream_t);\
-    template void launchPhaseRampGeneral<R>(cplx<R>*, bitCapInt, const RampArgs&, hipStream_t);
+    template void launchPhaseRampGeneral<R>(cplx<R>*, bitCapInt, const RampArgs&, hipStream_t);    \
+    template void launchQftColumn<R>(                                                               \
+        cplx<R>*, bitCapInt, bitLenInt, bitLenInt, bitCapInt, int, bool, hipStream_t);
 
 QA_INSTANTIATE(float)
 QA_INSTANTIATE(double)

@@ -177,6 +177,12 @@ struct RampArgs {
 template <typename R>
 void launchPhaseRampGeneral(cplx<R>* sv, bitCapInt maxQPower, const RampArgs& a, hipStream_t stream);
 
+// fully fused QFT column (H + the column's phase ramp in one pass);
+// pre=false: QFT order (H then ramp), pre=true: IQFT order (ramp then H)
+template <typename R>
+void launchQftColumn(cplx<R>* sv, bitCapInt maxQPower, bitLenInt rampStart, bitLenInt col,
+    bitCapInt tPow, int sign, bool pre, hipStream_t stream);
+
 // contiguous per-chunk |amp|^2 sums (inverse-CDF sampling support):
 // sums[c] = sum over [c*chunkLen, (c+1)*chunkLen)
 template <typename R>

@@ -479,22 +479,32 @@ void QEngineHIP<R>::PhaseRampGeneral(R scale, bitLenInt rampStart, bitCapInt inP
 
 template <typename R> void QEngineHIP<R>::QFT(bitLenInt start, bitLenInt length, bool)
 {
-    // H on the top column, then ONE fused ramp per column (the i controlled
-    // phases CPhaseRootN(i-j+1, j, i) collapse to exp(i*pi*(x mod 2^i)/2^i)
-    // on the bit-i-set half)
+    // per column, ONE fully-fused kernel applies H and the column's entire
+    // controlled-phase ladder (exp(i*pi*(x mod 2^i)/2^i) on the H output's
+    // bit-set half) — a single state pass per column
     if (!length) return;
+    QA_HIP_CHECK(hipSetDevice(deviceId));
     for (bitLenInt i = length; i-- > 0;) {
-        this->H(start + i);
-        if (i) PhaseRamp(PI_R<R> / (R)pow2(i), start, i, pow2(start + i));
+        if (!i) {
+            this->H(start);
+            break;
+        }
+        HipProfScope prof("qft_column", stream);
+        launchQftColumn<R>(dState, maxQPower, start, i, pow2(start + i), +1, false, stream);
     }
 }
 
 template <typename R> void QEngineHIP<R>::IQFT(bitLenInt start, bitLenInt length, bool)
 {
     if (!length) return;
+    QA_HIP_CHECK(hipSetDevice(deviceId));
     for (bitLenInt i = 0; i < length; ++i) {
-        if (i) PhaseRamp(-PI_R<R> / (R)pow2(i), start, i, pow2(start + i));
-        this->H(start + i);
+        if (!i) {
+            this->H(start);
+            continue;
+        }
+        HipProfScope prof("qft_column", stream);
+        launchQftColumn<R>(dState, maxQPower, start, i, pow2(start + i), -1, true, stream);
     }
 }
 

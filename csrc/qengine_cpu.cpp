@@ -263,12 +263,38 @@ void QEngineCPU<R>::PhaseRampGeneral(R scale, bitLenInt rampStart, bitCapInt inP
     }
 }
 
+template <typename R> void QEngineCPU<R>::QftColumn(bitLenInt start, bitLenInt col, int sign, bool pre)
+{
+    // fully fused column: H on bit (start+col) + the column phase ladder in
+    // one pass over the state
+    const bitCapInt tPow = pow2(start + col);
+    const bitCapInt rampMask = pow2Mask(col);
+    const R scale = (R)sign * PI_R<R> / (R)pow2(col);
+    const R s = SQRT1_2_R<R>;
+    cplx<R>* sv = stateVec.data();
+    this->par_for_skip(maxQPower >> 1u, tPow, [=](const bitCapInt& i, unsigned) {
+        const R theta = scale * (R)((i >> start) & rampMask);
+        const cplx<R> f = polar<R>(1, theta);
+        cplx<R> x = sv[i];
+        cplx<R> y = sv[i | tPow];
+        if (pre) y = f * y;
+        cplx<R> o0 = s * (x + y);
+        cplx<R> o1 = s * (x - y);
+        if (!pre) o1 = f * o1;
+        sv[i] = o0;
+        sv[i | tPow] = o1;
+    });
+}
+
 template <typename R> void QEngineCPU<R>::QFT(bitLenInt start, bitLenInt length, bool)
 {
     if (!length) return;
     for (bitLenInt i = length; i-- > 0;) {
-        this->H(start + i);
-        if (i) PhaseRamp(PI_R<R> / (R)pow2(i), start, i, pow2(start + i));
+        if (!i) {
+            this->H(start);
+            break;
+        }
+        QftColumn(start, i, +1, false);
     }
 }
 
@@ -276,8 +302,11 @@ template <typename R> void QEngineCPU<R>::IQFT(bitLenInt start, bitLenInt length
 {
     if (!length) return;
     for (bitLenInt i = 0; i < length; ++i) {
-        if (i) PhaseRamp(-PI_R<R> / (R)pow2(i), start, i, pow2(start + i));
-        this->H(start + i);
+        if (!i) {
+            this->H(start);
+            continue;
+        }
+        QftColumn(start, i, -1, true);
     }
 }
 

@@ -138,6 +138,7 @@ protected:
     void ControlledPermutationOp(
         bitCapInt controlMask, const std::function<bitCapInt(bitCapInt)>& f);
     bitCapInt SampleOnce();
+    void QftColumn(bitLenInt start, bitLenInt col, int sign, bool pre);
 
 };
 

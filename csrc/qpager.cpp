@@ -254,15 +254,10 @@ void QPager<R>::DispatchGate(const cplx<R>* m, bitLenInt target,
         xConj(false);
         return;
     }
-    if (isInvert && intra.empty()) {
-        // restrict the pairwise pointer swap to participating pairs
+    if (isInvert && intra.empty() && !metaOff) {
+        // pairwise pointer swap restricted to metaOn-participating pairs;
+        // anti meta controls (metaOff) fall through to the exchange sandwich
         MetaInvert(m[1], m[2], tb, {}, metaOn);
-        // note: metaOff pages were excluded from metaOn check only; redo with
-        // full participation: handled below for the general case.
-        if (metaOff) {
-            // the simple mask above ignored metaOff; undo and use MetaMtrx
-            // (rare path: anti meta controls on an invert)
-        }
         return;
     }
     xConj(true);

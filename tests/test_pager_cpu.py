@@ -158,3 +158,23 @@ def test_swap_meta_meta_on_pager():
     q.swap(n - 1, n - 2)  # pointer relabel
     assert abs(q.prob(n - 2) - 1.0) < 1e-6
     assert abs(q.prob(n - 1)) < 1e-6
+
+
+def test_anti_meta_controlled_invert():
+    # anti-control on one meta qubit, invert target on another meta qubit
+    n = 6
+    q = make_paged(n, 4, seed=3)
+    cp = make_cpu(n, seed=3)
+    for i in range(n):
+        q.h(i)
+        cp.h(i)
+    q.macinvert([n - 1], 1, 1, n - 2)  # anti-CNOT(meta, meta)
+    cp.macinvert([n - 1], 1, 1, n - 2)
+    assert_states_close(q.get_state_vector(), cp.get_state_vector(), 1e-5)
+    q2 = make_paged(n, 4, seed=4)
+    cp2 = make_cpu(n, seed=4)
+    q2.x(0)
+    cp2.x(0)
+    q2.macinvert([0], 1, 1, n - 1)  # anti-control intra (off) on meta target: no-op
+    cp2.macinvert([0], 1, 1, n - 1)
+    assert_states_close(q2.get_state_vector(), cp2.get_state_vector(), 1e-6)
