@@ -51,7 +51,10 @@ def main():
         import torch
         import torch.distributed as tdist
 
-        backend = "nccl" if engine == "hip" else "gloo"
+        # mixed backend: CUDA tensors (half-page exchanges) over RCCL/xGMI,
+        # CPU tensors (norm scalars, decision broadcasts, object gathers)
+        # over gloo — a pure nccl group rejects CPU-tensor collectives
+        backend = "cpu:gloo,cuda:nccl" if engine == "hip" else "gloo"
         tdist.init_process_group(backend=backend)
         if engine == "hip":
             torch.cuda.set_device(local_rank)

@@ -76,7 +76,9 @@ class DistQPager:
         lower half. Afterwards the page-index bit of the pair holds what the
         local top bit held (and vice versa) — callers record the slot swap."""
         self.q.finish()
-        nccl = dist.get_backend() == "nccl"
+        cfg = (dist.get_backend_config() if hasattr(dist, "get_backend_config")
+               else str(dist.get_backend()))
+        nccl = "nccl" in cfg
         if nccl or not self._is_hip():
             view = self._half_view(low_half=not i_am_low)
             tmp = torch.empty_like(view)
