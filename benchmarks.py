@@ -38,18 +38,38 @@ def run_ghz(q, n, rng, depth):
         q.cnot(i, i + 1)
 
 
+_SQRT2 = 0.7071067811865476
+_M1Q = {
+    "h": [_SQRT2, _SQRT2, _SQRT2, -_SQRT2],
+    "x": [0, 1, 1, 0],
+    "t": [1, 0, 0, np.exp(1j * np.pi / 4)],
+    "s": [1, 0, 0, 1j],
+    "sqrt_x": [0.5 + 0.5j, 0.5 - 0.5j, 0.5 - 0.5j, 0.5 + 0.5j],
+}
+
+
+def _u_mtrx(th, ph, lm):
+    c, s = np.cos(th / 2), np.sin(th / 2)
+    return [c, -s * np.exp(1j * lm), s * np.exp(1j * ph), c * np.exp(1j * (ph + lm))]
+
+
+def _apply_1q_layer(q, targets, mats):
+    """One layer of independent 1q gates → a single fused Mtrx1qBatch pass on
+    state-vector engines (layered sims lower it per-gate)."""
+    if hasattr(q, "mtrx_1q_batch"):
+        q.mtrx_1q_batch(list(targets), [complex(x) for m in mats for x in m])
+    else:
+        for t, m in zip(targets, mats):
+            q.mtrx([complex(x) for x in m], t)
+
+
 def run_random_circuit(q, n, rng, depth):
     q.set_permutation(0)
     d = depth or n
+    names = ["h", "x", "t"]
     for _ in range(d):
-        for i in range(n):
-            g = rng.integers(3)
-            if g == 0:
-                q.h(i)
-            elif g == 1:
-                q.x(i)
-            else:
-                q.t(i)
+        mats = [_M1Q[names[rng.integers(3)]] for _ in range(n)]
+        _apply_1q_layer(q, range(n), mats)
         for i in range(0, n - 1, 2):
             q.cnot(i, i + 1)
 
@@ -59,8 +79,8 @@ def run_supremacy(q, n, rng, depth):
     d = depth or n
     sq = ["sqrt_x", "s", "h"]
     for layer in range(d):
-        for i in range(n):
-            getattr(q, sq[rng.integers(3)])(i)
+        mats = [_M1Q[sq[rng.integers(3)]] for _ in range(n)]
+        _apply_1q_layer(q, range(n), mats)
         start = layer % 2
         for i in range(start, n - 1, 2):
             th, ph = rng.uniform(0, 2 * np.pi, 2)
@@ -71,11 +91,19 @@ def run_qv(q, n, rng, depth):
     q.set_permutation(0)
     for _ in range(depth or n):
         perm = rng.permutation(n)
+        # the layer's 1q rotations act on disjoint pairs: fuse them all into
+        # batched passes, then apply the pair CNOTs (commuting reorder)
+        targets, mats = [], []
+        pairs = []
         for k in range(0, n - 1, 2):
             a, b = int(perm[k]), int(perm[k + 1])
+            pairs.append((a, b))
             for t in (a, b):
                 th, ph, lm = rng.uniform(0, 2 * np.pi, 3)
-                q.u(t, float(th), float(ph), float(lm))
+                targets.append(t)
+                mats.append(_u_mtrx(float(th), float(ph), float(lm)))
+        _apply_1q_layer(q, targets, mats)
+        for a, b in pairs:
             q.cnot(a, b)
 
 

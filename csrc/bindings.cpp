@@ -68,6 +68,14 @@ template <typename R> static void bindQInterface(py::module_& m, const char* nam
                     from_std<R>(m[3]) };
                 q.Mtrx(mm, t);
             })
+        .def("mtrx_1q_batch",
+            [](QI& q, std::vector<bitLenInt> targets, std::vector<C> ms) {
+                if (ms.size() != 4u * targets.size())
+                    throw QrackError("mtrx_1q_batch: need 4 entries per target");
+                std::vector<cplx<R>> mm(ms.size());
+                for (size_t i = 0; i < ms.size(); ++i) mm[i] = from_std<R>(ms[i]);
+                q.Mtrx1qBatch(targets, mm);
+            })
         .def("phase",
             [](QI& q, C tl, C br, bitLenInt t) { q.Phase(from_std<R>(tl), from_std<R>(br), t); })
         .def("invert",

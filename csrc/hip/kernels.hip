@@ -1071,6 +1071,135 @@ void launchQftColumn(cplx<R>* sv, bitCapInt maxQPower, bitLenInt rampStart, bitL
     }
 }
 
+// ---- batched independent single-qubit gates ---------------------------------
+// k distinct-target 2x2s in ONE pass: each lane owns a 2^k-amplitude orbit in
+// registers (k <= 5 fp32 / 4 fp64), so k memory-bound passes collapse to one.
+
+template <typename R, int K> __global__ void k_mtrx_batch(cplx<R>* sv, Batch1qArgs<R> a)
+{
+    const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
+    for (bitCapInt j = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; j < a.maxI; j += stride) {
+        const bitCapInt base = expandBits(j, a.tPow, K);
+        cplx<R> v[1 << K];
+#pragma unroll
+        for (int s = 0; s < (1 << K); ++s) {
+            bitCapInt off = 0;
+#pragma unroll
+            for (int g = 0; g < K; ++g) {
+                if (s & (1 << g)) off |= a.tPow[g];
+            }
+            v[s] = sv[base | off];
+        }
+#pragma unroll
+        for (int g = 0; g < K; ++g) {
+            const cplx<R> m0 = a.m[4 * g], m1 = a.m[4 * g + 1], m2 = a.m[4 * g + 2],
+                          m3 = a.m[4 * g + 3];
+#pragma unroll
+            for (int s = 0; s < (1 << K); ++s) {
+                if (s & (1 << g)) continue;
+                const int t = s | (1 << g);
+                const cplx<R> x = v[s], y = v[t];
+                v[s] = m0 * x + m1 * y;
+                v[t] = m2 * x + m3 * y;
+            }
+        }
+#pragma unroll
+        for (int s = 0; s < (1 << K); ++s) {
+            bitCapInt off = 0;
+#pragma unroll
+            for (int g = 0; g < K; ++g) {
+                if (s & (1 << g)) off |= a.tPow[g];
+            }
+            sv[base | off] = v[s];
+        }
+    }
+}
+
+// fp32 float4 variant (requires tPow[0] >= 2): each lane handles TWO adjacent
+// orbits via 16 B loads/stores — full-line HBM streams like k_apply2x2_1v.
+template <int K> __global__ void k_mtrx_batch_v(cplx<float>* sv, Batch1qArgs<float> a)
+{
+    float4* sv4 = reinterpret_cast<float4*>(sv);
+    const bitCapInt half = a.maxI >> 1u;
+    const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
+    for (bitCapInt t = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; t < half; t += stride) {
+        const bitCapInt base = expandBits(2u * t, a.tPow, K); // even: bit0 non-target
+        float4 v[1 << K];
+#pragma unroll
+        for (int s = 0; s < (1 << K); ++s) {
+            bitCapInt off = 0;
+#pragma unroll
+            for (int g = 0; g < K; ++g) {
+                if (s & (1 << g)) off |= a.tPow[g];
+            }
+            v[s] = sv4[(base | off) >> 1u];
+        }
+#pragma unroll
+        for (int g = 0; g < K; ++g) {
+            const cplx<float> m0 = a.m[4 * g], m1 = a.m[4 * g + 1], m2 = a.m[4 * g + 2],
+                              m3 = a.m[4 * g + 3];
+#pragma unroll
+            for (int s = 0; s < (1 << K); ++s) {
+                if (s & (1 << g)) continue;
+                const int tt = s | (1 << g);
+                const cplx<float> x0{ v[s].x, v[s].y }, x1{ v[s].z, v[s].w };
+                const cplx<float> y0{ v[tt].x, v[tt].y }, y1{ v[tt].z, v[tt].w };
+                const cplx<float> a0 = m0 * x0 + m1 * y0, a1 = m0 * x1 + m1 * y1;
+                const cplx<float> b0 = m2 * x0 + m3 * y0, b1 = m2 * x1 + m3 * y1;
+                v[s] = make_float4(a0.re, a0.im, a1.re, a1.im);
+                v[tt] = make_float4(b0.re, b0.im, b1.re, b1.im);
+            }
+        }
+#pragma unroll
+        for (int s = 0; s < (1 << K); ++s) {
+            bitCapInt off = 0;
+#pragma unroll
+            for (int g = 0; g < K; ++g) {
+                if (s & (1 << g)) off |= a.tPow[g];
+            }
+            sv4[(base | off) >> 1u] = v[s];
+        }
+    }
+}
+
+template <typename R>
+void launchMtrx1qBatch(cplx<R>* sv, const Batch1qArgs<R>& a, hipStream_t stream)
+{
+    const int grid = gridFor(a.maxI);
+    if constexpr (std::is_same_v<R, float>) {
+        if (a.tPow[0] >= 2u && a.maxI >= 2u && (a.maxI & 1u) == 0u && a.k <= 4) {
+            const int gridv = gridFor(a.maxI >> 1u);
+            switch (a.k) {
+            case 2:
+                hipLaunchKernelGGL((k_mtrx_batch_v<2>), dim3(gridv), dim3(QA_BLOCK), 0, stream, sv, a);
+                return;
+            case 3:
+                hipLaunchKernelGGL((k_mtrx_batch_v<3>), dim3(gridv), dim3(QA_BLOCK), 0, stream, sv, a);
+                return;
+            case 4:
+                hipLaunchKernelGGL((k_mtrx_batch_v<4>), dim3(gridv), dim3(QA_BLOCK), 0, stream, sv, a);
+                return;
+            }
+        }
+    }
+    switch (a.k) {
+    case 2:
+        hipLaunchKernelGGL((k_mtrx_batch<R, 2>), dim3(grid), dim3(QA_BLOCK), 0, stream, sv, a);
+        break;
+    case 3:
+        hipLaunchKernelGGL((k_mtrx_batch<R, 3>), dim3(grid), dim3(QA_BLOCK), 0, stream, sv, a);
+        break;
+    case 4:
+        hipLaunchKernelGGL((k_mtrx_batch<R, 4>), dim3(grid), dim3(QA_BLOCK), 0, stream, sv, a);
+        break;
+    case 5:
+        hipLaunchKernelGGL((k_mtrx_batch<R, 5>), dim3(grid), dim3(QA_BLOCK), 0, stream, sv, a);
+        break;
+    default:
+        throw QrackError("launchMtrx1qBatch: k out of range");
+    }
+}
+
 template <typename R> __global__ void k_phase_ramp_gen(cplx<R>* sv, bitCapInt maxI, RampArgs a)
 {
     const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
@@ -1254,7 +1383,8 @@ void launchPartProbs(const cplx<R>* sv, bitCapInt maxQPower, bitLenInt start, bi
     template void launchPhaseRamp<R>(cplx<R>*, bitCapInt, bitLenInt, bitLenInt, bitCapInt, double, hipStream_t);\
     template void launchPhaseRampGeneral<R>(cplx<R>*, bitCapInt, const RampArgs&, hipStream_t);    \
     template void launchQftColumn<R>(                                                               \
-        cplx<R>*, bitCapInt, bitLenInt, bitLenInt, bitCapInt, int, bool, hipStream_t);
+        cplx<R>*, bitCapInt, bitLenInt, bitLenInt, bitCapInt, int, bool, hipStream_t);              \
+    template void launchMtrx1qBatch<R>(cplx<R>*, const Batch1qArgs<R>&, hipStream_t);
 
 QA_INSTANTIATE(float)
 QA_INSTANTIATE(double)

@@ -177,6 +177,21 @@ struct RampArgs {
 template <typename R>
 void launchPhaseRampGeneral(cplx<R>* sv, bitCapInt maxQPower, const RampArgs& a, hipStream_t stream);
 
+// batched independent single-qubit gates: k distinct-target 2x2s applied in
+// ONE full-state pass (2^k-amplitude orbits in registers). The memory-bound
+// fusion win: k passes -> 1. fp32 supports k in [2,5], fp64 [2,4].
+constexpr int QA_MAX_BATCH_1Q = 5;
+
+template <typename R> struct Batch1qArgs {
+    cplx<R> m[4 * QA_MAX_BATCH_1Q];
+    bitCapInt tPow[QA_MAX_BATCH_1Q]; // sorted ascending, distinct
+    int k;
+    bitCapInt maxI; // orbit count = maxQPower >> k
+};
+
+template <typename R>
+void launchMtrx1qBatch(cplx<R>* sv, const Batch1qArgs<R>& a, hipStream_t stream);
+
 // fully fused QFT column (H + the column's phase ramp in one pass);
 // pre=false: QFT order (H then ramp), pre=true: IQFT order (ramp then H)
 template <typename R>

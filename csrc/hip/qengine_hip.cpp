@@ -1,6 +1,8 @@
 // qrack_amd — HIP engine host implementation (see qengine_hip.hpp).
 #include "qengine_hip.hpp"
 
+#include <set>
+
 #include "../qfactory.hpp"
 
 #include <chrono>
@@ -396,6 +398,49 @@ void QEngineHIP<R>::Apply2x2(bitCapInt offset1, bitCapInt offset2, const cplx<R>
     QA_HIP_CHECK(hipSetDevice(deviceId));
     HipProfScope prof("apply2x2", stream);
     launchApply2x2<R>(dState, a, stream);
+}
+
+// batched independent 1q gates: k gates in one full-state pass (k_mtrx_batch).
+// fp32 fuses up to 5 gates per pass, fp64 up to 4 (register budget).
+template <typename R>
+void QEngineHIP<R>::Mtrx1qBatch(
+    const std::vector<bitLenInt>& targets, const std::vector<cplx<R>>& mtrxs)
+{
+    if (mtrxs.size() != 4u * targets.size())
+        throw QrackError("Mtrx1qBatch: need 4 entries per target");
+    std::set<bitLenInt> uniq(targets.begin(), targets.end());
+    if (uniq.size() != targets.size() || targets.size() < 2u) {
+        QInterface<R>::Mtrx1qBatch(targets, mtrxs);
+        return;
+    }
+    for (bitLenInt t : targets) {
+        if (t >= qubitCount) throw QrackError("Mtrx1qBatch: target out of range");
+    }
+    const size_t maxK = (sizeof(R) == 4u) ? 5u : 4u;
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    size_t i = 0;
+    while (i < targets.size()) {
+        const size_t k = std::min(maxK, targets.size() - i);
+        if (k == 1u) {
+            this->Mtrx(&mtrxs[4u * i], targets[i]);
+            ++i;
+            continue;
+        }
+        std::vector<size_t> ord(k);
+        for (size_t j = 0; j < k; ++j) ord[j] = i + j;
+        std::sort(ord.begin(), ord.end(),
+            [&](size_t a, size_t b) { return targets[a] < targets[b]; });
+        Batch1qArgs<R> a{};
+        for (size_t g = 0; g < k; ++g) {
+            a.tPow[g] = pow2(targets[ord[g]]);
+            for (int e = 0; e < 4; ++e) a.m[4u * g + e] = mtrxs[4u * ord[g] + e];
+        }
+        a.k = (int)k;
+        a.maxI = maxQPower >> k;
+        HipProfScope prof("mtrx_1q_batch", stream);
+        launchMtrx1qBatch<R>(dState, a, stream);
+        i += k;
+    }
 }
 
 template <typename R> void QEngineHIP<R>::XMask(bitCapInt mask)

@@ -98,19 +98,46 @@ public:
         AppendControlled(x, b, { a }, 1u);
     }
 
-    // replay into a simulator (parity: qcircuit.hpp Run)
+    // replay into a simulator (parity: qcircuit.hpp Run). Runs of uncontrolled
+    // single-qubit gates are fused: same-target products combine into one 2x2
+    // (the reference's Combine) and distinct-target runs dispatch through
+    // Mtrx1qBatch — ONE full-state pass per <=5 gates on state-vector engines.
     void Run(QInterfacePtr<R> qsim) const
     {
+        std::vector<bitLenInt> bt;
+        std::vector<cplx<R>> bm;
+        auto flush = [&]() {
+            if (bt.empty()) return;
+            if (bt.size() == 1u) {
+                qsim->Mtrx(bm.data(), bt[0]);
+            } else {
+                qsim->Mtrx1qBatch(bt, bm);
+            }
+            bt.clear();
+            bm.clear();
+        };
         for (const auto& g : gates) {
             if (g.controls.empty()) {
-                qsim->Mtrx(g.payloads.at(0).data(), g.target);
+                const auto& p = g.payloads.at(0);
+                size_t idx = 0;
+                while (idx < bt.size() && bt[idx] != g.target) ++idx;
+                if (idx < bt.size()) {
+                    cplx<R> out[4];
+                    mul2x2(p.data(), &bm[4u * idx], out);
+                    std::copy(out, out + 4, &bm[4u * idx]);
+                } else {
+                    bt.push_back(g.target);
+                    bm.insert(bm.end(), p.begin(), p.end());
+                }
                 continue;
             }
+            flush();
             std::vector<bitLenInt> ctrls(g.controls.begin(), g.controls.end());
             for (auto& kv : g.payloads) {
                 qsim->UCMtrx(ctrls, kv.second.data(), g.target, kv.first);
             }
         }
+        flush();
     }
 
     QCircuitPtr<R> Inverse() const

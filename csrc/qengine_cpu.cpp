@@ -3,6 +3,8 @@
 // utility}.cpp (gate apply, ALU permutation ops, reductions, sampling).
 #include "qengine_cpu.hpp"
 
+#include <set>
+
 #include <algorithm>
 #include <cstring>
 
@@ -110,6 +112,79 @@ template <typename R> void QEngineCPU<R>::ShuffleBuffers(QEnginePtr<R> engine)
 }
 
 // ---- gate primitives -------------------------------------------------------
+
+// batched independent 1q gates: one pass over 2^k-amplitude orbits held in
+// registers (the HIP engine's k_mtrx_batch, CPU flavor). Falls back to the
+// sequential default on duplicate targets.
+template <typename R>
+void QEngineCPU<R>::Mtrx1qBatch(
+    const std::vector<bitLenInt>& targets, const std::vector<cplx<R>>& mtrxs)
+{
+    if (mtrxs.size() != 4u * targets.size())
+        throw QrackError("Mtrx1qBatch: need 4 entries per target");
+    std::set<bitLenInt> uniq(targets.begin(), targets.end());
+    if (uniq.size() != targets.size() || targets.size() < 2u) {
+        QInterface<R>::Mtrx1qBatch(targets, mtrxs);
+        return;
+    }
+    for (bitLenInt t : targets) {
+        if (t >= qubitCount) throw QrackError("Mtrx1qBatch: target out of range");
+    }
+    constexpr size_t MAXK = 6;
+    size_t i = 0;
+    while (i < targets.size()) {
+        size_t k = std::min(MAXK, targets.size() - i);
+        if (k == 1u) {
+            this->Mtrx(&mtrxs[4u * i], targets[i]);
+            ++i;
+            continue;
+        }
+        std::vector<size_t> ord(k);
+        for (size_t j = 0; j < k; ++j) ord[j] = i + j;
+        std::sort(ord.begin(), ord.end(),
+            [&](size_t a, size_t b) { return targets[a] < targets[b]; });
+        std::vector<bitCapInt> pows(k);
+        std::vector<cplx<R>> m(4u * k);
+        for (size_t g = 0; g < k; ++g) {
+            pows[g] = pow2(targets[ord[g]]);
+            for (int e = 0; e < 4; ++e) m[4u * g + e] = mtrxs[4u * ord[g] + e];
+        }
+        cplx<R>* sv = stateVec.data();
+        const cplx<R>* mp = m.data();
+        const bitCapInt* pp = pows.data();
+        const int kk = (int)k;
+        this->par_for_mask(maxQPower >> k, pows, [sv, mp, pp, kk](const bitCapInt& base, unsigned) {
+            cplx<R> v[1u << MAXK];
+            const int nOrb = 1 << kk;
+            for (int s = 0; s < nOrb; ++s) {
+                bitCapInt off = 0;
+                for (int g = 0; g < kk; ++g) {
+                    if (s & (1 << g)) off |= pp[g];
+                }
+                v[s] = sv[base | off];
+            }
+            for (int g = 0; g < kk; ++g) {
+                const cplx<R> m0 = mp[4 * g], m1 = mp[4 * g + 1], m2 = mp[4 * g + 2],
+                              m3 = mp[4 * g + 3];
+                for (int s = 0; s < nOrb; ++s) {
+                    if (s & (1 << g)) continue;
+                    const int t = s | (1 << g);
+                    const cplx<R> x = v[s], y = v[t];
+                    v[s] = m0 * x + m1 * y;
+                    v[t] = m2 * x + m3 * y;
+                }
+            }
+            for (int s = 0; s < nOrb; ++s) {
+                bitCapInt off = 0;
+                for (int g = 0; g < kk; ++g) {
+                    if (s & (1 << g)) off |= pp[g];
+                }
+                sv[base | off] = v[s];
+            }
+        });
+        i += k;
+    }
+}
 
 template <typename R>
 void QEngineCPU<R>::Apply2x2(bitCapInt offset1, bitCapInt offset2, const cplx<R>* mtrx,

@@ -59,6 +59,7 @@ public:
     void UniformlyControlledSingleBit(
         const std::vector<bitLenInt>& controls, bitLenInt target, const cplx<R>* mtrxs) override;
     void ROL(bitLenInt shift, bitLenInt start, bitLenInt length) override;
+    void Mtrx1qBatch(const std::vector<bitLenInt>& targets, const std::vector<cplx<R>>& mtrxs) override;
     void QFT(bitLenInt start, bitLenInt length, bool trySeparate = false) override;
     void IQFT(bitLenInt start, bitLenInt length, bool trySeparate = false) override;
     void PhaseRamp(R scale, bitLenInt rampStart, bitLenInt rampBits, bitCapInt condPower) override;

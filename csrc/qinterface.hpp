@@ -78,6 +78,20 @@ public:
 
     // ---- primitive gate API (wrapper-interceptable virtuals) ---------------
     virtual void Mtrx(const cplx<R>* mtrx, bitLenInt target) = 0;
+
+    // batched independent single-qubit gates: mtrxs is k row-major 2x2s on k
+    // DISTINCT targets (all gates commute, so order is immaterial). Engines
+    // override with a single fused full-state pass; the default lowering
+    // applies them one by one. (MI355X-native addition — the reference fuses
+    // only same-target gates, qcircuit.hpp Combine.)
+    virtual void Mtrx1qBatch(const std::vector<bitLenInt>& targets, const std::vector<cplx<R>>& mtrxs)
+    {
+        if (mtrxs.size() != 4u * targets.size())
+            throw QrackError("Mtrx1qBatch: need 4 entries per target");
+        for (size_t i = 0; i < targets.size(); ++i) {
+            Mtrx(&mtrxs[4u * i], targets[i]);
+        }
+    }
     virtual void Phase(cplx<R> topLeft, cplx<R> bottomRight, bitLenInt target);
     virtual void Invert(cplx<R> topRight, cplx<R> bottomLeft, bitLenInt target);
     virtual void MCMtrx(const std::vector<bitLenInt>& controls, const cplx<R>* mtrx, bitLenInt target);

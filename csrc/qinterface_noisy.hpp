@@ -37,6 +37,13 @@ public:
     void SetNoiseParameter(R np) { noiseParam = np; }
     R GetNoiseParameter() const { return noiseParam; }
 
+    // per-gate noise injection must see every gate: undo the wrapper's fused
+    // forwarding and lower the batch through Mtrx one gate at a time
+    void Mtrx1qBatch(const std::vector<bitLenInt>& targets, const std::vector<cplx<R>>& mtrxs) override
+    {
+        QInterface<R>::Mtrx1qBatch(targets, mtrxs);
+    }
+
     void Mtrx(const cplx<R>* m, bitLenInt t) override
     {
         inner->Mtrx(m, t);

@@ -33,6 +33,10 @@ public:
 
     // ---- gates ----
     void Mtrx(const cplx<R>* m, bitLenInt t) override { inner->Mtrx(m, t); }
+    void Mtrx1qBatch(const std::vector<bitLenInt>& targets, const std::vector<cplx<R>>& mtrxs) override
+    {
+        inner->Mtrx1qBatch(targets, mtrxs);
+    }
     void Phase(cplx<R> tl, cplx<R> br, bitLenInt t) override { inner->Phase(tl, br, t); }
     void Invert(cplx<R> tr, cplx<R> bl, bitLenInt t) override { inner->Invert(tr, bl, t); }
     void MCMtrx(const std::vector<bitLenInt>& c, const cplx<R>* m, bitLenInt t) override

@@ -366,3 +366,27 @@ def test_mirror_circuit_24q_gpu():
         else:
             q.swap(op[1], op[2])
     assert q.m_all() == init
+
+
+@pytest.mark.parametrize("precision", ["fp32", "fp64"])
+def test_mtrx_1q_batch_gpu(precision):
+    # fused k-gate pass vs sequential application (vector path: no target 0;
+    # scalar path: target 0 included; chunking: k=7 > max batch)
+    n = 20
+    rng = np.random.default_rng(11)
+    for targets in ([3, 7, 12, 15, 18], [0, 1, 5, 9], list(range(7))):
+        ms = []
+        for _ in targets:
+            th, ph, lm = rng.uniform(0, 2 * np.pi, 3)
+            c, s = np.cos(th / 2), np.sin(th / 2)
+            ms.append([c, -s * np.exp(1j * lm), s * np.exp(1j * ph),
+                       c * np.exp(1j * (ph + lm))])
+        qb = qa.create_simulator(n, engine="hip", precision=precision, seed=3)
+        qs = qa.create_simulator(n, engine="hip", precision=precision, seed=3)
+        for i in range(0, n, 3):
+            qb.h(i)
+            qs.h(i)
+        qb.mtrx_1q_batch(targets, [complex(x) for m in ms for x in m])
+        for t, m in zip(targets, ms):
+            qs.mtrx([complex(x) for x in m], t)
+        assert float(qb.sum_sqr_diff(qs)) < 1e-5
