@@ -416,7 +416,10 @@ void QEngineHIP<R>::Mtrx1qBatch(
     for (bitLenInt t : targets) {
         if (t >= qubitCount) throw QrackError("Mtrx1qBatch: target out of range");
     }
-    const size_t maxK = (sizeof(R) == 4u) ? 5u : 4u;
+    // chunk at 4 for BOTH precisions: fp32 k=5 would leave the float4
+    // vector path (k_mtrx_batch_v caps at k=4 for register budget) and the
+    // scalar kernel's 8 B accesses cost more than the extra pass saves
+    const size_t maxK = 4u;
     QA_HIP_CHECK(hipSetDevice(deviceId));
     size_t i = 0;
     while (i < targets.size()) {
