@@ -128,6 +128,8 @@ def main():
         help="comma layer list (e.g. qunit,stabilizer_hybrid,hip) or auto")
     p.add_argument("--precision", default="fp32", choices=["fp32", "fp64"])
     p.add_argument("--seed", type=int, default=0, help="0 = time-seeded (printed for repro)")
+    p.add_argument("--sdrp", type=float, default=0.0,
+                   help="Schmidt-decomposition rounding parameter (0 = exact)")
     p.add_argument("--out", default="")
     args = p.parse_args()
 
@@ -143,6 +145,8 @@ def main():
         else:
             q = qa.create_simulator(
                 n, precision=args.precision, layers=args.layers.split(","), seed=seed + n)
+        if args.sdrp > 0.0:
+            q.set_sdrp(args.sdrp)
         rng = np.random.default_rng(seed + n)
         times = []
         for s in range(args.samples):

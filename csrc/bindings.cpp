@@ -294,6 +294,8 @@ template <typename R> static void bindQInterface(py::module_& m, const char* nam
         .def("finish", &QI::Finish)
         .def("is_finished", &QI::isFinished)
         .def("get_unitary_fidelity", &QI::GetUnitaryFidelity)
+        .def("set_sdrp", &QI::SetSdrp)
+        .def("get_sdrp", &QI::GetSdrp)
         .def("reset_unitary_fidelity", &QI::ResetUnitaryFidelity)
         .def("set_device", &QI::SetDevice)
         .def("get_device", &QI::GetDevice)

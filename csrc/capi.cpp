@@ -567,6 +567,11 @@ void qrack_reset_unitary_fidelity(quid sid)
     guarded(sid, [&](SimSlot& s) { FOR_SIM(s, q.ResetUnitaryFidelity()); });
 }
 
+void qrack_set_sdrp(quid sid, double sdrp)
+{
+    guarded(sid, [&](SimSlot& s) { FOR_SIM(s, q.SetSdrp(sdrp)); });
+}
+
 quid qrack_compose(quid sid, quid other)
 {
     SimSlot* a = slot(sid);

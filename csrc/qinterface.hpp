@@ -384,6 +384,14 @@ public:
     virtual bool isFinished() { return true; }
     virtual void Dump() {}
     virtual double GetUnitaryFidelity() { return 1.0; }
+
+    // Schmidt-decomposition rounding parameter: 0 = exact; > 0 rounds
+    // near-separable qubits to product states after entangling gates, trading
+    // fidelity (tracked in GetUnitaryFidelity) for separability — the
+    // reference's approximate-simulation headline knob (pinvoke SetSdrp,
+    // qunit.cpp separabilityThreshold). No-op on exact engines.
+    virtual void SetSdrp(double sdrp) { (void)sdrp; }
+    virtual double GetSdrp() { return 0.0; }
     virtual void ResetUnitaryFidelity() {}
     virtual void SetDevice(int64_t deviceId) {}
     virtual int64_t GetDevice() const { return -1; }

@@ -26,6 +26,13 @@ QUnit<R>::QUnit(bitLenInt qBitCount, bitCapInt initState, RngPtr rgp, EngineFact
     if (const char* env = std::getenv("QRACK_QUNIT_SEPARABILITY_THRESHOLD")) {
         separabilityThreshold = (R)std::atof(env);
     }
+    if (const char* env = std::getenv("QRACK_QUNIT_SDRP")) {
+        const double v = std::atof(env);
+        if (v > 0.0) {
+            sdrp = v;
+            separabilityThreshold = (R)v;
+        }
+    }
     if (const char* env = std::getenv("QRACK_QUNIT_ACE_MAX_QB")) {
         aceMaxQubits = (bitLenInt)std::atoi(env);
     }
@@ -255,6 +262,8 @@ void QUnit<R>::MCMtrx(const std::vector<bitLenInt>& controls, const cplx<R>* m, 
         std::vector<bitLenInt> mc;
         for (bitLenInt c : live) mc.push_back(shards[c].mapped);
         unit->MCMtrx(mc, m, shards[t].mapped);
+        MaybeSeparate(t);
+        for (bitLenInt c : live) MaybeSeparate(c);
     } catch (const std::bad_alloc&) {
         bool applies = false;
         ElideControls(live, false, applies);
@@ -285,6 +294,8 @@ void QUnit<R>::MACMtrx(const std::vector<bitLenInt>& controls, const cplx<R>* m,
         std::vector<bitLenInt> mc;
         for (bitLenInt c : live) mc.push_back(shards[c].mapped);
         unit->MACMtrx(mc, m, shards[t].mapped);
+        MaybeSeparate(t);
+        for (bitLenInt c : live) MaybeSeparate(c);
     } catch (const std::bad_alloc&) {
         bool applies = false;
         ElideControls(live, true, applies);
@@ -331,6 +342,8 @@ void QUnit<R>::MCPhase(
         std::vector<bitLenInt> mc;
         for (bitLenInt c : live) mc.push_back(shards[c].mapped);
         unit->MCPhase(mc, tl, br, shards[t].mapped);
+        MaybeSeparate(t);
+        for (bitLenInt c : live) MaybeSeparate(c);
     } catch (const std::bad_alloc&) {
         bool applies = false;
         ElideControls(live, false, applies);
@@ -372,6 +385,8 @@ void QUnit<R>::UCMtrx(
     std::vector<bitLenInt> mc;
     for (bitLenInt c : live) mc.push_back(shards[c].mapped);
     unit->UCMtrx(mc, m, shards[t].mapped, livePerm);
+    MaybeSeparate(t);
+    for (bitLenInt c : live) MaybeSeparate(c);
 }
 
 template <typename R>
@@ -405,6 +420,8 @@ template <typename R> void QUnit<R>::ISwap(bitLenInt q1, bitLenInt q2)
     if (q1 == q2) return;
     QInterfacePtr<R> unit = EntangleAll({ q1, q2 });
     unit->ISwap(shards[q1].mapped, shards[q2].mapped);
+    MaybeSeparate(q1);
+    MaybeSeparate(q2);
 }
 
 template <typename R> void QUnit<R>::IISwap(bitLenInt q1, bitLenInt q2)
@@ -412,6 +429,8 @@ template <typename R> void QUnit<R>::IISwap(bitLenInt q1, bitLenInt q2)
     if (q1 == q2) return;
     QInterfacePtr<R> unit = EntangleAll({ q1, q2 });
     unit->IISwap(shards[q1].mapped, shards[q2].mapped);
+    MaybeSeparate(q1);
+    MaybeSeparate(q2);
 }
 
 template <typename R> void QUnit<R>::SqrtSwap(bitLenInt q1, bitLenInt q2)
@@ -419,6 +438,8 @@ template <typename R> void QUnit<R>::SqrtSwap(bitLenInt q1, bitLenInt q2)
     if (q1 == q2) return;
     QInterfacePtr<R> unit = EntangleAll({ q1, q2 });
     unit->SqrtSwap(shards[q1].mapped, shards[q2].mapped);
+    MaybeSeparate(q1);
+    MaybeSeparate(q2);
 }
 
 template <typename R> void QUnit<R>::ISqrtSwap(bitLenInt q1, bitLenInt q2)
@@ -426,6 +447,8 @@ template <typename R> void QUnit<R>::ISqrtSwap(bitLenInt q1, bitLenInt q2)
     if (q1 == q2) return;
     QInterfacePtr<R> unit = EntangleAll({ q1, q2 });
     unit->ISqrtSwap(shards[q1].mapped, shards[q2].mapped);
+    MaybeSeparate(q1);
+    MaybeSeparate(q2);
 }
 
 template <typename R> void QUnit<R>::FSim(R theta, R phi, bitLenInt q1, bitLenInt q2)
@@ -434,6 +457,8 @@ template <typename R> void QUnit<R>::FSim(R theta, R phi, bitLenInt q1, bitLenIn
     try {
         QInterfacePtr<R> unit = EntangleAll({ q1, q2 });
         unit->FSim(theta, phi, shards[q1].mapped, shards[q2].mapped);
+        MaybeSeparate(q1);
+        MaybeSeparate(q2);
     } catch (const std::bad_alloc&) {
         // ACE: classically collapse q1 (the fsim "control-like" qubit)
         bool applies = false;

@@ -31,7 +31,14 @@ protected:
     EngineFactoryFn<R> subFactory;
     double logFidelity = 0.0;
     R separabilityThreshold;
+    double sdrp = 0.0; // Schmidt-decomposition rounding parameter (0 = exact)
     bitLenInt aceMaxQubits = 0; // 0 = unlimited; else entangle cap (ACE)
+
+    // reactive separation after an entangling gate (active only under SDRP)
+    void MaybeSeparate(bitLenInt q)
+    {
+        if (sdrp > 0.0 && shards[q].unit->GetQubitCount() > 1u) TrySeparate(q);
+    }
 
     QInterfacePtr<R> MakeUnit(bitLenInt n, bitCapInt perm)
     {
@@ -58,6 +65,16 @@ public:
 
     double GetUnitaryFidelity() override { return std::exp(logFidelity); }
     void ResetUnitaryFidelity() override { logFidelity = 0.0; }
+
+    // SDRP: sets the rounding tolerance AND enables reactive separation
+    // attempts after every entangling gate (reference qunit.cpp TrySeparate
+    // with separabilityThreshold = sdrp). sdrp = 0 restores exact behavior.
+    void SetSdrp(double sdrp) override
+    {
+        this->sdrp = sdrp;
+        if (sdrp > 0.0) separabilityThreshold = (R)sdrp;
+    }
+    double GetSdrp() override { return sdrp; }
 
     // ---- state ----
     void SetPermutation(bitCapInt perm, cplx<R> phase = cplx<R>((R)1, (R)0)) override;

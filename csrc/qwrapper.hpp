@@ -37,6 +37,8 @@ public:
     {
         inner->Mtrx1qBatch(targets, mtrxs);
     }
+    void SetSdrp(double sdrp) override { inner->SetSdrp(sdrp); }
+    double GetSdrp() override { return inner->GetSdrp(); }
     void Phase(cplx<R> tl, cplx<R> br, bitLenInt t) override { inner->Phase(tl, br, t); }
     void Invert(cplx<R> tr, cplx<R> bl, bitLenInt t) override { inner->Invert(tr, bl, t); }
     void MCMtrx(const std::vector<bitLenInt>& c, const cplx<R>* m, bitLenInt t) override

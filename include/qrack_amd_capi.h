@@ -103,6 +103,8 @@ int qrack_try_separate_1qb(quid sid, uint64_t q);
 int qrack_try_separate_2qb(quid sid, uint64_t q1, uint64_t q2);
 double qrack_get_unitary_fidelity(quid sid);
 void qrack_reset_unitary_fidelity(quid sid);
+// Schmidt-decomposition rounding parameter (0 = exact; >0 = approximate)
+void qrack_set_sdrp(quid sid, double sdrp);
 
 /* compose / decompose */
 quid qrack_compose(quid sid, quid other);
