@@ -32,6 +32,19 @@ static void battery(QInterfacePtr<float> q)
     } catch (const QrackError&) {
         // some stacks (pure BDT) have no ALU: expected capability hole
     }
+    // batched 1q gates + SDRP approximate mode
+    const float s2 = 0.70710678f;
+    const cplx<float> hh[4] = { { s2, 0 }, { s2, 0 }, { s2, 0 }, { -s2, 0 } };
+    std::vector<cplx<float>> ms;
+    for (int g = 0; g < 3; ++g) ms.insert(ms.end(), hh, hh + 4);
+    q->Mtrx1qBatch({ 0, 2, 4 }, ms);
+    q->SetSdrp(0.3);
+    q->H(1);
+    q->CNOT(1, 3);
+    q->RY(0.05f, 3);
+    q->CZ(1, 3);
+    (void)q->GetUnitaryFidelity();
+    q->SetSdrp(0.0);
     (void)res;
 }
 
