@@ -296,6 +296,8 @@ template <typename R> static void bindQInterface(py::module_& m, const char* nam
         .def("get_unitary_fidelity", &QI::GetUnitaryFidelity)
         .def("set_sdrp", &QI::SetSdrp)
         .def("get_sdrp", &QI::GetSdrp)
+        .def("set_ncrp", &QI::SetNcrp)
+        .def("get_ncrp", &QI::GetNcrp)
         .def("reset_unitary_fidelity", &QI::ResetUnitaryFidelity)
         .def("set_device", &QI::SetDevice)
         .def("get_device", &QI::GetDevice)

@@ -572,6 +572,11 @@ void qrack_set_sdrp(quid sid, double sdrp)
     guarded(sid, [&](SimSlot& s) { FOR_SIM(s, q.SetSdrp(sdrp)); });
 }
 
+void qrack_set_ncrp(quid sid, double ncrp)
+{
+    guarded(sid, [&](SimSlot& s) { FOR_SIM(s, q.SetNcrp(ncrp)); });
+}
+
 quid qrack_compose(quid sid, quid other)
 {
     SimSlot* a = slot(sid);

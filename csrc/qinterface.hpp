@@ -392,6 +392,13 @@ public:
     // qunit.cpp separabilityThreshold). No-op on exact engines.
     virtual void SetSdrp(double sdrp) { (void)sdrp; }
     virtual double GetSdrp() { return 0.0; }
+
+    // near-Clifford rounding parameter: 0 = exact; > 0 snaps buffered
+    // non-Clifford phase gates to the nearest Clifford when |sin(delta/2)|
+    // <= ncrp, keeping wide circuits inside the stabilizer tableau
+    // (reference pinvoke SetNcrp). No-op on layers without a tableau.
+    virtual void SetNcrp(double ncrp) { (void)ncrp; }
+    virtual double GetNcrp() { return 0.0; }
     virtual void ResetUnitaryFidelity() {}
     virtual void SetDevice(int64_t deviceId) {}
     virtual int64_t GetDevice() const { return -1; }

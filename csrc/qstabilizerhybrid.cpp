@@ -1,6 +1,9 @@
 // qrack_amd — QStabilizerHybrid implementation (see header).
 #include "qstabilizerhybrid.hpp"
 
+#include <cmath>
+#include <cstdlib>
+
 #include "qengine_cpu.hpp"
 
 namespace qrack_amd {
@@ -14,6 +17,10 @@ QStabilizerHybrid<R>::QStabilizerHybrid(bitLenInt qBitCount, bitCapInt initState
     , engineFactory(factory)
     , shards(qBitCount)
 {
+    if (const char* env = std::getenv("QRACK_NCRP")) {
+        const double v = std::atof(env);
+        if (v > 0.0) ncrp = v;
+    }
     if (!engineFactory) {
         RngPtr rng = this->rand_generator;
         engineFactory = [rng](bitLenInt n, bitCapInt perm) -> QInterfacePtr<R> {
@@ -63,6 +70,35 @@ template <typename R> bool QStabilizerHybrid<R>::TryShardFlushClifford(bitLenInt
     }
 }
 
+// NCRP rounding: shard must be ~diag(m0, m3); snap arg(m3/m0) to the nearest
+// multiple of pi/2 when |sin(delta/2)| <= ncrp. The per-rounding fidelity is
+// exact for this state: |<psi| diag(1, e^{i delta}) |psi>|^2 =
+// 1 - 2 p1 (1 - p1) (1 - cos delta), with p1 from the tableau.
+template <typename R> bool QStabilizerHybrid<R>::TryShardRoundClifford(bitLenInt q)
+{
+    if (!shards[q]) return true;
+    if (!stabilizer || ncrp <= 0.0 || !ShardIsPhase(q)) return false;
+    const auto& m = *shards[q];
+    if (norm(m[0]) <= (R)1e-24) return false;
+    const cplx<R> ratio = m[3] * conj(m[0]);
+    const double theta = std::atan2((double)ratio.im, (double)ratio.re);
+    const double half_pi = 1.5707963267948966;
+    const double k = std::nearbyint(theta / half_pi);
+    const double delta = theta - k * half_pi;
+    if (std::abs(std::sin(delta / 2.0)) > ncrp) return false;
+    const double p1 = (double)stabilizer->Prob(q);
+    const double fid = 1.0 - 2.0 * p1 * (1.0 - p1) * (1.0 - std::cos(delta));
+    logFidelity += std::log(std::max(fid, 1e-300));
+    // snapped Clifford phase: diag(m0, m0 * i^k) up to the shard's own
+    // global-phase convention
+    const int ki = ((int)k % 4 + 4) % 4;
+    static const cplx<R> IPOW[4] = { { 1, 0 }, { 0, 1 }, { -1, 0 }, { 0, -1 } };
+    const cplx<R> snapped[4] = { m[0], cplx<R>(0, 0), cplx<R>(0, 0), m[0] * IPOW[ki] };
+    stabilizer->Mtrx(snapped, q);
+    shards[q].reset();
+    return true;
+}
+
 template <typename R> void QStabilizerHybrid<R>::FlushShard(bitLenInt q)
 {
     if (!shards[q]) return;
@@ -71,7 +107,7 @@ template <typename R> void QStabilizerHybrid<R>::FlushShard(bitLenInt q)
         shards[q].reset();
         return;
     }
-    if (!TryShardFlushClifford(q)) {
+    if (!TryShardFlushClifford(q) && !TryShardRoundClifford(q)) {
         SwitchToEngine();
         if (shards[q]) {
             engine->Mtrx(shards[q]->data(), q);
@@ -492,6 +528,8 @@ template <typename R> QInterfacePtr<R> QStabilizerHybrid<R>::Clone()
             clone->shards[q] = std::make_unique<std::array<cplx<R>, 4>>(*shards[q]);
         }
     }
+    clone->ncrp = ncrp;
+    clone->logFidelity = logFidelity;
     return clone;
 }
 

@@ -31,6 +31,14 @@ protected:
     // per-qubit buffered 2x2 (identity when absent)
     std::vector<std::unique_ptr<std::array<cplx<R>, 4>>> shards;
 
+    double ncrp = 0.0;        // near-Clifford rounding parameter (0 = exact)
+    double logFidelity = 0.0; // rounding-fidelity accumulator
+
+    // NCRP: if the shard is a phase gate within ncrp of a Clifford phase,
+    // snap it into the tableau and log the exact overlap loss. Returns true
+    // if the shard was disposed of (flushed exactly or rounded).
+    bool TryShardRoundClifford(bitLenInt q);
+
     bool InEngineMode() const { return (bool)engine; }
     void FlushShard(bitLenInt q);        // apply buffered 2x2 to the active backend
     void DumpShardIfPhase(bitLenInt q);  // drop diagonal shards (safe before Z ops)
@@ -44,6 +52,12 @@ public:
         EngineFactoryFn<R> factory = nullptr, bool doNorm = true, R normThresh = eps<R>::value);
 
     void SwitchToEngine();
+
+    void SetNcrp(double v) override { ncrp = v; }
+    double GetNcrp() override { return ncrp; }
+    double GetUnitaryFidelity() override { return std::exp(logFidelity); }
+    void ResetUnitaryFidelity() override { logFidelity = 0.0; }
+
     // try to fold every pending shard into the tableau; false if any shard
     // is non-Clifford (the state is then NOT serializable as a tableau)
     bool TryFlushAllShards()

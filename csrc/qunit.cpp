@@ -803,6 +803,9 @@ template <typename R> QInterfacePtr<R> QUnit<R>::Clone()
         clone->shards[q].mapped = shards[q].mapped;
     }
     clone->logFidelity = logFidelity;
+    clone->sdrp = sdrp;
+    clone->ncrp = ncrp;
+    clone->separabilityThreshold = separabilityThreshold;
     return clone;
 }
 

@@ -13,6 +13,8 @@
 
 #include "qstabilizerhybrid.hpp"
 
+#include <set>
+
 namespace qrack_amd {
 
 template <typename R> class QUnit;
@@ -32,6 +34,7 @@ protected:
     double logFidelity = 0.0;
     R separabilityThreshold;
     double sdrp = 0.0; // Schmidt-decomposition rounding parameter (0 = exact)
+    double ncrp = 0.0; // near-Clifford rounding parameter, forwarded to units
     bitLenInt aceMaxQubits = 0; // 0 = unlimited; else entangle cap (ACE)
 
     // reactive separation after an entangling gate (active only under SDRP)
@@ -42,7 +45,9 @@ protected:
 
     QInterfacePtr<R> MakeUnit(bitLenInt n, bitCapInt perm)
     {
-        return subFactory(n, perm);
+        QInterfacePtr<R> u = subFactory(n, perm);
+        if (ncrp > 0.0) u->SetNcrp(ncrp);
+        return u;
     }
 
     // merge all units containing `qs` into one; returns it
@@ -63,8 +68,25 @@ public:
     QUnit(bitLenInt qBitCount, bitCapInt initState = 0u, RngPtr rgp = nullptr,
         EngineFactoryFn<R> factory = nullptr, bool doNorm = true, R normThresh = eps<R>::value);
 
-    double GetUnitaryFidelity() override { return std::exp(logFidelity); }
-    void ResetUnitaryFidelity() override { logFidelity = 0.0; }
+    // own rounding fidelity times every distinct unit's (units may round
+    // internally, e.g. NCRP in a stabilizer-hybrid sub-layer)
+    double GetUnitaryFidelity() override
+    {
+        double f = std::exp(logFidelity);
+        std::set<QInterface<R>*> seen;
+        for (auto& s : shards) {
+            if (seen.insert(s.unit.get()).second) f *= s.unit->GetUnitaryFidelity();
+        }
+        return f;
+    }
+    void ResetUnitaryFidelity() override
+    {
+        logFidelity = 0.0;
+        std::set<QInterface<R>*> seen;
+        for (auto& s : shards) {
+            if (seen.insert(s.unit.get()).second) s.unit->ResetUnitaryFidelity();
+        }
+    }
 
     // SDRP: sets the rounding tolerance AND enables reactive separation
     // attempts after every entangling gate (reference qunit.cpp TrySeparate
@@ -75,6 +97,14 @@ public:
         if (sdrp > 0.0) separabilityThreshold = (R)sdrp;
     }
     double GetSdrp() override { return sdrp; }
+
+    // NCRP forwards to every unit (current and future)
+    void SetNcrp(double v) override
+    {
+        ncrp = v;
+        for (auto& s : shards) s.unit->SetNcrp(v);
+    }
+    double GetNcrp() override { return ncrp; }
 
     // ---- state ----
     void SetPermutation(bitCapInt perm, cplx<R> phase = cplx<R>((R)1, (R)0)) override;

@@ -105,6 +105,8 @@ double qrack_get_unitary_fidelity(quid sid);
 void qrack_reset_unitary_fidelity(quid sid);
 // Schmidt-decomposition rounding parameter (0 = exact; >0 = approximate)
 void qrack_set_sdrp(quid sid, double sdrp);
+// near-Clifford rounding parameter (0 = exact)
+void qrack_set_ncrp(quid sid, double ncrp);
 
 /* compose / decompose */
 quid qrack_compose(quid sid, quid other);

@@ -74,3 +74,79 @@ def test_sdrp_exact_when_fully_entangling():
     s = 1 / np.sqrt(2)
     assert_states_close(sv, np.array([s, 0, 0, s]), 1e-5)
     assert q.get_unitary_fidelity() == pytest.approx(1.0)
+
+
+# ---- NCRP (near-Clifford rounding parameter) --------------------------------
+
+
+def test_ncrp_zero_switches_to_engine():
+    # a T gate then CNOT forces the dense engine when ncrp == 0
+    q = qa.create_simulator(4, layers=["stabilizer_hybrid", "cpu"], seed=1)
+    q.h(0)
+    q.t(0)
+    q.cnot(0, 1)
+    assert not q.is_clifford()
+
+
+def test_ncrp_rounds_near_clifford_phases():
+    # rz(0.1) is within ncrp=0.1 of identity: the tableau absorbs it and the
+    # hybrid never materializes a state vector
+    q = qa.create_simulator(30, layers=["stabilizer_hybrid", "sparse"], seed=2)
+    q.set_ncrp(0.1)
+    assert q.get_ncrp() == pytest.approx(0.1)
+    for i in range(30):
+        q.h(i)
+    for i in range(29):
+        q.rz(0.08, i)
+        q.cnot(i, i + 1)
+    assert q.is_clifford()  # still a tableau at width 30
+    f = q.get_unitary_fidelity()
+    assert 0.5 < f < 1.0  # rounding happened and was logged
+    res = q.multi_shot_measure_mask([1, 2, 4], 50)
+    assert sum(res.values()) == 50
+
+
+def test_ncrp_keeps_true_t_gates_exact():
+    # T (pi/4) is NOT within ncrp=0.1 of Clifford: must stay exact
+    q = qa.create_simulator(3, layers=["stabilizer_hybrid", "cpu"], seed=3)
+    q.set_ncrp(0.1)
+    cp = qa.create_simulator(3, engine="cpu", seed=3)
+    for s in (q, cp):
+        s.h(0)
+        s.t(0)
+        s.cnot(0, 1)
+        s.h(1)
+    assert_states_close(q.get_state_vector(), cp.get_state_vector(), 1e-5)
+    assert q.get_unitary_fidelity() == pytest.approx(1.0)
+
+
+def test_ncrp_snap_accuracy():
+    # rounding rz(delta) ~ identity: state equals the unrounded Clifford part
+    q = qa.create_simulator(2, layers=["stabilizer_hybrid", "cpu"], seed=4)
+    q.set_ncrp(0.05)
+    q.h(0)
+    q.rz(0.06, 0)  # |sin(0.03)| = 0.03 <= 0.05: rounded away
+    q.cnot(0, 1)
+    cp = qa.create_simulator(2, engine="cpu", seed=4)
+    cp.h(0)
+    cp.cnot(0, 1)
+    assert_states_close(q.get_state_vector(), cp.get_state_vector(), 1e-5)
+    # fidelity log matches 1 - 2 p (1-p) (1 - cos 0.06) with p = 0.5
+    expect = 1.0 - 2.0 * 0.25 * (1.0 - np.cos(0.06))
+    assert q.get_unitary_fidelity() == pytest.approx(expect, abs=1e-6)
+
+
+def test_ncrp_through_qunit_stack():
+    # QUnit forwards ncrp to its (current and future) units and aggregates
+    # the units' fidelities
+    q = qa.create_simulator(8, layers=["qunit", "stabilizer_hybrid", "cpu"], seed=5)
+    q.set_ncrp(0.1)
+    for i in range(8):
+        q.h(i)
+        q.rz(0.05, i)
+    for i in range(7):
+        q.cnot(i, i + 1)
+    f = q.get_unitary_fidelity()
+    assert 0.8 < f < 1.0
+    q.reset_unitary_fidelity()
+    assert q.get_unitary_fidelity() == pytest.approx(1.0)
