@@ -130,6 +130,8 @@ def main():
     p.add_argument("--seed", type=int, default=0, help="0 = time-seeded (printed for repro)")
     p.add_argument("--sdrp", type=float, default=0.0,
                    help="Schmidt-decomposition rounding parameter (0 = exact)")
+    p.add_argument("--ncrp", type=float, default=0.0,
+                   help="near-Clifford rounding parameter (0 = exact)")
     p.add_argument("--out", default="")
     args = p.parse_args()
 
@@ -147,6 +149,8 @@ def main():
                 n, precision=args.precision, layers=args.layers.split(","), seed=seed + n)
         if args.sdrp > 0.0:
             q.set_sdrp(args.sdrp)
+        if args.ncrp > 0.0:
+            q.set_ncrp(args.ncrp)
         rng = np.random.default_rng(seed + n)
         times = []
         for s in range(args.samples):

@@ -61,3 +61,23 @@ def test_qunit_sparse_hundreds():
     # packed results cap at 64 bits; wide registers read per qubit
     bits = [q.m(i) for i in range(0, n, 10)]
     assert all(b in (0, 1) for b in bits)
+
+
+def test_ncrp_100q_near_clifford_tableau():
+    # 100-qubit near-Clifford circuit stays a tableau under NCRP rounding
+    import qrack_amd as qa
+    n = 100
+    q = qa.create_simulator(n, layers=["stabilizer_hybrid", "sparse"], seed=11)
+    q.set_ncrp(0.12)
+    for i in range(n):
+        q.h(i)
+    for layer in range(4):
+        for i in range(n):
+            q.rz(0.1, i)
+        for i in range(layer % 2, n - 1, 2):
+            q.cnot(i, i + 1)
+    assert q.is_clifford()
+    f = q.get_unitary_fidelity()
+    assert 0.0 < f < 1.0
+    res = q.multi_shot_measure_mask([1 << i for i in range(8)], 64)
+    assert sum(res.values()) == 64
