@@ -298,6 +298,8 @@ template <typename R> static void bindQInterface(py::module_& m, const char* nam
         .def("get_sdrp", &QI::GetSdrp)
         .def("set_ncrp", &QI::SetNcrp)
         .def("get_ncrp", &QI::GetNcrp)
+        .def("set_reactive_separate", &QI::SetReactiveSeparate)
+        .def("get_reactive_separate", &QI::GetReactiveSeparate)
         .def("reset_unitary_fidelity", &QI::ResetUnitaryFidelity)
         .def("set_device", &QI::SetDevice)
         .def("get_device", &QI::GetDevice)

@@ -399,6 +399,12 @@ public:
     // (reference pinvoke SetNcrp). No-op on layers without a tableau.
     virtual void SetNcrp(double ncrp) { (void)ncrp; }
     virtual double GetNcrp() { return 0.0; }
+
+    // reactive separation: attempt TrySeparate on the targets of every
+    // entangling gate even without SDRP (exact rounding threshold). The
+    // reference's QUnit default behavior toggle (SetReactiveSeparate).
+    virtual void SetReactiveSeparate(bool on) { (void)on; }
+    virtual bool GetReactiveSeparate() { return false; }
     virtual void ResetUnitaryFidelity() {}
     virtual void SetDevice(int64_t deviceId) {}
     virtual int64_t GetDevice() const { return -1; }

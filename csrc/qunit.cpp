@@ -26,6 +26,9 @@ QUnit<R>::QUnit(bitLenInt qBitCount, bitCapInt initState, RngPtr rgp, EngineFact
     if (const char* env = std::getenv("QRACK_QUNIT_SEPARABILITY_THRESHOLD")) {
         separabilityThreshold = (R)std::atof(env);
     }
+    if (const char* env = std::getenv("QRACK_QUNIT_REACTIVE_SEPARATE")) {
+        reactiveSeparate = std::atoi(env) != 0;
+    }
     if (const char* env = std::getenv("QRACK_QUNIT_SDRP")) {
         const double v = std::atof(env);
         if (v > 0.0) {

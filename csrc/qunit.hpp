@@ -35,12 +35,14 @@ protected:
     R separabilityThreshold;
     double sdrp = 0.0; // Schmidt-decomposition rounding parameter (0 = exact)
     double ncrp = 0.0; // near-Clifford rounding parameter, forwarded to units
+    bool reactiveSeparate = false; // TrySeparate after entangling gates
     bitLenInt aceMaxQubits = 0; // 0 = unlimited; else entangle cap (ACE)
 
     // reactive separation after an entangling gate (active only under SDRP)
     void MaybeSeparate(bitLenInt q)
     {
-        if (sdrp > 0.0 && shards[q].unit->GetQubitCount() > 1u) TrySeparate(q);
+        if ((sdrp > 0.0 || reactiveSeparate) && shards[q].unit->GetQubitCount() > 1u)
+            TrySeparate(q);
     }
 
     QInterfacePtr<R> MakeUnit(bitLenInt n, bitCapInt perm)
@@ -105,6 +107,9 @@ public:
         for (auto& s : shards) s.unit->SetNcrp(v);
     }
     double GetNcrp() override { return ncrp; }
+
+    void SetReactiveSeparate(bool on) override { reactiveSeparate = on; }
+    bool GetReactiveSeparate() override { return reactiveSeparate; }
 
     // ---- state ----
     void SetPermutation(bitCapInt perm, cplx<R> phase = cplx<R>((R)1, (R)0)) override;

@@ -41,6 +41,8 @@ public:
     double GetSdrp() override { return inner->GetSdrp(); }
     void SetNcrp(double ncrp) override { inner->SetNcrp(ncrp); }
     double GetNcrp() override { return inner->GetNcrp(); }
+    void SetReactiveSeparate(bool on) override { inner->SetReactiveSeparate(on); }
+    bool GetReactiveSeparate() override { return inner->GetReactiveSeparate(); }
     void Phase(cplx<R> tl, cplx<R> br, bitLenInt t) override { inner->Phase(tl, br, t); }
     void Invert(cplx<R> tr, cplx<R> bl, bitLenInt t) override { inner->Invert(tr, bl, t); }
     void MCMtrx(const std::vector<bitLenInt>& c, const cplx<R>* m, bitLenInt t) override

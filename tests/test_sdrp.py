@@ -150,3 +150,26 @@ def test_ncrp_through_qunit_stack():
     assert 0.8 < f < 1.0
     q.reset_unitary_fidelity()
     assert q.get_unitary_fidelity() == pytest.approx(1.0)
+
+
+def test_reactive_separate_mirror_circuit():
+    # with reactive separation, a mirror circuit returns every qubit to its
+    # own 1-qubit unit and the state is exact
+    q = qa.create_simulator(6, layers=["qunit", "cpu"], seed=6)
+    q.set_reactive_separate(True)
+    assert q.get_reactive_separate()
+    ops = []
+    rng = np.random.default_rng(8)
+    for _ in range(10):
+        a, b = rng.choice(6, 2, replace=False)
+        th = float(rng.uniform(0, 2 * np.pi))
+        ops.append((int(a), int(b), th))
+    for a, b, th in ops:
+        q.ry(th, a)
+        q.cnot(a, b)
+    for a, b, th in reversed(ops):
+        q.cnot(a, b)
+        q.ry(-th, a)
+    for i in range(6):
+        assert q.prob(i) < 1e-4
+    assert q.get_unitary_fidelity() == pytest.approx(1.0, abs=1e-6)
