@@ -1,12 +1,13 @@
 """Isolated timing of the batched-1q-gate kernel vs sequential gates.
 Usage: python tools/batch_probe.py [qubits]
 """
+import os
 import sys
 import time
 
 import numpy as np
 
-sys.path.insert(0, ".")
+sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)), ".."))
 import qrack_amd as qa
 
 n = int(sys.argv[1]) if len(sys.argv) > 1 else 28
@@ -41,5 +42,5 @@ t1 = time.perf_counter()
 seq_ms = 1000 * (t1 - t0) / REPS
 state_gb = (1 << n) * 8 / 1e9
 print(f"n={n} batch(4 gates, 1 pass)={batch_ms:.3f} ms "
-      f"({2 * state_gb / batch_ms * 1000:.2f} TB/s RMW) "
+      f"({2 * state_gb / batch_ms:.2f} TB/s RMW) "
       f"sequential(4 passes)={seq_ms:.3f} ms speedup={seq_ms / batch_ms:.2f}x")
