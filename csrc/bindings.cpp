@@ -217,6 +217,60 @@ template <typename R> static void bindQInterface(py::module_& m, const char* nam
                 for (int p : paulis) ps.push_back((Pauli)p);
                 return q.PauliExpectation(bits, ps);
             })
+        .def("expectation_pauli_all",
+            [](QI& q, std::vector<bitLenInt> bits, std::vector<int> paulis) {
+                std::vector<Pauli> ps;
+                for (int p : paulis) ps.push_back((Pauli)p);
+                return q.ExpectationPauliAll(bits, ps);
+            })
+        .def("variance_pauli_all",
+            [](QI& q, std::vector<bitLenInt> bits, std::vector<int> paulis) {
+                std::vector<Pauli> ps;
+                for (int p : paulis) ps.push_back((Pauli)p);
+                return q.VariancePauliAll(bits, ps);
+            })
+        .def("prob_bits_all",
+            [](QI& q, std::vector<bitLenInt> bits) {
+                py::array_t<double> out((py::ssize_t)pow2((bitLenInt)bits.size()));
+                q.ProbBitsAll(bits, out.mutable_data());
+                return out;
+            })
+        .def("prob_mask_all",
+            [](QI& q, bitCapInt mask) {
+                int k = 0;
+                for (bitCapInt m = mask; m; m &= m - 1u) ++k;
+                py::array_t<double> out((py::ssize_t)pow2((bitLenInt)k));
+                q.ProbMaskAll(mask, out.mutable_data());
+                return out;
+            })
+        .def("expectation_bits_factorized",
+            [](QI& q, std::vector<bitLenInt> bits, std::vector<bitCapInt> perms, bitCapInt offset) {
+                return q.ExpectationBitsFactorized(bits, perms, offset);
+            }, py::arg("bits"), py::arg("perms"), py::arg("offset") = 0)
+        .def("variance_bits_factorized",
+            [](QI& q, std::vector<bitLenInt> bits, std::vector<bitCapInt> perms, bitCapInt offset) {
+                return q.VarianceBitsFactorized(bits, perms, offset);
+            }, py::arg("bits"), py::arg("perms"), py::arg("offset") = 0)
+        .def("expectation_floats_factorized",
+            [](QI& q, std::vector<bitLenInt> bits, std::vector<double> w) {
+                return q.ExpectationFloatsFactorized(bits, w);
+            })
+        .def("variance_floats_factorized",
+            [](QI& q, std::vector<bitLenInt> bits, std::vector<double> w) {
+                return q.VarianceFloatsFactorized(bits, w);
+            })
+        .def("expectation_unitary_all",
+            [](QI& q, std::vector<bitLenInt> bits, std::vector<C> ops, std::vector<double> ev) {
+                std::vector<cplx<R>> mm(ops.size());
+                for (size_t i = 0; i < ops.size(); ++i) mm[i] = from_std<R>(ops[i]);
+                return q.ExpectationUnitaryAll(bits, mm, ev);
+            }, py::arg("bits"), py::arg("basis_ops"), py::arg("eigen_vals") = std::vector<double>())
+        .def("variance_unitary_all",
+            [](QI& q, std::vector<bitLenInt> bits, std::vector<C> ops, std::vector<double> ev) {
+                std::vector<cplx<R>> mm(ops.size());
+                for (size_t i = 0; i < ops.size(); ++i) mm[i] = from_std<R>(ops[i]);
+                return q.VarianceUnitaryAll(bits, mm, ev);
+            }, py::arg("bits"), py::arg("basis_ops"), py::arg("eigen_vals") = std::vector<double>())
         // ---- ALU ----
         .def("inc", &QI::INC)
         .def("dec", &QI::DEC)

@@ -449,6 +449,184 @@ double QInterface<R>::VarianceBitsAll(const std::vector<bitLenInt>& bits, bitCap
 }
 
 template <typename R>
+void QInterface<R>::ProbBitsAll(const std::vector<bitLenInt>& bits, double* probsOut)
+{
+    const size_t k = bits.size();
+    if (k > 24u) throw QrackError("ProbBitsAll: more than 24 bits");
+    const bitCapInt outLen = pow2((bitLenInt)k);
+    std::fill(probsOut, probsOut + outLen, 0.0);
+    if (qubitCount <= 26u) {
+        // one pass over the dense distribution
+        std::vector<R> probs(maxQPower);
+        GetProbs(probs.data());
+        for (bitCapInt i = 0; i < maxQPower; ++i) {
+            bitCapInt p = 0;
+            for (size_t b = 0; b < k; ++b) {
+                if ((i >> bits[b]) & 1u) p |= (ONE_BCI << b);
+            }
+            probsOut[p] += (double)probs[i];
+        }
+        return;
+    }
+    // wide states: 2^k masked-probability queries (layers answer these
+    // without materializing the distribution)
+    bitCapInt mask = 0;
+    for (bitLenInt b : bits) mask |= pow2(b);
+    for (bitCapInt p = 0; p < outLen; ++p) {
+        bitCapInt perm = 0;
+        for (size_t b = 0; b < k; ++b) {
+            if ((p >> b) & 1u) perm |= pow2(bits[b]);
+        }
+        probsOut[p] = (double)ProbMask(mask, perm);
+    }
+}
+
+template <typename R> void QInterface<R>::ProbMaskAll(bitCapInt mask, double* probsOut)
+{
+    std::vector<bitLenInt> bits;
+    for (bitLenInt b = 0; b < qubitCount; ++b) {
+        if ((mask >> b) & 1u) bits.push_back(b);
+    }
+    ProbBitsAll(bits, probsOut);
+}
+
+template <typename R>
+double QInterface<R>::ExpectationFloatsFactorized(
+    const std::vector<bitLenInt>& bits, const std::vector<double>& weights)
+{
+    if (weights.size() < 2u * bits.size())
+        throw QrackError("ExpectationFloatsFactorized: need (w0, w1) per bit");
+    double e = 0;
+    for (size_t b = 0; b < bits.size(); ++b) {
+        const double p1 = (double)Prob(bits[b]);
+        e += weights[2u * b] * (1.0 - p1) + weights[2u * b + 1u] * p1;
+    }
+    return e;
+}
+
+template <typename R>
+double QInterface<R>::VarianceFloatsFactorized(
+    const std::vector<bitLenInt>& bits, const std::vector<double>& weights)
+{
+    if (weights.size() < 2u * bits.size())
+        throw QrackError("VarianceFloatsFactorized: need (w0, w1) per bit");
+    // the variance of a sum needs the joint distribution (cross terms)
+    const bitCapInt outLen = pow2((bitLenInt)bits.size());
+    std::vector<double> joint(outLen);
+    ProbBitsAll(bits, joint.data());
+    double mean = 0, e2 = 0;
+    for (bitCapInt p = 0; p < outLen; ++p) {
+        double val = 0;
+        for (size_t b = 0; b < bits.size(); ++b) {
+            val += ((p >> b) & 1u) ? weights[2u * b + 1u] : weights[2u * b];
+        }
+        mean += val * joint[p];
+        e2 += val * val * joint[p];
+    }
+    return e2 - mean * mean;
+}
+
+template <typename R>
+double QInterface<R>::VarianceBitsFactorized(
+    const std::vector<bitLenInt>& bits, const std::vector<bitCapInt>& perms, bitCapInt offset)
+{
+    if (perms.size() < bits.size())
+        throw QrackError("VarianceBitsFactorized: need one perm per bit");
+    std::vector<double> w(2u * bits.size());
+    for (size_t b = 0; b < bits.size(); ++b) {
+        w[2u * b] = 0.0;
+        w[2u * b + 1u] = (double)perms[b];
+    }
+    // offset shifts the value uniformly: variance is unaffected by it
+    return VarianceFloatsFactorized(bits, w);
+}
+
+template <typename R>
+double QInterface<R>::ExpectationUnitaryAll(const std::vector<bitLenInt>& bits,
+    const std::vector<cplx<R>>& basisOps, const std::vector<double>& eigenVals)
+{
+    if (basisOps.size() != 4u * bits.size())
+        throw QrackError("ExpectationUnitaryAll: need a 2x2 per bit");
+    QInterfacePtr<R> c = Clone();
+    std::vector<double> w(2u * bits.size());
+    for (size_t b = 0; b < bits.size(); ++b) {
+        c->Mtrx(&basisOps[4u * b], bits[b]);
+        w[2u * b] = eigenVals.empty() ? 1.0 : eigenVals[2u * b];
+        w[2u * b + 1u] = eigenVals.empty() ? -1.0 : eigenVals[2u * b + 1u];
+    }
+    return c->ExpectationFloatsFactorized(bits, w);
+}
+
+template <typename R>
+double QInterface<R>::VarianceUnitaryAll(const std::vector<bitLenInt>& bits,
+    const std::vector<cplx<R>>& basisOps, const std::vector<double>& eigenVals)
+{
+    if (basisOps.size() != 4u * bits.size())
+        throw QrackError("VarianceUnitaryAll: need a 2x2 per bit");
+    QInterfacePtr<R> c = Clone();
+    std::vector<double> w(2u * bits.size());
+    for (size_t b = 0; b < bits.size(); ++b) {
+        c->Mtrx(&basisOps[4u * b], bits[b]);
+        w[2u * b] = eigenVals.empty() ? 1.0 : eigenVals[2u * b];
+        w[2u * b + 1u] = eigenVals.empty() ? -1.0 : eigenVals[2u * b + 1u];
+    }
+    return c->VarianceFloatsFactorized(bits, w);
+}
+
+// reference-semantics SUM of single-qubit Paulis (basis-rotate a clone, then
+// factorized +1/-1 expectation; qinterface.cpp:715-769)
+template <typename R>
+static QInterfacePtr<R> pauliBasisClone(QInterface<R>* self, std::vector<bitLenInt>& bits,
+    std::vector<Pauli>& paulis, std::vector<double>& eig)
+{
+    for (size_t i = bits.size(); i-- > 0u;) {
+        if (paulis[i] == PauliI) {
+            bits.erase(bits.begin() + i);
+            paulis.erase(paulis.begin() + i);
+        }
+    }
+    QInterfacePtr<R> c = self->Clone();
+    eig.clear();
+    for (size_t i = 0; i < bits.size(); ++i) {
+        eig.push_back(1.0);
+        eig.push_back(-1.0);
+        if (paulis[i] == PauliX) {
+            c->H(bits[i]);
+        } else if (paulis[i] == PauliY) {
+            c->IS(bits[i]);
+            c->H(bits[i]);
+        }
+    }
+    return c;
+}
+
+template <typename R>
+double QInterface<R>::ExpectationPauliAll(
+    const std::vector<bitLenInt>& bitsIn, const std::vector<Pauli>& paulisIn)
+{
+    if (bitsIn.size() != paulisIn.size()) throw QrackError("ExpectationPauliAll: size mismatch");
+    std::vector<bitLenInt> bits(bitsIn);
+    std::vector<Pauli> paulis(paulisIn);
+    std::vector<double> eig;
+    QInterfacePtr<R> c = pauliBasisClone<R>(this, bits, paulis, eig);
+    if (bits.empty()) return 1.0;
+    return c->ExpectationFloatsFactorized(bits, eig);
+}
+
+template <typename R>
+double QInterface<R>::VariancePauliAll(
+    const std::vector<bitLenInt>& bitsIn, const std::vector<Pauli>& paulisIn)
+{
+    if (bitsIn.size() != paulisIn.size()) throw QrackError("VariancePauliAll: size mismatch");
+    std::vector<bitLenInt> bits(bitsIn);
+    std::vector<Pauli> paulis(paulisIn);
+    std::vector<double> eig;
+    QInterfacePtr<R> c = pauliBasisClone<R>(this, bits, paulis, eig);
+    if (bits.empty()) return 0.0;
+    return c->VarianceFloatsFactorized(bits, eig);
+}
+
+template <typename R>
 double QInterface<R>::PauliExpectation(
     const std::vector<bitLenInt>& bits, const std::vector<Pauli>& paulis)
 {

@@ -310,6 +310,92 @@ public:
         const std::vector<bitLenInt>& bits, const std::vector<bitCapInt>& perms, bitCapInt offset = 0);
     virtual double VarianceBitsAll(const std::vector<bitLenInt>& bits, bitCapInt offset = 0);
     virtual double PauliExpectation(const std::vector<bitLenInt>& bits, const std::vector<Pauli>& paulis);
+    // NOTE: PauliExpectation above is the tensor-PRODUCT observable
+    // <P_0 x P_1 x ...> (an extension kept for product-parity use); the
+    // reference's ExpectationPauliAll / VariancePauliAll are the factorized
+    // SUM of single-qubit Paulis (qinterface.cpp:715-769) — implemented here
+    // with matching semantics.
+    virtual double ExpectationPauliAll(
+        const std::vector<bitLenInt>& bits, const std::vector<Pauli>& paulis);
+    virtual double VariancePauliAll(
+        const std::vector<bitLenInt>& bits, const std::vector<Pauli>& paulis);
+    // product observable squares to identity: Var = 1 - E^2
+    double PauliProductVariance(const std::vector<bitLenInt>& bits, const std::vector<Pauli>& paulis)
+    {
+        const double e = PauliExpectation(bits, paulis);
+        return 1.0 - e * e;
+    }
+
+    // joint probabilities of the listed bits: probsOut[perm] with perm bit i
+    // = value of bits[i] (2^bits.size() entries; parity: ProbBitsAll)
+    virtual void ProbBitsAll(const std::vector<bitLenInt>& bits, double* probsOut);
+    // same over the set bits of a mask, low to high (parity: ProbMaskAll)
+    virtual void ProbMaskAll(bitCapInt mask, double* probsOut);
+
+    // per-qubit weighted observables: weights holds (w0, w1) per bit;
+    // E = sum_b w0_b P(b=0) + w1_b P(b=1) (parity: ExpectationFloatsFactorized)
+    virtual double ExpectationFloatsFactorized(
+        const std::vector<bitLenInt>& bits, const std::vector<double>& weights);
+    // variance of the same sum — needs the JOINT distribution (cross terms)
+    virtual double VarianceFloatsFactorized(
+        const std::vector<bitLenInt>& bits, const std::vector<double>& weights);
+    virtual double VarianceBitsFactorized(const std::vector<bitLenInt>& bits,
+        const std::vector<bitCapInt>& perms, bitCapInt offset = 0);
+
+    // expectation/variance in arbitrary single-qubit bases: basisOps holds one
+    // row-major 2x2 per bit mapping the measurement basis to computational;
+    // eigenVals holds (e0, e1) per bit (default +1, -1).
+    // (parity: ExpectationUnitaryAll / VarianceUnitaryAll)
+    virtual double ExpectationUnitaryAll(const std::vector<bitLenInt>& bits,
+        const std::vector<cplx<R>>& basisOps, const std::vector<double>& eigenVals = {});
+    virtual double VarianceUnitaryAll(const std::vector<bitLenInt>& bits,
+        const std::vector<cplx<R>>& basisOps, const std::vector<double>& eigenVals = {});
+
+    // ---- Rdm variants (parity: *Rdm family) --------------------------------
+    // On exact layers these equal the plain forms; the reference's Rdm forms
+    // trade exactness for avoiding ancilla-phase flushes in approximate QUnit
+    // states (roundRz flag), which this build's layers do not require.
+    double ProbRdm(bitLenInt q) { return Prob(q); }
+    double ProbAllRdm(bool roundRz, bitCapInt perm) { (void)roundRz; return (double)ProbAll(perm); }
+    double ProbMaskRdm(bool roundRz, bitCapInt mask, bitCapInt perm)
+    {
+        (void)roundRz;
+        return (double)ProbMask(mask, perm);
+    }
+    double ExpectationBitsAllRdm(bool roundRz, const std::vector<bitLenInt>& bits, bitCapInt offset = 0)
+    {
+        (void)roundRz;
+        return ExpectationBitsAll(bits, offset);
+    }
+    double ExpectationBitsFactorizedRdm(bool roundRz, const std::vector<bitLenInt>& bits,
+        const std::vector<bitCapInt>& perms, bitCapInt offset = 0)
+    {
+        (void)roundRz;
+        return ExpectationBitsFactorized(bits, perms, offset);
+    }
+    double ExpectationFloatsFactorizedRdm(
+        bool roundRz, const std::vector<bitLenInt>& bits, const std::vector<double>& weights)
+    {
+        (void)roundRz;
+        return ExpectationFloatsFactorized(bits, weights);
+    }
+    double VarianceBitsAllRdm(bool roundRz, const std::vector<bitLenInt>& bits, bitCapInt offset = 0)
+    {
+        (void)roundRz;
+        return VarianceBitsAll(bits, offset);
+    }
+    double VarianceBitsFactorizedRdm(bool roundRz, const std::vector<bitLenInt>& bits,
+        const std::vector<bitCapInt>& perms, bitCapInt offset = 0)
+    {
+        (void)roundRz;
+        return VarianceBitsFactorized(bits, perms, offset);
+    }
+    double VarianceFloatsFactorizedRdm(
+        bool roundRz, const std::vector<bitLenInt>& bits, const std::vector<double>& weights)
+    {
+        (void)roundRz;
+        return VarianceFloatsFactorized(bits, weights);
+    }
 
     // parity rotation family (parity: include/qparity.hpp)
     virtual void UniformParityRZ(bitCapInt mask, R angle);
