@@ -149,6 +149,106 @@ template <typename R> static void bindQInterface(py::module_& m, const char* nam
         .def("cs", [](QI& q, bitLenInt c, bitLenInt t) { q.CS(c, t); })
         .def("cphase_root_n", [](QI& q, bitLenInt n, bitLenInt c, bitLenInt t) { q.CPhaseRootN(n, c, t); })
         .def("crz", [](QI& q, R th, bitLenInt c, bitLenInt t) { q.CRZ(th, c, t); })
+        .def("crx", [](QI& q, R th, bitLenInt c, bitLenInt t) { q.CRX(th, c, t); })
+        .def("cry", [](QI& q, R th, bitLenInt c, bitLenInt t) { q.CRY(th, c, t); })
+        .def("crt", [](QI& q, R th, bitLenInt c, bitLenInt t) { q.CRT(th, c, t); })
+        .def("ct", &QI::CT)
+        .def("cit", &QI::CIT)
+        .def("ccy", &QI::CCY)
+        .def("anti_cy", &QI::AntiCY)
+        .def("anti_ccy", &QI::AntiCCY)
+        .def("anti_ccz", &QI::AntiCCZ)
+        .def("anti_ch", &QI::AntiCH)
+        .def("anti_cs", &QI::AntiCS)
+        .def("anti_cis", &QI::AntiCIS)
+        .def("anti_ct", &QI::AntiCT)
+        .def("anti_cit", &QI::AntiCIT)
+        .def("anti_cphase_root_n", &QI::AntiCPhaseRootN)
+        .def("anti_ciphase_root_n", &QI::AntiCIPhaseRootN)
+        .def("cu", [](QI& q, std::vector<bitLenInt> c, bitLenInt t, R th, R ph, R lm) {
+            q.CU(c, t, th, ph, lm);
+        })
+        .def("anti_cu", [](QI& q, std::vector<bitLenInt> c, bitLenInt t, R th, R ph, R lm) {
+            q.AntiCU(c, t, th, ph, lm);
+        })
+        .def("u2", [](QI& q, bitLenInt t, R ph, R lm) { q.U2(t, ph, lm); })
+        .def("iu2", [](QI& q, bitLenInt t, R ph, R lm) { q.IU2(t, ph, lm); })
+        .def("ai", [](QI& q, bitLenInt t, R az, R incl) { q.AI(t, az, incl); })
+        .def("iai", [](QI& q, bitLenInt t, R az, R incl) { q.IAI(t, az, incl); })
+        .def("cai", [](QI& q, bitLenInt c, bitLenInt t, R az, R incl) { q.CAI(c, t, az, incl); })
+        .def("anti_cai", [](QI& q, bitLenInt c, bitLenInt t, R az, R incl) { q.AntiCAI(c, t, az, incl); })
+        .def("ciai", [](QI& q, bitLenInt c, bitLenInt t, R az, R incl) { q.CIAI(c, t, az, incl); })
+        .def("anti_ciai", [](QI& q, bitLenInt c, bitLenInt t, R az, R incl) { q.AntiCIAI(c, t, az, incl); })
+        .def("sqrt_h", &QI::SqrtH)
+        .def("sh", &QI::SH)
+        .def("his", &QI::HIS)
+        .def("sqrt_w", &QI::SqrtW)
+        .def("isqrt_w", &QI::ISqrtW)
+        .def("sqrt_y", &QI::SqrtY)
+        .def("isqrt_y", &QI::ISqrtY)
+        .def("phase_root_n_mask", &QI::PhaseRootNMask)
+        .def("uc_phase", [](QI& q, std::vector<bitLenInt> c, C tl, C br, bitLenInt t, bitCapInt perm) {
+            q.UCPhase(c, from_std<R>(tl), from_std<R>(br), t, perm);
+        })
+        .def("uc_invert", [](QI& q, std::vector<bitLenInt> c, C tr, C bl, bitLenInt t, bitCapInt perm) {
+            q.UCInvert(c, from_std<R>(tr), from_std<R>(bl), t, perm);
+        })
+        .def("exp_", [](QI& q, R r, bitLenInt t) { q.Exp(r, t); })
+        .def("exp_x", [](QI& q, R r, bitLenInt t) { q.ExpX(r, t); })
+        .def("exp_y", [](QI& q, R r, bitLenInt t) { q.ExpY(r, t); })
+        .def("exp_z", [](QI& q, R r, bitLenInt t) { q.ExpZ(r, t); })
+        .def("rx_dyad", &QI::RXDyad)
+        .def("ry_dyad", &QI::RYDyad)
+        .def("rz_dyad", &QI::RZDyad)
+        .def("rt_dyad", &QI::RTDyad)
+        .def("exp_dyad", &QI::ExpDyad)
+        .def("exp_x_dyad", &QI::ExpXDyad)
+        .def("exp_y_dyad", &QI::ExpYDyad)
+        .def("exp_z_dyad", &QI::ExpZDyad)
+        .def("crx_dyad", &QI::CRXDyad)
+        .def("cry_dyad", &QI::CRYDyad)
+        .def("crz_dyad", &QI::CRZDyad)
+        .def("crt_dyad", &QI::CRTDyad)
+        .def("uniformly_controlled_ry",
+            [](QI& q, std::vector<bitLenInt> c, bitLenInt t, std::vector<R> angles) {
+                q.UniformlyControlledRY(c, t, angles);
+            })
+        .def("uniformly_controlled_rz",
+            [](QI& q, std::vector<bitLenInt> c, bitLenInt t, std::vector<R> angles) {
+                q.UniformlyControlledRZ(c, t, angles);
+            })
+        .def("cisqrt_swap", [](QI& q, std::vector<bitLenInt> c, bitLenInt a, bitLenInt b) {
+            q.CISqrtSwap(c, a, b);
+        })
+        .def("anti_cisqrt_swap", [](QI& q, std::vector<bitLenInt> c, bitLenInt a, bitLenInt b) {
+            q.AntiCISqrtSwap(c, a, b);
+        })
+        // boolean logic, shifts, adders, classical assignment
+        .def("and_", &QI::AND)
+        .def("or_", &QI::OR)
+        .def("xor_", &QI::XOR)
+        .def("nand", &QI::NAND)
+        .def("nor", &QI::NOR)
+        .def("xnor", &QI::XNOR)
+        .def("cland", &QI::CLAND)
+        .def("clor", &QI::CLOR)
+        .def("clxor", &QI::CLXOR)
+        .def("clnand", &QI::CLNAND)
+        .def("clnor", &QI::CLNOR)
+        .def("clxnor", &QI::CLXNOR)
+        .def("asl", &QI::ASL)
+        .def("asr", &QI::ASR)
+        .def("lsl", &QI::LSL)
+        .def("lsr", &QI::LSR)
+        .def("adc", &QI::ADC)
+        .def("iadc", &QI::IADC)
+        .def("cadc", &QI::CADC)
+        .def("ciadc", &QI::CIADC)
+        .def("cfull_add", &QI::CFullAdd)
+        .def("cifull_add", &QI::CIFullAdd)
+        .def("set_bit", &QI::SetBit)
+        .def("set_reg", &QI::SetReg)
+        .def("reverse", &QI::Reverse)
         // ---- swaps ----
         .def("swap", [](QI& q, bitLenInt a, bitLenInt b) { q.Swap(a, b); })
         .def("iswap", [](QI& q, bitLenInt a, bitLenInt b) { q.ISwap(a, b); })

@@ -802,6 +802,204 @@ template <typename R> void QInterface<R>::CPhaseFlipIfLess(bitCapInt, bitLenInt,
     aluThrow<R>();
 }
 
+// ---- boolean logic / shifts / ripple adders (parity: reference logic.cpp,
+// arithmetic.cpp, qinterface.cpp ASL family) --------------------------------
+
+template <typename R> void QInterface<R>::AND(bitLenInt in1, bitLenInt in2, bitLenInt out)
+{
+    if (in1 == in2 && in2 == out) return;
+    if (in1 == out || in2 == out) throw QrackError("AND: output cannot alias an input");
+    if (in1 == in2) {
+        CNOT(in1, out);
+    } else {
+        CCNOT(in1, in2, out);
+    }
+}
+
+template <typename R> void QInterface<R>::OR(bitLenInt in1, bitLenInt in2, bitLenInt out)
+{
+    if (in1 == in2 && in2 == out) return;
+    if (in1 == out || in2 == out) throw QrackError("OR: output cannot alias an input");
+    // De Morgan: out ^= NOT(NOT a AND NOT b)
+    X(out);
+    if (in1 == in2) {
+        AntiCNOT(in1, out);
+    } else {
+        X(in1);
+        X(in2);
+        CCNOT(in1, in2, out);
+        X(in2);
+        X(in1);
+    }
+}
+
+template <typename R> void QInterface<R>::XOR(bitLenInt in1, bitLenInt in2, bitLenInt out)
+{
+    if (in1 == in2 && in2 == out) {
+        SetBit(out, false);
+        return;
+    }
+    if (in1 == out) {
+        CNOT(in2, out);
+    } else if (in2 == out) {
+        CNOT(in1, out);
+    } else {
+        CNOT(in1, out);
+        CNOT(in2, out);
+    }
+}
+
+template <typename R> void QInterface<R>::CLAND(bitLenInt qIn, bool cIn, bitLenInt out)
+{
+    if (cIn && qIn != out) CNOT(qIn, out);
+}
+
+template <typename R> void QInterface<R>::CLOR(bitLenInt qIn, bool cIn, bitLenInt out)
+{
+    if (cIn) {
+        X(out);
+    } else if (qIn != out) {
+        CNOT(qIn, out);
+    }
+}
+
+template <typename R> void QInterface<R>::CLXOR(bitLenInt qIn, bool cIn, bitLenInt out)
+{
+    if (qIn != out) {
+        if (cIn) X(out);
+        CNOT(qIn, out);
+    } else if (cIn) {
+        X(out);
+    }
+}
+
+template <typename R> void QInterface<R>::ASL(bitLenInt shift, bitLenInt start, bitLenInt length)
+{
+    if (!length || !shift) return;
+    if (shift >= length) {
+        SetReg(start, length, 0u);
+        return;
+    }
+    // keep the top (sign) bit in place: park it next door, rotate, zero-fill
+    const bitLenInt end = start + length;
+    Swap(end - 1u, end - 2u);
+    ROL(shift, start, length);
+    SetReg(start, shift, 0u);
+    Swap(end - 1u, end - 2u);
+}
+
+template <typename R> void QInterface<R>::ASR(bitLenInt shift, bitLenInt start, bitLenInt length)
+{
+    if (!length || !shift) return;
+    if (shift >= length) {
+        SetReg(start, length, 0u);
+        return;
+    }
+    const bitLenInt end = start + length;
+    Swap(end - 1u, end - 2u);
+    ROR(shift, start, length);
+    SetReg(end - shift - 1u, shift, 0u);
+    Swap(end - 1u, end - 2u);
+}
+
+template <typename R> void QInterface<R>::LSL(bitLenInt shift, bitLenInt start, bitLenInt length)
+{
+    if (!length || !shift) return;
+    if (shift >= length) {
+        SetReg(start, length, 0u);
+        return;
+    }
+    ROL(shift, start, length);
+    SetReg(start, shift, 0u);
+}
+
+template <typename R> void QInterface<R>::LSR(bitLenInt shift, bitLenInt start, bitLenInt length)
+{
+    if (!length || !shift) return;
+    if (shift >= length) {
+        SetReg(start, length, 0u);
+        return;
+    }
+    SetReg(start, shift, 0u);
+    ROR(shift, start, length);
+}
+
+template <typename R>
+void QInterface<R>::CFullAdd(const std::vector<bitLenInt>& ctrl, bitLenInt in1, bitLenInt in2,
+    bitLenInt carryInSumOut, bitLenInt carryOut)
+{
+    // FullAdd with every gate lifted by the control set
+    auto ctrlPlus = [&](std::initializer_list<bitLenInt> extra) {
+        std::vector<bitLenInt> c(ctrl);
+        c.insert(c.end(), extra.begin(), extra.end());
+        return c;
+    };
+    MCInvert(ctrlPlus({ in1, in2 }), cplx<R>(1, 0), cplx<R>(1, 0), carryOut);
+    MCInvert(ctrlPlus({ in1 }), cplx<R>(1, 0), cplx<R>(1, 0), in2);
+    MCInvert(ctrlPlus({ in2, carryInSumOut }), cplx<R>(1, 0), cplx<R>(1, 0), carryOut);
+    MCInvert(ctrlPlus({ in2 }), cplx<R>(1, 0), cplx<R>(1, 0), carryInSumOut);
+    MCInvert(ctrlPlus({ in1 }), cplx<R>(1, 0), cplx<R>(1, 0), in2);
+}
+
+template <typename R>
+void QInterface<R>::CIFullAdd(const std::vector<bitLenInt>& ctrl, bitLenInt in1, bitLenInt in2,
+    bitLenInt carryInSumOut, bitLenInt carryOut)
+{
+    auto ctrlPlus = [&](std::initializer_list<bitLenInt> extra) {
+        std::vector<bitLenInt> c(ctrl);
+        c.insert(c.end(), extra.begin(), extra.end());
+        return c;
+    };
+    MCInvert(ctrlPlus({ in1 }), cplx<R>(1, 0), cplx<R>(1, 0), in2);
+    MCInvert(ctrlPlus({ in2 }), cplx<R>(1, 0), cplx<R>(1, 0), carryInSumOut);
+    MCInvert(ctrlPlus({ in2, carryInSumOut }), cplx<R>(1, 0), cplx<R>(1, 0), carryOut);
+    MCInvert(ctrlPlus({ in1 }), cplx<R>(1, 0), cplx<R>(1, 0), in2);
+    MCInvert(ctrlPlus({ in1, in2 }), cplx<R>(1, 0), cplx<R>(1, 0), carryOut);
+}
+
+template <typename R>
+void QInterface<R>::ADC(bitLenInt in1, bitLenInt in2, bitLenInt output, bitLenInt length,
+    bitLenInt carry)
+{
+    // output := in1 + in2 + carryIn (per-bit ripple); carry := carry-out;
+    // inputs preserved. FullAdd leaves the sum in its carryInSumOut slot, so
+    // each step computes into `carry` then swaps the sum into the output bit.
+    for (bitLenInt i = 0; i < length; ++i) {
+        FullAdd(in1 + i, in2 + i, carry, output + i);
+        Swap(carry, output + i);
+    }
+}
+
+template <typename R>
+void QInterface<R>::IADC(bitLenInt in1, bitLenInt in2, bitLenInt output, bitLenInt length,
+    bitLenInt carry)
+{
+    for (bitLenInt i = length; i-- > 0u;) {
+        Swap(carry, output + i);
+        IFullAdd(in1 + i, in2 + i, carry, output + i);
+    }
+}
+
+template <typename R>
+void QInterface<R>::CADC(const std::vector<bitLenInt>& ctrl, bitLenInt in1, bitLenInt in2,
+    bitLenInt output, bitLenInt length, bitLenInt carry)
+{
+    for (bitLenInt i = 0; i < length; ++i) {
+        CFullAdd(ctrl, in1 + i, in2 + i, carry, output + i);
+        CSwap(ctrl, carry, output + i);
+    }
+}
+
+template <typename R>
+void QInterface<R>::CIADC(const std::vector<bitLenInt>& ctrl, bitLenInt in1, bitLenInt in2,
+    bitLenInt output, bitLenInt length, bitLenInt carry)
+{
+    for (bitLenInt i = length; i-- > 0u;) {
+        CSwap(ctrl, carry, output + i);
+        CIFullAdd(ctrl, in1 + i, in2 + i, carry, output + i);
+    }
+}
+
 template <typename R>
 void QInterface<R>::FullAdd(bitLenInt in1, bitLenInt in2, bitLenInt carryInSumOut, bitLenInt carryOut)
 {

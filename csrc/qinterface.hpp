@@ -249,6 +249,273 @@ public:
         MCPhase({ c }, polar<R>(1, -theta / 2), polar<R>(1, theta / 2), t);
     }
 
+    // ---- extended named gates (parity: reference gates/rotational API) -----
+    // "Azimuth, Inclination" Bloch-direction prep (rotational.cpp:55-75)
+    void AI(bitLenInt t, R azimuth, R inclination)
+    {
+        const R cA = std::cos(azimuth), sA = std::sin(azimuth);
+        const R cI = std::cos(inclination / 2), sI = std::sin(inclination / 2);
+        const cplx<R> m[4] = { { cI, 0 }, cplx<R>(-cA, sA) * sI, cplx<R>(cA, sA) * sI, { cI, 0 } };
+        Mtrx(m, t);
+    }
+    void IAI(bitLenInt t, R azimuth, R inclination)
+    {
+        const R cA = std::cos(azimuth), sA = std::sin(azimuth);
+        const R cI = std::cos(inclination / 2), sI = std::sin(inclination / 2);
+        const cplx<R> m[4] = { { cI, 0 }, cplx<R>(cA, -sA) * sI, cplx<R>(-cA, -sA) * sI,
+            { cI, 0 } };
+        Mtrx(m, t);
+    }
+    void CAI(bitLenInt c, bitLenInt t, R az, R incl)
+    {
+        const R cA = std::cos(az), sA = std::sin(az);
+        const R cI = std::cos(incl / 2), sI = std::sin(incl / 2);
+        const cplx<R> m[4] = { { cI, 0 }, cplx<R>(-cA, sA) * sI, cplx<R>(cA, sA) * sI, { cI, 0 } };
+        MCMtrx({ c }, m, t);
+    }
+    void AntiCAI(bitLenInt c, bitLenInt t, R az, R incl)
+    {
+        const R cA = std::cos(az), sA = std::sin(az);
+        const R cI = std::cos(incl / 2), sI = std::sin(incl / 2);
+        const cplx<R> m[4] = { { cI, 0 }, cplx<R>(-cA, sA) * sI, cplx<R>(cA, sA) * sI, { cI, 0 } };
+        MACMtrx({ c }, m, t);
+    }
+    void CIAI(bitLenInt c, bitLenInt t, R az, R incl)
+    {
+        const R cA = std::cos(az), sA = std::sin(az);
+        const R cI = std::cos(incl / 2), sI = std::sin(incl / 2);
+        const cplx<R> m[4] = { { cI, 0 }, cplx<R>(cA, -sA) * sI, cplx<R>(-cA, -sA) * sI,
+            { cI, 0 } };
+        MCMtrx({ c }, m, t);
+    }
+    void AntiCIAI(bitLenInt c, bitLenInt t, R az, R incl)
+    {
+        const R cA = std::cos(az), sA = std::sin(az);
+        const R cI = std::cos(incl / 2), sI = std::sin(incl / 2);
+        const cplx<R> m[4] = { { cI, 0 }, cplx<R>(cA, -sA) * sI, cplx<R>(-cA, -sA) * sI,
+            { cI, 0 } };
+        MACMtrx({ c }, m, t);
+    }
+    void CT(bitLenInt c, bitLenInt t) { MCPhase({ c }, cplx<R>(1, 0), polar<R>(1, PI_R<R> / 4), t); }
+    void CIT(bitLenInt c, bitLenInt t) { MCPhase({ c }, cplx<R>(1, 0), polar<R>(1, -PI_R<R> / 4), t); }
+    void AntiCY(bitLenInt c, bitLenInt t) { MACInvert({ c }, cplx<R>(0, -1), cplx<R>(0, 1), t); }
+    void CCY(bitLenInt c1, bitLenInt c2, bitLenInt t)
+    {
+        MCInvert({ c1, c2 }, cplx<R>(0, -1), cplx<R>(0, 1), t);
+    }
+    void AntiCCY(bitLenInt c1, bitLenInt c2, bitLenInt t)
+    {
+        MACInvert({ c1, c2 }, cplx<R>(0, -1), cplx<R>(0, 1), t);
+    }
+    void AntiCCZ(bitLenInt c1, bitLenInt c2, bitLenInt t)
+    {
+        MACPhase({ c1, c2 }, cplx<R>(1, 0), cplx<R>(-1, 0), t);
+    }
+    void AntiCH(bitLenInt c, bitLenInt t)
+    {
+        const R s = SQRT1_2_R<R>;
+        const cplx<R> m[4] = { { s, 0 }, { s, 0 }, { s, 0 }, { -s, 0 } };
+        MACMtrx({ c }, m, t);
+    }
+    void AntiCS(bitLenInt c, bitLenInt t) { MACPhase({ c }, cplx<R>(1, 0), cplx<R>(0, 1), t); }
+    void AntiCIS(bitLenInt c, bitLenInt t) { MACPhase({ c }, cplx<R>(1, 0), cplx<R>(0, -1), t); }
+    void AntiCT(bitLenInt c, bitLenInt t)
+    {
+        MACPhase({ c }, cplx<R>(1, 0), polar<R>(1, PI_R<R> / 4), t);
+    }
+    void AntiCIT(bitLenInt c, bitLenInt t)
+    {
+        MACPhase({ c }, cplx<R>(1, 0), polar<R>(1, -PI_R<R> / 4), t);
+    }
+    void AntiCPhaseRootN(bitLenInt n, bitLenInt c, bitLenInt t)
+    {
+        if (n == 0) return;
+        MACPhase({ c }, cplx<R>(1, 0), polar<R>(1, PI_R<R> / (R)pow2(n - 1)), t);
+    }
+    void AntiCIPhaseRootN(bitLenInt n, bitLenInt c, bitLenInt t)
+    {
+        if (n == 0) return;
+        MACPhase({ c }, cplx<R>(1, 0), polar<R>(1, -PI_R<R> / (R)pow2(n - 1)), t);
+    }
+    void CU(const std::vector<bitLenInt>& controls, bitLenInt t, R theta, R phi, R lambda)
+    {
+        const R c = std::cos(theta / 2), s = std::sin(theta / 2);
+        const cplx<R> m[4] = { { c, 0 }, (R)(-s) * polar<R>(1, lambda), s * polar<R>(1, phi),
+            c * polar<R>(1, phi + lambda) };
+        MCMtrx(controls, m, t);
+    }
+    void AntiCU(const std::vector<bitLenInt>& controls, bitLenInt t, R theta, R phi, R lambda)
+    {
+        const R c = std::cos(theta / 2), s = std::sin(theta / 2);
+        const cplx<R> m[4] = { { c, 0 }, (R)(-s) * polar<R>(1, lambda), s * polar<R>(1, phi),
+            c * polar<R>(1, phi + lambda) };
+        MACMtrx(controls, m, t);
+    }
+    void IU2(bitLenInt q, R phi, R lambda)
+    {
+        U(q, PI_R<R> / 2, -lambda - PI_R<R>, -phi + PI_R<R>);
+    }
+    // sqrt(H): H = SqrtH * SqrtH (gates API parity)
+    void SqrtH(bitLenInt q)
+    {
+        const R s2 = (R)1.4142135623730951;
+        const cplx<R> m[4] = { { (R)((1 + s2) / (2 * s2)), (R)((-1 + s2) / (2 * s2)) },
+            { SQRT1_2_R<R> / 2, -SQRT1_2_R<R> / 2 }, { SQRT1_2_R<R> / 2, -SQRT1_2_R<R> / 2 },
+            { (R)((-1 + s2) / (2 * s2)), (R)((1 + s2) / (2 * s2)) } };
+        Mtrx(m, q);
+    }
+    // Y-basis transforms: SH = S*H (Z->Y), HIS = H*IS (Y->Z)
+    void SH(bitLenInt q)
+    {
+        const R s = SQRT1_2_R<R>;
+        const cplx<R> m[4] = { { s, 0 }, { s, 0 }, { 0, s }, { 0, -s } };
+        Mtrx(m, q);
+    }
+    void HIS(bitLenInt q)
+    {
+        const R s = SQRT1_2_R<R>;
+        const cplx<R> m[4] = { { s, 0 }, { 0, -s }, { s, 0 }, { 0, s } };
+        Mtrx(m, q);
+    }
+    // sqrt(W) and inverse, Sycamore gate set (2019 Arute)
+    void SqrtW(bitLenInt q)
+    {
+        const R s = SQRT1_2_R<R>;
+        const cplx<R> m[4] = { { s, 0 }, { (R)-0.5, (R)-0.5 }, { (R)0.5, (R)-0.5 }, { s, 0 } };
+        Mtrx(m, q);
+    }
+    void ISqrtW(bitLenInt q)
+    {
+        const R s = SQRT1_2_R<R>;
+        const cplx<R> m[4] = { { s, 0 }, { (R)0.5, (R)0.5 }, { (R)-0.5, (R)0.5 }, { s, 0 } };
+        Mtrx(m, q);
+    }
+    void CISqrtSwap(const std::vector<bitLenInt>& controls, bitLenInt q1, bitLenInt q2)
+    {
+        if (q1 == q2) return;
+        const cplx<R> m[4] = { { (R)0.5, (R)-0.5 }, { (R)0.5, (R)0.5 }, { (R)0.5, (R)0.5 },
+            { (R)0.5, (R)-0.5 } };
+        std::vector<bitLenInt> c2(controls);
+        c2.push_back(q2);
+        CNOT(q1, q2);
+        MCMtrx(c2, m, q1);
+        CNOT(q1, q2);
+    }
+    void AntiCISqrtSwap(const std::vector<bitLenInt>& controls, bitLenInt q1, bitLenInt q2)
+    {
+        for (bitLenInt c : controls) X(c);
+        CISqrtSwap(controls, q1, q2);
+        for (bitLenInt c : controls) X(c);
+    }
+    void PhaseRootNMask(bitLenInt n, bitCapInt mask)
+    {
+        bitCapInt m = mask;
+        while (m) {
+            PhaseRootN(n, log2Ocl(m & (~m + 1u)));
+            m &= m - 1u;
+        }
+    }
+    // mixed-polarity controlled phase / invert (UCMtrx specializations)
+    void UCPhase(const std::vector<bitLenInt>& controls, cplx<R> tl, cplx<R> br, bitLenInt t,
+        bitCapInt perm)
+    {
+        const cplx<R> m[4] = { tl, cplx<R>(0, 0), cplx<R>(0, 0), br };
+        UCMtrx(controls, m, t, perm);
+    }
+    void UCInvert(const std::vector<bitLenInt>& controls, cplx<R> tr, cplx<R> bl, bitLenInt t,
+        bitCapInt perm)
+    {
+        const cplx<R> m[4] = { cplx<R>(0, 0), tr, bl, cplx<R>(0, 0) };
+        UCMtrx(controls, m, t, perm);
+    }
+
+    // ---- extended rotation API (reference rotational.cpp + dyadic forms) ---
+    // Exp family: phase factor e^{i radians} times the (I/X/Y/Z) operator
+    void Exp(R radians, bitLenInt q)
+    {
+        const cplx<R> f = polar<R>(1, radians);
+        Phase(f, f, q);
+    }
+    void ExpX(R radians, bitLenInt q)
+    {
+        const cplx<R> f = polar<R>(1, radians);
+        Invert(f, f, q);
+    }
+    void ExpY(R radians, bitLenInt q)
+    {
+        const cplx<R> f = polar<R>(1, radians);
+        Invert(f * cplx<R>(0, -1), f * cplx<R>(0, 1), q);
+    }
+    void ExpZ(R radians, bitLenInt q)
+    {
+        const cplx<R> f = polar<R>(1, radians);
+        Phase(f, cplx<R>(0, 0) - f, q);
+    }
+    void CRX(R theta, bitLenInt c, bitLenInt t)
+    {
+        const R co = std::cos(theta / 2), si = std::sin(theta / 2);
+        const cplx<R> m[4] = { { co, 0 }, { 0, -si }, { 0, -si }, { co, 0 } };
+        MCMtrx({ c }, m, t);
+    }
+    void CRY(R theta, bitLenInt c, bitLenInt t)
+    {
+        const R co = std::cos(theta / 2), si = std::sin(theta / 2);
+        const cplx<R> m[4] = { { co, 0 }, { -si, 0 }, { si, 0 }, { co, 0 } };
+        MCMtrx({ c }, m, t);
+    }
+    void CRT(R theta, bitLenInt c, bitLenInt t)
+    {
+        MCPhase({ c }, cplx<R>(1, 0), polar<R>(1, theta), t);
+    }
+    // dyadic fractions: angle = -2 pi * numerator / 2^denomPower
+    // (reference qinterface.cpp:1310 dyadAngle)
+    R DyadAngle(int numerator, int denomPower) const
+    {
+        return (R)((-PI_R<R> * numerator * 2) / (double)pow2(denomPower));
+    }
+    void RXDyad(int n, int d, bitLenInt q) { RX(DyadAngle(n, d), q); }
+    void RYDyad(int n, int d, bitLenInt q) { RY(DyadAngle(n, d), q); }
+    void RZDyad(int n, int d, bitLenInt q) { RZ(DyadAngle(n, d), q); }
+    void RTDyad(int n, int d, bitLenInt q) { RT(DyadAngle(n, d), q); }
+    void ExpDyad(int n, int d, bitLenInt q) { Exp(DyadAngle(n, d), q); }
+    void ExpXDyad(int n, int d, bitLenInt q) { ExpX(DyadAngle(n, d), q); }
+    void ExpYDyad(int n, int d, bitLenInt q) { ExpY(DyadAngle(n, d), q); }
+    void ExpZDyad(int n, int d, bitLenInt q) { ExpZ(DyadAngle(n, d), q); }
+    void CRXDyad(int n, int d, bitLenInt c, bitLenInt t) { CRX(DyadAngle(n, d), c, t); }
+    void CRYDyad(int n, int d, bitLenInt c, bitLenInt t) { CRY(DyadAngle(n, d), c, t); }
+    void CRZDyad(int n, int d, bitLenInt c, bitLenInt t) { CRZ(DyadAngle(n, d), c, t); }
+    void CRTDyad(int n, int d, bitLenInt c, bitLenInt t) { CRT(DyadAngle(n, d), c, t); }
+    // multiplexed rotations: one angle per control permutation
+    void UniformlyControlledRY(
+        const std::vector<bitLenInt>& controls, bitLenInt t, const std::vector<R>& angles)
+    {
+        const bitCapInt n = pow2((bitLenInt)controls.size());
+        if ((bitCapInt)angles.size() < n) throw QrackError("UniformlyControlledRY: need 2^k angles");
+        std::vector<cplx<R>> ms(4u * n);
+        for (bitCapInt i = 0; i < n; ++i) {
+            const R co = std::cos(angles[i] / 2), si = std::sin(angles[i] / 2);
+            ms[4u * i] = { co, 0 };
+            ms[4u * i + 1u] = { -si, 0 };
+            ms[4u * i + 2u] = { si, 0 };
+            ms[4u * i + 3u] = { co, 0 };
+        }
+        UniformlyControlledSingleBit(controls, t, ms.data());
+    }
+    void UniformlyControlledRZ(
+        const std::vector<bitLenInt>& controls, bitLenInt t, const std::vector<R>& angles)
+    {
+        const bitCapInt n = pow2((bitLenInt)controls.size());
+        if ((bitCapInt)angles.size() < n) throw QrackError("UniformlyControlledRZ: need 2^k angles");
+        std::vector<cplx<R>> ms(4u * n);
+        for (bitCapInt i = 0; i < n; ++i) {
+            ms[4u * i] = polar<R>(1, -angles[i] / 2);
+            ms[4u * i + 1u] = { 0, 0 };
+            ms[4u * i + 2u] = { 0, 0 };
+            ms[4u * i + 3u] = polar<R>(1, angles[i] / 2);
+        }
+        UniformlyControlledSingleBit(controls, t, ms.data());
+    }
+
     // ---- register-wide helpers ---------------------------------------------
     void X(bitLenInt start, bitLenInt length)
     {
@@ -293,6 +560,27 @@ public:
 
     virtual bool ForceM(bitLenInt q, bool result, bool doForce = true, bool doApply = true) = 0;
     bool M(bitLenInt q) { return ForceM(q, false, false, true); }
+    // measure-and-correct classical assignment (parity: SetBit/SetReg)
+    virtual void SetBit(bitLenInt q, bool value)
+    {
+        if (M(q) != value) X(q);
+    }
+    virtual void SetReg(bitLenInt start, bitLenInt length, bitCapInt value)
+    {
+        if (!length) return;
+        const bitCapInt cur = MReg(start, length);
+        const bitCapInt diff = cur ^ (value & pow2Mask(length));
+        if (diff) XMask(diff << start);
+    }
+    // reverse qubit order in [first, last) via swaps (parity: Reverse)
+    virtual void Reverse(bitLenInt first, bitLenInt last)
+    {
+        while ((last > 0u) && (first < (last - 1u))) {
+            --last;
+            Swap(first, last);
+            ++first;
+        }
+    }
     virtual bool ForceMParity(bitCapInt mask, bool result, bool doForce = true);
     virtual bitCapInt ForceMReg(
         bitLenInt start, bitLenInt length, bitCapInt result, bool doForce = true, bool doApply = true);
@@ -444,6 +732,66 @@ public:
         bitLenInt carryOut);
     virtual void IFullAdd(bitLenInt inputBit1, bitLenInt inputBit2, bitLenInt carryInSumOut,
         bitLenInt carryOut);
+
+    // controlled full adders + register ripple adders with carry
+    // (parity: reference arithmetic.cpp CFullAdd/CIFullAdd/ADC/IADC/CADC/CIADC)
+    virtual void CFullAdd(const std::vector<bitLenInt>& controls, bitLenInt inputBit1,
+        bitLenInt inputBit2, bitLenInt carryInSumOut, bitLenInt carryOut);
+    virtual void CIFullAdd(const std::vector<bitLenInt>& controls, bitLenInt inputBit1,
+        bitLenInt inputBit2, bitLenInt carryInSumOut, bitLenInt carryOut);
+    virtual void ADC(bitLenInt input1, bitLenInt input2, bitLenInt output, bitLenInt length,
+        bitLenInt carry);
+    virtual void IADC(bitLenInt input1, bitLenInt input2, bitLenInt output, bitLenInt length,
+        bitLenInt carry);
+    virtual void CADC(const std::vector<bitLenInt>& controls, bitLenInt input1, bitLenInt input2,
+        bitLenInt output, bitLenInt length, bitLenInt carry);
+    virtual void CIADC(const std::vector<bitLenInt>& controls, bitLenInt input1, bitLenInt input2,
+        bitLenInt output, bitLenInt length, bitLenInt carry);
+
+    // boolean logic on qubits (output assumed |0> where the reference assumes
+    // it; parity: reference logic.cpp)
+    virtual void AND(bitLenInt in1, bitLenInt in2, bitLenInt out);
+    virtual void OR(bitLenInt in1, bitLenInt in2, bitLenInt out);
+    virtual void XOR(bitLenInt in1, bitLenInt in2, bitLenInt out);
+    void NAND(bitLenInt in1, bitLenInt in2, bitLenInt out)
+    {
+        AND(in1, in2, out);
+        X(out);
+    }
+    void NOR(bitLenInt in1, bitLenInt in2, bitLenInt out)
+    {
+        OR(in1, in2, out);
+        X(out);
+    }
+    void XNOR(bitLenInt in1, bitLenInt in2, bitLenInt out)
+    {
+        XOR(in1, in2, out);
+        X(out);
+    }
+    virtual void CLAND(bitLenInt qIn, bool cIn, bitLenInt out);
+    virtual void CLOR(bitLenInt qIn, bool cIn, bitLenInt out);
+    virtual void CLXOR(bitLenInt qIn, bool cIn, bitLenInt out);
+    void CLNAND(bitLenInt qIn, bool cIn, bitLenInt out)
+    {
+        CLAND(qIn, cIn, out);
+        X(out);
+    }
+    void CLNOR(bitLenInt qIn, bool cIn, bitLenInt out)
+    {
+        CLOR(qIn, cIn, out);
+        X(out);
+    }
+    void CLXNOR(bitLenInt qIn, bool cIn, bitLenInt out)
+    {
+        CLXOR(qIn, cIn, out);
+        X(out);
+    }
+
+    // register shifts (parity: reference qinterface.cpp ASL/ASR/LSL/LSR)
+    virtual void ASL(bitLenInt shift, bitLenInt start, bitLenInt length);
+    virtual void ASR(bitLenInt shift, bitLenInt start, bitLenInt length);
+    virtual void LSL(bitLenInt shift, bitLenInt start, bitLenInt length);
+    virtual void LSR(bitLenInt shift, bitLenInt start, bitLenInt length);
     virtual void PhaseFlipIfLess(bitCapInt greaterPerm, bitLenInt start, bitLenInt length);
     virtual void CPhaseFlipIfLess(
         bitCapInt greaterPerm, bitLenInt start, bitLenInt length, bitLenInt flagIndex);
