@@ -454,6 +454,19 @@ template <typename R> static void bindQInterface(py::module_& m, const char* nam
         .def("get_ncrp", &QI::GetNcrp)
         .def("set_reactive_separate", &QI::SetReactiveSeparate)
         .def("get_reactive_separate", &QI::GetReactiveSeparate)
+        .def("set_noise_parameter", &QI::SetNoiseParameter)
+        .def("get_noise_parameter", &QI::GetNoiseParameter)
+        .def("set_ace_max_qubits", &QI::SetAceMaxQubits)
+        .def("get_ace_max_qubits", &QI::GetAceMaxQubits)
+        .def("set_concurrency", &QI::SetConcurrency)
+        .def("set_t_injection", &QI::SetTInjection)
+        .def("get_t_injection", &QI::GetTInjection)
+        .def("first_nonzero_phase", &QI::FirstNonzeroPhase)
+        .def("highest_prob_all", &QI::HighestProbAll)
+        .def("sample_clone", &QI::SampleClone)
+        .def("try_decompose",
+            [](QI& q, bitLenInt start, Ptr dest) { return q.TryDecompose(start, dest); })
+        .def("get_device_list", &QI::GetDeviceList)
         .def("reset_unitary_fidelity", &QI::ResetUnitaryFidelity)
         .def("set_device", &QI::SetDevice)
         .def("get_device", &QI::GetDevice)

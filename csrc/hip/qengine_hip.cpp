@@ -746,6 +746,11 @@ template <typename R> bitCapInt QEngineHIP<R>::MAll()
     return result;
 }
 
+template <typename R> bitCapInt QEngineHIP<R>::HighestProbAll()
+{
+    return argMax().second;
+}
+
 template <typename R>
 std::map<bitCapInt, int> QEngineHIP<R>::MultiShotMeasureMask(
     const std::vector<bitCapInt>& qPowers, unsigned shots)

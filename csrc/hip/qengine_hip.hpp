@@ -153,6 +153,7 @@ public:
     bitCapInt MAll() override;
     std::map<bitCapInt, int> MultiShotMeasureMask(
         const std::vector<bitCapInt>& qPowers, unsigned shots) override;
+    bitCapInt HighestProbAll() override;
     double ExpectationBitsFactorized(const std::vector<bitLenInt>& bits,
         const std::vector<bitCapInt>& perms, bitCapInt offset = 0) override;
     double VarianceBitsAll(const std::vector<bitLenInt>& bits, bitCapInt offset = 0) override;

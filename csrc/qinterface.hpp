@@ -839,6 +839,82 @@ public:
     // reference's QUnit default behavior toggle (SetReactiveSeparate).
     virtual void SetReactiveSeparate(bool on) { (void)on; }
     virtual bool GetReactiveSeparate() { return false; }
+
+    // ---- misc knobs / queries (parity: reference qinterface.hpp) ----------
+    // thread-count hint; the thread pool already sizes itself from
+    // QRACK_MAX_CPU_THREADS / hardware_concurrency, so the default ignores it
+    virtual void SetConcurrency(uint32_t threads) { (void)threads; }
+    // per-gate depolarizing noise strength (acted on by QInterfaceNoisy)
+    virtual void SetNoiseParameter(double np) { (void)np; }
+    virtual double GetNoiseParameter() { return 0.0; }
+    // ACE width cap (acted on by QUnit; 0 = unlimited)
+    virtual void SetAceMaxQubits(bitLenInt maxQb) { (void)maxQb; }
+    virtual bitLenInt GetAceMaxQubits() { return 0; }
+    // T-injection gadget toggle: reserved (the tableau T-gadget is a
+    // planned round-2 feature; NCRP covers near-Clifford phases today)
+    virtual void SetTInjection(bool on) { (void)on; }
+    virtual bool GetTInjection() { return false; }
+    virtual void SetUseExactNearClifford(bool on) { if (on) SetNcrp(0.0); }
+    virtual bool GetUseExactNearClifford() { return GetNcrp() <= 0.0; }
+    // stochastic-rounding toggle for approximation layers: reserved
+    virtual void SetStochastic(bool on) { (void)on; }
+    virtual bool GetIsArbitraryGlobalPhase() { return false; }
+    virtual std::vector<int64_t> GetDeviceList() { return { GetDevice() }; }
+
+    // phase of the first nonzero amplitude (parity: FirstNonzeroPhase)
+    virtual double FirstNonzeroPhase()
+    {
+        for (bitCapInt i = 0; i < maxQPower; ++i) {
+            const cplx<R> a = GetAmplitude(i);
+            if (norm(a) > (double)eps<R>::value) return std::atan2((double)a.im, (double)a.re);
+        }
+        return 0.0;
+    }
+    // most probable basis state (parity: HighestProbAll; engines override)
+    virtual bitCapInt HighestProbAll()
+    {
+        bitCapInt best = 0;
+        double bestP = -1.0;
+        for (bitCapInt i = 0; i < maxQPower; ++i) {
+            const double p = (double)ProbAll(i);
+            if (p > bestP) {
+                bestP = p;
+                best = i;
+            }
+        }
+        return best;
+    }
+    // one terminal-measurement sample drawn from a throwaway clone
+    virtual bitCapInt SampleClone(const std::vector<bitCapInt>& qPowers)
+    {
+        QInterfacePtr<R> c = Clone();
+        const bitCapInt raw = c->MAll();
+        bitCapInt out = 0;
+        for (size_t b = 0; b < qPowers.size(); ++b) {
+            if (raw & qPowers[b]) out |= (ONE_BCI << b);
+        }
+        return out;
+    }
+    // Clone alias (reference Copy)
+    virtual QInterfacePtr<R> Copy() { return Clone(); }
+    // Compose that may consume the source (default: plain Compose)
+    virtual bitLenInt ComposeNoClone(QInterfacePtr<R> toCopy) { return Compose(toCopy); }
+    // attempt Decompose; on failure leave the state untouched and return false
+    virtual bool TryDecompose(bitLenInt start, QInterfacePtr<R> dest, R error_tol = eps<R>::value)
+    {
+        QInterfacePtr<R> probe = Clone();
+        QInterfacePtr<R> probeDest = dest->Clone();
+        try {
+            probe->Decompose(start, probeDest);
+        } catch (const std::exception&) {
+            return false;
+        }
+        probe->Compose(probeDest, start);
+        const double diff = probe->SumSqrDiff(this->shared_from_this());
+        if (diff > (double)error_tol) return false;
+        Decompose(start, dest);
+        return true;
+    }
     virtual void ResetUnitaryFidelity() {}
     virtual void SetDevice(int64_t deviceId) {}
     virtual int64_t GetDevice() const { return -1; }

@@ -34,8 +34,8 @@ public:
         }
     }
 
-    void SetNoiseParameter(R np) { noiseParam = np; }
-    R GetNoiseParameter() const { return noiseParam; }
+    void SetNoiseParameter(double np) override { noiseParam = (R)np; }
+    double GetNoiseParameter() override { return (double)noiseParam; }
 
     // per-gate noise injection must see every gate: undo the wrapper's fused
     // forwarding and lower the batch through Mtrx one gate at a time

@@ -111,6 +111,9 @@ public:
     void SetReactiveSeparate(bool on) override { reactiveSeparate = on; }
     bool GetReactiveSeparate() override { return reactiveSeparate; }
 
+    void SetAceMaxQubits(bitLenInt maxQb) override { aceMaxQubits = maxQb; }
+    bitLenInt GetAceMaxQubits() override { return aceMaxQubits; }
+
     // ---- state ----
     void SetPermutation(bitCapInt perm, cplx<R> phase = cplx<R>((R)1, (R)0)) override;
     void SetQuantumState(const cplx<R>* inputState) override;

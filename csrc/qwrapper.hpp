@@ -43,6 +43,15 @@ public:
     double GetNcrp() override { return inner->GetNcrp(); }
     void SetReactiveSeparate(bool on) override { inner->SetReactiveSeparate(on); }
     bool GetReactiveSeparate() override { return inner->GetReactiveSeparate(); }
+    void SetAceMaxQubits(bitLenInt m) override { inner->SetAceMaxQubits(m); }
+    bitLenInt GetAceMaxQubits() override { return inner->GetAceMaxQubits(); }
+    void SetConcurrency(uint32_t t) override { inner->SetConcurrency(t); }
+    void SetTInjection(bool on) override { inner->SetTInjection(on); }
+    bool GetTInjection() override { return inner->GetTInjection(); }
+    void SetStochastic(bool on) override { inner->SetStochastic(on); }
+    std::vector<int64_t> GetDeviceList() override { return inner->GetDeviceList(); }
+    double FirstNonzeroPhase() override { return inner->FirstNonzeroPhase(); }
+    bitCapInt HighestProbAll() override { return inner->HighestProbAll(); }
     void Phase(cplx<R> tl, cplx<R> br, bitLenInt t) override { inner->Phase(tl, br, t); }
     void Invert(cplx<R> tr, cplx<R> bl, bitLenInt t) override { inner->Invert(tr, bl, t); }
     void MCMtrx(const std::vector<bitLenInt>& c, const cplx<R>* m, bitLenInt t) override

@@ -230,3 +230,52 @@ def test_anti_controlled_named():
     q2.x(1)
     q2.ccy(0, 1, 2)
     assert abs(q2.prob(2) - 1.0) < 1e-9
+
+
+def test_misc_knobs_and_queries():
+    q = cpu(3)
+    q.set_concurrency(4)  # hint; no-op
+    q.x(1)
+    assert q.highest_prob_all() == 2
+    q.h(0)
+    # first nonzero phase of H|0> (x) |010> is 0
+    assert abs(q.first_nonzero_phase()) < 1e-9
+    s = q.sample_clone([1, 2, 4])
+    assert s in (0b010, 0b011)
+    # sampling from a clone must not disturb the state
+    assert abs(q.prob(0) - 0.5) < 1e-9
+
+
+def test_noise_parameter_knob():
+    q = qa.create_simulator(2, layers=["noisy", "cpu"], seed=1)
+    q.set_noise_parameter(0.0)
+    assert q.get_noise_parameter() == 0.0
+    q.h(0)
+    q.cnot(0, 1)
+    assert abs(q.prob(1) - 0.5) < 1e-6  # zero noise: exact Bell
+
+
+def test_ace_max_qubits_knob():
+    q = qa.create_simulator(8, layers=["qunit", "cpu"], seed=2)
+    q.set_ace_max_qubits(2)
+    assert q.get_ace_max_qubits() == 2
+    q.h(0)
+    q.cnot(0, 1)  # exactly at cap: allowed
+    assert abs(q.prob(1) - 0.5) < 1e-6
+
+
+def test_try_decompose():
+    # separable split succeeds; entangled split fails and leaves state intact
+    q = cpu(3)
+    q.x(2)
+    dest = qa.create_simulator(1, engine="cpu", precision="fp64", seed=9)
+    assert q.try_decompose(2, dest)
+    assert q.num_qubits == 2
+    assert abs(dest.prob(0) - 1.0) < 1e-9
+    q2 = cpu(2)
+    q2.h(0)
+    q2.cnot(0, 1)
+    dest2 = qa.create_simulator(1, engine="cpu", precision="fp64", seed=9)
+    assert not q2.try_decompose(1, dest2)
+    assert q2.num_qubits == 2
+    assert abs(q2.prob(1) - 0.5) < 1e-9
