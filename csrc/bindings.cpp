@@ -467,6 +467,13 @@ template <typename R> static void bindQInterface(py::module_& m, const char* nam
         .def("try_decompose",
             [](QI& q, bitLenInt start, Ptr dest) { return q.TryDecompose(start, dest); })
         .def("get_device_list", &QI::GetDeviceList)
+        .def("are_factorized",
+            [](QI& q, std::vector<bitLenInt> a, std::vector<bitLenInt> b, bool flush) {
+                return q.AreFactorized(a, b, flush);
+            }, py::arg("a"), py::arg("b"), py::arg("flush_cache") = false)
+        .def("get_amplitude_count", &QI::GetAmplitudeCount)
+        .def("set_sparse_ace_max_mb", &QI::SetSparseAceMaxMb)
+        .def("set_sparse_probability_floor", &QI::SetSparseProbabilityFloor)
         .def("reset_unitary_fidelity", &QI::ResetUnitaryFidelity)
         .def("set_device", &QI::SetDevice)
         .def("get_device", &QI::GetDevice)

@@ -44,6 +44,14 @@ protected:
     void MapPermutation(const std::function<bitCapInt(bitCapInt)>& f);
 
 public:
+    void SetSparseAceMaxMb(size_t mb) override
+    {
+        maxEntries = (mb << 20) / (sizeof(cplx<R>) + sizeof(bitCapInt));
+        TruncateToCap();
+    }
+    void SetSparseProbabilityFloor(double floorNorm) override { truncThresh = (R)floorNorm; }
+
+
     QEngineSparse(bitLenInt qBitCount, bitCapInt initState = 0u, RngPtr rgp = nullptr,
         bool doNorm = true, R normThresh = eps<R>::value);
 

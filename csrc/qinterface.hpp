@@ -860,6 +860,33 @@ public:
     virtual void SetStochastic(bool on) { (void)on; }
     virtual bool GetIsArbitraryGlobalPhase() { return false; }
     virtual std::vector<int64_t> GetDeviceList() { return { GetDevice() }; }
+    virtual void SetDeviceList(const std::vector<int64_t>& devices) { (void)devices; }
+    // pager-style nonzero-amplitude budget query; dense layers report the
+    // full register size (parity: GetAmplitudeCount)
+    virtual bitCapInt GetAmplitudeCount() { return maxQPower; }
+    // are the two qubit sets stored in disjoint internal factors? Default
+    // false, matching the reference's base (qinterface.hpp:2937); QUnit
+    // answers from its shard map.
+    virtual bool AreFactorized(
+        const std::vector<bitLenInt>& a, const std::vector<bitLenInt>& b, bool flushCache = false)
+    {
+        (void)a;
+        (void)b;
+        (void)flushCache;
+        return false;
+    }
+    // closest-Clifford S-quadrant hints (reference no-op defaults,
+    // qinterface.hpp:3040-3048; acted on by rounding layers when relevant)
+    virtual void SetMajorQuadrant(bool q) { (void)q; }
+    virtual void SetMajorQuadrant(bitLenInt t, bool q)
+    {
+        (void)t;
+        (void)q;
+    }
+    virtual void FlipQuadrant(bitLenInt t) { (void)t; }
+    // sparse-engine caps (acted on by QEngineSparse)
+    virtual void SetSparseAceMaxMb(size_t mb) { (void)mb; }
+    virtual void SetSparseProbabilityFloor(double floorNorm) { (void)floorNorm; }
 
     // phase of the first nonzero amplitude (parity: FirstNonzeroPhase)
     virtual double FirstNonzeroPhase()

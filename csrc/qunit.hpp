@@ -114,6 +114,21 @@ public:
     void SetAceMaxQubits(bitLenInt maxQb) override { aceMaxQubits = maxQb; }
     bitLenInt GetAceMaxQubits() override { return aceMaxQubits; }
 
+    bool AreFactorized(const std::vector<bitLenInt>& a, const std::vector<bitLenInt>& b,
+        bool flushCache = false) override
+    {
+        if (flushCache) {
+            for (bitLenInt x : a) TrySeparate(x);
+            for (bitLenInt x : b) TrySeparate(x);
+        }
+        std::set<QInterface<R>*> ua;
+        for (bitLenInt x : a) ua.insert(shards[x].unit.get());
+        for (bitLenInt x : b) {
+            if (ua.count(shards[x].unit.get())) return false;
+        }
+        return true;
+    }
+
     // ---- state ----
     void SetPermutation(bitCapInt perm, cplx<R> phase = cplx<R>((R)1, (R)0)) override;
     void SetQuantumState(const cplx<R>* inputState) override;

@@ -279,3 +279,21 @@ def test_try_decompose():
     assert not q2.try_decompose(1, dest2)
     assert q2.num_qubits == 2
     assert abs(q2.prob(1) - 0.5) < 1e-9
+
+
+def test_are_factorized():
+    q = qa.create_simulator(4, layers=["qunit", "cpu"], seed=3)
+    q.h(0)
+    q.cnot(0, 1)
+    assert not q.are_factorized([0], [1])
+    assert q.are_factorized([0, 1], [2, 3])
+    assert qa.create_simulator(2, engine="cpu").are_factorized([0], [1]) is False
+
+
+def test_sparse_knobs():
+    q = qa.create_simulator(10, engine="sparse", seed=4)
+    q.set_sparse_probability_floor(1e-10)
+    q.set_sparse_ace_max_mb(64)
+    for i in range(10):
+        q.h(i)
+    assert abs(q.prob(0) - 0.5) < 1e-4
