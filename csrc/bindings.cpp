@@ -505,6 +505,18 @@ template <typename R> static void bindQInterface(py::module_& m, const char* nam
                 if (!eng) throw QrackError("phase_ramp_general requires an engine");
                 eng->PhaseRampGeneral(scale, rampStart, inPlaceRelMask, sPows, sWeights, condPower);
             })
+        .def("qft_column_general",
+            [](Ptr q, bitLenInt target, double scale, bitLenInt rampStart,
+                bitCapInt inPlaceRelMask, std::vector<bitCapInt> sPows,
+                std::vector<uint64_t> sWeights, double phase0, bool pre) {
+                auto eng = std::dynamic_pointer_cast<QEngine<R>>(q);
+                if (!eng) throw QrackError("qft_column_general requires an engine");
+                eng->QftColumnGeneral(
+                    target, scale, rampStart, inPlaceRelMask, sPows, sWeights, phase0, pre);
+            },
+            py::arg("target"), py::arg("scale"), py::arg("ramp_start"),
+            py::arg("in_place_mask"), py::arg("pows"), py::arg("weights"),
+            py::arg("phase0") = 0.0, py::arg("pre") = false)
         .def("norm_total",
             [](Ptr q) {
                 auto eng = std::dynamic_pointer_cast<QEngine<R>>(q);

@@ -1043,6 +1043,103 @@ __global__ void k_qft_col_v(
     }
 }
 
+// generalized fused column: the ramp's bits may be relocated (distributed
+// pager lazy qubit maps) — frac(i) = ((i >> rampStart) & inPlaceRelMask) +
+// sum_k (i & sPow[k] ? sWeight[k] : 0), plus a constant phase0 (the
+// meta-page scalar term folded into the same pass).
+template <typename R, bool PRE>
+__global__ void k_qft_col_gen(cplx<R>* sv, bitCapInt maxI, bitCapInt tPow, RampArgs a, R phase0)
+{
+    const R s = (R)0.70710678118654752440;
+    const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
+    for (bitCapInt j = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; j < maxI; j += stride) {
+        const bitCapInt i = ((j & ~(tPow - 1u)) << 1u) | (j & (tPow - 1u));
+        uint64_t frac = (uint64_t)((i >> a.rampStart) & a.inPlaceRelMask);
+        for (int k = 0; k < a.nScattered; ++k) {
+            if (i & a.sPow[k]) frac += a.sWeight[k];
+        }
+        R sn, cs;
+        devSinCos<R>((R)a.scale * (R)frac + phase0, &sn, &cs);
+        const cplx<R> f{ cs, sn };
+        cplx<R> x = sv[i];
+        cplx<R> y = sv[i | tPow];
+        if (PRE) y = f * y;
+        cplx<R> o0 = s * (x + y);
+        cplx<R> o1 = s * (x - y);
+        if (!PRE) o1 = f * o1;
+        sv[i] = o0;
+        sv[i | tPow] = o1;
+    }
+}
+
+template <bool PRE>
+__global__ void k_qft_col_gen_v(
+    cplx<float>* sv, bitCapInt maxI, bitCapInt tPow, RampArgs a, float phase0)
+{
+    float4* sv4 = reinterpret_cast<float4*>(sv);
+    const float s = 0.70710678f;
+    const bitCapInt half = maxI >> 1u;
+    const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
+    for (bitCapInt k = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; k < half; k += stride) {
+        const bitCapInt j = 2u * k;
+        const bitCapInt i = ((j & ~(tPow - 1u)) << 1u) | (j & (tPow - 1u));
+        const bitCapInt lo4 = i >> 1u;
+        const bitCapInt hi4 = (i | tPow) >> 1u;
+        float4 vlo = sv4[lo4];
+        float4 vhi = sv4[hi4];
+        uint64_t fr0 = (uint64_t)((i >> a.rampStart) & a.inPlaceRelMask);
+        uint64_t fr1 = (uint64_t)(((i + 1u) >> a.rampStart) & a.inPlaceRelMask);
+        for (int t = 0; t < a.nScattered; ++t) {
+            if (i & a.sPow[t]) fr0 += a.sWeight[t];
+            if ((i + 1u) & a.sPow[t]) fr1 += a.sWeight[t];
+        }
+        float s0, c0, s1, c1;
+        __sincosf((float)a.scale * (float)fr0 + phase0, &s0, &c0);
+        __sincosf((float)a.scale * (float)fr1 + phase0, &s1, &c1);
+        const cplx<float> f0{ c0, s0 }, f1{ c1, s1 };
+        cplx<float> x0{ vlo.x, vlo.y }, x1{ vlo.z, vlo.w };
+        cplx<float> y0{ vhi.x, vhi.y }, y1{ vhi.z, vhi.w };
+        if (PRE) {
+            y0 = f0 * y0;
+            y1 = f1 * y1;
+        }
+        cplx<float> a0 = s * (x0 + y0), a1 = s * (x1 + y1);
+        cplx<float> b0 = s * (x0 - y0), b1 = s * (x1 - y1);
+        if (!PRE) {
+            b0 = f0 * b0;
+            b1 = f1 * b1;
+        }
+        sv4[lo4] = make_float4(a0.re, a0.im, a1.re, a1.im);
+        sv4[hi4] = make_float4(b0.re, b0.im, b1.re, b1.im);
+    }
+}
+
+template <typename R>
+void launchQftColumnGeneral(cplx<R>* sv, bitCapInt maxQPower, bitCapInt tPow, const RampArgs& a,
+    double phase0, bool pre, hipStream_t stream)
+{
+    const bitCapInt maxI = maxQPower >> 1u;
+    if constexpr (std::is_same_v<R, float>) {
+        if (tPow >= 2u && (maxI & 1u) == 0u) {
+            if (pre) {
+                hipLaunchKernelGGL((k_qft_col_gen_v<true>), dim3(gridFor(maxI >> 1u)),
+                    dim3(QA_BLOCK), 0, stream, sv, maxI, tPow, a, (float)phase0);
+            } else {
+                hipLaunchKernelGGL((k_qft_col_gen_v<false>), dim3(gridFor(maxI >> 1u)),
+                    dim3(QA_BLOCK), 0, stream, sv, maxI, tPow, a, (float)phase0);
+            }
+            return;
+        }
+    }
+    if (pre) {
+        hipLaunchKernelGGL((k_qft_col_gen<R, true>), dim3(gridFor(maxI)), dim3(QA_BLOCK), 0,
+            stream, sv, maxI, tPow, a, (R)phase0);
+    } else {
+        hipLaunchKernelGGL((k_qft_col_gen<R, false>), dim3(gridFor(maxI)), dim3(QA_BLOCK), 0,
+            stream, sv, maxI, tPow, a, (R)phase0);
+    }
+}
+
 template <typename R>
 void launchQftColumn(cplx<R>* sv, bitCapInt maxQPower, bitLenInt rampStart, bitLenInt col,
     bitCapInt tPow, int sign, bool pre, hipStream_t stream)
@@ -1384,7 +1481,9 @@ void launchPartProbs(const cplx<R>* sv, bitCapInt maxQPower, bitLenInt start, bi
     template void launchPhaseRampGeneral<R>(cplx<R>*, bitCapInt, const RampArgs&, hipStream_t);    \
     template void launchQftColumn<R>(                                                               \
         cplx<R>*, bitCapInt, bitLenInt, bitLenInt, bitCapInt, int, bool, hipStream_t);              \
-    template void launchMtrx1qBatch<R>(cplx<R>*, const Batch1qArgs<R>&, hipStream_t);
+    template void launchMtrx1qBatch<R>(cplx<R>*, const Batch1qArgs<R>&, hipStream_t);                               \
+    template void launchQftColumnGeneral<R>(                                                        \
+        cplx<R>*, bitCapInt, bitCapInt, const RampArgs&, double, bool, hipStream_t);
 
 QA_INSTANTIATE(float)
 QA_INSTANTIATE(double)

@@ -177,6 +177,13 @@ struct RampArgs {
 template <typename R>
 void launchPhaseRampGeneral(cplx<R>* sv, bitCapInt maxQPower, const RampArgs& a, hipStream_t stream);
 
+// generalized fused QFT column: H on tPow + the (possibly relocated-bit)
+// ramp of RampArgs in ONE pass; phase0 is a constant phase folded onto the
+// target=1 side (distributed pager meta-page scalar)
+template <typename R>
+void launchQftColumnGeneral(cplx<R>* sv, bitCapInt maxQPower, bitCapInt tPow, const RampArgs& a,
+    double phase0, bool pre, hipStream_t stream);
+
 // batched independent single-qubit gates: k distinct-target 2x2s applied in
 // ONE full-state pass (2^k-amplitude orbits in registers). The memory-bound
 // fusion win: k passes -> 1. fp32 supports k in [2,5], fp64 [2,4].

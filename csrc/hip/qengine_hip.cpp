@@ -503,6 +503,27 @@ void QEngineHIP<R>::PhaseRamp(R scale, bitLenInt rampStart, bitLenInt rampBits, 
 }
 
 template <typename R>
+void QEngineHIP<R>::QftColumnGeneral(bitLenInt target, double scale, bitLenInt rampStart,
+    bitCapInt inPlaceRelMask, const std::vector<bitCapInt>& sPows,
+    const std::vector<uint64_t>& sWeights, double phase0, bool pre)
+{
+    if (sPows.size() > 8u) throw QrackError("QftColumnGeneral: more than 8 relocated bits");
+    RampArgs a{};
+    a.rampStart = rampStart;
+    a.inPlaceRelMask = inPlaceRelMask;
+    a.nScattered = (int)sPows.size();
+    for (size_t k = 0; k < sPows.size(); ++k) {
+        a.sPow[k] = sPows[k];
+        a.sWeight[k] = sWeights[k];
+    }
+    a.condPow = 0;
+    a.scale = scale;
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    HipProfScope prof("qft_column", stream);
+    launchQftColumnGeneral<R>(dState, maxQPower, pow2(target), a, phase0, pre, stream);
+}
+
+template <typename R>
 void QEngineHIP<R>::PhaseRampGeneral(R scale, bitLenInt rampStart, bitCapInt inPlaceRelMask,
     const std::vector<bitCapInt>& sPows, const std::vector<uint64_t>& sWeights, bitCapInt condPower)
 {

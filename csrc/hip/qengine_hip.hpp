@@ -142,6 +142,9 @@ public:
     void PhaseRampGeneral(R scale, bitLenInt rampStart, bitCapInt inPlaceRelMask,
         const std::vector<bitCapInt>& sPows, const std::vector<uint64_t>& sWeights,
         bitCapInt condPower) override;
+    void QftColumnGeneral(bitLenInt target, double scale, bitLenInt rampStart,
+        bitCapInt inPlaceRelMask, const std::vector<bitCapInt>& sPows,
+        const std::vector<uint64_t>& sWeights, double phase0, bool pre) override;
 
     // ---- probability / measurement ----
     R Prob(bitLenInt q) override;

@@ -214,6 +214,30 @@ public:
         }
     }
 
+    // generalized fused QFT column (distributed pager lazy maps): H on
+    // `target` + the relocated-bit ramp e^{i(scale*frac + phase0)} applied on
+    // the target=1 side — ONE state pass on engines. pre=true applies the
+    // ramp before H (IQFT order). Default lowering: H then general ramp.
+    virtual void QftColumnGeneral(bitLenInt target, double scale, bitLenInt rampStart,
+        bitCapInt inPlaceRelMask, const std::vector<bitCapInt>& sPows,
+        const std::vector<uint64_t>& sWeights, double phase0, bool pre)
+    {
+        const bitCapInt cond = pow2(target);
+        auto ramp = [&]() {
+            PhaseRampGeneral((R)scale, rampStart, inPlaceRelMask, sPows, sWeights, cond);
+            if (phase0 != 0.0) {
+                this->Phase(cplx<R>(1, 0), polar<R>(1, (R)phase0), target);
+            }
+        };
+        if (pre) {
+            ramp();
+            this->H(target);
+        } else {
+            this->H(target);
+            ramp();
+        }
+    }
+
     // ---- measurement --------------------------------------------------------
     bool ForceM(bitLenInt qubit, bool result, bool doForce = true, bool doApply = true) override;
 

@@ -338,6 +338,40 @@ void QEngineCPU<R>::PhaseRampGeneral(R scale, bitLenInt rampStart, bitCapInt inP
     }
 }
 
+template <typename R>
+void QEngineCPU<R>::QftColumnGeneral(bitLenInt target, double scale, bitLenInt rampStart,
+    bitCapInt inPlaceRelMask, const std::vector<bitCapInt>& sPows,
+    const std::vector<uint64_t>& sWeights, double phase0, bool pre)
+{
+    const bitCapInt tPow = pow2(target);
+    const R s = SQRT1_2_R<R>;
+    const int nSc = (int)sPows.size();
+    bitCapInt sp[8] = {};
+    uint64_t sw[8] = {};
+    if (nSc > 8) throw QrackError("QftColumnGeneral: more than 8 relocated bits");
+    for (int k = 0; k < nSc; ++k) {
+        sp[k] = sPows[k];
+        sw[k] = sWeights[k];
+    }
+    cplx<R>* sv = stateVec.data();
+    const R sc = (R)scale, p0 = (R)phase0;
+    this->par_for_skip(maxQPower >> 1u, tPow, [=](const bitCapInt& i, unsigned) {
+        uint64_t frac = (uint64_t)((i >> rampStart) & inPlaceRelMask);
+        for (int k = 0; k < nSc; ++k) {
+            if (i & sp[k]) frac += sw[k];
+        }
+        const cplx<R> f = polar<R>(1, sc * (R)frac + p0);
+        cplx<R> x = sv[i];
+        cplx<R> y = sv[i | tPow];
+        if (pre) y = f * y;
+        cplx<R> o0 = s * (x + y);
+        cplx<R> o1 = s * (x - y);
+        if (!pre) o1 = f * o1;
+        sv[i] = o0;
+        sv[i | tPow] = o1;
+    });
+}
+
 template <typename R> void QEngineCPU<R>::QftColumn(bitLenInt start, bitLenInt col, int sign, bool pre)
 {
     // fully fused column: H on bit (start+col) + the column phase ladder in

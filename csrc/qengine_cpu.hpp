@@ -60,6 +60,9 @@ public:
         const std::vector<bitLenInt>& controls, bitLenInt target, const cplx<R>* mtrxs) override;
     void ROL(bitLenInt shift, bitLenInt start, bitLenInt length) override;
     void Mtrx1qBatch(const std::vector<bitLenInt>& targets, const std::vector<cplx<R>>& mtrxs) override;
+    void QftColumnGeneral(bitLenInt target, double scale, bitLenInt rampStart,
+        bitCapInt inPlaceRelMask, const std::vector<bitCapInt>& sPows,
+        const std::vector<uint64_t>& sWeights, double phase0, bool pre) override;
     void QFT(bitLenInt start, bitLenInt length, bool trySeparate = false) override;
     void IQFT(bitLenInt start, bitLenInt length, bool trySeparate = false) override;
     void PhaseRamp(R scale, bitLenInt rampStart, bitLenInt rampBits, bitCapInt condPower) override;

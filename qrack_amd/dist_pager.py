@@ -399,17 +399,83 @@ class DistQPager:
             else:
                 self.q.global_phase(f)
 
+    def _realize_local(self, t_logical):
+        """Make the logical qubit's slot local: meta slots cost ONE half-page
+        exchange (which itself realizes the swap meta<->local-top)."""
+        s = self.slot_of[t_logical]
+        if s < self.qpp:
+            return s
+        tb = s - self.qpp
+        partner = self._rank_of_page(self.my_page ^ (1 << tb))
+        i_am_low = ((self.my_page >> tb) & 1) == 0
+        self._shuffle(partner, i_am_low)
+        self._swap_slots(self.qpp - 1, s)
+        return self.qpp - 1
+
+    def _ramp_parts(self, start, i):
+        """Decompose the column-i ramp under the lazy map into the engine's
+        in-place mask, scattered (pow, weight) terms, and the per-page meta
+        weight (same split as _column_ramp)."""
+        in_place = 0
+        scattered = []
+        meta_weight = 0
+        for j in range(start, start + i):
+            s = self.slot_of[j]
+            w = 1 << (j - start)
+            if s < self.qpp:
+                if s == j and j < self.qpp:
+                    in_place |= 1 << (j - start)
+                else:
+                    scattered.append((1 << s, w))
+            else:
+                if (self.my_page >> (s - self.qpp)) & 1:
+                    meta_weight += w
+        ramp_start = start
+        if start >= self.qpp or in_place == 0:
+            ramp_start = 0
+            m = in_place
+            j = 0
+            while m:
+                if m & 1:
+                    scattered.append((1 << (start + j), 1 << j))
+                m >>= 1
+                j += 1
+            in_place = 0
+        return ramp_start, in_place, scattered, meta_weight
+
+    def _fused_column(self, start, i, sign, pre):
+        """One engine pass per column: realize the target locally (one
+        exchange at most), then H + the whole (relocated) phase ladder +
+        the meta scalar in a single fused kernel."""
+        t_slot = self._realize_local(start + i)
+        if i == 0:
+            s = 1 / np.sqrt(2)
+            self._dispatch([s, s, s, -s], t_slot, [], 0)
+            return
+        rs, in_place, scattered, meta_w = self._ramp_parts(start, i)
+        if len(scattered) > 8 or not hasattr(self.q, "qft_column_general"):
+            # fallback: plain H + the multi-kernel ramp path
+            s = 1 / np.sqrt(2)
+            if pre:
+                self._column_ramp(start, i, sign)
+                self._dispatch([s, s, s, -s], t_slot, [], 0)
+            else:
+                self._dispatch([s, s, s, -s], t_slot, [], 0)
+                self._column_ramp(start, i, sign)
+            return
+        scale = sign * np.pi / (1 << i)
+        pows = [p for p, _ in scattered]
+        ws = [w for _, w in scattered]
+        self.q.qft_column_general(
+            t_slot, float(scale), rs, in_place, pows, ws, float(scale * meta_w), pre)
+
     def qft(self, start, length):
         for i in range(length - 1, -1, -1):
-            self.h(start + i)
-            if i:
-                self._column_ramp(start, i, +1)
+            self._fused_column(start, i, +1, False)
 
     def iqft(self, start, length):
         for i in range(length):
-            if i:
-                self._column_ramp(start, i, -1)
-            self.h(start + i)
+            self._fused_column(start, i, -1, True)
 
     # ---- measurement ------------------------------------------------------------
 
