@@ -684,11 +684,12 @@ PYBIND11_MODULE(_qrack, m)
 
     m.def("create", &CreateStack<float>, py::arg("qubits"), py::arg("layers") = std::vector<std::string>{ "cpu" },
         py::arg("init_perm") = (bitCapInt)0, py::arg("seed") = (int64_t)-1,
-        py::arg("device_id") = (int64_t)-1, py::arg("pages_per_device") = (bitLenInt)1);
+        py::arg("device_id") = (int64_t)-1, py::arg("pages_per_device") = (bitLenInt)1,
+        py::arg("devices") = std::vector<int64_t>{});
     m.def("create_d", &CreateStack<double>, py::arg("qubits"),
         py::arg("layers") = std::vector<std::string>{ "cpu" }, py::arg("init_perm") = (bitCapInt)0,
         py::arg("seed") = (int64_t)-1, py::arg("device_id") = (int64_t)-1,
-        py::arg("pages_per_device") = (bitLenInt)1);
+        py::arg("pages_per_device") = (bitLenInt)1, py::arg("devices") = std::vector<int64_t>{});
 
     m.def("hip_device_count", &HipDeviceCount);
 #ifdef QRACK_AMD_HIP_ENGINE

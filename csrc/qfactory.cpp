@@ -44,8 +44,8 @@ int HipDeviceCount()
 // Recursive layer-stack factory: layers[0] is outermost.
 // (parity: qfactory.hpp CreateQuantumInterface / CreateArrangedLayersFull)
 template <typename R>
-EngineFactoryFn<R> LayerFactory(
-    std::vector<std::string> layers, RngPtr rng, int64_t deviceId, bitLenInt pagesPerDevice)
+EngineFactoryFn<R> LayerFactory(std::vector<std::string> layers, RngPtr rng, int64_t deviceId,
+    bitLenInt pagesPerDevice, const std::vector<int64_t>& devices = {})
 {
     if (layers.empty()) layers.push_back("cpu");
     const std::string head = layers.front();
@@ -72,19 +72,19 @@ EngineFactoryFn<R> LayerFactory(
         };
     }
     if (head == "stabilizer_hybrid") {
-        EngineFactoryFn<R> sub = LayerFactory<R>(tail, rng, deviceId, pagesPerDevice);
+        EngineFactoryFn<R> sub = LayerFactory<R>(tail, rng, deviceId, pagesPerDevice, devices);
         return [rng, sub](bitLenInt n, bitCapInt perm) -> QInterfacePtr<R> {
             return std::make_shared<QStabilizerHybrid<R>>(n, perm, rng, sub);
         };
     }
     if (head == "tensor_network") {
-        EngineFactoryFn<R> sub = LayerFactory<R>(tail, rng, deviceId, pagesPerDevice);
+        EngineFactoryFn<R> sub = LayerFactory<R>(tail, rng, deviceId, pagesPerDevice, devices);
         return [rng, sub](bitLenInt n, bitCapInt perm) -> QInterfacePtr<R> {
             return std::make_shared<QTensorNetwork<R>>(n, perm, rng, sub);
         };
     }
     if (head == "noisy") {
-        EngineFactoryFn<R> sub = LayerFactory<R>(tail, rng, deviceId, pagesPerDevice);
+        EngineFactoryFn<R> sub = LayerFactory<R>(tail, rng, deviceId, pagesPerDevice, devices);
         return [rng, sub](bitLenInt n, bitCapInt perm) -> QInterfacePtr<R> {
             return std::make_shared<QInterfaceNoisy<R>>(n, sub(n, perm), (R)-1, rng);
         };
@@ -111,17 +111,17 @@ EngineFactoryFn<R> LayerFactory(
         };
     }
     if (head == "bdt_hybrid") {
-        EngineFactoryFn<R> sub = LayerFactory<R>(tail, rng, deviceId, pagesPerDevice);
+        EngineFactoryFn<R> sub = LayerFactory<R>(tail, rng, deviceId, pagesPerDevice, devices);
         return [rng, sub](bitLenInt n, bitCapInt perm) -> QInterfacePtr<R> {
             return std::make_shared<QBdtHybridImpl<R>>(n, perm, rng, sub);
         };
     }
     if (head == "pager") {
-        EngineFactoryFn<R> sub = LayerFactory<R>(tail, rng, deviceId, pagesPerDevice);
+        EngineFactoryFn<R> sub = LayerFactory<R>(tail, rng, deviceId, pagesPerDevice, devices);
         // device list: QRACK_QPAGER_DEVICES "N.id,..." (reference syntax) or
         // the single requested device
-        std::vector<int64_t> devs;
-        if (const char* env = std::getenv("QRACK_QPAGER_DEVICES")) {
+        std::vector<int64_t> devs(devices);
+        if (const char* env = devs.empty() ? std::getenv("QRACK_QPAGER_DEVICES") : nullptr) {
             std::string spec(env);
             size_t pos = 0;
             while (pos < spec.size()) {
@@ -149,13 +149,13 @@ EngineFactoryFn<R> LayerFactory(
         };
     }
     if (head == "qunit") {
-        EngineFactoryFn<R> sub = LayerFactory<R>(tail, rng, deviceId, pagesPerDevice);
+        EngineFactoryFn<R> sub = LayerFactory<R>(tail, rng, deviceId, pagesPerDevice, devices);
         return [rng, sub](bitLenInt n, bitCapInt perm) -> QInterfacePtr<R> {
             return std::make_shared<QUnit<R>>(n, perm, rng, sub);
         };
     }
     if (head == "qunit_multi") {
-        EngineFactoryFn<R> sub = LayerFactory<R>(tail, rng, deviceId, pagesPerDevice);
+        EngineFactoryFn<R> sub = LayerFactory<R>(tail, rng, deviceId, pagesPerDevice, devices);
         return [rng, sub](bitLenInt n, bitCapInt perm) -> QInterfacePtr<R> {
             return std::make_shared<QUnitMulti<R>>(n, perm, rng, sub);
         };
@@ -165,10 +165,10 @@ EngineFactoryFn<R> LayerFactory(
 
 template <typename R>
 QInterfacePtr<R> CreateStack(bitLenInt qubits, std::vector<std::string> layers, bitCapInt initPerm,
-    int64_t seed, int64_t deviceId, bitLenInt pagesPerDevice)
+    int64_t seed, int64_t deviceId, bitLenInt pagesPerDevice, std::vector<int64_t> devices)
 {
     RngPtr rng = (seed < 0) ? std::make_shared<Rng>() : std::make_shared<Rng>((uint64_t)seed);
-    return LayerFactory<R>(layers, rng, deviceId, pagesPerDevice)(qubits, initPerm);
+    return LayerFactory<R>(layers, rng, deviceId, pagesPerDevice, devices)(qubits, initPerm);
 }
 
 #ifndef QRACK_AMD_HIP_ENGINE
@@ -183,8 +183,10 @@ template QInterfacePtr<double> MakeHipEngine<double>(bitLenInt, bitCapInt, RngPt
 #endif
 
 template QInterfacePtr<float> CreateStack<float>(
-    bitLenInt, std::vector<std::string>, bitCapInt, int64_t, int64_t, bitLenInt);
+    bitLenInt, std::vector<std::string>, bitCapInt, int64_t, int64_t, bitLenInt,
+    std::vector<int64_t>);
 template QInterfacePtr<double> CreateStack<double>(
-    bitLenInt, std::vector<std::string>, bitCapInt, int64_t, int64_t, bitLenInt);
+    bitLenInt, std::vector<std::string>, bitCapInt, int64_t, int64_t, bitLenInt,
+    std::vector<int64_t>);
 
 } // namespace qrack_amd

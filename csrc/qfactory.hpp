@@ -22,6 +22,6 @@ QInterfacePtr<R> MakeHipEngine(bitLenInt qubits, bitCapInt initPerm, RngPtr rng,
 // Implemented in qfactory.cpp; grows as layers land.
 template <typename R>
 QInterfacePtr<R> CreateStack(bitLenInt qubits, std::vector<std::string> layers, bitCapInt initPerm,
-    int64_t seed, int64_t deviceId, bitLenInt pagesPerDevice);
+    int64_t seed, int64_t deviceId, bitLenInt pagesPerDevice, std::vector<int64_t> devices = {});
 
 } // namespace qrack_amd

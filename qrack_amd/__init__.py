@@ -46,6 +46,7 @@ def create_simulator(
     seed=-1,
     device_id=-1,
     pages_per_device=1,
+    devices=None,
 ):
     """Create a simulator stack.
 
@@ -70,4 +71,5 @@ def create_simulator(
         seed=seed,
         device_id=device_id,
         pages_per_device=pages_per_device,
+        devices=list(devices) if devices else [],
     )
