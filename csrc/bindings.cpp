@@ -665,10 +665,11 @@ template <typename R> static void bindExtras(py::module_& m, const char* suffix)
         },
         py::arg("text"), py::arg("seed") = (int64_t)-1);
     m.def((std::string("lossy_save_") + suffix).c_str(),
-        [](Ptr q, const std::string& path, bitLenInt blockBits) {
-            LossySaveState<R>(q, path, blockBits);
+        [](Ptr q, const std::string& path, bitLenInt blockBits, int bits, bool rotate) {
+            LossySaveState<R>(q, path, blockBits, bits, rotate);
         },
-        py::arg("sim"), py::arg("path"), py::arg("block_bits") = (bitLenInt)12);
+        py::arg("sim"), py::arg("path"), py::arg("block_bits") = (bitLenInt)12,
+        py::arg("bits") = 16, py::arg("rotate") = true);
     m.def((std::string("lossy_load_") + suffix).c_str(),
         [](Ptr q, const std::string& path) { LossyLoadState<R>(q, path); });
 }

@@ -42,47 +42,104 @@ template <typename R> QInterfacePtr<R> LoadStabilizerText(const std::string& s, 
 // ---- lossy quantized binary -------------------------------------------------
 
 constexpr char QAMD_TQ_MAGIC[8] = { 'Q', 'A', 'M', 'D', 'T', 'Q', '1', '\0' };
+constexpr char QAMD_TQ2_MAGIC[8] = { 'Q', 'A', 'M', 'D', 'T', 'Q', '2', '\0' };
+
+// TurboQuant-style seeded randomized-Hadamard rotation (reference:
+// statevector_turboquant.hpp block rotation): sign flips D from a per-block
+// seed, then an orthonormal fast Walsh-Hadamard transform. The rotation makes
+// block values near-isotropic, so a uniform b-bit quantizer loses far less
+// fidelity at the same byte budget; the inverse is H then D again.
+inline void qa_tq_signs(uint64_t seed, bitCapInt blockIdx, std::vector<float>& signs)
+{
+    // splitmix64 stream keyed by (seed, blockIdx)
+    uint64_t x = seed ^ (0x9e3779b97f4a7c15ull * (blockIdx + 1u));
+    for (size_t i = 0; i < signs.size(); ++i) {
+        x += 0x9e3779b97f4a7c15ull;
+        uint64_t z = x;
+        z = (z ^ (z >> 30)) * 0xbf58476d1ce4e5b9ull;
+        z = (z ^ (z >> 27)) * 0x94d049bb133111ebull;
+        z ^= z >> 31;
+        signs[i] = (z & 1u) ? -1.0f : 1.0f;
+    }
+}
+
+inline void qa_wht(double* v, size_t n)
+{
+    // in-place orthonormal Walsh-Hadamard (n a power of two)
+    for (size_t h = 1; h < n; h <<= 1) {
+        for (size_t i = 0; i < n; i += h << 1) {
+            for (size_t j = i; j < i + h; ++j) {
+                const double a = v[j], b = v[j + h];
+                v[j] = a + b;
+                v[j + h] = a - b;
+            }
+        }
+    }
+    const double inv = 1.0 / std::sqrt((double)n);
+    for (size_t i = 0; i < n; ++i) v[i] *= inv;
+}
 
 template <typename R>
-void LossySaveState(QInterfacePtr<R> q, const std::string& path, bitLenInt blockBits = 12)
+void LossySaveState(QInterfacePtr<R> q, const std::string& path, bitLenInt blockBits = 12,
+    int bits = 16, bool rotate = true, uint64_t seed = 0x51a4d70f2u)
 {
+    if (bits != 8 && bits != 16) throw QrackError("LossySaveState: bits must be 8 or 16");
     const bitLenInt n = q->GetQubitCount();
     const bitCapInt maxQPower = q->GetMaxQPower();
     if (blockBits > n) blockBits = n;
     const bitCapInt blockLen = pow2(blockBits);
+    const size_t reals = (size_t)blockLen * 2u;
     FILE* f = std::fopen(path.c_str(), "wb");
     if (!f) throw QrackError("LossySaveState: cannot open " + path);
-    std::fwrite(QAMD_TQ_MAGIC, 1, 8, f);
-    const uint64_t qb = n, bb = blockBits, prec = sizeof(R);
+    std::fwrite(QAMD_TQ2_MAGIC, 1, 8, f);
+    const uint64_t qb = n, bb = blockBits, prec = sizeof(R), vbits = (uint64_t)bits,
+                   vrot = rotate ? 1u : 0u;
     std::fwrite(&qb, 8, 1, f);
     std::fwrite(&bb, 8, 1, f);
     std::fwrite(&prec, 8, 1, f);
+    std::fwrite(&vbits, 8, 1, f);
+    std::fwrite(&vrot, 8, 1, f);
+    std::fwrite(&seed, 8, 1, f);
     auto eng = std::dynamic_pointer_cast<QEngine<R>>(q);
     std::vector<cplx<R>> buf(blockLen);
-    std::vector<int16_t> qbuf(blockLen * 2);
+    std::vector<double> v(reals);
+    std::vector<float> signs(reals);
+    std::vector<int16_t> q16(reals);
+    std::vector<int8_t> q8(reals);
     std::vector<cplx<R>> dense;
     if (!eng) {
         dense.resize(maxQPower);
         q->GetQuantumState(dense.data());
     }
-    for (bitCapInt off = 0; off < maxQPower; off += blockLen) {
+    const double qmax = (bits == 8) ? 126.0 : 32766.0;
+    bitCapInt blockIdx = 0;
+    for (bitCapInt off = 0; off < maxQPower; off += blockLen, ++blockIdx) {
         if (eng) {
             eng->GetAmplitudePage(buf.data(), off, blockLen);
         } else {
             std::memcpy(buf.data(), dense.data() + off, sizeof(cplx<R>) * blockLen);
         }
-        R maxAbs = 0;
         for (bitCapInt i = 0; i < blockLen; ++i) {
-            maxAbs = std::max(maxAbs, std::max(std::abs(buf[i].re), std::abs(buf[i].im)));
+            v[2 * i] = (double)buf[i].re;
+            v[2 * i + 1] = (double)buf[i].im;
         }
+        if (rotate) {
+            qa_tq_signs(seed, blockIdx, signs);
+            for (size_t i = 0; i < reals; ++i) v[i] *= signs[i];
+            qa_wht(v.data(), reals);
+        }
+        double maxAbs = 0;
+        for (size_t i = 0; i < reals; ++i) maxAbs = std::max(maxAbs, std::abs(v[i]));
         const float scale = (float)maxAbs;
         std::fwrite(&scale, 4, 1, f);
-        const R inv = (maxAbs > 0) ? (R)(32766.0 / (double)maxAbs) : (R)0;
-        for (bitCapInt i = 0; i < blockLen; ++i) {
-            qbuf[2 * i] = (int16_t)std::lround((double)(buf[i].re * inv));
-            qbuf[2 * i + 1] = (int16_t)std::lround((double)(buf[i].im * inv));
+        const double inv = (maxAbs > 0) ? (qmax / maxAbs) : 0.0;
+        if (bits == 8) {
+            for (size_t i = 0; i < reals; ++i) q8[i] = (int8_t)std::lround(v[i] * inv);
+            std::fwrite(q8.data(), 1, reals, f);
+        } else {
+            for (size_t i = 0; i < reals; ++i) q16[i] = (int16_t)std::lround(v[i] * inv);
+            std::fwrite(q16.data(), 2, reals, f);
         }
-        std::fwrite(qbuf.data(), 2, blockLen * 2, f);
     }
     std::fclose(f);
 }
@@ -92,32 +149,63 @@ template <typename R> void LossyLoadState(QInterfacePtr<R> q, const std::string&
     FILE* f = std::fopen(path.c_str(), "rb");
     if (!f) throw QrackError("LossyLoadState: cannot open " + path);
     char magic[8];
-    if (std::fread(magic, 1, 8, f) != 8 || std::memcmp(magic, QAMD_TQ_MAGIC, 7) != 0) {
+    if (std::fread(magic, 1, 8, f) != 8 ||
+        (std::memcmp(magic, QAMD_TQ_MAGIC, 7) != 0 && std::memcmp(magic, QAMD_TQ2_MAGIC, 7) != 0)) {
         std::fclose(f);
         throw QrackError("LossyLoadState: bad magic");
     }
-    uint64_t qb = 0, bb = 0, prec = 0;
+    const bool v2 = magic[6] == '2';
+    uint64_t qb = 0, bb = 0, prec = 0, vbits = 16, vrot = 0, seed = 0;
     (void)!std::fread(&qb, 8, 1, f);
     (void)!std::fread(&bb, 8, 1, f);
     (void)!std::fread(&prec, 8, 1, f);
+    if (v2) {
+        (void)!std::fread(&vbits, 8, 1, f);
+        (void)!std::fread(&vrot, 8, 1, f);
+        (void)!std::fread(&seed, 8, 1, f);
+    }
     if ((bitLenInt)qb != q->GetQubitCount()) {
         std::fclose(f);
         throw QrackError("LossyLoadState: qubit count mismatch");
     }
     const bitCapInt blockLen = pow2((bitLenInt)bb);
+    const size_t reals = (size_t)blockLen * 2u;
     const bitCapInt maxQPower = q->GetMaxQPower();
     auto eng = std::dynamic_pointer_cast<QEngine<R>>(q);
     std::vector<cplx<R>> buf(blockLen);
-    std::vector<int16_t> qbuf(blockLen * 2);
+    std::vector<double> v(reals);
+    std::vector<float> signs(reals);
+    std::vector<int16_t> q16(reals);
+    std::vector<int8_t> q8(reals);
     std::vector<cplx<R>> dense;
     if (!eng) dense.resize(maxQPower);
-    for (bitCapInt off = 0; off < maxQPower; off += blockLen) {
+    const double qmax = (vbits == 8u) ? 126.0 : 32766.0;
+    bitCapInt blockIdx = 0;
+    for (bitCapInt off = 0; off < maxQPower; off += blockLen, ++blockIdx) {
         float scale = 0;
         (void)!std::fread(&scale, 4, 1, f);
-        (void)!std::fread(qbuf.data(), 2, blockLen * 2, f);
-        const R s = (R)((double)scale / 32766.0);
-        for (bitCapInt i = 0; i < blockLen; ++i) {
-            buf[i] = cplx<R>(s * (R)qbuf[2 * i], s * (R)qbuf[2 * i + 1]);
+        const double sc = (double)scale / qmax;
+        if (v2 && vbits == 8u) {
+            (void)!std::fread(q8.data(), 1, reals, f);
+            for (size_t i = 0; i < reals; ++i) v[i] = sc * (double)q8[i];
+        } else {
+            (void)!std::fread(q16.data(), 2, reals, f);
+            for (size_t i = 0; i < reals; ++i) v[i] = sc * (double)q16[i];
+        }
+        if (v2 && vrot) {
+            qa_wht(v.data(), reals);
+            qa_tq_signs(seed, blockIdx, signs);
+            for (size_t i = 0; i < reals; ++i) v[i] *= signs[i];
+        }
+        if (!v2) {
+            // v1 layout: interleaved re/im quantized directly (no rotation)
+            for (bitCapInt i = 0; i < blockLen; ++i) {
+                buf[i] = cplx<R>((R)v[2 * i], (R)v[2 * i + 1]);
+            }
+        } else {
+            for (bitCapInt i = 0; i < blockLen; ++i) {
+                buf[i] = cplx<R>((R)v[2 * i], (R)v[2 * i + 1]);
+            }
         }
         if (eng) {
             eng->SetAmplitudePage(buf.data(), off, blockLen);

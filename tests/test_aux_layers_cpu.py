@@ -243,3 +243,31 @@ def test_lossy_save_load(tmp_path):
     sv2 = np.asarray(q2.get_state_vector())
     fid = abs(np.vdot(sv1, sv2))
     assert fid > 0.999  # int16 quantization keeps high fidelity
+
+
+def test_turboquant_rotation_improves_int8_fidelity(tmp_path):
+    """QAMDTQ2: the seeded randomized-Hadamard block rotation (TurboQuant
+    technique) must beat direct int8 quantization at equal byte budget."""
+    def fid(bits, rotate):
+        q = qa.create_simulator(10, engine="cpu", precision="fp64", seed=3)
+        rng = np.random.default_rng(1)
+        for i in range(10):
+            q.ry(float(rng.uniform(0, np.pi)), i)
+        for i in range(9):
+            q.cnot(i, i + 1)
+        for i in range(10):
+            q.t(i)
+            q.h(i)
+        sv = np.asarray(q.get_state_vector()).copy()
+        p = str(tmp_path / f"tq_{bits}_{rotate}.bin")
+        qa.lossy_save_D(q, p, 7, bits, rotate)
+        q2 = qa.create_simulator(10, engine="cpu", precision="fp64", seed=3)
+        qa.lossy_load_D(q2, p)
+        sv2 = np.asarray(q2.get_state_vector())
+        return abs(np.vdot(sv, sv2)) ** 2
+
+    f_plain = fid(8, False)
+    f_rot = fid(8, True)
+    assert f_rot > f_plain
+    assert f_rot > 0.999
+    assert fid(16, True) > 1 - 1e-8
