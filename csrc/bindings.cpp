@@ -298,6 +298,32 @@ template <typename R> static void bindQInterface(py::module_& m, const char* nam
             py::arg("do_apply") = true)
         .def("m", [](QI& q, bitLenInt t) { return q.M(t); })
         .def("m_all", [](QI& q) { return q.MAll(); })
+        .def("m_all_big",
+            [](QI& q) {
+                // >64-qubit terminal measurement: per-qubit collapse composed
+                // into an arbitrary-precision Python int (the uint64 packed
+                // MAll saturates past 63 qubits)
+                py::int_ out(0);
+                const py::int_ one(1);
+                for (bitLenInt i = 0; i < q.GetQubitCount(); ++i) {
+                    if (q.M(i)) {
+                        out = py::int_(out | (one << py::int_(i)));
+                    }
+                }
+                return out;
+            })
+        .def("sample_clone_big",
+            [](Ptr q) {
+                QInterfacePtr<R> c = q->Clone();
+                py::int_ out(0);
+                const py::int_ one(1);
+                for (bitLenInt i = 0; i < c->GetQubitCount(); ++i) {
+                    if (c->M(i)) {
+                        out = py::int_(out | (one << py::int_(i)));
+                    }
+                }
+                return out;
+            })
         .def("m_reg", [](QI& q, bitLenInt s, bitLenInt l) { return q.MReg(s, l); })
         .def("force_m_reg", &QI::ForceMReg, py::arg("start"), py::arg("length"), py::arg("result"),
             py::arg("do_force") = true, py::arg("do_apply") = true)

@@ -81,3 +81,24 @@ def test_ncrp_100q_near_clifford_tableau():
     assert 0.0 < f < 1.0
     res = q.multi_shot_measure_mask([1 << i for i in range(8)], 64)
     assert sum(res.values()) == 64
+
+
+def test_m_all_big_120q():
+    import qrack_amd as qa
+    n = 120
+    q = qa.create_simulator(n, layers=["qunit", "stabilizer"], seed=5)
+    for i in range(0, n, 2):
+        q.x(i)
+    r = q.m_all_big()
+    expect = sum(1 << i for i in range(0, n, 2))
+    assert r == expect
+    q2 = qa.create_simulator(n, layers=["qunit", "stabilizer"], seed=6)
+    q2.h(0)
+    for i in range(n - 1):
+        q2.cnot(i, i + 1)  # 120-qubit GHZ
+    s = q2.sample_clone_big()
+    assert s in (0, (1 << n) - 1)
+    # sampling a clone leaves the superposition intact
+    assert abs(q2.prob(n - 1) - 0.5) < 1e-6
+    r2 = q2.m_all_big()
+    assert r2 in (0, (1 << n) - 1)
