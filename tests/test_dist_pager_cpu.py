@@ -243,3 +243,43 @@ def test_force_m_meta_world2():
 
 def test_swap_meta_world2():
     run_distributed("_body_swap_meta", world=2, qubits=4, port_off=10)
+
+
+def _body_qft_after_swaps(pager, rank):
+    # scramble the lazy map with logical swaps, then full QFT: exercises the
+    # fused column kernel's scattered-bit ramps under a permuted map
+    n = pager.num_qubits
+    x = 3
+    pager.set_permutation(x)
+    pager.swap(0, n - 1)
+    pager.swap(1, n - 2)
+    pager.swap(0, n - 1)  # permutation is now a single (1, n-2) swap
+    # logical value after swaps: bit1 <-> bit(n-2) of x=3 -> 1 | (1 << (n-2))
+    xl = 1 | (1 << (n - 2))
+    pager.qft(0, n)
+    sv = pager.get_state_vector().astype(np.complex128)
+    N = 1 << n
+    k = np.arange(N)
+    expected = np.exp(2j * np.pi * xl * k / N) / np.sqrt(N)
+    rev = np.array([int(format(i, f"0{n}b")[::-1], 2) for i in range(N)])
+    inner = np.vdot(expected, sv[rev])
+    assert abs(abs(inner) - 1.0) < 1e-4
+
+
+def _body_iqft_roundtrip_swapped(pager, rank):
+    n = pager.num_qubits
+    pager.set_permutation(9 % (1 << n))
+    pager.swap(0, n - 1)
+    pager.qft(0, n)
+    pager.iqft(0, n)
+    pager.swap(0, n - 1)
+    sv = pager.get_state_vector()
+    assert abs(abs(sv[9 % (1 << n)]) - 1.0) < 1e-4
+
+
+def test_qft_after_swaps_world4():
+    run_distributed("_body_qft_after_swaps", world=4, qubits=6, seed=11, port_off=10)
+
+
+def test_iqft_roundtrip_swapped_world2():
+    run_distributed("_body_iqft_roundtrip_swapped", world=2, qubits=5, seed=12, port_off=11)
