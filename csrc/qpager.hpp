@@ -65,6 +65,30 @@ public:
 
     // ---- gates ----
     void Mtrx(const cplx<R>* m, bitLenInt target) override;
+    // intra-page targets batch into one fused pass per page; meta targets
+    // fall back to the per-gate page tricks
+    void Mtrx1qBatch(
+        const std::vector<bitLenInt>& targets, const std::vector<cplx<R>>& mtrxs) override
+    {
+        if (mtrxs.size() != 4u * targets.size())
+            throw QrackError("Mtrx1qBatch: need 4 entries per target");
+        std::vector<bitLenInt> intra;
+        std::vector<cplx<R>> intraM;
+        for (size_t i = 0; i < targets.size(); ++i) {
+            if (targets[i] < qpp) {
+                intra.push_back(targets[i]);
+                intraM.insert(intraM.end(), &mtrxs[4u * i], &mtrxs[4u * i] + 4);
+            }
+        }
+        if (intra.size() > 1u) {
+            for (auto& pg : qPages) pg->Mtrx1qBatch(intra, intraM);
+        } else if (intra.size() == 1u) {
+            for (auto& pg : qPages) pg->Mtrx(intraM.data(), intra[0]);
+        }
+        for (size_t i = 0; i < targets.size(); ++i) {
+            if (targets[i] >= qpp) Mtrx(&mtrxs[4u * i], targets[i]);
+        }
+    }
     void UCMtrx(const std::vector<bitLenInt>& controls, const cplx<R>* m, bitLenInt target,
         bitCapInt controlPerm) override;
     void UniformlyControlledSingleBit(

@@ -84,6 +84,15 @@ public:
 
     // ---- gates ----
     void Mtrx(const cplx<R>* mtrx, bitLenInt target) override;
+    void Mtrx1qBatch(
+        const std::vector<bitLenInt>& targets, const std::vector<cplx<R>>& mtrxs) override
+    {
+        if (engine) {
+            engine->Mtrx1qBatch(targets, mtrxs);
+            return;
+        }
+        QInterface<R>::Mtrx1qBatch(targets, mtrxs); // per-gate through shards
+    }
     void Phase(cplx<R> topLeft, cplx<R> bottomRight, bitLenInt target) override;
     void Invert(cplx<R> topRight, cplx<R> bottomLeft, bitLenInt target) override;
     void MCMtrx(const std::vector<bitLenInt>& controls, const cplx<R>* mtrx, bitLenInt target) override;

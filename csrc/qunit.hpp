@@ -138,6 +138,30 @@ public:
 
     // ---- gates ----
     void Mtrx(const cplx<R>* mtrx, bitLenInt target) override;
+    // group the batch by unit and forward fused sub-batches: a layer of 1q
+    // gates on a merged unit costs ceil(k/4) passes instead of k
+    void Mtrx1qBatch(
+        const std::vector<bitLenInt>& targets, const std::vector<cplx<R>>& mtrxs) override
+    {
+        if (mtrxs.size() != 4u * targets.size())
+            throw QrackError("Mtrx1qBatch: need 4 entries per target");
+        std::map<QInterface<R>*, std::pair<std::vector<bitLenInt>, std::vector<cplx<R>>>> groups;
+        std::map<QInterface<R>*, QInterfacePtr<R>> keep;
+        for (size_t i = 0; i < targets.size(); ++i) {
+            Shard& s = shards[targets[i]];
+            auto& g = groups[s.unit.get()];
+            keep[s.unit.get()] = s.unit;
+            g.first.push_back(s.mapped);
+            g.second.insert(g.second.end(), &mtrxs[4u * i], &mtrxs[4u * i] + 4);
+        }
+        for (auto& kv : groups) {
+            if (kv.second.first.size() == 1u) {
+                kv.first->Mtrx(kv.second.second.data(), kv.second.first[0]);
+            } else {
+                kv.first->Mtrx1qBatch(kv.second.first, kv.second.second);
+            }
+        }
+    }
     void Phase(cplx<R> topLeft, cplx<R> bottomRight, bitLenInt target) override;
     void Invert(cplx<R> topRight, cplx<R> bottomLeft, bitLenInt target) override;
     void MCMtrx(const std::vector<bitLenInt>& controls, const cplx<R>* mtrx, bitLenInt target) override;

@@ -57,3 +57,29 @@ def test_mtrx_1q_batch_on_layered_stack():
     qu.mtrx_1q_batch(targets, flat)
     cp.mtrx_1q_batch(targets, flat)
     assert_states_close(qu.get_state_vector(), cp.get_state_vector(), 1e-5)
+
+
+@pytest.mark.parametrize("layers", [
+    ["qunit", "cpu"],
+    ["qunit", "stabilizer_hybrid", "cpu"],
+    ["pager", "cpu"],
+    ["stabilizer_hybrid", "cpu"],
+])
+def test_batch_routes_through_stacks(layers):
+    # entangle first so QUnit holds merged units, then batch a full 1q layer
+    n = 6
+    rng = np.random.default_rng(21)
+    qb = qa.create_simulator(n, layers=layers, seed=4, pages_per_device=2)
+    qs = qa.create_simulator(n, engine="cpu", seed=4)
+    for s in (qb, qs):
+        for i in range(n):
+            s.ry(0.4 + 0.1 * i, i)
+        for i in range(0, n - 1, 2):
+            s.cnot(i, i + 1)
+    targets = list(range(n))
+    ms = [_rand_u2(rng) for _ in targets]
+    flat = [complex(x) for m in ms for x in m]
+    qb.mtrx_1q_batch(targets, flat)
+    for t, m in zip(targets, ms):
+        qs.mtrx([complex(x) for x in m], t)
+    assert_states_close(qb.get_state_vector(), qs.get_state_vector(), 1e-4)
