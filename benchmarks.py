@@ -63,6 +63,14 @@ def _apply_1q_layer(q, targets, mats):
             q.mtrx([complex(x) for x in m], t)
 
 
+def _apply_cnot_layer(q, controls, targets):
+    if hasattr(q, "cnot_batch"):
+        q.cnot_batch(list(controls), list(targets))
+    else:
+        for c, t in zip(controls, targets):
+            q.cnot(c, t)
+
+
 def run_random_circuit(q, n, rng, depth):
     q.set_permutation(0)
     d = depth or n
@@ -70,8 +78,7 @@ def run_random_circuit(q, n, rng, depth):
     for _ in range(d):
         mats = [_M1Q[names[rng.integers(3)]] for _ in range(n)]
         _apply_1q_layer(q, range(n), mats)
-        for i in range(0, n - 1, 2):
-            q.cnot(i, i + 1)
+        _apply_cnot_layer(q, range(0, n - 1, 2), range(1, n, 2))
 
 
 def run_supremacy(q, n, rng, depth):
@@ -103,8 +110,7 @@ def run_qv(q, n, rng, depth):
                 targets.append(t)
                 mats.append(_u_mtrx(float(th), float(ph), float(lm)))
         _apply_1q_layer(q, targets, mats)
-        for a, b in pairs:
-            q.cnot(a, b)
+        _apply_cnot_layer(q, [a for a, _ in pairs], [b for _, b in pairs])
 
 
 WORKLOADS = {

@@ -68,6 +68,10 @@ template <typename R> static void bindQInterface(py::module_& m, const char* nam
                     from_std<R>(m[3]) };
                 q.Mtrx(mm, t);
             })
+        .def("cnot_batch",
+            [](QI& q, std::vector<bitLenInt> controls, std::vector<bitLenInt> targets) {
+                q.CnotBatch(controls, targets);
+            })
         .def("mtrx_1q_batch",
             [](QI& q, std::vector<bitLenInt> targets, std::vector<C> ms) {
                 if (ms.size() != 4u * targets.size())

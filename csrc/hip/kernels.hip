@@ -1259,6 +1259,63 @@ template <int K> __global__ void k_mtrx_batch_v(cplx<float>* sv, Batch1qArgs<flo
     }
 }
 
+// ---- batched disjoint CNOTs: one permutation pass per layer ----------------
+
+template <typename R> __global__ void k_cnot_batch(cplx<R>* sv, CnotBatchArgs a)
+{
+    const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
+    for (bitCapInt i = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; i < a.maxI; i += stride) {
+        bitCapInt xm = 0;
+        for (int j = 0; j < a.k; ++j) {
+            if (i & a.cPow[j]) xm |= a.tPow[j];
+        }
+        const bitCapInt p = i ^ xm;
+        if (p <= i) continue; // partner handles the swap (or xm == 0)
+        const cplx<R> t = sv[i];
+        sv[i] = sv[p];
+        sv[p] = t;
+    }
+}
+
+// fp32 float4 variant: adjacent amplitude pairs share xm when neither a
+// control nor a target sits at bit 0
+__global__ void k_cnot_batch_v(cplx<float>* sv, CnotBatchArgs a)
+{
+    float4* sv4 = reinterpret_cast<float4*>(sv);
+    const bitCapInt half = a.maxI >> 1u;
+    const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
+    for (bitCapInt q = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; q < half; q += stride) {
+        const bitCapInt i = q << 1u;
+        bitCapInt xm = 0;
+        for (int j = 0; j < a.k; ++j) {
+            if (i & a.cPow[j]) xm |= a.tPow[j];
+        }
+        const bitCapInt p = i ^ xm;
+        if (p <= i) continue;
+        const bitCapInt qi = i >> 1u, qp = p >> 1u;
+        const float4 t = sv4[qi];
+        sv4[qi] = sv4[qp];
+        sv4[qp] = t;
+    }
+}
+
+template <typename R>
+void launchCnotBatch(cplx<R>* sv, const CnotBatchArgs& a, hipStream_t stream)
+{
+    if constexpr (std::is_same_v<R, float>) {
+        bool low = false;
+        for (int j = 0; j < a.k; ++j) {
+            if (a.cPow[j] < 2u || a.tPow[j] < 2u) low = true;
+        }
+        if (!low && (a.maxI & 1u) == 0u) {
+            hipLaunchKernelGGL((k_cnot_batch_v), dim3(gridFor(a.maxI >> 1u)), dim3(QA_BLOCK), 0,
+                stream, sv, a);
+            return;
+        }
+    }
+    hipLaunchKernelGGL((k_cnot_batch<R>), dim3(gridFor(a.maxI)), dim3(QA_BLOCK), 0, stream, sv, a);
+}
+
 template <typename R>
 void launchMtrx1qBatch(cplx<R>* sv, const Batch1qArgs<R>& a, hipStream_t stream)
 {
@@ -1482,6 +1539,7 @@ void launchPartProbs(const cplx<R>* sv, bitCapInt maxQPower, bitLenInt start, bi
     template void launchQftColumn<R>(                                                               \
         cplx<R>*, bitCapInt, bitLenInt, bitLenInt, bitCapInt, int, bool, hipStream_t);              \
     template void launchMtrx1qBatch<R>(cplx<R>*, const Batch1qArgs<R>&, hipStream_t);                               \
+    template void launchCnotBatch<R>(cplx<R>*, const CnotBatchArgs&, hipStream_t);                               \
     template void launchQftColumnGeneral<R>(                                                        \
         cplx<R>*, bitCapInt, bitCapInt, const RampArgs&, double, bool, hipStream_t);
 

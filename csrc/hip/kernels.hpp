@@ -199,6 +199,22 @@ template <typename R> struct Batch1qArgs {
 template <typename R>
 void launchMtrx1qBatch(cplx<R>* sv, const Batch1qArgs<R>& a, hipStream_t stream);
 
+// batched disjoint CNOTs: a whole layer of k control/target pairs (no qubit
+// repeated) applied as ONE in-place permutation pass — amp[i] swaps with
+// amp[i ^ xm(i)] where xm(i) XORs tPow[j] for every set control bit. One
+// full-state RMW replaces k half-state passes.
+constexpr int QA_MAX_BATCH_CNOT = 16;
+
+struct CnotBatchArgs {
+    bitCapInt cPow[QA_MAX_BATCH_CNOT];
+    bitCapInt tPow[QA_MAX_BATCH_CNOT];
+    int k;
+    bitCapInt maxI; // full state size
+};
+
+template <typename R>
+void launchCnotBatch(cplx<R>* sv, const CnotBatchArgs& a, hipStream_t stream);
+
 // fully fused QFT column (H + the column's phase ramp in one pass);
 // pre=false: QFT order (H then ramp), pre=true: IQFT order (ramp then H)
 template <typename R>

@@ -400,6 +400,37 @@ void QEngineHIP<R>::Apply2x2(bitCapInt offset1, bitCapInt offset2, const cplx<R>
     launchApply2x2<R>(dState, a, stream);
 }
 
+// batched disjoint CNOTs: one permutation pass per layer (k_cnot_batch).
+template <typename R>
+void QEngineHIP<R>::CnotBatch(
+    const std::vector<bitLenInt>& controls, const std::vector<bitLenInt>& targets)
+{
+    if (controls.size() != targets.size())
+        throw QrackError("CnotBatch: need one target per control");
+    const size_t k = controls.size();
+    std::set<bitLenInt> uniq;
+    for (size_t i = 0; i < k; ++i) {
+        uniq.insert(controls[i]);
+        uniq.insert(targets[i]);
+        if (controls[i] >= qubitCount || targets[i] >= qubitCount)
+            throw QrackError("CnotBatch: qubit out of range");
+    }
+    if (uniq.size() != 2u * k || k == 0u || k > (size_t)QA_MAX_BATCH_CNOT) {
+        QInterface<R>::CnotBatch(controls, targets);
+        return;
+    }
+    CnotBatchArgs a{};
+    for (size_t i = 0; i < k; ++i) {
+        a.cPow[i] = pow2(controls[i]);
+        a.tPow[i] = pow2(targets[i]);
+    }
+    a.k = (int)k;
+    a.maxI = maxQPower;
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    HipProfScope prof("cnot_batch", stream);
+    launchCnotBatch<R>(dState, a, stream);
+}
+
 // batched independent 1q gates: k gates in one full-state pass (k_mtrx_batch).
 // fp32 fuses up to 5 gates per pass, fp64 up to 4 (register budget).
 template <typename R>

@@ -117,6 +117,8 @@ public:
     void Apply2x2(bitCapInt offset1, bitCapInt offset2, const cplx<R>* mtrx,
         const std::vector<bitCapInt>& qPowersSorted) override;
     void Mtrx1qBatch(const std::vector<bitLenInt>& targets, const std::vector<cplx<R>>& mtrxs) override;
+    void CnotBatch(
+        const std::vector<bitLenInt>& controls, const std::vector<bitLenInt>& targets) override;
     void ApplyM(bitCapInt regMask, bitCapInt result, cplx<R> nrm) override;
     void GetAmplitudePage(cplx<R>* pagePtr, bitCapInt offset, bitCapInt length) override;
     void SetAmplitudePage(const cplx<R>* pagePtr, bitCapInt offset, bitCapInt length) override;

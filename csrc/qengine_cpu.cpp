@@ -5,6 +5,10 @@
 
 #include <set>
 
+namespace qrack_amd {
+constexpr int QA_CNOT_BATCH_MAX = 16;
+} // namespace qrack_amd
+
 #include <algorithm>
 #include <cstring>
 
@@ -112,6 +116,44 @@ template <typename R> void QEngineCPU<R>::ShuffleBuffers(QEnginePtr<R> engine)
 }
 
 // ---- gate primitives -------------------------------------------------------
+
+// batched disjoint CNOTs: the layer is an involutive permutation
+// i <-> i ^ xm(i); one parallel pass swaps each orbit pair once.
+template <typename R>
+void QEngineCPU<R>::CnotBatch(
+    const std::vector<bitLenInt>& controls, const std::vector<bitLenInt>& targets)
+{
+    if (controls.size() != targets.size())
+        throw QrackError("CnotBatch: need one target per control");
+    const size_t k = controls.size();
+    std::set<bitLenInt> uniq;
+    for (size_t i = 0; i < k; ++i) {
+        uniq.insert(controls[i]);
+        uniq.insert(targets[i]);
+        if (controls[i] >= qubitCount || targets[i] >= qubitCount)
+            throw QrackError("CnotBatch: qubit out of range");
+    }
+    if (uniq.size() != 2u * k || k == 0u || k > (size_t)QA_CNOT_BATCH_MAX) {
+        QInterface<R>::CnotBatch(controls, targets); // overlap or size: per-gate
+        return;
+    }
+    bitCapInt cp[QA_CNOT_BATCH_MAX], tp[QA_CNOT_BATCH_MAX];
+    for (size_t i = 0; i < k; ++i) {
+        cp[i] = pow2(controls[i]);
+        tp[i] = pow2(targets[i]);
+    }
+    cplx<R>* sv = stateVec.data();
+    const int kk = (int)k;
+    this->par_for(0, maxQPower, [sv, cp, tp, kk](const bitCapInt& i, unsigned) {
+        bitCapInt xm = 0;
+        for (int j = 0; j < kk; ++j) {
+            if (i & cp[j]) xm |= tp[j];
+        }
+        const bitCapInt p = i ^ xm;
+        if (p <= i) return;
+        std::swap(sv[i], sv[p]);
+    });
+}
 
 // batched independent 1q gates: one pass over 2^k-amplitude orbits held in
 // registers (the HIP engine's k_mtrx_batch, CPU flavor). Falls back to the

@@ -92,6 +92,15 @@ public:
             Mtrx(&mtrxs[4u * i], targets[i]);
         }
     }
+    // batched disjoint CNOTs: one permutation pass per layer on engines
+    // (controls/targets must not share qubits; default lowering loops CNOT)
+    virtual void CnotBatch(
+        const std::vector<bitLenInt>& controls, const std::vector<bitLenInt>& targets)
+    {
+        if (controls.size() != targets.size())
+            throw QrackError("CnotBatch: need one target per control");
+        for (size_t i = 0; i < controls.size(); ++i) CNOT(controls[i], targets[i]);
+    }
     virtual void Phase(cplx<R> topLeft, cplx<R> bottomRight, bitLenInt target);
     virtual void Invert(cplx<R> topRight, cplx<R> bottomLeft, bitLenInt target);
     virtual void MCMtrx(const std::vector<bitLenInt>& controls, const cplx<R>* mtrx, bitLenInt target);

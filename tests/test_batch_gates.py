@@ -83,3 +83,35 @@ def test_batch_routes_through_stacks(layers):
     for t, m in zip(targets, ms):
         qs.mtrx([complex(x) for x in m], t)
     assert_states_close(qb.get_state_vector(), qs.get_state_vector(), 1e-4)
+
+
+@pytest.mark.parametrize("precision", ["fp32", "fp64"])
+def test_cnot_batch_matches_sequential(precision):
+    n = 8
+    rng = np.random.default_rng(31)
+    for trial in range(3):
+        perm = rng.permutation(n)
+        controls = [int(perm[i]) for i in range(0, n, 2)]
+        targets = [int(perm[i + 1]) for i in range(0, n, 2)]
+        qb = qa.create_simulator(n, engine="cpu", precision=precision, seed=5)
+        qs = qa.create_simulator(n, engine="cpu", precision=precision, seed=5)
+        for i in range(n):
+            th = float(rng.uniform(0, np.pi))
+            qb.ry(th, i)
+            qs.ry(th, i)
+        qb.cnot_batch(controls, targets)
+        for c, t in zip(controls, targets):
+            qs.cnot(c, t)
+        assert_states_close(qb.get_state_vector(), qs.get_state_vector(), 1e-5)
+
+
+def test_cnot_batch_overlapping_falls_back():
+    # shared qubit across pairs: sequential semantics (in order)
+    q1 = qa.create_simulator(3, engine="cpu", seed=1)
+    q2 = qa.create_simulator(3, engine="cpu", seed=1)
+    q1.x(0)
+    q2.x(0)
+    q1.cnot_batch([0, 1], [1, 2])  # chain: overlapping qubit 1
+    q2.cnot(0, 1)
+    q2.cnot(1, 2)
+    assert_states_close(q1.get_state_vector(), q2.get_state_vector(), 1e-6)
