@@ -390,3 +390,20 @@ def test_mtrx_1q_batch_gpu(precision):
         for t, m in zip(targets, ms):
             qs.mtrx([complex(x) for x in m], t)
         assert float(qb.sum_sqr_diff(qs)) < 1e-5
+
+
+def test_cnot_batch_gpu():
+    n = 22
+    rng = np.random.default_rng(41)
+    for controls, targets in (([2, 8, 14], [5, 11, 19]), ([0, 3], [1, 4])):
+        qb = qa.create_simulator(n, engine="hip", seed=6)
+        qs = qa.create_simulator(n, engine="hip", seed=6)
+        for i in range(0, n, 2):
+            qb.h(i)
+            qs.h(i)
+            qb.t(i)
+            qs.t(i)
+        qb.cnot_batch(controls, targets)
+        for c, t in zip(controls, targets):
+            qs.cnot(c, t)
+        assert float(qb.sum_sqr_diff(qs)) < 1e-5
