@@ -19,7 +19,7 @@ namespace qrack_amd {
 constexpr int QA_BLOCK = 256;
 // 8192 blocks measured ~5% faster than 2048 on the 30-qubit H stream
 // (gpurun A/B, profiles/KERNELS.md); grid-stride keeps residency bounded
-constexpr int QA_MAX_BLOCKS = 8192;
+constexpr int QA_MAX_BLOCKS = QA_REDUCE_MAX_BLOCKS;
 
 static inline int maxBlocks()
 {

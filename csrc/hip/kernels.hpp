@@ -16,6 +16,8 @@
 namespace qrack_amd {
 
 constexpr int QA_MAX_SKIP_POWERS = 16;
+// grid-size cap for reduction launches; partials buffers must hold this many
+constexpr int QA_REDUCE_MAX_BLOCKS = 8192;
 
 // by-value argument block for gate kernels
 template <typename R> struct GateArgs {

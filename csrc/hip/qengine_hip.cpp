@@ -132,7 +132,7 @@ QEngineHIP<R>::QEngineHIP(bitLenInt qBitCount, bitCapInt initState, RngPtr rgp, 
     QA_HIP_CHECK(hipSetDevice(deviceId));
     QA_HIP_CHECK(hipStreamCreateWithFlags(&stream, hipStreamNonBlocking));
     dState = allocDev(maxQPower);
-    const int maxGrid = 4096;
+    const int maxGrid = QA_REDUCE_MAX_BLOCKS;
     QA_HIP_CHECK(hipMalloc(&dPartials, maxGrid * 2 * sizeof(double)));
     QA_HIP_CHECK(hipMalloc(&dIdx, maxGrid * sizeof(bitCapInt)));
     hPartials.resize(maxGrid * 2);
@@ -211,8 +211,8 @@ template <typename R> void QEngineHIP<R>::SetDevice(int64_t devId)
     QA_HIP_CHECK(hipSetDevice(deviceId));
     QA_HIP_CHECK(hipStreamCreateWithFlags(&stream, hipStreamNonBlocking));
     dState = allocDev(maxQPower);
-    QA_HIP_CHECK(hipMalloc(&dPartials, 4096 * 2 * sizeof(double)));
-    QA_HIP_CHECK(hipMalloc(&dIdx, 4096 * sizeof(bitCapInt)));
+    QA_HIP_CHECK(hipMalloc(&dPartials, QA_REDUCE_MAX_BLOCKS * 2 * sizeof(double)));
+    QA_HIP_CHECK(hipMalloc(&dIdx, QA_REDUCE_MAX_BLOCKS * sizeof(bitCapInt)));
     SetQuantumState(host.data());
 }
 
@@ -872,16 +872,17 @@ template <typename R> double QEngineHIP<R>::SumSqrDiff(QInterfacePtr<R> other)
             tmp, host.data(), sizeof(cplx<R>) * maxQPower, hipMemcpyHostToDevice, stream));
         otherBuf = tmp;
     }
-    const int grid = launchInner<R>(dState, otherBuf, maxQPower, dPartials, dPartials + 4096, stream);
+    const int grid = launchInner<R>(
+        dState, otherBuf, maxQPower, dPartials, dPartials + QA_REDUCE_MAX_BLOCKS, stream);
     QA_HIP_CHECK(hipMemcpyAsync(
         hPartials.data(), dPartials, grid * sizeof(double), hipMemcpyDeviceToHost, stream));
-    QA_HIP_CHECK(hipMemcpyAsync(hPartials.data() + 4096, dPartials + 4096, grid * sizeof(double),
-        hipMemcpyDeviceToHost, stream));
+    QA_HIP_CHECK(hipMemcpyAsync(hPartials.data() + QA_REDUCE_MAX_BLOCKS,
+        dPartials + QA_REDUCE_MAX_BLOCKS, grid * sizeof(double), hipMemcpyDeviceToHost, stream));
     Finish();
     double re = 0, im = 0;
     for (int i = 0; i < grid; ++i) {
         re += hPartials[i];
-        im += hPartials[4096 + i];
+        im += hPartials[QA_REDUCE_MAX_BLOCKS + i];
     }
     if (tmp) freeDev(tmp, maxQPower);
     const double inner = std::sqrt(re * re + im * im);
