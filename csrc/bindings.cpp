@@ -68,6 +68,12 @@ template <typename R> static void bindQInterface(py::module_& m, const char* nam
                     from_std<R>(m[3]) };
                 q.Mtrx(mm, t);
             })
+        .def("cphase_pairs",
+            [](QI& q, std::vector<bitLenInt> c, std::vector<bitLenInt> t, std::vector<double> a) {
+                q.CPhasePairs(c, t, a);
+            })
+        .def("cz_batch",
+            [](QI& q, std::vector<bitLenInt> c, std::vector<bitLenInt> t) { q.CzBatch(c, t); })
         .def("cnot_batch",
             [](QI& q, std::vector<bitLenInt> controls, std::vector<bitLenInt> targets) {
                 q.CnotBatch(controls, targets);

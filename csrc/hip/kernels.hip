@@ -1299,6 +1299,28 @@ __global__ void k_cnot_batch_v(cplx<float>* sv, CnotBatchArgs a)
     }
 }
 
+template <typename R> __global__ void k_cphase_pairs(cplx<R>* sv, CPhasePairsArgs a)
+{
+    const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
+    for (bitCapInt i = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; i < a.maxI; i += stride) {
+        double th = 0;
+        for (int j = 0; j < a.k; ++j) {
+            if ((i & a.cPow[j]) && (i & a.tPow[j])) th += a.angle[j];
+        }
+        if (th == 0.0) continue;
+        R s, c;
+        devSinCos<R>((R)th, &s, &c);
+        sv[i] = cplx<R>{ c, s } * sv[i];
+    }
+}
+
+template <typename R>
+void launchCPhasePairs(cplx<R>* sv, const CPhasePairsArgs& a, hipStream_t stream)
+{
+    hipLaunchKernelGGL(
+        (k_cphase_pairs<R>), dim3(gridFor(a.maxI)), dim3(QA_BLOCK), 0, stream, sv, a);
+}
+
 template <typename R>
 void launchCnotBatch(cplx<R>* sv, const CnotBatchArgs& a, hipStream_t stream)
 {
@@ -1539,7 +1561,8 @@ void launchPartProbs(const cplx<R>* sv, bitCapInt maxQPower, bitLenInt start, bi
     template void launchQftColumn<R>(                                                               \
         cplx<R>*, bitCapInt, bitLenInt, bitLenInt, bitCapInt, int, bool, hipStream_t);              \
     template void launchMtrx1qBatch<R>(cplx<R>*, const Batch1qArgs<R>&, hipStream_t);                               \
-    template void launchCnotBatch<R>(cplx<R>*, const CnotBatchArgs&, hipStream_t);                               \
+    template void launchCnotBatch<R>(cplx<R>*, const CnotBatchArgs&, hipStream_t);                              \
+    template void launchCPhasePairs<R>(cplx<R>*, const CPhasePairsArgs&, hipStream_t);                               \
     template void launchQftColumnGeneral<R>(                                                        \
         cplx<R>*, bitCapInt, bitCapInt, const RampArgs&, double, bool, hipStream_t);
 

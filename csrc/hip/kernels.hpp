@@ -217,6 +217,20 @@ struct CnotBatchArgs {
 template <typename R>
 void launchCnotBatch(cplx<R>* sv, const CnotBatchArgs& a, hipStream_t stream);
 
+// batched controlled-phase pairs: amp[i] *= exp(i * sum_j angle_j) over pairs
+// with both bits set — one diagonal pass applies a whole CZ/CPhase layer
+// (pairs may share qubits; diagonal ops commute).
+struct CPhasePairsArgs {
+    bitCapInt cPow[QA_MAX_BATCH_CNOT];
+    bitCapInt tPow[QA_MAX_BATCH_CNOT];
+    double angle[QA_MAX_BATCH_CNOT];
+    int k;
+    bitCapInt maxI;
+};
+
+template <typename R>
+void launchCPhasePairs(cplx<R>* sv, const CPhasePairsArgs& a, hipStream_t stream);
+
 // fully fused QFT column (H + the column's phase ramp in one pass);
 // pre=false: QFT order (H then ramp), pre=true: IQFT order (ramp then H)
 template <typename R>

@@ -400,6 +400,33 @@ void QEngineHIP<R>::Apply2x2(bitCapInt offset1, bitCapInt offset2, const cplx<R>
     launchApply2x2<R>(dState, a, stream);
 }
 
+// batched controlled-phase pairs: one diagonal pass per layer.
+template <typename R>
+void QEngineHIP<R>::CPhasePairs(const std::vector<bitLenInt>& controls,
+    const std::vector<bitLenInt>& targets, const std::vector<double>& angles)
+{
+    if (controls.size() != targets.size() || angles.size() != controls.size())
+        throw QrackError("CPhasePairs: need (control, target, angle) triples");
+    const size_t k = controls.size();
+    if (k == 0u || k > (size_t)QA_MAX_BATCH_CNOT) {
+        QInterface<R>::CPhasePairs(controls, targets, angles);
+        return;
+    }
+    CPhasePairsArgs a{};
+    for (size_t i = 0; i < k; ++i) {
+        if (controls[i] >= qubitCount || targets[i] >= qubitCount)
+            throw QrackError("CPhasePairs: qubit out of range");
+        a.cPow[i] = pow2(controls[i]);
+        a.tPow[i] = pow2(targets[i]);
+        a.angle[i] = angles[i];
+    }
+    a.k = (int)k;
+    a.maxI = maxQPower;
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    HipProfScope prof("cphase_pairs", stream);
+    launchCPhasePairs<R>(dState, a, stream);
+}
+
 // batched disjoint CNOTs: one permutation pass per layer (k_cnot_batch).
 template <typename R>
 void QEngineHIP<R>::CnotBatch(

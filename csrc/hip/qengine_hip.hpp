@@ -119,6 +119,8 @@ public:
     void Mtrx1qBatch(const std::vector<bitLenInt>& targets, const std::vector<cplx<R>>& mtrxs) override;
     void CnotBatch(
         const std::vector<bitLenInt>& controls, const std::vector<bitLenInt>& targets) override;
+    void CPhasePairs(const std::vector<bitLenInt>& controls, const std::vector<bitLenInt>& targets,
+        const std::vector<double>& angles) override;
     void ApplyM(bitCapInt regMask, bitCapInt result, cplx<R> nrm) override;
     void GetAmplitudePage(cplx<R>* pagePtr, bitCapInt offset, bitCapInt length) override;
     void SetAmplitudePage(const cplx<R>* pagePtr, bitCapInt offset, bitCapInt length) override;

@@ -117,6 +117,38 @@ template <typename R> void QEngineCPU<R>::ShuffleBuffers(QEnginePtr<R> engine)
 
 // ---- gate primitives -------------------------------------------------------
 
+// batched controlled-phase pairs: one diagonal pass for a whole layer
+template <typename R>
+void QEngineCPU<R>::CPhasePairs(const std::vector<bitLenInt>& controls,
+    const std::vector<bitLenInt>& targets, const std::vector<double>& angles)
+{
+    if (controls.size() != targets.size() || angles.size() != controls.size())
+        throw QrackError("CPhasePairs: need (control, target, angle) triples");
+    const size_t k = controls.size();
+    if (k == 0u || k > (size_t)QA_CNOT_BATCH_MAX) {
+        QInterface<R>::CPhasePairs(controls, targets, angles);
+        return;
+    }
+    bitCapInt cp[QA_CNOT_BATCH_MAX], tp[QA_CNOT_BATCH_MAX];
+    double an[QA_CNOT_BATCH_MAX];
+    for (size_t i = 0; i < k; ++i) {
+        if (controls[i] >= qubitCount || targets[i] >= qubitCount)
+            throw QrackError("CPhasePairs: qubit out of range");
+        cp[i] = pow2(controls[i]);
+        tp[i] = pow2(targets[i]);
+        an[i] = angles[i];
+    }
+    cplx<R>* sv = stateVec.data();
+    const int kk = (int)k;
+    this->par_for(0, maxQPower, [sv, cp, tp, an, kk](const bitCapInt& i, unsigned) {
+        double th = 0;
+        for (int j = 0; j < kk; ++j) {
+            if ((i & cp[j]) && (i & tp[j])) th += an[j];
+        }
+        if (th != 0.0) sv[i] = polar<R>(1, (R)th) * sv[i];
+    });
+}
+
 // batched disjoint CNOTs: the layer is an involutive permutation
 // i <-> i ^ xm(i); one parallel pass swaps each orbit pair once.
 template <typename R>

@@ -62,6 +62,8 @@ public:
     void Mtrx1qBatch(const std::vector<bitLenInt>& targets, const std::vector<cplx<R>>& mtrxs) override;
     void CnotBatch(
         const std::vector<bitLenInt>& controls, const std::vector<bitLenInt>& targets) override;
+    void CPhasePairs(const std::vector<bitLenInt>& controls, const std::vector<bitLenInt>& targets,
+        const std::vector<double>& angles) override;
     void QftColumnGeneral(bitLenInt target, double scale, bitLenInt rampStart,
         bitCapInt inPlaceRelMask, const std::vector<bitCapInt>& sPows,
         const std::vector<uint64_t>& sWeights, double phase0, bool pre) override;

@@ -101,6 +101,21 @@ public:
             throw QrackError("CnotBatch: need one target per control");
         for (size_t i = 0; i < controls.size(); ++i) CNOT(controls[i], targets[i]);
     }
+    // batched controlled-phase pairs (one diagonal pass per layer on engines;
+    // pairs may share qubits — diagonal gates commute)
+    virtual void CPhasePairs(const std::vector<bitLenInt>& controls,
+        const std::vector<bitLenInt>& targets, const std::vector<double>& angles)
+    {
+        if (controls.size() != targets.size() || angles.size() != controls.size())
+            throw QrackError("CPhasePairs: need (control, target, angle) triples");
+        for (size_t i = 0; i < controls.size(); ++i) {
+            MCPhase({ controls[i] }, cplx<R>(1, 0), polar<R>(1, (R)angles[i]), targets[i]);
+        }
+    }
+    void CzBatch(const std::vector<bitLenInt>& controls, const std::vector<bitLenInt>& targets)
+    {
+        CPhasePairs(controls, targets, std::vector<double>(controls.size(), 3.14159265358979323846));
+    }
     virtual void Phase(cplx<R> topLeft, cplx<R> bottomRight, bitLenInt target);
     virtual void Invert(cplx<R> topRight, cplx<R> bottomLeft, bitLenInt target);
     virtual void MCMtrx(const std::vector<bitLenInt>& controls, const cplx<R>* mtrx, bitLenInt target);

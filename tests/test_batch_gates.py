@@ -115,3 +115,36 @@ def test_cnot_batch_overlapping_falls_back():
     q2.cnot(0, 1)
     q2.cnot(1, 2)
     assert_states_close(q1.get_state_vector(), q2.get_state_vector(), 1e-6)
+
+
+@pytest.mark.parametrize("precision", ["fp32", "fp64"])
+def test_cphase_pairs_matches_sequential(precision):
+    n = 7
+    rng = np.random.default_rng(51)
+    controls = [0, 2, 4, 1]  # overlapping qubit 1 allowed: diagonal commutes
+    targets = [1, 3, 5, 6]
+    angles = [float(a) for a in rng.uniform(0.2, 2.8, 4)]
+    qb = qa.create_simulator(n, engine="cpu", precision=precision, seed=5)
+    qs = qa.create_simulator(n, engine="cpu", precision=precision, seed=5)
+    for i in range(n):
+        qb.h(i)
+        qs.h(i)
+    qb.cphase_pairs(controls, targets, angles)
+    for c, t, a in zip(controls, targets, angles):
+        qs.mcphase([c], 1, complex(np.exp(1j * a)), t)
+    assert_states_close(qb.get_state_vector(), qs.get_state_vector(), 1e-5)
+
+
+def test_cz_batch_graph_state():
+    # 1D cluster state: H all + CZ chain in two one-pass layers
+    n = 6
+    q1 = qa.create_simulator(n, engine="cpu", seed=2)
+    q2 = qa.create_simulator(n, engine="cpu", seed=2)
+    for i in range(n):
+        q1.h(i)
+        q2.h(i)
+    q1.cz_batch(list(range(0, n - 1, 2)), list(range(1, n, 2)))
+    q1.cz_batch(list(range(1, n - 1, 2)), list(range(2, n, 2)))
+    for i in range(n - 1):
+        q2.cz(i, i + 1)
+    assert_states_close(q1.get_state_vector(), q2.get_state_vector(), 1e-5)

@@ -407,3 +407,20 @@ def test_cnot_batch_gpu():
         for c, t in zip(controls, targets):
             qs.cnot(c, t)
         assert float(qb.sum_sqr_diff(qs)) < 1e-5
+
+
+def test_cphase_pairs_gpu():
+    n = 21
+    qb = qa.create_simulator(n, engine="hip", seed=7)
+    qs = qa.create_simulator(n, engine="hip", seed=7)
+    for i in range(n):
+        qb.h(i)
+        qs.h(i)
+    controls = [0, 5, 10, 15]
+    targets = [3, 8, 13, 18]
+    angles = [0.3, 1.1, 2.2, 0.7]
+    qb.cphase_pairs(controls, targets, angles)
+    import numpy as np
+    for c, t, a in zip(controls, targets, angles):
+        qs.mcphase([c], 1, complex(np.exp(1j * a)), t)
+    assert float(qb.sum_sqr_diff(qs)) < 1e-5
