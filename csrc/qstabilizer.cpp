@@ -252,12 +252,13 @@ template <typename R> void QStabilizer<R>::Phase(cplx<R> tl, cplx<R> br, bitLenI
 
 template <typename R> void QStabilizer<R>::Invert(cplx<R> tr, cplx<R> bl, bitLenInt t)
 {
-    // X-type: bl/tr in {1, i, -1, -i}: X, then a phase on the target
+    // [[0, tr], [bl, 0]] = X * diag(bl, tr). Apply the PHASE FIRST: Phase()
+    // throws on non-Clifford ratios before mutating anything, which keeps
+    // this gate transactional — callers (QStabilizerHybrid shard flushes)
+    // probe with throwing gates and must not see a half-applied tableau.
     if (norm(tr) <= 0 || norm(bl) <= 0) throw QrackError("QStabilizer: non-unitary invert");
+    Phase(bl, tr, t);
     XGate(t);
-    // after X, remaining diag(tr', bl') with tr'=tr? X*diag(bl,tr)?? derive:
-    // [[0,tr],[bl,0]] = diag(tr, bl) * X  ->  apply X first, then Phase(tr, bl)
-    Phase(tr, bl, t);
 }
 
 template <typename R> void QStabilizer<R>::Mtrx(const cplx<R>* m, bitLenInt t)

@@ -97,6 +97,9 @@ QInterfacePtr<R> QUnit<R>::EntangleAll(const std::vector<bitLenInt>& qs)
 template <typename R>
 QInterfacePtr<R> QUnit<R>::EntangleOrdered(const std::vector<bitLenInt>& qs)
 {
+    // callers of the ordered form follow with non-diagonal register ops
+    // (ALU, decompose, state access): pending phase pairs must land first
+    for (bitLenInt q : qs) FlushPhasePairs(q);
     QInterfacePtr<R> unit = EntangleAll(qs);
     // in-unit swaps until shard[qs[i]].mapped == i
     for (bitLenInt i = 0; i < (bitLenInt)qs.size(); ++i) {
@@ -173,6 +176,7 @@ bool QUnit<R>::ElideControls(const std::vector<bitLenInt>& controls, bool anti, 
 
 template <typename R> void QUnit<R>::SetPermutation(bitCapInt perm, cplx<R> phase)
 {
+    pendingPairs.clear();
     for (bitLenInt q = 0; q < qubitCount; ++q) {
         shards[q].unit = MakeUnit(1u, (perm >> q) & 1u);
         shards[q].mapped = 0;
@@ -182,6 +186,7 @@ template <typename R> void QUnit<R>::SetPermutation(bitCapInt perm, cplx<R> phas
 
 template <typename R> void QUnit<R>::SetQuantumState(const cplx<R>* inputState)
 {
+    pendingPairs.clear();
     QInterfacePtr<R> unit = MakeUnit(qubitCount, 0u);
     unit->SetQuantumState(inputState);
     for (bitLenInt q = 0; q < qubitCount; ++q) {
@@ -229,6 +234,7 @@ template <typename R> void QUnit<R>::SetAmplitude(bitCapInt perm, cplx<R> amp)
 
 template <typename R> void QUnit<R>::Mtrx(const cplx<R>* m, bitLenInt t)
 {
+    if (norm(m[1]) > (R)1e-24 || norm(m[2]) > (R)1e-24) FlushPhasePairs(t);
     shards[t].unit->Mtrx(m, shards[t].mapped);
 }
 
@@ -239,12 +245,16 @@ template <typename R> void QUnit<R>::Phase(cplx<R> tl, cplx<R> br, bitLenInt t)
 
 template <typename R> void QUnit<R>::Invert(cplx<R> tr, cplx<R> bl, bitLenInt t)
 {
+    FlushPhasePairs(t);
     shards[t].unit->Invert(tr, bl, shards[t].mapped);
 }
 
 template <typename R>
 void QUnit<R>::MCMtrx(const std::vector<bitLenInt>& controls, const cplx<R>* m, bitLenInt t)
 {
+    // a pending pair on a CONTROL commutes (controlled ops are diagonal on
+    // the control); only the target's pairs must flush
+    FlushPhasePairs(t);
     std::vector<bitLenInt> live;
     for (bitLenInt c : controls) {
         bool on = false;
@@ -277,6 +287,7 @@ void QUnit<R>::MCMtrx(const std::vector<bitLenInt>& controls, const cplx<R>* m, 
 template <typename R>
 void QUnit<R>::MACMtrx(const std::vector<bitLenInt>& controls, const cplx<R>* m, bitLenInt t)
 {
+    FlushPhasePairs(t);
     std::vector<bitLenInt> live;
     for (bitLenInt c : controls) {
         bool on = false;
@@ -338,6 +349,22 @@ void QUnit<R>::MCPhase(
         }
         return;
     }
+    if (live.size() == 1u && shards[live[0]].unit != shards[t].unit) {
+        const double mt = std::hypot((double)tl.re, (double)tl.im);
+        const double mb = std::hypot((double)br.re, (double)br.im);
+        if (std::abs(mt - 1.0) < 1e-9 && std::abs(mb - 1.0) < 1e-9) {
+            // defer the cross-unit phase pair: factor diag(1,1,tl,br) into a
+            // 1q phase on the control times CPhase(arg(br/tl)) and buffer it
+            const double at = std::atan2((double)tl.im, (double)tl.re);
+            const double ab = std::atan2((double)br.im, (double)br.re);
+            if (std::abs(at) > 1e-14) {
+                shards[live[0]].unit->Phase(
+                    cplx<R>(1, 0), polar<R>(1, (R)at), shards[live[0]].mapped);
+            }
+            BufferPhasePair(live[0], t, ab - at);
+            return;
+        }
+    }
     std::vector<bitLenInt> qs(live);
     qs.push_back(t);
     try {
@@ -366,6 +393,7 @@ template <typename R>
 void QUnit<R>::UCMtrx(
     const std::vector<bitLenInt>& controls, const cplx<R>* m, bitLenInt t, bitCapInt perm)
 {
+    FlushPhasePairs(t);
     std::vector<bitLenInt> live;
     bitCapInt livePerm = 0;
     for (size_t i = 0; i < controls.size(); ++i) {
@@ -396,6 +424,7 @@ template <typename R>
 void QUnit<R>::UniformlyControlledSingleBit(
     const std::vector<bitLenInt>& controls, bitLenInt t, const cplx<R>* mtrxs)
 {
+    FlushPhasePairs(t);
     if (controls.empty()) {
         Mtrx(mtrxs, t);
         return;
@@ -411,6 +440,14 @@ void QUnit<R>::UniformlyControlledSingleBit(
 template <typename R> void QUnit<R>::Swap(bitLenInt q1, bitLenInt q2)
 {
     if (q1 == q2) return;
+    // pending pairs follow the swapped labels
+    for (auto& p : pendingPairs) {
+        if (p.a == q1) p.a = q2;
+        else if (p.a == q2) p.a = q1;
+        if (p.b == q1) p.b = q2;
+        else if (p.b == q2) p.b = q1;
+        if (p.a > p.b) std::swap(p.a, p.b);
+    }
     if (shards[q1].unit == shards[q2].unit) {
         shards[q1].unit->Swap(shards[q1].mapped, shards[q2].mapped);
         return;
@@ -421,6 +458,8 @@ template <typename R> void QUnit<R>::Swap(bitLenInt q1, bitLenInt q2)
 template <typename R> void QUnit<R>::ISwap(bitLenInt q1, bitLenInt q2)
 {
     if (q1 == q2) return;
+    FlushPhasePairs(q1);
+    FlushPhasePairs(q2);
     QInterfacePtr<R> unit = EntangleAll({ q1, q2 });
     unit->ISwap(shards[q1].mapped, shards[q2].mapped);
     MaybeSeparate(q1);
@@ -430,6 +469,8 @@ template <typename R> void QUnit<R>::ISwap(bitLenInt q1, bitLenInt q2)
 template <typename R> void QUnit<R>::IISwap(bitLenInt q1, bitLenInt q2)
 {
     if (q1 == q2) return;
+    FlushPhasePairs(q1);
+    FlushPhasePairs(q2);
     QInterfacePtr<R> unit = EntangleAll({ q1, q2 });
     unit->IISwap(shards[q1].mapped, shards[q2].mapped);
     MaybeSeparate(q1);
@@ -439,6 +480,8 @@ template <typename R> void QUnit<R>::IISwap(bitLenInt q1, bitLenInt q2)
 template <typename R> void QUnit<R>::SqrtSwap(bitLenInt q1, bitLenInt q2)
 {
     if (q1 == q2) return;
+    FlushPhasePairs(q1);
+    FlushPhasePairs(q2);
     QInterfacePtr<R> unit = EntangleAll({ q1, q2 });
     unit->SqrtSwap(shards[q1].mapped, shards[q2].mapped);
     MaybeSeparate(q1);
@@ -448,6 +491,8 @@ template <typename R> void QUnit<R>::SqrtSwap(bitLenInt q1, bitLenInt q2)
 template <typename R> void QUnit<R>::ISqrtSwap(bitLenInt q1, bitLenInt q2)
 {
     if (q1 == q2) return;
+    FlushPhasePairs(q1);
+    FlushPhasePairs(q2);
     QInterfacePtr<R> unit = EntangleAll({ q1, q2 });
     unit->ISqrtSwap(shards[q1].mapped, shards[q2].mapped);
     MaybeSeparate(q1);
@@ -457,6 +502,8 @@ template <typename R> void QUnit<R>::ISqrtSwap(bitLenInt q1, bitLenInt q2)
 template <typename R> void QUnit<R>::FSim(R theta, R phi, bitLenInt q1, bitLenInt q2)
 {
     if (q1 == q2) throw QrackError("FSim: identical qubits");
+    FlushPhasePairs(q1);
+    FlushPhasePairs(q2);
     try {
         QInterfacePtr<R> unit = EntangleAll({ q1, q2 });
         unit->FSim(theta, phi, shards[q1].mapped, shards[q2].mapped);
@@ -497,6 +544,7 @@ template <typename R> bool QUnit<R>::ForceM(bitLenInt q, bool result, bool doFor
     if (prob <= 0) throw QrackError("QUnit::ForceM: impossible outcome");
     s.unit->ForceM(s.mapped, outcome, true, true);
     SeparateBit(q, outcome);
+    ResolvePhasePairsOnMeasure(q, outcome);
     return outcome;
 }
 
@@ -625,6 +673,7 @@ double QUnit<R>::ExpectationBitsFactorized(
 
 template <typename R> bool QUnit<R>::TrySeparate(bitLenInt q)
 {
+    FlushPhasePairs(q);
     Shard& s = shards[q];
     QInterfacePtr<R> unit = s.unit;
     if (unit->GetQubitCount() == 1u) return true;
@@ -732,6 +781,15 @@ template <typename R> bitLenInt QUnit<R>::Compose(QInterfacePtr<R> toCopy, bitLe
         }
         shards.insert(shards.begin() + start, ns.begin(), ns.end());
     }
+    for (auto& p : pendingPairs) {
+        if (p.a >= start) p.a += oQubits;
+        if (p.b >= start) p.b += oQubits;
+    }
+    if (o) {
+        for (const auto& p : o->pendingPairs) {
+            pendingPairs.push_back({ (bitLenInt)(p.a + start), (bitLenInt)(p.b + start), p.angle });
+        }
+    }
     this->SetQubitCount(qubitCount + oQubits);
     return start;
 }
@@ -755,6 +813,10 @@ template <typename R> void QUnit<R>::Decompose(bitLenInt start, QInterfacePtr<R>
     }
     dest->SetQuantumState(buf.data());
     shards.erase(shards.begin() + start, shards.begin() + start + len);
+    for (auto& p : pendingPairs) {
+        if (p.a >= start + len) p.a -= len;
+        if (p.b >= start + len) p.b -= len;
+    }
     this->SetQubitCount(qubitCount - len);
 }
 
@@ -770,6 +832,10 @@ template <typename R> void QUnit<R>::Dispose(bitLenInt start, bitLenInt length)
         }
     }
     shards.erase(shards.begin() + start, shards.begin() + start + length);
+    for (auto& p : pendingPairs) {
+        if (p.a >= start + length) p.a -= length;
+        if (p.b >= start + length) p.b -= length;
+    }
     this->SetQubitCount(qubitCount - length);
 }
 
@@ -790,6 +856,10 @@ template <typename R> bitLenInt QUnit<R>::Allocate(bitLenInt start, bitLenInt le
         ns[q].mapped = 0;
     }
     shards.insert(shards.begin() + start, ns.begin(), ns.end());
+    for (auto& p : pendingPairs) {
+        if (p.a >= start) p.a += length;
+        if (p.b >= start) p.b += length;
+    }
     this->SetQubitCount(qubitCount + length);
     return start;
 }
@@ -806,6 +876,7 @@ template <typename R> QInterfacePtr<R> QUnit<R>::Clone()
         clone->shards[q].mapped = shards[q].mapped;
     }
     clone->logFidelity = logFidelity;
+    clone->pendingPairs = pendingPairs;
     clone->sdrp = sdrp;
     clone->ncrp = ncrp;
     clone->separabilityThreshold = separabilityThreshold;

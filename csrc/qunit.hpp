@@ -38,6 +38,79 @@ protected:
     bool reactiveSeparate = false; // TrySeparate after entangling gates
     bitLenInt aceMaxQubits = 0; // 0 = unlimited; else entangle cap (ACE)
 
+    // deferred cross-unit controlled-phase pairs (the core of the reference's
+    // phase-shard optimization, qengineshard.hpp PhaseShards): a CPhase
+    // between SEPARATE units is buffered instead of entangling. Diagonal
+    // pending pairs leave every Z-basis probability/sampling query exact;
+    // they flush only when a non-diagonal gate (or state access / structural
+    // op) touches a buffered qubit. Same-pair buffers combine (and cancel).
+    struct PhasePair {
+        bitLenInt a, b; // logical ids, a < b
+        double angle;   // phase on |11>
+    };
+    std::vector<PhasePair> pendingPairs;
+
+    void BufferPhasePair(bitLenInt q1, bitLenInt q2, double angle)
+    {
+        if (q1 > q2) std::swap(q1, q2);
+        for (size_t i = 0; i < pendingPairs.size(); ++i) {
+            if (pendingPairs[i].a == q1 && pendingPairs[i].b == q2) {
+                pendingPairs[i].angle += angle;
+                const double rem = std::fmod(pendingPairs[i].angle, 2.0 * 3.14159265358979323846);
+                if (std::abs(rem) < 1e-12 || std::abs(std::abs(rem) - 2.0 * 3.14159265358979323846) < 1e-12) {
+                    pendingPairs.erase(pendingPairs.begin() + i); // cancelled
+                }
+                return;
+            }
+        }
+        pendingPairs.push_back({ q1, q2, angle });
+    }
+
+    void ApplyPairNow(const PhasePair& p)
+    {
+        QInterfacePtr<R> unit = EntangleAll({ p.a, p.b });
+        unit->MCPhase(
+            { shards[p.a].mapped }, cplx<R>(1, 0), polar<R>(1, (R)p.angle), shards[p.b].mapped);
+    }
+
+    // flush every pending pair touching q (non-diagonal op incoming)
+    void FlushPhasePairs(bitLenInt q)
+    {
+        if (pendingPairs.empty()) return;
+        std::vector<PhasePair> todo;
+        for (size_t i = pendingPairs.size(); i-- > 0;) {
+            if (pendingPairs[i].a == q || pendingPairs[i].b == q) {
+                todo.push_back(pendingPairs[i]);
+                pendingPairs.erase(pendingPairs.begin() + i);
+            }
+        }
+        for (const auto& p : todo) ApplyPairNow(p);
+    }
+
+    void FlushAllPhasePairs()
+    {
+        std::vector<PhasePair> todo;
+        todo.swap(pendingPairs);
+        for (const auto& p : todo) ApplyPairNow(p);
+    }
+
+    // measurement resolution: q collapsed to `outcome` — each pending pair
+    // involving q degenerates to a 1-qubit phase on the partner (outcome=1)
+    // or vanishes (outcome=0)
+    void ResolvePhasePairsOnMeasure(bitLenInt q, bool outcome)
+    {
+        if (pendingPairs.empty()) return;
+        for (size_t i = pendingPairs.size(); i-- > 0;) {
+            const PhasePair p = pendingPairs[i];
+            if (p.a != q && p.b != q) continue;
+            pendingPairs.erase(pendingPairs.begin() + i);
+            if (!outcome) continue;
+            const bitLenInt other = (p.a == q) ? p.b : p.a;
+            shards[other].unit->Phase(
+                cplx<R>(1, 0), polar<R>(1, (R)p.angle), shards[other].mapped);
+        }
+    }
+
     // reactive separation after an entangling gate (active only under SDRP)
     void MaybeSeparate(bitLenInt q)
     {
@@ -145,6 +218,11 @@ public:
     {
         if (mtrxs.size() != 4u * targets.size())
             throw QrackError("Mtrx1qBatch: need 4 entries per target");
+        for (size_t i = 0; i < targets.size(); ++i) {
+            if (norm(mtrxs[4u * i + 1u]) > (R)1e-24 || norm(mtrxs[4u * i + 2u]) > (R)1e-24) {
+                FlushPhasePairs(targets[i]);
+            }
+        }
         std::map<QInterface<R>*, std::pair<std::vector<bitLenInt>, std::vector<cplx<R>>>> groups;
         std::map<QInterface<R>*, QInterfacePtr<R>> keep;
         for (size_t i = 0; i < targets.size(); ++i) {
@@ -194,6 +272,7 @@ public:
         const std::vector<bitCapInt>& perms, bitCapInt offset = 0) override;
     void GetReducedDensityMatrix(bitLenInt q, cplx<R>* out) override
     {
+        FlushPhasePairs(q);
         shards[q].unit->GetReducedDensityMatrix(shards[q].mapped, out);
     }
 

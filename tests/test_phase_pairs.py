@@ -1,0 +1,159 @@
+"""QUnit deferred cross-unit controlled-phase pairs (the core of the
+reference's phase-shard optimization, qengineshard.hpp PhaseShards):
+cross-unit CZ/CPhase gates buffer instead of entangling, flush exactly on
+non-diagonal contact, resolve under measurement, and cancel pairwise."""
+
+import numpy as np
+import pytest
+
+import qrack_amd as qa
+from ref_sim import assert_states_close
+
+
+def make(n, seed=3):
+    return qa.create_simulator(n, layers=["qunit", "cpu"], seed=seed)
+
+
+def cpu(n, seed=3):
+    return qa.create_simulator(n, engine="cpu", seed=seed)
+
+
+def test_cz_chain_stays_separable_and_exact():
+    n = 8
+    q = make(n)
+    cp = cpu(n)
+    for i in range(n):
+        q.h(i)
+        cp.h(i)
+    for i in range(n - 1):
+        q.cz(i, i + 1)
+        cp.cz(i, i + 1)
+    # Z-basis probabilities exact without any entangling
+    for i in range(n):
+        assert abs(q.prob(i) - 0.5) < 1e-6
+    # state access flushes the pairs exactly
+    assert_states_close(q.get_state_vector(), cp.get_state_vector(), 1e-4)
+
+
+def test_cz_cancellation_never_entangles():
+    n = 4
+    q = make(n)
+    for i in range(n):
+        q.h(i)
+    q.cz(0, 2)
+    q.cz(2, 0)  # same pair: angles sum to 2 pi -> cancelled
+    q.cz(1, 3)
+    q.cz(1, 3)
+    # mirror: H back must give |0...0> exactly with everything separable
+    for i in range(n):
+        q.h(i)  # flushes nothing: all pairs cancelled
+    for i in range(n):
+        assert q.prob(i) < 1e-6
+    assert q.get_unitary_fidelity() == pytest.approx(1.0)
+
+
+def test_graph_state_matches_dense():
+    # 2D-ish graph state: H all + CZ edges, then single-qubit rotations
+    n = 6
+    edges = [(0, 1), (1, 2), (2, 3), (3, 4), (4, 5), (0, 3), (1, 4)]
+    q = make(n)
+    cp = cpu(n)
+    for i in range(n):
+        q.h(i)
+        cp.h(i)
+    for a, b in edges:
+        q.cz(a, b)
+        cp.cz(a, b)
+    for i in range(n):
+        q.ry(0.3 + 0.1 * i, i)  # non-diagonal: flushes that qubit's pairs
+        cp.ry(0.3 + 0.1 * i, i)
+    assert_states_close(q.get_state_vector(), cp.get_state_vector(), 1e-4)
+
+
+def test_cphase_angles_combine():
+    q = make(3)
+    cp = cpu(3)
+    for s in (q, cp):
+        s.h(0)
+        s.h(2)
+    for s in (q, cp):
+        s.mcphase([0], 1, complex(np.exp(0.4j)), 2)
+        s.mcphase([2], 1, complex(np.exp(0.5j)), 0)  # symmetric: combines
+    assert_states_close(q.get_state_vector(), cp.get_state_vector(), 1e-5)
+
+
+def test_measurement_resolves_pairs():
+    # CZ then measure the control: partner gets the phase iff outcome is 1
+    for forced in (False, True):
+        q = make(2)
+        cp = cpu(2)
+        for s in (q, cp):
+            s.h(0)
+            s.h(1)
+            s.mcphase([0], 1, complex(np.exp(0.8j)), 1)
+        ra = q.force_m(0, forced)
+        rb = cp.force_m(0, forced)
+        assert ra == rb == forced
+        assert_states_close(q.get_state_vector(), cp.get_state_vector(), 1e-5)
+
+
+def test_swap_renames_pending_pairs():
+    n = 4
+    q = make(n)
+    cp = cpu(n)
+    for s in (q, cp):
+        s.h(0)
+        s.h(1)
+        s.cz(0, 1)
+        s.swap(1, 3)
+        s.h(3)  # must flush the renamed (0,3) pair
+    assert_states_close(q.get_state_vector(), cp.get_state_vector(), 1e-5)
+
+
+def test_multishot_with_pending_pairs():
+    # diagonal pending pairs leave Z sampling exact (no flush, no entangle)
+    n = 10
+    q = make(n)
+    for i in range(n):
+        q.h(i)
+    for i in range(0, n - 1):
+        q.cz(i, i + 1)
+    res = q.multi_shot_measure_mask([1 << i for i in range(n)], 200)
+    assert sum(res.values()) == 200
+    assert q.get_unitary_fidelity() == pytest.approx(1.0)
+
+
+def test_qft_uses_buffered_phases():
+    # QFT lowers to H + cross-unit CPhases: buffered pairs must still give
+    # the exact transform
+    n = 6
+    x = 44
+    q = make(n)
+    q.set_permutation(x % (1 << n))
+    q.qft(0, n)
+    sv = np.asarray(q.get_state_vector()).astype(np.complex128)
+    N = 1 << n
+    k = np.arange(N)
+    expected = np.exp(2j * np.pi * (x % N) * k / N) / np.sqrt(N)
+    rev = np.array([int(format(i, f"0{n}b")[::-1], 2) for i in range(N)])
+    inner = np.vdot(expected, sv[rev])
+    assert abs(abs(inner) - 1.0) < 1e-4
+
+
+def test_compose_dispose_with_pending_pairs():
+    a = make(3, seed=1)
+    for i in range(3):
+        a.h(i)
+    a.cz(0, 2)
+    b = cpu(2, seed=2)
+    b.x(0)
+    a.compose(b)  # ids shift? (append at end: unchanged) pairs survive
+    assert a.num_qubits == 5
+    a.h(2)  # flush (0,2)
+    ref = cpu(5, seed=9)
+    for i in range(3):
+        ref.h(i)
+    ref.cz(0, 2)
+    ref.x(3)
+    ref.h(2)
+    assert_states_close(a.get_state_vector(), ref.get_state_vector(), 1e-4)

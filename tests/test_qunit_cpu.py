@@ -249,16 +249,25 @@ def test_ace_elision_caps_entanglement():
         q = make(n, ["qunit", "cpu"], seed=13)
         for i in range(n):
             q.h(i)
-        # chain of CZs would entangle everything; the cap forces elision
+        # a CZ chain is now EXACT under the cap: cross-unit phase pairs
+        # buffer instead of entangling (deferred-CZ optimization)
         for i in range(n - 1):
             q.cz(i, i + 1)
-        # still functional: probabilities remain sane and fidelity < 1
         for i in range(n):
-            p = q.prob(i)
+            assert abs(q.prob(i) - 0.5) < 1e-5
+        assert q.get_unitary_fidelity() == pytest.approx(1.0)
+        # a CNOT chain is non-diagonal: the cap forces classical elision
+        q2 = make(n, ["qunit", "cpu"], seed=14)
+        for i in range(n):
+            q2.h(i)
+        for i in range(n - 1):
+            q2.cnot(i, i + 1)
+        for i in range(n):
+            p = q2.prob(i)
             assert 0.0 <= p <= 1.0
-        assert q.get_unitary_fidelity() < 1.0
-        assert q.get_unitary_fidelity() > 0.0
-        r = q.m_all()
+        assert q2.get_unitary_fidelity() < 1.0
+        assert q2.get_unitary_fidelity() > 0.0
+        r = q2.m_all()
         assert 0 <= r < (1 << n)
     finally:
         del os.environ["QRACK_QUNIT_ACE_MAX_QB"]
