@@ -29,6 +29,8 @@ def main():
     p.add_argument("--qubits", type=int, default=0, help="base (per-GPU) qubit count; 0 = auto")
     p.add_argument("--precision", default="fp32", choices=["fp32", "fp64"])
     p.add_argument("--engine", default="auto", choices=["auto", "hip", "cpu"])
+    p.add_argument("--backend", default="auto",
+                   help="torch.distributed backend override (auto = cpu:gloo,cuda:nccl for hip)")
     args = p.parse_args()
 
     import qrack_amd as qa
@@ -54,7 +56,9 @@ def main():
         # mixed backend: CUDA tensors (half-page exchanges) over RCCL/xGMI,
         # CPU tensors (norm scalars, decision broadcasts, object gathers)
         # over gloo — a pure nccl group rejects CPU-tensor collectives
-        backend = "cpu:gloo,cuda:nccl" if engine == "hip" else "gloo"
+        backend = args.backend
+        if backend == "auto":
+            backend = "cpu:gloo,cuda:nccl" if engine == "hip" else "gloo"
         tdist.init_process_group(backend=backend)
         if engine == "hip":
             torch.cuda.set_device(local_rank)
