@@ -61,6 +61,9 @@ def main():
             backend = "cpu:gloo,cuda:nccl" if engine == "hip" else "gloo"
         tdist.init_process_group(backend=backend)
         if engine == "hip":
+            # clamp for single-GPU multi-rank rehearsals; identity on real
+            # one-rank-per-GPU launches
+            local_rank = local_rank % max(1, qa.hip_device_count())
             torch.cuda.set_device(local_rank)
         dist = tdist
 
