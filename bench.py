@@ -33,12 +33,21 @@ def main():
                    help="torch.distributed backend override (auto = cpu:gloo,cuda:nccl for hip)")
     args = p.parse_args()
 
-    import qrack_amd as qa
-
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
     n_gpus = max(args.gpus, world)
+
+    if world > 1:
+        # ORDER MATTERS: torch.cuda must initialize the HIP runtime BEFORE
+        # the qrack_amd extension touches it — the reverse order leaves
+        # torch.cuda reporting "No HIP GPUs are available" in this process
+        import torch
+
+        if torch.cuda.is_available():
+            torch.cuda.init()
+
+    import qrack_amd as qa
 
     engine = args.engine
     if engine == "auto":
