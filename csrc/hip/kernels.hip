@@ -1259,6 +1259,53 @@ template <int K> __global__ void k_mtrx_batch_v(cplx<float>* sv, Batch1qArgs<flo
     }
 }
 
+// ---- LDS-tiled low-bit gate batch -------------------------------------------
+// Each workgroup stages a contiguous 2^QA_LDS_TILE_BITS-amplitude tile in
+// LDS, applies every gate (targets all inside the tile) at LDS bandwidth,
+// then writes the tile back: ONE global RMW pass for up to 12 fused gates,
+// with bit-0 targets handled as cheaply as any other.
+
+template <typename R> __global__ void k_mtrx_batch_lds(cplx<R>* sv, BatchLdsArgs<R> a)
+{
+    constexpr int TB = qaLdsTileBits<R>();
+    __shared__ cplx<R> tile[1u << TB];
+    constexpr unsigned TILE = 1u << TB;
+    const bitCapInt nTiles = a.maxQPower >> TB;
+    for (bitCapInt t = blockIdx.x; t < nTiles; t += gridDim.x) {
+        const bitCapInt base = t << TB;
+        for (unsigned i = threadIdx.x; i < TILE; i += blockDim.x) {
+            tile[i] = sv[base + i];
+        }
+        __syncthreads();
+        for (int g = 0; g < a.k; ++g) {
+            const unsigned tp = (unsigned)a.tPow[g];
+            const cplx<R> m0 = a.m[4 * g], m1 = a.m[4 * g + 1], m2 = a.m[4 * g + 2],
+                          m3 = a.m[4 * g + 3];
+            for (unsigned j = threadIdx.x; j < (TILE >> 1); j += blockDim.x) {
+                const unsigned i0 = ((j & ~(tp - 1u)) << 1u) | (j & (tp - 1u));
+                const unsigned i1 = i0 | tp;
+                const cplx<R> x = tile[i0];
+                const cplx<R> y = tile[i1];
+                tile[i0] = m0 * x + m1 * y;
+                tile[i1] = m2 * x + m3 * y;
+            }
+            __syncthreads();
+        }
+        for (unsigned i = threadIdx.x; i < TILE; i += blockDim.x) {
+            sv[base + i] = tile[i];
+        }
+        __syncthreads();
+    }
+}
+
+template <typename R>
+void launchMtrx1qBatchLds(cplx<R>* sv, const BatchLdsArgs<R>& a, hipStream_t stream)
+{
+    const bitCapInt nTiles = a.maxQPower >> qaLdsTileBits<R>();
+    const int grid = (int)std::min<bitCapInt>(nTiles, (bitCapInt)QA_MAX_BLOCKS);
+    hipLaunchKernelGGL((k_mtrx_batch_lds<R>), dim3(grid), dim3(QA_BLOCK), 0, stream, sv, a);
+}
+
 // ---- batched disjoint CNOTs: one permutation pass per layer ----------------
 
 template <typename R> __global__ void k_cnot_batch(cplx<R>* sv, CnotBatchArgs a)
@@ -1562,6 +1609,7 @@ void launchPartProbs(const cplx<R>* sv, bitCapInt maxQPower, bitLenInt start, bi
         cplx<R>*, bitCapInt, bitLenInt, bitLenInt, bitCapInt, int, bool, hipStream_t);              \
     template void launchMtrx1qBatch<R>(cplx<R>*, const Batch1qArgs<R>&, hipStream_t);                               \
     template void launchCnotBatch<R>(cplx<R>*, const CnotBatchArgs&, hipStream_t);                              \
+    template void launchMtrx1qBatchLds<R>(cplx<R>*, const BatchLdsArgs<R>&, hipStream_t);                              \
     template void launchCPhasePairs<R>(cplx<R>*, const CPhasePairsArgs&, hipStream_t);                               \
     template void launchQftColumnGeneral<R>(                                                        \
         cplx<R>*, bitCapInt, bitCapInt, const RampArgs&, double, bool, hipStream_t);

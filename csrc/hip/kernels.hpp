@@ -201,6 +201,25 @@ template <typename R> struct Batch1qArgs {
 template <typename R>
 void launchMtrx1qBatch(cplx<R>* sv, const Batch1qArgs<R>& a, hipStream_t stream);
 
+// LDS-tiled low-bit batch: gates whose targets all sit below QA_LDS_TILE_BITS
+// apply inside a shared-memory tile — up to 12 gates in ONE global RMW pass
+// (the register-orbit kernels degrade on low-bit targets; LDS does not).
+// 32 KB LDS tiles both ways: 4096 fp32 amps / 2048 fp64 amps (a 64 KB fp64
+// tile would sit exactly on the per-workgroup LDS limit)
+template <typename R> constexpr int qaLdsTileBits() { return sizeof(R) == 4 ? 12 : 11; }
+constexpr int QA_LDS_TILE_BITS = 12; // fp32 value; use qaLdsTileBits<R>()
+constexpr int QA_MAX_BATCH_LDS = 12;
+
+template <typename R> struct BatchLdsArgs {
+    cplx<R> m[4 * QA_MAX_BATCH_LDS];
+    bitCapInt tPow[QA_MAX_BATCH_LDS]; // sorted ascending, all < 2^QA_LDS_TILE_BITS
+    int k;
+    bitCapInt maxQPower;
+};
+
+template <typename R>
+void launchMtrx1qBatchLds(cplx<R>* sv, const BatchLdsArgs<R>& a, hipStream_t stream);
+
 // batched disjoint CNOTs: a whole layer of k control/target pairs (no qubit
 // repeated) applied as ONE in-place permutation pass — amp[i] swaps with
 // amp[i ^ xm(i)] where xm(i) XORs tPow[j] for every set control bit. One

@@ -479,16 +479,47 @@ void QEngineHIP<R>::Mtrx1qBatch(
     // scalar kernel's 8 B accesses cost more than the extra pass saves
     const size_t maxK = 4u;
     QA_HIP_CHECK(hipSetDevice(deviceId));
-    size_t i = 0;
-    while (i < targets.size()) {
-        const size_t k = std::min(maxK, targets.size() - i);
-        if (k == 1u) {
-            this->Mtrx(&mtrxs[4u * i], targets[i]);
+    // partition: targets below the LDS tile go through the shared-memory
+    // kernel (up to 12 per ONE pass, bit 0 included); the rest chunk into
+    // float4 register batches of 4
+    std::vector<size_t> low, high;
+    const bitLenInt ldsBits = (bitLenInt)qaLdsTileBits<R>();
+    if (qubitCount > ldsBits) {
+        for (size_t j = 0; j < targets.size(); ++j) {
+            (targets[j] < ldsBits ? low : high).push_back(j);
+        }
+    } else {
+        for (size_t j = 0; j < targets.size(); ++j) high.push_back(j);
+    }
+    std::sort(low.begin(), low.end(),
+        [&](size_t a, size_t b) { return targets[a] < targets[b]; });
+    for (size_t i = 0; i < low.size();) {
+        const size_t k = std::min((size_t)QA_MAX_BATCH_LDS, low.size() - i);
+        if (k == 1u && high.empty() && low.size() == 1u) {
+            this->Mtrx(&mtrxs[4u * low[i]], targets[low[i]]);
             ++i;
             continue;
         }
-        std::vector<size_t> ord(k);
-        for (size_t j = 0; j < k; ++j) ord[j] = i + j;
+        BatchLdsArgs<R> a{};
+        for (size_t g = 0; g < k; ++g) {
+            a.tPow[g] = pow2(targets[low[i + g]]);
+            for (int e = 0; e < 4; ++e) a.m[4u * g + e] = mtrxs[4u * low[i + g] + e];
+        }
+        a.k = (int)k;
+        a.maxQPower = maxQPower;
+        HipProfScope prof("mtrx_1q_batch_lds", stream);
+        launchMtrx1qBatchLds<R>(dState, a, stream);
+        i += k;
+    }
+    size_t i = 0;
+    while (i < high.size()) {
+        const size_t k = std::min(maxK, high.size() - i);
+        if (k == 1u) {
+            this->Mtrx(&mtrxs[4u * high[i]], targets[high[i]]);
+            ++i;
+            continue;
+        }
+        std::vector<size_t> ord(high.begin() + i, high.begin() + i + k);
         std::sort(ord.begin(), ord.end(),
             [&](size_t a, size_t b) { return targets[a] < targets[b]; });
         Batch1qArgs<R> a{};
