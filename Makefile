@@ -46,3 +46,14 @@ asan:
 	    csrc/qpager.cpp csrc/qfactory.cpp csrc/common/parallel_for.cpp \
 	    tools/asan_smoke.cpp -o build/asan_smoke -lpthread
 	./build/asan_smoke
+
+# TSAN lane: exercises the thread-pool ParallelFor under the race detector
+# (the reference has no sanitizer lanes; the thread-safety
+# contract — one engine instance per thread — is enforced here instead)
+tsan:
+	g++ -O1 -g -std=c++17 -fsanitize=thread -fno-omit-frame-pointer \
+	    -Icsrc csrc/qinterface.cpp csrc/qengine_cpu.cpp csrc/qengine_sparse.cpp \
+	    csrc/qstabilizer.cpp csrc/qstabilizerhybrid.cpp csrc/qunit.cpp csrc/qbdt.cpp \
+	    csrc/qpager.cpp csrc/qfactory.cpp csrc/common/parallel_for.cpp \
+	    tools/asan_smoke.cpp -o build/tsan_smoke -lpthread
+	./build/tsan_smoke

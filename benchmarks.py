@@ -31,6 +31,14 @@ def run_qft(q, n, rng, depth):
     q.qft(0, n)
 
 
+def run_qft_cosmology(q, n, rng, depth):
+    # the reference's test_qft_cosmology: QFT of a fully superposed register
+    q.set_permutation(0)
+    mats = [_M1Q["h"]] * n
+    _apply_1q_layer(q, range(n), mats)
+    q.qft(0, n)
+
+
 def run_ghz(q, n, rng, depth):
     q.set_permutation(0)
     q.h(0)
@@ -115,6 +123,7 @@ def run_qv(q, n, rng, depth):
 
 WORKLOADS = {
     "qft": run_qft,
+    "qft_cosmology": run_qft_cosmology,
     "ghz": run_ghz,
     "random_circuit": run_random_circuit,
     "supremacy": run_supremacy,
