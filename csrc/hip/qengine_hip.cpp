@@ -495,7 +495,8 @@ void QEngineHIP<R>::Mtrx1qBatch(
         [&](size_t a, size_t b) { return targets[a] < targets[b]; });
     for (size_t i = 0; i < low.size();) {
         const size_t k = std::min((size_t)QA_MAX_BATCH_LDS, low.size() - i);
-        if (k == 1u && high.empty() && low.size() == 1u) {
+        if (k == 1u) {
+            // a lone gate is cheaper on the plain pair kernel than an LDS pass
             this->Mtrx(&mtrxs[4u * low[i]], targets[low[i]]);
             ++i;
             continue;

