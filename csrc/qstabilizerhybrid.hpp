@@ -68,6 +68,24 @@ public:
         }
         return true;
     }
+    // serialization access (tableau + buffered shards; reference
+    // qstabilizerhybrid.cpp:2235-2291 writes both)
+    QStabilizerPtr<R> Tableau() { return stabilizer; }
+    bool HasShard(bitLenInt q) const { return (bool)shards[q]; }
+    const std::array<cplx<R>, 4>& ShardData(bitLenInt q) const { return *shards[q]; }
+    void InjectShard(bitLenInt q, const cplx<R>* m)
+    {
+        shards[q] = std::make_unique<std::array<cplx<R>, 4>>();
+        for (int i = 0; i < 4; ++i) (*shards[q])[i] = m[i];
+    }
+    void ReplaceTableau(QStabilizerPtr<R> st)
+    {
+        if (st->GetQubitCount() != qubitCount)
+            throw QrackError("ReplaceTableau: width mismatch");
+        stabilizer = st;
+        engine.reset();
+    }
+
     bool isClifford() const override { return !InEngineMode(); }
     bool isClifford(bitLenInt q) const override
     {

@@ -24,19 +24,63 @@ template <typename R> std::string SaveStabilizerText(QInterfacePtr<R> q)
         return st->Serialize();
     }
     if (auto hy = std::dynamic_pointer_cast<QStabilizerHybrid<R>>(q)) {
-        if (!hy->isClifford() || !hy->TryFlushAllShards()) {
+        if (!hy->isClifford()) {
             throw QrackError("stabilizer save: state is no longer Clifford");
         }
-        auto st = std::dynamic_pointer_cast<QStabilizer<R>>(hy->ActiveBackend());
-        if (!st) throw QrackError("stabilizer save: no tableau backend");
-        return st->Serialize();
+        // fold Clifford-product shards into the tableau, keep the rest
+        hy->TryFlushAllShards();
+        std::string out = hy->Tableau()->Serialize();
+        bool any = false;
+        for (bitLenInt i = 0; i < q->GetQubitCount(); ++i) {
+            if (hy->HasShard(i)) any = true;
+        }
+        if (any) {
+            // non-Clifford 1q buffers ride along (reference: tableau +
+            // 4-complex MpsShard per qubit, qstabilizerhybrid.cpp:2235-2291)
+            std::ostringstream ss;
+            ss << out << "SHARDS\n";
+            ss.setf(std::ios::scientific);
+            ss.precision(17);
+            for (bitLenInt i = 0; i < q->GetQubitCount(); ++i) {
+                if (!hy->HasShard(i)) {
+                    ss << "I\n";
+                    continue;
+                }
+                const auto& m = hy->ShardData(i);
+                ss << "S";
+                for (int e = 0; e < 4; ++e) ss << " " << m[e].re << " " << m[e].im;
+                ss << "\n";
+            }
+            return ss.str();
+        }
+        return out;
     }
     throw QrackError("stabilizer save: not a stabilizer-capable layer");
 }
 
 template <typename R> QInterfacePtr<R> LoadStabilizerText(const std::string& s, RngPtr rng)
 {
-    return QStabilizer<R>::Deserialize(s, rng);
+    const size_t shardPos = s.find("SHARDS\n");
+    if (shardPos == std::string::npos) {
+        return QStabilizer<R>::Deserialize(s, rng);
+    }
+    QStabilizerPtr<R> st = QStabilizer<R>::Deserialize(s.substr(0, shardPos), rng);
+    auto hy = std::make_shared<QStabilizerHybrid<R>>(st->GetQubitCount(), 0u, rng);
+    hy->ReplaceTableau(st);
+    std::istringstream ss(s.substr(shardPos + 7u));
+    std::string line;
+    for (bitLenInt q = 0; q < st->GetQubitCount() && std::getline(ss, line); ++q) {
+        if (line.empty() || line[0] == 'I') continue;
+        std::istringstream ls(line.substr(1));
+        cplx<R> m[4];
+        for (int e = 0; e < 4; ++e) {
+            double re = 0, im = 0;
+            ls >> re >> im;
+            m[e] = cplx<R>((R)re, (R)im);
+        }
+        hy->InjectShard(q, m);
+    }
+    return hy;
 }
 
 // ---- lossy quantized binary -------------------------------------------------

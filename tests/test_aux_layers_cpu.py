@@ -221,9 +221,16 @@ def test_stabilizer_hybrid_save():
     text = qa.save_stabilizer_F(q)
     q2 = qa.load_stabilizer_F(text)
     assert abs(q2.prob(0) - 0.5) < 1e-6
-    # non-Clifford state must refuse
+    # buffered non-Clifford shards now SAVE (SHARDS block); an
+    # engine-materialized state must still refuse
     q.t(0)
     q.ry(0.3, 0)
+    text2 = qa.save_stabilizer_F(q)
+    assert "SHARDS" in text2
+    q3 = qa.load_stabilizer_F(text2)
+    assert abs(q3.prob(0) - q.prob(0)) < 1e-5
+    q.cnot(0, 1)  # non-Clifford shard forced onto the tableau -> engine mode
+    assert not q.is_clifford()
     with pytest.raises(Exception):
         qa.save_stabilizer_F(q)
 
@@ -271,3 +278,25 @@ def test_turboquant_rotation_improves_int8_fidelity(tmp_path):
     assert f_rot > f_plain
     assert f_rot > 0.999
     assert fid(16, True) > 1 - 1e-8
+
+
+def test_stabilizer_hybrid_save_with_shards(tmp_path):
+    """The Clifford text stream carries non-Clifford 1q shard buffers
+    (reference: tableau + MpsShard per qubit)."""
+    q = qa.create_simulator(3, layers=["stabilizer_hybrid", "cpu"], seed=4)
+    q.h(0)
+    q.cnot(0, 1)
+    q.t(1)          # buffered non-Clifford shard
+    q.rz(0.3, 2)    # another
+    text = qa.save_stabilizer_F(q)
+    assert "SHARDS" in text
+    q2 = qa.load_stabilizer_F(text, 4)
+    sv1 = np.asarray(q.get_state_vector())
+    sv2 = np.asarray(q2.get_state_vector())
+    inner = abs(np.vdot(sv1, sv2))
+    assert inner > 1 - 1e-5
+    # clean Clifford states keep the plain tableau format
+    q3 = qa.create_simulator(2, layers=["stabilizer_hybrid", "cpu"], seed=5)
+    q3.h(0)
+    q3.cnot(0, 1)
+    assert "SHARDS" not in qa.save_stabilizer_F(q3)
