@@ -181,6 +181,45 @@ public:
     }
     double GetNcrp() override { return ncrp; }
 
+    // ---- subsystem-aware serialization access (QUNTQ-container parity) ----
+    void FlushAllForSerialize() { FlushAllPhasePairs(); }
+    QInterfacePtr<R> NewUnit(bitLenInt w) { return MakeUnit(w, 0u); }
+    // units[] = distinct sub-states; qmap[q] = (index into units, mapped bit)
+    void GetUnitMap(std::vector<QInterfacePtr<R>>& units,
+        std::vector<std::pair<uint32_t, uint32_t>>& qmap)
+    {
+        units.clear();
+        qmap.assign(qubitCount, { 0u, 0u });
+        std::map<QInterface<R>*, uint32_t> idx;
+        for (bitLenInt q = 0; q < qubitCount; ++q) {
+            auto it = idx.find(shards[q].unit.get());
+            uint32_t u;
+            if (it == idx.end()) {
+                u = (uint32_t)units.size();
+                idx[shards[q].unit.get()] = u;
+                units.push_back(shards[q].unit);
+            } else {
+                u = it->second;
+            }
+            qmap[q] = { u, (uint32_t)shards[q].mapped };
+        }
+    }
+    // install freshly loaded units: qmap as above, states[] the new units
+    void RebuildFromUnits(const std::vector<QInterfacePtr<R>>& states,
+        const std::vector<std::pair<uint32_t, uint32_t>>& qmap)
+    {
+        if (qmap.size() != (size_t)qubitCount)
+            throw QrackError("RebuildFromUnits: width mismatch");
+        pendingPairs.clear();
+        for (bitLenInt q = 0; q < qubitCount; ++q) {
+            if (qmap[q].first >= states.size())
+                throw QrackError("RebuildFromUnits: unit index out of range");
+            shards[q].unit = states[qmap[q].first];
+            shards[q].mapped = (bitLenInt)qmap[q].second;
+        }
+        logFidelity = 0.0;
+    }
+
     void SetReactiveSeparate(bool on) override { reactiveSeparate = on; }
     bool GetReactiveSeparate() override { return reactiveSeparate; }
 

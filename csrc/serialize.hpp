@@ -12,6 +12,7 @@
 #include "qengine.hpp"
 #include "qstabilizer.hpp"
 #include "qstabilizerhybrid.hpp"
+#include "qunit.hpp"
 
 #include <cstdio>
 #include <cstring>
@@ -124,8 +125,8 @@ inline void qa_wht(double* v, size_t n)
 }
 
 template <typename R>
-void LossySaveState(QInterfacePtr<R> q, const std::string& path, bitLenInt blockBits = 12,
-    int bits = 16, bool rotate = true, uint64_t seed = 0x51a4d70f2u)
+void LossySaveToStream(QInterfacePtr<R> q, FILE* f, bitLenInt blockBits = 12, int bits = 16,
+    bool rotate = true, uint64_t seed = 0x51a4d70f2u)
 {
     if (bits != 8 && bits != 16) throw QrackError("LossySaveState: bits must be 8 or 16");
     const bitLenInt n = q->GetQubitCount();
@@ -133,8 +134,6 @@ void LossySaveState(QInterfacePtr<R> q, const std::string& path, bitLenInt block
     if (blockBits > n) blockBits = n;
     const bitCapInt blockLen = pow2(blockBits);
     const size_t reals = (size_t)blockLen * 2u;
-    FILE* f = std::fopen(path.c_str(), "wb");
-    if (!f) throw QrackError("LossySaveState: cannot open " + path);
     std::fwrite(QAMD_TQ2_MAGIC, 1, 8, f);
     const uint64_t qb = n, bb = blockBits, prec = sizeof(R), vbits = (uint64_t)bits,
                    vrot = rotate ? 1u : 0u;
@@ -185,17 +184,91 @@ void LossySaveState(QInterfacePtr<R> q, const std::string& path, bitLenInt block
             std::fwrite(q16.data(), 2, reals, f);
         }
     }
+}
+
+constexpr char QAMD_TU_MAGIC[8] = { 'Q', 'A', 'M', 'D', 'T', 'U', '1', '\0' };
+
+// QUnit-aware container (reference QUNTQ parity, qunit_turboquant.cpp):
+// each Schmidt unit compresses separately — a product-heavy wide state
+// costs the SUM of its small units, not the dense 2^n blob.
+template <typename R> class QUnit;
+template <typename R>
+void LossySaveQUnit(std::shared_ptr<QUnit<R>> qu, FILE* f, bitLenInt blockBits, int bits,
+    bool rotate)
+{
+    qu->FlushAllForSerialize();
+    std::vector<QInterfacePtr<R>> units;
+    std::vector<std::pair<uint32_t, uint32_t>> qmap;
+    qu->GetUnitMap(units, qmap);
+    std::fwrite(QAMD_TU_MAGIC, 1, 8, f);
+    const uint64_t qb = qu->GetQubitCount(), nu = units.size();
+    std::fwrite(&qb, 8, 1, f);
+    std::fwrite(&nu, 8, 1, f);
+    for (const auto& m : qmap) {
+        std::fwrite(&m.first, 4, 1, f);
+        std::fwrite(&m.second, 4, 1, f);
+    }
+    for (const auto& u : units) {
+        const bitLenInt bb = std::min<bitLenInt>(blockBits, u->GetQubitCount());
+        LossySaveToStream<R>(u, f, bb, bits, rotate);
+    }
+}
+
+template <typename R> void LossyLoadFromStream(QInterfacePtr<R> q, FILE* f);
+
+template <typename R>
+void LossyLoadQUnit(std::shared_ptr<QUnit<R>> qu, FILE* f)
+{
+    char magic[8];
+    if (std::fread(magic, 1, 8, f) != 8 || std::memcmp(magic, QAMD_TU_MAGIC, 7) != 0)
+        throw QrackError("LossyLoadQUnit: bad magic");
+    uint64_t qb = 0, nu = 0;
+    (void)!std::fread(&qb, 8, 1, f);
+    (void)!std::fread(&nu, 8, 1, f);
+    if ((bitLenInt)qb != qu->GetQubitCount()) throw QrackError("LossyLoadQUnit: width mismatch");
+    std::vector<std::pair<uint32_t, uint32_t>> qmap(qb);
+    for (auto& m : qmap) {
+        (void)!std::fread(&m.first, 4, 1, f);
+        (void)!std::fread(&m.second, 4, 1, f);
+    }
+    // unit widths = (max mapped + 1) per unit index
+    std::vector<bitLenInt> widths(nu, 0);
+    for (const auto& m : qmap) {
+        widths[m.first] = std::max<bitLenInt>(widths[m.first], (bitLenInt)(m.second + 1u));
+    }
+    std::vector<QInterfacePtr<R>> units;
+    for (uint64_t u = 0; u < nu; ++u) {
+        QInterfacePtr<R> nuip = qu->NewUnit(widths[u]);
+        LossyLoadFromStream<R>(nuip, f);
+        units.push_back(nuip);
+    }
+    qu->RebuildFromUnits(units, qmap);
+}
+
+template <typename R>
+void LossySaveState(QInterfacePtr<R> q, const std::string& path, bitLenInt blockBits = 12,
+    int bits = 16, bool rotate = true, uint64_t seed = 0x51a4d70f2u)
+{
+    FILE* f = std::fopen(path.c_str(), "wb");
+    if (!f) throw QrackError("LossySaveState: cannot open " + path);
+    try {
+        if (auto qu = std::dynamic_pointer_cast<QUnit<R>>(q)) {
+            LossySaveQUnit<R>(qu, f, blockBits, bits, rotate);
+        } else {
+            LossySaveToStream<R>(q, f, blockBits, bits, rotate, seed);
+        }
+    } catch (...) {
+        std::fclose(f);
+        throw;
+    }
     std::fclose(f);
 }
 
-template <typename R> void LossyLoadState(QInterfacePtr<R> q, const std::string& path)
+template <typename R> void LossyLoadFromStream(QInterfacePtr<R> q, FILE* f)
 {
-    FILE* f = std::fopen(path.c_str(), "rb");
-    if (!f) throw QrackError("LossyLoadState: cannot open " + path);
     char magic[8];
     if (std::fread(magic, 1, 8, f) != 8 ||
         (std::memcmp(magic, QAMD_TQ_MAGIC, 7) != 0 && std::memcmp(magic, QAMD_TQ2_MAGIC, 7) != 0)) {
-        std::fclose(f);
         throw QrackError("LossyLoadState: bad magic");
     }
     const bool v2 = magic[6] == '2';
@@ -209,7 +282,6 @@ template <typename R> void LossyLoadState(QInterfacePtr<R> q, const std::string&
         (void)!std::fread(&seed, 8, 1, f);
     }
     if ((bitLenInt)qb != q->GetQubitCount()) {
-        std::fclose(f);
         throw QrackError("LossyLoadState: qubit count mismatch");
     }
     const bitCapInt blockLen = pow2((bitLenInt)bb);
@@ -257,9 +329,30 @@ template <typename R> void LossyLoadState(QInterfacePtr<R> q, const std::string&
             std::memcpy(dense.data() + off, buf.data(), sizeof(cplx<R>) * blockLen);
         }
     }
-    std::fclose(f);
     if (!eng) q->SetQuantumState(dense.data());
     q->NormalizeState();
+}
+
+template <typename R> void LossyLoadState(QInterfacePtr<R> q, const std::string& path)
+{
+    FILE* f = std::fopen(path.c_str(), "rb");
+    if (!f) throw QrackError("LossyLoadState: cannot open " + path);
+    try {
+        char peek[8] = {};
+        const size_t got = std::fread(peek, 1, 8, f);
+        std::fseek(f, 0, SEEK_SET);
+        if (got == 8 && std::memcmp(peek, QAMD_TU_MAGIC, 7) == 0) {
+            auto qu = std::dynamic_pointer_cast<QUnit<R>>(q);
+            if (!qu) throw QrackError("LossyLoadState: QUNIT container needs a qunit layer");
+            LossyLoadQUnit<R>(qu, f);
+        } else {
+            LossyLoadFromStream<R>(q, f);
+        }
+    } catch (...) {
+        std::fclose(f);
+        throw;
+    }
+    std::fclose(f);
 }
 
 } // namespace qrack_amd

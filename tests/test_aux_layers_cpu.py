@@ -300,3 +300,28 @@ def test_stabilizer_hybrid_save_with_shards(tmp_path):
     q3.h(0)
     q3.cnot(0, 1)
     assert "SHARDS" not in qa.save_stabilizer_F(q3)
+
+
+def test_qunit_aware_lossy_container(tmp_path):
+    """QUNTQ-parity container: each Schmidt unit compresses separately, so a
+    wide product-heavy state costs the sum of its units, not 2^n."""
+    import os as _os
+
+    n = 40  # dense would be 2^40 amplitudes: impossible; units are small
+    q = qa.create_simulator(n, layers=["qunit", "cpu"], seed=6)
+    rng = np.random.default_rng(4)
+    for i in range(n):
+        q.ry(float(rng.uniform(0, np.pi)), i)
+    for i in range(0, n - 1, 4):
+        q.cnot(i, i + 1)  # pairs entangle: units of width <= 2
+    p = str(tmp_path / "qu.bin")
+    qa.lossy_save_F(q, p, 12, 16, True)
+    assert _os.path.getsize(p) < (1 << 20)  # far below dense
+    q2 = qa.create_simulator(n, layers=["qunit", "cpu"], seed=7)
+    qa.lossy_load_F(q2, p)
+    for i in range(n):
+        assert abs(q2.prob(i) - q.prob(i)) < 1e-3
+    # correlations preserved within units
+    r1 = q.pauli_expectation([0, 1], [2, 2])
+    r2 = q2.pauli_expectation([0, 1], [2, 2])
+    assert abs(r1 - r2) < 1e-3
