@@ -97,9 +97,18 @@ def run_supremacy(q, n, rng, depth):
         mats = [_M1Q[sq[rng.integers(3)]] for _ in range(n)]
         _apply_1q_layer(q, range(n), mats)
         start = layer % 2
+        thetas, phis, q1s, q2s = [], [], [], []
         for i in range(start, n - 1, 2):
             th, ph = rng.uniform(0, 2 * np.pi, 2)
-            q.fsim(float(th), float(ph), i, i + 1)
+            thetas.append(float(th))
+            phis.append(float(ph))
+            q1s.append(i)
+            q2s.append(i + 1)
+        if hasattr(q, "fsim_batch"):
+            q.fsim_batch(thetas, phis, q1s, q2s)
+        else:
+            for th, ph, a, b in zip(thetas, phis, q1s, q2s):
+                q.fsim(th, ph, a, b)
 
 
 def run_qv(q, n, rng, depth):

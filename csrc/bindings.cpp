@@ -68,6 +68,9 @@ template <typename R> static void bindQInterface(py::module_& m, const char* nam
                     from_std<R>(m[3]) };
                 q.Mtrx(mm, t);
             })
+        .def("fsim_batch",
+            [](QI& q, std::vector<R> thetas, std::vector<R> phis, std::vector<bitLenInt> q1s,
+                std::vector<bitLenInt> q2s) { q.FSimBatch(thetas, phis, q1s, q2s); })
         .def("cphase_pairs",
             [](QI& q, std::vector<bitLenInt> c, std::vector<bitLenInt> t, std::vector<double> a) {
                 q.CPhasePairs(c, t, a);

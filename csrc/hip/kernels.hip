@@ -1306,6 +1306,55 @@ void launchMtrx1qBatchLds(cplx<R>* sv, const BatchLdsArgs<R>& a, hipStream_t str
     hipLaunchKernelGGL((k_mtrx_batch_lds<R>), dim3(grid), dim3(QA_BLOCK), 0, stream, sv, a);
 }
 
+// LDS-tiled two-qubit batch: each workgroup stages one tile, then applies
+// every 4x4 (pair bits inside the tile) at LDS speed — one global RMW pass
+// for a whole disjoint two-qubit layer segment.
+template <typename R> __global__ void k_mtrx2q_batch_lds(cplx<R>* sv, Batch2qLdsArgs<R> a)
+{
+    constexpr int TB = qaLdsTileBits<R>();
+    __shared__ cplx<R> tile[1u << TB];
+    constexpr unsigned TILE = 1u << TB;
+    const bitCapInt nTiles = a.maxQPower >> TB;
+    for (bitCapInt t = blockIdx.x; t < nTiles; t += gridDim.x) {
+        const bitCapInt base = t << TB;
+        for (unsigned i = threadIdx.x; i < TILE; i += blockDim.x) {
+            tile[i] = sv[base + i];
+        }
+        __syncthreads();
+        for (int g = 0; g < a.k; ++g) {
+            const unsigned lo = (unsigned)a.p1[g];
+            const unsigned hi = (unsigned)a.p2[g];
+            const cplx<R>* m = &a.m[16 * g];
+            for (unsigned j = threadIdx.x; j < (TILE >> 2); j += blockDim.x) {
+                // expand j around lo then hi (lo < hi)
+                unsigned i0 = ((j & ~(lo - 1u)) << 1u) | (j & (lo - 1u));
+                i0 = ((i0 & ~(hi - 1u)) << 1u) | (i0 & (hi - 1u));
+                const unsigned i1 = i0 | lo;
+                const unsigned i2 = i0 | hi;
+                const unsigned i3 = i0 | lo | hi;
+                const cplx<R> v0 = tile[i0], v1 = tile[i1], v2 = tile[i2], v3 = tile[i3];
+                tile[i0] = m[0] * v0 + m[1] * v1 + m[2] * v2 + m[3] * v3;
+                tile[i1] = m[4] * v0 + m[5] * v1 + m[6] * v2 + m[7] * v3;
+                tile[i2] = m[8] * v0 + m[9] * v1 + m[10] * v2 + m[11] * v3;
+                tile[i3] = m[12] * v0 + m[13] * v1 + m[14] * v2 + m[15] * v3;
+            }
+            __syncthreads();
+        }
+        for (unsigned i = threadIdx.x; i < TILE; i += blockDim.x) {
+            sv[base + i] = tile[i];
+        }
+        __syncthreads();
+    }
+}
+
+template <typename R>
+void launchMtrx2qBatchLds(cplx<R>* sv, const Batch2qLdsArgs<R>& a, hipStream_t stream)
+{
+    const bitCapInt nTiles = a.maxQPower >> qaLdsTileBits<R>();
+    const int grid = (int)std::min<bitCapInt>(nTiles, (bitCapInt)QA_MAX_BLOCKS);
+    hipLaunchKernelGGL((k_mtrx2q_batch_lds<R>), dim3(grid), dim3(QA_BLOCK), 0, stream, sv, a);
+}
+
 // ---- batched disjoint CNOTs: one permutation pass per layer ----------------
 
 template <typename R> __global__ void k_cnot_batch(cplx<R>* sv, CnotBatchArgs a)
@@ -1609,7 +1658,8 @@ void launchPartProbs(const cplx<R>* sv, bitCapInt maxQPower, bitLenInt start, bi
         cplx<R>*, bitCapInt, bitLenInt, bitLenInt, bitCapInt, int, bool, hipStream_t);              \
     template void launchMtrx1qBatch<R>(cplx<R>*, const Batch1qArgs<R>&, hipStream_t);                               \
     template void launchCnotBatch<R>(cplx<R>*, const CnotBatchArgs&, hipStream_t);                              \
-    template void launchMtrx1qBatchLds<R>(cplx<R>*, const BatchLdsArgs<R>&, hipStream_t);                              \
+    template void launchMtrx1qBatchLds<R>(cplx<R>*, const BatchLdsArgs<R>&, hipStream_t);                \
+    template void launchMtrx2qBatchLds<R>(cplx<R>*, const Batch2qLdsArgs<R>&, hipStream_t);                              \
     template void launchCPhasePairs<R>(cplx<R>*, const CPhasePairsArgs&, hipStream_t);                               \
     template void launchQftColumnGeneral<R>(                                                        \
         cplx<R>*, bitCapInt, bitCapInt, const RampArgs&, double, bool, hipStream_t);

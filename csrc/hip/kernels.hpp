@@ -220,6 +220,21 @@ template <typename R> struct BatchLdsArgs {
 template <typename R>
 void launchMtrx1qBatchLds(cplx<R>* sv, const BatchLdsArgs<R>& a, hipStream_t stream);
 
+// LDS-tiled batch of disjoint TWO-qubit 4x4 gates (all four qubit bits below
+// the tile): a whole fsim/SU(4) layer segment in ONE global pass.
+constexpr int QA_MAX_BATCH_2Q = 6;
+
+template <typename R> struct Batch2qLdsArgs {
+    cplx<R> m[16 * QA_MAX_BATCH_2Q]; // row-major 4x4 per pair, basis |q2 q1>
+    bitCapInt p1[QA_MAX_BATCH_2Q];   // pow2(q1) (low bit of the pair index)
+    bitCapInt p2[QA_MAX_BATCH_2Q];   // pow2(q2) (high bit)
+    int k;
+    bitCapInt maxQPower;
+};
+
+template <typename R>
+void launchMtrx2qBatchLds(cplx<R>* sv, const Batch2qLdsArgs<R>& a, hipStream_t stream);
+
 // batched disjoint CNOTs: a whole layer of k control/target pairs (no qubit
 // repeated) applied as ONE in-place permutation pass — amp[i] swaps with
 // amp[i ^ xm(i)] where xm(i) XORs tPow[j] for every set control bit. One

@@ -400,6 +400,63 @@ void QEngineHIP<R>::Apply2x2(bitCapInt offset1, bitCapInt offset2, const cplx<R>
     launchApply2x2<R>(dState, a, stream);
 }
 
+// batched disjoint fsim layer: pairs whose bits BOTH sit inside the LDS
+// tile fuse as 4x4s (up to 6 per single global pass); the rest apply via
+// the normal swap-block + one-sided-phase path.
+template <typename R>
+void QEngineHIP<R>::FSimBatch(const std::vector<R>& thetas, const std::vector<R>& phis,
+    const std::vector<bitLenInt>& q1s, const std::vector<bitLenInt>& q2s)
+{
+    if (thetas.size() != phis.size() || q1s.size() != q2s.size() || thetas.size() != q1s.size())
+        throw QrackError("FSimBatch: need (theta, phi, q1, q2) per gate");
+    const bitLenInt ldsBits = (bitLenInt)qaLdsTileBits<R>();
+    std::set<bitLenInt> uniq;
+    bool disjoint = true;
+    for (size_t i = 0; i < q1s.size(); ++i) {
+        if (!uniq.insert(q1s[i]).second) disjoint = false;
+        if (!uniq.insert(q2s[i]).second) disjoint = false;
+    }
+    std::vector<size_t> low, rest;
+    for (size_t i = 0; i < q1s.size(); ++i) {
+        if (disjoint && qubitCount > ldsBits && q1s[i] < ldsBits && q2s[i] < ldsBits) {
+            low.push_back(i);
+        } else {
+            rest.push_back(i);
+        }
+    }
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    for (size_t i = 0; i < low.size();) {
+        const size_t k = std::min((size_t)QA_MAX_BATCH_2Q, low.size() - i);
+        Batch2qLdsArgs<R> a{};
+        for (size_t g = 0; g < k; ++g) {
+            const size_t ix = low[i + g];
+            bitLenInt qa_ = q1s[ix], qb_ = q2s[ix];
+            if (qa_ > qb_) std::swap(qa_, qb_);
+            a.p1[g] = pow2(qa_);
+            a.p2[g] = pow2(qb_);
+            // fsim(theta, phi): |01>,|10> mix by [[c,-is],[-is,c]];
+            // |11> phase e^{-i phi} (basis |q2 q1| = i2 i1)
+            const R ct = std::cos(thetas[ix]), st = std::sin(thetas[ix]);
+            cplx<R>* m = &a.m[16 * g];
+            for (int e = 0; e < 16; ++e) m[e] = cplx<R>(0, 0);
+            m[0] = cplx<R>(1, 0);
+            m[5] = cplx<R>(ct, 0);
+            m[6] = cplx<R>(0, -st);
+            m[9] = cplx<R>(0, -st);
+            m[10] = cplx<R>(ct, 0);
+            m[15] = polar<R>(1, -phis[ix]);
+        }
+        a.k = (int)k;
+        a.maxQPower = maxQPower;
+        HipProfScope prof("fsim_batch_lds", stream);
+        launchMtrx2qBatchLds<R>(dState, a, stream);
+        i += k;
+    }
+    for (size_t ix : rest) {
+        this->FSim(thetas[ix], phis[ix], q1s[ix], q2s[ix]);
+    }
+}
+
 // batched controlled-phase pairs: one diagonal pass per layer.
 template <typename R>
 void QEngineHIP<R>::CPhasePairs(const std::vector<bitLenInt>& controls,

@@ -121,6 +121,8 @@ public:
         const std::vector<bitLenInt>& controls, const std::vector<bitLenInt>& targets) override;
     void CPhasePairs(const std::vector<bitLenInt>& controls, const std::vector<bitLenInt>& targets,
         const std::vector<double>& angles) override;
+    void FSimBatch(const std::vector<R>& thetas, const std::vector<R>& phis,
+        const std::vector<bitLenInt>& q1s, const std::vector<bitLenInt>& q2s) override;
     void ApplyM(bitCapInt regMask, bitCapInt result, cplx<R> nrm) override;
     void GetAmplitudePage(cplx<R>* pagePtr, bitCapInt offset, bitCapInt length) override;
     void SetAmplitudePage(const cplx<R>* pagePtr, bitCapInt offset, bitCapInt length) override;

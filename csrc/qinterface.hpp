@@ -143,6 +143,15 @@ public:
     virtual void SqrtSwap(bitLenInt q1, bitLenInt q2);
     virtual void ISqrtSwap(bitLenInt q1, bitLenInt q2);
     virtual void FSim(R theta, R phi, bitLenInt q1, bitLenInt q2);
+    // batched disjoint fsim layer: engines fuse in-LDS-tile pairs into one
+    // pass; default lowering applies them one by one
+    virtual void FSimBatch(const std::vector<R>& thetas, const std::vector<R>& phis,
+        const std::vector<bitLenInt>& q1s, const std::vector<bitLenInt>& q2s)
+    {
+        if (thetas.size() != phis.size() || q1s.size() != q2s.size() || thetas.size() != q1s.size())
+            throw QrackError("FSimBatch: need (theta, phi, q1, q2) per gate");
+        for (size_t i = 0; i < thetas.size(); ++i) FSim(thetas[i], phis[i], q1s[i], q2s[i]);
+    }
     virtual void CSwap(const std::vector<bitLenInt>& controls, bitLenInt q1, bitLenInt q2);
     virtual void AntiCSwap(const std::vector<bitLenInt>& controls, bitLenInt q1, bitLenInt q2);
     virtual void CSqrtSwap(const std::vector<bitLenInt>& controls, bitLenInt q1, bitLenInt q2);

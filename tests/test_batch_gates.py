@@ -148,3 +148,19 @@ def test_cz_batch_graph_state():
     for i in range(n - 1):
         q2.cz(i, i + 1)
     assert_states_close(q1.get_state_vector(), q2.get_state_vector(), 1e-5)
+
+
+def test_fsim_batch_cpu_default():
+    # the default lowering applies gates sequentially: equality is exact
+    n = 6
+    q1 = qa.create_simulator(n, engine="cpu", seed=8)
+    q2 = qa.create_simulator(n, engine="cpu", seed=8)
+    for i in range(n):
+        q1.h(i)
+        q2.h(i)
+    thetas, phis = [0.4, 1.2, 0.9], [0.7, 0.2, 1.5]
+    a, b = [0, 2, 4], [1, 3, 5]
+    q1.fsim_batch(thetas, phis, a, b)
+    for th, ph, x, y in zip(thetas, phis, a, b):
+        q2.fsim(th, ph, x, y)
+    assert_states_close(q1.get_state_vector(), q2.get_state_vector(), 1e-5)

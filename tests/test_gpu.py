@@ -424,3 +424,21 @@ def test_cphase_pairs_gpu():
     for c, t, a in zip(controls, targets, angles):
         qs.mcphase([c], 1, complex(np.exp(1j * a)), t)
     assert float(qb.sum_sqr_diff(qs)) < 1e-5
+
+
+def test_fsim_batch_gpu():
+    n = 20
+    qb = qa.create_simulator(n, engine="hip", seed=9)
+    qs = qa.create_simulator(n, engine="hip", seed=9)
+    for i in range(n):
+        qb.h(i)
+        qs.h(i)
+    # mix of in-tile (low) and high pairs
+    thetas = [0.4, 1.2, 0.9, 0.3]
+    phis = [0.7, 0.2, 1.5, 2.0]
+    a = [0, 4, 8, 14]
+    b = [1, 5, 9, 17]
+    qb.fsim_batch(thetas, phis, a, b)
+    for th, ph, x, y in zip(thetas, phis, a, b):
+        qs.fsim(th, ph, x, y)
+    assert float(qb.sum_sqr_diff(qs)) < 1e-5
