@@ -68,6 +68,13 @@ template <typename R> static void bindQInterface(py::module_& m, const char* nam
                     from_std<R>(m[3]) };
                 q.Mtrx(mm, t);
             })
+        .def("mtrx_2q",
+            [](QI& q, std::vector<C> m, bitLenInt a, bitLenInt b) {
+                if (m.size() != 16) throw QrackError("mtrx_2q: need 16 entries");
+                std::vector<cplx<R>> mm(16);
+                for (int i = 0; i < 16; ++i) mm[i] = from_std<R>(m[i]);
+                q.Mtrx2q(mm.data(), a, b);
+            })
         .def("fsim_batch",
             [](QI& q, std::vector<R> thetas, std::vector<R> phis, std::vector<bitLenInt> q1s,
                 std::vector<bitLenInt> q2s) { q.FSimBatch(thetas, phis, q1s, q2s); })

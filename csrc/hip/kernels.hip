@@ -1355,6 +1355,29 @@ void launchMtrx2qBatchLds(cplx<R>* sv, const Batch2qLdsArgs<R>& a, hipStream_t s
     hipLaunchKernelGGL((k_mtrx2q_batch_lds<R>), dim3(grid), dim3(QA_BLOCK), 0, stream, sv, a);
 }
 
+template <typename R> __global__ void k_mtrx_2q(cplx<R>* sv, Gate4x4Args<R> a)
+{
+    const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
+    for (bitCapInt j = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; j < a.maxI; j += stride) {
+        bitCapInt i0 = ((j & ~(a.p1 - 1u)) << 1u) | (j & (a.p1 - 1u));
+        i0 = ((i0 & ~(a.p2 - 1u)) << 1u) | (i0 & (a.p2 - 1u));
+        const bitCapInt i1 = i0 | a.p1;
+        const bitCapInt i2 = i0 | a.p2;
+        const bitCapInt i3 = i0 | a.p1 | a.p2;
+        const cplx<R> v0 = sv[i0], v1 = sv[i1], v2 = sv[i2], v3 = sv[i3];
+        sv[i0] = a.m[0] * v0 + a.m[1] * v1 + a.m[2] * v2 + a.m[3] * v3;
+        sv[i1] = a.m[4] * v0 + a.m[5] * v1 + a.m[6] * v2 + a.m[7] * v3;
+        sv[i2] = a.m[8] * v0 + a.m[9] * v1 + a.m[10] * v2 + a.m[11] * v3;
+        sv[i3] = a.m[12] * v0 + a.m[13] * v1 + a.m[14] * v2 + a.m[15] * v3;
+    }
+}
+
+template <typename R>
+void launchMtrx2q(cplx<R>* sv, const Gate4x4Args<R>& a, hipStream_t stream)
+{
+    hipLaunchKernelGGL((k_mtrx_2q<R>), dim3(gridFor(a.maxI)), dim3(QA_BLOCK), 0, stream, sv, a);
+}
+
 // ---- batched disjoint CNOTs: one permutation pass per layer ----------------
 
 template <typename R> __global__ void k_cnot_batch(cplx<R>* sv, CnotBatchArgs a)
@@ -1659,7 +1682,8 @@ void launchPartProbs(const cplx<R>* sv, bitCapInt maxQPower, bitLenInt start, bi
     template void launchMtrx1qBatch<R>(cplx<R>*, const Batch1qArgs<R>&, hipStream_t);                               \
     template void launchCnotBatch<R>(cplx<R>*, const CnotBatchArgs&, hipStream_t);                              \
     template void launchMtrx1qBatchLds<R>(cplx<R>*, const BatchLdsArgs<R>&, hipStream_t);                \
-    template void launchMtrx2qBatchLds<R>(cplx<R>*, const Batch2qLdsArgs<R>&, hipStream_t);                              \
+    template void launchMtrx2qBatchLds<R>(cplx<R>*, const Batch2qLdsArgs<R>&, hipStream_t);                \
+    template void launchMtrx2q<R>(cplx<R>*, const Gate4x4Args<R>&, hipStream_t);                              \
     template void launchCPhasePairs<R>(cplx<R>*, const CPhasePairsArgs&, hipStream_t);                               \
     template void launchQftColumnGeneral<R>(                                                        \
         cplx<R>*, bitCapInt, bitCapInt, const RampArgs&, double, bool, hipStream_t);

@@ -235,6 +235,19 @@ template <typename R> struct Batch2qLdsArgs {
 template <typename R>
 void launchMtrx2qBatchLds(cplx<R>* sv, const Batch2qLdsArgs<R>& a, hipStream_t stream);
 
+// general global two-qubit 4x4 apply (register orbit of 4 amplitudes; any
+// qubit positions) — the reference decomposes SU(4) into gate strings, this
+// build applies it in ONE pass.
+template <typename R> struct Gate4x4Args {
+    cplx<R> m[16]; // row-major, basis |q2 q1>
+    bitCapInt p1;  // pow2(min qubit)
+    bitCapInt p2;  // pow2(max qubit)
+    bitCapInt maxI; // maxQPower >> 2
+};
+
+template <typename R>
+void launchMtrx2q(cplx<R>* sv, const Gate4x4Args<R>& a, hipStream_t stream);
+
 // batched disjoint CNOTs: a whole layer of k control/target pairs (no qubit
 // repeated) applied as ONE in-place permutation pass — amp[i] swaps with
 // amp[i ^ xm(i)] where xm(i) XORs tPow[j] for every set control bit. One

@@ -400,6 +400,29 @@ void QEngineHIP<R>::Apply2x2(bitCapInt offset1, bitCapInt offset2, const cplx<R>
     launchApply2x2<R>(dState, a, stream);
 }
 
+// general two-qubit 4x4 apply: one pass (k_mtrx_2q).
+template <typename R>
+void QEngineHIP<R>::Mtrx2q(const cplx<R>* m, bitLenInt q1, bitLenInt q2)
+{
+    if (q1 == q2 || q1 >= qubitCount || q2 >= qubitCount)
+        throw QrackError("Mtrx2q: bad qubit indices");
+    Gate4x4Args<R> a{};
+    a.p1 = pow2(std::min(q1, q2));
+    a.p2 = pow2(std::max(q1, q2));
+    if (q1 < q2) {
+        std::copy(m, m + 16, a.m);
+    } else {
+        static const int permIdx[4] = { 0, 2, 1, 3 };
+        for (int r = 0; r < 4; ++r) {
+            for (int cc = 0; cc < 4; ++cc) a.m[4 * r + cc] = m[4 * permIdx[r] + permIdx[cc]];
+        }
+    }
+    a.maxI = maxQPower >> 2u;
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    HipProfScope prof("mtrx_2q", stream);
+    launchMtrx2q<R>(dState, a, stream);
+}
+
 // batched disjoint fsim layer: pairs whose bits BOTH sit inside the LDS
 // tile fuse as 4x4s (up to 6 per single global pass); the rest apply via
 // the normal swap-block + one-sided-phase path.

@@ -117,6 +117,39 @@ template <typename R> void QEngineCPU<R>::ShuffleBuffers(QEnginePtr<R> engine)
 
 // ---- gate primitives -------------------------------------------------------
 
+// general two-qubit 4x4 apply (orbit of 4 amplitudes per iteration)
+template <typename R>
+void QEngineCPU<R>::Mtrx2q(const cplx<R>* m, bitLenInt q1, bitLenInt q2)
+{
+    if (q1 == q2 || q1 >= qubitCount || q2 >= qubitCount)
+        throw QrackError("Mtrx2q: bad qubit indices");
+    const bitCapInt p1 = pow2(std::min(q1, q2));
+    const bitCapInt p2 = pow2(std::max(q1, q2));
+    // caller convention: m is in basis |q2 q1>; after sorting the powers the
+    // low power is min(q1,q2) — if q1 > q2, swap the middle rows/columns
+    cplx<R> mm[16];
+    if (q1 < q2) {
+        std::copy(m, m + 16, mm);
+    } else {
+        static const int permIdx[4] = { 0, 2, 1, 3 };
+        for (int r = 0; r < 4; ++r) {
+            for (int cc = 0; cc < 4; ++cc) mm[4 * r + cc] = m[4 * permIdx[r] + permIdx[cc]];
+        }
+    }
+    cplx<R>* sv = stateVec.data();
+    std::vector<bitCapInt> pows{ p1, p2 };
+    this->par_for_mask(maxQPower >> 2u, pows, [sv, mm, p1, p2](const bitCapInt& i0, unsigned) {
+        const bitCapInt i1 = i0 | p1;
+        const bitCapInt i2 = i0 | p2;
+        const bitCapInt i3 = i0 | p1 | p2;
+        const cplx<R> v0 = sv[i0], v1 = sv[i1], v2 = sv[i2], v3 = sv[i3];
+        sv[i0] = mm[0] * v0 + mm[1] * v1 + mm[2] * v2 + mm[3] * v3;
+        sv[i1] = mm[4] * v0 + mm[5] * v1 + mm[6] * v2 + mm[7] * v3;
+        sv[i2] = mm[8] * v0 + mm[9] * v1 + mm[10] * v2 + mm[11] * v3;
+        sv[i3] = mm[12] * v0 + mm[13] * v1 + mm[14] * v2 + mm[15] * v3;
+    });
+}
+
 // batched controlled-phase pairs: one diagonal pass for a whole layer
 template <typename R>
 void QEngineCPU<R>::CPhasePairs(const std::vector<bitLenInt>& controls,

@@ -64,6 +64,7 @@ public:
         const std::vector<bitLenInt>& controls, const std::vector<bitLenInt>& targets) override;
     void CPhasePairs(const std::vector<bitLenInt>& controls, const std::vector<bitLenInt>& targets,
         const std::vector<double>& angles) override;
+    void Mtrx2q(const cplx<R>* m16, bitLenInt q1, bitLenInt q2) override;
     void QftColumnGeneral(bitLenInt target, double scale, bitLenInt rampStart,
         bitCapInt inPlaceRelMask, const std::vector<bitCapInt>& sPows,
         const std::vector<uint64_t>& sWeights, double phase0, bool pre) override;

@@ -143,6 +143,17 @@ public:
     virtual void SqrtSwap(bitLenInt q1, bitLenInt q2);
     virtual void ISqrtSwap(bitLenInt q1, bitLenInt q2);
     virtual void FSim(R theta, R phi, bitLenInt q1, bitLenInt q2);
+    // general two-qubit 4x4 gate in basis |q2 q1> (row-major m16): engines
+    // apply it in ONE pass — the reference decomposes SU(4) into gate
+    // strings instead. Layers forward to their sub-state.
+    virtual void Mtrx2q(const cplx<R>* m16, bitLenInt q1, bitLenInt q2)
+    {
+        (void)m16;
+        (void)q1;
+        (void)q2;
+        throw QrackError("Mtrx2q: not supported on this layer (use an engine/qunit stack)");
+    }
+
     // batched disjoint fsim layer: engines fuse in-LDS-tile pairs into one
     // pass; default lowering applies them one by one
     virtual void FSimBatch(const std::vector<R>& thetas, const std::vector<R>& phis,

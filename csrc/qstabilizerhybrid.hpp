@@ -119,6 +119,11 @@ public:
         bitCapInt controlPerm) override;
     void UniformlyControlledSingleBit(
         const std::vector<bitLenInt>& controls, bitLenInt target, const cplx<R>* mtrxs) override;
+    void Mtrx2q(const cplx<R>* m16, bitLenInt q1, bitLenInt q2) override
+    {
+        SwitchToEngine();
+        engine->Mtrx2q(m16, q1, q2);
+    }
     void Swap(bitLenInt q1, bitLenInt q2) override;
     void ISwap(bitLenInt q1, bitLenInt q2) override;
     void IISwap(bitLenInt q1, bitLenInt q2) override;

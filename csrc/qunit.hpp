@@ -250,6 +250,15 @@ public:
 
     // ---- gates ----
     void Mtrx(const cplx<R>* mtrx, bitLenInt target) override;
+    void Mtrx2q(const cplx<R>* m16, bitLenInt q1, bitLenInt q2) override
+    {
+        FlushPhasePairs(q1);
+        FlushPhasePairs(q2);
+        QInterfacePtr<R> unit = EntangleAll({ q1, q2 });
+        unit->Mtrx2q(m16, shards[q1].mapped, shards[q2].mapped);
+        MaybeSeparate(q1);
+        MaybeSeparate(q2);
+    }
     // group the batch by unit and forward fused sub-batches: a layer of 1q
     // gates on a merged unit costs ceil(k/4) passes instead of k
     void Mtrx1qBatch(

@@ -94,6 +94,10 @@ public:
     void SqrtSwap(bitLenInt a, bitLenInt b) override { inner->SqrtSwap(a, b); }
     void ISqrtSwap(bitLenInt a, bitLenInt b) override { inner->ISqrtSwap(a, b); }
     void FSim(R th, R ph, bitLenInt a, bitLenInt b) override { inner->FSim(th, ph, a, b); }
+    void Mtrx2q(const cplx<R>* m16, bitLenInt a, bitLenInt b) override
+    {
+        inner->Mtrx2q(m16, a, b);
+    }
     void CSwap(const std::vector<bitLenInt>& c, bitLenInt a, bitLenInt b) override
     {
         inner->CSwap(c, a, b);

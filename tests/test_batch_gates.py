@@ -164,3 +164,36 @@ def test_fsim_batch_cpu_default():
     for th, ph, x, y in zip(thetas, phis, a, b):
         q2.fsim(th, ph, x, y)
     assert_states_close(q1.get_state_vector(), q2.get_state_vector(), 1e-5)
+
+
+def _rand_u4(rng):
+    # Haar-ish random 4x4 unitary via QR
+    z = rng.normal(size=(4, 4)) + 1j * rng.normal(size=(4, 4))
+    qm, r = np.linalg.qr(z)
+    return qm * (np.diag(r) / np.abs(np.diag(r)))
+
+
+@pytest.mark.parametrize("layers", [["cpu"], ["qunit", "cpu"], ["stabilizer_hybrid", "cpu"]])
+@pytest.mark.parametrize("pair", [(1, 3), (3, 1), (0, 2)])
+def test_mtrx_2q_vs_numpy(layers, pair):
+    n = 4
+    rng = np.random.default_rng(61)
+    u = _rand_u4(rng)
+    q = qa.create_simulator(n, layers=layers, precision="fp64", seed=5)
+    cp = qa.create_simulator(n, engine="cpu", precision="fp64", seed=5)
+    for s in (q, cp):
+        for i in range(n):
+            s.ry(0.3 + 0.2 * i, i)
+    q.mtrx_2q([complex(x) for x in u.flatten()], pair[0], pair[1])
+    # numpy reference: basis |q2 q1> on (pair[0]=q1, pair[1]=q2)
+    sv = np.asarray(cp.get_state_vector()).astype(np.complex128)
+    q1b, q2b = pair
+    out = sv.copy()
+    for i0 in range(1 << n):
+        if (i0 >> q1b) & 1 or (i0 >> q2b) & 1:
+            continue
+        idx = [i0, i0 | (1 << q1b), i0 | (1 << q2b), i0 | (1 << q1b) | (1 << q2b)]
+        vec = sv[idx]
+        out[idx] = u @ vec
+    got = np.asarray(q.get_state_vector()).astype(np.complex128)
+    assert np.abs(got - out).max() < 1e-9

@@ -442,3 +442,23 @@ def test_fsim_batch_gpu():
     for th, ph, x, y in zip(thetas, phis, a, b):
         qs.fsim(th, ph, x, y)
     assert float(qb.sum_sqr_diff(qs)) < 1e-5
+
+
+def test_mtrx_2q_gpu():
+    n = 20
+    rng = np.random.default_rng(71)
+    z = rng.normal(size=(4, 4)) + 1j * rng.normal(size=(4, 4))
+    qm, r = np.linalg.qr(z)
+    u = qm * (np.diag(r) / np.abs(np.diag(r)))
+    flat = [complex(x) for x in u.flatten()]
+    for pair in ((3, 15), (15, 3), (0, 1)):
+        qh = qa.create_simulator(n, engine="hip", seed=5)
+        qc = qa.create_simulator(n, engine="cpu", seed=5)
+        for s in (qh, qc):
+            for i in range(0, n, 2):
+                s.h(i)
+        qh.mtrx_2q(flat, pair[0], pair[1])
+        qc.mtrx_2q(flat, pair[0], pair[1])
+        svh = np.asarray(qh.get_state_vector())
+        svc = np.asarray(qc.get_state_vector())
+        assert np.abs(svh - svc).max() < 1e-5
