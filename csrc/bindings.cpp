@@ -75,6 +75,12 @@ template <typename R> static void bindQInterface(py::module_& m, const char* nam
                 for (int i = 0; i < 16; ++i) mm[i] = from_std<R>(m[i]);
                 q.Mtrx2q(mm.data(), a, b);
             })
+        .def("mtrx_2q_batch",
+            [](QI& q, std::vector<C> ms, std::vector<bitLenInt> q1s, std::vector<bitLenInt> q2s) {
+                std::vector<cplx<R>> mm(ms.size());
+                for (size_t i = 0; i < ms.size(); ++i) mm[i] = from_std<R>(ms[i]);
+                q.Mtrx2qBatch(mm, q1s, q2s);
+            })
         .def("fsim_batch",
             [](QI& q, std::vector<R> thetas, std::vector<R> phis, std::vector<bitLenInt> q1s,
                 std::vector<bitLenInt> q2s) { q.FSimBatch(thetas, phis, q1s, q2s); })

@@ -423,6 +423,62 @@ void QEngineHIP<R>::Mtrx2q(const cplx<R>* m, bitLenInt q1, bitLenInt q2)
     launchMtrx2q<R>(dState, a, stream);
 }
 
+// batched disjoint two-qubit 4x4 layer: in-tile pairs fuse through the LDS
+// kernel; the rest apply as single-pass 4x4s.
+template <typename R>
+void QEngineHIP<R>::Mtrx2qBatch(const std::vector<cplx<R>>& ms,
+    const std::vector<bitLenInt>& q1s, const std::vector<bitLenInt>& q2s)
+{
+    if (q1s.size() != q2s.size() || ms.size() != 16u * q1s.size())
+        throw QrackError("Mtrx2qBatch: need a 4x4 per pair");
+    const bitLenInt ldsBits = (bitLenInt)qaLdsTileBits<R>();
+    std::set<bitLenInt> uniq;
+    bool disjoint = true;
+    for (size_t i = 0; i < q1s.size(); ++i) {
+        if (!uniq.insert(q1s[i]).second) disjoint = false;
+        if (!uniq.insert(q2s[i]).second) disjoint = false;
+    }
+    std::vector<size_t> low, rest;
+    for (size_t i = 0; i < q1s.size(); ++i) {
+        if (disjoint && qubitCount > ldsBits && q1s[i] < ldsBits && q2s[i] < ldsBits &&
+            q1s[i] != q2s[i]) {
+            low.push_back(i);
+        } else {
+            rest.push_back(i);
+        }
+    }
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    auto permuted = [&](const cplx<R>* m, bool swapped, cplx<R>* out) {
+        if (!swapped) {
+            std::copy(m, m + 16, out);
+            return;
+        }
+        static const int permIdx[4] = { 0, 2, 1, 3 };
+        for (int r = 0; r < 4; ++r) {
+            for (int cc = 0; cc < 4; ++cc) out[4 * r + cc] = m[4 * permIdx[r] + permIdx[cc]];
+        }
+    };
+    for (size_t i = 0; i < low.size();) {
+        const size_t k = std::min((size_t)QA_MAX_BATCH_2Q, low.size() - i);
+        Batch2qLdsArgs<R> a{};
+        for (size_t g = 0; g < k; ++g) {
+            const size_t ix = low[i + g];
+            const bool swapped = q1s[ix] > q2s[ix];
+            a.p1[g] = pow2(std::min(q1s[ix], q2s[ix]));
+            a.p2[g] = pow2(std::max(q1s[ix], q2s[ix]));
+            permuted(&ms[16u * ix], swapped, &a.m[16 * g]);
+        }
+        a.k = (int)k;
+        a.maxQPower = maxQPower;
+        HipProfScope prof("mtrx_2q_batch_lds", stream);
+        launchMtrx2qBatchLds<R>(dState, a, stream);
+        i += k;
+    }
+    for (size_t ix : rest) {
+        this->Mtrx2q(&ms[16u * ix], q1s[ix], q2s[ix]);
+    }
+}
+
 // batched disjoint fsim layer: pairs whose bits BOTH sit inside the LDS
 // tile fuse as 4x4s (up to 6 per single global pass); the rest apply via
 // the normal swap-block + one-sided-phase path.

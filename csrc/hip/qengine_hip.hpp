@@ -124,6 +124,8 @@ public:
     void FSimBatch(const std::vector<R>& thetas, const std::vector<R>& phis,
         const std::vector<bitLenInt>& q1s, const std::vector<bitLenInt>& q2s) override;
     void Mtrx2q(const cplx<R>* m16, bitLenInt q1, bitLenInt q2) override;
+    void Mtrx2qBatch(const std::vector<cplx<R>>& ms, const std::vector<bitLenInt>& q1s,
+        const std::vector<bitLenInt>& q2s) override;
     void ApplyM(bitCapInt regMask, bitCapInt result, cplx<R> nrm) override;
     void GetAmplitudePage(cplx<R>* pagePtr, bitCapInt offset, bitCapInt length) override;
     void SetAmplitudePage(const cplx<R>* pagePtr, bitCapInt offset, bitCapInt length) override;

@@ -154,6 +154,16 @@ public:
         throw QrackError("Mtrx2q: not supported on this layer (use an engine/qunit stack)");
     }
 
+    // batched disjoint two-qubit 4x4 layer: engines fuse in-LDS-tile pairs
+    // into one pass and apply the rest as single-pass 4x4s
+    virtual void Mtrx2qBatch(const std::vector<cplx<R>>& ms, const std::vector<bitLenInt>& q1s,
+        const std::vector<bitLenInt>& q2s)
+    {
+        if (q1s.size() != q2s.size() || ms.size() != 16u * q1s.size())
+            throw QrackError("Mtrx2qBatch: need a 4x4 per pair");
+        for (size_t i = 0; i < q1s.size(); ++i) Mtrx2q(&ms[16u * i], q1s[i], q2s[i]);
+    }
+
     // batched disjoint fsim layer: engines fuse in-LDS-tile pairs into one
     // pass; default lowering applies them one by one
     virtual void FSimBatch(const std::vector<R>& thetas, const std::vector<R>& phis,

@@ -197,3 +197,20 @@ def test_mtrx_2q_vs_numpy(layers, pair):
         out[idx] = u @ vec
     got = np.asarray(q.get_state_vector()).astype(np.complex128)
     assert np.abs(got - out).max() < 1e-9
+
+
+def test_mtrx_2q_batch_cpu_default():
+    n = 6
+    rng = np.random.default_rng(81)
+    us = [_rand_u4(rng) for _ in range(3)]
+    q1s, q2s = [0, 2, 5], [1, 3, 4]
+    flat = [complex(x) for u in us for x in u.flatten()]
+    qb = qa.create_simulator(n, engine="cpu", precision="fp64", seed=6)
+    qs = qa.create_simulator(n, engine="cpu", precision="fp64", seed=6)
+    for s in (qb, qs):
+        for i in range(n):
+            s.ry(0.2 + 0.1 * i, i)
+    qb.mtrx_2q_batch(flat, q1s, q2s)
+    for u, a, b in zip(us, q1s, q2s):
+        qs.mtrx_2q([complex(x) for x in u.flatten()], a, b)
+    assert_states_close(qb.get_state_vector(), qs.get_state_vector(), 1e-9)

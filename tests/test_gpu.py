@@ -462,3 +462,24 @@ def test_mtrx_2q_gpu():
         svh = np.asarray(qh.get_state_vector())
         svc = np.asarray(qc.get_state_vector())
         assert np.abs(svh - svc).max() < 1e-5
+
+
+def test_mtrx_2q_batch_gpu():
+    n = 20
+    rng = np.random.default_rng(91)
+    def u4():
+        z = rng.normal(size=(4, 4)) + 1j * rng.normal(size=(4, 4))
+        qm, r = np.linalg.qr(z)
+        return qm * (np.diag(r) / np.abs(np.diag(r)))
+    us = [u4() for _ in range(4)]
+    q1s, q2s = [0, 5, 9, 14], [3, 6, 2, 18]  # mix of in-tile / high / swapped
+    flat = [complex(x) for u in us for x in u.flatten()]
+    qb = qa.create_simulator(n, engine="hip", seed=7)
+    qs = qa.create_simulator(n, engine="hip", seed=7)
+    for s in (qb, qs):
+        for i in range(0, n, 3):
+            s.h(i)
+    qb.mtrx_2q_batch(flat, q1s, q2s)
+    for u, a, b in zip(us, q1s, q2s):
+        qs.mtrx_2q([complex(x) for x in u.flatten()], a, b)
+    assert float(qb.sum_sqr_diff(qs)) < 1e-5
