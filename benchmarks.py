@@ -111,23 +111,36 @@ def run_supremacy(q, n, rng, depth):
                 q.fsim(th, ph, a, b)
 
 
+def _haar_su4(rng):
+    z = rng.normal(size=(4, 4)) + 1j * rng.normal(size=(4, 4))
+    qm, r = np.linalg.qr(z)
+    return qm * (np.diag(r) / np.abs(np.diag(r)))
+
+
 def run_qv(q, n, rng, depth):
+    """Quantum volume, faithful protocol: depth layers of Haar-random
+    SU(4) blocks on a random qubit pairing (2018 Cross et al.). Engines and
+    qunit-family stacks apply each block as ONE 4x4 (Mtrx2qBatch, in-LDS-tile
+    pairs fused); other stacks fall back to a u-u-CNOT-u-u approximation.
+    """
     q.set_permutation(0)
     for _ in range(depth or n):
         perm = rng.permutation(n)
-        # the layer's 1q rotations act on disjoint pairs: fuse them all into
-        # batched passes, then apply the pair CNOTs (commuting reorder)
-        targets, mats = [], []
-        pairs = []
-        for k in range(0, n - 1, 2):
-            a, b = int(perm[k]), int(perm[k + 1])
-            pairs.append((a, b))
-            for t in (a, b):
-                th, ph, lm = rng.uniform(0, 2 * np.pi, 3)
-                targets.append(t)
-                mats.append(_u_mtrx(float(th), float(ph), float(lm)))
-        _apply_1q_layer(q, targets, mats)
-        _apply_cnot_layer(q, [a for a, _ in pairs], [b for _, b in pairs])
+        pairs = [(int(perm[k]), int(perm[k + 1])) for k in range(0, n - 1, 2)]
+        us = [_haar_su4(rng) for _ in pairs]
+        try:
+            flat = [complex(x) for u in us for x in u.flatten()]
+            q.mtrx_2q_batch(flat, [a for a, _ in pairs], [b for _, b in pairs])
+        except (AttributeError, RuntimeError):
+            # stacks without a native 4x4: SU(4)-ish approximation
+            targets, mats = [], []
+            for a, b in pairs:
+                for t in (a, b):
+                    th, ph, lm = rng.uniform(0, 2 * np.pi, 3)
+                    targets.append(t)
+                    mats.append(_u_mtrx(float(th), float(ph), float(lm)))
+            _apply_1q_layer(q, targets, mats)
+            _apply_cnot_layer(q, [a for a, _ in pairs], [b for _, b in pairs])
 
 
 WORKLOADS = {
