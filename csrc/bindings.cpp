@@ -312,6 +312,9 @@ template <typename R> static void bindQInterface(py::module_& m, const char* nam
         .def("clone", [](QI& q) { return q.Clone(); })
         .def("try_separate_1", [](QI& q, bitLenInt qb) { return q.TrySeparate(qb); })
         .def("try_separate_2", [](QI& q, bitLenInt q1, bitLenInt q2) { return q.TrySeparate(q1, q2); })
+        .def("try_separate", [](QI& q, bitLenInt x) { return q.TrySeparate(x); })
+        .def("try_separate",
+            [](QI& q, bitLenInt x, bitLenInt y) { return q.TrySeparate(x, y); })
         .def("try_separate",
             [](QI& q, std::vector<bitLenInt> qs, R tol) { return q.TrySeparate(qs, tol); })
         // ---- probability / measurement ----

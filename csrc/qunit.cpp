@@ -677,6 +677,33 @@ template <typename R> bool QUnit<R>::TrySeparate(bitLenInt q)
     Shard& s = shards[q];
     QInterfacePtr<R> unit = s.unit;
     if (unit->GetQubitCount() == 1u) return true;
+    // Clifford units answer separability EXACTLY from the tableau (the
+    // QUnitClifford specialization folded into QUnit; reference
+    // qunitclifford.cpp TrySeparate): move the bit to position 0 and ask
+    // CanDecomposeDispose — no tomography, no rounding, no fidelity cost.
+    if (auto st = std::dynamic_pointer_cast<QStabilizer<R>>(unit)) {
+        if (s.mapped != 0u) {
+            st->Swap(0u, s.mapped);
+            for (bitLenInt x = 0; x < qubitCount; ++x) {
+                if (shards[x].unit == unit && shards[x].mapped == 0u && x != q) {
+                    shards[x].mapped = s.mapped;
+                    break;
+                }
+            }
+            s.mapped = 0u;
+        }
+        if (!st->CanDecomposeDispose(0u, 1u)) return false;
+        QInterfacePtr<R> solo = MakeUnit(1u, 0u);
+        st->Decompose(0u, solo);
+        for (bitLenInt x = 0; x < qubitCount; ++x) {
+            if (shards[x].unit == unit && shards[x].mapped >= 1u && x != q) {
+                shards[x].mapped -= 1u;
+            }
+        }
+        s.unit = solo;
+        s.mapped = 0u;
+        return true;
+    }
     // 3-axis Bloch tomography (parity: qunit.cpp:696-855)
     const bitLenInt m = s.mapped;
     const R pz = unit->Prob(m);

@@ -157,3 +157,31 @@ def test_compose_dispose_with_pending_pairs():
     ref.x(3)
     ref.h(2)
     assert_states_close(a.get_state_vector(), ref.get_state_vector(), 1e-4)
+
+
+def test_clifford_exact_separation():
+    """Clifford units separate EXACTLY from the tableau (QUnitClifford
+    specialization): a mirrored entangler leaves every qubit in its own
+    1-qubit unit with fidelity exactly 1."""
+    n = 30
+    q = qa.create_simulator(n, layers=["qunit", "stabilizer"], seed=8)
+    q.set_reactive_separate(True)
+    for i in range(n):
+        q.h(i)
+    for i in range(n - 1):
+        q.cnot(i, i + 1)
+    for i in reversed(range(n - 1)):
+        q.cnot(i, i + 1)
+    # mirror: back to product |+>^n; exact separation must have fired
+    for i in range(n):
+        assert abs(q.prob(i) - 0.5) < 1e-6
+        assert q.try_separate(i)
+    assert q.get_unitary_fidelity() == pytest.approx(1.0)
+    # an entangled Bell pair must refuse exact separation
+    q2 = qa.create_simulator(4, layers=["qunit", "stabilizer"], seed=9)
+    q2.h(0)
+    q2.cnot(0, 1)
+    assert not q2.try_separate(0)
+    sv = np.asarray(q2.get_state_vector())
+    s = 1 / np.sqrt(2)
+    assert abs(abs(sv[0]) - s) < 1e-6 and abs(abs(sv[3]) - s) < 1e-6
