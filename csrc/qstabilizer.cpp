@@ -765,14 +765,18 @@ template <typename R> bitLenInt QStabilizer<R>::Compose(QInterfacePtr<R> toCopy,
 
 template <typename R> bool QStabilizer<R>::CanDecomposeDispose(bitLenInt start, bitLenInt length)
 {
-    // separable iff the stabilizer group has exactly `length` generators
-    // supported entirely inside [start, start+length) after elimination
+    // Separable iff, after Gaussian elimination pivoting on the COMPLEMENT's
+    // columns, exactly `length` stabilizer generators have support only
+    // inside [start, start+length). (Pivoting on the part's own columns —
+    // the previous form — false-negatives on separable states whose
+    // generator basis mixes part and complement, e.g. {X0X1X2, X1X2, X2}
+    // for |+++>.)
     QStabilizerPtr<R> clone = std::static_pointer_cast<QStabilizer<R>>(Clone());
     const bitLenInt n = qubitCount;
     auto inPart = [&](bitLenInt q) { return q >= start && q < start + length; };
     size_t i = n;
-    // eliminate part-columns to the top rows
-    for (bitLenInt j = start; j < start + length; ++j) {
+    for (bitLenInt j = 0; j < n; ++j) {
+        if (inPart(j)) continue;
         for (int pass = 0; pass < 2; ++pass) {
             const bool isX = (pass == 0);
             for (size_t k = i; k < 2u * (size_t)n; ++k) {
@@ -789,10 +793,8 @@ template <typename R> bool QStabilizer<R>::CanDecomposeDispose(bitLenInt start, 
             }
         }
     }
-    const size_t pivotRows = i - n;
-    if (pivotRows != length) return false;
-    // pivot rows must have no support outside the part
-    for (size_t r = n; r < i; ++r) {
+    if ((2u * (size_t)n - i) != (size_t)length) return false;
+    for (size_t r = i; r < 2u * (size_t)n; ++r) {
         for (bitLenInt q = 0; q < n; ++q) {
             if (!inPart(q) && (clone->getX(r, q) || clone->getZ(r, q))) return false;
         }
