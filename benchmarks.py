@@ -39,6 +39,22 @@ def run_qft_cosmology(q, n, rng, depth):
     q.qft(0, n)
 
 
+def run_clifford(q, n, rng, depth):
+    # wide random Clifford circuit (reference QUnitClifford regime: hundreds
+    # of qubits on the tableau); run with --layers qunit,stabilizer
+    q.set_permutation(0)
+    d = depth or 20
+    names = ["h", "s", "x", "z", "sqrt_x"]
+    for _ in range(d):
+        for i in range(n):
+            getattr(q, names[rng.integers(len(names))])(i)
+        for i in range(0, n - 1, 2):
+            if rng.integers(2):
+                q.cnot(i, i + 1)
+            else:
+                q.cz(i, i + 1)
+
+
 def run_ghz(q, n, rng, depth):
     q.set_permutation(0)
     q.h(0)
@@ -146,6 +162,7 @@ def run_qv(q, n, rng, depth):
 WORKLOADS = {
     "qft": run_qft,
     "qft_cosmology": run_qft_cosmology,
+    "clifford": run_clifford,
     "ghz": run_ghz,
     "random_circuit": run_random_circuit,
     "supremacy": run_supremacy,
