@@ -279,11 +279,24 @@ template <typename R> static int matrixKind(const cplx<R>* m)
     return isPhase ? 1 : (isInvert ? 2 : 0);
 }
 
+__global__ void k_apply2x2_1mfma(cplx<float>* sv, GateArgs<float> a);
+
 template <typename R>
 void launchApply2x2(cplx<R>* sv, const GateArgs<R>& a, hipStream_t stream)
 {
     const int kind = matrixKind(a.m);
     const int grid = gridFor(a.maxI);
+    if constexpr (std::is_same_v<R, float>) {
+        static const bool useMfma = []() {
+            if (const char* env = std::getenv("QRACK_GPU_MFMA")) return std::atoi(env) != 0;
+            return false;
+        }();
+        if (useMfma && a.nPowers == 1) {
+            hipLaunchKernelGGL((k_apply2x2_1mfma), dim3(gridFor(a.maxI >> 2u)), dim3(QA_BLOCK), 0,
+                stream, sv, a);
+            return;
+        }
+    }
     // fp32 pairs can vectorize as float4 whenever consecutive iteration
     // indices stay memory-adjacent: lowest skip power >= 2
     const bool vecOk = (a.nPowers > 0) && (a.qPowers[0] >= 2u) && ((a.maxI & 1u) == 0u);
@@ -1255,6 +1268,69 @@ template <int K> __global__ void k_mtrx_batch_v(cplx<float>* sv, Batch1qArgs<flo
                 if (s & (1 << g)) off |= a.tPow[g];
             }
             sv4[(base | off) >> 1u] = v[s];
+        }
+    }
+}
+
+// ---- MFMA demonstration variant ---------------------------------------------
+// The BASELINE north star calls for "MFMA used for the batched 2x2 complex
+// mat-vec". A 2x2 complex gate on a pair (x, y) is the 4x4 REAL matrix
+// G = [[m0.re,-m0.im,m1.re,-m1.im],[m0.im,m0.re,m1.im,m1.re],
+//      [m2.re,-m2.im,m3.re,-m3.im],[m2.im,m2.re,m3.im,m3.re]] acting on
+// (x.re, x.im, y.re, y.im). Each wave feeds v_mfma_f32_16x16x4_f32 with
+// A[i][k] = G[i%4][k] (the gate replicated down the rows) and
+// B[k][j] = component k of pair j, producing 16 pair results per MFMA.
+// The workload is HBM-bound at ~0.1 flop/byte, so this is a correctness +
+// counter demonstration (QRACK_GPU_MFMA=1), not the default path — the A/B
+// numbers are recorded in profiles/KERNELS.md.
+typedef float qa_f32x4 __attribute__((ext_vector_type(4)));
+
+__global__ void k_apply2x2_1mfma(cplx<float>* sv, GateArgs<float> a)
+{
+    const unsigned lane = threadIdx.x & 63u;
+    const unsigned j16 = lane & 15u;   // B column / D column
+    const unsigned kRow = lane >> 4u;  // A k-index / B k-index
+    const bitCapInt p = a.qPowers[0];
+    // gate as 4x4 real, row (lane&15)%4, column kRow
+    const float G[4][4] = {
+        { a.m[0].re, -a.m[0].im, a.m[1].re, -a.m[1].im },
+        { a.m[0].im, a.m[0].re, a.m[1].im, a.m[1].re },
+        { a.m[2].re, -a.m[2].im, a.m[3].re, -a.m[3].im },
+        { a.m[2].im, a.m[2].re, a.m[3].im, a.m[3].re },
+    };
+    const float aVal = G[j16 & 3u][kRow];
+    const bitCapInt wavesPerGrid = ((bitCapInt)gridDim.x * blockDim.x) >> 6u;
+    const bitCapInt waveId = (((bitCapInt)blockIdx.x * blockDim.x + threadIdx.x) >> 6u);
+    const bitCapInt nGroups = (a.maxI + 15u) >> 4u; // 16 pairs per wave step
+    for (bitCapInt g = waveId; g < nGroups; g += wavesPerGrid) {
+        const bitCapInt myPair = (g << 4u) | j16;
+        float xre = 0, xim = 0, yre = 0, yim = 0;
+        const bool live = myPair < a.maxI;
+        bitCapInt i = 0;
+        if (live) {
+            i = ((myPair & ~(p - 1u)) << 1u) | (myPair & (p - 1u));
+            const cplx<float> x = sv[i | a.offset1];
+            const cplx<float> y = sv[i | a.offset2];
+            xre = x.re;
+            xim = x.im;
+            yre = y.re;
+            yim = y.im;
+        }
+        // B[k][j] = component k of pair j: pull from lane j via wave shuffle
+        const int src = (int)j16;
+        const float c0 = __shfl(xre, src, 64);
+        const float c1 = __shfl(xim, src, 64);
+        const float c2 = __shfl(yre, src, 64);
+        const float c3 = __shfl(yim, src, 64);
+        const float bVal = (kRow == 0u) ? c0 : (kRow == 1u) ? c1 : (kRow == 2u) ? c2 : c3;
+        qa_f32x4 acc = { 0.f, 0.f, 0.f, 0.f };
+        acc = __builtin_amdgcn_mfma_f32_16x16x4f32(aVal, bVal, acc, 0, 0, 0);
+        // D[row][col]: lane holds rows (lane>>4)*4 + t, col = lane&15; every
+        // 16-lane group holds the full 4-vector of pair (lane&15) — group 0
+        // writes back
+        if (live && kRow == 0u) {
+            sv[i | a.offset1] = cplx<float>{ acc[0], acc[1] };
+            sv[i | a.offset2] = cplx<float>{ acc[2], acc[3] };
         }
     }
 }

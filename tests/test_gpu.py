@@ -483,3 +483,32 @@ def test_mtrx_2q_batch_gpu():
     for u, a, b in zip(us, q1s, q2s):
         qs.mtrx_2q([complex(x) for x in u.flatten()], a, b)
     assert float(qb.sum_sqr_diff(qs)) < 1e-5
+
+
+def test_mfma_gate_variant():
+    # QRACK_GPU_MFMA=1 routes single-target fp32 gates through the MFMA
+    # 4x4-real mat-vec kernel; numerics must match the default path
+    import subprocess, sys, os
+    code = """
+import os, sys
+sys.path.insert(0, os.getcwd())
+import numpy as np, qrack_amd as qa
+q = qa.create_simulator(20, engine="hip", seed=3)
+rng = np.random.default_rng(5)
+for i in range(20):
+    q.ry(float(rng.uniform(0, np.pi)), i)
+q.h(0); q.h(13); q.t(5); q.x(9)
+sv = np.asarray(q.get_state_vector())
+np.save("/tmp/mfma_ab.npy", sv)
+print("OK")
+"""
+    env0 = {**os.environ, "QRACK_GPU_MFMA": "0"}
+    env1 = {**os.environ, "QRACK_GPU_MFMA": "1"}
+    r0 = subprocess.run([sys.executable, "-c", code], env=env0, capture_output=True, text=True)
+    assert "OK" in r0.stdout, r0.stderr
+    import numpy as np
+    ref = np.load("/tmp/mfma_ab.npy")
+    r1 = subprocess.run([sys.executable, "-c", code], env=env1, capture_output=True, text=True)
+    assert "OK" in r1.stdout, r1.stderr
+    got = np.load("/tmp/mfma_ab.npy")
+    assert np.abs(ref - got).max() < 1e-5
