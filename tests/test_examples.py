@@ -23,3 +23,19 @@ def test_example_runs(script):
         env={**os.environ, "PYTHONPATH": os.path.dirname(EXDIR)},
     )
     assert out.returncode == 0, f"{script}: {out.stdout}\n{out.stderr}"
+
+
+def test_dist_qft_example_two_ranks():
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", str(21000 + os.getpid() % 8000),
+         os.path.join(EXDIR, "dist_qft.py"), "10"],
+        cwd=os.path.dirname(EXDIR),
+        capture_output=True,
+        text=True,
+        timeout=300,
+        env={**os.environ, "PYTHONPATH": os.path.dirname(EXDIR)},
+    )
+    assert out.returncode == 0, f"{out.stdout}\n{out.stderr}"
+    assert "QFT across 2 rank(s)" in out.stdout
