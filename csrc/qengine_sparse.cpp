@@ -114,19 +114,22 @@ void QEngineSparse<R>::Apply2x2(bitCapInt offset1, bitCapInt offset2, const cplx
         for (auto& a : adds) Put(a.first, a.second);
         return;
     }
-    // general: gather pair bases
+    // general: gather pair bases with ALL skip bits stripped — the pair is
+    // (b | offset1, b | offset2). (Stripping only targetPow and assuming
+    // offset1 is the clear side breaks swap-block gates, where offset1 and
+    // offset2 are DIFFERENT single bits: fsim/sqrt-swap on sparse.)
     std::vector<bitCapInt> bases;
     for (auto& kv : amps) {
         const bitCapInt pat = kv.first & skipMask;
         if (pat == offset1 || pat == offset2) {
-            bases.push_back(kv.first & ~targetPow);
+            bases.push_back(kv.first & ~skipMask);
         }
     }
     std::sort(bases.begin(), bases.end());
     bases.erase(std::unique(bases.begin(), bases.end()), bases.end());
     for (bitCapInt b : bases) {
-        const bitCapInt i1 = b; // has offset1 pattern (target bit clear)
-        const bitCapInt i2 = b | targetPow;
+        const bitCapInt i1 = b | offset1;
+        const bitCapInt i2 = b | offset2;
         const cplx<R> x = Get(i1), y = Get(i2);
         Put(i1, m[0] * x + m[1] * y);
         Put(i2, m[2] * x + m[3] * y);

@@ -96,3 +96,20 @@ def test_sparse_under_qunit():
     q.cnot(0, 15)
     q.cnot(15, 29)
     assert abs(q.prob(29) - 0.5) < 1e-5
+
+
+def test_sparse_swap_block_gates():
+    """Regression: swap-block Apply2x2 (offset1 and offset2 both single DIFFERENT
+    bits) — fsim/sqrt-swap were mispaired in the sparse general path."""
+    import numpy as np
+    for gate, args in (("sqrt_swap", (1, 3)), ("isqrt_swap", (0, 2)),
+                       ("fsim", (1.41, 0.54, 2, 3))):
+        q = qa.create_simulator(4, engine="sparse", seed=3)
+        cp = qa.create_simulator(4, engine="cpu", seed=3)
+        for s in (q, cp):
+            for i in range(4):
+                s.ry(0.5 + 0.3 * i, i)
+            getattr(s, gate)(*args)
+        fid = abs(np.vdot(np.asarray(cp.get_state_vector()),
+                          np.asarray(q.get_state_vector())))
+        assert fid > 1 - 1e-5, (gate, fid)
