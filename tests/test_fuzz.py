@@ -21,7 +21,7 @@ GATE_1Q = ["h", "x", "y", "z", "s", "t", "sqrt_x"]
 def circuits(draw):
     ops = []
     for _ in range(draw(st.integers(3, 14))):
-        kind = draw(st.integers(0, 3))
+        kind = draw(st.integers(0, 6))
         if kind == 0:
             ops.append(("g1", draw(st.sampled_from(GATE_1Q)), draw(st.integers(0, N_QUBITS - 1))))
         elif kind == 1:
@@ -31,10 +31,26 @@ def circuits(draw):
             a = draw(st.integers(0, N_QUBITS - 1))
             b = draw(st.integers(0, N_QUBITS - 1).filter(lambda x: x != a))
             ops.append(("cnot", a, b))
-        else:
+        elif kind == 3:
             a = draw(st.integers(0, N_QUBITS - 1))
             b = draw(st.integers(0, N_QUBITS - 1).filter(lambda x: x != a))
             ops.append(("cz", a, b))
+        elif kind == 4:
+            a = draw(st.integers(0, N_QUBITS - 1))
+            b = draw(st.integers(0, N_QUBITS - 1).filter(lambda x: x != a))
+            ops.append(("swap", a, b))
+        elif kind == 5:
+            a = draw(st.integers(0, N_QUBITS - 1))
+            b = draw(st.integers(0, N_QUBITS - 1).filter(lambda x: x != a))
+            ops.append(("fsim", draw(st.floats(0.1, 3.0)), draw(st.floats(0.1, 3.0)), a, b))
+        else:
+            # batched 1q layer on two distinct targets
+            a = draw(st.integers(0, N_QUBITS - 1))
+            b = draw(st.integers(0, N_QUBITS - 1).filter(lambda x: x != a))
+            th = draw(st.floats(0.1, 3.0))
+            c, sn = np.cos(th / 2), np.sin(th / 2)
+            m = [complex(c), complex(-sn), complex(sn), complex(c)]
+            ops.append(("batch", [a, b], m + m))
     return ops
 
 
@@ -46,8 +62,14 @@ def apply(q, ops):
             q.ry(op[1], op[2])
         elif op[0] == "cnot":
             q.cnot(op[1], op[2])
-        else:
+        elif op[0] == "cz":
             q.cz(op[1], op[2])
+        elif op[0] == "swap":
+            q.swap(op[1], op[2])
+        elif op[0] == "fsim":
+            q.fsim(op[1], op[2], op[3], op[4])
+        else:
+            q.mtrx_1q_batch(op[1], op[2])
 
 
 STACKS = [
