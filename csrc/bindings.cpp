@@ -309,6 +309,8 @@ template <typename R> static void bindQInterface(py::module_& m, const char* nam
         .def("dispose_perm",
             [](QI& q, bitLenInt start, bitLenInt len, bitCapInt perm) { q.Dispose(start, len, perm); })
         .def("allocate", [](QI& q, bitLenInt len) { return q.Allocate(len); })
+        .def("allocate",
+            [](QI& q, bitLenInt start, bitLenInt len) { return q.Allocate(start, len); })
         .def("clone", [](QI& q) { return q.Clone(); })
         .def("try_separate_1", [](QI& q, bitLenInt qb) { return q.TrySeparate(qb); })
         .def("try_separate_2", [](QI& q, bitLenInt q1, bitLenInt q2) { return q.TrySeparate(q1, q2); })

@@ -980,9 +980,14 @@ public:
     virtual QInterfacePtr<R> Copy() { return Clone(); }
     // Compose that may consume the source (default: plain Compose)
     virtual bitLenInt ComposeNoClone(QInterfacePtr<R> toCopy) { return Compose(toCopy); }
-    // attempt Decompose; on failure leave the state untouched and return false
-    virtual bool TryDecompose(bitLenInt start, QInterfacePtr<R> dest, R error_tol = eps<R>::value)
+    // attempt Decompose; on failure leave the state untouched and return
+    // false. The verify compares a decompose+recompose probe against the
+    // original; the default tolerance allows for a few hundred ULP of
+    // round-trip float noise (machine epsilon alone false-negatives on
+    // separable fp32 states).
+    virtual bool TryDecompose(bitLenInt start, QInterfacePtr<R> dest, R error_tol = (R)0)
     {
+        if (error_tol <= (R)0) error_tol = (R)1024 * eps<R>::value;
         QInterfacePtr<R> probe = Clone();
         QInterfacePtr<R> probeDest = dest->Clone();
         try {
