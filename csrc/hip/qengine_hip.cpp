@@ -13,6 +13,15 @@
 
 namespace qrack_amd {
 
+template <typename R>
+static void qaCheckModN(bitCapInt modN, bitLenInt length, const char* op)
+{
+    if (modN == 0u || modN > pow2(length)) {
+        throw QrackError(std::string(op) + ": modN must be in (0, 2^length]");
+    }
+}
+
+
 // ---- profiler ---------------------------------------------------------------
 
 namespace {
@@ -1478,6 +1487,7 @@ template <typename R>
 void QEngineHIP<R>::MULModNOut(
     bitCapInt toMul, bitCapInt modN, bitLenInt inStart, bitLenInt outStart, bitLenInt length)
 {
+    qaCheckModN<R>(modN, length, "MULModNOut");
     checkAluRangeHip<R>(inStart, length, qubitCount, "MULModNOut");
     checkAluRangeHip<R>(outStart, length, qubitCount, "MULModNOut");
     PermArgs a{};
@@ -1489,6 +1499,7 @@ template <typename R>
 void QEngineHIP<R>::IMULModNOut(
     bitCapInt toMul, bitCapInt modN, bitLenInt inStart, bitLenInt outStart, bitLenInt length)
 {
+    qaCheckModN<R>(modN, length, "IMULModNOut");
     PermArgs a{};
     setupModArgs<R>(a, PermOp::IMULMODN, toMul, modN, inStart, outStart, length, maxQPower, {});
     permuteOp(a, true, false);
@@ -1498,6 +1509,7 @@ template <typename R>
 void QEngineHIP<R>::POWModNOut(
     bitCapInt base, bitCapInt modN, bitLenInt inStart, bitLenInt outStart, bitLenInt length)
 {
+    qaCheckModN<R>(modN, length, "POWModNOut");
     checkAluRangeHip<R>(inStart, length, qubitCount, "POWModNOut");
     checkAluRangeHip<R>(outStart, length, qubitCount, "POWModNOut");
     PermArgs a{};
@@ -1563,6 +1575,7 @@ template <typename R>
 void QEngineHIP<R>::CMULModNOut(bitCapInt toMul, bitCapInt modN, bitLenInt inStart, bitLenInt outStart,
     bitLenInt length, const std::vector<bitLenInt>& controls)
 {
+    qaCheckModN<R>(modN, length, "CMULModNOut");
     if (controls.empty()) {
         MULModNOut(toMul, modN, inStart, outStart, length);
         return;
@@ -1576,6 +1589,7 @@ template <typename R>
 void QEngineHIP<R>::CIMULModNOut(bitCapInt toMul, bitCapInt modN, bitLenInt inStart, bitLenInt outStart,
     bitLenInt length, const std::vector<bitLenInt>& controls)
 {
+    qaCheckModN<R>(modN, length, "CIMULModNOut");
     if (controls.empty()) {
         IMULModNOut(toMul, modN, inStart, outStart, length);
         return;
@@ -1589,6 +1603,7 @@ template <typename R>
 void QEngineHIP<R>::CPOWModNOut(bitCapInt base, bitCapInt modN, bitLenInt inStart, bitLenInt outStart,
     bitLenInt length, const std::vector<bitLenInt>& controls)
 {
+    qaCheckModN<R>(modN, length, "CPOWModNOut");
     if (controls.empty()) {
         POWModNOut(base, modN, inStart, outStart, length);
         return;

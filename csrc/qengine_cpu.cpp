@@ -927,6 +927,16 @@ void QEngineCPU<R>::ControlledPermutationOp(
 // ---- ALU -------------------------------------------------------------------
 
 template <typename R>
+static void checkModN(bitCapInt modN, bitLenInt length, const char* op)
+{
+    // the residue must FIT the out register: modN > 2^length would scatter
+    // writes past the register (caught by fuzzing as heap corruption)
+    if (modN == 0u || modN > pow2(length)) {
+        throw QrackError(std::string(op) + ": modN must be in (0, 2^length]");
+    }
+}
+
+template <typename R>
 static void checkAluRange(bitLenInt start, bitLenInt length, bitLenInt qubitCount, const char* op)
 {
     if ((bitCapInt)start + length > qubitCount) {
@@ -1145,6 +1155,7 @@ template <typename R>
 void QEngineCPU<R>::MULModNOut(
     bitCapInt toMul, bitCapInt modN, bitLenInt inStart, bitLenInt outStart, bitLenInt length)
 {
+    checkModN<R>(modN, length, "MULModNOut");
     checkAluRange<R>(inStart, length, qubitCount, "MULModNOut");
     checkAluRange<R>(outStart, length, qubitCount, "MULModNOut");
     const bitCapInt lenMask = pow2Mask(length);
@@ -1168,6 +1179,7 @@ template <typename R>
 void QEngineCPU<R>::IMULModNOut(
     bitCapInt toMul, bitCapInt modN, bitLenInt inStart, bitLenInt outStart, bitLenInt length)
 {
+    checkModN<R>(modN, length, "IMULModNOut");
     const bitCapInt lenMask = pow2Mask(length);
     const bitCapInt inMask = lenMask << inStart;
     const bitCapInt outMask = lenMask << outStart;
@@ -1201,6 +1213,7 @@ template <typename R>
 void QEngineCPU<R>::POWModNOut(
     bitCapInt base, bitCapInt modN, bitLenInt inStart, bitLenInt outStart, bitLenInt length)
 {
+    checkModN<R>(modN, length, "POWModNOut");
     checkAluRange<R>(inStart, length, qubitCount, "POWModNOut");
     checkAluRange<R>(outStart, length, qubitCount, "POWModNOut");
     const bitCapInt lenMask = pow2Mask(length);
@@ -1292,6 +1305,7 @@ template <typename R>
 void QEngineCPU<R>::CMULModNOut(bitCapInt toMul, bitCapInt modN, bitLenInt inStart, bitLenInt outStart,
     bitLenInt length, const std::vector<bitLenInt>& controls)
 {
+    checkModN<R>(modN, length, "CMULModNOut");
     if (controls.empty()) {
         MULModNOut(toMul, modN, inStart, outStart, length);
         return;
@@ -1322,6 +1336,7 @@ template <typename R>
 void QEngineCPU<R>::CIMULModNOut(bitCapInt toMul, bitCapInt modN, bitLenInt inStart, bitLenInt outStart,
     bitLenInt length, const std::vector<bitLenInt>& controls)
 {
+    checkModN<R>(modN, length, "CIMULModNOut");
     if (controls.empty()) {
         IMULModNOut(toMul, modN, inStart, outStart, length);
         return;
@@ -1352,6 +1367,7 @@ template <typename R>
 void QEngineCPU<R>::CPOWModNOut(bitCapInt base, bitCapInt modN, bitLenInt inStart, bitLenInt outStart,
     bitLenInt length, const std::vector<bitLenInt>& controls)
 {
+    checkModN<R>(modN, length, "CPOWModNOut");
     if (controls.empty()) {
         POWModNOut(base, modN, inStart, outStart, length);
         return;

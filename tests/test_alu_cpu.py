@@ -5,6 +5,8 @@ test_indexed* / test_hash / test_phaseflip family.
 
 import numpy as np
 
+import pytest
+
 import qrack_amd as qa
 
 
@@ -253,3 +255,17 @@ def test_incbcd_superposition():
     q.h(4)  # superpose a spectator qubit
     q.incbcd(3, 0, 4)
     assert q.m_reg(0, 4) == 0x3
+
+
+def test_mod_alu_rejects_oversized_modn():
+    """Regression (fuzz-found heap overflow): modN > 2^length scattered
+    writes past the out register; it must throw instead."""
+    q = qa.create_simulator(8, engine="cpu", seed=1)
+    q.x(0)
+    with pytest.raises(RuntimeError):
+        q.mul_mod_n_out(7, 31, 0, 5, 3)  # modN 31 > 2^3
+    with pytest.raises(RuntimeError):
+        q.pow_mod_n_out(3, 300, 0, 4, 4)
+    # valid modN still works
+    q.mul_mod_n_out(3, 8, 0, 5, 3)
+    assert (q.m_all() >> 5) == 3
