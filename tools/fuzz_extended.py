@@ -18,14 +18,19 @@ sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)), ".."
 def fuzz_state():
     import numpy as np
     import qrack_amd as qa
-    import numpy as np
-    import qrack_amd as qa
 
-    STACKS = [
-        ["cpu"], ["sparse"], ["bdt"], ["stabilizer_hybrid", "cpu"],
-        ["qunit", "cpu"], ["qunit", "stabilizer_hybrid", "cpu"],
-        ["pager", "cpu"], ["qunit", "stabilizer", ], ["hybrid"], ["tensor_network", "cpu"],
-    ]
+    if os.environ.get("QA_FUZZ_GPU"):
+        # GPU-box variant: HIP engine paths vs the dense CPU reference
+        STACKS = [
+            ["hip"], ["pager", "hip"], ["hybrid"],
+            ["qunit", "hybrid"], ["qunit", "stabilizer_hybrid", "hybrid"],
+        ]
+    else:
+        STACKS = [
+            ["cpu"], ["sparse"], ["bdt"], ["stabilizer_hybrid", "cpu"],
+            ["qunit", "cpu"], ["qunit", "stabilizer_hybrid", "cpu"],
+            ["pager", "cpu"], ["qunit", "stabilizer", ], ["hybrid"], ["tensor_network", "cpu"],
+        ]
     N = 5
     rng = np.random.default_rng(20260913)
     fails = 0
@@ -211,7 +216,7 @@ def fuzz_struct():
 def fuzz_serial():
     import numpy as np
     import qrack_amd as qa
-    import sys, tempfile, os
+    import tempfile
     sys.path.insert(0, "/root/repo")
 
     rng = np.random.default_rng(77)
