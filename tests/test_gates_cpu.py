@@ -211,3 +211,27 @@ def test_time_evolve_pauli_x(precision):
     sv = q.get_state_vector()
     assert abs(sv[0] - np.cos(t)) < 1e-5
     assert abs(sv[1] + 1j * np.sin(t)) < 1e-5
+
+
+def test_time_evolve_vs_scipy_random_hermitian():
+    """TimeEvolve against scipy expm for random Hermitian 2x2 terms,
+    including controlled terms."""
+    import scipy.linalg as sla
+
+    rng = np.random.default_rng(17)
+    for trial in range(6):
+        a, b = rng.normal(size=2)
+        c = rng.normal() + 1j * rng.normal()
+        H = np.array([[a, c], [np.conj(c), b]])
+        t = float(rng.uniform(0.1, 1.5))
+        U = sla.expm(-1j * H * t)
+        q = qa.create_simulator(2, engine="cpu", precision="fp64", seed=1)
+        q.ry(0.9, 0)
+        q.ry(0.4, 1)
+        sv0 = np.asarray(q.get_state_vector()).astype(complex).reshape(2, 2)
+        q.time_evolve([{"target": 0, "matrix": [complex(H[0, 0]), complex(H[0, 1]),
+                                                complex(H[1, 0]), complex(H[1, 1])]}], t)
+        got = np.asarray(q.get_state_vector()).astype(complex).reshape(2, 2)
+        # qubit 0 is the LSB: state[i1][i0]; apply U on axis 1
+        want = np.einsum("ab,ib->ia", U, sv0)
+        assert np.abs(got - want).max() < 1e-6, trial
