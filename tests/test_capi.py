@@ -104,3 +104,19 @@ def test_error_latch(lib):
     lib.qrack_muln(u64(sid), u64(3), u64(15), u64(0), u64(4), u64(4))
     assert lib.qrack_get_error(u64(sid)) != 0
     lib.qrack_destroy(u64(sid))
+
+
+def test_approximation_knobs(lib):
+    # SetSdrp / SetNcrp via the C ABI on the canonical stack
+    layers = b"qunit,stabilizer_hybrid,cpu"
+    sid = lib.qrack_init_count_type(u64(6), layers, 0, ctypes.c_int64(7))
+    lib.qrack_set_sdrp(u64(sid), ctypes.c_double(0.3))
+    lib.qrack_set_ncrp(u64(sid), ctypes.c_double(0.1))
+    for i in range(6):
+        lib.qrack_h(u64(sid), u64(i))
+    for i in range(5):
+        lib.qrack_mcx1(u64(sid), u64(i), u64(i + 1)) if hasattr(lib, "qrack_mcx1") else None
+    fid = lib.qrack_get_unitary_fidelity(u64(sid))
+    assert 0.0 < fid <= 1.0 + 1e-9
+    assert lib.qrack_get_error(u64(sid)) == 0
+    lib.qrack_destroy(u64(sid))
