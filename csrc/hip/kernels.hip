@@ -1127,6 +1127,68 @@ __global__ void k_qft_col_gen_v(
     }
 }
 
+// A/B experiment (QRACK_GPU_QFT_PIPE=1): software-pipelined column — loads
+// for iteration i+1 issue before iteration i's stores, hiding the RMW
+// turnaround on the two streams. Same math as k_qft_col_v<false>.
+template <bool PRE>
+__global__ void k_qft_col_v_pipe(
+    cplx<float>* sv, bitCapInt maxI, bitCapInt tPow, bitLenInt rampStart, bitCapInt rampMask,
+    float scale)
+{
+    float4* sv4 = reinterpret_cast<float4*>(sv);
+    const float s = 0.70710678f;
+    const bitCapInt half = maxI >> 1u;
+    const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
+    bitCapInt kIt = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x;
+    if (kIt >= half) return;
+    auto idx = [&](bitCapInt kk, bitCapInt& lo4, bitCapInt& hi4, bitCapInt& i) {
+        const bitCapInt j = 2u * kk;
+        i = ((j & ~(tPow - 1u)) << 1u) | (j & (tPow - 1u));
+        lo4 = i >> 1u;
+        hi4 = (i | tPow) >> 1u;
+    };
+    bitCapInt lo4, hi4, i;
+    idx(kIt, lo4, hi4, i);
+    float4 vlo = sv4[lo4];
+    float4 vhi = sv4[hi4];
+    while (true) {
+        const bitCapInt kNext = kIt + stride;
+        bitCapInt nlo4 = 0, nhi4 = 0, ni = 0;
+        float4 nlo{}, nhi{};
+        const bool more = kNext < half;
+        if (more) {
+            idx(kNext, nlo4, nhi4, ni);
+            nlo = sv4[nlo4]; // prefetch next pair before this one's stores
+            nhi = sv4[nhi4];
+        }
+        float s0, c0, s1, c1;
+        __sincosf(scale * (float)((i >> rampStart) & rampMask), &s0, &c0);
+        __sincosf(scale * (float)(((i + 1u) >> rampStart) & rampMask), &s1, &c1);
+        const cplx<float> f0{ c0, s0 }, f1{ c1, s1 };
+        cplx<float> x0{ vlo.x, vlo.y }, x1{ vlo.z, vlo.w };
+        cplx<float> y0{ vhi.x, vhi.y }, y1{ vhi.z, vhi.w };
+        if (PRE) {
+            y0 = f0 * y0;
+            y1 = f1 * y1;
+        }
+        cplx<float> a0 = s * (x0 + y0), a1 = s * (x1 + y1);
+        cplx<float> b0 = s * (x0 - y0), b1 = s * (x1 - y1);
+        if (!PRE) {
+            b0 = f0 * b0;
+            b1 = f1 * b1;
+        }
+        sv4[lo4] = make_float4(a0.re, a0.im, a1.re, a1.im);
+        sv4[hi4] = make_float4(b0.re, b0.im, b1.re, b1.im);
+        if (!more) break;
+        kIt = kNext;
+        lo4 = nlo4;
+        hi4 = nhi4;
+        i = ni;
+        vlo = nlo;
+        vhi = nhi;
+    }
+}
+
 template <typename R>
 void launchQftColumnGeneral(cplx<R>* sv, bitCapInt maxQPower, bitCapInt tPow, const RampArgs& a,
     double phase0, bool pre, hipStream_t stream)
@@ -1162,6 +1224,20 @@ void launchQftColumn(cplx<R>* sv, bitCapInt maxQPower, bitLenInt rampStart, bitL
     const R scale = (R)sign * (R)3.14159265358979323846 / (R)(ONE_BCI << col);
     if constexpr (std::is_same_v<R, float>) {
         if (tPow >= 2u && (maxI & 1u) == 0u) {
+            static const bool pipe = []() {
+                if (const char* env = std::getenv("QRACK_GPU_QFT_PIPE")) return std::atoi(env) != 0;
+                return false;
+            }();
+            if (pipe) {
+                if (pre) {
+                    hipLaunchKernelGGL((k_qft_col_v_pipe<true>), dim3(gridFor(maxI >> 1u)),
+                        dim3(QA_BLOCK), 0, stream, sv, maxI, tPow, rampStart, rampMask, (float)scale);
+                } else {
+                    hipLaunchKernelGGL((k_qft_col_v_pipe<false>), dim3(gridFor(maxI >> 1u)),
+                        dim3(QA_BLOCK), 0, stream, sv, maxI, tPow, rampStart, rampMask, (float)scale);
+                }
+                return;
+            }
             if (pre) {
                 hipLaunchKernelGGL((k_qft_col_v<true>), dim3(gridFor(maxI >> 1u)), dim3(QA_BLOCK),
                     0, stream, sv, maxI, tPow, rampStart, rampMask, (float)scale);
