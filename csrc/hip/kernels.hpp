@@ -17,9 +17,10 @@ namespace qrack_amd {
 
 constexpr int QA_MAX_SKIP_POWERS = 16;
 // grid-size cap for gate/reduction launches; partials buffers must hold
-// this many (A/B-swept on the 30q QFT: 16384 beats 8192 by ~1.5%, 8192
-// beats 2048 by ~4%)
-constexpr int QA_REDUCE_MAX_BLOCKS = 16384;
+// this many (A/B-swept on the 30q QFT: 32768 = 104.4 ms < 16384 = 106.1 <
+// 8192 = 108.0 < 2048 = 112.1 — more blocks shrink the grid-stride tails
+// and balance the 8 XCDs)
+constexpr int QA_REDUCE_MAX_BLOCKS = 32768;
 
 // by-value argument block for gate kernels
 template <typename R> struct GateArgs {
