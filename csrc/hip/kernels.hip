@@ -17,15 +17,21 @@
 namespace qrack_amd {
 
 constexpr int QA_BLOCK = 256;
-// 8192 blocks measured ~5% faster than 2048 on the 30-qubit H stream
-// (gpurun A/B, profiles/KERNELS.md); grid-stride keeps residency bounded
+// Streaming gate kernels launch an EXACT grid (one iteration per thread):
+// the block-count A/B on the 30q QFT is monotone all the way up —
+// 2048 = 112.1 ms, 8192 = 108.0, 32768 = 104.4, exact (~0.5-1M blocks) =
+// 96.0 ms. Consecutive blocks touch consecutive tiles, which keeps each
+// XCD's L2 on a contiguous slice; the grid-stride remainder loop only
+// handles env-capped experiments (QRACK_GPU_BLOCKS). Reductions stay
+// capped at QA_REDUCE_MAX_BLOCKS to bound their partials buffers.
 constexpr int QA_MAX_BLOCKS = QA_REDUCE_MAX_BLOCKS;
+constexpr bitCapInt QA_GRID_HW_MAX = (bitCapInt)1 << 30; // dispatch sanity cap
 
 static inline int maxBlocks()
 {
     static int v = [] {
         if (const char* env = std::getenv("QRACK_GPU_BLOCKS")) return std::atoi(env);
-        return QA_MAX_BLOCKS;
+        return 0; // 0 = exact grid
     }();
     return v;
 }
@@ -42,7 +48,9 @@ static inline bool useNontemporal()
 static inline int gridFor(bitCapInt n)
 {
     bitCapInt b = (n + QA_BLOCK - 1) / QA_BLOCK;
-    if (b > (bitCapInt)maxBlocks()) b = maxBlocks();
+    const int cap = maxBlocks();
+    if (cap > 0 && b > (bitCapInt)cap) b = cap;
+    if (b > QA_GRID_HW_MAX) b = QA_GRID_HW_MAX;
     if (b < 1) b = 1;
     return (int)b;
 }
@@ -68,7 +76,18 @@ template <bool NT> __device__ __forceinline__ void st4(float4* p, float4 v)
     }
 }
 
-int reduceGridSize(bitCapInt n) { return gridFor(n); }
+// reductions/argmax/inner write one partial PER BLOCK: their grids must
+// stay within the fixed partials buffers regardless of the exact-grid
+// default for streaming kernels
+static inline int gridForReduce(bitCapInt n)
+{
+    bitCapInt b = (n + QA_BLOCK - 1) / QA_BLOCK;
+    if (b > (bitCapInt)QA_REDUCE_MAX_BLOCKS) b = QA_REDUCE_MAX_BLOCKS;
+    if (b < 1) b = 1;
+    return (int)b;
+}
+
+int reduceGridSize(bitCapInt n) { return gridForReduce(n); }
 
 __device__ __forceinline__ bitCapInt expandBits(bitCapInt j, const bitCapInt* pows, int n)
 {
@@ -519,7 +538,7 @@ template <typename R>
 int launchReduce(const cplx<R>* sv, const ReduceArgs& a, int op, double* partialsDev,
     hipStream_t stream)
 {
-    const int grid = gridFor(a.maxI);
+    const int grid = gridForReduce(a.maxI);
     hipLaunchKernelGGL(
         (k_reduce<R>), dim3(grid), dim3(QA_BLOCK), 0, stream, sv, a, op, partialsDev);
     return grid;
@@ -571,7 +590,7 @@ template <typename R>
 int launchArgMax(const cplx<R>* sv, bitCapInt maxI, double* valsDev, bitCapInt* idxDev,
     hipStream_t stream)
 {
-    const int grid = gridFor(maxI);
+    const int grid = gridForReduce(maxI);
     hipLaunchKernelGGL(
         (k_argmax<R>), dim3(grid), dim3(QA_BLOCK), 0, stream, sv, maxI, valsDev, idxDev);
     return grid;
@@ -1746,7 +1765,7 @@ template <typename R>
 int launchInner(const cplx<R>* a, const cplx<R>* b, bitCapInt maxI, double* partialsRe,
     double* partialsIm, hipStream_t stream)
 {
-    const int grid = gridFor(maxI);
+    const int grid = gridForReduce(maxI);
     hipLaunchKernelGGL(
         (k_inner<R>), dim3(grid), dim3(QA_BLOCK), 0, stream, a, b, maxI, partialsRe, partialsIm);
     return grid;
