@@ -25,7 +25,10 @@ constexpr int QA_BLOCK = 256;
 // handles env-capped experiments (QRACK_GPU_BLOCKS). Reductions stay
 // capped at QA_REDUCE_MAX_BLOCKS to bound their partials buffers.
 constexpr int QA_MAX_BLOCKS = QA_REDUCE_MAX_BLOCKS;
-constexpr bitCapInt QA_GRID_HW_MAX = (bitCapInt)1 << 30; // dispatch sanity cap
+// the HSA dispatch packet's grid size is a 32-bit WORK-ITEM count: blocks
+// * QA_BLOCK must stay below 2^32 or the launch silently wraps (caught at
+// 34 qubits, where the exact column grid needs exactly 2^32 threads)
+constexpr bitCapInt QA_GRID_HW_MAX = ((bitCapInt)1 << 24) - 1u;
 
 static inline int maxBlocks()
 {
