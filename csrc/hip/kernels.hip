@@ -16,7 +16,11 @@
 
 namespace qrack_amd {
 
+#ifndef QA_BLOCK_OVERRIDE
 constexpr int QA_BLOCK = 256;
+#else
+constexpr int QA_BLOCK = QA_BLOCK_OVERRIDE;
+#endif
 // Streaming gate kernels launch an EXACT grid (one iteration per thread):
 // the block-count A/B on the 30q QFT is monotone all the way up —
 // 2048 = 112.1 ms, 8192 = 108.0, 32768 = 104.4, exact (~0.5-1M blocks) =
