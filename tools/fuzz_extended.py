@@ -96,8 +96,12 @@ def fuzz_measure():
     import numpy as np
     import qrack_amd as qa
 
-    STACKS = [["cpu"], ["sparse"], ["stabilizer_hybrid", "cpu"], ["qunit", "cpu"],
-              ["qunit", "stabilizer_hybrid", "cpu"], ["pager", "cpu"], ["bdt"], ["hybrid"]]
+    if os.environ.get("QA_FUZZ_GPU"):
+        STACKS = [["hip"], ["pager", "hip"], ["hybrid"], ["qunit", "hybrid"],
+                  ["qunit", "stabilizer_hybrid", "hybrid"]]
+    else:
+        STACKS = [["cpu"], ["sparse"], ["stabilizer_hybrid", "cpu"], ["qunit", "cpu"],
+                  ["qunit", "stabilizer_hybrid", "cpu"], ["pager", "cpu"], ["bdt"], ["hybrid"]]
     N = 5
     rng = np.random.default_rng(99)
     fails = 0
