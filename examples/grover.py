@@ -1,6 +1,7 @@
 """Grover search for a marked item (parity: /root/reference/examples/grovers.cpp)."""
 import sys, math
-sys.path.insert(0, "..")
+import os
+sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)), ".."))
 import qrack_amd as qa
 
 

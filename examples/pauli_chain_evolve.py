@@ -1,7 +1,8 @@
 """Trotterized time evolution of a transverse-field Ising chain
 (parity: /root/reference/examples/ - Hamiltonian/TimeEvolve usage)."""
 import sys
-sys.path.insert(0, "..")
+import os
+sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)), ".."))
 import qrack_amd as qa
 
 if __name__ == "__main__":

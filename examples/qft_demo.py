@@ -1,6 +1,7 @@
 """QFT period finding demo (parity: /root/reference/examples/qft.cpp)."""
 import sys
-sys.path.insert(0, "..")
+import os
+sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)), ".."))
 import qrack_amd as qa
 
 if __name__ == "__main__":

@@ -3,7 +3,8 @@ shors_factoring.cpp), using the coherent modular-exponentiation ALU
 (POWModNOut) + the fused QFT."""
 import sys, math
 from fractions import Fraction
-sys.path.insert(0, "..")
+import os
+sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)), ".."))
 import qrack_amd as qa
 
 

@@ -1,6 +1,7 @@
 """Quantum teleportation (parity: /root/reference/examples/teleport.cpp)."""
 import sys
-sys.path.insert(0, "..")
+import os
+sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)), ".."))
 import qrack_amd as qa
 
 
