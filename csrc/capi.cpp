@@ -17,24 +17,39 @@ struct SimSlot {
     QInterfacePtr<float> f;
     QInterfacePtr<double> d;
     int error = 0;
+    // Per-simulator operation lock: held across every op so a concurrent
+    // qrack_destroy() cannot free state mid-operation (the shared_ptr keeps
+    // the slot itself alive; this mutex serializes ops on one handle, same
+    // contract as the reference pinvoke per-simulator locks).
+    std::mutex op;
     bitLenInt Qubits() const { return f ? f->GetQubitCount() : (d ? d->GetQubitCount() : 0); }
 };
+using SimSlotPtr = std::shared_ptr<SimSlot>;
 
 std::mutex g_mtx;
-std::map<quid, SimSlot> g_sims;
+std::map<quid, SimSlotPtr> g_sims;
 quid g_next = 1;
 
-SimSlot* slot(quid sid)
+SimSlotPtr slot(quid sid)
 {
     std::lock_guard<std::mutex> lk(g_mtx);
     auto it = g_sims.find(sid);
-    return (it == g_sims.end()) ? nullptr : &it->second;
+    return (it == g_sims.end()) ? nullptr : it->second;
+}
+
+quid registerSlot(SimSlotPtr s)
+{
+    std::lock_guard<std::mutex> lk(g_mtx);
+    const quid sid = g_next++;
+    g_sims[sid] = std::move(s);
+    return sid;
 }
 
 template <typename F> void guarded(quid sid, F&& fn)
 {
-    SimSlot* s = slot(sid);
+    SimSlotPtr s = slot(sid);
     if (!s) return;
+    std::lock_guard<std::mutex> lk(s->op);
     try {
         fn(*s);
     } catch (const std::bad_alloc&) {
@@ -46,8 +61,9 @@ template <typename F> void guarded(quid sid, F&& fn)
 
 template <typename F> double guardedD(quid sid, F&& fn, double dflt = 0.0)
 {
-    SimSlot* s = slot(sid);
+    SimSlotPtr s = slot(sid);
     if (!s) return dflt;
+    std::lock_guard<std::mutex> lk(s->op);
     try {
         return fn(*s);
     } catch (const std::exception&) {
@@ -101,20 +117,17 @@ quid qrack_init_count_type(
     } else {
         layers.push_back("cpu");
     }
-    SimSlot s;
+    SimSlotPtr s = std::make_shared<SimSlot>();
     try {
         if (dbl) {
-            s.d = CreateStack<double>((bitLenInt)qubits, layers, 0u, -1, -1, 1u);
+            s->d = CreateStack<double>((bitLenInt)qubits, layers, 0u, -1, -1, 1u);
         } else {
-            s.f = CreateStack<float>((bitLenInt)qubits, layers, 0u, -1, -1, 1u);
+            s->f = CreateStack<float>((bitLenInt)qubits, layers, 0u, -1, -1, 1u);
         }
     } catch (const std::exception&) {
         return 0;
     }
-    std::lock_guard<std::mutex> lk(g_mtx);
-    const quid sid = g_next++;
-    g_sims[sid] = std::move(s);
-    return sid;
+    return registerSlot(std::move(s));
 }
 
 quid qrack_init_count(uint64_t qubits, int gpu)
@@ -124,19 +137,17 @@ quid qrack_init_count(uint64_t qubits, int gpu)
 
 quid qrack_init_clone(quid sid)
 {
-    SimSlot* s = slot(sid);
+    SimSlotPtr s = slot(sid);
     if (!s) return 0;
-    SimSlot c;
+    SimSlotPtr c = std::make_shared<SimSlot>();
+    std::lock_guard<std::mutex> lk(s->op);
     try {
-        if (s->f) c.f = s->f->Clone();
-        if (s->d) c.d = s->d->Clone();
+        if (s->f) c->f = s->f->Clone();
+        if (s->d) c->d = s->d->Clone();
     } catch (const std::exception&) {
         return 0;
     }
-    std::lock_guard<std::mutex> lk(g_mtx);
-    const quid nid = g_next++;
-    g_sims[nid] = std::move(c);
-    return nid;
+    return registerSlot(std::move(c));
 }
 
 void qrack_destroy(quid sid)
@@ -152,13 +163,15 @@ void qrack_seed(quid sid, uint64_t sd)
 
 uint64_t qrack_num_qubits(quid sid)
 {
-    SimSlot* s = slot(sid);
-    return s ? s->Qubits() : 0;
+    SimSlotPtr s = slot(sid);
+    if (!s) return 0;
+    std::lock_guard<std::mutex> lk(s->op);
+    return s->Qubits();
 }
 
 int qrack_get_error(quid sid)
 {
-    SimSlot* s = slot(sid);
+    SimSlotPtr s = slot(sid);
     return s ? s->error : -1;
 }
 
@@ -579,9 +592,18 @@ void qrack_set_ncrp(quid sid, double ncrp)
 
 quid qrack_compose(quid sid, quid other)
 {
-    SimSlot* a = slot(sid);
-    SimSlot* b = slot(other);
+    SimSlotPtr a = slot(sid);
+    SimSlotPtr b = slot(other);
     if (!a || !b) return 0;
+    // Lock both slots deadlock-free (same handle twice: single lock).
+    std::unique_lock<std::mutex> la(a->op, std::defer_lock);
+    std::unique_lock<std::mutex> lb;
+    if (a.get() == b.get()) {
+        la.lock();
+    } else {
+        lb = std::unique_lock<std::mutex>(b->op, std::defer_lock);
+        std::lock(la, lb);
+    }
     try {
         if (a->f && b->f) a->f->Compose(b->f);
         if (a->d && b->d) a->d->Compose(b->d);
@@ -594,8 +616,9 @@ quid qrack_compose(quid sid, quid other)
 
 int qrack_qstabilizer_out_to_file(quid sid, const char* path)
 {
-    SimSlot* s = slot(sid);
+    SimSlotPtr s = slot(sid);
     if (!s) return -1;
+    std::lock_guard<std::mutex> lk(s->op);
     try {
         std::string text;
         if (s->f) text = SaveStabilizerText<float>(s->f);
@@ -614,12 +637,9 @@ quid qrack_qstabilizer_in_from_file(const char* path)
     try {
         std::ifstream f(path);
         std::string text((std::istreambuf_iterator<char>(f)), std::istreambuf_iterator<char>());
-        SimSlot s;
-        s.f = LoadStabilizerText<float>(text, nullptr);
-        std::lock_guard<std::mutex> lk(g_mtx);
-        const quid sid = g_next++;
-        g_sims[sid] = std::move(s);
-        return sid;
+        SimSlotPtr s = std::make_shared<SimSlot>();
+        s->f = LoadStabilizerText<float>(text, nullptr);
+        return registerSlot(std::move(s));
     } catch (const std::exception&) {
         return 0;
     }
@@ -627,8 +647,9 @@ quid qrack_qstabilizer_in_from_file(const char* path)
 
 int qrack_lossy_out_to_file(quid sid, const char* path)
 {
-    SimSlot* s = slot(sid);
+    SimSlotPtr s = slot(sid);
     if (!s) return -1;
+    std::lock_guard<std::mutex> lk(s->op);
     try {
         if (s->f) LossySaveState<float>(s->f, path);
         if (s->d) LossySaveState<double>(s->d, path);
@@ -641,8 +662,9 @@ int qrack_lossy_out_to_file(quid sid, const char* path)
 
 int qrack_lossy_in_from_file(quid sid, const char* path)
 {
-    SimSlot* s = slot(sid);
+    SimSlotPtr s = slot(sid);
     if (!s) return -1;
+    std::lock_guard<std::mutex> lk(s->op);
     try {
         if (s->f) LossyLoadState<float>(s->f, path);
         if (s->d) LossyLoadState<double>(s->d, path);

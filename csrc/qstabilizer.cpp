@@ -1009,16 +1009,19 @@ template <typename R> double QStabilizer<R>::SumSqrDiff(QInterfacePtr<R> other)
 
 template <typename R> std::string QStabilizer<R>::Serialize() const
 {
-    // text stream: qubit count, then per row "x-bits z-bits r" (parity model:
-    // qstabilizer.cpp:3407-3489 tableau dump)
+    // Reference-interchangeable text stream (operator<< at the reference's
+    // qstabilizer.cpp:3407-3437): qubit count, then 2n rows of
+    // "x0 x1 ... z0 z1 ... r" with space-separated single bits. The reference
+    // canonicalizes with gaussian(false) before dumping; we dump the tableau
+    // as-is (both canonical and non-canonical tableaus parse identically and
+    // describe the same state up to row operations).
     std::ostringstream os;
     const bitLenInt n = qubitCount;
     os << (uint64_t)n << "\n";
     for (size_t i = 0; i < 2u * (size_t)n; ++i) {
-        for (bitLenInt q = 0; q < n; ++q) os << (getX(i, q) ? '1' : '0');
-        os << ' ';
-        for (bitLenInt q = 0; q < n; ++q) os << (getZ(i, q) ? '1' : '0');
-        os << ' ' << (int)rPhase[i] << "\n";
+        for (bitLenInt q = 0; q < n; ++q) os << (getX(i, q) ? 1 : 0) << ' ';
+        for (bitLenInt q = 0; q < n; ++q) os << (getZ(i, q) ? 1 : 0) << ' ';
+        os << (int)rPhase[i] << "\n";
     }
     return os.str();
 }
@@ -1029,15 +1032,37 @@ template <typename R> QStabilizerPtr<R> QStabilizer<R>::Deserialize(const std::s
     uint64_t n = 0;
     is >> n;
     auto q = std::make_shared<QStabilizer<R>>((bitLenInt)n, 0u, rgp);
+    // Accept both the reference's space-separated bit rows and the legacy
+    // qrack_amd packed rows ("0101 0011 r") for old checkpoints.
+    std::string tok;
     for (size_t i = 0; i < 2u * n; ++i) {
-        std::string xs, zs;
-        int r;
-        is >> xs >> zs >> r;
-        for (uint64_t j = 0; j < n; ++j) {
-            q->setX(i, (bitLenInt)j, xs[j] == '1');
-            q->setZ(i, (bitLenInt)j, zs[j] == '1');
+        if (!(is >> tok)) break;
+        if (n > 1 && tok.size() == n) {
+            // legacy packed: x-string, z-string, r
+            std::string zs;
+            int r;
+            is >> zs >> r;
+            for (uint64_t j = 0; j < n; ++j) {
+                q->setX(i, (bitLenInt)j, tok[j] == '1');
+                q->setZ(i, (bitLenInt)j, zs[j] == '1');
+            }
+            q->rPhase[i] = (uint8_t)r;
+        } else {
+            // reference space-separated: 2n single bits then r
+            q->setX(i, 0, tok == "1");
+            int b;
+            for (uint64_t j = 1; j < n; ++j) {
+                is >> b;
+                q->setX(i, (bitLenInt)j, b != 0);
+            }
+            for (uint64_t j = 0; j < n; ++j) {
+                is >> b;
+                q->setZ(i, (bitLenInt)j, b != 0);
+            }
+            int r;
+            is >> r;
+            q->rPhase[i] = (uint8_t)r;
         }
-        q->rPhase[i] = (uint8_t)r;
     }
     return q;
 }

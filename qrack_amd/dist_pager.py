@@ -264,7 +264,6 @@ class DistQPager:
         if not local_c:
             self.q.global_phase(factor)
             return
-        diag = [1, 0, 0, factor]
         if len(local_c) == 1:
             if local_perm & 1:
                 self.q.phase(1, factor, local_c[0])

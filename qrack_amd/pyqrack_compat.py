@@ -42,6 +42,7 @@ class QrackSimulator:
         if cloneSim is not None:
             self.sim = cloneSim.sim.clone()
             self.num_qubits = cloneSim.num_qubits
+            self._shot_rng = __import__("random").Random()
             return
         layers = []
         if isTensorNetwork:
@@ -62,6 +63,7 @@ class QrackSimulator:
             layers.append("hip" if qa.hip_device_count() > 0 else "cpu")
         self.sim = qa.create_simulator(qubitCount, precision=precision, layers=layers, seed=seed)
         self.num_qubits = qubitCount
+        self._shot_rng = __import__("random").Random(seed if seed >= 0 else None)
         if pyzxCircuit is not None:
             raise NotImplementedError("pyzx circuits are not supported")
 
@@ -230,9 +232,9 @@ class QrackSimulator:
         out = []
         for val, cnt in res.items():
             out.extend([int(val)] * cnt)
-        import random
-
-        random.shuffle(out)
+        # Shuffle with the simulator-seeded RNG so seeded runs are reproducible
+        # (PyQrack returns shots in per-shot sampled order; we reconstruct one).
+        self._shot_rng.shuffle(out)
         return out
 
     def joint_ensemble_probability(self, b, q):

@@ -215,6 +215,32 @@ def test_stabilizer_save_load():
     assert q.approx_compare(q2)
 
 
+def test_stabilizer_text_reference_format():
+    # The tableau dump must use the reference's stream shape (operator<< at
+    # reference qstabilizer.cpp:3407-3437): "n\n" then 2n rows of 2n+1
+    # space-separated tokens "x0 .. xn-1 z0 .. zn-1 r".
+    q = qa.create_simulator(3, layers=["stabilizer"], seed=5)
+    q.h(0)
+    q.cnot(0, 1)
+    text = qa.save_stabilizer_F(q)
+    lines = [ln for ln in text.splitlines() if ln.strip()]
+    assert lines[0] == "3"
+    assert len(lines) == 1 + 6
+    for ln in lines[1:]:
+        toks = ln.split()
+        assert len(toks) == 7
+        assert all(t in ("0", "1", "2", "3") for t in toks)
+    # a reference-produced stream (space-separated bits) loads correctly
+    ref_stream = "2\n1 0 0 0 0\n0 1 0 0 0\n0 0 1 0 0\n0 0 0 1 0\n"
+    q2 = qa.load_stabilizer_F(ref_stream)
+    assert q2.num_qubits == 2
+    assert abs(q2.prob(0)) < 1e-6 and abs(q2.prob(1)) < 1e-6
+    # legacy packed rows ("0101 0011 r") still load (old checkpoints)
+    legacy = "2\n10 00 0\n01 00 0\n00 10 0\n00 01 0\n"
+    q3 = qa.load_stabilizer_F(legacy)
+    assert abs(q3.prob(0)) < 1e-6 and abs(q3.prob(1)) < 1e-6
+
+
 def test_stabilizer_hybrid_save():
     q = qa.create_simulator(2, layers=["stabilizer_hybrid", "cpu"], seed=2)
     q.h(0)
