@@ -129,6 +129,24 @@ def main():
     gates_per_sec = n_gates * args.steps / elapsed
     single_gate_ms = ms_per_step / n_gates
 
+    # Isolated single-gate wall-clock (the BASELINE.md metric): one H on a
+    # low qubit, synchronized each application — not amortized across fused
+    # QFT columns. Measured on the single-rank path only.
+    isolated_h_ms = None
+    if world == 1:
+        sim.set_permutation(0)
+        reps = 10
+        sim.h(0)
+        if hasattr(sim, "finish"):
+            sim.finish()
+        th0 = time.perf_counter()
+        for _ in range(reps):
+            sim.h(0)
+            if hasattr(sim, "finish"):
+                sim.finish()
+        th1 = time.perf_counter()
+        isolated_h_ms = 1000.0 * (th1 - th0) / reps
+
     if rank == 0:
         out = {
             "metric": "qft_gates_per_sec",
@@ -149,6 +167,7 @@ def main():
                 "base_qubits_per_gpu": base_qubits,
                 "n_gates": n_gates,
                 "single_gate_ms": single_gate_ms,
+                "isolated_h_ms": isolated_h_ms,
                 "engine": engine,
                 "global_batch": 1,
                 "seq_len": qubits,

