@@ -574,6 +574,22 @@ template <typename R> static void bindQInterface(py::module_& m, const char* nam
             py::arg("target"), py::arg("scale"), py::arg("ramp_start"),
             py::arg("in_place_mask"), py::arg("pows"), py::arg("weights"),
             py::arg("phase0") = 0.0, py::arg("pre") = false)
+#ifdef QRACK_AMD_HIP_ENGINE
+        .def("qft_column_top_range",
+            [](Ptr q, double scale, bitLenInt rampStart, bitCapInt inPlaceRelMask,
+                std::vector<bitCapInt> sPows, std::vector<uint64_t> sWeights, double phase0,
+                bool pre, uint64_t itLo, uint64_t itHi, uintptr_t recvPtr, bool recvIsLow,
+                uintptr_t extStream) {
+                auto eng = std::dynamic_pointer_cast<QEngineHIP<R>>(q);
+                if (!eng) throw QrackError("qft_column_top_range requires the HIP engine");
+                eng->QftColumnTopRange(scale, rampStart, inPlaceRelMask, sPows, sWeights, phase0,
+                    pre, itLo, itHi, recvPtr, recvIsLow, extStream);
+            },
+            py::arg("scale"), py::arg("ramp_start"), py::arg("in_place_mask"), py::arg("pows"),
+            py::arg("weights"), py::arg("phase0"), py::arg("pre"), py::arg("it_lo"),
+            py::arg("it_hi"), py::arg("recv_ptr"), py::arg("recv_is_low"),
+            py::arg("ext_stream") = 0)
+#endif
         .def("norm_total",
             [](Ptr q) {
                 auto eng = std::dynamic_pointer_cast<QEngine<R>>(q);

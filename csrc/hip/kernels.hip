@@ -1241,6 +1241,130 @@ void launchQftColumnGeneral(cplx<R>* sv, bitCapInt maxQPower, bitCapInt tPow, co
     }
 }
 
+// Ranged top-target fused column with receive-buffer fusion (pipelined
+// distributed page exchange; see kernels.hpp). Pair rows r in [itLo, itHi):
+//   x (target=0 side) = recvIsLow ? recvSrc[r-itLo] : sv[r]
+//   y (target=1 side) = recvIsLow ? sv[r|tPow]      : recvSrc[r-itLo]
+// outputs written in place to sv[r], sv[r|tPow]. Because the target is the
+// TOP bit, the ramp fraction depends only on r (< tPow) and the pair map is
+// the identity — no bit interleave needed.
+template <typename R, bool PRE, bool RECV_LOW>
+__global__ void k_qft_col_top_range(cplx<R>* sv, bitCapInt tPow, RampArgs a, R phase0,
+    bitCapInt itLo, bitCapInt itHi, const cplx<R>* __restrict__ recvSrc)
+{
+    const R s = (R)0.70710678118654752440;
+    const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
+    for (bitCapInt r = itLo + (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; r < itHi;
+         r += stride) {
+        uint64_t frac = (uint64_t)((r >> a.rampStart) & a.inPlaceRelMask);
+        for (int k = 0; k < a.nScattered; ++k) {
+            if (r & a.sPow[k]) frac += a.sWeight[k];
+        }
+        R sn, cs;
+        devSinCos<R>((R)a.scale * (R)frac + phase0, &sn, &cs);
+        const cplx<R> f{ cs, sn };
+        cplx<R> x = RECV_LOW ? recvSrc[r - itLo] : sv[r];
+        cplx<R> y = RECV_LOW ? sv[r | tPow] : recvSrc[r - itLo];
+        if (PRE) y = f * y;
+        cplx<R> o0 = s * (x + y);
+        cplx<R> o1 = s * (x - y);
+        if (!PRE) o1 = f * o1;
+        sv[r] = o0;
+        sv[r | tPow] = o1;
+    }
+}
+
+// float4-vectorized variant: two adjacent pair rows per thread (requires
+// even itLo/itHi, which the chunking always produces for qpp >= 2)
+template <bool PRE, bool RECV_LOW>
+__global__ void k_qft_col_top_range_v(cplx<float>* sv, bitCapInt tPow, RampArgs a, float phase0,
+    bitCapInt itLo, bitCapInt itHi, const cplx<float>* __restrict__ recvSrc)
+{
+    float4* sv4 = reinterpret_cast<float4*>(sv);
+    const float4* rc4 = reinterpret_cast<const float4*>(recvSrc);
+    const float s = 0.70710678f;
+    const bitCapInt lo4 = itLo >> 1u, hi4 = itHi >> 1u;
+    const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
+    for (bitCapInt k = lo4 + (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; k < hi4;
+         k += stride) {
+        const bitCapInt r = 2u * k;
+        const bitCapInt t4 = (r | tPow) >> 1u;
+        float4 vlo = RECV_LOW ? rc4[k - lo4] : sv4[k];
+        float4 vhi = RECV_LOW ? sv4[t4] : rc4[k - lo4];
+        uint64_t fr0 = (uint64_t)((r >> a.rampStart) & a.inPlaceRelMask);
+        uint64_t fr1 = (uint64_t)(((r + 1u) >> a.rampStart) & a.inPlaceRelMask);
+        for (int t = 0; t < a.nScattered; ++t) {
+            if (r & a.sPow[t]) fr0 += a.sWeight[t];
+            if ((r + 1u) & a.sPow[t]) fr1 += a.sWeight[t];
+        }
+        float s0, c0, s1, c1;
+        __sincosf((float)a.scale * (float)fr0 + phase0, &s0, &c0);
+        __sincosf((float)a.scale * (float)fr1 + phase0, &s1, &c1);
+        const cplx<float> f0{ c0, s0 }, f1{ c1, s1 };
+        cplx<float> x0{ vlo.x, vlo.y }, x1{ vlo.z, vlo.w };
+        cplx<float> y0{ vhi.x, vhi.y }, y1{ vhi.z, vhi.w };
+        if (PRE) {
+            y0 = f0 * y0;
+            y1 = f1 * y1;
+        }
+        cplx<float> a0 = s * (x0 + y0), a1 = s * (x1 + y1);
+        cplx<float> b0 = s * (x0 - y0), b1 = s * (x1 - y1);
+        if (!PRE) {
+            b0 = f0 * b0;
+            b1 = f1 * b1;
+        }
+        sv4[k] = make_float4(a0.re, a0.im, a1.re, a1.im);
+        sv4[t4] = make_float4(b0.re, b0.im, b1.re, b1.im);
+    }
+}
+
+template <typename R>
+void launchQftColumnTopRange(cplx<R>* sv, bitCapInt maxQPower, const RampArgs& a, double phase0,
+    bool pre, bitCapInt itLo, bitCapInt itHi, const cplx<R>* recvSrc, bool recvIsLow,
+    hipStream_t stream)
+{
+    const bitCapInt tPow = maxQPower >> 1u;
+    const bitCapInt n = itHi - itLo;
+    if (!n) return;
+    if constexpr (std::is_same_v<R, float>) {
+        if ((itLo & 1u) == 0u && (itHi & 1u) == 0u) {
+            const dim3 g(gridFor(n >> 1u)), b(QA_BLOCK);
+            if (pre) {
+                if (recvIsLow)
+                    hipLaunchKernelGGL((k_qft_col_top_range_v<true, true>), g, b, 0, stream, sv,
+                        tPow, a, (float)phase0, itLo, itHi, recvSrc);
+                else
+                    hipLaunchKernelGGL((k_qft_col_top_range_v<true, false>), g, b, 0, stream, sv,
+                        tPow, a, (float)phase0, itLo, itHi, recvSrc);
+            } else {
+                if (recvIsLow)
+                    hipLaunchKernelGGL((k_qft_col_top_range_v<false, true>), g, b, 0, stream, sv,
+                        tPow, a, (float)phase0, itLo, itHi, recvSrc);
+                else
+                    hipLaunchKernelGGL((k_qft_col_top_range_v<false, false>), g, b, 0, stream, sv,
+                        tPow, a, (float)phase0, itLo, itHi, recvSrc);
+            }
+            return;
+        }
+    }
+    const dim3 g(gridFor(n)), b(QA_BLOCK);
+    if (pre) {
+        if (recvIsLow)
+            hipLaunchKernelGGL((k_qft_col_top_range<R, true, true>), g, b, 0, stream, sv, tPow, a,
+                (R)phase0, itLo, itHi, recvSrc);
+        else
+            hipLaunchKernelGGL((k_qft_col_top_range<R, true, false>), g, b, 0, stream, sv, tPow,
+                a, (R)phase0, itLo, itHi, recvSrc);
+    } else {
+        if (recvIsLow)
+            hipLaunchKernelGGL((k_qft_col_top_range<R, false, true>), g, b, 0, stream, sv, tPow,
+                a, (R)phase0, itLo, itHi, recvSrc);
+        else
+            hipLaunchKernelGGL((k_qft_col_top_range<R, false, false>), g, b, 0, stream, sv, tPow,
+                a, (R)phase0, itLo, itHi, recvSrc);
+    }
+}
+
 template <typename R>
 void launchQftColumn(cplx<R>* sv, bitCapInt maxQPower, bitLenInt rampStart, bitLenInt col,
     bitCapInt tPow, int sign, bool pre, hipStream_t stream)
@@ -1863,6 +1987,8 @@ void launchPartProbs(const cplx<R>* sv, bitCapInt maxQPower, bitLenInt start, bi
     template void launchMtrx2qBatchLds<R>(cplx<R>*, const Batch2qLdsArgs<R>&, hipStream_t);                \
     template void launchMtrx2q<R>(cplx<R>*, const Gate4x4Args<R>&, hipStream_t);                              \
     template void launchCPhasePairs<R>(cplx<R>*, const CPhasePairsArgs&, hipStream_t);                               \
+    template void launchQftColumnTopRange<R>(cplx<R>*, bitCapInt, const RampArgs&, double, bool,    \
+        bitCapInt, bitCapInt, const cplx<R>*, bool, hipStream_t);                                   \
     template void launchQftColumnGeneral<R>(                                                        \
         cplx<R>*, bitCapInt, bitCapInt, const RampArgs&, double, bool, hipStream_t);
 

@@ -189,6 +189,21 @@ template <typename R>
 void launchQftColumnGeneral(cplx<R>* sv, bitCapInt maxQPower, bitCapInt tPow, const RampArgs& a,
     double phase0, bool pre, hipStream_t stream);
 
+// Ranged top-target fused QFT column for the distributed pager's PIPELINED
+// page exchange: processes pair rows r in [itLo, itHi) of the H on the TOP
+// local qubit (tPow = maxQPower/2), reading ONE side of each pair straight
+// from the RCCL receive buffer `recvSrc` (indexed r - itLo) instead of the
+// state vector — the received chunk is consumed in place, no staging copy.
+// recvIsLow: true when the received chunk is the pair's low (target=0) side
+// (the HIGH page of an exchange receives the low side). Launched per chunk
+// on the caller-supplied stream (torch's current stream) so NCCL chunk c+1
+// overlaps with chunk c's compute. Replaces the host-staged ShuffleBuffers
+// swap of the reference (opencl.cpp:254-264) with exchange+apply fusion.
+template <typename R>
+void launchQftColumnTopRange(cplx<R>* sv, bitCapInt maxQPower, const RampArgs& a, double phase0,
+    bool pre, bitCapInt itLo, bitCapInt itHi, const cplx<R>* recvSrc, bool recvIsLow,
+    hipStream_t stream);
+
 // batched independent single-qubit gates: k distinct-target 2x2s applied in
 // ONE full-state pass (2^k-amplitude orbits in registers). The memory-bound
 // fusion win: k passes -> 1. fp32 supports k in [2,5], fp64 [2,4].

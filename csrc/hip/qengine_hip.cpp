@@ -759,6 +759,29 @@ void QEngineHIP<R>::QftColumnGeneral(bitLenInt target, double scale, bitLenInt r
 }
 
 template <typename R>
+void QEngineHIP<R>::QftColumnTopRange(double scale, bitLenInt rampStart, bitCapInt inPlaceRelMask,
+    const std::vector<bitCapInt>& sPows, const std::vector<uint64_t>& sWeights, double phase0,
+    bool pre, uint64_t itLo, uint64_t itHi, uintptr_t recvPtr, bool recvIsLow, uintptr_t extStream)
+{
+    if (sPows.size() > 8u) throw QrackError("QftColumnTopRange: more than 8 relocated bits");
+    if (!recvPtr) throw QrackError("QftColumnTopRange: null receive buffer");
+    RampArgs a{};
+    a.rampStart = rampStart;
+    a.inPlaceRelMask = inPlaceRelMask;
+    a.nScattered = (int)sPows.size();
+    for (size_t k = 0; k < sPows.size(); ++k) {
+        a.sPow[k] = sPows[k];
+        a.sWeight[k] = sWeights[k];
+    }
+    a.condPow = 0;
+    a.scale = scale;
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    hipStream_t st = extStream ? (hipStream_t)extStream : stream;
+    launchQftColumnTopRange<R>(dState, maxQPower, a, phase0, pre, (bitCapInt)itLo,
+        (bitCapInt)itHi, (const cplx<R>*)recvPtr, recvIsLow, st);
+}
+
+template <typename R>
 void QEngineHIP<R>::PhaseRampGeneral(R scale, bitLenInt rampStart, bitCapInt inPlaceRelMask,
     const std::vector<bitCapInt>& sPows, const std::vector<uint64_t>& sWeights, bitCapInt condPower)
 {

@@ -154,6 +154,16 @@ public:
     void QftColumnGeneral(bitLenInt target, double scale, bitLenInt rampStart,
         bitCapInt inPlaceRelMask, const std::vector<bitCapInt>& sPows,
         const std::vector<uint64_t>& sWeights, double phase0, bool pre) override;
+    // Pipelined-exchange building block (distributed pager): fused top-target
+    // column over pair rows [itLo, itHi), one pair side read straight from
+    // the RCCL receive buffer `recvPtr` (device pointer, chunk-local), and the
+    // launch placed on `extStream` (torch's current HIP stream) so per-chunk
+    // compute interleaves with the in-flight NCCL chunks. recvIsLow: received
+    // chunk is the target=0 side (true on the high page of an exchange).
+    void QftColumnTopRange(double scale, bitLenInt rampStart, bitCapInt inPlaceRelMask,
+        const std::vector<bitCapInt>& sPows, const std::vector<uint64_t>& sWeights, double phase0,
+        bool pre, uint64_t itLo, uint64_t itHi, uintptr_t recvPtr, bool recvIsLow,
+        uintptr_t extStream);
 
     // ---- probability / measurement ----
     R Prob(bitLenInt q) override;

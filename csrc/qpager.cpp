@@ -406,7 +406,7 @@ template <typename R> void QPager<R>::QFT(bitLenInt start, bitLenInt length, boo
             const bitLenInt metaStart = (start < qpp) ? 0u : (start - qpp);
             for (bitCapInt p = 0; p < PageCount(); ++p) {
                 if (!((p >> (t - qpp)) & 1u)) continue;
-                if (intraBits) p /* keep */, qPages[p]->PhaseRamp(scale, start, intraBits, 0u);
+                if (intraBits) qPages[p]->PhaseRamp(scale, start, intraBits, 0u);
                 // meta contribution: bits metaStart .. (t-qpp-1) of the page
                 const bitCapInt metaVal =
                     (p >> metaStart) & (pow2((bitLenInt)(t - qpp - metaStart)) - 1u);

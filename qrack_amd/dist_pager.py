@@ -48,6 +48,14 @@ class DistQPager:
         self.logical_at = list(range(qubits))  # slot -> logical qubit
         self.rng = np.random.default_rng(seed)  # replicated decision RNG
         self.page_len = 1 << self.qpp
+        # pipelined-exchange chunk count (meta QFT columns): the half-page
+        # sendrecv is split into this many NCCL chunks, each chunk's fused
+        # column kernel launching on torch's stream as soon as it lands —
+        # comm for chunk c+1 overlaps compute for chunk c
+        import os as _os
+
+        self.pipe_chunks = max(1, int(_os.environ.get("QRACK_DIST_PIPE_CHUNKS", "4")))
+        self.pipe_enable = _os.environ.get("QRACK_DIST_PIPE", "1") != "0"  # A/B switch
 
     # ---- helpers ------------------------------------------------------------
 
@@ -442,10 +450,84 @@ class DistQPager:
             in_place = 0
         return ramp_start, in_place, scattered, meta_weight
 
+    def _nccl_active(self):
+        cfg = (dist.get_backend_config() if hasattr(dist, "get_backend_config")
+               else str(dist.get_backend()))
+        return "nccl" in cfg
+
+    def _fused_column_meta_pipelined(self, start, i, sign, pre):
+        """Meta-target column with the exchange PIPELINED against compute:
+        the half-page sendrecv is chunked; as each NCCL chunk lands, the
+        ranged fused column kernel consumes it straight from the receive
+        buffer (exchange+apply fusion, no staging copy) on torch's stream,
+        while the next chunk is still on the xGMI link. The concurrent page
+        pairs of the exchange are disjoint rank pairs, so with one process
+        per GPU they ride disjoint xGMI links by construction (the reference
+        instead host-stages each pair serially, opencl.cpp:254-264)."""
+        t_logical = start + i
+        s = self.slot_of[t_logical]
+        tb = s - self.qpp
+        tb_bit = 1 << tb
+        partner = self._rank_of_page(self.my_page ^ tb_bit)
+        i_am_low = ((self.my_page >> tb) & 1) == 0
+        # record the exchange-realized swap FIRST: the ramp decomposition
+        # must see the post-exchange map (target at local top)
+        self._swap_slots(self.qpp - 1, s)
+        if i:
+            rs, in_place, scattered, meta_w = self._ramp_parts(start, i)
+        else:
+            rs, in_place, scattered, meta_w = 0, 0, [], 0
+        if len(scattered) > 8:
+            # rare wide-scatter fallback: plain exchange then unfused column
+            self._shuffle(partner, i_am_low)
+            sH = 1 / np.sqrt(2)
+            if pre:
+                self._column_ramp(start, i, sign)
+                self._dispatch([sH, sH, sH, -sH], self.qpp - 1, [], 0)
+            else:
+                self._dispatch([sH, sH, sH, -sH], self.qpp - 1, [], 0)
+                self._column_ramp(start, i, sign)
+            return
+        scale = sign * np.pi / (1 << i)
+        pows = [p for p, _ in scattered]
+        ws = [w for _, w in scattered]
+        self.q.finish()
+        half = self.page_len // 2
+        view = self._half_view(low_half=not i_am_low)
+        tmp = torch.empty_like(view)
+        # power-of-two chunk count dividing the (power-of-two) half length,
+        # so every chunk boundary stays even for the float4 kernel path
+        n_chunks = 1 << (self.pipe_chunks.bit_length() - 1)
+        while n_chunks > 1 and ((half // n_chunks) < (1 << 16) or half % n_chunks):
+            n_chunks //= 2
+        step = half // n_chunks
+        bounds = [(c * step, (c + 1) * step) for c in range(n_chunks)]
+        reqs = []
+        for lo, hi in bounds:
+            reqs.append(dist.batch_isend_irecv([
+                dist.P2POp(dist.isend, view[lo:hi], partner),
+                dist.P2POp(dist.irecv, tmp[lo:hi], partner),
+            ]))
+        ext = torch.cuda.current_stream(self.device_id).cuda_stream
+        elem = tmp.element_size()
+        recv_is_low = not i_am_low  # high page receives the target=0 side
+        base = tmp.data_ptr()
+        for (lo, hi), rq in zip(bounds, reqs):
+            for r in rq:
+                r.wait()  # stream-wait on torch's current stream (NCCL)
+            self.q.qft_column_top_range(
+                float(scale), rs, in_place, pows, ws, float(scale * meta_w), pre,
+                lo, hi, base + lo * elem, recv_is_low, ext)
+        torch.cuda.synchronize(self.device_id)
+
     def _fused_column(self, start, i, sign, pre):
         """One engine pass per column: realize the target locally (one
         exchange at most), then H + the whole (relocated) phase ladder +
         the meta scalar in a single fused kernel."""
+        if (self.pipe_enable and self.slot_of[start + i] >= self.qpp and self._is_hip()
+                and self._nccl_active() and hasattr(self.q, "qft_column_top_range")):
+            self._fused_column_meta_pipelined(start, i, sign, pre)
+            return
         t_slot = self._realize_local(start + i)
         if i == 0:
             s = 1 / np.sqrt(2)
