@@ -206,8 +206,15 @@ template <typename R> void QUnit<R>::GetQuantumState(cplx<R>* outputState)
 
 template <typename R> cplx<R> QUnit<R>::GetAmplitude(bitCapInt perm)
 {
-    // product over independent units
+    // invert buffers change which basis states carry weight: land them.
+    // CP buffers only contribute phases, folded in below without flushing.
+    FlushInvAll();
     cplx<R> amp(1, 0);
+    for (const auto& p : pendingPairs) {
+        if (((perm >> p.c) & 1u) && ((perm >> p.t) & 1u)) {
+            amp = amp * polar<R>(1, (R)p.angle);
+        }
+    }
     std::set<QInterfacePtr<R>> seen;
     for (bitLenInt q = 0; q < qubitCount; ++q) {
         QInterfacePtr<R> u = shards[q].unit;
@@ -234,27 +241,43 @@ template <typename R> void QUnit<R>::SetAmplitude(bitCapInt perm, cplx<R> amp)
 
 template <typename R> void QUnit<R>::Mtrx(const cplx<R>* m, bitLenInt t)
 {
-    if (norm(m[1]) > (R)1e-24 || norm(m[2]) > (R)1e-24) FlushPhasePairs(t);
-    shards[t].unit->Mtrx(m, shards[t].mapped);
+    // diagonal / anti-diagonal / exact-H gates COMMUTE through the pending
+    // buffers (possibly adjusting the forwarded matrix); everything else
+    // flushes t's buffers (reference Mtrx commutation, qunit.cpp:2433-2487)
+    cplx<R> mm[4] = { m[0], m[1], m[2], m[3] };
+    CommuteBuffers1q(t, mm);
+    shards[t].unit->Mtrx(mm, shards[t].mapped);
 }
 
 template <typename R> void QUnit<R>::Phase(cplx<R> tl, cplx<R> br, bitLenInt t)
 {
+    CommuteDiag(t, tl, br);
     shards[t].unit->Phase(tl, br, shards[t].mapped);
 }
 
 template <typename R> void QUnit<R>::Invert(cplx<R> tr, cplx<R> bl, bitLenInt t)
 {
-    FlushPhasePairs(t);
+    CommuteInvert(t, tr, bl);
     shards[t].unit->Invert(tr, bl, shards[t].mapped);
 }
 
 template <typename R>
 void QUnit<R>::MCMtrx(const std::vector<bitLenInt>& controls, const cplx<R>* m, bitLenInt t)
 {
-    // a pending pair on a CONTROL commutes (controlled ops are diagonal on
-    // the control); only the target's pairs must flush
+    // route diagonal / anti-diagonal payloads to the buffering paths
+    if (norm(m[1]) <= (R)1e-24 && norm(m[2]) <= (R)1e-24) {
+        MCPhase(controls, m[0], m[3], t);
+        return;
+    }
+    if (norm(m[0]) <= (R)1e-24 && norm(m[3]) <= (R)1e-24) {
+        MCInvert(controls, m[1], m[2], t);
+        return;
+    }
+    // a pending CP/controlling buffer on a CONTROL commutes (controlled ops
+    // are diagonal on the control); buffers targeting a control, and all of
+    // the target's buffers, must flush
     FlushPhasePairs(t);
+    for (bitLenInt c : controls) FlushInvTargeting(c);
     std::vector<bitLenInt> live;
     for (bitLenInt c : controls) {
         bool on = false;
@@ -288,6 +311,7 @@ template <typename R>
 void QUnit<R>::MACMtrx(const std::vector<bitLenInt>& controls, const cplx<R>* m, bitLenInt t)
 {
     FlushPhasePairs(t);
+    for (bitLenInt c : controls) FlushInvTargeting(c);
     std::vector<bitLenInt> live;
     for (bitLenInt c : controls) {
         bool on = false;
@@ -321,6 +345,10 @@ template <typename R>
 void QUnit<R>::MCPhase(
     const std::vector<bitLenInt>& controls, cplx<R> tl, cplx<R> br, bitLenInt t)
 {
+    // the gate is diagonal on every involved qubit: only buffers whose X
+    // TARGET is involved fail to commute
+    FlushInvTargeting(t);
+    for (bitLenInt c : controls) FlushInvTargeting(c);
     // for diagonal gates, a deterministic |1> target also short-circuits
     std::vector<bitLenInt> live;
     for (bitLenInt c : controls) {
@@ -361,7 +389,7 @@ void QUnit<R>::MCPhase(
                 shards[live[0]].unit->Phase(
                     cplx<R>(1, 0), polar<R>(1, (R)at), shards[live[0]].mapped);
             }
-            BufferPhasePair(live[0], t, ab - at);
+            BufferCPhase(live[0], t, ab - at);
             return;
         }
     }
@@ -385,8 +413,53 @@ template <typename R>
 void QUnit<R>::MCInvert(
     const std::vector<bitLenInt>& controls, cplx<R> tr, cplx<R> bl, bitLenInt t)
 {
-    const cplx<R> m[4] = { cplx<R>(0, 0), tr, bl, cplx<R>(0, 0) };
-    MCMtrx(controls, m, t);
+    for (bitLenInt c : controls) FlushInvTargeting(c);
+    std::vector<bitLenInt> live;
+    for (bitLenInt c : controls) {
+        bool on = false;
+        if (ControlShortcut(c, false, on)) {
+            if (!on) return;
+        } else {
+            live.push_back(c);
+        }
+    }
+    if (live.empty()) {
+        Invert(tr, bl, t);
+        return;
+    }
+    if (live.size() == 1u && shards[live[0]].unit != shards[t].unit) {
+        const double mt = std::hypot((double)tr.re, (double)tr.im);
+        const double mb = std::hypot((double)bl.re, (double)bl.im);
+        if (std::abs(mt - 1.0) < 1e-9 && std::abs(mb - 1.0) < 1e-9) {
+            // defer the cross-unit controlled invert (reference isInvert
+            // PhaseShards, qengineshard.hpp:29-47): V = CX(c,t) ∘
+            // CP(arg(tr/bl)) ∘ Phase(1,bl)_c — the 1q phase lands now, the
+            // CX+CP buffer joins/composes with any pending same-pair buffer
+            const bitLenInt c = live[0];
+            const double at = std::atan2((double)tr.im, (double)tr.re);
+            const double ab = std::atan2((double)bl.im, (double)bl.re);
+            if (std::abs(ab) > 1e-14) {
+                shards[c].unit->Phase(cplx<R>(1, 0), polar<R>(1, (R)ab), shards[c].mapped);
+            }
+            BufferCInvert(c, t, at - ab);
+            return;
+        }
+    }
+    FlushPhasePairs(t);
+    std::vector<bitLenInt> qs(live);
+    qs.push_back(t);
+    try {
+        QInterfacePtr<R> unit = EntangleAll(qs);
+        std::vector<bitLenInt> mc;
+        for (bitLenInt c : live) mc.push_back(shards[c].mapped);
+        unit->MCInvert(mc, tr, bl, shards[t].mapped);
+        MaybeSeparate(t);
+        for (bitLenInt c : live) MaybeSeparate(c);
+    } catch (const std::bad_alloc&) {
+        bool applies = false;
+        ElideControls(live, false, applies);
+        if (applies) Invert(tr, bl, t);
+    }
 }
 
 template <typename R>
@@ -440,13 +513,12 @@ void QUnit<R>::UniformlyControlledSingleBit(
 template <typename R> void QUnit<R>::Swap(bitLenInt q1, bitLenInt q2)
 {
     if (q1 == q2) return;
-    // pending pairs follow the swapped labels
+    // pending buffers follow the swapped labels (roles travel with them)
     for (auto& p : pendingPairs) {
-        if (p.a == q1) p.a = q2;
-        else if (p.a == q2) p.a = q1;
-        if (p.b == q1) p.b = q2;
-        else if (p.b == q2) p.b = q1;
-        if (p.a > p.b) std::swap(p.a, p.b);
+        if (p.c == q1) p.c = q2;
+        else if (p.c == q2) p.c = q1;
+        if (p.t == q1) p.t = q2;
+        else if (p.t == q2) p.t = q1;
     }
     if (shards[q1].unit == shards[q2].unit) {
         shards[q1].unit->Swap(shards[q1].mapped, shards[q2].mapped);
@@ -526,11 +598,15 @@ template <typename R> void QUnit<R>::FSim(R theta, R phi, bitLenInt q1, bitLenIn
 
 template <typename R> R QUnit<R>::Prob(bitLenInt q)
 {
+    // pending CP buffers and q-as-control inverts never shift q's Z-basis
+    // marginal; only inverts TARGETING q must land first
+    FlushInvTargeting(q);
     return shards[q].unit->Prob(shards[q].mapped);
 }
 
 template <typename R> bool QUnit<R>::ForceM(bitLenInt q, bool result, bool doForce, bool doApply)
 {
+    FlushInvTargeting(q);
     Shard& s = shards[q];
     const R p1 = s.unit->Prob(s.mapped);
     bool outcome;
@@ -564,6 +640,7 @@ std::map<bitCapInt, int> QUnit<R>::MultiShotMeasureMask(
     // units are independent subsystems: sample each separately and combine
     // shots elementwise (no entanglement, no width blow-up)
     if (!shots) return {};
+    for (bitCapInt p : qPowers) FlushInvTargeting(log2Ocl(p));
     std::vector<bitCapInt> joint(shots, 0u);
     std::set<QInterfacePtr<R>> seen;
     for (size_t b = 0; b < qPowers.size(); ++b) {
@@ -604,6 +681,9 @@ std::map<bitCapInt, int> QUnit<R>::MultiShotMeasureMask(
 
 template <typename R> R QUnit<R>::ProbMask(bitCapInt mask, bitCapInt permutation)
 {
+    for (bitLenInt q = 0; q < qubitCount; ++q) {
+        if ((mask >> q) & 1u) FlushInvTargeting(q);
+    }
     // product over independent units
     double p = 1.0;
     std::set<QInterfacePtr<R>> seen;
@@ -627,6 +707,9 @@ template <typename R> R QUnit<R>::ProbMask(bitCapInt mask, bitCapInt permutation
 
 template <typename R> R QUnit<R>::ProbParity(bitCapInt mask)
 {
+    for (bitLenInt q = 0; q < qubitCount; ++q) {
+        if ((mask >> q) & 1u) FlushInvTargeting(q);
+    }
     double pOdd = 0.0;
     std::set<QInterfacePtr<R>> seen;
     for (bitLenInt q = 0; q < qubitCount; ++q) {
@@ -646,6 +729,9 @@ template <typename R> R QUnit<R>::ProbParity(bitCapInt mask)
 
 template <typename R> bool QUnit<R>::ForceMParity(bitCapInt mask, bool result, bool doForce)
 {
+    for (bitLenInt q = 0; q < qubitCount; ++q) {
+        if ((mask >> q) & 1u) FlushPhasePairs(q);
+    }
     std::vector<bitLenInt> qs;
     for (bitLenInt q = 0; q < qubitCount; ++q) {
         if ((mask >> q) & 1u) qs.push_back(q);
@@ -809,12 +895,13 @@ template <typename R> bitLenInt QUnit<R>::Compose(QInterfacePtr<R> toCopy, bitLe
         shards.insert(shards.begin() + start, ns.begin(), ns.end());
     }
     for (auto& p : pendingPairs) {
-        if (p.a >= start) p.a += oQubits;
-        if (p.b >= start) p.b += oQubits;
+        if (p.c >= start) p.c += oQubits;
+        if (p.t >= start) p.t += oQubits;
     }
     if (o) {
         for (const auto& p : o->pendingPairs) {
-            pendingPairs.push_back({ (bitLenInt)(p.a + start), (bitLenInt)(p.b + start), p.angle });
+            pendingPairs.push_back(
+                { (bitLenInt)(p.c + start), (bitLenInt)(p.t + start), p.angle, p.inv });
         }
     }
     this->SetQubitCount(qubitCount + oQubits);
@@ -841,8 +928,8 @@ template <typename R> void QUnit<R>::Decompose(bitLenInt start, QInterfacePtr<R>
     dest->SetQuantumState(buf.data());
     shards.erase(shards.begin() + start, shards.begin() + start + len);
     for (auto& p : pendingPairs) {
-        if (p.a >= start + len) p.a -= len;
-        if (p.b >= start + len) p.b -= len;
+        if (p.c >= start + len) p.c -= len;
+        if (p.t >= start + len) p.t -= len;
     }
     this->SetQubitCount(qubitCount - len);
 }
@@ -860,8 +947,8 @@ template <typename R> void QUnit<R>::Dispose(bitLenInt start, bitLenInt length)
     }
     shards.erase(shards.begin() + start, shards.begin() + start + length);
     for (auto& p : pendingPairs) {
-        if (p.a >= start + length) p.a -= length;
-        if (p.b >= start + length) p.b -= length;
+        if (p.c >= start + length) p.c -= length;
+        if (p.t >= start + length) p.t -= length;
     }
     this->SetQubitCount(qubitCount - length);
 }
@@ -884,8 +971,8 @@ template <typename R> bitLenInt QUnit<R>::Allocate(bitLenInt start, bitLenInt le
     }
     shards.insert(shards.begin() + start, ns.begin(), ns.end());
     for (auto& p : pendingPairs) {
-        if (p.a >= start) p.a += length;
-        if (p.b >= start) p.b += length;
+        if (p.c >= start) p.c += length;
+        if (p.t >= start) p.t += length;
     }
     this->SetQubitCount(qubitCount + length);
     return start;

@@ -38,76 +38,367 @@ protected:
     bool reactiveSeparate = false; // TrySeparate after entangling gates
     bitLenInt aceMaxQubits = 0; // 0 = unlimited; else entangle cap (ACE)
 
-    // deferred cross-unit controlled-phase pairs (the core of the reference's
-    // phase-shard optimization, qengineshard.hpp PhaseShards): a CPhase
-    // between SEPARATE units is buffered instead of entangling. Diagonal
-    // pending pairs leave every Z-basis probability/sampling query exact;
-    // they flush only when a non-diagonal gate (or state access / structural
-    // op) touches a buffered qubit. Same-pair buffers combine (and cancel).
-    struct PhasePair {
-        bitLenInt a, b; // logical ids, a < b
-        double angle;   // phase on |11>
+    // Deferred cross-unit 2-qubit buffers — the capability of the reference's
+    // PhaseShard machinery (qengineshard.hpp:29-73, commutation through Mtrx
+    // at qunit.cpp:2433-2487), re-designed as a flat pairwise-commuting
+    // buffer set instead of 4 per-shard maps:
+    //   buffer B(c,t; angle, inv) = [inv ? CX(c->t) : I] ∘ CP(angle)
+    // where CP(angle) = diag(1,1,1,e^{i*angle}) on (c,t). A controlled
+    // phase/invert between SEPARATE units is buffered instead of entangling;
+    // later 1-qubit diagonal, anti-diagonal and (exact-H) gates COMMUTE
+    // through the buffers with only angle updates and free 1-qubit phases,
+    // so circuit families like graph-state prep, QFT ladders and mirror
+    // sequences never touch an engine until something genuinely entangling
+    // arrives. Invariant: all pending buffers pairwise commute (enforced on
+    // insertion), so flush order is irrelevant.
+    struct PhaseBuffer {
+        bitLenInt c, t; // endpoints; for inv: c = control, t = X target
+        double angle;   // CP angle on |11>
+        bool inv;       // true: op = CX(c,t) ∘ CP(angle)
     };
-    std::vector<PhasePair> pendingPairs;
+    std::vector<PhaseBuffer> pendingPairs;
 
-    void BufferPhasePair(bitLenInt q1, bitLenInt q2, double angle)
+    static bool AngleZero(double a)
     {
-        if (q1 > q2) std::swap(q1, q2);
+        const double rem = std::fmod(a, 2.0 * 3.14159265358979323846);
+        return std::abs(rem) < 1e-12 ||
+            std::abs(std::abs(rem) - 2.0 * 3.14159265358979323846) < 1e-12;
+    }
+    static bool AnglePi(double a)
+    {
+        const double rem = std::fmod(std::abs(a), 2.0 * 3.14159265358979323846);
+        return std::abs(rem - 3.14159265358979323846) < 1e-12;
+    }
+
+    // do two buffers commute as operators? (CPs always do; an X target must
+    // not collide with the other buffer's diagonal support)
+    static bool BuffersCommute(const PhaseBuffer& x, const PhaseBuffer& y)
+    {
+        if (!x.inv && !y.inv) return true;
+        // x's X target collides if it's in y's diagonal support; for inv y
+        // the diagonal support is {y.c} plus the CP part touches y.t too
+        auto collides = [](const PhaseBuffer& a, const PhaseBuffer& b) {
+            if (!a.inv) return false;
+            // a has X on a.t; b is diagonal on b.c and (if CP angle or !inv)
+            // on b.t, except b's own X target b.t when b.inv && AngleZero
+            if (b.c == a.t) return true;
+            if (b.t == a.t) {
+                if (!b.inv) return true;                 // CP diagonal on b.t
+                if (!AngleZero(b.angle)) return true;    // CP part of inv b
+                return false;                            // pure CX same target: X⊗X ok
+            }
+            return false;
+        };
+        return !collides(x, y) && !collides(y, x);
+    }
+
+    int FindBuffer(bitLenInt a, bitLenInt b) const
+    {
         for (size_t i = 0; i < pendingPairs.size(); ++i) {
-            if (pendingPairs[i].a == q1 && pendingPairs[i].b == q2) {
-                pendingPairs[i].angle += angle;
-                const double rem = std::fmod(pendingPairs[i].angle, 2.0 * 3.14159265358979323846);
-                if (std::abs(rem) < 1e-12 || std::abs(std::abs(rem) - 2.0 * 3.14159265358979323846) < 1e-12) {
-                    pendingPairs.erase(pendingPairs.begin() + i); // cancelled
+            const PhaseBuffer& p = pendingPairs[i];
+            if ((p.c == a && p.t == b) || (p.c == b && p.t == a)) return (int)i;
+        }
+        return -1;
+    }
+
+    void ApplyBufferNow(const PhaseBuffer& p)
+    {
+        try {
+            QInterfacePtr<R> unit = EntangleAll({ p.c, p.t });
+            if (!AngleZero(p.angle)) {
+                unit->MCPhase({ shards[p.c].mapped }, cplx<R>(1, 0), polar<R>(1, (R)p.angle),
+                    shards[p.t].mapped);
+            }
+            if (p.inv) {
+                unit->MCInvert(
+                    { shards[p.c].mapped }, cplx<R>(1, 0), cplx<R>(1, 0), shards[p.t].mapped);
+            }
+        } catch (const std::bad_alloc&) {
+            // ACE: the deferred coupler hit the entangle cap — classically
+            // collapse the control with fidelity bookkeeping (ElideCz)
+            bool applies = false;
+            ElideControls({ p.c }, false, applies);
+            if (applies) {
+                if (p.inv) {
+                    Invert(polar<R>(1, (R)p.angle), cplx<R>(1, 0), p.t);
+                } else if (!AngleZero(p.angle)) {
+                    Phase(cplx<R>(1, 0), polar<R>(1, (R)p.angle), p.t);
                 }
-                return;
             }
         }
-        pendingPairs.push_back({ q1, q2, angle });
     }
 
-    void ApplyPairNow(const PhasePair& p)
-    {
-        QInterfacePtr<R> unit = EntangleAll({ p.a, p.b });
-        unit->MCPhase(
-            { shards[p.a].mapped }, cplx<R>(1, 0), polar<R>(1, (R)p.angle), shards[p.b].mapped);
-    }
-
-    // flush every pending pair touching q (non-diagonal op incoming)
+    // flush every pending buffer touching q
     void FlushPhasePairs(bitLenInt q)
     {
         if (pendingPairs.empty()) return;
-        std::vector<PhasePair> todo;
+        std::vector<PhaseBuffer> todo;
         for (size_t i = pendingPairs.size(); i-- > 0;) {
-            if (pendingPairs[i].a == q || pendingPairs[i].b == q) {
+            if (pendingPairs[i].c == q || pendingPairs[i].t == q) {
                 todo.push_back(pendingPairs[i]);
                 pendingPairs.erase(pendingPairs.begin() + i);
             }
         }
-        for (const auto& p : todo) ApplyPairNow(p);
+        for (const auto& p : todo) ApplyBufferNow(p);
+    }
+
+    // flush only invert buffers whose X TARGET is q (q's unit marginals are
+    // stale only in that case; CP buffers and q-as-control never shift
+    // Z-basis probabilities of q)
+    void FlushInvTargeting(bitLenInt q)
+    {
+        if (pendingPairs.empty()) return;
+        std::vector<PhaseBuffer> todo;
+        for (size_t i = pendingPairs.size(); i-- > 0;) {
+            if (pendingPairs[i].inv && pendingPairs[i].t == q) {
+                todo.push_back(pendingPairs[i]);
+                pendingPairs.erase(pendingPairs.begin() + i);
+            }
+        }
+        for (const auto& p : todo) ApplyBufferNow(p);
+    }
+
+    void FlushInvAll()
+    {
+        if (pendingPairs.empty()) return;
+        std::vector<PhaseBuffer> todo;
+        for (size_t i = pendingPairs.size(); i-- > 0;) {
+            if (pendingPairs[i].inv) {
+                todo.push_back(pendingPairs[i]);
+                pendingPairs.erase(pendingPairs.begin() + i);
+            }
+        }
+        for (const auto& p : todo) ApplyBufferNow(p);
     }
 
     void FlushAllPhasePairs()
     {
-        std::vector<PhasePair> todo;
+        std::vector<PhaseBuffer> todo;
         todo.swap(pendingPairs);
-        for (const auto& p : todo) ApplyPairNow(p);
+        for (const auto& p : todo) ApplyBufferNow(p);
     }
 
-    // measurement resolution: q collapsed to `outcome` — each pending pair
-    // involving q degenerates to a 1-qubit phase on the partner (outcome=1)
-    // or vanishes (outcome=0)
+    // flush existing buffers that would not commute with a tentative new one
+    void FlushNonCommuting(const PhaseBuffer& nb)
+    {
+        if (pendingPairs.empty()) return;
+        std::vector<PhaseBuffer> todo;
+        for (size_t i = pendingPairs.size(); i-- > 0;) {
+            const PhaseBuffer& p = pendingPairs[i];
+            const bool samePair = (p.c == nb.c && p.t == nb.t) || (p.c == nb.t && p.t == nb.c);
+            if (samePair) continue; // same-pair composition handled by caller
+            if (!BuffersCommute(p, nb)) {
+                todo.push_back(p);
+                pendingPairs.erase(pendingPairs.begin() + i);
+            }
+        }
+        for (const auto& p : todo) ApplyBufferNow(p);
+    }
+
+    void PushBuffer(PhaseBuffer nb)
+    {
+        if (nb.inv || !AngleZero(nb.angle)) pendingPairs.push_back(nb);
+    }
+
+    // buffer CP(angle) on (q1,q2), composing with any same-pair buffer.
+    // Composition onto an inv buffer: CP(a)·CX·CP(θ) = CX·CP(θ-a)·P_c(a).
+    void BufferCPhase(bitLenInt q1, bitLenInt q2, double angle)
+    {
+        if (AngleZero(angle)) return;
+        const int i = FindBuffer(q1, q2);
+        if (i >= 0) {
+            PhaseBuffer& p = pendingPairs[(size_t)i];
+            if (!p.inv) {
+                p.angle += angle;
+                if (AngleZero(p.angle)) pendingPairs.erase(pendingPairs.begin() + i);
+                return;
+            }
+            // same-pair inv buffer (either orientation): incoming CP is
+            // symmetric, keep p's orientation
+            shards[p.c].unit->Phase(cplx<R>(1, 0), polar<R>(1, (R)angle), shards[p.c].mapped);
+            p.angle -= angle;
+            return;
+        }
+        PhaseBuffer nb{ q1, q2, angle, false };
+        FlushNonCommuting(nb);
+        PushBuffer(nb);
+    }
+
+    // buffer CX(c,t)∘CP(angle) on top of any same-pair buffer.
+    // Onto pure CP(θ): CX·CP(a)·CP(θ) = CX·CP(a+θ).
+    // Onto same-direction inv: CX·CP(a)·CX·CP(θ) = P_c(a)·CP(θ-a) — CX pair
+    // cancels to a pure phase buffer. Opposite direction: flush first.
+    void BufferCInvert(bitLenInt c, bitLenInt t, double angle)
+    {
+        int i = FindBuffer(c, t);
+        if (i >= 0) {
+            PhaseBuffer p = pendingPairs[(size_t)i];
+            if (!p.inv) {
+                pendingPairs.erase(pendingPairs.begin() + i);
+                PhaseBuffer nb{ c, t, angle + p.angle, true };
+                FlushNonCommuting(nb);
+                PushBuffer(nb);
+                return;
+            }
+            if (p.c == c && p.t == t) {
+                pendingPairs.erase(pendingPairs.begin() + i);
+                if (!AngleZero(angle)) {
+                    shards[c].unit->Phase(
+                        cplx<R>(1, 0), polar<R>(1, (R)angle), shards[c].mapped);
+                }
+                BufferCPhase(c, t, p.angle - angle);
+                return;
+            }
+            // opposite orientation: apply the old one, then buffer fresh
+            pendingPairs.erase(pendingPairs.begin() + i);
+            ApplyBufferNow(p);
+        }
+        PhaseBuffer nb{ c, t, angle, true };
+        FlushNonCommuting(nb);
+        PushBuffer(nb);
+    }
+
+    // Commute an incoming 1-qubit gate on q through the pending buffers,
+    // flushing only what cannot commute. May emit free 1-qubit phase/invert
+    // ops on partner shards and may ADJUST the forwarded matrix in place.
+    // Returns with m ready to forward to q's unit.
+    void CommuteBuffers1q(bitLenInt q, cplx<R>* m)
+    {
+        if (pendingPairs.empty()) return;
+        const bool diag = norm(m[1]) <= (R)1e-24 && norm(m[2]) <= (R)1e-24;
+        const bool anti = norm(m[0]) <= (R)1e-24 && norm(m[3]) <= (R)1e-24;
+        if (diag) {
+            CommuteDiag(q, m[0], m[3]);
+            return;
+        }
+        if (anti) {
+            CommuteInvert(q, m[1], m[2]);
+            return;
+        }
+        if (IsHadamard(m) && TryCommuteH(q)) return;
+        FlushPhasePairs(q);
+    }
+
+    static bool IsHadamard(const cplx<R>* m)
+    {
+        const R s = (R)0.70710678118654752440;
+        // m == f * [[s,s],[s,-s]] for unimodular f
+        return std::abs(std::sqrt((double)norm(m[0])) - (double)s) < 1e-9 &&
+            norm(m[0] - m[1]) < (R)1e-18 && norm(m[0] - m[2]) < (R)1e-18 &&
+            norm(m[0] + m[3]) < (R)1e-18;
+    }
+
+    // incoming diag(p0,p1) on q (caller forwards the diag itself):
+    //   pure CP / inv with q==c: commutes unchanged
+    //   inv with q==t: angle += 2*arg(p0/p1); free Phase(1, p1/p0) on c
+    void CommuteDiag(bitLenInt q, cplx<R> p0, cplx<R> p1)
+    {
+        for (auto& p : pendingPairs) {
+            if (!(p.inv && p.t == q)) continue;
+            const double a0 = std::atan2((double)p0.im, (double)p0.re);
+            const double a1 = std::atan2((double)p1.im, (double)p1.re);
+            p.angle += 2.0 * (a0 - a1);
+            shards[p.c].unit->Phase(cplx<R>(1, 0), polar<R>(1, (R)(a1 - a0)), shards[p.c].mapped);
+        }
+    }
+
+    // incoming [0,tr;bl,0] on q = diag(tr,bl)·X. X-stage per buffer:
+    //   pure CP(θ), q either end: θ -> -θ; free Phase(1, e^{iθ}) on partner
+    //   inv, q==t: θ -> -θ; free Phase(1, e^{iθ}) on c
+    //   inv, q==c: θ same; free Invert(e^{-iθ}, 1) on t; forwarded tr *= e^{iθ}
+    // then diag(tr',bl)-stage = CommuteDiag. m adjusted in place via tr/bl.
+    void CommuteInvert(bitLenInt q, cplx<R>& tr, cplx<R>& bl)
+    {
+        for (auto& p : pendingPairs) {
+            if (p.c != q && p.t != q) continue;
+            if (!p.inv) {
+                const double th = p.angle;
+                const bitLenInt o = (p.c == q) ? p.t : p.c;
+                p.angle = -th;
+                if (!AngleZero(th)) {
+                    shards[o].unit->Phase(cplx<R>(1, 0), polar<R>(1, (R)th), shards[o].mapped);
+                }
+            } else if (p.t == q) {
+                const double th = p.angle;
+                p.angle = -th;
+                if (!AngleZero(th)) {
+                    shards[p.c].unit->Phase(
+                        cplx<R>(1, 0), polar<R>(1, (R)th), shards[p.c].mapped);
+                }
+            } else { // p.c == q
+                const double th = p.angle;
+                shards[p.t].unit->Invert(
+                    polar<R>(1, (R)(-th)), cplx<R>(1, 0), shards[p.t].mapped);
+                tr = tr * polar<R>(1, (R)th);
+            }
+        }
+        CommuteDiag(q, tr, bl);
+    }
+
+    // incoming exact H on q: CP(π) buffers become CX(partner->q, 0); pure-CX
+    // buffers targeting q (angle 0) become CP(π). Anything else on q, or a
+    // transform that would break the pairwise-commuting invariant, aborts
+    // (returns false; caller flushes). Global phase of the H is irrelevant.
+    bool TryCommuteH(bitLenInt q)
+    {
+        std::vector<size_t> mine;
+        for (size_t i = 0; i < pendingPairs.size(); ++i) {
+            const PhaseBuffer& p = pendingPairs[i];
+            if (p.c != q && p.t != q) continue;
+            const bool czLike = !p.inv && AnglePi(p.angle);
+            const bool cxLike = p.inv && p.t == q && AngleZero(p.angle);
+            if (!czLike && !cxLike) return false;
+            mine.push_back(i);
+        }
+        if (mine.empty()) return true;
+        // tentative transforms
+        std::vector<PhaseBuffer> nb;
+        for (size_t i : mine) {
+            const PhaseBuffer& p = pendingPairs[i];
+            if (!p.inv) {
+                const bitLenInt o = (p.c == q) ? p.t : p.c;
+                nb.push_back({ o, q, 0.0, true }); // CZ -> CX(o->q)
+            } else {
+                nb.push_back({ p.c, q, 3.14159265358979323846, false }); // CX -> CZ
+            }
+        }
+        // invariant check vs untouched buffers and among the new set
+        for (size_t i = 0; i < pendingPairs.size(); ++i) {
+            if (std::find(mine.begin(), mine.end(), i) != mine.end()) continue;
+            for (const auto& x : nb) {
+                if (!BuffersCommute(pendingPairs[i], x)) return false;
+            }
+        }
+        for (size_t a = 0; a < nb.size(); ++a) {
+            for (size_t b = a + 1; b < nb.size(); ++b) {
+                if (!BuffersCommute(nb[a], nb[b])) return false;
+            }
+        }
+        for (size_t k = 0; k < mine.size(); ++k) pendingPairs[mine[k]] = nb[k];
+        return true;
+    }
+
+    // measurement resolution: q collapsed to `outcome`. CP buffers touching
+    // q degenerate to a 1-qubit phase on the partner (outcome=1) or vanish;
+    // inv buffers with q==c degenerate to Invert(e^{iθ},1) on t (outcome=1)
+    // or vanish. (inv buffers with q==t were flushed before collapse.)
     void ResolvePhasePairsOnMeasure(bitLenInt q, bool outcome)
     {
         if (pendingPairs.empty()) return;
         for (size_t i = pendingPairs.size(); i-- > 0;) {
-            const PhasePair p = pendingPairs[i];
-            if (p.a != q && p.b != q) continue;
+            const PhaseBuffer p = pendingPairs[i];
+            if (p.c != q && p.t != q) continue;
             pendingPairs.erase(pendingPairs.begin() + i);
             if (!outcome) continue;
-            const bitLenInt other = (p.a == q) ? p.b : p.a;
-            shards[other].unit->Phase(
-                cplx<R>(1, 0), polar<R>(1, (R)p.angle), shards[other].mapped);
+            if (!p.inv) {
+                const bitLenInt other = (p.c == q) ? p.t : p.c;
+                shards[other].unit->Phase(
+                    cplx<R>(1, 0), polar<R>(1, (R)p.angle), shards[other].mapped);
+            } else {
+                // q == p.c (collapsed control): X·diag(1,e^{iθ}) on t
+                shards[p.t].unit->Invert(
+                    polar<R>(1, (R)p.angle), cplx<R>(1, 0), shards[p.t].mapped);
+            }
         }
     }
 
@@ -233,10 +524,18 @@ public:
             for (bitLenInt x : a) TrySeparate(x);
             for (bitLenInt x : b) TrySeparate(x);
         }
-        std::set<QInterface<R>*> ua;
+        std::set<QInterface<R>*> ua, ub;
         for (bitLenInt x : a) ua.insert(shards[x].unit.get());
         for (bitLenInt x : b) {
             if (ua.count(shards[x].unit.get())) return false;
+            ub.insert(shards[x].unit.get());
+        }
+        // a pending buffer crossing the two unit groups is a (deferred)
+        // entangling link: the sets are not factorized
+        for (const auto& p : pendingPairs) {
+            QInterface<R>* uc = shards[p.c].unit.get();
+            QInterface<R>* ut = shards[p.t].unit.get();
+            if ((ua.count(uc) && ub.count(ut)) || (ua.count(ut) && ub.count(uc))) return false;
         }
         return true;
     }
@@ -266,11 +565,8 @@ public:
     {
         if (mtrxs.size() != 4u * targets.size())
             throw QrackError("Mtrx1qBatch: need 4 entries per target");
-        for (size_t i = 0; i < targets.size(); ++i) {
-            if (norm(mtrxs[4u * i + 1u]) > (R)1e-24 || norm(mtrxs[4u * i + 2u]) > (R)1e-24) {
-                FlushPhasePairs(targets[i]);
-            }
-        }
+        std::vector<cplx<R>> mm(mtrxs);
+        for (size_t i = 0; i < targets.size(); ++i) CommuteBuffers1q(targets[i], &mm[4u * i]);
         std::map<QInterface<R>*, std::pair<std::vector<bitLenInt>, std::vector<cplx<R>>>> groups;
         std::map<QInterface<R>*, QInterfacePtr<R>> keep;
         for (size_t i = 0; i < targets.size(); ++i) {
@@ -278,7 +574,7 @@ public:
             auto& g = groups[s.unit.get()];
             keep[s.unit.get()] = s.unit;
             g.first.push_back(s.mapped);
-            g.second.insert(g.second.end(), &mtrxs[4u * i], &mtrxs[4u * i] + 4);
+            g.second.insert(g.second.end(), &mm[4u * i], &mm[4u * i] + 4);
         }
         for (auto& kv : groups) {
             if (kv.second.first.size() == 1u) {

@@ -29,6 +29,11 @@ import qrack_amd as qa
 class DistQPager:
     def __init__(self, qubits, precision="fp32", engine="hip", seed=1234, device_id=0):
         assert dist.is_initialized(), "torch.distributed must be initialized"
+        if engine == "hip" and torch.cuda.is_available():
+            # torch's HIP runtime must win the init race against the qrack
+            # extension, or torch.cuda reports "No HIP GPUs are available"
+            # in this process (bench.py ordering, enforced here for all users)
+            torch.cuda.init()
         self.world = dist.get_world_size()
         self.rank = dist.get_rank()
         assert self.world & (self.world - 1) == 0, "world size must be a power of 2"
@@ -502,22 +507,45 @@ class DistQPager:
             n_chunks //= 2
         step = half // n_chunks
         bounds = [(c * step, (c + 1) * step) for c in range(n_chunks)]
-        reqs = []
-        for lo, hi in bounds:
-            reqs.append(dist.batch_isend_irecv([
-                dist.P2POp(dist.isend, view[lo:hi], partner),
-                dist.P2POp(dist.irecv, tmp[lo:hi], partner),
-            ]))
         ext = torch.cuda.current_stream(self.device_id).cuda_stream
         elem = tmp.element_size()
         recv_is_low = not i_am_low  # high page receives the target=0 side
         base = tmp.data_ptr()
-        for (lo, hi), rq in zip(bounds, reqs):
-            for r in rq:
-                r.wait()  # stream-wait on torch's current stream (NCCL)
-            self.q.qft_column_top_range(
-                float(scale), rs, in_place, pows, ws, float(scale * meta_w), pre,
-                lo, hi, base + lo * elem, recv_is_low, ext)
+        if self._nccl_active():
+            # RCCL over xGMI: all chunk sendrecvs issued up front (they run
+            # back-to-back on NCCL's comm stream); each req.wait() is a
+            # stream-wait on torch's current stream, so chunk c's kernel
+            # overlaps chunk c+1's transfer
+            reqs = []
+            for lo, hi in bounds:
+                reqs.append(dist.batch_isend_irecv([
+                    dist.P2POp(dist.isend, view[lo:hi], partner),
+                    dist.P2POp(dist.irecv, tmp[lo:hi], partner),
+                ]))
+            for (lo, hi), rq in zip(bounds, reqs):
+                for r in rq:
+                    r.wait()
+                self.q.qft_column_top_range(
+                    float(scale), rs, in_place, pows, ws, float(scale * meta_w), pre,
+                    lo, hi, base + lo * elem, recv_is_low, ext)
+        else:
+            # gloo transport (1-GPU rehearsals: RCCL refuses two ranks on one
+            # device — "Duplicate GPU detected"): host-staged chunks through
+            # the IDENTICAL chunk/ranged-kernel flow, so multi-rank CI on one
+            # GPU covers everything but the RCCL transport itself
+            for lo, hi in bounds:
+                send_cpu = view[lo:hi].cpu()
+                tmp_cpu = torch.empty_like(send_cpu)
+                rq = dist.batch_isend_irecv([
+                    dist.P2POp(dist.isend, send_cpu, partner),
+                    dist.P2POp(dist.irecv, tmp_cpu, partner),
+                ])
+                for r in rq:
+                    r.wait()
+                tmp[lo : hi].copy_(tmp_cpu)
+                self.q.qft_column_top_range(
+                    float(scale), rs, in_place, pows, ws, float(scale * meta_w), pre,
+                    lo, hi, base + lo * elem, recv_is_low, ext)
         torch.cuda.synchronize(self.device_id)
 
     def _fused_column(self, start, i, sign, pre):
@@ -525,7 +553,7 @@ class DistQPager:
         exchange at most), then H + the whole (relocated) phase ladder +
         the meta scalar in a single fused kernel."""
         if (self.pipe_enable and self.slot_of[start + i] >= self.qpp and self._is_hip()
-                and self._nccl_active() and hasattr(self.q, "qft_column_top_range")):
+                and hasattr(self.q, "qft_column_top_range")):
             self._fused_column_meta_pipelined(start, i, sign, pre)
             return
         t_slot = self._realize_local(start + i)

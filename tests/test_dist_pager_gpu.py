@@ -13,9 +13,12 @@ import os
 import numpy as np
 import pytest
 
-import qrack_amd as qa
-
 pytestmark = pytest.mark.gpu
+
+# NOTE: qrack_amd is deliberately NOT imported at module level — spawn
+# children re-import this module, and the extension .so must not load
+# before torch.cuda.init() in a process that will use torch.cuda (the
+# init-order constraint bench.py documents).
 
 import os as _os
 PORT = 28000 + (_os.getpid() % 8000)
@@ -30,9 +33,9 @@ def _worker(rank, world, qubits, seed, port, fn_name, backend="gloo", marker_dir
     import torch
     import torch.distributed as dist
 
-    if "nccl" in backend:
-        # torch first, engine second (bench.py ordering); all rehearsal
-        # ranks share device 0 on the 1-GPU box
+    # torch first, engine second (bench.py ordering); all rehearsal ranks
+    # share device 0 on the 1-GPU box
+    if torch.cuda.is_available():
         torch.cuda.init()
         torch.cuda.set_device(0)
     dist.init_process_group(backend, rank=rank, world_size=world)
