@@ -661,6 +661,14 @@ template <typename R> static void bindQInterface(py::module_& m, const char* nam
                 });
             })
         .def("is_clifford", [](QI& q) { return q.isClifford(); })
+        .def("ancilla_count",
+            [](Ptr q) -> int {
+                // T-gadget ancillae pending on a stabilizer-hybrid layer
+                if (auto hy = std::dynamic_pointer_cast<QStabilizerHybrid<R>>(q)) {
+                    return (int)hy->GetAncillaCount();
+                }
+                return 0;
+            })
         .def("depolarizing_channel_weak_1qb", &QI::DepolarizingChannelWeak1Qb)
         .def("reduced_density_matrix", [](QI& q, bitLenInt qb) {
             cplx<R> rho[4];

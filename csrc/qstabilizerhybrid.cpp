@@ -21,6 +21,12 @@ QStabilizerHybrid<R>::QStabilizerHybrid(bitLenInt qBitCount, bitCapInt initState
         const double v = std::atof(env);
         if (v > 0.0) ncrp = v;
     }
+    if (const char* env = std::getenv("QRACK_USE_T_GADGET")) {
+        useTGadget = std::atoi(env) != 0;
+    }
+    if (const char* env = std::getenv("QRACK_MAX_ANCILLA")) {
+        maxAncilla = (bitLenInt)std::atoi(env);
+    }
     if (!engineFactory) {
         RngPtr rng = this->rand_generator;
         engineFactory = [rng](bitLenInt n, bitCapInt perm) -> QInterfacePtr<R> {
@@ -99,7 +105,80 @@ template <typename R> bool QStabilizerHybrid<R>::TryShardRoundClifford(bitLenInt
     return true;
 }
 
+// Reverse T-injection (see header): absorb a blocked non-Clifford phase (or
+// invert = X·phase) shard into the tableau exactly via a gadget ancilla.
+template <typename R> bool QStabilizerHybrid<R>::TryShardGadget(bitLenInt q)
+{
+    if (!shards[q] || !stabilizer || !useTGadget) return false;
+    if (ancillaCount >= maxAncilla) return false;
+    auto m = *shards[q];
+    const bool isPhase = (norm(m[1]) <= (R)1e-12) && (norm(m[2]) <= (R)1e-12);
+    const bool isInvert = (norm(m[0]) <= (R)1e-12) && (norm(m[3]) <= (R)1e-12);
+    if (!isPhase && !isInvert) return false;
+    if (isInvert) {
+        // [0,tr;bl,0] = diag(tr,bl)·X — the X is Clifford, peel it off
+        try {
+            stabilizer->Invert(cplx<R>(1, 0), cplx<R>(1, 0), q);
+        } catch (const QrackError&) {
+            return false;
+        }
+        m = { m[1], cplx<R>(0, 0), cplx<R>(0, 0), m[2] };
+    }
+    // shard = m0 · diag(1, e^{iθ}); snap the Clifford quarter-turns into the
+    // tableau, gadget the fractional remainder δ ∈ (-π/4, π/4]
+    const cplx<R> ratio = m[3] * conj(m[0]);
+    const double theta = std::atan2((double)ratio.im, (double)ratio.re);
+    const double half_pi = 1.5707963267948966;
+    const double k = std::nearbyint(theta / half_pi);
+    const double delta = theta - k * half_pi;
+    const int ki = ((int)k % 4 + 4) % 4;
+    static const cplx<R> IPOW[4] = { { 1, 0 }, { 0, 1 }, { -1, 0 }, { 0, -1 } };
+    // Clifford part including the shard's own global phase and the e^{iδ/2}
+    // split so the remainder is exactly RZ(δ) = diag(e^{-iδ/2}, e^{iδ/2})
+    const cplx<R> g = m[0] * polar<R>(1, (R)(delta / 2.0));
+    const cplx<R> cliffordPart[4] = { g, cplx<R>(0, 0), cplx<R>(0, 0), g * IPOW[ki] };
+    try {
+        stabilizer->Mtrx(cliffordPart, q);
+    } catch (const QrackError&) {
+        return false;
+    }
+    shards[q].reset();
+    if (std::abs(delta) <= 1e-14) return true;
+    // ancilla |0> at the tableau top; CNOT(q -> a); shard H·RZ(δ) on a;
+    // deferred ⟨0|_a postselection completes RZ(δ) on q exactly
+    const bitLenInt a = stabilizer->GetQubitCount();
+    stabilizer->Allocate(a, 1u);
+    stabilizer->CNOTGate(q, a);
+    const R s2 = (R)0.70710678118654752440;
+    const cplx<R> e0 = polar<R>(1, (R)(-delta / 2.0)), e1 = polar<R>(1, (R)(delta / 2.0));
+    // H · diag(e0, e1)
+    const cplx<R> anc[4] = { s2 * e0, s2 * e1, s2 * e0, cplx<R>(0, 0) - s2 * e1 };
+    shards.emplace_back(std::make_unique<std::array<cplx<R>, 4>>());
+    for (int i = 0; i < 4; ++i) (*shards.back())[i] = anc[i];
+    ++ancillaCount;
+    return true;
+}
+
 template <typename R> void QStabilizerHybrid<R>::FlushShard(bitLenInt q)
+{
+    if (!shards[q]) return;
+    if (engine) {
+        engine->Mtrx(shards[q]->data(), q);
+        shards[q].reset();
+        return;
+    }
+    if (!TryShardFlushClifford(q) && !TryShardRoundClifford(q) && !TryShardGadget(q)) {
+        SwitchToEngine();
+        if (shards[q]) {
+            engine->Mtrx(shards[q]->data(), q);
+            shards[q].reset();
+        }
+    }
+}
+
+// flush WITHOUT the ancilla gadget: structural separations need the qubit
+// free of gadget links, so a non-Clifford shard forces the engine instead
+template <typename R> void QStabilizerHybrid<R>::FlushShardNoGadget(bitLenInt q)
 {
     if (!shards[q]) return;
     if (engine) {
@@ -125,14 +204,16 @@ template <typename R> void QStabilizerHybrid<R>::DumpShardIfPhase(bitLenInt q)
 
 template <typename R> void QStabilizerHybrid<R>::SwitchToEngine()
 {
+    InvalidateCache();
     if (engine) return;
     // materialize the tableau state into a fresh state-vector engine
     // (parity: qstabilizerhybrid.cpp:435-511). Wide states skip the dense
     // 2^n buffer: the tableau's 2^g nonzero amplitudes stream directly into
     // the engine (the sparse engine handles thousands of qubits this way).
-    engine = engineFactory(qubitCount, 0u);
-    if (qubitCount <= 26u) {
-        std::vector<cplx<R>> buf(maxQPower);
+    const bitLenInt w = TableauWidth();
+    engine = engineFactory(w, 0u);
+    if (w <= 26u) {
+        std::vector<cplx<R>> buf(pow2(w));
         stabilizer->GetQuantumState(buf.data());
         engine->SetQuantumState(buf.data());
     } else {
@@ -144,12 +225,23 @@ template <typename R> void QStabilizerHybrid<R>::SwitchToEngine()
             [&](bitCapInt idx, cplx<R> amp) { e->SetAmplitude(idx, amp); });
     }
     stabilizer.reset();
-    // flush every pending shard into the engine
-    for (bitLenInt q = 0; q < qubitCount; ++q) {
+    // flush every pending shard (logical + gadget-ancilla) into the engine
+    for (bitLenInt q = 0; q < w; ++q) {
         if (shards[q]) {
             engine->Mtrx(shards[q]->data(), q);
             shards[q].reset();
         }
+    }
+    if (ancillaCount) {
+        // act the deferred gadget postselections: every ancilla to |0>
+        // (probability exactly 1/2 each — never impossible), renormalized
+        // by ForceM, then disposed (ancillae are separable after collapse)
+        for (bitLenInt a = 0; a < ancillaCount; ++a) {
+            engine->ForceM(qubitCount + a, false, true, true);
+        }
+        engine->Dispose(qubitCount, ancillaCount);
+        shards.resize(qubitCount);
+        ancillaCount = 0;
     }
 }
 
@@ -157,13 +249,17 @@ template <typename R> void QStabilizerHybrid<R>::SwitchToEngine()
 
 template <typename R> void QStabilizerHybrid<R>::SetPermutation(bitCapInt perm, cplx<R> phase)
 {
+    InvalidateCache();
     engine.reset();
+    ancillaCount = 0;
+    shards.resize(qubitCount);
     for (auto& s : shards) s.reset();
     stabilizer = std::make_shared<QStabilizer<R>>(qubitCount, perm, this->rand_generator);
 }
 
 template <typename R> void QStabilizerHybrid<R>::SetQuantumState(const cplx<R>* inputState)
 {
+    InvalidateCache();
     SwitchToEngine();
     engine->SetQuantumState(inputState);
 }
@@ -187,7 +283,7 @@ template <typename R> cplx<R> QStabilizerHybrid<R>::GetAmplitude(bitCapInt perm)
     for (auto& s : shards) {
         if (s) anyShard = true;
     }
-    if (!engine && !anyShard) return stabilizer->GetAmplitude(perm);
+    if (!engine && !anyShard && !ancillaCount) return stabilizer->GetAmplitude(perm);
     if (engine && !anyShard) return engine->GetAmplitude(perm);
     QStabilizerHybridPtr<R> c = std::static_pointer_cast<QStabilizerHybrid<R>>(Clone());
     c->SwitchToEngine();
@@ -196,6 +292,7 @@ template <typename R> cplx<R> QStabilizerHybrid<R>::GetAmplitude(bitCapInt perm)
 
 template <typename R> void QStabilizerHybrid<R>::SetAmplitude(bitCapInt perm, cplx<R> amp)
 {
+    InvalidateCache();
     SwitchToEngine();
     engine->SetAmplitude(perm, amp);
 }
@@ -204,6 +301,7 @@ template <typename R> void QStabilizerHybrid<R>::SetAmplitude(bitCapInt perm, cp
 
 template <typename R> void QStabilizerHybrid<R>::Mtrx(const cplx<R>* m, bitLenInt t)
 {
+    InvalidateCache();
     if (engine) {
         engine->Mtrx(m, t);
         return;
@@ -227,6 +325,7 @@ template <typename R> void QStabilizerHybrid<R>::Invert(cplx<R> tr, cplx<R> bl, 
 template <typename R>
 void QStabilizerHybrid<R>::MCMtrx(const std::vector<bitLenInt>& controls, const cplx<R>* m, bitLenInt t)
 {
+    InvalidateCache();
     if (controls.empty()) {
         Mtrx(m, t);
         return;
@@ -250,6 +349,7 @@ void QStabilizerHybrid<R>::MCMtrx(const std::vector<bitLenInt>& controls, const 
 template <typename R>
 void QStabilizerHybrid<R>::MACMtrx(const std::vector<bitLenInt>& controls, const cplx<R>* m, bitLenInt t)
 {
+    InvalidateCache();
     if (controls.empty()) {
         Mtrx(m, t);
         return;
@@ -273,6 +373,7 @@ template <typename R>
 void QStabilizerHybrid<R>::UCMtrx(
     const std::vector<bitLenInt>& controls, const cplx<R>* m, bitLenInt t, bitCapInt perm)
 {
+    InvalidateCache();
     if (controls.empty()) {
         Mtrx(m, t);
         return;
@@ -285,6 +386,7 @@ template <typename R>
 void QStabilizerHybrid<R>::UniformlyControlledSingleBit(
     const std::vector<bitLenInt>& controls, bitLenInt t, const cplx<R>* mtrxs)
 {
+    InvalidateCache();
     if (controls.empty()) {
         Mtrx(mtrxs, t);
         return;
@@ -295,6 +397,7 @@ void QStabilizerHybrid<R>::UniformlyControlledSingleBit(
 
 template <typename R> void QStabilizerHybrid<R>::Swap(bitLenInt q1, bitLenInt q2)
 {
+    InvalidateCache();
     if (q1 == q2) return;
     std::swap(shards[q1], shards[q2]);
     if (engine) {
@@ -306,6 +409,7 @@ template <typename R> void QStabilizerHybrid<R>::Swap(bitLenInt q1, bitLenInt q2
 
 template <typename R> void QStabilizerHybrid<R>::ISwap(bitLenInt q1, bitLenInt q2)
 {
+    InvalidateCache();
     FlushShard(q1);
     if (!engine) FlushShard(q2);
     if (engine) {
@@ -317,6 +421,7 @@ template <typename R> void QStabilizerHybrid<R>::ISwap(bitLenInt q1, bitLenInt q
 
 template <typename R> void QStabilizerHybrid<R>::IISwap(bitLenInt q1, bitLenInt q2)
 {
+    InvalidateCache();
     FlushShard(q1);
     if (!engine) FlushShard(q2);
     if (engine) {
@@ -330,6 +435,11 @@ template <typename R> void QStabilizerHybrid<R>::IISwap(bitLenInt q1, bitLenInt 
 
 template <typename R> R QStabilizerHybrid<R>::Prob(bitLenInt q)
 {
+    if (!engine && ancillaCount) {
+        // pending gadget postselections reweight Z marginals once any
+        // Clifford acted after the gadget: query the cached resolved clone
+        return RdmClone()->Prob(q);
+    }
     if (!engine && ShardIsPhase(q)) {
         return stabilizer->Prob(q);
     }
@@ -354,6 +464,8 @@ template <typename R> R QStabilizerHybrid<R>::Prob(bitLenInt q)
 template <typename R>
 bool QStabilizerHybrid<R>::ForceM(bitLenInt q, bool result, bool doForce, bool doApply)
 {
+    InvalidateCache();
+    if (!engine && ancillaCount) SwitchToEngine();
     if (!engine && !ShardIsPhase(q)) {
         FlushShard(q);
     } else if (!engine && shards[q]) {
@@ -419,17 +531,36 @@ template <typename R> R QStabilizerHybrid<R>::ProbMask(bitCapInt mask, bitCapInt
 
 template <typename R> bitLenInt QStabilizerHybrid<R>::Compose(QInterfacePtr<R> toCopy, bitLenInt start)
 {
+    InvalidateCache();
     QStabilizerHybrid<R>* o = dynamic_cast<QStabilizerHybrid<R>*>(toCopy.get());
     const bitLenInt oQubits = toCopy->GetQubitCount();
     const bitLenInt nQubits = qubitCount + oQubits;
     if (o && !engine && !o->engine) {
+        const bitLenInt m = o->qubitCount;
+        const bitLenInt j = o->ancillaCount;
+        const bitLenInt W = TableauWidth() + m + j; // combined tableau width
         stabilizer->Compose(o->stabilizer, start);
-        for (bitLenInt q = 0; q < oQubits; ++q) shards.emplace(shards.begin() + start);
-        for (bitLenInt q = 0; q < oQubits; ++q) {
+        // o's gadget ancillae landed at [start+m, start+m+j): bubble them to
+        // the global end so ancillae always sit above every logical qubit
+        for (bitLenInt x = 0; x < j; ++x) {
+            for (bitLenInt pos = (bitLenInt)(start + m + j - 1 - x); pos < (bitLenInt)(W - 1 - x);
+                 ++pos) {
+                stabilizer->SwapGate(pos, pos + 1);
+            }
+        }
+        for (bitLenInt q = 0; q < m; ++q) shards.emplace(shards.begin() + start);
+        for (bitLenInt q = 0; q < m; ++q) {
             if (o->shards[q]) {
                 shards[start + q] = std::make_unique<std::array<cplx<R>, 4>>(*o->shards[q]);
             }
         }
+        for (bitLenInt q = 0; q < j; ++q) {
+            shards.emplace_back();
+            if (o->shards[m + q]) {
+                shards.back() = std::make_unique<std::array<cplx<R>, 4>>(*o->shards[m + q]);
+            }
+        }
+        ancillaCount += j;
         this->SetQubitCount(nQubits);
         return start;
     }
@@ -448,14 +579,17 @@ template <typename R> bitLenInt QStabilizerHybrid<R>::Compose(QInterfacePtr<R> t
 
 template <typename R> void QStabilizerHybrid<R>::Decompose(bitLenInt start, QInterfacePtr<R> dest)
 {
+    InvalidateCache();
     const bitLenInt len = dest->GetQubitCount();
     QStabilizerHybrid<R>* o = dynamic_cast<QStabilizerHybrid<R>*>(dest.get());
     if (!engine && o && stabilizer->CanDecomposeDispose(start, len)) {
-        for (bitLenInt q = start; q < start + len; ++q) FlushShard(q);
+        for (bitLenInt q = start; q < start + len; ++q) FlushShardNoGadget(q);
         if (!engine) {
             o->engine.reset();
+            o->ancillaCount = 0;
             o->stabilizer = std::make_shared<QStabilizer<R>>(len, 0u, this->rand_generator);
             stabilizer->Decompose(start, o->stabilizer);
+            o->shards.resize(len);
             for (auto& s : o->shards) s.reset();
             shards.erase(shards.begin() + start, shards.begin() + start + len);
             this->SetQubitCount(qubitCount - len);
@@ -475,7 +609,8 @@ template <typename R> void QStabilizerHybrid<R>::Decompose(bitLenInt start, QInt
 
 template <typename R> void QStabilizerHybrid<R>::Dispose(bitLenInt start, bitLenInt length)
 {
-    for (bitLenInt q = start; q < start + length; ++q) FlushShard(q);
+    InvalidateCache();
+    for (bitLenInt q = start; q < start + length; ++q) FlushShardNoGadget(q);
     if (!engine && stabilizer->CanDecomposeDispose(start, length)) {
         stabilizer->Dispose(start, length);
     } else {
@@ -489,7 +624,8 @@ template <typename R> void QStabilizerHybrid<R>::Dispose(bitLenInt start, bitLen
 template <typename R>
 void QStabilizerHybrid<R>::Dispose(bitLenInt start, bitLenInt length, bitCapInt disposedPerm)
 {
-    for (bitLenInt q = start; q < start + length; ++q) FlushShard(q);
+    InvalidateCache();
+    for (bitLenInt q = start; q < start + length; ++q) FlushShardNoGadget(q);
     if (!engine) {
         stabilizer->Dispose(start, length, disposedPerm);
     } else {
@@ -501,6 +637,7 @@ void QStabilizerHybrid<R>::Dispose(bitLenInt start, bitLenInt length, bitCapInt 
 
 template <typename R> bitLenInt QStabilizerHybrid<R>::Allocate(bitLenInt start, bitLenInt length)
 {
+    InvalidateCache();
     if (!length) return start;
     if (!engine) {
         stabilizer->Allocate(start, length);
@@ -523,12 +660,16 @@ template <typename R> QInterfacePtr<R> QStabilizerHybrid<R>::Clone()
         clone->stabilizer = std::static_pointer_cast<QStabilizer<R>>(stabilizer->Clone());
         clone->engine.reset();
     }
-    for (bitLenInt q = 0; q < qubitCount; ++q) {
+    clone->ancillaCount = ancillaCount;
+    clone->shards.resize(shards.size());
+    for (bitLenInt q = 0; q < (bitLenInt)shards.size(); ++q) {
         if (shards[q]) {
             clone->shards[q] = std::make_unique<std::array<cplx<R>, 4>>(*shards[q]);
         }
     }
     clone->ncrp = ncrp;
+    clone->useTGadget = useTGadget;
+    clone->maxAncilla = maxAncilla;
     clone->logFidelity = logFidelity;
     return clone;
 }

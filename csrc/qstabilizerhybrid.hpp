@@ -34,13 +34,45 @@ protected:
     double ncrp = 0.0;        // near-Clifford rounding parameter (0 = exact)
     double logFidelity = 0.0; // rounding-fidelity accumulator
 
+    // Reverse T-injection gadget (reference qstabilizerhybrid.cpp:195-240,
+    // PRX Quantum 3.020361 App. A): a blocked non-Clifford PHASE shard
+    // RZ(δ) on q is realized EXACTLY in the tableau by allocating an
+    // ancilla a=|0>, CNOT(q,a), and moving H·RZ(δ) onto a as a shard; the
+    // deferred ⟨0|_a postselection (probability exactly 1/2 per ancilla)
+    // happens at SwitchToEngine. Z-basis marginals and samples of the
+    // LOGICAL qubits are exact straight from the tableau with ancillae
+    // pending, because the deferred op is diagonal on q and unitary on a.
+    // Ancillae live at tableau indices [qubitCount, qubitCount+ancillaCount).
+    bitLenInt ancillaCount = 0;
+    bitLenInt maxAncilla = 16;
+    bool useTGadget = true;
+
     // NCRP: if the shard is a phase gate within ncrp of a Clifford phase,
     // snap it into the tableau and log the exact overlap loss. Returns true
     // if the shard was disposed of (flushed exactly or rounded).
     bool TryShardRoundClifford(bitLenInt q);
+    // the gadget step above; returns true if the shard was absorbed
+    bool TryShardGadget(bitLenInt q);
+    bitLenInt TableauWidth() const { return qubitCount + ancillaCount; }
+
+    // rdm-clone cache (reference qstabilizerhybrid.hpp:68): read-only
+    // queries on a tableau with pending gadget ancillae need the deferred
+    // postselections acted; cache ONE engine-mode clone across consecutive
+    // queries, invalidated by any mutation.
+    QInterfacePtr<R> rdmClone;
+    void InvalidateCache() { rdmClone.reset(); }
+    QInterfacePtr<R> RdmClone()
+    {
+        if (rdmClone) return rdmClone;
+        auto c = std::static_pointer_cast<QStabilizerHybrid<R>>(this->Clone());
+        c->SwitchToEngine();
+        rdmClone = c->engine;
+        return rdmClone;
+    }
 
     bool InEngineMode() const { return (bool)engine; }
     void FlushShard(bitLenInt q);        // apply buffered 2x2 to the active backend
+    void FlushShardNoGadget(bitLenInt q); // same, but never spends an ancilla
     void DumpShardIfPhase(bitLenInt q);  // drop diagonal shards (safe before Z ops)
     bool ShardIsPhase(bitLenInt q) const;
     bool ShardIsIdentity(bitLenInt q) const;
@@ -68,6 +100,13 @@ public:
         }
         return true;
     }
+    bitLenInt GetAncillaCount() const { return ancillaCount; }
+    // loader hook: mark the top `k` tableau qubits as gadget ancillae
+    void SetAncillae(bitLenInt k)
+    {
+        ancillaCount = k;
+        shards.resize(qubitCount + k);
+    }
     // serialization access (tableau + buffered shards; reference
     // qstabilizerhybrid.cpp:2235-2291 writes both)
     QStabilizerPtr<R> Tableau() { return stabilizer; }
@@ -80,7 +119,7 @@ public:
     }
     void ReplaceTableau(QStabilizerPtr<R> st)
     {
-        if (st->GetQubitCount() != qubitCount)
+        if (st->GetQubitCount() != TableauWidth())
             throw QrackError("ReplaceTableau: width mismatch");
         stabilizer = st;
         engine.reset();

@@ -80,12 +80,24 @@ def test_sdrp_exact_when_fully_entangling():
 
 
 def test_ncrp_zero_switches_to_engine():
-    # a T gate then CNOT forces the dense engine when ncrp == 0
-    q = qa.create_simulator(4, layers=["stabilizer_hybrid", "cpu"], seed=1)
-    q.h(0)
-    q.t(0)
-    q.cnot(0, 1)
-    assert not q.is_clifford()
+    # a T gate then CNOT stays Clifford via the reverse T-injection gadget
+    # (round 2); with the gadget disabled it must force the dense engine
+    import os
+
+    os.environ["QRACK_USE_T_GADGET"] = "0"
+    try:
+        q = qa.create_simulator(4, layers=["stabilizer_hybrid", "cpu"], seed=1)
+        q.h(0)
+        q.t(0)
+        q.cnot(0, 1)
+        assert not q.is_clifford()
+    finally:
+        del os.environ["QRACK_USE_T_GADGET"]
+    q2 = qa.create_simulator(4, layers=["stabilizer_hybrid", "cpu"], seed=1)
+    q2.h(0)
+    q2.t(0)
+    q2.cnot(0, 1)
+    assert q2.is_clifford()  # gadget kept the tableau
 
 
 def test_ncrp_rounds_near_clifford_phases():
