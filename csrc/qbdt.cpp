@@ -24,6 +24,9 @@ QBdt<R>::QBdt(bitLenInt qBitCount, bitCapInt initState, RngPtr rgp)
     if (const char* env = std::getenv("QRACK_QBDT_MAX_NODES")) {
         maxNodes = (size_t)std::atoll(env);
     }
+    if (const char* env = std::getenv("QRACK_QBDT_SEPARABILITY_THRESHOLD")) {
+        sepThreshold = (R)std::atof(env);
+    }
     root = MakeBasis(0, qBitCount, initState);
 }
 
@@ -217,6 +220,7 @@ void QBdt<R>::UCMtrx(
     cplx<R> w = rootWeight;
     root = Apply(root, 0u, m, target, controls, sortedPerm, w);
     rootWeight = w;
+    MaybeRound();
     CheckGuard();
 }
 
@@ -510,6 +514,39 @@ template <typename R> size_t QBdt<R>::CountNodes() const
 template <typename R> void QBdt<R>::CheckGuard() const
 {
     if (CountNodes() > maxNodes) throw std::bad_alloc();
+}
+
+template <typename R>
+QBdtNodePtr<R> QBdt<R>::RoundTree(
+    QBdtNodePtr<R> n, bitLenInt depth, std::map<const QBdtNode<R>*, QBdtNodePtr<R>>& memo)
+{
+    if (!n || depth >= qubitCount) return n;
+    auto it = memo.find(n.get());
+    if (it != memo.end()) return it->second;
+    auto nn = std::make_shared<QBdtNode<R>>(*n);
+    const double n0 = (double)norm(nn->w[0]);
+    const double n1 = (double)norm(nn->w[1]);
+    const double tot = n0 + n1;
+    const double thr = (double)sepThreshold;
+    if (tot > 0.0) {
+        if (n0 > 0.0 && n0 / tot <= thr) {
+            // round the 0-branch away; sibling renormalized; the discarded
+            // node-local relative mass bounds the fidelity loss
+            nn->w[0] = cplx<R>(0, 0);
+            nn->c[0] = nullptr;
+            nn->w[1] = nn->w[1] * (R)std::sqrt(tot / n1);
+            logFidelity += std::log(std::max(1.0 - n0 / tot, 1e-300));
+        } else if (n1 > 0.0 && n1 / tot <= thr) {
+            nn->w[1] = cplx<R>(0, 0);
+            nn->c[1] = nullptr;
+            nn->w[0] = nn->w[0] * (R)std::sqrt(tot / n0);
+            logFidelity += std::log(std::max(1.0 - n1 / tot, 1e-300));
+        }
+    }
+    nn->c[0] = RoundTree(nn->c[0], depth + 1u, memo);
+    nn->c[1] = RoundTree(nn->c[1], depth + 1u, memo);
+    memo[n.get()] = nn;
+    return nn;
 }
 
 template class QBdt<float>;

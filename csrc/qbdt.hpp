@@ -13,6 +13,8 @@
 
 #include "qstabilizerhybrid.hpp" // EngineFactoryFn
 
+#include <cmath>
+#include <map>
 #include <set>
 
 namespace qrack_amd {
@@ -36,6 +38,13 @@ protected:
     QBdtNodePtr<R> root;
     cplx<R> rootWeight;
     size_t maxNodes; // RAM guard (QRACK_QBDT_MAX_ALLOC_MB equivalent)
+    // approximate branch rounding (reference QRACK_QBDT_SEPARABILITY_THRESHOLD
+    // + node.cpp Prune): a branch whose node-local relative probability is
+    // below sepThreshold is rounded to zero (sibling renormalized) with
+    // fidelity accounting, so compression DEGRADES instead of hitting the
+    // node cap. 0 = exact.
+    R sepThreshold = (R)0;
+    double logFidelity = 0.0;
 
     // ---- tree primitives ----
     static QBdtNodePtr<R> MakeBasis(bitLenInt depth, bitLenInt nQubits, bitCapInt perm);
@@ -58,11 +67,24 @@ protected:
     size_t CountNodes() const;
     static void CountNodesRec(QBdtNodePtr<R> n, std::set<const QBdtNode<R>*>& seen);
     void CheckGuard() const;
+    // copy-on-write rounding pass (memoized across shared subtrees)
+    QBdtNodePtr<R> RoundTree(
+        QBdtNodePtr<R> n, bitLenInt depth, std::map<const QBdtNode<R>*, QBdtNodePtr<R>>& memo);
+    void MaybeRound()
+    {
+        if (sepThreshold <= (R)0 || !root) return;
+        std::map<const QBdtNode<R>*, QBdtNodePtr<R>> memo;
+        root = RoundTree(root, 0u, memo);
+    }
 
 public:
     QBdt(bitLenInt qBitCount, bitCapInt initState = 0u, RngPtr rgp = nullptr);
 
     size_t NodeCount() const { return CountNodes(); }
+
+    double GetUnitaryFidelity() override { return std::exp(logFidelity); }
+    void ResetUnitaryFidelity() override { logFidelity = 0.0; }
+    void SetQbdtSeparabilityThreshold(double v) { sepThreshold = (R)v; }
 
     // ---- state ----
     void SetPermutation(bitCapInt perm, cplx<R> phase = cplx<R>((R)1, (R)0)) override;

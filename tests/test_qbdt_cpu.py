@@ -204,3 +204,79 @@ def test_bdt_hybrid_switches_to_engine():
         assert_states_close(q.get_state_vector(), cp.get_state_vector(), 1e-4)
     finally:
         del os.environ["QRACK_QBDT_HYBRID_THRESHOLD"]
+
+
+def test_qbdt_branch_rounding_degrades_gracefully():
+    """with QRACK_QBDT_SEPARABILITY_THRESHOLD set,
+    a random circuit that would exceed the exact node budget completes with
+    reported fidelity < 1 instead of raising the RAM guard."""
+    import os
+
+    import numpy as np
+
+    # weakly-entangling circuit: small rotations + CNOT chains produce many
+    # low-weight branches — exactly what rounding absorbs; the exact tree
+    # still exceeds the node cap
+    os.environ["QRACK_QBDT_MAX_NODES"] = "200"
+    os.environ["QRACK_QBDT_SEPARABILITY_THRESHOLD"] = "0.05"
+    try:
+        n = 12
+        rng = np.random.default_rng(7)
+        q = qa.create_simulator(n, layers=["bdt"], seed=7)
+        for layer in range(8):
+            for t in range(n):
+                th = float(rng.uniform(0.05, 0.25))
+                q.ry(th, t)
+            for t in range(0, n - 1):
+                q.cnot(t, t + 1)
+        # completed under a node cap that the exact tree blows through
+        fid = q.get_unitary_fidelity()
+        assert 0.0 < fid <= 1.0
+        p = q.prob(0)
+        assert 0.0 <= p <= 1.0 + 1e-6
+    finally:
+        del os.environ["QRACK_QBDT_MAX_NODES"]
+        del os.environ["QRACK_QBDT_SEPARABILITY_THRESHOLD"]
+    # and without rounding, the same circuit must hit the guard
+    os.environ["QRACK_QBDT_MAX_NODES"] = "200"
+    try:
+        q2 = qa.create_simulator(n, layers=["bdt"], seed=7)
+        rng = np.random.default_rng(7)
+        with pytest.raises(Exception):
+            for layer in range(8):
+                for t in range(n):
+                    th = float(rng.uniform(0.05, 0.25))
+                    q2.ry(th, t)
+                for t in range(0, n - 1):
+                    q2.cnot(t, t + 1)
+    finally:
+        del os.environ["QRACK_QBDT_MAX_NODES"]
+
+
+def test_qbdt_rounding_accuracy_small_threshold():
+    """Tiny thresholds keep the state numerically close to exact."""
+    import os
+
+    import numpy as np
+
+    os.environ["QRACK_QBDT_SEPARABILITY_THRESHOLD"] = "1e-9"
+    try:
+        n = 6
+        q = qa.create_simulator(n, layers=["bdt"], seed=3)
+        cp = qa.create_simulator(n, engine="cpu", seed=3)
+        rng = np.random.default_rng(3)
+        for _ in range(30):
+            t = int(rng.integers(n))
+            th = float(rng.uniform(0, 2 * np.pi))
+            q.ry(th, t)
+            cp.ry(th, t)
+            c = int(rng.integers(n))
+            if c != t:
+                q.cz(c, t)
+                cp.cz(c, t)
+        sv1 = np.asarray(q.get_state_vector())
+        sv2 = np.asarray(cp.get_state_vector())
+        f = abs(np.vdot(sv1, sv2))
+        assert f > 1.0 - 1e-5
+    finally:
+        del os.environ["QRACK_QBDT_SEPARABILITY_THRESHOLD"]
