@@ -14,6 +14,7 @@
 #include "qcircuit.hpp"
 #include "qengine_cpu.hpp"
 #include "qfactory.hpp"
+#include "qhybrid.hpp"
 #include "qneuron.hpp"
 #include "serialize.hpp"
 #ifdef QRACK_AMD_HIP_ENGINE
@@ -661,6 +662,11 @@ template <typename R> static void bindQInterface(py::module_& m, const char* nam
                 });
             })
         .def("is_clifford", [](QI& q) { return q.isClifford(); })
+        .def("hybrid_mode",
+            [](Ptr q) -> std::string {
+                if (auto hy = std::dynamic_pointer_cast<QHybrid<R>>(q)) return hy->ModeName();
+                return "n/a";
+            })
         .def("ancilla_count",
             [](Ptr q) -> int {
                 // T-gadget ancillae pending on a stabilizer-hybrid layer

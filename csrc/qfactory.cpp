@@ -101,8 +101,21 @@ EngineFactoryFn<R> LayerFactory(std::vector<std::string> layers, RngPtr rng, int
         } else {
             gpuF = cpuF;
         }
-        return [rng, cpuF, gpuF](bitLenInt n, bitCapInt perm) -> QInterfacePtr<R> {
-            return std::make_shared<QHybrid<R>>(n, perm, rng, cpuF, gpuF);
+        // pager promotion tier (reference qhybrid.cpp:43-53): past the max
+        // single-alloc width the hybrid transparently re-shards onto a
+        // QPager whose pages reuse the gpu (or cpu) engine factory
+        EngineFactoryFn<R> bestF = (HipDeviceCount() > 0)
+            ? gpuF
+            : cpuF;
+        EngineFactoryFn<R> pagerF = [rng, bestF](bitLenInt n, bitCapInt perm) -> QInterfacePtr<R> {
+            bitLenInt pq = (sizeof(R) == 4) ? 33u : 32u;
+            if (const char* env = std::getenv("QRACK_MAX_PAGE_QB")) {
+                pq = (bitLenInt)std::atoi(env);
+            }
+            return std::make_shared<QPager<R>>(n, perm, rng, bestF, pq);
+        };
+        return [rng, cpuF, gpuF, pagerF](bitLenInt n, bitCapInt perm) -> QInterfacePtr<R> {
+            return std::make_shared<QHybrid<R>>(n, perm, rng, cpuF, gpuF, 0u, pagerF);
         };
     }
     if (head == "bdt") {

@@ -56,6 +56,34 @@ public:
     bitLenInt GetQubitsPerPage() const { return qpp; }
     bitLenInt GetMetaBits() const { return metaBits; }
 
+    // bulk amplitude-range access routed to the owning pages (QHybrid pager
+    // promotion migrates engine<->pager in bounded chunks through these —
+    // reference SetAmplitudePage/CombineEngines, qpager.cpp:316-367)
+    void GetAmplitudePage(cplx<R>* out, bitCapInt offset, bitCapInt length)
+    {
+        while (length) {
+            const bitCapInt p = offset >> qpp;
+            const bitCapInt in = offset & (PageLen() - 1u);
+            const bitCapInt take = std::min<bitCapInt>(length, PageLen() - in);
+            qPages[p]->GetAmplitudePage(out, in, take);
+            out += take;
+            offset += take;
+            length -= take;
+        }
+    }
+    void SetAmplitudePage(const cplx<R>* in, bitCapInt offset, bitCapInt length)
+    {
+        while (length) {
+            const bitCapInt p = offset >> qpp;
+            const bitCapInt at = offset & (PageLen() - 1u);
+            const bitCapInt take = std::min<bitCapInt>(length, PageLen() - at);
+            qPages[p]->SetAmplitudePage(in, at, take);
+            in += take;
+            offset += take;
+            length -= take;
+        }
+    }
+
     // ---- state ----
     void SetPermutation(bitCapInt perm, cplx<R> phase = cplx<R>((R)1, (R)0)) override;
     void SetQuantumState(const cplx<R>* inputState) override;

@@ -203,3 +203,36 @@ def test_sum_sqr_diff_and_approx_compare():
     c.h(0)
     c.phase_flip()
     assert a.approx_compare(c)
+
+
+def test_qhybrid_pager_promotion():
+    """growing past the max single-alloc width promotes
+    the hybrid transparently onto a QPager (and demotes on shrink), with the
+    state migrated in bounded chunks — verified by amplitude continuity."""
+    import os
+
+    import numpy as np
+
+    os.environ["QRACK_MAX_PAGE_QB"] = "4"
+    try:
+        q = qa.create_simulator(4, layers=["hybrid"], seed=5)
+        assert q.hybrid_mode() in ("cpu", "gpu")
+        q.h(0)
+        q.cnot(0, 3)
+        sv_before = np.asarray(q.get_state_vector()).copy()
+        q.allocate(4, 2)  # 6 qubits > max page 4 -> paged
+        assert q.hybrid_mode() == "paged"
+        sv_after = np.asarray(q.get_state_vector())
+        # original amplitudes live in the |00> block of the new qubits
+        assert np.allclose(sv_after[: len(sv_before)], sv_before, atol=1e-6)
+        # gates still work while paged
+        q.h(5)
+        q.cnot(5, 0)
+        q.cnot(5, 0)
+        q.h(5)
+        assert abs(q.prob(5)) < 1e-6
+        q.dispose(4, 2)
+        assert q.hybrid_mode() in ("cpu", "gpu")
+        assert np.allclose(np.asarray(q.get_state_vector()), sv_before, atol=1e-6)
+    finally:
+        del os.environ["QRACK_MAX_PAGE_QB"]
