@@ -332,9 +332,13 @@ template <typename R> static void bindQInterface(py::module_& m, const char* nam
         .def("m_all", [](QI& q) { return q.MAll(); })
         .def("m_all_big",
             [](QI& q) {
-                // >64-qubit terminal measurement: per-qubit collapse composed
-                // into an arbitrary-precision Python int (the uint64 packed
-                // MAll saturates past 63 qubits)
+                // >64-qubit terminal measurement via the packed BigCap path
+                // (C++ MAllWide; Python int only for >128q tails)
+                if (q.GetQubitCount() <= 128u) {
+                    const BigCap r = q.MAllWide();
+                    return py::int_(
+                        (py::int_((uint64_t)r.hi) << py::int_(64)) | py::int_((uint64_t)r.lo));
+                }
                 py::int_ out(0);
                 const py::int_ one(1);
                 for (bitLenInt i = 0; i < q.GetQubitCount(); ++i) {
@@ -343,6 +347,22 @@ template <typename R> static void bindQInterface(py::module_& m, const char* nam
                     }
                 }
                 return out;
+            })
+        .def("set_permutation_big",
+            [](QI& q, py::object perm) {
+                // packed wide permutation init (BigCap SetPermutationWide)
+                py::int_ p(perm);
+                const uint64_t lo =
+                    py::cast<uint64_t>(p & py::int_(0xFFFFFFFFFFFFFFFFull));
+                const uint64_t hi = py::cast<uint64_t>(
+                    (p >> py::int_(64)) & py::int_(0xFFFFFFFFFFFFFFFFull));
+                q.SetPermutationWide(BigCap(lo, hi));
+            })
+        .def("multi_shot_measure_qubits",
+            [](QI& q, std::vector<bitLenInt> qubits, unsigned shots) {
+                // wide-safe sampling: qubit INDICES in, position-packed
+                // values out (valid for any logical width)
+                return q.MultiShotMeasureQubits(qubits, shots);
             })
         .def("sample_clone_big",
             [](Ptr q) {

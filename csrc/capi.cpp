@@ -481,6 +481,37 @@ void qrack_measure_shots(
     });
 }
 
+void qrack_m_all_wide(quid sid, uint64_t* lo, uint64_t* hi)
+{
+    guarded(sid, [&](SimSlot& s) {
+        BigCap r;
+        FOR_SIM(s, r = q.MAllWide());
+        if (lo) *lo = r.lo;
+        if (hi) *hi = r.hi;
+    });
+}
+
+void qrack_set_permutation_wide(quid sid, uint64_t lo, uint64_t hi)
+{
+    guarded(sid, [&](SimSlot& s) { FOR_SIM(s, q.SetPermutationWide(BigCap(lo, hi))); });
+}
+
+void qrack_measure_shots_qubits(
+    quid sid, const uint64_t* qubits, uint64_t nq, uint64_t shots, uint64_t* shotsArray)
+{
+    guarded(sid, [&](SimSlot& s) {
+        std::vector<bitLenInt> qs;
+        for (uint64_t i = 0; i < nq; ++i) qs.push_back((bitLenInt)qubits[i]);
+        FOR_SIM(s, {
+            auto res = q.MultiShotMeasureQubits(qs, (unsigned)shots);
+            uint64_t idx = 0;
+            for (auto& kv : res) {
+                for (int k = 0; k < kv.second && idx < shots; ++k) shotsArray[idx++] = kv.first;
+            }
+        });
+    });
+}
+
 double qrack_joint_ensemble_probability(
     quid sid, const int* paulis, const uint64_t* qs, uint64_t n)
 {

@@ -41,6 +41,41 @@ inline bitLenInt log2Ocl(bitCapInt n) {
 }
 inline bool isPowerOfTwo(bitCapInt x) { return x && !(x & (x - 1u)); }
 
+// Packed WIDE capacity integer for >64-qubit logical masks (capability
+// parity with the reference's QBCAPPOW>6 BigInteger / bi_* shims,
+// /root/reference/include/common/big_integer.hpp, qrack_types.hpp:64-86):
+// 128 bits as two uint64 limbs. Used at the LOGICAL layer (QUnit /
+// stabilizer / QBdt stacks past 64 qubits) — dense engines never index
+// past 2^64, so the hot kernels keep the 64-bit bitCapInt.
+struct BigCap {
+    uint64_t lo = 0;
+    uint64_t hi = 0;
+    constexpr BigCap() = default;
+    constexpr BigCap(uint64_t l, uint64_t h = 0) : lo(l), hi(h) {}
+    bool bit(bitLenInt q) const { return (q < 64u) ? ((lo >> q) & 1u) : ((q < 128u) && ((hi >> (q - 64u)) & 1u)); }
+    void setBit(bitLenInt q, bool v)
+    {
+        if (q < 64u) {
+            lo = v ? (lo | (1ull << q)) : (lo & ~(1ull << q));
+        } else if (q < 128u) {
+            hi = v ? (hi | (1ull << (q - 64u))) : (hi & ~(1ull << (q - 64u)));
+        }
+    }
+    bool operator==(const BigCap& o) const { return lo == o.lo && hi == o.hi; }
+    bool operator!=(const BigCap& o) const { return !(*this == o); }
+    bool operator<(const BigCap& o) const { return hi != o.hi ? hi < o.hi : lo < o.lo; }
+    BigCap operator|(const BigCap& o) const { return { lo | o.lo, hi | o.hi }; }
+    BigCap operator&(const BigCap& o) const { return { lo & o.lo, hi & o.hi }; }
+    BigCap operator^(const BigCap& o) const { return { lo ^ o.lo, hi ^ o.hi }; }
+    bool any() const { return lo || hi; }
+};
+inline BigCap pow2w(bitLenInt p)
+{
+    BigCap b;
+    b.setBit(p, true);
+    return b;
+}
+
 // POD complex usable identically in host C++ and HIP device code
 // (std::complex is not device-friendly; layout is the same: {re, im}).
 template <typename R> struct cplx {

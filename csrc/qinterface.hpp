@@ -640,6 +640,52 @@ public:
         bitLenInt start, bitLenInt length, bitCapInt result, bool doForce = true, bool doApply = true);
     bitCapInt MReg(bitLenInt start, bitLenInt length) { return ForceMReg(start, length, 0, false, true); }
     virtual bitCapInt MAll() { return MReg(0, qubitCount); }
+    // ---- packed >64-qubit paths (BigCap; reference BigInteger parity) ----
+    // terminal measurement of ANY width: per-qubit collapse into 128 packed
+    // bits (identical collapse semantics to MAll)
+    virtual BigCap MAllWide()
+    {
+        BigCap out;
+        for (bitLenInt q = 0; q < qubitCount; ++q) {
+            if (M(q)) out.setBit(q, true);
+        }
+        return out;
+    }
+    virtual void SetPermutationWide(const BigCap& perm)
+    {
+        SetPermutation(0u);
+        for (bitLenInt q = 0; q < qubitCount; ++q) {
+            if (perm.bit(q)) X(q);
+        }
+    }
+    // sampling keyed by POSITION in `qubits` (wide-safe input: qubit indices
+    // instead of 64-bit powers; outputs pack list positions, so up to 64
+    // sampled qubits of an arbitrarily wide state)
+    virtual std::map<bitCapInt, int> MultiShotMeasureQubits(
+        const std::vector<bitLenInt>& qubits, unsigned shots)
+    {
+        if (qubits.size() > 64u) throw QrackError("MultiShotMeasureQubits: > 64 sampled qubits");
+        bool narrow = true;
+        for (bitLenInt q : qubits) {
+            if (q >= 63u) narrow = false;
+        }
+        if (narrow) {
+            std::vector<bitCapInt> powers;
+            for (bitLenInt q : qubits) powers.push_back(pow2(q));
+            return MultiShotMeasureMask(powers, shots);
+        }
+        // wide fallback: per-shot clone + per-qubit collapse
+        std::map<bitCapInt, int> results;
+        for (unsigned s = 0; s < shots; ++s) {
+            QInterfacePtr<R> c = Clone();
+            bitCapInt val = 0;
+            for (size_t b = 0; b < qubits.size(); ++b) {
+                if (c->M(qubits[b])) val |= (ONE_BCI << b);
+            }
+            results[val]++;
+        }
+        return results;
+    }
     virtual std::map<bitCapInt, int> MultiShotMeasureMask(
         const std::vector<bitCapInt>& qPowers, unsigned shots);
     virtual void MultiShotMeasureMask(

@@ -171,6 +171,30 @@ public:
     R Prob(bitLenInt q) override;
     bool ForceM(bitLenInt q, bool result, bool doForce = true, bool doApply = true) override;
     bitCapInt MAll() override;
+    // wide-safe sampling: tableau states clone + collapse per listed qubit
+    // (no 64-bit packed MAll in the path), so 100+ qubit Clifford stacks
+    // sample exactly
+    std::map<bitCapInt, int> MultiShotMeasureQubits(
+        const std::vector<bitLenInt>& qubits, unsigned shots) override
+    {
+        bool blocking = ancillaCount > 0 || (bool)engine;
+        for (bitLenInt q = 0; q < qubitCount && !blocking; ++q) {
+            if (shards[q] && !ShardIsPhase(q)) blocking = true;
+        }
+        if (!blocking && stabilizer) {
+            std::map<bitCapInt, int> results;
+            for (unsigned s = 0; s < shots; ++s) {
+                auto c = std::static_pointer_cast<QStabilizer<R>>(stabilizer->Clone());
+                bitCapInt val = 0;
+                for (size_t b = 0; b < qubits.size(); ++b) {
+                    if (c->M(qubits[b])) val |= (ONE_BCI << b);
+                }
+                results[val]++;
+            }
+            return results;
+        }
+        return QInterface<R>::MultiShotMeasureQubits(qubits, shots);
+    }
     std::map<bitCapInt, int> MultiShotMeasureMask(
         const std::vector<bitCapInt>& qPowers, unsigned shots) override;
     R ProbMask(bitCapInt mask, bitCapInt permutation) override;

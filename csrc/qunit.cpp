@@ -637,22 +637,32 @@ template <typename R>
 std::map<bitCapInt, int> QUnit<R>::MultiShotMeasureMask(
     const std::vector<bitCapInt>& qPowers, unsigned shots)
 {
+    std::vector<bitLenInt> qubits;
+    for (bitCapInt p : qPowers) qubits.push_back(log2Ocl(p));
+    return MultiShotMeasureQubits(qubits, shots);
+}
+
+template <typename R>
+std::map<bitCapInt, int> QUnit<R>::MultiShotMeasureQubits(
+    const std::vector<bitLenInt>& qubits, unsigned shots)
+{
     // units are independent subsystems: sample each separately and combine
-    // shots elementwise (no entanglement, no width blow-up)
+    // shots elementwise (no entanglement, no width blow-up). Qubit-INDEX
+    // addressing keeps this exact past 64 logical qubits (BigCap parity).
     if (!shots) return {};
-    for (bitCapInt p : qPowers) FlushInvTargeting(log2Ocl(p));
+    for (bitLenInt q : qubits) FlushInvTargeting(q);
     std::vector<bitCapInt> joint(shots, 0u);
     std::set<QInterfacePtr<R>> seen;
-    for (size_t b = 0; b < qPowers.size(); ++b) {
-        const bitLenInt q = log2Ocl(qPowers[b]);
+    for (size_t b = 0; b < qubits.size(); ++b) {
+        const bitLenInt q = qubits[b];
         QInterfacePtr<R> u = shards[q].unit;
         if (seen.count(u)) continue;
         seen.insert(u);
         // output-bit <-> local-power mapping for every masked qubit of u
         std::vector<size_t> outBits;
         std::vector<bitCapInt> localPowers;
-        for (size_t k = 0; k < qPowers.size(); ++k) {
-            const bitLenInt lq = log2Ocl(qPowers[k]);
+        for (size_t k = 0; k < qubits.size(); ++k) {
+            const bitLenInt lq = qubits[k];
             if (shards[lq].unit == u) {
                 outBits.push_back(k);
                 localPowers.push_back(pow2(shards[lq].mapped));

@@ -609,6 +609,8 @@ public:
     bitCapInt MAll() override;
     std::map<bitCapInt, int> MultiShotMeasureMask(
         const std::vector<bitCapInt>& qPowers, unsigned shots) override;
+    std::map<bitCapInt, int> MultiShotMeasureQubits(
+        const std::vector<bitLenInt>& qubits, unsigned shots) override;
     R ProbMask(bitCapInt mask, bitCapInt permutation) override;
     R ProbParity(bitCapInt mask) override;
     bool ForceMParity(bitCapInt mask, bool result, bool doForce = true) override;

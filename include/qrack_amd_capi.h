@@ -79,6 +79,12 @@ uint64_t qrack_m_all(quid sid);
 /* shotsArray receives `shots` packed results over the `qs` bit order */
 void qrack_measure_shots(quid sid, const uint64_t* qs, uint64_t nq, uint64_t shots,
     uint64_t* shotsArray);
+/* packed >64-qubit paths (two uint64 limbs per 128-bit value) */
+void qrack_m_all_wide(quid sid, uint64_t* lo, uint64_t* hi);
+void qrack_set_permutation_wide(quid sid, uint64_t lo, uint64_t hi);
+/* wide-safe sampling: qubit INDICES (any width); results pack list positions */
+void qrack_measure_shots_qubits(quid sid, const uint64_t* qubits, uint64_t nq, uint64_t shots,
+    uint64_t* shotsArray);
 double qrack_joint_ensemble_probability(quid sid, const int* paulis, const uint64_t* qs, uint64_t n);
 
 /* QFT */
