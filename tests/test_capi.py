@@ -120,3 +120,26 @@ def test_approximation_knobs(lib):
     assert 0.0 < fid <= 1.0 + 1e-9
     assert lib.qrack_get_error(u64(sid)) == 0
     lib.qrack_destroy(u64(sid))
+
+
+def test_capi_wide_masks(lib):
+    """Packed >64-qubit C ABI: wide permutation in,
+    two-limb measurement out, qubit-index shot sampling."""
+    sid = lib.qrack_init_count_type(80, 0, 1, 1, 0, 0, 0, 0, 0, 0)
+    assert sid != 0
+    lo_in = (1 << 63) | 0b101
+    hi_in = (1 << 6) | 1  # qubits 64 and 70
+    lib.qrack_set_permutation_wide(sid, ctypes.c_uint64(lo_in), ctypes.c_uint64(hi_in))
+    lo = ctypes.c_uint64(0)
+    hi = ctypes.c_uint64(0)
+    lib.qrack_m_all_wide(sid, ctypes.byref(lo), ctypes.byref(hi))
+    assert lo.value == lo_in
+    assert hi.value == hi_in
+    # x then sample high qubits by index
+    lib.qrack_x(sid, 75)
+    qubits = (ctypes.c_uint64 * 3)(64, 70, 75)
+    shots = (ctypes.c_uint64 * 10)()
+    lib.qrack_measure_shots_qubits(sid, qubits, 3, 10, shots)
+    for k in range(10):
+        assert shots[k] == 0b111
+    lib.qrack_destroy(sid)
