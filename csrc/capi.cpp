@@ -3,7 +3,9 @@
 // per-simulator error latch polled by get_error).
 #include "../include/qrack_amd_capi.h"
 
+#include "qcircuit.hpp"
 #include "qfactory.hpp"
+#include "qneuron.hpp"
 #include "serialize.hpp"
 
 #include <fstream>
@@ -704,6 +706,467 @@ int qrack_lossy_in_from_file(quid sid, const char* path)
         s->error = 1;
         return -1;
     }
+}
+
+/* ---- approximation / separability controls ---- */
+
+void qrack_set_ace_max_qb(quid sid, uint64_t maxQb)
+{
+    guarded(sid, [&](SimSlot& s) { FOR_SIM(s, q.SetAceMaxQubits((bitLenInt)maxQb)); });
+}
+
+void qrack_set_reactive_separate(quid sid, int on)
+{
+    guarded(sid, [&](SimSlot& s) { FOR_SIM(s, q.SetReactiveSeparate(on != 0)); });
+}
+
+int qrack_try_separate_tol(quid sid, const uint64_t* qs, uint64_t n, double tol)
+{
+    int out = 0;
+    guarded(sid, [&](SimSlot& s) {
+        std::vector<bitLenInt> v = ctrlVec(qs, n);
+        FOR_SIM(s, out = q.TrySeparate(v, (decltype(q.Prob(0)))tol) ? 1 : 0);
+    });
+    return out;
+}
+
+int qrack_are_factorized(
+    quid sid, const uint64_t* qa, uint64_t na, const uint64_t* qb, uint64_t nb)
+{
+    int out = 0;
+    guarded(sid, [&](SimSlot& s) {
+        std::vector<bitLenInt> a = ctrlVec(qa, na), b = ctrlVec(qb, nb);
+        FOR_SIM(s, out = q.AreFactorized(a, b) ? 1 : 0);
+    });
+    return out;
+}
+
+/* ---- multiplexer ---- */
+
+void qrack_multiplex_1mtrx(
+    quid sid, const uint64_t* cs, uint64_t nc, uint64_t qb, const double* m8)
+{
+    guarded(sid, [&](SimSlot& s) {
+        std::vector<bitLenInt> ctrls = ctrlVec(cs, nc);
+        const size_t nMats = (size_t)1 << nc;
+        if (s.f) {
+            std::vector<cplx<float>> m(4u * nMats);
+            for (size_t i = 0; i < 4u * nMats; ++i)
+                m[i] = cplx<float>((float)m8[2 * i], (float)m8[2 * i + 1]);
+            s.f->UniformlyControlledSingleBit(ctrls, (bitLenInt)qb, m.data());
+        } else if (s.d) {
+            std::vector<cplx<double>> m(4u * nMats);
+            for (size_t i = 0; i < 4u * nMats; ++i)
+                m[i] = cplx<double>(m8[2 * i], m8[2 * i + 1]);
+            s.d->UniformlyControlledSingleBit(ctrls, (bitLenInt)qb, m.data());
+        }
+    });
+}
+
+/* ---- boolean logic + parity phase ---- */
+
+#define LOGIC3(name, call)                                                                         \
+    void name(quid sid, uint64_t a, uint64_t b, uint64_t o)                                        \
+    {                                                                                              \
+        guarded(sid,                                                                               \
+            [&](SimSlot& s) { FOR_SIM(s, q.call((bitLenInt)a, (bitLenInt)b, (bitLenInt)o)); });    \
+    }
+
+LOGIC3(qrack_and, AND)
+LOGIC3(qrack_or, OR)
+LOGIC3(qrack_xor, XOR)
+LOGIC3(qrack_nand, NAND)
+LOGIC3(qrack_nor, NOR)
+
+#define CLOGIC(name, call)                                                                         \
+    void name(quid sid, uint64_t qi, int ci, uint64_t qo)                                          \
+    {                                                                                              \
+        guarded(sid,                                                                               \
+            [&](SimSlot& s) { FOR_SIM(s, q.call((bitLenInt)qi, ci != 0, (bitLenInt)qo)); });       \
+    }
+
+CLOGIC(qrack_cland, CLAND)
+CLOGIC(qrack_clor, CLOR)
+CLOGIC(qrack_clxor, CLXOR)
+
+void qrack_phase_parity(quid sid, double lambda, const uint64_t* qs, uint64_t n)
+{
+    guarded(sid, [&](SimSlot& s) {
+        bitCapInt mask = 0;
+        for (uint64_t i = 0; i < n; ++i) mask |= pow2((bitLenInt)qs[i]);
+        if (s.f) s.f->PhaseParity((float)lambda, mask);
+        if (s.d) s.d->PhaseParity(lambda, mask);
+    });
+}
+
+/* ---- modular arithmetic (Shor) ---- */
+
+void qrack_divn(quid sid, uint64_t a, uint64_t m, uint64_t in, uint64_t out, uint64_t len)
+{
+    guarded(sid, [&](SimSlot& s) {
+        FOR_SIM(s, q.IMULModNOut(a, m, (bitLenInt)in, (bitLenInt)out, (bitLenInt)len));
+    });
+}
+
+void qrack_mcmuln(quid sid, uint64_t a, const uint64_t* cs, uint64_t nc, uint64_t m, uint64_t in,
+    uint64_t out, uint64_t len)
+{
+    guarded(sid, [&](SimSlot& s) {
+        std::vector<bitLenInt> ctrls = ctrlVec(cs, nc);
+        FOR_SIM(s, q.CMULModNOut(a, m, (bitLenInt)in, (bitLenInt)out, (bitLenInt)len, ctrls));
+    });
+}
+
+void qrack_mcdivn(quid sid, uint64_t a, const uint64_t* cs, uint64_t nc, uint64_t m, uint64_t in,
+    uint64_t out, uint64_t len)
+{
+    guarded(sid, [&](SimSlot& s) {
+        std::vector<bitLenInt> ctrls = ctrlVec(cs, nc);
+        FOR_SIM(s, q.CIMULModNOut(a, m, (bitLenInt)in, (bitLenInt)out, (bitLenInt)len, ctrls));
+    });
+}
+
+} // extern "C"
+
+/* ---- quantum neuron sub-API ---- */
+
+namespace {
+
+struct NeuronSlot {
+    std::shared_ptr<QNeuron<float>> f;
+    std::shared_ptr<QNeuron<double>> d;
+    SimSlotPtr sim; // keeps the simulator alive
+    std::mutex op;
+};
+using NeuronSlotPtr = std::shared_ptr<NeuronSlot>;
+std::mutex g_nmtx;
+std::map<quid, NeuronSlotPtr> g_neurons;
+quid g_nnext = 1;
+
+NeuronSlotPtr neuronSlot(quid nid)
+{
+    std::lock_guard<std::mutex> lk(g_nmtx);
+    auto it = g_neurons.find(nid);
+    return (it == g_neurons.end()) ? nullptr : it->second;
+}
+
+template <typename F> void nguarded(quid nid, F&& fn)
+{
+    NeuronSlotPtr s = neuronSlot(nid);
+    if (!s) return;
+    std::lock_guard<std::mutex> lk(s->op);
+    try {
+        fn(*s);
+    } catch (const std::exception&) {
+        if (s->sim) s->sim->error = 1;
+    }
+}
+
+} // namespace
+
+extern "C" {
+
+quid qrack_init_qneuron(quid sid, const uint64_t* inputs, uint64_t n, uint64_t output,
+    int activationFn, double alpha, double tolerance)
+{
+    SimSlotPtr sim = slot(sid);
+    if (!sim) return 0;
+    NeuronSlotPtr ns = std::make_shared<NeuronSlot>();
+    ns->sim = sim;
+    std::vector<bitLenInt> ins = ctrlVec(inputs, n);
+    try {
+        if (sim->f) {
+            ns->f = std::make_shared<QNeuron<float>>(sim->f, ins, (bitLenInt)output,
+                (QNeuronActivationFn)activationFn, (float)alpha, (float)tolerance);
+        }
+        if (sim->d) {
+            ns->d = std::make_shared<QNeuron<double>>(sim->d, ins, (bitLenInt)output,
+                (QNeuronActivationFn)activationFn, alpha, tolerance);
+        }
+    } catch (const std::exception&) {
+        return 0;
+    }
+    std::lock_guard<std::mutex> lk(g_nmtx);
+    const quid nid = g_nnext++;
+    g_neurons[nid] = std::move(ns);
+    return nid;
+}
+
+void qrack_destroy_qneuron(quid nid)
+{
+    std::lock_guard<std::mutex> lk(g_nmtx);
+    g_neurons.erase(nid);
+}
+
+uint64_t qrack_get_qneuron_qubit_count(quid nid)
+{
+    uint64_t out = 0;
+    nguarded(nid, [&](NeuronSlot& s) {
+        if (s.f) out = (uint64_t)s.f->GetAngles().size();
+        if (s.d) out = (uint64_t)s.d->GetAngles().size();
+    });
+    return out;
+}
+
+void qrack_set_qneuron_angles(quid nid, const double* angles, uint64_t n)
+{
+    nguarded(nid, [&](NeuronSlot& s) {
+        if (s.f) {
+            std::vector<float> a(angles, angles + n);
+            s.f->SetAngles(a);
+        }
+        if (s.d) {
+            std::vector<double> a(angles, angles + n);
+            s.d->SetAngles(a);
+        }
+    });
+}
+
+void qrack_get_qneuron_angles(quid nid, double* angles, uint64_t n)
+{
+    nguarded(nid, [&](NeuronSlot& s) {
+        if (s.f) {
+            const auto& a = s.f->GetAngles();
+            for (uint64_t i = 0; i < n && i < a.size(); ++i) angles[i] = (double)a[i];
+        } else if (s.d) {
+            const auto& a = s.d->GetAngles();
+            for (uint64_t i = 0; i < n && i < a.size(); ++i) angles[i] = a[i];
+        }
+    });
+}
+
+void qrack_set_qneuron_alpha(quid nid, double alpha)
+{
+    nguarded(nid, [&](NeuronSlot& s) {
+        if (s.f) s.f->SetAlpha((float)alpha);
+        if (s.d) s.d->SetAlpha(alpha);
+    });
+}
+
+void qrack_set_qneuron_activation_fn(quid nid, int fn)
+{
+    nguarded(nid, [&](NeuronSlot& s) {
+        if (s.f) s.f->SetActivationFn((QNeuronActivationFn)fn);
+        if (s.d) s.d->SetActivationFn((QNeuronActivationFn)fn);
+    });
+}
+
+double qrack_qneuron_predict(quid nid, int expected, int resetInit)
+{
+    double out = 0;
+    nguarded(nid, [&](NeuronSlot& s) {
+        if (s.f) out = (double)s.f->Predict(expected != 0, resetInit != 0);
+        if (s.d) out = s.d->Predict(expected != 0, resetInit != 0);
+    });
+    return out;
+}
+
+double qrack_qneuron_unpredict(quid nid, int expected)
+{
+    double out = 0;
+    nguarded(nid, [&](NeuronSlot& s) {
+        if (s.f) out = (double)s.f->Unpredict(expected != 0);
+        if (s.d) out = s.d->Unpredict(expected != 0);
+    });
+    return out;
+}
+
+void qrack_qneuron_learn(quid nid, double eta, int expected, int resetInit)
+{
+    nguarded(nid, [&](NeuronSlot& s) {
+        if (s.f) s.f->Learn((float)eta, expected != 0, resetInit != 0);
+        if (s.d) s.d->Learn(eta, expected != 0, resetInit != 0);
+    });
+}
+
+void qrack_qneuron_learn_cycle(quid nid, double eta, int expected)
+{
+    nguarded(nid, [&](NeuronSlot& s) {
+        if (s.f) s.f->Learn((float)eta, expected != 0, false);
+        if (s.d) s.d->Learn(eta, expected != 0, false);
+    });
+}
+
+void qrack_qneuron_learn_permutation(quid nid, double eta, int expected, uint64_t perm)
+{
+    nguarded(nid, [&](NeuronSlot& s) {
+        if (s.f) s.f->LearnPermutation((float)eta, expected != 0, perm);
+        if (s.d) s.d->LearnPermutation(eta, expected != 0, perm);
+    });
+}
+
+} // extern "C"
+
+/* ---- serializable circuit sub-API ---- */
+
+namespace {
+
+struct CircSlot {
+    QCircuitPtr<float> f;
+    QCircuitPtr<double> d;
+    std::mutex op;
+};
+using CircSlotPtr = std::shared_ptr<CircSlot>;
+std::mutex g_cmtx;
+std::map<quid, CircSlotPtr> g_circs;
+quid g_cnext = 1;
+
+CircSlotPtr circSlot(quid cid)
+{
+    std::lock_guard<std::mutex> lk(g_cmtx);
+    auto it = g_circs.find(cid);
+    return (it == g_circs.end()) ? nullptr : it->second;
+}
+
+quid registerCirc(CircSlotPtr c)
+{
+    std::lock_guard<std::mutex> lk(g_cmtx);
+    const quid cid = g_cnext++;
+    g_circs[cid] = std::move(c);
+    return cid;
+}
+
+} // namespace
+
+extern "C" {
+
+quid qrack_init_qcircuit(uint64_t qubits)
+{
+    CircSlotPtr c = std::make_shared<CircSlot>();
+    c->f = std::make_shared<QCircuit<float>>((bitLenInt)qubits);
+    c->d = std::make_shared<QCircuit<double>>((bitLenInt)qubits);
+    return registerCirc(std::move(c));
+}
+
+void qrack_destroy_qcircuit(quid cid)
+{
+    std::lock_guard<std::mutex> lk(g_cmtx);
+    g_circs.erase(cid);
+}
+
+uint64_t qrack_qcircuit_qubit_count(quid cid)
+{
+    CircSlotPtr c = circSlot(cid);
+    return c ? (uint64_t)c->f->GetQubitCount() : 0;
+}
+
+void qrack_qcircuit_append_1qb(quid cid, const double* m8, uint64_t q)
+{
+    CircSlotPtr c = circSlot(cid);
+    if (!c) return;
+    std::lock_guard<std::mutex> lk(c->op);
+    cplx<float> mf[4];
+    cplx<double> md[4];
+    mtrxFrom<float>(m8, mf);
+    mtrxFrom<double>(m8, md);
+    c->f->AppendMtrx(mf, (bitLenInt)q);
+    c->d->AppendMtrx(md, (bitLenInt)q);
+}
+
+void qrack_qcircuit_append_mc(
+    quid cid, const double* m8, const uint64_t* cs, uint64_t nc, uint64_t q, uint64_t perm)
+{
+    CircSlotPtr c = circSlot(cid);
+    if (!c) return;
+    std::lock_guard<std::mutex> lk(c->op);
+    std::vector<bitLenInt> ctrls = ctrlVec(cs, nc);
+    cplx<float> mf[4];
+    cplx<double> md[4];
+    mtrxFrom<float>(m8, mf);
+    mtrxFrom<double>(m8, md);
+    c->f->AppendControlled(mf, (bitLenInt)q, ctrls, perm);
+    c->d->AppendControlled(md, (bitLenInt)q, ctrls, perm);
+}
+
+void qrack_qcircuit_swap(quid cid, uint64_t q1, uint64_t q2)
+{
+    CircSlotPtr c = circSlot(cid);
+    if (!c) return;
+    std::lock_guard<std::mutex> lk(c->op);
+    c->f->Swap((bitLenInt)q1, (bitLenInt)q2);
+    c->d->Swap((bitLenInt)q1, (bitLenInt)q2);
+}
+
+void qrack_qcircuit_run(quid cid, quid sid)
+{
+    CircSlotPtr c = circSlot(cid);
+    if (!c) return;
+    std::lock_guard<std::mutex> clk(c->op);
+    guarded(sid, [&](SimSlot& s) {
+        if (s.f) c->f->Run(s.f);
+        if (s.d) c->d->Run(s.d);
+    });
+}
+
+quid qrack_qcircuit_inverse(quid cid)
+{
+    CircSlotPtr c = circSlot(cid);
+    if (!c) return 0;
+    std::lock_guard<std::mutex> lk(c->op);
+    CircSlotPtr o = std::make_shared<CircSlot>();
+    o->f = c->f->Inverse();
+    o->d = c->d->Inverse();
+    return registerCirc(std::move(o));
+}
+
+quid qrack_qcircuit_past_light_cone(quid cid, const uint64_t* qs, uint64_t n)
+{
+    CircSlotPtr c = circSlot(cid);
+    if (!c) return 0;
+    std::lock_guard<std::mutex> lk(c->op);
+    std::set<bitLenInt> m;
+    for (uint64_t i = 0; i < n; ++i) m.insert((bitLenInt)qs[i]);
+    CircSlotPtr o = std::make_shared<CircSlot>();
+    o->f = c->f->PastLightCone(m);
+    o->d = c->d->PastLightCone(m);
+    return registerCirc(std::move(o));
+}
+
+int qrack_qcircuit_out_to_file(quid cid, const char* path)
+{
+    CircSlotPtr c = circSlot(cid);
+    if (!c) return -1;
+    std::lock_guard<std::mutex> lk(c->op);
+    try {
+        std::ofstream f(path);
+        f << c->d->Serialize();
+        return 0;
+    } catch (const std::exception&) {
+        return -1;
+    }
+}
+
+quid qrack_qcircuit_in_from_file(const char* path)
+{
+    try {
+        std::ifstream f(path);
+        std::string text((std::istreambuf_iterator<char>(f)), std::istreambuf_iterator<char>());
+        CircSlotPtr c = std::make_shared<CircSlot>();
+        c->f = QCircuit<float>::Deserialize(text);
+        c->d = QCircuit<double>::Deserialize(text);
+        return registerCirc(std::move(c));
+    } catch (const std::exception&) {
+        return 0;
+    }
+}
+
+uint64_t qrack_qcircuit_out_to_string_length(quid cid)
+{
+    CircSlotPtr c = circSlot(cid);
+    if (!c) return 0;
+    std::lock_guard<std::mutex> lk(c->op);
+    return (uint64_t)c->d->Serialize().size() + 1u;
+}
+
+void qrack_qcircuit_out_to_string(quid cid, char* out, uint64_t cap)
+{
+    CircSlotPtr c = circSlot(cid);
+    if (!c || !cap) return;
+    std::lock_guard<std::mutex> lk(c->op);
+    const std::string s = c->d->Serialize();
+    const uint64_t n = std::min<uint64_t>(cap - 1u, s.size());
+    std::memcpy(out, s.data(), n);
+    out[n] = 0;
 }
 
 } // extern "C"

@@ -97,6 +97,14 @@ QInterfacePtr<R> QUnit<R>::EntangleAll(const std::vector<bitLenInt>& qs)
 template <typename R>
 QInterfacePtr<R> QUnit<R>::EntangleOrdered(const std::vector<bitLenInt>& qs)
 {
+    // reject duplicate / out-of-range qubits BEFORE any mutation (e.g. an
+    // ALU call with overlapping in/out registers must fail cleanly, not
+    // corrupt the shard map)
+    std::set<bitLenInt> uniq(qs.begin(), qs.end());
+    if (uniq.size() != qs.size()) throw QrackError("EntangleOrdered: duplicate qubits");
+    for (bitLenInt q : qs) {
+        if (q >= qubitCount) throw QrackError("EntangleOrdered: qubit out of range");
+    }
     // callers of the ordered form follow with non-diagonal register ops
     // (ALU, decompose, state access): pending phase pairs must land first
     for (bitLenInt q : qs) FlushPhasePairs(q);

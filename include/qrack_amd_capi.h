@@ -117,6 +117,68 @@ void qrack_set_ncrp(quid sid, double ncrp);
 /* compose / decompose */
 quid qrack_compose(quid sid, quid other);
 
+/* approximation / separability controls (reference SetSdrp/SetNcrp family) */
+void qrack_set_ace_max_qb(quid sid, uint64_t maxQb);
+void qrack_set_reactive_separate(quid sid, int on);
+int qrack_try_separate_tol(quid sid, const uint64_t* qs, uint64_t n, double tol);
+int qrack_are_factorized(quid sid, const uint64_t* qa, uint64_t na, const uint64_t* qb,
+    uint64_t nb);
+
+/* multiplexer (reference Multiplex1Mtrx): m8 holds 2^nc row-major 2x2s,
+   interleaved re/im doubles */
+void qrack_multiplex_1mtrx(quid sid, const uint64_t* cs, uint64_t nc, uint64_t q,
+    const double* m8);
+
+/* boolean logic + parity phase */
+void qrack_and(quid sid, uint64_t qi1, uint64_t qi2, uint64_t qo);
+void qrack_or(quid sid, uint64_t qi1, uint64_t qi2, uint64_t qo);
+void qrack_xor(quid sid, uint64_t qi1, uint64_t qi2, uint64_t qo);
+void qrack_nand(quid sid, uint64_t qi1, uint64_t qi2, uint64_t qo);
+void qrack_nor(quid sid, uint64_t qi1, uint64_t qi2, uint64_t qo);
+void qrack_cland(quid sid, uint64_t qi, int ci, uint64_t qo);
+void qrack_clor(quid sid, uint64_t qi, int ci, uint64_t qo);
+void qrack_clxor(quid sid, uint64_t qi, int ci, uint64_t qo);
+void qrack_phase_parity(quid sid, double lambda, const uint64_t* qs, uint64_t n);
+
+/* modular arithmetic (Shor building blocks; reference MULN/DIVN/MCMULN/MCDIVN) */
+void qrack_divn(quid sid, uint64_t a, uint64_t m, uint64_t inStart, uint64_t outStart,
+    uint64_t len);
+void qrack_mcmuln(quid sid, uint64_t a, const uint64_t* cs, uint64_t nc, uint64_t m,
+    uint64_t inStart, uint64_t outStart, uint64_t len);
+void qrack_mcdivn(quid sid, uint64_t a, const uint64_t* cs, uint64_t nc, uint64_t m,
+    uint64_t inStart, uint64_t outStart, uint64_t len);
+
+/* quantum neuron sub-API (reference init_qneuron/qneuron_* family) */
+quid qrack_init_qneuron(quid sid, const uint64_t* inputs, uint64_t n, uint64_t output,
+    int activationFn, double alpha, double tolerance);
+void qrack_destroy_qneuron(quid nid);
+uint64_t qrack_get_qneuron_qubit_count(quid nid);
+void qrack_set_qneuron_angles(quid nid, const double* angles, uint64_t n);
+void qrack_get_qneuron_angles(quid nid, double* angles, uint64_t n);
+void qrack_set_qneuron_alpha(quid nid, double alpha);
+void qrack_set_qneuron_activation_fn(quid nid, int fn);
+double qrack_qneuron_predict(quid nid, int expected, int resetInit);
+double qrack_qneuron_unpredict(quid nid, int expected);
+void qrack_qneuron_learn(quid nid, double eta, int expected, int resetInit);
+void qrack_qneuron_learn_cycle(quid nid, double eta, int expected);
+void qrack_qneuron_learn_permutation(quid nid, double eta, int expected, uint64_t perm);
+
+/* serializable circuit sub-API (reference init_qcircuit/qcircuit_* family) */
+quid qrack_init_qcircuit(uint64_t qubits);
+void qrack_destroy_qcircuit(quid cid);
+uint64_t qrack_qcircuit_qubit_count(quid cid);
+void qrack_qcircuit_append_1qb(quid cid, const double* m8, uint64_t q);
+void qrack_qcircuit_append_mc(quid cid, const double* m8, const uint64_t* cs, uint64_t nc,
+    uint64_t q, uint64_t perm);
+void qrack_qcircuit_swap(quid cid, uint64_t q1, uint64_t q2);
+void qrack_qcircuit_run(quid cid, quid sid);
+quid qrack_qcircuit_inverse(quid cid);
+quid qrack_qcircuit_past_light_cone(quid cid, const uint64_t* qs, uint64_t n);
+int qrack_qcircuit_out_to_file(quid cid, const char* path);
+quid qrack_qcircuit_in_from_file(const char* path);
+uint64_t qrack_qcircuit_out_to_string_length(quid cid);
+void qrack_qcircuit_out_to_string(quid cid, char* out, uint64_t cap);
+
 /* file I/O (SURVEY.md §5 checkpoint formats) */
 int qrack_qstabilizer_out_to_file(quid sid, const char* path);
 quid qrack_qstabilizer_in_from_file(const char* path);
