@@ -16,6 +16,7 @@
 #include "qfactory.hpp"
 #include "qhybrid.hpp"
 #include "qneuron.hpp"
+#include "qunitmulti.hpp"
 #include "serialize.hpp"
 #ifdef QRACK_AMD_HIP_ENGINE
 #include "hip/qengine_hip.hpp"
@@ -682,6 +683,17 @@ template <typename R> static void bindQInterface(py::module_& m, const char* nam
                 });
             })
         .def("is_clifford", [](QI& q) { return q.isClifford(); })
+        .def("unit_placement",
+            [](Ptr q) {
+                // (unit width, device id) per distinct QUnitMulti unit
+                std::vector<std::pair<int, long long>> out;
+                if (auto um = std::dynamic_pointer_cast<QUnitMulti<R>>(q)) {
+                    for (auto& p : um->UnitPlacement()) {
+                        out.push_back({ (int)p.first, (long long)p.second });
+                    }
+                }
+                return out;
+            })
         .def("hybrid_mode",
             [](Ptr q) -> std::string {
                 if (auto hy = std::dynamic_pointer_cast<QHybrid<R>>(q)) return hy->ModeName();

@@ -91,6 +91,7 @@ QInterfacePtr<R> QUnit<R>::EntangleAll(const std::vector<bitLenInt>& qs)
             }
         }
     }
+    if (units.size() > 1u) OnStructureChanged();
     return base;
 }
 
@@ -139,6 +140,7 @@ template <typename R> void QUnit<R>::SeparateBit(bitLenInt q, bool value)
     s.unit = MakeUnit(1u, value ? 1u : 0u);
     s.mapped = 0;
     FixMappedAfterRemoval(unit, mapped);
+    OnStructureChanged();
 }
 
 template <typename R>
@@ -806,6 +808,7 @@ template <typename R> bool QUnit<R>::TrySeparate(bitLenInt q)
         }
         s.unit = solo;
         s.mapped = 0u;
+        OnStructureChanged();
         return true;
     }
     // 3-axis Bloch tomography (parity: qunit.cpp:696-855)
@@ -843,6 +846,7 @@ template <typename R> bool QUnit<R>::TrySeparate(bitLenInt q)
     // restore the local state: V^dagger |outcome>
     const cplx<R> Vd[4] = { conj(V[0]), conj(V[2]), conj(V[1]), conj(V[3]) };
     s.unit->Mtrx(Vd, 0);
+    OnStructureChanged();
     return true;
 }
 
@@ -881,6 +885,7 @@ template <typename R> bool QUnit<R>::TrySeparate(bitLenInt q1, bitLenInt q2)
     shards[q1].mapped = 0;
     shards[q2].unit = pairUnit;
     shards[q2].mapped = 1;
+    OnStructureChanged();
     return true;
 }
 

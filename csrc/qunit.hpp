@@ -409,12 +409,17 @@ protected:
             TrySeparate(q);
     }
 
-    QInterfacePtr<R> MakeUnit(bitLenInt n, bitCapInt perm)
+    virtual QInterfacePtr<R> MakeUnit(bitLenInt n, bitCapInt perm)
     {
         QInterfacePtr<R> u = subFactory(n, perm);
         if (ncrp > 0.0) u->SetNcrp(ncrp);
         return u;
     }
+
+    // notification hook: fired after any entangle/separate/structural change
+    // (QUnitMulti rebalances shard units across devices here — reference
+    // qunitmulti.cpp:217-274 redistributes after every entangle/separate)
+    virtual void OnStructureChanged() {}
 
     // merge all units containing `qs` into one; returns it
     QInterfacePtr<R> EntangleAll(const std::vector<bitLenInt>& qs);

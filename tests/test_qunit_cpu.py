@@ -326,3 +326,36 @@ def test_optimal_stack_alias():
     cp.t(1)
     cp.cnot(1, 2)
     assert_states_close(q.get_state_vector(), cp.get_state_vector(), 1e-4)
+
+
+def test_qunitmulti_placement_balances():
+    """QUnitMulti rebalances after EVERY
+    entangle/separate (OnStructureChanged hook), largest units first —
+    verified through the fake-device placement seam on CPU."""
+    import os
+
+    os.environ["QRACK_FAKE_DEVICES"] = "2"
+    try:
+        n = 8
+        q = qa.create_simulator(n, layers=["qunit_multi", "cpu"], seed=3)
+        # two entangled 4-qubit groups
+        for base in (0, 4):
+            q.h(base)
+            for i in range(3):
+                q.ry(0.4, base + i)  # non-Clifford: no phase-buffer deferral
+                q.cnot(base + i, base + i + 1)
+                q.ry(0.3, base + i + 1)
+        placement = q.unit_placement()
+        big = [(w, d) for (w, d) in placement if w == 4]
+        assert len(big) == 2, placement
+        # the two 4-qubit units must land on DIFFERENT devices
+        assert big[0][1] != big[1][1], placement
+        # measuring one group splits it back; rebalancing keeps devices valid
+        for i in range(4):
+            q.m(i)
+        placement2 = q.unit_placement()
+        assert all(d in (0, 1) for (_, d) in placement2)
+        # states still correct
+        assert 0.0 <= q.prob(5) <= 1.0
+    finally:
+        del os.environ["QRACK_FAKE_DEVICES"]
