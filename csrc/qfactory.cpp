@@ -6,6 +6,7 @@
 #include "qinterface_noisy.hpp"
 #include "qbdt.hpp"
 #include "qengine_sparse.hpp"
+#include "qengine_turboquant.hpp"
 #include "qbdthybrid.hpp"
 #include "qpager.hpp"
 #include "qtensornetwork.hpp"
@@ -64,6 +65,13 @@ EngineFactoryFn<R> LayerFactory(std::vector<std::string> layers, RngPtr rng, int
     if (head == "sparse") {
         return [rng](bitLenInt n, bitCapInt perm) -> QInterfacePtr<R> {
             return std::make_shared<QEngineSparse<R>>(n, perm, rng);
+        };
+    }
+    if (head == "turboquant") {
+        // block-compressed RUNTIME storage backend (reference
+        // statevector_turboquant.hpp:449-530 as a live engine)
+        return [rng](bitLenInt n, bitCapInt perm) -> QInterfacePtr<R> {
+            return std::make_shared<QEngineTurboQuant<R>>(n, perm, rng);
         };
     }
     if (head == "stabilizer") {
