@@ -6,6 +6,7 @@
 // stream format).
 #pragma once
 
+#include <complex>
 #include "qinterface.hpp"
 
 #include <list>
@@ -183,22 +184,40 @@ public:
     // ASCII stream format (parity model: qcircuit.cpp:17-101)
     std::string Serialize() const
     {
+        // Reference-interchangeable stream (operator<< at the reference's
+        // qcircuit.cpp:17-80): whitespace-separated "n ngates" then per gate
+        // "target nc c... np key (re,im)x4 ..." with std::complex formatting.
         std::ostringstream os;
-        os << (uint64_t)qubitCount << "\n" << gates.size() << "\n";
+        os.precision(17);
+        os << (uint64_t)qubitCount << " " << gates.size() << " ";
         for (const auto& g : gates) {
-            os << (uint64_t)g.target << "\n";
-            os << g.controls.size();
-            for (bitLenInt c : g.controls) os << " " << (uint64_t)c;
-            os << "\n" << g.payloads.size() << "\n";
+            os << (uint64_t)g.target << " ";
+            os << g.controls.size() << " ";
+            for (bitLenInt c : g.controls) os << (uint64_t)c << " ";
+            os << g.payloads.size() << " ";
             for (auto& kv : g.payloads) {
-                os << kv.first;
+                os << kv.first << " ";
                 for (int i = 0; i < 4; ++i) {
-                    os << " " << kv.second[i].re << " " << kv.second[i].im;
+                    os << "(" << kv.second[i].re << "," << kv.second[i].im << ") ";
                 }
-                os << "\n";
             }
         }
         return os.str();
+    }
+
+    // complex token reader: accepts the reference's "(re,im)" AND the
+    // legacy qrack_amd "re im" pair
+    static cplx<R> ReadCplx(std::istream& is)
+    {
+        is >> std::ws;
+        if (is.peek() == '(') {
+            std::complex<double> c;
+            is >> c;
+            return cplx<R>((R)c.real(), (R)c.imag());
+        }
+        double re = 0, im = 0;
+        is >> re >> im;
+        return cplx<R>((R)re, (R)im);
     }
 
     static QCircuitPtr<R> Deserialize(const std::string& s)
@@ -226,9 +245,7 @@ public:
                 bitCapInt perm = 0;
                 is >> perm;
                 std::array<cplx<R>, 4> m;
-                for (int j = 0; j < 4; ++j) {
-                    is >> m[j].re >> m[j].im;
-                }
+                for (int j = 0; j < 4; ++j) m[j] = ReadCplx(is);
                 g.payloads[perm] = m;
             }
             c->gates.push_back(g);

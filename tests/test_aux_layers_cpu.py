@@ -351,3 +351,32 @@ def test_qunit_aware_lossy_container(tmp_path):
     r1 = q.pauli_expectation([0, 1], [2, 2])
     r2 = q2.pauli_expectation([0, 1], [2, 2])
     assert abs(r1 - r2) < 1e-3
+
+
+def test_qcircuit_text_reference_format():
+    """Circuit stream uses the reference's token shape (qcircuit.cpp:17-80):
+    whitespace-separated with std::complex '(re,im)' payload entries; a
+    reference-produced stream loads, and legacy 're im' pairs still load."""
+    c = qa.QCircuitF(2)
+    s = 0.7071067811865476
+    c.append_mtrx([s, s, s, -s], 0)
+    c.append_controlled([0, 1, 1, 0], 1, [0], 1)
+    text = c.serialize()
+    assert "(" in text and "," in text  # std::complex formatting
+    c2 = qa.QCircuitF.deserialize(text)
+    assert c2.num_qubits == 2
+    assert c2.gate_count == 2
+    # roundtrip runs identically
+    q1 = make_cpu(2)
+    q2 = make_cpu(2)
+    c.run(q1)
+    c2.run(q2)
+    assert_states_close(q1.get_state_vector(), q2.get_state_vector(), 1e-6)
+    # reference-shaped stream parses (CNOT with control 0, target 1)
+    ref_stream = "2 1 1 1 0 1 1 (0,0) (1,0) (1,0) (0,0) "
+    c3 = qa.QCircuitF.deserialize(ref_stream)
+    assert c3.num_qubits == 2
+    # legacy pair format parses
+    legacy = "2\n1\n1\n1 0\n1\n1 0 0 1 0 1 0 0 0\n"
+    c4 = qa.QCircuitF.deserialize(legacy)
+    assert c4.num_qubits == 2
