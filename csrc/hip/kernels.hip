@@ -947,6 +947,11 @@ void launchPhaseFlipIfLess(cplx<R>* sv, bitCapInt maxQPower, bitCapInt greaterPe
 
 // ---- fused QFT phase ramp ------------------------------------------------------
 
+__device__ __forceinline__ bitCapInt insertZeroBitDev(bitCapInt i, bitCapInt p)
+{
+    return ((i & ~(p - 1u)) << 1u) | (i & (p - 1u));
+}
+
 template <typename R> __device__ __forceinline__ void devSinCos(R t, R* s, R* c);
 template <> __device__ __forceinline__ void devSinCos<float>(float t, float* s, float* c)
 {
@@ -1362,6 +1367,154 @@ void launchQftColumnTopRange(cplx<R>* sv, bitCapInt maxQPower, const RampArgs& a
         else
             hipLaunchKernelGGL((k_qft_col_top_range<R, false, false>), g, b, 0, stream, sv, tPow,
                 a, (R)phase0, itLo, itHi, recvSrc);
+    }
+}
+
+// TWO QFT columns fused in ONE full-state pass (halves the pass count of
+// the already-fused per-column ladder): columns (col, col-1) process
+// 4-amplitude orbits over bits (tHi, tLo). The cross-column ramp term is a
+// CONSTANT factor iF = e^{i·sign·π/2} and the lower column's ramp is the
+// square of the upper's, so ONE sincos per orbit drives all three phase
+// factors:  f_hi(bLo) = f0·iF^bLo,  f_lo = f0².
+// Forward (PRE=false, QFT): H_hi, ramp_hi, H_lo, ramp_lo.
+// Inverse (PRE=true, IQFT): ramp_lo, H_lo, ramp_hi, H_hi (exact adjoint).
+template <typename R, bool PRE>
+__global__ void k_qft_col2(cplx<R>* sv, bitCapInt orbits, bitCapInt tHi, bitCapInt tLo,
+    bitLenInt rampStart, bitCapInt lowMask, R scaleHi)
+{
+    const R s = (R)0.70710678118654752440;
+    const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
+    // iF = e^{i·sign·π/2}; sign is carried in scaleHi's sign
+    const R iSign = (scaleHi >= 0) ? (R)1 : (R)-1;
+    for (bitCapInt k = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; k < orbits;
+         k += stride) {
+        const bitCapInt r = insertZeroBitDev(insertZeroBitDev(k, tLo), tHi);
+        const uint64_t lf = (uint64_t)((r >> rampStart) & lowMask);
+        R sn, cs;
+        devSinCos<R>(scaleHi * (R)lf, &sn, &cs);
+        const cplx<R> f0{ cs, sn };
+        const cplx<R> iF{ 0, iSign };
+        const cplx<R> fHi0 = f0, fHi1 = f0 * iF;
+        const cplx<R> fLo = f0 * f0;
+        cplx<R> a00 = sv[r];
+        cplx<R> a01 = sv[r | tLo];
+        cplx<R> a10 = sv[r | tHi];
+        cplx<R> a11 = sv[r | tHi | tLo];
+        if (!PRE) {
+            // H_hi then ramp_hi (phase on bHi=1, depends on bLo)
+            cplx<R> b00 = s * (a00 + a10), b01 = s * (a01 + a11);
+            cplx<R> b10 = fHi0 * (s * (a00 - a10)), b11 = fHi1 * (s * (a01 - a11));
+            // H_lo then ramp_lo (phase on bLo=1)
+            sv[r] = s * (b00 + b01);
+            sv[r | tLo] = fLo * (s * (b00 - b01));
+            sv[r | tHi] = s * (b10 + b11);
+            sv[r | tHi | tLo] = fLo * (s * (b10 - b11));
+        } else {
+            // ramp_lo then H_lo
+            a01 = fLo * a01;
+            a11 = fLo * a11;
+            cplx<R> b00 = s * (a00 + a01), b01 = s * (a00 - a01);
+            cplx<R> b10 = s * (a10 + a11), b11 = s * (a10 - a11);
+            // ramp_hi then H_hi
+            b10 = fHi0 * b10;
+            b11 = fHi1 * b11;
+            sv[r] = s * (b00 + b10);
+            sv[r | tHi] = s * (b00 - b10);
+            sv[r | tLo] = s * (b01 + b11);
+            sv[r | tHi | tLo] = s * (b01 - b11);
+        }
+    }
+}
+
+// float4-vectorized: each thread drives TWO adjacent orbits (8 amplitudes,
+// 4 float4 RMWs). Requires tLo >= 2.
+template <bool PRE>
+__global__ void k_qft_col2_v(cplx<float>* sv, bitCapInt orbitPairs, bitCapInt tHi, bitCapInt tLo,
+    bitLenInt rampStart, bitCapInt lowMask, float scaleHi)
+{
+    float4* sv4 = reinterpret_cast<float4*>(sv);
+    const float s = 0.70710678f;
+    const float iSign = (scaleHi >= 0) ? 1.0f : -1.0f;
+    const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
+    for (bitCapInt m = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; m < orbitPairs;
+         m += stride) {
+        const bitCapInt r = insertZeroBitDev(insertZeroBitDev(2u * m, tLo), tHi);
+        const bitCapInt i00 = r >> 1u;
+        const bitCapInt i01 = (r | tLo) >> 1u;
+        const bitCapInt i10 = (r | tHi) >> 1u;
+        const bitCapInt i11 = (r | tHi | tLo) >> 1u;
+        float4 v00 = sv4[i00], v01 = sv4[i01], v10 = sv4[i10], v11 = sv4[i11];
+        const uint64_t lf0 = (uint64_t)((r >> rampStart) & lowMask);
+        const uint64_t lf1 = (uint64_t)(((r + 1u) >> rampStart) & lowMask);
+        float s0, c0, s1, c1;
+        __sincosf(scaleHi * (float)lf0, &s0, &c0);
+        __sincosf(scaleHi * (float)lf1, &s1, &c1);
+        const cplx<float> f0a{ c0, s0 }, f0b{ c1, s1 };
+        const cplx<float> iF{ 0.0f, iSign };
+#define QA_COL2_BODY(aA, aB, aC, aD, f0)                                                           \
+    {                                                                                              \
+        const cplx<float> fHi0 = f0, fHi1 = f0 * iF, fLo = f0 * f0;                                \
+        if (!PRE) {                                                                                \
+            cplx<float> b00 = s * (aA + aC), b01 = s * (aB + aD);                                  \
+            cplx<float> b10 = fHi0 * (s * (aA - aC)), b11 = fHi1 * (s * (aB - aD));                \
+            aA = s * (b00 + b01);                                                                  \
+            aB = fLo * (s * (b00 - b01));                                                          \
+            aC = s * (b10 + b11);                                                                  \
+            aD = fLo * (s * (b10 - b11));                                                          \
+        } else {                                                                                   \
+            cplx<float> t01 = fLo * aB, t11 = fLo * aD;                                            \
+            cplx<float> b00 = s * (aA + t01), b01 = s * (aA - t01);                                \
+            cplx<float> b10 = s * (aC + t11), b11 = s * (aC - t11);                                \
+            b10 = fHi0 * b10;                                                                      \
+            b11 = fHi1 * b11;                                                                      \
+            aA = s * (b00 + b10);                                                                  \
+            aC = s * (b00 - b10);                                                                  \
+            aB = s * (b01 + b11);                                                                  \
+            aD = s * (b01 - b11);                                                                  \
+        }                                                                                          \
+    }
+        cplx<float> x00{ v00.x, v00.y }, y00{ v00.z, v00.w };
+        cplx<float> x01{ v01.x, v01.y }, y01{ v01.z, v01.w };
+        cplx<float> x10{ v10.x, v10.y }, y10{ v10.z, v10.w };
+        cplx<float> x11{ v11.x, v11.y }, y11{ v11.z, v11.w };
+        QA_COL2_BODY(x00, x01, x10, x11, f0a);
+        QA_COL2_BODY(y00, y01, y10, y11, f0b);
+#undef QA_COL2_BODY
+        sv4[i00] = make_float4(x00.re, x00.im, y00.re, y00.im);
+        sv4[i01] = make_float4(x01.re, x01.im, y01.re, y01.im);
+        sv4[i10] = make_float4(x10.re, x10.im, y10.re, y10.im);
+        sv4[i11] = make_float4(x11.re, x11.im, y11.re, y11.im);
+    }
+}
+
+template <typename R>
+void launchQftColumn2(cplx<R>* sv, bitCapInt maxQPower, bitLenInt rampStart, bitLenInt col,
+    bitCapInt tHi, bitCapInt tLo, int sign, bool pre, hipStream_t stream)
+{
+    // columns (col, col-1): ramp low-bit count = col-1 bits below tLo
+    const bitCapInt lowMask = (ONE_BCI << (col - 1u)) - 1u;
+    const R scaleHi = (R)sign * (R)3.14159265358979323846 / (R)(ONE_BCI << col);
+    const bitCapInt orbits = maxQPower >> 2u;
+    if constexpr (std::is_same_v<R, float>) {
+        if (tLo >= 2u && (orbits & 1u) == 0u) {
+            if (pre) {
+                hipLaunchKernelGGL((k_qft_col2_v<true>), dim3(gridFor(orbits >> 1u)),
+                    dim3(QA_BLOCK), 0, stream, sv, orbits >> 1u, tHi, tLo, rampStart, lowMask,
+                    (float)scaleHi);
+            } else {
+                hipLaunchKernelGGL((k_qft_col2_v<false>), dim3(gridFor(orbits >> 1u)),
+                    dim3(QA_BLOCK), 0, stream, sv, orbits >> 1u, tHi, tLo, rampStart, lowMask,
+                    (float)scaleHi);
+            }
+            return;
+        }
+    }
+    if (pre) {
+        hipLaunchKernelGGL((k_qft_col2<R, true>), dim3(gridFor(orbits)), dim3(QA_BLOCK), 0,
+            stream, sv, orbits, tHi, tLo, rampStart, lowMask, scaleHi);
+    } else {
+        hipLaunchKernelGGL((k_qft_col2<R, false>), dim3(gridFor(orbits)), dim3(QA_BLOCK), 0,
+            stream, sv, orbits, tHi, tLo, rampStart, lowMask, scaleHi);
     }
 }
 
@@ -1979,6 +2132,8 @@ void launchPartProbs(const cplx<R>* sv, bitCapInt maxQPower, bitLenInt start, bi
         const cplx<R>*, bitCapInt, bitLenInt, bitLenInt, double*, hipStream_t);                     \
     template void launchPhaseRamp<R>(cplx<R>*, bitCapInt, bitLenInt, bitLenInt, bitCapInt, double, hipStream_t);\
     template void launchPhaseRampGeneral<R>(cplx<R>*, bitCapInt, const RampArgs&, hipStream_t);    \
+    template void launchQftColumn2<R>(cplx<R>*, bitCapInt, bitLenInt, bitLenInt, bitCapInt,        \
+        bitCapInt, int, bool, hipStream_t);                                                         \
     template void launchQftColumn<R>(                                                               \
         cplx<R>*, bitCapInt, bitLenInt, bitLenInt, bitCapInt, int, bool, hipStream_t);              \
     template void launchMtrx1qBatch<R>(cplx<R>*, const Batch1qArgs<R>&, hipStream_t);                               \

@@ -182,6 +182,13 @@ struct RampArgs {
 template <typename R>
 void launchPhaseRampGeneral(cplx<R>* sv, bitCapInt maxQPower, const RampArgs& a, hipStream_t stream);
 
+// TWO fused QFT columns in one state pass (see kernels.hip k_qft_col2):
+// columns (col, col-1) as 4-amplitude orbits — ONE sincos per orbit drives
+// both ramps (cross term is a constant ±i factor, lower ramp = f0^2)
+template <typename R>
+void launchQftColumn2(cplx<R>* sv, bitCapInt maxQPower, bitLenInt rampStart, bitLenInt col,
+    bitCapInt tHi, bitCapInt tLo, int sign, bool pre, hipStream_t stream);
+
 // generalized fused QFT column: H on tPow + the (possibly relocated-bit)
 // ramp of RampArgs in ONE pass; phase0 is a constant phase folded onto the
 // target=1 side (distributed pager meta-page scalar)

@@ -811,13 +811,28 @@ template <typename R> void QEngineHIP<R>::QFT(bitLenInt start, bitLenInt length,
     // bit-set half) — a single state pass per column
     if (!length) return;
     QA_HIP_CHECK(hipSetDevice(deviceId));
-    for (bitLenInt i = length; i-- > 0;) {
-        if (!i) {
+    static const bool fuse2 = []() {
+        if (const char* env = std::getenv("QRACK_GPU_QFT_FUSE2")) return std::atoi(env) != 0;
+        return true;
+    }();
+    bitLenInt i = length;
+    while (i > 0) {
+        const bitLenInt col = i - 1u;
+        if (fuse2 && col >= 1u && maxQPower >= 8u) {
+            // two columns per pass: halves the full-state pass count
+            HipProfScope prof("qft_column2", stream);
+            launchQftColumn2<R>(dState, maxQPower, start, col, pow2(start + col),
+                pow2(start + col - 1u), +1, false, stream);
+            i -= 2u;
+            continue;
+        }
+        if (!col) {
             this->H(start);
             break;
         }
         HipProfScope prof("qft_column", stream);
-        launchQftColumn<R>(dState, maxQPower, start, i, pow2(start + i), +1, false, stream);
+        launchQftColumn<R>(dState, maxQPower, start, col, pow2(start + col), +1, false, stream);
+        --i;
     }
 }
 
@@ -825,13 +840,29 @@ template <typename R> void QEngineHIP<R>::IQFT(bitLenInt start, bitLenInt length
 {
     if (!length) return;
     QA_HIP_CHECK(hipSetDevice(deviceId));
-    for (bitLenInt i = 0; i < length; ++i) {
+    static const bool fuse2 = []() {
+        if (const char* env = std::getenv("QRACK_GPU_QFT_FUSE2")) return std::atoi(env) != 0;
+        return true;
+    }();
+    bitLenInt i = 0;
+    while (i < length) {
+        if (fuse2 && (i + 1u) < length && maxQPower >= 8u) {
+            // pair (lo=i, hi=i+1) in one pass (exact adjoint of the forward
+            // pair; a lo column of 0 degenerates to the plain H inside)
+            HipProfScope prof("qft_column2", stream);
+            launchQftColumn2<R>(dState, maxQPower, start, (bitLenInt)(i + 1u),
+                pow2(start + i + 1u), pow2(start + i), -1, true, stream);
+            i += 2u;
+            continue;
+        }
         if (!i) {
             this->H(start);
+            ++i;
             continue;
         }
         HipProfScope prof("qft_column", stream);
         launchQftColumn<R>(dState, maxQPower, start, i, pow2(start + i), -1, true, stream);
+        ++i;
     }
 }
 

@@ -547,3 +547,28 @@ print("OK")
     assert "OK" in r1.stdout, r1.stderr
     got = np.load("/tmp/mfma_ab.npy")
     assert np.abs(ref - got).max() < 1e-5
+
+
+def test_qft_fused2_numerics_vs_cpu():
+    """The 2-column fused QFT kernel (k_qft_col2) must match the CPU engine
+    exactly (both directions, odd and even lengths)."""
+    for n in (9, 12):
+        for init in (0, 0b101101, (1 << n) - 3):
+            q = make(n, seed=n)
+            cp = qa.create_simulator(n, engine="cpu", seed=n)
+            q.set_permutation(init & ((1 << n) - 1))
+            cp.set_permutation(init & ((1 << n) - 1))
+            q.qft(0, n)
+            cp.qft(0, n)
+            assert_states_close(q.get_state_vector(), cp.get_state_vector(), 2e-4)
+            q.iqft(0, n)
+            cp.iqft(0, n)
+            assert_states_close(q.get_state_vector(), cp.get_state_vector(), 2e-4)
+    # offset register QFT (rampStart > 0)
+    q = make(10, seed=3)
+    cp = qa.create_simulator(10, engine="cpu", seed=3)
+    q.set_permutation(0b1011010010)
+    cp.set_permutation(0b1011010010)
+    q.qft(3, 6)
+    cp.qft(3, 6)
+    assert_states_close(q.get_state_vector(), cp.get_state_vector(), 2e-4)
