@@ -1487,6 +1487,219 @@ __global__ void k_qft_col2_v(cplx<float>* sv, bitCapInt orbitPairs, bitCapInt tH
     }
 }
 
+// THREE QFT columns fused per pass (8-amplitude orbits). Phase structure:
+// with f0 = e^{i·s_hi·lf}, A = e^{i·sign·π/2} = ±i, B = e^{i·sign·π/4}:
+//   ramp_hi(bMid,bLo) = f0·A^bMid·B^bLo, ramp_mid(bLo) = f0²·A^bLo,
+//   ramp_lo = f0⁴ — still ONE sincos per orbit.
+template <typename R, bool PRE>
+__global__ void k_qft_col3(cplx<R>* sv, bitCapInt orbits, bitCapInt tHi, bitCapInt tMid,
+    bitCapInt tLo, bitLenInt rampStart, bitCapInt lowMask, R scaleHi)
+{
+    const R s = (R)0.70710678118654752440;
+    const R c45 = (R)0.70710678118654752440;
+    const R iSign = (scaleHi >= 0) ? (R)1 : (R)-1;
+    const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
+    for (bitCapInt k = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; k < orbits;
+         k += stride) {
+        const bitCapInt r =
+            insertZeroBitDev(insertZeroBitDev(insertZeroBitDev(k, tLo), tMid), tHi);
+        bitCapInt idx[8];
+        for (int b = 0; b < 8; ++b) {
+            idx[b] = r | ((b & 4) ? tHi : 0u) | ((b & 2) ? tMid : 0u) | ((b & 1) ? tLo : 0u);
+        }
+        cplx<R> v[8];
+        for (int b = 0; b < 8; ++b) v[b] = sv[idx[b]];
+        const uint64_t lf = (uint64_t)((r >> rampStart) & lowMask);
+        R sn, cs;
+        devSinCos<R>(scaleHi * (R)lf, &sn, &cs);
+        const cplx<R> f0{ cs, sn };
+        const cplx<R> A{ 0, iSign };
+        const cplx<R> B{ c45, iSign * c45 };
+        const cplx<R> f2 = f0 * f0;
+        const cplx<R> f4 = f2 * f2;
+        // per-slot hi-ramp factors for bHi=1: f0 * A^bMid * B^bLo
+        cplx<R> fh[4];
+        fh[0] = f0;
+        fh[1] = f0 * B;
+        fh[2] = f0 * A;
+        fh[3] = f0 * A * B;
+        auto hPair = [&](int lo, int hi) {
+            const cplx<R> t = s * (v[lo] + v[hi]);
+            const cplx<R> u = s * (v[lo] - v[hi]);
+            v[lo] = t;
+            v[hi] = u;
+        };
+        if (!PRE) {
+            for (int b = 0; b < 4; ++b) hPair(b, b | 4);
+            for (int m = 0; m < 4; ++m) v[4 | m] = fh[m] * v[4 | m];
+            hPair(0, 2);
+            hPair(1, 3);
+            hPair(4, 6);
+            hPair(5, 7);
+            v[2] = f2 * v[2];
+            v[3] = f2 * A * v[3];
+            v[6] = f2 * v[6];
+            v[7] = f2 * A * v[7];
+            hPair(0, 1);
+            hPair(2, 3);
+            hPair(4, 5);
+            hPair(6, 7);
+            v[1] = f4 * v[1];
+            v[3] = f4 * v[3];
+            v[5] = f4 * v[5];
+            v[7] = f4 * v[7];
+        } else {
+            // exact adjoint order (angles carry the sign): ramp_lo, H_lo,
+            // ramp_mid, H_mid, ramp_hi, H_hi
+            v[1] = f4 * v[1];
+            v[3] = f4 * v[3];
+            v[5] = f4 * v[5];
+            v[7] = f4 * v[7];
+            hPair(0, 1);
+            hPair(2, 3);
+            hPair(4, 5);
+            hPair(6, 7);
+            v[2] = f2 * v[2];
+            v[3] = f2 * A * v[3];
+            v[6] = f2 * v[6];
+            v[7] = f2 * A * v[7];
+            hPair(0, 2);
+            hPair(1, 3);
+            hPair(4, 6);
+            hPair(5, 7);
+            for (int m = 0; m < 4; ++m) v[4 | m] = fh[m] * v[4 | m];
+            for (int b = 0; b < 4; ++b) hPair(b, b | 4);
+        }
+        for (int b = 0; b < 8; ++b) sv[idx[b]] = v[b];
+    }
+}
+
+// float4-vectorized 3-column kernel: two adjacent orbits per thread
+// (16 amplitudes, 8 float4 RMWs); requires tLo >= 2.
+template <bool PRE>
+__global__ void k_qft_col3_v(cplx<float>* sv, bitCapInt orbitPairs, bitCapInt tHi, bitCapInt tMid,
+    bitCapInt tLo, bitLenInt rampStart, bitCapInt lowMask, float scaleHi)
+{
+    float4* sv4 = reinterpret_cast<float4*>(sv);
+    const float s = 0.70710678f;
+    const float iSign = (scaleHi >= 0) ? 1.0f : -1.0f;
+    const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
+    for (bitCapInt m = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; m < orbitPairs;
+         m += stride) {
+        const bitCapInt r =
+            insertZeroBitDev(insertZeroBitDev(insertZeroBitDev(2u * m, tLo), tMid), tHi);
+        bitCapInt idx[8];
+        for (int b = 0; b < 8; ++b) {
+            idx[b] =
+                (r | ((b & 4) ? tHi : 0u) | ((b & 2) ? tMid : 0u) | ((b & 1) ? tLo : 0u)) >> 1u;
+        }
+        float4 raw[8];
+        for (int b = 0; b < 8; ++b) raw[b] = sv4[idx[b]];
+        const uint64_t lf0 = (uint64_t)((r >> rampStart) & lowMask);
+        const uint64_t lf1 = (uint64_t)(((r + 1u) >> rampStart) & lowMask);
+        float s0, c0, s1, c1;
+        __sincosf(scaleHi * (float)lf0, &s0, &c0);
+        __sincosf(scaleHi * (float)lf1, &s1, &c1);
+        const cplx<float> A{ 0.0f, iSign };
+        const cplx<float> B{ 0.70710678f, iSign * 0.70710678f };
+        for (int half = 0; half < 2; ++half) {
+            const cplx<float> f0 = half ? cplx<float>{ c1, s1 } : cplx<float>{ c0, s0 };
+            const cplx<float> f2 = f0 * f0;
+            const cplx<float> f4 = f2 * f2;
+            cplx<float> fh[4] = { f0, f0 * B, f0 * A, f0 * A * B };
+            cplx<float> v[8];
+            for (int b = 0; b < 8; ++b) {
+                v[b] = half ? cplx<float>{ raw[b].z, raw[b].w } : cplx<float>{ raw[b].x, raw[b].y };
+            }
+            auto hPair = [&](int lo, int hi) {
+                const cplx<float> t = s * (v[lo] + v[hi]);
+                const cplx<float> u = s * (v[lo] - v[hi]);
+                v[lo] = t;
+                v[hi] = u;
+            };
+            if (!PRE) {
+                for (int b = 0; b < 4; ++b) hPair(b, b | 4);
+                for (int q = 0; q < 4; ++q) v[4 | q] = fh[q] * v[4 | q];
+                hPair(0, 2);
+                hPair(1, 3);
+                hPair(4, 6);
+                hPair(5, 7);
+                v[2] = f2 * v[2];
+                v[3] = f2 * A * v[3];
+                v[6] = f2 * v[6];
+                v[7] = f2 * A * v[7];
+                hPair(0, 1);
+                hPair(2, 3);
+                hPair(4, 5);
+                hPair(6, 7);
+                v[1] = f4 * v[1];
+                v[3] = f4 * v[3];
+                v[5] = f4 * v[5];
+                v[7] = f4 * v[7];
+            } else {
+                v[1] = f4 * v[1];
+                v[3] = f4 * v[3];
+                v[5] = f4 * v[5];
+                v[7] = f4 * v[7];
+                hPair(0, 1);
+                hPair(2, 3);
+                hPair(4, 5);
+                hPair(6, 7);
+                v[2] = f2 * v[2];
+                v[3] = f2 * A * v[3];
+                v[6] = f2 * v[6];
+                v[7] = f2 * A * v[7];
+                hPair(0, 2);
+                hPair(1, 3);
+                hPair(4, 6);
+                hPair(5, 7);
+                for (int q = 0; q < 4; ++q) v[4 | q] = fh[q] * v[4 | q];
+                for (int b = 0; b < 4; ++b) hPair(b, b | 4);
+            }
+            for (int b = 0; b < 8; ++b) {
+                if (half) {
+                    raw[b].z = v[b].re;
+                    raw[b].w = v[b].im;
+                } else {
+                    raw[b].x = v[b].re;
+                    raw[b].y = v[b].im;
+                }
+            }
+        }
+        for (int b = 0; b < 8; ++b) sv4[idx[b]] = raw[b];
+    }
+}
+
+template <typename R>
+void launchQftColumn3(cplx<R>* sv, bitCapInt maxQPower, bitLenInt rampStart, bitLenInt col,
+    bitCapInt tHi, bitCapInt tMid, bitCapInt tLo, int sign, bool pre, hipStream_t stream)
+{
+    const bitCapInt lowMask = (ONE_BCI << (col - 2u)) - 1u;
+    const R scaleHi = (R)sign * (R)3.14159265358979323846 / (R)(ONE_BCI << col);
+    const bitCapInt orbits = maxQPower >> 3u;
+    if constexpr (std::is_same_v<R, float>) {
+        if (tLo >= 2u && (orbits & 1u) == 0u) {
+            if (pre) {
+                hipLaunchKernelGGL((k_qft_col3_v<true>), dim3(gridFor(orbits >> 1u)),
+                    dim3(QA_BLOCK), 0, stream, sv, orbits >> 1u, tHi, tMid, tLo, rampStart,
+                    lowMask, (float)scaleHi);
+            } else {
+                hipLaunchKernelGGL((k_qft_col3_v<false>), dim3(gridFor(orbits >> 1u)),
+                    dim3(QA_BLOCK), 0, stream, sv, orbits >> 1u, tHi, tMid, tLo, rampStart,
+                    lowMask, (float)scaleHi);
+            }
+            return;
+        }
+    }
+    if (pre) {
+        hipLaunchKernelGGL((k_qft_col3<R, true>), dim3(gridFor(orbits)), dim3(QA_BLOCK), 0,
+            stream, sv, orbits, tHi, tMid, tLo, rampStart, lowMask, scaleHi);
+    } else {
+        hipLaunchKernelGGL((k_qft_col3<R, false>), dim3(gridFor(orbits)), dim3(QA_BLOCK), 0,
+            stream, sv, orbits, tHi, tMid, tLo, rampStart, lowMask, scaleHi);
+    }
+}
+
 template <typename R>
 void launchQftColumn2(cplx<R>* sv, bitCapInt maxQPower, bitLenInt rampStart, bitLenInt col,
     bitCapInt tHi, bitCapInt tLo, int sign, bool pre, hipStream_t stream)
@@ -2134,6 +2347,8 @@ void launchPartProbs(const cplx<R>* sv, bitCapInt maxQPower, bitLenInt start, bi
     template void launchPhaseRampGeneral<R>(cplx<R>*, bitCapInt, const RampArgs&, hipStream_t);    \
     template void launchQftColumn2<R>(cplx<R>*, bitCapInt, bitLenInt, bitLenInt, bitCapInt,        \
         bitCapInt, int, bool, hipStream_t);                                                         \
+    template void launchQftColumn3<R>(cplx<R>*, bitCapInt, bitLenInt, bitLenInt, bitCapInt,        \
+        bitCapInt, bitCapInt, int, bool, hipStream_t);                                              \
     template void launchQftColumn<R>(                                                               \
         cplx<R>*, bitCapInt, bitLenInt, bitLenInt, bitCapInt, int, bool, hipStream_t);              \
     template void launchMtrx1qBatch<R>(cplx<R>*, const Batch1qArgs<R>&, hipStream_t);                               \

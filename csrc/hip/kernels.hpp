@@ -189,6 +189,11 @@ template <typename R>
 void launchQftColumn2(cplx<R>* sv, bitCapInt maxQPower, bitLenInt rampStart, bitLenInt col,
     bitCapInt tHi, bitCapInt tLo, int sign, bool pre, hipStream_t stream);
 
+// THREE fused QFT columns per pass (8-amplitude orbits; see k_qft_col3)
+template <typename R>
+void launchQftColumn3(cplx<R>* sv, bitCapInt maxQPower, bitLenInt rampStart, bitLenInt col,
+    bitCapInt tHi, bitCapInt tMid, bitCapInt tLo, int sign, bool pre, hipStream_t stream);
+
 // generalized fused QFT column: H on tPow + the (possibly relocated-bit)
 // ramp of RampArgs in ONE pass; phase0 is a constant phase folded onto the
 // target=1 side (distributed pager meta-page scalar)

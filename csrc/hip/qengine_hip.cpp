@@ -811,15 +811,26 @@ template <typename R> void QEngineHIP<R>::QFT(bitLenInt start, bitLenInt length,
     // bit-set half) — a single state pass per column
     if (!length) return;
     QA_HIP_CHECK(hipSetDevice(deviceId));
-    static const bool fuse2 = []() {
-        if (const char* env = std::getenv("QRACK_GPU_QFT_FUSE2")) return std::atoi(env) != 0;
-        return true;
+    static const int fuseMax = []() {
+        if (const char* env = std::getenv("QRACK_GPU_QFT_FUSE2")) {
+            return std::atoi(env) != 0 ? 3 : 1;
+        }
+        if (const char* env = std::getenv("QRACK_GPU_QFT_FUSE")) return std::atoi(env);
+        return 3;
     }();
     bitLenInt i = length;
     while (i > 0) {
         const bitLenInt col = i - 1u;
-        if (fuse2 && col >= 1u && maxQPower >= 8u) {
-            // two columns per pass: halves the full-state pass count
+        if (fuseMax >= 3 && col >= 2u && maxQPower >= 16u) {
+            // three columns per pass (8-amplitude orbits)
+            HipProfScope prof("qft_column3", stream);
+            launchQftColumn3<R>(dState, maxQPower, start, col, pow2(start + col),
+                pow2(start + col - 1u), pow2(start + col - 2u), +1, false, stream);
+            i -= 3u;
+            continue;
+        }
+        if (fuseMax >= 2 && col >= 1u && maxQPower >= 8u) {
+            // two columns per pass
             HipProfScope prof("qft_column2", stream);
             launchQftColumn2<R>(dState, maxQPower, start, col, pow2(start + col),
                 pow2(start + col - 1u), +1, false, stream);
@@ -840,15 +851,25 @@ template <typename R> void QEngineHIP<R>::IQFT(bitLenInt start, bitLenInt length
 {
     if (!length) return;
     QA_HIP_CHECK(hipSetDevice(deviceId));
-    static const bool fuse2 = []() {
-        if (const char* env = std::getenv("QRACK_GPU_QFT_FUSE2")) return std::atoi(env) != 0;
-        return true;
+    static const int fuseMax = []() {
+        if (const char* env = std::getenv("QRACK_GPU_QFT_FUSE2")) {
+            return std::atoi(env) != 0 ? 3 : 1;
+        }
+        if (const char* env = std::getenv("QRACK_GPU_QFT_FUSE")) return std::atoi(env);
+        return 3;
     }();
     bitLenInt i = 0;
     while (i < length) {
-        if (fuse2 && (i + 1u) < length && maxQPower >= 8u) {
-            // pair (lo=i, hi=i+1) in one pass (exact adjoint of the forward
-            // pair; a lo column of 0 degenerates to the plain H inside)
+        if (fuseMax >= 3 && (i + 2u) < length && maxQPower >= 16u) {
+            // triple (lo=i, mid=i+1, hi=i+2): exact adjoint of the forward
+            // triple (a lo column of 0 degenerates to the plain H inside)
+            HipProfScope prof("qft_column3", stream);
+            launchQftColumn3<R>(dState, maxQPower, start, (bitLenInt)(i + 2u),
+                pow2(start + i + 2u), pow2(start + i + 1u), pow2(start + i), -1, true, stream);
+            i += 3u;
+            continue;
+        }
+        if (fuseMax >= 2 && (i + 1u) < length && maxQPower >= 8u) {
             HipProfScope prof("qft_column2", stream);
             launchQftColumn2<R>(dState, maxQPower, start, (bitLenInt)(i + 1u),
                 pow2(start + i + 1u), pow2(start + i), -1, true, stream);

@@ -157,15 +157,17 @@ def test_capi_shor_order_finding(lib):
         lib.qrack_h(sid, u64(i))
     # |ctrl>|0> -> |ctrl>|7^ctrl mod 15> (POWModNOut Shor building block)
     lib.qrack_pown(sid, u64(7), u64(15), u64(0), u64(n_ctrl), u64(n_ctrl))
-    lib.qrack_iqft(sid, u64(0), u64(n_ctrl))
+    # forward no-swap QFT + bit-reversed read = phase-estimation inverse
+    lib.qrack_qft(sid, u64(0), u64(n_ctrl))
     qubits = (ctypes.c_uint64 * n_ctrl)(*range(n_ctrl))
     shots = (ctypes.c_uint64 * 256)()
     lib.qrack_measure_shots_qubits(sid, qubits, n_ctrl, 256, shots)
-    # order r=4: control register collapses near multiples of 64/4 = 16
-    peaks = [0, 16, 32, 48]
-    near_peak = sum(1 for k in range(256)
-                    if min(abs(int(shots[k]) - p) for p in peaks + [64]) <= 2)
-    assert near_peak >= 200, near_peak
+    # order r=4: bit-reversed peaks are EXACTLY the values 0..3 (multiples
+    # of 2^6/4 = 16 reversed in 6 bits); every shot must land on one
+    for k in range(256):
+        v = int(shots[k])
+        rev = int(format(v, f"0{n_ctrl}b")[::-1], 2)
+        assert rev % 16 == 0, (v, rev)
     lib.qrack_destroy(sid)
 
 
