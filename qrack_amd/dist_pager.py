@@ -515,9 +515,22 @@ class DistQPager:
             # RCCL over xGMI: all chunk sendrecvs issued up front (they run
             # back-to-back on NCCL's comm stream); each req.wait() is a
             # stream-wait on torch's current stream, so chunk c's kernel
-            # overlaps chunk c+1's transfer
+            # overlaps chunk c+1's transfer. If the VERY FIRST chunk issue
+            # fails (communicator-level refusal — page data untouched), fall
+            # back to the gloo-host staging path instead of dying.
             reqs = []
-            for lo, hi in bounds:
+            try:
+                lo0, hi0 = bounds[0]
+                reqs.append(dist.batch_isend_irecv([
+                    dist.P2POp(dist.isend, view[lo0:hi0], partner),
+                    dist.P2POp(dist.irecv, tmp[lo0:hi0], partner),
+                ]))
+            except Exception:
+                self._pipe_gloo_chunks(bounds, view, tmp, partner, scale, rs, in_place,
+                                       pows, ws, meta_w, pre, base, elem, recv_is_low, ext)
+                torch.cuda.synchronize(self.device_id)
+                return
+            for lo, hi in bounds[1:]:
                 reqs.append(dist.batch_isend_irecv([
                     dist.P2POp(dist.isend, view[lo:hi], partner),
                     dist.P2POp(dist.irecv, tmp[lo:hi], partner),
@@ -533,20 +546,25 @@ class DistQPager:
             # device — "Duplicate GPU detected"): host-staged chunks through
             # the IDENTICAL chunk/ranged-kernel flow, so multi-rank CI on one
             # GPU covers everything but the RCCL transport itself
-            for lo, hi in bounds:
-                send_cpu = view[lo:hi].cpu()
-                tmp_cpu = torch.empty_like(send_cpu)
-                rq = dist.batch_isend_irecv([
-                    dist.P2POp(dist.isend, send_cpu, partner),
-                    dist.P2POp(dist.irecv, tmp_cpu, partner),
-                ])
-                for r in rq:
-                    r.wait()
-                tmp[lo : hi].copy_(tmp_cpu)
-                self.q.qft_column_top_range(
-                    float(scale), rs, in_place, pows, ws, float(scale * meta_w), pre,
-                    lo, hi, base + lo * elem, recv_is_low, ext)
+            self._pipe_gloo_chunks(bounds, view, tmp, partner, scale, rs, in_place, pows, ws,
+                                   meta_w, pre, base, elem, recv_is_low, ext)
         torch.cuda.synchronize(self.device_id)
+
+    def _pipe_gloo_chunks(self, bounds, view, tmp, partner, scale, rs, in_place, pows, ws,
+                          meta_w, pre, base, elem, recv_is_low, ext):
+        for lo, hi in bounds:
+            send_cpu = view[lo:hi].cpu()
+            tmp_cpu = torch.empty_like(send_cpu)
+            rq = dist.batch_isend_irecv([
+                dist.P2POp(dist.isend, send_cpu, partner),
+                dist.P2POp(dist.irecv, tmp_cpu, partner),
+            ])
+            for r in rq:
+                r.wait()
+            tmp[lo:hi].copy_(tmp_cpu)
+            self.q.qft_column_top_range(
+                float(scale), rs, in_place, pows, ws, float(scale * meta_w), pre,
+                lo, hi, base + lo * elem, recv_is_low, ext)
 
     def _fused_column(self, start, i, sign, pre):
         """One engine pass per column: realize the target locally (one
