@@ -1670,6 +1670,120 @@ __global__ void k_qft_col3_v(cplx<float>* sv, bitCapInt orbitPairs, bitCapInt tH
     }
 }
 
+// GENERIC K-column fused QFT kernel (2^K-amplitude orbits). The per-column
+// ramp factor for the slot with column-c's bit set is
+//   f0^(2^(K-1-c)) · Π_{j<c} U(c-j)^(b_j),   U(d) = e^{i·sign·π/2^d}
+// — one sincos (f0) plus K-1 constant roots of unity drive every ramp.
+template <typename R, int K, bool PRE>
+__global__ void k_qft_colK(cplx<R>* sv, bitCapInt orbits, const bitCapInt tPows0,
+    const bitCapInt tPows1, const bitCapInt tPows2, const bitCapInt tPows3, bitLenInt rampStart,
+    bitCapInt lowMask, R scaleHi)
+{
+    // tPows: [0]=lowest column bit ... [K-1]=highest
+    const bitCapInt tP[4] = { tPows0, tPows1, tPows2, tPows3 };
+    const R s = (R)0.70710678118654752440;
+    const R iSign = (scaleHi >= 0) ? (R)1 : (R)-1;
+    const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
+    // U(d) = e^{i·sign·π/2^d}, d = 1..K-1
+    cplx<R> U[4];
+    {
+        R ang = iSign * (R)1.57079632679489661923; // π/2
+        for (int d = 1; d < K; ++d) {
+            R sn, cs;
+            devSinCos<R>(ang, &sn, &cs);
+            U[d] = cplx<R>{ cs, sn };
+            ang = ang * (R)0.5;
+        }
+    }
+    const int NS = 1 << K;
+    for (bitCapInt k = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; k < orbits;
+         k += stride) {
+        bitCapInt r = k;
+        for (int b = 0; b < K; ++b) r = insertZeroBitDev(r, tP[b]);
+        bitCapInt idx[1 << K];
+        for (int b = 0; b < NS; ++b) {
+            bitCapInt x = r;
+            for (int c = 0; c < K; ++c) {
+                if (b & (1 << c)) x |= tP[c];
+            }
+            idx[b] = x;
+        }
+        cplx<R> v[1 << K];
+        for (int b = 0; b < NS; ++b) v[b] = sv[idx[b]];
+        const uint64_t lf = (uint64_t)((r >> rampStart) & lowMask);
+        R sn, cs;
+        devSinCos<R>(scaleHi * (R)lf, &sn, &cs);
+        cplx<R> fPow[4]; // f0^(2^m): fPow[0]=f0, fPow[m]=fPow[m-1]^2
+        fPow[0] = cplx<R>{ cs, sn };
+        for (int m = 1; m < K; ++m) fPow[m] = fPow[m - 1] * fPow[m - 1];
+        // slot bit c corresponds to column index (from hi): c=K-1 is hi
+        auto column = [&](int c) {
+            // H over axis bit c
+            for (int b = 0; b < NS; ++b) {
+                if (b & (1 << c)) continue;
+                const int hb = b | (1 << c);
+                const cplx<R> t = s * (v[b] + v[hb]);
+                const cplx<R> u = s * (v[b] - v[hb]);
+                v[b] = t;
+                v[hb] = u;
+            }
+            // ramp on slots with bit c set
+            for (int b = 0; b < NS; ++b) {
+                if (!(b & (1 << c))) continue;
+                cplx<R> f = fPow[K - 1 - c];
+                for (int j = 0; j < c; ++j) {
+                    if (b & (1 << j)) f = f * U[c - j];
+                }
+                v[b] = f * v[b];
+            }
+        };
+        auto columnInv = [&](int c) {
+            for (int b = 0; b < NS; ++b) {
+                if (!(b & (1 << c))) continue;
+                cplx<R> f = fPow[K - 1 - c];
+                for (int j = 0; j < c; ++j) {
+                    if (b & (1 << j)) f = f * U[c - j];
+                }
+                v[b] = f * v[b];
+            }
+            for (int b = 0; b < NS; ++b) {
+                if (b & (1 << c)) continue;
+                const int hb = b | (1 << c);
+                const cplx<R> t = s * (v[b] + v[hb]);
+                const cplx<R> u = s * (v[b] - v[hb]);
+                v[b] = t;
+                v[hb] = u;
+            }
+        };
+        if (!PRE) {
+            for (int c = K - 1; c >= 0; --c) column(c);
+        } else {
+            for (int c = 0; c < K; ++c) columnInv(c);
+        }
+        for (int b = 0; b < NS; ++b) sv[idx[b]] = v[b];
+    }
+}
+
+template <typename R>
+void launchQftColumnK(cplx<R>* sv, bitCapInt maxQPower, bitLenInt rampStart, bitLenInt col,
+    int kCols, const bitCapInt* tPows, int sign, bool pre, hipStream_t stream)
+{
+    // tPows[0..kCols-1] ascending (lowest column first); col = highest column
+    const bitCapInt lowMask = (ONE_BCI << (col - (kCols - 1))) - 1u;
+    const R scaleHi = (R)sign * (R)3.14159265358979323846 / (R)(ONE_BCI << col);
+    const bitCapInt orbits = maxQPower >> kCols;
+    if (kCols != 4) return; // only K=4 instantiated; 2/3 have tuned kernels
+    if (pre) {
+        hipLaunchKernelGGL((k_qft_colK<R, 4, true>), dim3(gridFor(orbits)), dim3(QA_BLOCK), 0,
+            stream, sv, orbits, tPows[0], tPows[1], tPows[2], tPows[3], rampStart, lowMask,
+            scaleHi);
+    } else {
+        hipLaunchKernelGGL((k_qft_colK<R, 4, false>), dim3(gridFor(orbits)), dim3(QA_BLOCK), 0,
+            stream, sv, orbits, tPows[0], tPows[1], tPows[2], tPows[3], rampStart, lowMask,
+            scaleHi);
+    }
+}
+
 template <typename R>
 void launchQftColumn3(cplx<R>* sv, bitCapInt maxQPower, bitLenInt rampStart, bitLenInt col,
     bitCapInt tHi, bitCapInt tMid, bitCapInt tLo, int sign, bool pre, hipStream_t stream)
@@ -2349,6 +2463,8 @@ void launchPartProbs(const cplx<R>* sv, bitCapInt maxQPower, bitLenInt start, bi
         bitCapInt, int, bool, hipStream_t);                                                         \
     template void launchQftColumn3<R>(cplx<R>*, bitCapInt, bitLenInt, bitLenInt, bitCapInt,        \
         bitCapInt, bitCapInt, int, bool, hipStream_t);                                              \
+    template void launchQftColumnK<R>(cplx<R>*, bitCapInt, bitLenInt, bitLenInt, int,               \
+        const bitCapInt*, int, bool, hipStream_t);                                                  \
     template void launchQftColumn<R>(                                                               \
         cplx<R>*, bitCapInt, bitLenInt, bitLenInt, bitCapInt, int, bool, hipStream_t);              \
     template void launchMtrx1qBatch<R>(cplx<R>*, const Batch1qArgs<R>&, hipStream_t);                               \

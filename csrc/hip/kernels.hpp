@@ -189,6 +189,11 @@ template <typename R>
 void launchQftColumn2(cplx<R>* sv, bitCapInt maxQPower, bitLenInt rampStart, bitLenInt col,
     bitCapInt tHi, bitCapInt tLo, int sign, bool pre, hipStream_t stream);
 
+// generic K-column fused QFT pass (2^K-amplitude orbits; K=4 instantiated)
+template <typename R>
+void launchQftColumnK(cplx<R>* sv, bitCapInt maxQPower, bitLenInt rampStart, bitLenInt col,
+    int kCols, const bitCapInt* tPows, int sign, bool pre, hipStream_t stream);
+
 // THREE fused QFT columns per pass (8-amplitude orbits; see k_qft_col3)
 template <typename R>
 void launchQftColumn3(cplx<R>* sv, bitCapInt maxQPower, bitLenInt rampStart, bitLenInt col,

@@ -821,6 +821,14 @@ template <typename R> void QEngineHIP<R>::QFT(bitLenInt start, bitLenInt length,
     bitLenInt i = length;
     while (i > 0) {
         const bitLenInt col = i - 1u;
+        if (fuseMax >= 4 && col >= 3u && maxQPower >= 32u) {
+            HipProfScope prof("qft_column4", stream);
+            const bitCapInt tPows[4] = { pow2(start + col - 3u), pow2(start + col - 2u),
+                pow2(start + col - 1u), pow2(start + col) };
+            launchQftColumnK<R>(dState, maxQPower, start, col, 4, tPows, +1, false, stream);
+            i -= 4u;
+            continue;
+        }
         if (fuseMax >= 3 && col >= 2u && maxQPower >= 16u) {
             // three columns per pass (8-amplitude orbits)
             HipProfScope prof("qft_column3", stream);
@@ -860,6 +868,15 @@ template <typename R> void QEngineHIP<R>::IQFT(bitLenInt start, bitLenInt length
     }();
     bitLenInt i = 0;
     while (i < length) {
+        if (fuseMax >= 4 && (i + 3u) < length && maxQPower >= 32u) {
+            HipProfScope prof("qft_column4", stream);
+            const bitCapInt tPows[4] = { pow2(start + i), pow2(start + i + 1u),
+                pow2(start + i + 2u), pow2(start + i + 3u) };
+            launchQftColumnK<R>(
+                dState, maxQPower, start, (bitLenInt)(i + 3u), 4, tPows, -1, true, stream);
+            i += 4u;
+            continue;
+        }
         if (fuseMax >= 3 && (i + 2u) < length && maxQPower >= 16u) {
             // triple (lo=i, mid=i+1, hi=i+2): exact adjoint of the forward
             // triple (a lo column of 0 degenerates to the plain H inside)
