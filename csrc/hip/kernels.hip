@@ -1676,16 +1676,16 @@ __global__ void k_qft_col3_v(cplx<float>* sv, bitCapInt orbitPairs, bitCapInt tH
 // — one sincos (f0) plus K-1 constant roots of unity drive every ramp.
 template <typename R, int K, bool PRE>
 __global__ void k_qft_colK(cplx<R>* sv, bitCapInt orbits, const bitCapInt tPows0,
-    const bitCapInt tPows1, const bitCapInt tPows2, const bitCapInt tPows3, bitLenInt rampStart,
-    bitCapInt lowMask, R scaleHi)
+    const bitCapInt tPows1, const bitCapInt tPows2, const bitCapInt tPows3, const bitCapInt tPows4,
+    bitLenInt rampStart, bitCapInt lowMask, R scaleHi)
 {
     // tPows: [0]=lowest column bit ... [K-1]=highest
-    const bitCapInt tP[4] = { tPows0, tPows1, tPows2, tPows3 };
+    const bitCapInt tP[5] = { tPows0, tPows1, tPows2, tPows3, tPows4 };
     const R s = (R)0.70710678118654752440;
     const R iSign = (scaleHi >= 0) ? (R)1 : (R)-1;
     const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
     // U(d) = e^{i·sign·π/2^d}, d = 1..K-1
-    cplx<R> U[4];
+    cplx<R> U[5];
     {
         R ang = iSign * (R)1.57079632679489661923; // π/2
         for (int d = 1; d < K; ++d) {
@@ -1713,7 +1713,7 @@ __global__ void k_qft_colK(cplx<R>* sv, bitCapInt orbits, const bitCapInt tPows0
         const uint64_t lf = (uint64_t)((r >> rampStart) & lowMask);
         R sn, cs;
         devSinCos<R>(scaleHi * (R)lf, &sn, &cs);
-        cplx<R> fPow[4]; // f0^(2^m): fPow[0]=f0, fPow[m]=fPow[m-1]^2
+        cplx<R> fPow[5]; // f0^(2^m): fPow[0]=f0, fPow[m]=fPow[m-1]^2
         fPow[0] = cplx<R>{ cs, sn };
         for (int m = 1; m < K; ++m) fPow[m] = fPow[m - 1] * fPow[m - 1];
         // slot bit c corresponds to column index (from hi): c=K-1 is hi
@@ -1772,14 +1772,26 @@ void launchQftColumnK(cplx<R>* sv, bitCapInt maxQPower, bitLenInt rampStart, bit
     const bitCapInt lowMask = (ONE_BCI << (col - (kCols - 1))) - 1u;
     const R scaleHi = (R)sign * (R)3.14159265358979323846 / (R)(ONE_BCI << col);
     const bitCapInt orbits = maxQPower >> kCols;
-    if (kCols != 4) return; // only K=4 instantiated; 2/3 have tuned kernels
+    if (kCols == 5) {
+        if (pre) {
+            hipLaunchKernelGGL((k_qft_colK<R, 5, true>), dim3(gridFor(orbits)), dim3(QA_BLOCK),
+                0, stream, sv, orbits, tPows[0], tPows[1], tPows[2], tPows[3], tPows[4],
+                rampStart, lowMask, scaleHi);
+        } else {
+            hipLaunchKernelGGL((k_qft_colK<R, 5, false>), dim3(gridFor(orbits)), dim3(QA_BLOCK),
+                0, stream, sv, orbits, tPows[0], tPows[1], tPows[2], tPows[3], tPows[4],
+                rampStart, lowMask, scaleHi);
+        }
+        return;
+    }
+    if (kCols != 4) return; // 2/3 have tuned float4 kernels
     if (pre) {
         hipLaunchKernelGGL((k_qft_colK<R, 4, true>), dim3(gridFor(orbits)), dim3(QA_BLOCK), 0,
-            stream, sv, orbits, tPows[0], tPows[1], tPows[2], tPows[3], rampStart, lowMask,
+            stream, sv, orbits, tPows[0], tPows[1], tPows[2], tPows[3], 0u, rampStart, lowMask,
             scaleHi);
     } else {
         hipLaunchKernelGGL((k_qft_colK<R, 4, false>), dim3(gridFor(orbits)), dim3(QA_BLOCK), 0,
-            stream, sv, orbits, tPows[0], tPows[1], tPows[2], tPows[3], rampStart, lowMask,
+            stream, sv, orbits, tPows[0], tPows[1], tPows[2], tPows[3], 0u, rampStart, lowMask,
             scaleHi);
     }
 }

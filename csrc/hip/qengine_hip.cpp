@@ -816,15 +816,23 @@ template <typename R> void QEngineHIP<R>::QFT(bitLenInt start, bitLenInt length,
             return std::atoi(env) != 0 ? 3 : 1;
         }
         if (const char* env = std::getenv("QRACK_GPU_QFT_FUSE")) return std::atoi(env);
-        return 3;
+        return 4;
     }();
     bitLenInt i = length;
     while (i > 0) {
         const bitLenInt col = i - 1u;
+        if (fuseMax >= 5 && col >= 4u && maxQPower >= 64u) {
+            HipProfScope prof("qft_column5", stream);
+            const bitCapInt tPows[5] = { pow2(start + col - 4u), pow2(start + col - 3u),
+                pow2(start + col - 2u), pow2(start + col - 1u), pow2(start + col) };
+            launchQftColumnK<R>(dState, maxQPower, start, col, 5, tPows, +1, false, stream);
+            i -= 5u;
+            continue;
+        }
         if (fuseMax >= 4 && col >= 3u && maxQPower >= 32u) {
             HipProfScope prof("qft_column4", stream);
-            const bitCapInt tPows[4] = { pow2(start + col - 3u), pow2(start + col - 2u),
-                pow2(start + col - 1u), pow2(start + col) };
+            const bitCapInt tPows[5] = { pow2(start + col - 3u), pow2(start + col - 2u),
+                pow2(start + col - 1u), pow2(start + col), 0u };
             launchQftColumnK<R>(dState, maxQPower, start, col, 4, tPows, +1, false, stream);
             i -= 4u;
             continue;
@@ -864,14 +872,23 @@ template <typename R> void QEngineHIP<R>::IQFT(bitLenInt start, bitLenInt length
             return std::atoi(env) != 0 ? 3 : 1;
         }
         if (const char* env = std::getenv("QRACK_GPU_QFT_FUSE")) return std::atoi(env);
-        return 3;
+        return 4;
     }();
     bitLenInt i = 0;
     while (i < length) {
+        if (fuseMax >= 5 && (i + 4u) < length && maxQPower >= 64u) {
+            HipProfScope prof("qft_column5", stream);
+            const bitCapInt tPows[5] = { pow2(start + i), pow2(start + i + 1u),
+                pow2(start + i + 2u), pow2(start + i + 3u), pow2(start + i + 4u) };
+            launchQftColumnK<R>(
+                dState, maxQPower, start, (bitLenInt)(i + 4u), 5, tPows, -1, true, stream);
+            i += 5u;
+            continue;
+        }
         if (fuseMax >= 4 && (i + 3u) < length && maxQPower >= 32u) {
             HipProfScope prof("qft_column4", stream);
-            const bitCapInt tPows[4] = { pow2(start + i), pow2(start + i + 1u),
-                pow2(start + i + 2u), pow2(start + i + 3u) };
+            const bitCapInt tPows[5] = { pow2(start + i), pow2(start + i + 1u),
+                pow2(start + i + 2u), pow2(start + i + 3u), 0u };
             launchQftColumnK<R>(
                 dState, maxQPower, start, (bitLenInt)(i + 3u), 4, tPows, -1, true, stream);
             i += 4u;
