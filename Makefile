@@ -43,7 +43,7 @@ asan:
 	g++ -O1 -g -std=c++17 -fsanitize=address,undefined -fno-omit-frame-pointer \
 	    -Icsrc csrc/qinterface.cpp csrc/qengine_cpu.cpp csrc/qengine_sparse.cpp \
 	    csrc/qstabilizer.cpp csrc/qstabilizerhybrid.cpp csrc/qunit.cpp csrc/qbdt.cpp \
-	    csrc/qpager.cpp csrc/qfactory.cpp csrc/common/parallel_for.cpp \
+	    csrc/qpager.cpp csrc/qfactory.cpp csrc/qengine_turboquant.cpp csrc/common/parallel_for.cpp \
 	    tools/asan_smoke.cpp -o build/asan_smoke -lpthread
 	./build/asan_smoke
 
@@ -54,6 +54,6 @@ tsan:
 	g++ -O1 -g -std=c++17 -fsanitize=thread -fno-omit-frame-pointer \
 	    -Icsrc csrc/qinterface.cpp csrc/qengine_cpu.cpp csrc/qengine_sparse.cpp \
 	    csrc/qstabilizer.cpp csrc/qstabilizerhybrid.cpp csrc/qunit.cpp csrc/qbdt.cpp \
-	    csrc/qpager.cpp csrc/qfactory.cpp csrc/common/parallel_for.cpp \
+	    csrc/qpager.cpp csrc/qfactory.cpp csrc/qengine_turboquant.cpp csrc/common/parallel_for.cpp \
 	    tools/asan_smoke.cpp -o build/tsan_smoke -lpthread
 	./build/tsan_smoke

@@ -121,3 +121,23 @@ def test_hybrid_stack_over_turboquant():
             q.ry(0.3, t)
             cp.ry(0.3, t)
     assert_states_close(q.get_state_vector(), cp.get_state_vector(), 2e-3)
+
+
+def test_carry_alu_vs_dense():
+    import numpy as np
+
+    for seed in (1, 2):
+        q = make(9, seed=seed)
+        cp = qa.create_simulator(9, engine="cpu", seed=seed)
+        rng = np.random.default_rng(seed)
+        for s in (q, cp):
+            s.set_permutation(0b0110)
+            s.h(0)
+        for _ in range(6):
+            a = int(rng.integers(1, 15))
+            q.incc(a, 0, 4, 8)
+            cp.incc(a, 0, 4, 8)
+            b = int(rng.integers(1, 15))
+            q.decc(b, 0, 4, 8)
+            cp.decc(b, 0, 4, 8)
+        assert_states_close(q.get_state_vector(), cp.get_state_vector(), 1e-3)

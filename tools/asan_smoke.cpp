@@ -45,6 +45,19 @@ static void battery(QInterfacePtr<float> q)
     q->CZ(1, 3);
     (void)q->GetUnitaryFidelity();
     q->SetSdrp(0.0);
+    // round-2 paths: T-gadget ancillae + buffered cross-unit CX + wide masks
+    q->SetPermutation(0);
+    q->H(0);
+    q->T(0);
+    q->CNOT(0, 1);
+    q->H(2);
+    q->CNOT(2, 3);
+    q->CNOT(2, 3);
+    (void)q->Prob(1);
+    BigCap wide = q->MAllWide();
+    (void)wide;
+    q->SetPermutationWide(BigCap(5, 0));
+    (void)q->MultiShotMeasureQubits({ 0, 3, 5 }, 8);
     (void)res;
 }
 
@@ -58,6 +71,8 @@ int main()
         { "qunit", "cpu" },
         { "qunit", "stabilizer_hybrid", "cpu" },
         { "pager", "cpu" },
+        { "turboquant" },
+        { "qunit", "stabilizer_hybrid", "turboquant" },
     };
     for (const auto& layers : stacks) {
         auto q = CreateStack<float>(6, layers, 0, 42, -1, 2);
