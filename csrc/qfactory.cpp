@@ -3,6 +3,7 @@
 #include "qstabilizer.hpp"
 #include "qstabilizerhybrid.hpp"
 #include "qhybrid.hpp"
+#include "qfuser.hpp"
 #include "qinterface_noisy.hpp"
 #include "qbdt.hpp"
 #include "qengine_sparse.hpp"
@@ -89,6 +90,15 @@ EngineFactoryFn<R> LayerFactory(std::vector<std::string> layers, RngPtr rng, int
         EngineFactoryFn<R> sub = LayerFactory<R>(tail, rng, deviceId, pagesPerDevice, devices);
         return [rng, sub](bitLenInt n, bitCapInt perm) -> QInterfacePtr<R> {
             return std::make_shared<QTensorNetwork<R>>(n, perm, rng, sub);
+        };
+    }
+    if (head == "fuser") {
+        // transparent gate-fusion decorator (reference QCircuit combining /
+        // MpsShard fusion as a standalone layer): pending-2x2 compose +
+        // disjoint 2q-layer batching into the engine batch entry points
+        EngineFactoryFn<R> sub = LayerFactory<R>(tail, rng, deviceId, pagesPerDevice, devices);
+        return [rng, sub](bitLenInt n, bitCapInt perm) -> QInterfacePtr<R> {
+            return std::make_shared<QFuser<R>>(n, sub(n, perm), rng);
         };
     }
     if (head == "noisy") {
