@@ -572,3 +572,20 @@ def test_qft_fused2_numerics_vs_cpu():
     q.qft(3, 6)
     cp.qft(3, 6)
     assert_states_close(q.get_state_vector(), cp.get_state_vector(), 2e-4)
+
+
+def test_fuser_over_hip_numerics():
+    """QFuser batching over the HIP engine matches the bare engine."""
+    n = 12
+    rng = np.random.default_rng(91)
+    f = qa.create_simulator(n, layers=["fuser", "hip"], seed=91)
+    b = make(n, seed=91)
+    for d in range(5):
+        for t in range(n):
+            u = rand_unitary_2x2(rng)
+            f.mtrx(list(u.flatten()), t)
+            b.mtrx(list(u.flatten()), t)
+        for a in range(d % 2, n - 1, 2):
+            f.cnot(a, a + 1)
+            b.cnot(a, a + 1)
+    assert_states_close(f.get_state_vector(), b.get_state_vector(), 2e-4)
