@@ -597,6 +597,18 @@ template <typename R> static void bindQInterface(py::module_& m, const char* nam
             py::arg("in_place_mask"), py::arg("pows"), py::arg("weights"),
             py::arg("phase0") = 0.0, py::arg("pre") = false)
 #ifdef QRACK_AMD_HIP_ENGINE
+        .def("qft_column2_general",
+            [](Ptr q, bitLenInt targetHi, bitLenInt targetLo, double scale, bitLenInt rampStart,
+                bitCapInt inPlaceRelMask, std::vector<bitCapInt> sPows,
+                std::vector<uint64_t> sWeights, double phase0Hi, double phase0Lo, bool pre) {
+                auto eng = std::dynamic_pointer_cast<QEngineHIP<R>>(q);
+                if (!eng) throw QrackError("qft_column2_general requires the HIP engine");
+                eng->QftColumn2General(targetHi, targetLo, scale, rampStart, inPlaceRelMask,
+                    sPows, sWeights, phase0Hi, phase0Lo, pre);
+            },
+            py::arg("target_hi"), py::arg("target_lo"), py::arg("scale"), py::arg("ramp_start"),
+            py::arg("in_place_mask"), py::arg("pows"), py::arg("weights"), py::arg("phase0_hi"),
+            py::arg("phase0_lo"), py::arg("pre") = false)
         .def("qft_column_top_range",
             [](Ptr q, double scale, bitLenInt rampStart, bitCapInt inPlaceRelMask,
                 std::vector<bitCapInt> sPows, std::vector<uint64_t> sWeights, double phase0,

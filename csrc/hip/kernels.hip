@@ -1246,6 +1246,79 @@ void launchQftColumnGeneral(cplx<R>* sv, bitCapInt maxQPower, bitCapInt tPow, co
     }
 }
 
+// TWO generalized QFT columns per pass for the distributed pager's local
+// ladder: targets at ARBITRARY local slots (tPowHi/tPowLo by ROLE, not
+// numeric order), ramp bits possibly relocated (RampArgs, which must
+// EXCLUDE the low column's target bit — its contribution is the constant
+// A = e^{i·sign·π/2}); per-column meta-page scalars phase0Hi/phase0Lo.
+// Same algebra as k_qft_col2: f_hi(bLo) = f0·A^bLo·e^{i·phase0Hi},
+// f_lo = f0²·e^{i·phase0Lo} with ONE sincos per orbit.
+template <typename R, bool PRE>
+__global__ void k_qft_col2_gen(cplx<R>* sv, bitCapInt orbits, bitCapInt tPowHi, bitCapInt tPowLo,
+    RampArgs a, R phase0Hi, R phase0Lo)
+{
+    const R s = (R)0.70710678118654752440;
+    const R iSign = (a.scale >= 0) ? (R)1 : (R)-1;
+    const bitCapInt pLow = (tPowHi < tPowLo) ? tPowHi : tPowLo;
+    const bitCapInt pHigh = (tPowHi < tPowLo) ? tPowLo : tPowHi;
+    const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
+    for (bitCapInt k = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; k < orbits;
+         k += stride) {
+        const bitCapInt r = insertZeroBitDev(insertZeroBitDev(k, pLow), pHigh);
+        uint64_t frac = (uint64_t)((r >> a.rampStart) & a.inPlaceRelMask);
+        for (int t = 0; t < a.nScattered; ++t) {
+            if (r & a.sPow[t]) frac += a.sWeight[t];
+        }
+        R sn, cs;
+        devSinCos<R>((R)a.scale * (R)frac, &sn, &cs);
+        const cplx<R> f0{ cs, sn };
+        const cplx<R> A{ 0, iSign };
+        R snh, csh, snl, csl;
+        devSinCos<R>(phase0Hi, &snh, &csh);
+        devSinCos<R>(phase0Lo, &snl, &csl);
+        const cplx<R> fHi0 = f0 * cplx<R>{ csh, snh };
+        const cplx<R> fHi1 = fHi0 * A;
+        const cplx<R> fLo = f0 * f0 * cplx<R>{ csl, snl };
+        cplx<R> a00 = sv[r];
+        cplx<R> a01 = sv[r | tPowLo];
+        cplx<R> a10 = sv[r | tPowHi];
+        cplx<R> a11 = sv[r | tPowHi | tPowLo];
+        if (!PRE) {
+            cplx<R> b00 = s * (a00 + a10), b01 = s * (a01 + a11);
+            cplx<R> b10 = fHi0 * (s * (a00 - a10)), b11 = fHi1 * (s * (a01 - a11));
+            sv[r] = s * (b00 + b01);
+            sv[r | tPowLo] = fLo * (s * (b00 - b01));
+            sv[r | tPowHi] = s * (b10 + b11);
+            sv[r | tPowHi | tPowLo] = fLo * (s * (b10 - b11));
+        } else {
+            a01 = fLo * a01;
+            a11 = fLo * a11;
+            cplx<R> b00 = s * (a00 + a01), b01 = s * (a00 - a01);
+            cplx<R> b10 = s * (a10 + a11), b11 = s * (a10 - a11);
+            b10 = fHi0 * b10;
+            b11 = fHi1 * b11;
+            sv[r] = s * (b00 + b10);
+            sv[r | tPowHi] = s * (b00 - b10);
+            sv[r | tPowLo] = s * (b01 + b11);
+            sv[r | tPowHi | tPowLo] = s * (b01 - b11);
+        }
+    }
+}
+
+template <typename R>
+void launchQftColumn2General(cplx<R>* sv, bitCapInt maxQPower, bitCapInt tPowHi, bitCapInt tPowLo,
+    const RampArgs& a, double phase0Hi, double phase0Lo, bool pre, hipStream_t stream)
+{
+    const bitCapInt orbits = maxQPower >> 2u;
+    if (pre) {
+        hipLaunchKernelGGL((k_qft_col2_gen<R, true>), dim3(gridFor(orbits)), dim3(QA_BLOCK), 0,
+            stream, sv, orbits, tPowHi, tPowLo, a, (R)phase0Hi, (R)phase0Lo);
+    } else {
+        hipLaunchKernelGGL((k_qft_col2_gen<R, false>), dim3(gridFor(orbits)), dim3(QA_BLOCK), 0,
+            stream, sv, orbits, tPowHi, tPowLo, a, (R)phase0Hi, (R)phase0Lo);
+    }
+}
+
 // Ranged top-target fused column with receive-buffer fusion (pipelined
 // distributed page exchange; see kernels.hpp). Pair rows r in [itLo, itHi):
 //   x (target=0 side) = recvIsLow ? recvSrc[r-itLo] : sv[r]
@@ -2485,6 +2558,8 @@ void launchPartProbs(const cplx<R>* sv, bitCapInt maxQPower, bitLenInt start, bi
     template void launchMtrx2qBatchLds<R>(cplx<R>*, const Batch2qLdsArgs<R>&, hipStream_t);                \
     template void launchMtrx2q<R>(cplx<R>*, const Gate4x4Args<R>&, hipStream_t);                              \
     template void launchCPhasePairs<R>(cplx<R>*, const CPhasePairsArgs&, hipStream_t);                               \
+    template void launchQftColumn2General<R>(cplx<R>*, bitCapInt, bitCapInt, bitCapInt,            \
+        const RampArgs&, double, double, bool, hipStream_t);                                        \
     template void launchQftColumnTopRange<R>(cplx<R>*, bitCapInt, const RampArgs&, double, bool,    \
         bitCapInt, bitCapInt, const cplx<R>*, bool, hipStream_t);                                   \
     template void launchQftColumnGeneral<R>(                                                        \

@@ -206,6 +206,13 @@ template <typename R>
 void launchQftColumnGeneral(cplx<R>* sv, bitCapInt maxQPower, bitCapInt tPow, const RampArgs& a,
     double phase0, bool pre, hipStream_t stream);
 
+// TWO generalized columns per pass (arbitrary local target slots,
+// relocated ramp bits, per-column meta scalars) — the distributed pager's
+// local-ladder pass-halving (see k_qft_col2_gen)
+template <typename R>
+void launchQftColumn2General(cplx<R>* sv, bitCapInt maxQPower, bitCapInt tPowHi, bitCapInt tPowLo,
+    const RampArgs& a, double phase0Hi, double phase0Lo, bool pre, hipStream_t stream);
+
 // Ranged top-target fused QFT column for the distributed pager's PIPELINED
 // page exchange: processes pair rows r in [itLo, itHi) of the H on the TOP
 // local qubit (tPow = maxQPower/2), reading ONE side of each pair straight

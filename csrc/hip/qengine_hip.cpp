@@ -759,6 +759,28 @@ void QEngineHIP<R>::QftColumnGeneral(bitLenInt target, double scale, bitLenInt r
 }
 
 template <typename R>
+void QEngineHIP<R>::QftColumn2General(bitLenInt targetHi, bitLenInt targetLo, double scale,
+    bitLenInt rampStart, bitCapInt inPlaceRelMask, const std::vector<bitCapInt>& sPows,
+    const std::vector<uint64_t>& sWeights, double phase0Hi, double phase0Lo, bool pre)
+{
+    if (sPows.size() > 8u) throw QrackError("QftColumn2General: more than 8 relocated bits");
+    RampArgs a{};
+    a.rampStart = rampStart;
+    a.inPlaceRelMask = inPlaceRelMask;
+    a.nScattered = (int)sPows.size();
+    for (size_t k = 0; k < sPows.size(); ++k) {
+        a.sPow[k] = sPows[k];
+        a.sWeight[k] = sWeights[k];
+    }
+    a.condPow = 0;
+    a.scale = scale;
+    QA_HIP_CHECK(hipSetDevice(deviceId));
+    HipProfScope prof("qft_column2_gen", stream);
+    launchQftColumn2General<R>(
+        dState, maxQPower, pow2(targetHi), pow2(targetLo), a, phase0Hi, phase0Lo, pre, stream);
+}
+
+template <typename R>
 void QEngineHIP<R>::QftColumnTopRange(double scale, bitLenInt rampStart, bitCapInt inPlaceRelMask,
     const std::vector<bitCapInt>& sPows, const std::vector<uint64_t>& sWeights, double phase0,
     bool pre, uint64_t itLo, uint64_t itHi, uintptr_t recvPtr, bool recvIsLow, uintptr_t extStream)

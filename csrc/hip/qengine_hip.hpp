@@ -154,6 +154,13 @@ public:
     void QftColumnGeneral(bitLenInt target, double scale, bitLenInt rampStart,
         bitCapInt inPlaceRelMask, const std::vector<bitCapInt>& sPows,
         const std::vector<uint64_t>& sWeights, double phase0, bool pre) override;
+    // TWO generalized columns in one pass (distributed pager local ladder):
+    // targetHi's column (scale, ramp EXCLUDING targetLo's bit) then
+    // targetLo's column (2*scale on the same ramp bits); per-column
+    // meta-page scalars phase0Hi/phase0Lo
+    void QftColumn2General(bitLenInt targetHi, bitLenInt targetLo, double scale,
+        bitLenInt rampStart, bitCapInt inPlaceRelMask, const std::vector<bitCapInt>& sPows,
+        const std::vector<uint64_t>& sWeights, double phase0Hi, double phase0Lo, bool pre);
     // Pipelined-exchange building block (distributed pager): fused top-target
     // column over pair rows [itLo, itHi), one pair side read straight from
     // the RCCL receive buffer `recvPtr` (device pointer, chunk-local), and the

@@ -596,13 +596,46 @@ class DistQPager:
         self.q.qft_column_general(
             t_slot, float(scale), rs, in_place, pows, ws, float(scale * meta_w), pre)
 
+    def _try_pair_local(self, start, hi_col, sign, pre):
+        """Apply columns (hi_col, hi_col-1) in ONE engine pass when both
+        targets are local (k_qft_col2_gen: same lf drives both ramps, the
+        cross term is the constant ±i) — halves the local ladder's
+        full-state passes, matching the single-GPU engine's multi-column
+        fusion so weak scaling isn't penalized for using the pager."""
+        lo_col = hi_col - 1
+        s_hi = self.slot_of[start + hi_col]
+        s_lo = self.slot_of[start + lo_col]
+        if s_hi >= self.qpp or s_lo >= self.qpp:
+            return False
+        if not self._is_hip() or not hasattr(self.q, "qft_column2_general"):
+            return False
+        rs, in_place, scattered, meta_w = self._ramp_parts(start, lo_col)
+        if len(scattered) > 8:
+            return False
+        scale_hi = sign * np.pi / (1 << hi_col)
+        self.q.qft_column2_general(
+            s_hi, s_lo, float(scale_hi), rs, in_place,
+            [p for p, _ in scattered], [w for _, w in scattered],
+            float(scale_hi * meta_w), float(2.0 * scale_hi * meta_w), pre)
+        return True
+
     def qft(self, start, length):
-        for i in range(length - 1, -1, -1):
+        i = length - 1
+        while i >= 0:
+            if i >= 1 and self._try_pair_local(start, i, +1, False):
+                i -= 2
+                continue
             self._fused_column(start, i, +1, False)
+            i -= 1
 
     def iqft(self, start, length):
-        for i in range(length):
+        i = 0
+        while i < length:
+            if i + 1 < length and self._try_pair_local(start, i + 1, -1, True):
+                i += 2
+                continue
             self._fused_column(start, i, -1, True)
+            i += 1
 
     # ---- measurement ------------------------------------------------------------
 

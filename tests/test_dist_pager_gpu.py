@@ -280,3 +280,67 @@ def test_dlpack_view_roundtrip():
         timeout=240,
     )
     assert "DLPACK_OK" in out.stdout, f"stdout={out.stdout}\nstderr={out.stderr}"
+
+
+_COL2_GEN_BODY = """
+import torch
+torch.cuda.init()
+import sys
+sys.path.insert(0, {repo!r})
+import numpy as np
+import qrack_amd as qa
+
+n = 12
+N = 1 << n
+rng = np.random.default_rng(13)
+v = rng.normal(size=N) + 1j * rng.normal(size=N)
+sv = (v / np.linalg.norm(v)).astype(np.complex64)
+
+# two generalized columns (hi=9@slot 4, lo=8@slot 7 — scrambled slots) vs
+# the single-column general kernel applied twice
+scale_hi = np.pi / (1 << 9)
+rs = 0
+in_place = 0b1011          # some in-place ramp bits
+pows = [1 << 6, 1 << 10]   # scattered ramp bits
+ws = [16, 64]
+p0h, p0l = 0.21, 0.42
+
+for pre in (False, True):
+    q1 = qa.create_simulator(n, engine="hip", seed=1)
+    q1.set_amplitude_page(sv, 0)
+    q2 = qa.create_simulator(n, engine="hip", seed=1)
+    q2.set_amplitude_page(sv, 0)
+    # reference: column hi then column lo via the 1-col kernel.
+    # hi ramp = same bits PLUS the lo target's bit (slot 7, weight 1<<8
+    # at scale_hi -> pi/2 = the A factor).
+    hi_pows = pows + [1 << 7]
+    hi_ws = ws + [1 << 8]
+    if not pre:
+        q2.qft_column_general(4, scale_hi, rs, in_place, hi_pows, hi_ws, p0h, False)
+        q2.qft_column_general(7, 2 * scale_hi, rs, in_place, pows, ws, p0l, False)
+    else:
+        q2.qft_column_general(7, 2 * scale_hi, rs, in_place, pows, ws, p0l, True)
+        q2.qft_column_general(4, scale_hi, rs, in_place, hi_pows, hi_ws, p0h, True)
+    q1.qft_column2_general(4, 7, scale_hi, rs, in_place, pows, ws, p0h, p0l, pre)
+    a = np.asarray(q1.get_state_vector())
+    b = np.asarray(q2.get_state_vector())
+    err = np.abs(a - b).max()
+    assert err < 2e-6, (pre, err)
+print("COL2_GEN_OK")
+"""
+
+
+def test_qft_col2_general_kernel():
+    """The generalized 2-column kernel must equal two generalized 1-column
+    passes (scrambled target slots, relocated ramp bits, meta scalars)."""
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, "-c", _COL2_GEN_BODY.format(repo=repo)],
+        capture_output=True,
+        text=True,
+        timeout=300,
+    )
+    assert "COL2_GEN_OK" in out.stdout, f"stdout={out.stdout}\nstderr={out.stderr}"
