@@ -22,6 +22,9 @@ STACKS = [
     ["qunit", "stabilizer_hybrid", "cpu"],
     ["qunit_multi", "stabilizer_hybrid", "cpu"],
     ["pager", "cpu"],
+    ["turboquant"],
+    ["fuser", "cpu"],
+    ["fuser", "qunit", "stabilizer_hybrid", "cpu"],
     ["tensor_network", "cpu"],
     ["noisy", "cpu"],  # QRACK_GATE_DEPOLARIZATION defaults irrelevant: set 0
 ]
