@@ -158,6 +158,8 @@ public:
     using QInterface<R>::Compose;
     bitLenInt Compose(QInterfacePtr<R> o, bitLenInt s) override
     {
+        FlushAll();
+        if (auto* of = dynamic_cast<QFuser<R>*>(o.get())) of->FlushAll();
         QInterfaceWrapper<R>* w = dynamic_cast<QInterfaceWrapper<R>*>(o.get());
         const bitLenInt r = inner->Compose(w ? w->Inner() : o, s);
         this->SetQubitCount(inner->GetQubitCount());
@@ -165,6 +167,7 @@ public:
     }
     void Decompose(bitLenInt s, QInterfacePtr<R> d) override
     {
+        FlushAll();
         QInterfaceWrapper<R>* w = dynamic_cast<QInterfaceWrapper<R>*>(d.get());
         inner->Decompose(s, w ? w->Inner() : d);
         if (w) w->SetQubitCountFromInner();
@@ -278,6 +281,8 @@ public:
     }
     double SumSqrDiff(QInterfacePtr<R> o) override
     {
+        FlushAll();
+        if (auto* of = dynamic_cast<QFuser<R>*>(o.get())) of->FlushAll();
         QInterfaceWrapper<R>* w = dynamic_cast<QInterfaceWrapper<R>*>(o.get());
         return inner->SumSqrDiff(w ? w->Inner() : o);
     }
