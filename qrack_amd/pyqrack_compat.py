@@ -228,7 +228,8 @@ class QrackSimulator:
         return int(self.sim.m_all())
 
     def measure_shots(self, qubits, shots):
-        res = self.sim.multi_shot_measure_mask([1 << q for q in qubits], shots)
+        # qubit-index addressing keeps this exact past 64 logical qubits
+        res = self.sim.multi_shot_measure_qubits(list(qubits), shots)
         out = []
         for val, cnt in res.items():
             out.extend([int(val)] * cnt)
