@@ -283,3 +283,50 @@ def test_qft_after_swaps_world4():
 
 def test_iqft_roundtrip_swapped_world2():
     run_distributed("_body_iqft_roundtrip_swapped", world=2, qubits=5, seed=12, port_off=11)
+
+
+def _body_deep_equivalence(pager, rank):
+    """Deeper/wider equivalence sweep: many meta-ops,
+    swaps and mid-circuit probability probes at 10 qubits, world up to 8."""
+    import qrack_amd as qa
+
+    n = pager.num_qubits
+    ref = qa.create_simulator(n, engine="cpu", seed=99)
+    rng = np.random.default_rng(29)
+    pager.set_permutation(0)
+    for layer in range(6):
+        for i in range(n):
+            th = float(rng.uniform(0, 2 * np.pi))
+            s, c = np.sin(th / 2), np.cos(th / 2)
+            pager.mtrx([c, -s, s, c], i)
+            ref.ry(th, i)
+        # meta-heavy two-qubit ops: always touch the top qubits
+        for hi in range(n - 1, n - 4, -1):
+            lo = int(rng.integers(n - 4))
+            pager.cnot(lo, hi)
+            ref.cnot(lo, hi)
+        a, b = rng.choice(n, 2, replace=False)
+        pager.swap(int(a), int(b))
+        ref.swap(int(a), int(b))
+        t = int(rng.integers(n))
+        pager.rz(0.37, t)
+        ref.rz(0.37, t)
+        # probability probe mid-circuit on a (possibly meta) qubit
+        q = int(rng.integers(n))
+        assert abs(pager.prob(q) - ref.prob(q)) < 1e-5
+    pager.qft(0, n)
+    ref.qft(0, n)
+    pager.iqft(0, n)
+    ref.iqft(0, n)
+    sv = pager.get_state_vector().astype(np.complex128)
+    rv = np.asarray(ref.get_state_vector()).astype(np.complex128)
+    inner = abs(np.vdot(rv, sv))
+    assert inner > 1 - 1e-4, f"fidelity {inner}"
+
+
+def test_deep_equivalence_world4():
+    run_distributed("_body_deep_equivalence", world=4, qubits=10, port_off=12)
+
+
+def test_deep_equivalence_world8():
+    run_distributed("_body_deep_equivalence", world=8, qubits=10, port_off=13)
