@@ -102,7 +102,28 @@ public:
     void IISwap(bitLenInt a, bitLenInt b) override { FlushAll(); inner->IISwap(a, b); }
     void SqrtSwap(bitLenInt a, bitLenInt b) override { FlushAll(); inner->SqrtSwap(a, b); }
     void ISqrtSwap(bitLenInt a, bitLenInt b) override { FlushAll(); inner->ISqrtSwap(a, b); }
-    void FSim(R th, R ph, bitLenInt a, bitLenInt b) override { FlushAll(); inner->FSim(th, ph, a, b); }
+    void FSim(R th, R ph, bitLenInt a, bitLenInt b) override
+    {
+        if (a == b) throw QrackError("FSim: identical qubits");
+        // fSim(theta, phi) as a symmetric 4x4 in |q2 q1> basis: |01>,|10>
+        // mix by [[c,-is],[-is,c]], |11> gains e^{-i phi} (Google fSim
+        // convention, matching QInterface::FSim) — joins the pending
+        // disjoint 2q layer with both endpoints' pending 1q absorbed
+        const R c = std::cos(th), sn = std::sin(th);
+        cplx<R> m16[16] = {};
+        m16[0] = cplx<R>(1, 0);
+        m16[1 * 4 + 1] = cplx<R>(c, 0);
+        m16[1 * 4 + 2] = cplx<R>(0, -sn);
+        m16[2 * 4 + 1] = cplx<R>(0, -sn);
+        m16[2 * 4 + 2] = cplx<R>(c, 0);
+        m16[3 * 4 + 3] = polar<R>(1, -ph);
+        Queue2q(m16, a, b);
+    }
+    void FSimBatch(const std::vector<R>& thetas, const std::vector<R>& phis,
+        const std::vector<bitLenInt>& q1s, const std::vector<bitLenInt>& q2s) override
+    {
+        for (size_t i = 0; i < thetas.size(); ++i) FSim(thetas[i], phis[i], q1s[i], q2s[i]);
+    }
     
     void CSwap(const std::vector<bitLenInt>& c, bitLenInt a, bitLenInt b) override
     {

@@ -92,3 +92,20 @@ def test_mtrx2q_and_swap_through_fuser():
         s.swap(0, 4)
         s.cnot(4, 2)
     assert_states_close(f.get_state_vector(), c.get_state_vector(), 1e-5)
+
+
+def test_fsim_through_fuser():
+    import numpy as np
+
+    n = 6
+    rng = np.random.default_rng(31)
+    f, c = pair(n, 31)
+    for layer in range(4):
+        for t in range(n):
+            th = float(rng.uniform(0, 2 * np.pi))
+            f.ry(th, t); c.ry(th, t)
+        for a in range(layer % 2, n - 1, 2):
+            th, ph = rng.uniform(0, 2 * np.pi, 2)
+            f.fsim(float(th), float(ph), a, a + 1)
+            c.fsim(float(th), float(ph), a, a + 1)
+    assert_states_close(f.get_state_vector(), c.get_state_vector(), 1e-5)
