@@ -532,6 +532,7 @@ template <typename R> static void bindQInterface(py::module_& m, const char* nam
         .def("set_ncrp", &QI::SetNcrp)
         .def("get_ncrp", &QI::GetNcrp)
         .def("set_reactive_separate", &QI::SetReactiveSeparate)
+        .def("set_stochastic", &QI::SetStochastic)
         .def("get_reactive_separate", &QI::GetReactiveSeparate)
         .def("set_noise_parameter", &QI::SetNoiseParameter)
         .def("get_noise_parameter", &QI::GetNoiseParameter)

@@ -27,6 +27,9 @@ QStabilizerHybrid<R>::QStabilizerHybrid(bitLenInt qBitCount, bitCapInt initState
     if (const char* env = std::getenv("QRACK_MAX_ANCILLA")) {
         maxAncilla = (bitLenInt)std::atoi(env);
     }
+    if (const char* env = std::getenv("QRACK_USE_APPROX_NEAR_CLIFFORD")) {
+        stochasticNC = std::atoi(env) != 0;
+    }
     if (!engineFactory) {
         RngPtr rng = this->rand_generator;
         engineFactory = [rng](bitLenInt n, bitCapInt perm) -> QInterfacePtr<R> {
@@ -105,6 +108,38 @@ template <typename R> bool QStabilizerHybrid<R>::TryShardRoundClifford(bitLenInt
     return true;
 }
 
+// Stochastic near-Clifford rounding (reference RZRaw's coin flip,
+// qstabilizer.cpp:1820-1871): snap the phase shard to the floor or ceil
+// Clifford quarter-turn with probability given by the fractional part —
+// unbiased over shots, zero ancilla cost.
+template <typename R> bool QStabilizerHybrid<R>::TryShardStochastic(bitLenInt q)
+{
+    if (!shards[q] || !stabilizer || !stochasticNC) return false;
+    if (!ShardIsPhase(q)) return false;
+    const auto& m = *shards[q];
+    if (norm(m[0]) <= (R)1e-24) return false;
+    const cplx<R> ratio = m[3] * conj(m[0]);
+    const double theta = std::atan2((double)ratio.im, (double)ratio.re);
+    const double half_pi = 1.5707963267948966;
+    const double kf = std::floor(theta / half_pi);
+    const double frac = theta / half_pi - kf; // in [0,1)
+    const double k = (this->Rand() < frac) ? (kf + 1.0) : kf;
+    const double delta = theta - k * half_pi; // residual being dropped
+    const double p1 = (double)stabilizer->Prob(q);
+    const double fid = 1.0 - 2.0 * p1 * (1.0 - p1) * (1.0 - std::cos(delta));
+    logFidelity += std::log(std::max(fid, 1e-300));
+    const int ki = ((int)k % 4 + 4) % 4;
+    static const cplx<R> IPOW[4] = { { 1, 0 }, { 0, 1 }, { -1, 0 }, { 0, -1 } };
+    const cplx<R> snapped[4] = { m[0], cplx<R>(0, 0), cplx<R>(0, 0), m[0] * IPOW[ki] };
+    try {
+        stabilizer->Mtrx(snapped, q);
+    } catch (const QrackError&) {
+        return false;
+    }
+    shards[q].reset();
+    return true;
+}
+
 // Reverse T-injection (see header): absorb a blocked non-Clifford phase (or
 // invert = X·phase) shard into the tableau exactly via a gadget ancilla.
 template <typename R> bool QStabilizerHybrid<R>::TryShardGadget(bitLenInt q)
@@ -167,7 +202,8 @@ template <typename R> void QStabilizerHybrid<R>::FlushShard(bitLenInt q)
         shards[q].reset();
         return;
     }
-    if (!TryShardFlushClifford(q) && !TryShardRoundClifford(q) && !TryShardGadget(q)) {
+    if (!TryShardFlushClifford(q) && !TryShardRoundClifford(q) && !TryShardStochastic(q) &&
+        !TryShardGadget(q)) {
         SwitchToEngine();
         if (shards[q]) {
             engine->Mtrx(shards[q]->data(), q);

@@ -46,6 +46,12 @@ protected:
     bitLenInt ancillaCount = 0;
     bitLenInt maxAncilla = 16;
     bool useTGadget = true;
+    // stochastic near-Clifford rounding (reference isStochastic /
+    // QRACK_USE_APPROX_NEAR_CLIFFORD): a blocked non-Clifford phase shard
+    // snaps to the floor/ceil Clifford quarter-turn with probability
+    // proportional to the fractional part — unbiased per shot, no ancilla
+    // spent; the expected per-event overlap is logged like NCRP
+    bool stochasticNC = false;
 
     // NCRP: if the shard is a phase gate within ncrp of a Clifford phase,
     // snap it into the tableau and log the exact overlap loss. Returns true
@@ -53,6 +59,8 @@ protected:
     bool TryShardRoundClifford(bitLenInt q);
     // the gadget step above; returns true if the shard was absorbed
     bool TryShardGadget(bitLenInt q);
+    // stochastic rounding step (always succeeds for phase shards)
+    bool TryShardStochastic(bitLenInt q);
     bitLenInt TableauWidth() const { return qubitCount + ancillaCount; }
 
     // rdm-clone cache (reference qstabilizerhybrid.hpp:68): read-only
@@ -86,6 +94,7 @@ public:
     void SwitchToEngine();
 
     void SetNcrp(double v) override { ncrp = v; }
+    void SetStochastic(bool on) override { stochasticNC = on; }
     double GetNcrp() override { return ncrp; }
     double GetUnitaryFidelity() override { return std::exp(logFidelity); }
     void ResetUnitaryFidelity() override { logFidelity = 0.0; }
