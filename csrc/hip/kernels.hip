@@ -1246,6 +1246,62 @@ void launchQftColumnGeneral(cplx<R>* sv, bitCapInt maxQPower, bitCapInt tPow, co
     }
 }
 
+// TWO disjoint 4x4 gates per full-state pass (see kernels.hpp
+// Gate4x4Pair2Args): each thread owns a 16-amplitude orbit over the two
+// gates' four bit positions, applies gate A across its axis pair then
+// gate B across the other — one state read+write for two SU(4)s.
+template <typename R>
+__global__ void k_mtrx2q_pair2(cplx<R>* sv, Gate4x4Pair2Args<R> a)
+{
+    const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
+    for (bitCapInt k = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; k < a.orbits;
+         k += stride) {
+        bitCapInt r = k;
+        for (int b = 0; b < 4; ++b) r = insertZeroBitDev(r, a.sorted4[b]);
+        bitCapInt idx[16];
+        for (int ib = 0; ib < 4; ++ib) {
+            for (int ia = 0; ia < 4; ++ia) {
+                bitCapInt x = r;
+                if (ia & 1) x |= a.pA1;
+                if (ia & 2) x |= a.pA2;
+                if (ib & 1) x |= a.pB1;
+                if (ib & 2) x |= a.pB2;
+                idx[4 * ib + ia] = x;
+            }
+        }
+        cplx<R> v[16];
+        for (int b = 0; b < 16; ++b) v[b] = sv[idx[b]];
+        // gate A mixes the ia axis for each ib
+        for (int ib = 0; ib < 4; ++ib) {
+            cplx<R> out[4];
+            for (int rI = 0; rI < 4; ++rI) {
+                cplx<R> s(0, 0);
+                for (int cI = 0; cI < 4; ++cI) s = s + a.mA[4 * rI + cI] * v[4 * ib + cI];
+                out[rI] = s;
+            }
+            for (int rI = 0; rI < 4; ++rI) v[4 * ib + rI] = out[rI];
+        }
+        // gate B mixes the ib axis for each ia
+        for (int ia = 0; ia < 4; ++ia) {
+            cplx<R> out[4];
+            for (int rI = 0; rI < 4; ++rI) {
+                cplx<R> s(0, 0);
+                for (int cI = 0; cI < 4; ++cI) s = s + a.mB[4 * rI + cI] * v[4 * cI + ia];
+                out[rI] = s;
+            }
+            for (int rI = 0; rI < 4; ++rI) v[4 * rI + ia] = out[rI];
+        }
+        for (int b = 0; b < 16; ++b) sv[idx[b]] = v[b];
+    }
+}
+
+template <typename R>
+void launchMtrx2qPair2(cplx<R>* sv, const Gate4x4Pair2Args<R>& a, hipStream_t stream)
+{
+    hipLaunchKernelGGL(
+        (k_mtrx2q_pair2<R>), dim3(gridFor(a.orbits)), dim3(QA_BLOCK), 0, stream, sv, a);
+}
+
 // TWO generalized QFT columns per pass for the distributed pager's local
 // ladder: targets at ARBITRARY local slots (tPowHi/tPowLo by ROLE, not
 // numeric order), ramp bits possibly relocated (RampArgs, which must
@@ -2558,6 +2614,7 @@ void launchPartProbs(const cplx<R>* sv, bitCapInt maxQPower, bitLenInt start, bi
     template void launchMtrx2qBatchLds<R>(cplx<R>*, const Batch2qLdsArgs<R>&, hipStream_t);                \
     template void launchMtrx2q<R>(cplx<R>*, const Gate4x4Args<R>&, hipStream_t);                              \
     template void launchCPhasePairs<R>(cplx<R>*, const CPhasePairsArgs&, hipStream_t);                               \
+    template void launchMtrx2qPair2<R>(cplx<R>*, const Gate4x4Pair2Args<R>&, hipStream_t);         \
     template void launchQftColumn2General<R>(cplx<R>*, bitCapInt, bitCapInt, bitCapInt,            \
         const RampArgs&, double, double, bool, hipStream_t);                                        \
     template void launchQftColumnTopRange<R>(cplx<R>*, bitCapInt, const RampArgs&, double, bool,    \

@@ -290,6 +290,21 @@ template <typename R> struct Gate4x4Args {
 template <typename R>
 void launchMtrx2q(cplx<R>* sv, const Gate4x4Args<R>& a, hipStream_t stream);
 
+// TWO disjoint 4x4s in ONE full-state pass (16-amplitude orbits — the same
+// register-orbit budget the 4-column QFT kernel proved runs at full HBM
+// bandwidth). Matrices row-major in |q2 q1> basis with powers pre-sorted
+// per gate (pA1<pA2, pB1<pB2); all four bit positions distinct.
+template <typename R> struct Gate4x4Pair2Args {
+    cplx<R> mA[16];
+    cplx<R> mB[16];
+    bitCapInt pA1, pA2, pB1, pB2;
+    bitCapInt sorted4[4]; // the four powers ascending (orbit insertion)
+    bitCapInt orbits;     // maxQPower >> 4
+};
+
+template <typename R>
+void launchMtrx2qPair2(cplx<R>* sv, const Gate4x4Pair2Args<R>& a, hipStream_t stream);
+
 // batched disjoint CNOTs: a whole layer of k control/target pairs (no qubit
 // repeated) applied as ONE in-place permutation pass — amp[i] swaps with
 // amp[i ^ xm(i)] where xm(i) XORs tPow[j] for every set control bit. One

@@ -483,7 +483,26 @@ void QEngineHIP<R>::Mtrx2qBatch(const std::vector<cplx<R>>& ms,
         launchMtrx2qBatchLds<R>(dState, a, stream);
         i += k;
     }
-    for (size_t ix : rest) {
+    // the rest: TWO disjoint 4x4s per pass (16-amplitude orbits)
+    size_t j = 0;
+    for (; disjoint && j + 1 < rest.size(); j += 2) {
+        const size_t ixA = rest[j], ixB = rest[j + 1];
+        Gate4x4Pair2Args<R> a{};
+        permuted(&ms[16u * ixA], q1s[ixA] > q2s[ixA], a.mA);
+        permuted(&ms[16u * ixB], q1s[ixB] > q2s[ixB], a.mB);
+        a.pA1 = pow2(std::min(q1s[ixA], q2s[ixA]));
+        a.pA2 = pow2(std::max(q1s[ixA], q2s[ixA]));
+        a.pB1 = pow2(std::min(q1s[ixB], q2s[ixB]));
+        a.pB2 = pow2(std::max(q1s[ixB], q2s[ixB]));
+        bitCapInt s4[4] = { a.pA1, a.pA2, a.pB1, a.pB2 };
+        std::sort(s4, s4 + 4);
+        for (int b = 0; b < 4; ++b) a.sorted4[b] = s4[b];
+        a.orbits = maxQPower >> 4u;
+        HipProfScope prof("mtrx_2q_pair2", stream);
+        launchMtrx2qPair2<R>(dState, a, stream);
+    }
+    for (; j < rest.size(); ++j) {
+        const size_t ix = rest[j];
         this->Mtrx2q(&ms[16u * ix], q1s[ix], q2s[ix]);
     }
 }

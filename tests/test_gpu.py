@@ -589,3 +589,30 @@ def test_fuser_over_hip_numerics():
             f.cnot(a, a + 1)
             b.cnot(a, a + 1)
     assert_states_close(f.get_state_vector(), b.get_state_vector(), 2e-4)
+
+
+def test_mtrx2q_pair2_high_bits():
+    """Two disjoint HIGH-bit 4x4s must route through the 16-amplitude-orbit
+    pair kernel and match per-gate application."""
+    n = 18
+    rng = np.random.default_rng(87)
+
+    def u4():
+        z = rng.normal(size=(4, 4)) + 1j * rng.normal(size=(4, 4))
+        qm, r = np.linalg.qr(z)
+        return qm * (np.diag(r) / np.abs(np.diag(r)))
+
+    for pairs in [[(13, 16), (14, 17)], [(16, 13), (12, 15)],
+                  [(2, 14), (15, 5), (13, 16), (17, 3)]]:
+        us = [u4() for _ in pairs]
+        qb = make(n, seed=3)
+        qs = make(n, seed=3)
+        for i in range(n):
+            th = float(rng.uniform(0, np.pi))
+            qb.ry(th, i)
+            qs.ry(th, i)
+        flat = [complex(x) for u in us for x in u.flatten()]
+        qb.mtrx_2q_batch(flat, [a for a, _ in pairs], [b for _, b in pairs])
+        for u, (a, b) in zip(us, pairs):
+            qs.mtrx_2q([complex(x) for x in u.flatten()], a, b)
+        assert float(qb.sum_sqr_diff(qs)) < 1e-5
