@@ -559,8 +559,25 @@ void QEngineHIP<R>::FSimBatch(const std::vector<R>& thetas, const std::vector<R>
         launchMtrx2qBatchLds<R>(dState, a, stream);
         i += k;
     }
-    for (size_t ix : rest) {
-        this->FSim(thetas[ix], phis[ix], q1s[ix], q2s[ix]);
+    if (!rest.empty()) {
+        // route high-bit fsims through Mtrx2qBatch as 4x4s: the pair-of-4x4
+        // orbit kernel applies TWO per full-state pass
+        std::vector<cplx<R>> ms(16u * rest.size(), cplx<R>(0, 0));
+        std::vector<bitLenInt> ra, rb;
+        for (size_t g = 0; g < rest.size(); ++g) {
+            const size_t ix = rest[g];
+            const R ct = std::cos(thetas[ix]), st = std::sin(thetas[ix]);
+            cplx<R>* m = &ms[16u * g];
+            m[0] = cplx<R>(1, 0);
+            m[5] = cplx<R>(ct, 0);
+            m[6] = cplx<R>(0, -st);
+            m[9] = cplx<R>(0, -st);
+            m[10] = cplx<R>(ct, 0);
+            m[15] = polar<R>(1, -phis[ix]);
+            ra.push_back(q1s[ix]);
+            rb.push_back(q2s[ix]);
+        }
+        Mtrx2qBatch(ms, ra, rb);
     }
 }
 
