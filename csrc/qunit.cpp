@@ -39,6 +39,9 @@ QUnit<R>::QUnit(bitLenInt qBitCount, bitCapInt initState, RngPtr rgp, EngineFact
     if (const char* env = std::getenv("QRACK_QUNIT_ACE_MAX_QB")) {
         aceMaxQubits = (bitLenInt)std::atoi(env);
     }
+    if (std::getenv("QRACK_DISABLE_QUNIT_FIDELITY_GUARD")) {
+        fidelityGuard = false;
+    }
     for (bitLenInt q = 0; q < qBitCount; ++q) {
         shards[q].unit = MakeUnit(1u, (initState >> q) & 1u);
         shards[q].mapped = 0;
@@ -179,6 +182,7 @@ bool QUnit<R>::ElideControls(const std::vector<bitLenInt>& controls, bool anti, 
         ForceM(c, outcome, true, true);
         if (anti ? outcome : !outcome) gateApplies = false;
     }
+    CheckFidelity();
     return true;
 }
 
@@ -838,6 +842,7 @@ template <typename R> bool QUnit<R>::TrySeparate(bitLenInt q)
     const R p0 = (R)1 - p1;
     const bool outcome = (p1 > p0);
     logFidelity += std::log(std::max((double)(outcome ? p1 : p0), 1e-300));
+    CheckFidelity();
     unit->ForceM(m, outcome, true, true);
     unit->Dispose(m, 1u, outcome ? 1u : 0u);
     FixMappedAfterRemoval(unit, m);

@@ -33,6 +33,10 @@ protected:
     EngineFactoryFn<R> subFactory;
     double logFidelity = 0.0;
     R separabilityThreshold;
+    // fidelity guard (reference qunit.hpp:105-117 CheckFidelity): once the
+    // accumulated rounding-fidelity estimate is effectively zero, further
+    // results are meaningless — throw unless the user opts out
+    bool fidelityGuard = true;
     double sdrp = 0.0; // Schmidt-decomposition rounding parameter (0 = exact)
     double ncrp = 0.0; // near-Clifford rounding parameter, forwarded to units
     bool reactiveSeparate = false; // TrySeparate after entangling gates
@@ -399,6 +403,16 @@ protected:
                 shards[p.t].unit->Invert(
                     polar<R>(1, (R)p.angle), cplx<R>(1, 0), shards[p.t].mapped);
             }
+        }
+    }
+
+    static constexpr double QA_FIDELITY_MIN = -46.05170185988091; // ln(1e-20)
+    void CheckFidelity()
+    {
+        if (fidelityGuard && logFidelity <= QA_FIDELITY_MIN) {
+            throw QrackError(
+                "QUnit fidelity estimate is effectively 0 (this does not prove the true fidelity "
+                "is 0 — see README; set QRACK_DISABLE_QUNIT_FIDELITY_GUARD=1 to continue anyway)");
         }
     }
 

@@ -213,3 +213,35 @@ def test_stochastic_near_clifford():
     # terms (reference's caveat); accept a broad band around the exact
     # value but far from 0/0.5 degeneracy
     assert 0.05 < p1 < 0.35, p1
+
+
+def test_fidelity_guard_throws_and_opts_out():
+    """Reference CheckFidelity parity: deep ACE-elided circuits that drive
+    the fidelity estimate to ~0 raise; QRACK_DISABLE_QUNIT_FIDELITY_GUARD
+    opts out."""
+    import os
+
+    import numpy as np
+
+    def grind(n_rounds):
+        os.environ["QRACK_QUNIT_ACE_MAX_QB"] = "2"
+        try:
+            q = qa.create_simulator(6, layers=["qunit", "cpu"], seed=5)
+            rng = np.random.default_rng(5)
+            for _ in range(n_rounds):
+                for t in range(6):
+                    q.ry(float(rng.uniform(0.5, 1.0)), t)
+                for a in range(5):
+                    q.cnot(a, a + 1)  # every coupler elides under the cap
+            return q.get_unitary_fidelity()
+        finally:
+            del os.environ["QRACK_QUNIT_ACE_MAX_QB"]
+
+    with pytest.raises(Exception):
+        grind(60)
+    os.environ["QRACK_DISABLE_QUNIT_FIDELITY_GUARD"] = "1"
+    try:
+        f = grind(60)
+        assert 0.0 <= f < 1e-10
+    finally:
+        del os.environ["QRACK_DISABLE_QUNIT_FIDELITY_GUARD"]
