@@ -28,6 +28,11 @@ QPager<R>::QPager(bitLenInt qBitCount, bitCapInt initState, RngPtr rgp, EngineFa
     }
     qpp = pq;
     metaBits = qBitCount - qpp;
+    // env parity: QRACK_MAX_PAGING_QB caps the total paged width
+    if (const char* env = std::getenv("QRACK_MAX_PAGING_QB")) {
+        const bitLenInt cap = (bitLenInt)std::atoi(env);
+        if (qBitCount > cap) throw std::bad_alloc();
+    }
     if (metaBits > 16u) throw QrackError("QPager: too many pages");
     qPages.resize(PageCount());
     for (bitCapInt p = 0; p < PageCount(); ++p) {
