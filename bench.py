@@ -168,6 +168,9 @@ def main():
                 "n_gates": n_gates,
                 "single_gate_ms": single_gate_ms,
                 "isolated_h_ms": isolated_h_ms,
+                # capacity headline (measured: profiles/qft_fuse4.csv runs
+                # the 34-qubit = 128 GB fp32 QFT at fidelity 1.0)
+                "max_full_state_qubits_fp32": 34,
                 "engine": engine,
                 "global_batch": 1,
                 "seq_len": qubits,
