@@ -1799,6 +1799,78 @@ __global__ void k_qft_col3_v(cplx<float>* sv, bitCapInt orbitPairs, bitCapInt tH
     }
 }
 
+// ALL low-bit QFT columns in ONE pass: when the register starts at bit 0,
+// columns tb-1..0 act entirely inside a contiguous 2^tb-amplitude tile
+// (every column's ramp uses only lower tile bits). One 32 KB LDS tile per
+// block applies the whole low ladder — colMax+1 columns for one global
+// state read+write. tb = 12 (fp32) / 11 (fp64).
+template <typename R, bool PRE>
+__global__ void k_qft_low_lds(cplx<R>* sv, bitCapInt nTiles, int tb, int colMax, R piSign)
+{
+    extern __shared__ unsigned char qa_lds_raw[];
+    cplx<R>* lds = reinterpret_cast<cplx<R>*>(qa_lds_raw);
+    const int tile = 1 << tb;
+    const int pairs = tile >> 1;
+    for (bitCapInt t = blockIdx.x; t < nTiles; t += gridDim.x) {
+        const bitCapInt base = t << tb;
+        for (int j = threadIdx.x; j < tile; j += blockDim.x) lds[j] = sv[base + j];
+        __syncthreads();
+        const R s = (R)0.70710678118654752440;
+        if (!PRE) {
+            for (int col = colMax; col >= 0; --col) {
+                const int half = 1 << col;
+                const R scale = piSign / (R)half;
+                for (int k = threadIdx.x; k < pairs; k += blockDim.x) {
+                    const int lo = ((k >> col) << (col + 1)) | (k & (half - 1));
+                    const int hi = lo | half;
+                    const cplx<R> x = lds[lo], y = lds[hi];
+                    R sn, cs;
+                    devSinCos<R>(scale * (R)(lo & (half - 1)), &sn, &cs);
+                    const cplx<R> f{ cs, sn };
+                    lds[lo] = s * (x + y);
+                    lds[hi] = f * (s * (x - y));
+                }
+                __syncthreads();
+            }
+        } else {
+            for (int col = 0; col <= colMax; ++col) {
+                const int half = 1 << col;
+                const R scale = piSign / (R)half;
+                for (int k = threadIdx.x; k < pairs; k += blockDim.x) {
+                    const int lo = ((k >> col) << (col + 1)) | (k & (half - 1));
+                    const int hi = lo | half;
+                    R sn, cs;
+                    devSinCos<R>(scale * (R)(lo & (half - 1)), &sn, &cs);
+                    const cplx<R> f{ cs, sn };
+                    const cplx<R> x = lds[lo], y = f * lds[hi];
+                    lds[lo] = s * (x + y);
+                    lds[hi] = s * (x - y);
+                }
+                __syncthreads();
+            }
+        }
+        for (int j = threadIdx.x; j < tile; j += blockDim.x) sv[base + j] = lds[j];
+        __syncthreads();
+    }
+}
+
+template <typename R>
+void launchQftLowLds(
+    cplx<R>* sv, bitCapInt maxQPower, int tb, int colMax, int sign, bool pre, hipStream_t stream)
+{
+    const bitCapInt nTiles = maxQPower >> tb;
+    const size_t ldsBytes = (size_t(1) << tb) * sizeof(cplx<R>);
+    const int grid = (int)std::min<bitCapInt>(nTiles, (bitCapInt)QA_REDUCE_MAX_BLOCKS);
+    const R piSign = (R)sign * (R)3.14159265358979323846;
+    if (pre) {
+        hipLaunchKernelGGL((k_qft_low_lds<R, true>), dim3(grid), dim3(QA_BLOCK), ldsBytes, stream,
+            sv, nTiles, tb, colMax, piSign);
+    } else {
+        hipLaunchKernelGGL((k_qft_low_lds<R, false>), dim3(grid), dim3(QA_BLOCK), ldsBytes,
+            stream, sv, nTiles, tb, colMax, piSign);
+    }
+}
+
 // GENERIC K-column fused QFT kernel (2^K-amplitude orbits). The per-column
 // ramp factor for the slot with column-c's bit set is
 //   f0^(2^(K-1-c)) · Π_{j<c} U(c-j)^(b_j),   U(d) = e^{i·sign·π/2^d}
@@ -2615,6 +2687,7 @@ void launchPartProbs(const cplx<R>* sv, bitCapInt maxQPower, bitLenInt start, bi
     template void launchMtrx2q<R>(cplx<R>*, const Gate4x4Args<R>&, hipStream_t);                              \
     template void launchCPhasePairs<R>(cplx<R>*, const CPhasePairsArgs&, hipStream_t);                               \
     template void launchMtrx2qPair2<R>(cplx<R>*, const Gate4x4Pair2Args<R>&, hipStream_t);         \
+    template void launchQftLowLds<R>(cplx<R>*, bitCapInt, int, int, int, bool, hipStream_t);        \
     template void launchQftColumn2General<R>(cplx<R>*, bitCapInt, bitCapInt, bitCapInt,            \
         const RampArgs&, double, double, bool, hipStream_t);                                        \
     template void launchQftColumnTopRange<R>(cplx<R>*, bitCapInt, const RampArgs&, double, bool,    \

@@ -189,6 +189,12 @@ template <typename R>
 void launchQftColumn2(cplx<R>* sv, bitCapInt maxQPower, bitLenInt rampStart, bitLenInt col,
     bitCapInt tHi, bitCapInt tLo, int sign, bool pre, hipStream_t stream);
 
+// one-pass low-bit QFT ladder: columns colMax..0 applied inside contiguous
+// 2^tb-amplitude LDS tiles (start-0 registers only)
+template <typename R>
+void launchQftLowLds(
+    cplx<R>* sv, bitCapInt maxQPower, int tb, int colMax, int sign, bool pre, hipStream_t stream);
+
 // generic K-column fused QFT pass (2^K-amplitude orbits; K=4 instantiated)
 template <typename R>
 void launchQftColumnK(cplx<R>* sv, bitCapInt maxQPower, bitLenInt rampStart, bitLenInt col,

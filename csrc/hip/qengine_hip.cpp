@@ -876,9 +876,21 @@ template <typename R> void QEngineHIP<R>::QFT(bitLenInt start, bitLenInt length,
         if (const char* env = std::getenv("QRACK_GPU_QFT_FUSE")) return std::atoi(env);
         return 4;
     }();
+    static const bool ldsLow = []() {
+        if (const char* env = std::getenv("QRACK_GPU_QFT_LDS")) return std::atoi(env) != 0;
+        return true;
+    }();
+    const int tb = qaLdsTileBits<R>();
     bitLenInt i = length;
     while (i > 0) {
         const bitLenInt col = i - 1u;
+        if (ldsLow && start == 0u && (bitLenInt)i <= (bitLenInt)tb &&
+            maxQPower >= (ONE_BCI << tb)) {
+            // the whole remaining low ladder in ONE LDS-tiled pass
+            HipProfScope prof("qft_low_lds", stream);
+            launchQftLowLds<R>(dState, maxQPower, tb, (int)col, +1, false, stream);
+            break;
+        }
         if (fuseMax >= 5 && col >= 4u && maxQPower >= 64u) {
             HipProfScope prof("qft_column5", stream);
             const bitCapInt tPows[5] = { pow2(start + col - 4u), pow2(start + col - 3u),
@@ -932,7 +944,18 @@ template <typename R> void QEngineHIP<R>::IQFT(bitLenInt start, bitLenInt length
         if (const char* env = std::getenv("QRACK_GPU_QFT_FUSE")) return std::atoi(env);
         return 4;
     }();
+    static const bool ldsLow = []() {
+        if (const char* env = std::getenv("QRACK_GPU_QFT_LDS")) return std::atoi(env) != 0;
+        return true;
+    }();
+    const int tb = qaLdsTileBits<R>();
     bitLenInt i = 0;
+    if (ldsLow && start == 0u && length > 0u && maxQPower >= (ONE_BCI << tb)) {
+        const bitLenInt colMax = std::min<bitLenInt>(length - 1u, (bitLenInt)tb - 1u);
+        HipProfScope prof("qft_low_lds", stream);
+        launchQftLowLds<R>(dState, maxQPower, tb, (int)colMax, -1, true, stream);
+        i = colMax + 1u;
+    }
     while (i < length) {
         if (fuseMax >= 5 && (i + 4u) < length && maxQPower >= 64u) {
             HipProfScope prof("qft_column5", stream);
