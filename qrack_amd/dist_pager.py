@@ -619,9 +619,18 @@ class DistQPager:
             float(scale_hi * meta_w), float(2.0 * scale_hi * meta_w), pre)
         return True
 
+    def _identity_low(self, start, upto):
+        return start == 0 and all(self.slot_of[j] == j for j in range(upto + 1))
+
     def qft(self, start, length):
         i = length - 1
         while i >= 0:
+            if (i < 12 and self._is_hip() and self._identity_low(start, i)
+                    and hasattr(self.q, "qft_column2_general")):
+                # the whole remaining low ladder rides the engine's
+                # LDS-tiled one-pass kernel (identity map, no meta bits)
+                self.q.qft(0, i + 1)
+                return
             if i >= 1 and self._try_pair_local(start, i, +1, False):
                 i -= 2
                 continue
@@ -630,6 +639,13 @@ class DistQPager:
 
     def iqft(self, start, length):
         i = 0
+        if self._is_hip() and length > 0 and hasattr(self.q, "qft_column2_general"):
+            k = min(length, 12)
+            while k > 0 and not self._identity_low(start, k - 1):
+                k -= 1
+            if k >= 2:
+                self.q.iqft(0, k)
+                i = k
         while i < length:
             if i + 1 < length and self._try_pair_local(start, i + 1, -1, True):
                 i += 2
