@@ -1871,6 +1871,91 @@ void launchQftLowLds(
     }
 }
 
+// SIX mid-range QFT columns per pass through a 2D LDS tile: 64 contiguous
+// low amplitudes (coalesced 512 B runs) x 2^nCols column-bit combinations.
+// Ramp for column col: theta = scale_col * (x mod 2^col) where
+// x mod 2^col = low6 + midFixed + cbLow*2^colLo — low6 varies in-tile,
+// midFixed (bits 6..colLo-1) is tile-constant, cbLow is the tile's lower
+// column bits. Requires a start-0 register and colLo >= 6.
+template <typename R, bool PRE>
+__global__ void k_qft_mid_lds(cplx<R>* sv, bitCapInt nTiles, int colLo, int nCols, R piSign)
+{
+    extern __shared__ unsigned char qa_lds_raw2[];
+    cplx<R>* lds = reinterpret_cast<cplx<R>*>(qa_lds_raw2);
+    const int C = 1 << nCols;
+    const int tileAmps = 64 * C;
+    const int pairs = tileAmps >> 1;
+    const bitCapInt midMask = (ONE_BCI << (colLo - 6)) - 1u;
+    const R s = (R)0.70710678118654752440;
+    for (bitCapInt t = blockIdx.x; t < nTiles; t += gridDim.x) {
+        const bitCapInt xBase =
+            ((t >> (colLo - 6)) << (colLo + nCols)) | ((t & midMask) << 6);
+        const R midFixed = (R)(uint64_t)((t & midMask) << 6);
+        // load: row cb = 64 contiguous amps
+        for (int j = threadIdx.x; j < tileAmps; j += blockDim.x) {
+            const int cb = j >> 6;
+            const int low = j & 63;
+            lds[j] = sv[xBase | ((bitCapInt)cb << colLo) | (bitCapInt)low];
+        }
+        __syncthreads();
+        // columns colLo+nCols-1 .. colLo (forward) or reverse (PRE)
+        for (int step = 0; step < nCols; ++step) {
+            const int c = PRE ? step : (nCols - 1 - step);
+            const int col = colLo + c;
+            const int cbHalf = 1 << c; // pair stride in cb space
+            const R scale = piSign / (R)(ONE_BCI << col);
+            const R colScale = scale * (R)(ONE_BCI << colLo);
+            for (int k = threadIdx.x; k < pairs; k += blockDim.x) {
+                // pair index k over (cb-without-bit-c, low)
+                const int low = k & 63;
+                const int cbr = k >> 6;
+                const int cb0 = ((cbr >> c) << (c + 1)) | (cbr & (cbHalf - 1));
+                const int lo = (cb0 << 6) | low;
+                const int hi = ((cb0 | cbHalf) << 6) | low;
+                const int cbLow = cb0 & (cbHalf - 1);
+                R sn, cs;
+                devSinCos<R>(
+                    scale * ((R)low + midFixed) + colScale * (R)cbLow, &sn, &cs);
+                const cplx<R> f{ cs, sn };
+                const cplx<R> x = lds[lo];
+                if (!PRE) {
+                    const cplx<R> y = lds[hi];
+                    lds[lo] = s * (x + y);
+                    lds[hi] = f * (s * (x - y));
+                } else {
+                    const cplx<R> y = f * lds[hi];
+                    lds[lo] = s * (x + y);
+                    lds[hi] = s * (x - y);
+                }
+            }
+            __syncthreads();
+        }
+        for (int j = threadIdx.x; j < tileAmps; j += blockDim.x) {
+            const int cb = j >> 6;
+            const int low = j & 63;
+            sv[xBase | ((bitCapInt)cb << colLo) | (bitCapInt)low] = lds[j];
+        }
+        __syncthreads();
+    }
+}
+
+template <typename R>
+void launchQftMidLds(
+    cplx<R>* sv, bitCapInt maxQPower, int colLo, int nCols, int sign, bool pre, hipStream_t stream)
+{
+    const bitCapInt nTiles = maxQPower >> (6 + nCols);
+    const size_t ldsBytes = (size_t(64) << nCols) * sizeof(cplx<R>);
+    const int grid = (int)std::min<bitCapInt>(nTiles, (bitCapInt)QA_REDUCE_MAX_BLOCKS);
+    const R piSign = (R)sign * (R)3.14159265358979323846;
+    if (pre) {
+        hipLaunchKernelGGL((k_qft_mid_lds<R, true>), dim3(grid), dim3(QA_BLOCK), ldsBytes, stream,
+            sv, nTiles, colLo, nCols, piSign);
+    } else {
+        hipLaunchKernelGGL((k_qft_mid_lds<R, false>), dim3(grid), dim3(QA_BLOCK), ldsBytes,
+            stream, sv, nTiles, colLo, nCols, piSign);
+    }
+}
+
 // GENERIC K-column fused QFT kernel (2^K-amplitude orbits). The per-column
 // ramp factor for the slot with column-c's bit set is
 //   f0^(2^(K-1-c)) · Π_{j<c} U(c-j)^(b_j),   U(d) = e^{i·sign·π/2^d}
@@ -2688,6 +2773,7 @@ void launchPartProbs(const cplx<R>* sv, bitCapInt maxQPower, bitLenInt start, bi
     template void launchCPhasePairs<R>(cplx<R>*, const CPhasePairsArgs&, hipStream_t);                               \
     template void launchMtrx2qPair2<R>(cplx<R>*, const Gate4x4Pair2Args<R>&, hipStream_t);         \
     template void launchQftLowLds<R>(cplx<R>*, bitCapInt, int, int, int, bool, hipStream_t);        \
+    template void launchQftMidLds<R>(cplx<R>*, bitCapInt, int, int, int, bool, hipStream_t);        \
     template void launchQftColumn2General<R>(cplx<R>*, bitCapInt, bitCapInt, bitCapInt,            \
         const RampArgs&, double, double, bool, hipStream_t);                                        \
     template void launchQftColumnTopRange<R>(cplx<R>*, bitCapInt, const RampArgs&, double, bool,    \

@@ -195,6 +195,12 @@ template <typename R>
 void launchQftLowLds(
     cplx<R>* sv, bitCapInt maxQPower, int tb, int colMax, int sign, bool pre, hipStream_t stream);
 
+// nCols mid-range QFT columns per pass through a 2D LDS tile (64 contiguous
+// low amps x 2^nCols column-bit combos; start-0 registers, colLo >= 6)
+template <typename R>
+void launchQftMidLds(cplx<R>* sv, bitCapInt maxQPower, int colLo, int nCols, int sign, bool pre,
+    hipStream_t stream);
+
 // generic K-column fused QFT pass (2^K-amplitude orbits; K=4 instantiated)
 template <typename R>
 void launchQftColumnK(cplx<R>* sv, bitCapInt maxQPower, bitLenInt rampStart, bitLenInt col,

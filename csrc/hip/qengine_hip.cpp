@@ -880,7 +880,13 @@ template <typename R> void QEngineHIP<R>::QFT(bitLenInt start, bitLenInt length,
         if (const char* env = std::getenv("QRACK_GPU_QFT_LDS")) return std::atoi(env) != 0;
         return true;
     }();
+    static const bool ldsMid = []() {
+        if (const char* env = std::getenv("QRACK_GPU_QFT_MIDLDS")) return std::atoi(env) != 0;
+        return true;
+    }();
     const int tb = qaLdsTileBits<R>();
+    // mid-LDS group width: 64 x 2^G tile (fp32 32 KB, fp64 32 KB at G=5)
+    const int G = sizeof(R) == 4 ? 6 : 5;
     bitLenInt i = length;
     while (i > 0) {
         const bitLenInt col = i - 1u;
@@ -890,6 +896,15 @@ template <typename R> void QEngineHIP<R>::QFT(bitLenInt start, bitLenInt length,
             HipProfScope prof("qft_low_lds", stream);
             launchQftLowLds<R>(dState, maxQPower, tb, (int)col, +1, false, stream);
             break;
+        }
+        if (ldsLow && ldsMid && start == 0u && (bitLenInt)i > (bitLenInt)tb &&
+            maxQPower >= (ONE_BCI << tb)) {
+            // up to G mid columns per pass through a 2D LDS tile
+            const int nc = std::min<int>(G, (int)(i - (bitLenInt)tb));
+            HipProfScope prof("qft_mid_lds", stream);
+            launchQftMidLds<R>(dState, maxQPower, (int)i - nc, nc, +1, false, stream);
+            i -= (bitLenInt)nc;
+            continue;
         }
         if (fuseMax >= 5 && col >= 4u && maxQPower >= 64u) {
             HipProfScope prof("qft_column5", stream);
@@ -948,7 +963,12 @@ template <typename R> void QEngineHIP<R>::IQFT(bitLenInt start, bitLenInt length
         if (const char* env = std::getenv("QRACK_GPU_QFT_LDS")) return std::atoi(env) != 0;
         return true;
     }();
+    static const bool ldsMid = []() {
+        if (const char* env = std::getenv("QRACK_GPU_QFT_MIDLDS")) return std::atoi(env) != 0;
+        return true;
+    }();
     const int tb = qaLdsTileBits<R>();
+    const int G = sizeof(R) == 4 ? 6 : 5;
     bitLenInt i = 0;
     if (ldsLow && start == 0u && length > 0u && maxQPower >= (ONE_BCI << tb)) {
         const bitLenInt colMax = std::min<bitLenInt>(length - 1u, (bitLenInt)tb - 1u);
@@ -957,6 +977,14 @@ template <typename R> void QEngineHIP<R>::IQFT(bitLenInt start, bitLenInt length
         i = colMax + 1u;
     }
     while (i < length) {
+        if (ldsLow && ldsMid && start == 0u && i >= (bitLenInt)tb &&
+            maxQPower >= (ONE_BCI << tb)) {
+            const int nc = std::min<int>(G, (int)(length - i));
+            HipProfScope prof("qft_mid_lds", stream);
+            launchQftMidLds<R>(dState, maxQPower, (int)i, nc, -1, true, stream);
+            i += (bitLenInt)nc;
+            continue;
+        }
         if (fuseMax >= 5 && (i + 4u) < length && maxQPower >= 64u) {
             HipProfScope prof("qft_column5", stream);
             const bitCapInt tPows[5] = { pow2(start + i), pow2(start + i + 1u),

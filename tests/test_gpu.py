@@ -574,6 +574,30 @@ def test_qft_fused2_numerics_vs_cpu():
     assert_states_close(q.get_state_vector(), cp.get_state_vector(), 2e-4)
 
 
+def test_qft_mid_lds_numerics_vs_cpu():
+    """The 2D-tile mid-column LDS QFT kernel (k_qft_mid_lds: fires for
+    start-0 registers wider than the low-ladder tile — n > 12 fp32,
+    n > 11 fp64) must match the CPU engine in both directions, including
+    partial trailing groups (nCols < 6)."""
+    for n, prec in ((14, "fp32"), (16, "fp32"), (19, "fp32"), (14, "fp64"), (17, "fp64")):
+        rng = np.random.default_rng(n)
+        q = qa.create_simulator(n, engine="hip", precision=prec, seed=n)
+        cp = qa.create_simulator(n, engine="cpu", precision=prec, seed=n)
+        init = int(rng.integers(0, 1 << n))
+        q.set_permutation(init)
+        cp.set_permutation(init)
+        for i in range(0, n, 3):  # non-basis state: superpose a few qubits
+            q.ry(0.7 + 0.1 * i, i)
+            cp.ry(0.7 + 0.1 * i, i)
+        q.qft(0, n)
+        cp.qft(0, n)
+        tol = 2e-4 if prec == "fp32" else 1e-9
+        assert_states_close(q.get_state_vector(), cp.get_state_vector(), tol)
+        q.iqft(0, n)
+        cp.iqft(0, n)
+        assert_states_close(q.get_state_vector(), cp.get_state_vector(), tol)
+
+
 def test_fuser_over_hip_numerics():
     """QFuser batching over the HIP engine matches the bare engine."""
     n = 12
