@@ -620,13 +620,16 @@ class DistQPager:
         return True
 
     def _identity_low(self, start, upto):
-        return start == 0 and all(self.slot_of[j] == j for j in range(upto + 1))
+        # columns 0..upto ride the local engine's one-call QFT ladder only
+        # when every one of them is a LOCAL slot mapped identically (a meta
+        # qubit j has slot_of[j] == j too, but lives on the page index)
+        return (start == 0 and upto < self.qpp
+                and all(self.slot_of[j] == j for j in range(upto + 1)))
 
     def qft(self, start, length):
         i = length - 1
         while i >= 0:
-            if (i < 12 and self._is_hip() and self._identity_low(start, i)
-                    and hasattr(self.q, "qft_column2_general")):
+            if i < 12 and self._identity_low(start, i):
                 # the whole remaining low ladder rides the engine's
                 # LDS-tiled one-pass kernel (identity map, no meta bits)
                 self.q.qft(0, i + 1)
@@ -639,7 +642,7 @@ class DistQPager:
 
     def iqft(self, start, length):
         i = 0
-        if self._is_hip() and length > 0 and hasattr(self.q, "qft_column2_general"):
+        if length > 0:
             k = min(length, 12)
             while k > 0 and not self._identity_low(start, k - 1):
                 k -= 1
