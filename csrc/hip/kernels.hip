@@ -1799,57 +1799,164 @@ __global__ void k_qft_col3_v(cplx<float>* sv, bitCapInt orbitPairs, bitCapInt tH
     }
 }
 
-// ALL low-bit QFT columns in ONE pass: when the register starts at bit 0,
-// columns tb-1..0 act entirely inside a contiguous 2^tb-amplitude tile
-// (every column's ramp uses only lower tile bits). One 32 KB LDS tile per
-// block applies the whole low ladder — colMax+1 columns for one global
-// state read+write. tb = 12 (fp32) / 11 (fp64).
-template <typename R, bool PRE>
-__global__ void k_qft_low_lds(cplx<R>* sv, bitCapInt nTiles, int tb, int colMax, R piSign)
+// K-column orbit apply over 16 register-resident amplitudes — the same math
+// as k_qft_colK (one f0 plus constant roots of unity drive every ramp), as a
+// building block for the LDS-staged ladder kernels. v[b] is the amplitude
+// whose group-column bits spell b; U[d] = e^{i·sign·π/2^d}; f0 encodes the
+// below-group ramp contribution at the group's HIGHEST column scale.
+template <typename R, int K, bool PRE>
+__device__ __forceinline__ void qftOrbitColumns(cplx<R>* v, R iSign, cplx<R> f0)
 {
+    const R s = (R)0.70710678118654752440;
+    constexpr int NS = 1 << K;
+    // U(d) = e^{i·sign·π/2^d} — compile-time roots of unity (indices below
+    // are constants after full unrolling, so these fold into immediates)
+    const cplx<R> U[4] = { cplx<R>{ (R)1, (R)0 }, cplx<R>{ (R)0, iSign },
+        cplx<R>{ (R)0.70710678118654752440, iSign * (R)0.70710678118654752440 },
+        cplx<R>{ (R)0.92387953251128675613, iSign * (R)0.38268343236508977173 } };
+    cplx<R> fPow[K];
+    fPow[0] = f0;
+#pragma unroll
+    for (int m = 1; m < K; ++m) fPow[m] = fPow[m - 1] * fPow[m - 1];
+    auto column = [&](int c) {
+#pragma unroll
+        for (int b = 0; b < NS; ++b) {
+            if (b & (1 << c)) continue;
+            const int hb = b | (1 << c);
+            const cplx<R> t = s * (v[b] + v[hb]);
+            const cplx<R> u = s * (v[b] - v[hb]);
+            v[b] = t;
+            v[hb] = u;
+        }
+#pragma unroll
+        for (int b = 0; b < NS; ++b) {
+            if (!(b & (1 << c))) continue;
+            cplx<R> f = fPow[K - 1 - c];
+#pragma unroll
+            for (int j = 0; j < K; ++j) {
+                if (j < c && (b & (1 << j))) f = f * U[c - j];
+            }
+            v[b] = f * v[b];
+        }
+    };
+    auto columnInv = [&](int c) {
+#pragma unroll
+        for (int b = 0; b < NS; ++b) {
+            if (!(b & (1 << c))) continue;
+            cplx<R> f = fPow[K - 1 - c];
+#pragma unroll
+            for (int j = 0; j < K; ++j) {
+                if (j < c && (b & (1 << j))) f = f * U[c - j];
+            }
+            v[b] = f * v[b];
+        }
+#pragma unroll
+        for (int b = 0; b < NS; ++b) {
+            if (b & (1 << c)) continue;
+            const int hb = b | (1 << c);
+            const cplx<R> t = s * (v[b] + v[hb]);
+            const cplx<R> u = s * (v[b] - v[hb]);
+            v[b] = t;
+            v[hb] = u;
+        }
+    };
+    if (!PRE) {
+#pragma unroll
+        for (int c = K - 1; c >= 0; --c) column(c);
+    } else {
+#pragma unroll
+        for (int c = 0; c < K; ++c) columnInv(c);
+    }
+}
+
+// LDS bank swizzle for the low-ladder kernel: XOR index bits 4..7 into
+// bits 0..3 so every access pattern the orbit groups generate (element
+// strides 1, 16, 256 within a wave) lands on 16 distinct bank pairs per
+// 16 consecutive lanes — the optimum for 8/16-byte LDS elements.
+__device__ __forceinline__ int qaSwz(int j) { return j ^ ((j >> 4) & 15); }
+
+// One group of K columns applied over a 2^tb-amplitude tile as 16-amp
+// register orbits. K is a template parameter so v[] stays in registers
+// (runtime K forces the array to scratch — measured 2.5x slower). Groups
+// with gLo >= 6 may read/write HBM directly (lane-contiguous within the
+// orbit slot loops); gLo < 6 groups always go through (swizzled) LDS.
+template <typename R, int K, bool PRE>
+__device__ __forceinline__ void qftLowGroup(cplx<R>* svBase, cplx<R>* lds, int tile, int gLo,
+    R scaleHi, R iSign, bool fromGlobal, bool toGlobal)
+{
+    constexpr int NS = 1 << K;
+    const int orbitCount = tile >> K;
+    for (int o = threadIdx.x; o < orbitCount; o += blockDim.x) {
+        const int below = o & ((1 << gLo) - 1);
+        const int r = ((o >> gLo) << (gLo + K)) | below;
+        cplx<R> v[NS];
+        if (fromGlobal) {
+#pragma unroll
+            for (int b = 0; b < NS; ++b) v[b] = svBase[r | (b << gLo)];
+        } else {
+#pragma unroll
+            for (int b = 0; b < NS; ++b) v[b] = lds[qaSwz(r | (b << gLo))];
+        }
+        R sn, cs;
+        devSinCos<R>(scaleHi * (R)below, &sn, &cs);
+        qftOrbitColumns<R, K, PRE>(v, iSign, cplx<R>{ cs, sn });
+        if (toGlobal) {
+#pragma unroll
+            for (int b = 0; b < NS; ++b) svBase[r | (b << gLo)] = v[b];
+        } else {
+#pragma unroll
+            for (int b = 0; b < NS; ++b) lds[qaSwz(r | (b << gLo))] = v[b];
+        }
+    }
+}
+
+
+
+// ALL low-bit QFT columns in ONE pass: when the register starts at bit 0,
+// columns colMax..0 act entirely inside a contiguous 2^tb-amplitude tile.
+// Columns are processed in groups of (up to) 4 as 16-amplitude register
+// orbits — the first group gathers straight from HBM, the last scatters
+// straight back, and LDS carries the tile between groups. One global
+// read+write, 2 barriers, and 1 sincos per orbit per group replace the
+// former per-pair sincos ladder. tb = 12 (fp32) / 11 (fp64).
+template <typename R, int K, bool PRE>
+__global__ void __launch_bounds__(256, 3)
+    k_qft_low_lds(cplx<R>* sv, bitCapInt nTiles, int tb, int colMax, R piSign)
+{
+    // REQUIRES colMax+1 to be a multiple of K (the engine aligns the ladder
+    // length; K = 4 fp32 / 3 fp64) so every group is the single kernel-wide
+    // instantiation — keeps the orbit registers allocated without spills.
     extern __shared__ unsigned char qa_lds_raw[];
     cplx<R>* lds = reinterpret_cast<cplx<R>*>(qa_lds_raw);
     const int tile = 1 << tb;
-    const int pairs = tile >> 1;
+    const int nCols = colMax + 1;
+    const int nG = nCols / K;
+    const R iSign = (piSign >= 0) ? (R)1 : (R)-1;
+    // a group may touch HBM directly only if its orbit slot loops are
+    // lane-contiguous, i.e. gLo >= 6; otherwise a contiguous LDS<->HBM
+    // copy brackets the ladder on that side
+    const int firstG = PRE ? 0 : (nG - 1);
+    const int lastG = PRE ? (nG - 1) : 0;
+    const bool copyIn = (K * firstG) < 6;
+    const bool copyOut = (K * lastG) < 6;
     for (bitCapInt t = blockIdx.x; t < nTiles; t += gridDim.x) {
-        const bitCapInt base = t << tb;
-        for (int j = threadIdx.x; j < tile; j += blockDim.x) lds[j] = sv[base + j];
-        __syncthreads();
-        const R s = (R)0.70710678118654752440;
-        if (!PRE) {
-            for (int col = colMax; col >= 0; --col) {
-                const int half = 1 << col;
-                const R scale = piSign / (R)half;
-                for (int k = threadIdx.x; k < pairs; k += blockDim.x) {
-                    const int lo = ((k >> col) << (col + 1)) | (k & (half - 1));
-                    const int hi = lo | half;
-                    const cplx<R> x = lds[lo], y = lds[hi];
-                    R sn, cs;
-                    devSinCos<R>(scale * (R)(lo & (half - 1)), &sn, &cs);
-                    const cplx<R> f{ cs, sn };
-                    lds[lo] = s * (x + y);
-                    lds[hi] = f * (s * (x - y));
-                }
-                __syncthreads();
-            }
-        } else {
-            for (int col = 0; col <= colMax; ++col) {
-                const int half = 1 << col;
-                const R scale = piSign / (R)half;
-                for (int k = threadIdx.x; k < pairs; k += blockDim.x) {
-                    const int lo = ((k >> col) << (col + 1)) | (k & (half - 1));
-                    const int hi = lo | half;
-                    R sn, cs;
-                    devSinCos<R>(scale * (R)(lo & (half - 1)), &sn, &cs);
-                    const cplx<R> f{ cs, sn };
-                    const cplx<R> x = lds[lo], y = f * lds[hi];
-                    lds[lo] = s * (x + y);
-                    lds[hi] = s * (x - y);
-                }
-                __syncthreads();
-            }
+        cplx<R>* svBase = sv + (t << tb);
+        if (copyIn) {
+            for (int j = threadIdx.x; j < tile; j += blockDim.x) lds[qaSwz(j)] = svBase[j];
         }
-        for (int j = threadIdx.x; j < tile; j += blockDim.x) sv[base + j] = lds[j];
+        for (int gi = 0; gi < nG; ++gi) {
+            const int g = PRE ? gi : (nG - 1 - gi);
+            const int gLo = K * g;
+            const bool fromGlobal = (gi == 0) && !copyIn;
+            const bool toGlobal = (gi == nG - 1) && !copyOut;
+            const R scaleHi = piSign / (R)(1 << (gLo + K - 1));
+            if (gi != 0 || copyIn) __syncthreads();
+            qftLowGroup<R, K, PRE>(svBase, lds, tile, gLo, scaleHi, iSign, fromGlobal, toGlobal);
+        }
+        if (copyOut) {
+            __syncthreads();
+            for (int j = threadIdx.x; j < tile; j += blockDim.x) svBase[j] = lds[qaSwz(j)];
+        }
         __syncthreads();
     }
 }
@@ -1858,82 +1965,97 @@ template <typename R>
 void launchQftLowLds(
     cplx<R>* sv, bitCapInt maxQPower, int tb, int colMax, int sign, bool pre, hipStream_t stream)
 {
+    constexpr int K = qaLowLadderK<R>();
     const bitCapInt nTiles = maxQPower >> tb;
     const size_t ldsBytes = (size_t(1) << tb) * sizeof(cplx<R>);
     const int grid = (int)std::min<bitCapInt>(nTiles, (bitCapInt)QA_REDUCE_MAX_BLOCKS);
     const R piSign = (R)sign * (R)3.14159265358979323846;
     if (pre) {
-        hipLaunchKernelGGL((k_qft_low_lds<R, true>), dim3(grid), dim3(QA_BLOCK), ldsBytes, stream,
-            sv, nTiles, tb, colMax, piSign);
+        hipLaunchKernelGGL((k_qft_low_lds<R, K, true>), dim3(grid), dim3(QA_BLOCK), ldsBytes,
+            stream, sv, nTiles, tb, colMax, piSign);
     } else {
-        hipLaunchKernelGGL((k_qft_low_lds<R, false>), dim3(grid), dim3(QA_BLOCK), ldsBytes,
+        hipLaunchKernelGGL((k_qft_low_lds<R, K, false>), dim3(grid), dim3(QA_BLOCK), ldsBytes,
             stream, sv, nTiles, tb, colMax, piSign);
     }
 }
 
-// SIX mid-range QFT columns per pass through a 2D LDS tile: 64 contiguous
-// low amplitudes (coalesced 512 B runs) x 2^nCols column-bit combinations.
-// Ramp for column col: theta = scale_col * (x mod 2^col) where
-// x mod 2^col = low6 + midFixed + cbLow*2^colLo — low6 varies in-tile,
-// midFixed (bits 6..colLo-1) is tile-constant, cbLow is the tile's lower
-// column bits. Requires a start-0 register and colLo >= 6.
-template <typename R, bool PRE>
-__global__ void k_qft_mid_lds(cplx<R>* sv, bitCapInt nTiles, int colLo, int nCols, R piSign)
+// One group of K mid columns over a 64 x 2^nCols 2D tile as 16-amp register
+// orbits (K templated — see qftLowGroup). Every access, LDS or HBM, keeps
+// the 64-value `low` dimension in the lane index, so loads/stores are
+// 512 B-contiguous per slot and LDS is bank-conflict-free without swizzle.
+template <typename R, int K, bool PRE>
+__device__ __forceinline__ void qftMidGroup(cplx<R>* sv, cplx<R>* lds, bitCapInt xBase,
+    int colLo, int tileAmps, int gLo, R midFixed, R hbScale, R scaleHi, R iSign,
+    bool fromGlobal, bool toGlobal)
 {
+    constexpr int NS = 1 << K;
+    const int orbitCount = tileAmps >> K;
+    for (int o = threadIdx.x; o < orbitCount; o += blockDim.x) {
+        const int low = o & 63;
+        const int cbr = o >> 6;
+        const int cbBelow = cbr & ((1 << gLo) - 1);
+        const int cb0 = ((cbr >> gLo) << (gLo + K)) | cbBelow;
+        cplx<R> v[NS];
+        if (fromGlobal) {
+#pragma unroll
+            for (int b = 0; b < NS; ++b) {
+                v[b] = sv[xBase | ((bitCapInt)(cb0 | (b << gLo)) << colLo) | (bitCapInt)low];
+            }
+        } else {
+#pragma unroll
+            for (int b = 0; b < NS; ++b) v[b] = lds[((cb0 | (b << gLo)) << 6) | low];
+        }
+        R sn, cs;
+        devSinCos<R>(scaleHi * ((R)low + midFixed + hbScale * (R)cbBelow), &sn, &cs);
+        qftOrbitColumns<R, K, PRE>(v, iSign, cplx<R>{ cs, sn });
+        if (toGlobal) {
+#pragma unroll
+            for (int b = 0; b < NS; ++b) {
+                sv[xBase | ((bitCapInt)(cb0 | (b << gLo)) << colLo) | (bitCapInt)low] = v[b];
+            }
+        } else {
+#pragma unroll
+            for (int b = 0; b < NS; ++b) lds[((cb0 | (b << gLo)) << 6) | low] = v[b];
+        }
+    }
+}
+
+// Up to SIX mid-range QFT columns per pass through a 2D LDS tile: 64
+// contiguous low amplitudes (coalesced 512 B runs) x 2^nCols column-bit
+// combinations. Column bits are processed in groups of (up to) 4 as
+// 16-amplitude register orbits (qftOrbitColumns): the first group gathers
+// straight from HBM, the last scatters straight back, LDS carries the tile
+// between groups. Ramp for column col: theta = scale_col * (x mod 2^col)
+// where x mod 2^col = low6 + midFixed + cbBelow*2^colLo — low6 varies
+// in-tile, midFixed (bits 6..colLo-1) is tile-constant, cbBelow is the
+// tile's below-group column bits. Requires a start-0 register, colLo >= 6.
+template <typename R, int K, bool PRE>
+__global__ void __launch_bounds__(256, 3)
+    k_qft_mid_lds(cplx<R>* sv, bitCapInt nTiles, int colLo, int nCols, R piSign)
+{
+    // REQUIRES nCols to be a multiple of K (launcher picks K) so every
+    // group is the single kernel-wide instantiation — no register spills.
     extern __shared__ unsigned char qa_lds_raw2[];
     cplx<R>* lds = reinterpret_cast<cplx<R>*>(qa_lds_raw2);
     const int C = 1 << nCols;
     const int tileAmps = 64 * C;
-    const int pairs = tileAmps >> 1;
+    const int nG = nCols / K;
     const bitCapInt midMask = (ONE_BCI << (colLo - 6)) - 1u;
-    const R s = (R)0.70710678118654752440;
+    const R hbScale = (R)(uint64_t)(ONE_BCI << colLo);
+    const R iSign = (piSign >= 0) ? (R)1 : (R)-1;
     for (bitCapInt t = blockIdx.x; t < nTiles; t += gridDim.x) {
         const bitCapInt xBase =
             ((t >> (colLo - 6)) << (colLo + nCols)) | ((t & midMask) << 6);
         const R midFixed = (R)(uint64_t)((t & midMask) << 6);
-        // load: row cb = 64 contiguous amps
-        for (int j = threadIdx.x; j < tileAmps; j += blockDim.x) {
-            const int cb = j >> 6;
-            const int low = j & 63;
-            lds[j] = sv[xBase | ((bitCapInt)cb << colLo) | (bitCapInt)low];
-        }
-        __syncthreads();
-        // columns colLo+nCols-1 .. colLo (forward) or reverse (PRE)
-        for (int step = 0; step < nCols; ++step) {
-            const int c = PRE ? step : (nCols - 1 - step);
-            const int col = colLo + c;
-            const int cbHalf = 1 << c; // pair stride in cb space
-            const R scale = piSign / (R)(ONE_BCI << col);
-            const R colScale = scale * (R)(ONE_BCI << colLo);
-            for (int k = threadIdx.x; k < pairs; k += blockDim.x) {
-                // pair index k over (cb-without-bit-c, low)
-                const int low = k & 63;
-                const int cbr = k >> 6;
-                const int cb0 = ((cbr >> c) << (c + 1)) | (cbr & (cbHalf - 1));
-                const int lo = (cb0 << 6) | low;
-                const int hi = ((cb0 | cbHalf) << 6) | low;
-                const int cbLow = cb0 & (cbHalf - 1);
-                R sn, cs;
-                devSinCos<R>(
-                    scale * ((R)low + midFixed) + colScale * (R)cbLow, &sn, &cs);
-                const cplx<R> f{ cs, sn };
-                const cplx<R> x = lds[lo];
-                if (!PRE) {
-                    const cplx<R> y = lds[hi];
-                    lds[lo] = s * (x + y);
-                    lds[hi] = f * (s * (x - y));
-                } else {
-                    const cplx<R> y = f * lds[hi];
-                    lds[lo] = s * (x + y);
-                    lds[hi] = s * (x - y);
-                }
-            }
-            __syncthreads();
-        }
-        for (int j = threadIdx.x; j < tileAmps; j += blockDim.x) {
-            const int cb = j >> 6;
-            const int low = j & 63;
-            sv[xBase | ((bitCapInt)cb << colLo) | (bitCapInt)low] = lds[j];
+        for (int gi = 0; gi < nG; ++gi) {
+            const int g = PRE ? gi : (nG - 1 - gi);
+            const int gLo = K * g;
+            const bool fromGlobal = (gi == 0);
+            const bool toGlobal = (gi == nG - 1);
+            const R scaleHi = piSign / (R)(ONE_BCI << (colLo + gLo + K - 1));
+            if (gi != 0) __syncthreads();
+            qftMidGroup<R, K, PRE>(sv, lds, xBase, colLo, tileAmps, gLo, midFixed, hbScale,
+                scaleHi, iSign, fromGlobal, toGlobal);
         }
         __syncthreads();
     }
@@ -1947,12 +2069,32 @@ void launchQftMidLds(
     const size_t ldsBytes = (size_t(64) << nCols) * sizeof(cplx<R>);
     const int grid = (int)std::min<bitCapInt>(nTiles, (bitCapInt)QA_REDUCE_MAX_BLOCKS);
     const R piSign = (R)sign * (R)3.14159265358979323846;
-    if (pre) {
-        hipLaunchKernelGGL((k_qft_mid_lds<R, true>), dim3(grid), dim3(QA_BLOCK), ldsBytes, stream,
-            sv, nTiles, colLo, nCols, piSign);
+    // uniform group width: 4 | nCols -> K=4, else 3 | nCols -> K=3 (the
+    // engine only requests nCols in {4, 6}; other multiples also work)
+    if ((nCols & 3) == 0) {
+        if (pre) {
+            hipLaunchKernelGGL((k_qft_mid_lds<R, 4, true>), dim3(grid), dim3(QA_BLOCK), ldsBytes,
+                stream, sv, nTiles, colLo, nCols, piSign);
+        } else {
+            hipLaunchKernelGGL((k_qft_mid_lds<R, 4, false>), dim3(grid), dim3(QA_BLOCK), ldsBytes,
+                stream, sv, nTiles, colLo, nCols, piSign);
+        }
+    } else if ((nCols % 3) == 0) {
+        if (pre) {
+            hipLaunchKernelGGL((k_qft_mid_lds<R, 3, true>), dim3(grid), dim3(QA_BLOCK), ldsBytes,
+                stream, sv, nTiles, colLo, nCols, piSign);
+        } else {
+            hipLaunchKernelGGL((k_qft_mid_lds<R, 3, false>), dim3(grid), dim3(QA_BLOCK), ldsBytes,
+                stream, sv, nTiles, colLo, nCols, piSign);
+        }
     } else {
-        hipLaunchKernelGGL((k_qft_mid_lds<R, false>), dim3(grid), dim3(QA_BLOCK), ldsBytes,
-            stream, sv, nTiles, colLo, nCols, piSign);
+        if (pre) {
+            hipLaunchKernelGGL((k_qft_mid_lds<R, 2, true>), dim3(grid), dim3(QA_BLOCK), ldsBytes,
+                stream, sv, nTiles, colLo, nCols, piSign);
+        } else {
+            hipLaunchKernelGGL((k_qft_mid_lds<R, 2, false>), dim3(grid), dim3(QA_BLOCK), ldsBytes,
+                stream, sv, nTiles, colLo, nCols, piSign);
+        }
     }
 }
 

@@ -189,8 +189,13 @@ template <typename R>
 void launchQftColumn2(cplx<R>* sv, bitCapInt maxQPower, bitLenInt rampStart, bitLenInt col,
     bitCapInt tHi, bitCapInt tLo, int sign, bool pre, hipStream_t stream);
 
+// ladder group width: uniform K=4 fp32 / K=3 fp64 — the engine must align
+// low-ladder lengths to a multiple of this
+template <typename R> constexpr int qaLowLadderK() { return sizeof(R) == 4 ? 4 : 3; }
+
 // one-pass low-bit QFT ladder: columns colMax..0 applied inside contiguous
-// 2^tb-amplitude LDS tiles (start-0 registers only)
+// 2^tb-amplitude LDS tiles (start-0 registers only; colMax+1 must be a
+// multiple of qaLowLadderK<R>())
 template <typename R>
 void launchQftLowLds(
     cplx<R>* sv, bitCapInt maxQPower, int tb, int colMax, int sign, bool pre, hipStream_t stream);

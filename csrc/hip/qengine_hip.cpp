@@ -881,30 +881,60 @@ template <typename R> void QEngineHIP<R>::QFT(bitLenInt start, bitLenInt length,
         return true;
     }();
     static const bool ldsMid = []() {
+        // fp64 mid groups would need 64 KB LDS tiles (2 blocks/CU); the K4
+        // fusion path serves fp64 as well there, so mid-LDS is fp32-only
+        if (sizeof(R) != 4) return false;
         if (const char* env = std::getenv("QRACK_GPU_QFT_MIDLDS")) return std::atoi(env) != 0;
         return true;
     }();
     const int tb = qaLdsTileBits<R>();
-    // mid-LDS group width: 64 x 2^G tile (fp32 32 KB, fp64 32 KB at G=5)
-    const int G = sizeof(R) == 4 ? 6 : 5;
     bitLenInt i = length;
     while (i > 0) {
         const bitLenInt col = i - 1u;
         if (ldsLow && start == 0u && (bitLenInt)i <= (bitLenInt)tb &&
             maxQPower >= (ONE_BCI << tb)) {
-            // the whole remaining low ladder in ONE LDS-tiled pass
-            HipProfScope prof("qft_low_lds", stream);
-            launchQftLowLds<R>(dState, maxQPower, tb, (int)col, +1, false, stream);
-            break;
+            const bitLenInt AK = (bitLenInt)qaLowLadderK<R>();
+            const bitLenInt r = i % AK;
+            if (!r) {
+                // the whole remaining (K-aligned) ladder in ONE LDS pass of
+                // uniform register-orbit groups
+                HipProfScope prof("qft_low_lds", stream);
+                launchQftLowLds<R>(dState, maxQPower, tb, (int)col, +1, false, stream);
+                break;
+            }
+            if (i != r) {
+                // peel the top r columns so the ladder below is 4-aligned
+                if (r == 3u) {
+                    HipProfScope prof("qft_column3", stream);
+                    launchQftColumn3<R>(dState, maxQPower, start, col, pow2(start + col),
+                        pow2(start + col - 1u), pow2(start + col - 2u), +1, false, stream);
+                } else if (r == 2u) {
+                    HipProfScope prof("qft_column2", stream);
+                    launchQftColumn2<R>(dState, maxQPower, start, col, pow2(start + col),
+                        pow2(start + col - 1u), +1, false, stream);
+                } else {
+                    HipProfScope prof("qft_column", stream);
+                    launchQftColumn<R>(
+                        dState, maxQPower, start, col, pow2(start + col), +1, false, stream);
+                }
+                i -= r;
+                continue;
+            }
+            // i < 4: the small fused kernels below finish the register
         }
         if (ldsLow && ldsMid && start == 0u && (bitLenInt)i > (bitLenInt)tb &&
             maxQPower >= (ONE_BCI << tb)) {
-            // up to G mid columns per pass through a 2D LDS tile
-            const int nc = std::min<int>(G, (int)(i - (bitLenInt)tb));
-            HipProfScope prof("qft_mid_lds", stream);
-            launchQftMidLds<R>(dState, maxQPower, (int)i - nc, nc, +1, false, stream);
-            i -= (bitLenInt)nc;
-            continue;
+            // mid columns per pass through a 2D LDS tile (6 -> two K=3
+            // groups; trailing 4/3/2 -> one uniform group)
+            const int rem = (int)(i - (bitLenInt)tb);
+            const int nc = rem >= 6 ? 6 : (rem == 5 ? 4 : rem);
+            if (nc >= 2) {
+                HipProfScope prof("qft_mid_lds", stream);
+                launchQftMidLds<R>(dState, maxQPower, (int)i - nc, nc, +1, false, stream);
+                i -= (bitLenInt)nc;
+                continue;
+            }
+            // rem == 1: the plain fused kernels below handle it
         }
         if (fuseMax >= 5 && col >= 4u && maxQPower >= 64u) {
             HipProfScope prof("qft_column5", stream);
@@ -964,26 +994,34 @@ template <typename R> void QEngineHIP<R>::IQFT(bitLenInt start, bitLenInt length
         return true;
     }();
     static const bool ldsMid = []() {
+        if (sizeof(R) != 4) return false;
         if (const char* env = std::getenv("QRACK_GPU_QFT_MIDLDS")) return std::atoi(env) != 0;
         return true;
     }();
     const int tb = qaLdsTileBits<R>();
-    const int G = sizeof(R) == 4 ? 6 : 5;
     bitLenInt i = 0;
     if (ldsLow && start == 0u && length > 0u && maxQPower >= (ONE_BCI << tb)) {
-        const bitLenInt colMax = std::min<bitLenInt>(length - 1u, (bitLenInt)tb - 1u);
-        HipProfScope prof("qft_low_lds", stream);
-        launchQftLowLds<R>(dState, maxQPower, tb, (int)colMax, -1, true, stream);
-        i = colMax + 1u;
+        // the bottom K-aligned stretch of the ladder in ONE LDS pass; the
+        // few columns above it ride the ascending fusion below
+        const bitLenInt AK = (bitLenInt)qaLowLadderK<R>();
+        const bitLenInt L = (std::min<bitLenInt>(length, (bitLenInt)tb) / AK) * AK;
+        if (L >= AK) {
+            HipProfScope prof("qft_low_lds", stream);
+            launchQftLowLds<R>(dState, maxQPower, tb, (int)L - 1, -1, true, stream);
+            i = L;
+        }
     }
     while (i < length) {
-        if (ldsLow && ldsMid && start == 0u && i >= (bitLenInt)tb &&
+        if (ldsLow && ldsMid && start == 0u && i >= (bitLenInt)tb && i < length &&
             maxQPower >= (ONE_BCI << tb)) {
-            const int nc = std::min<int>(G, (int)(length - i));
-            HipProfScope prof("qft_mid_lds", stream);
-            launchQftMidLds<R>(dState, maxQPower, (int)i, nc, -1, true, stream);
-            i += (bitLenInt)nc;
-            continue;
+            const int rem = (int)(length - i);
+            const int nc = rem >= 6 ? 6 : (rem == 5 ? 4 : rem);
+            if (nc >= 2) {
+                HipProfScope prof("qft_mid_lds", stream);
+                launchQftMidLds<R>(dState, maxQPower, (int)i, nc, -1, true, stream);
+                i += (bitLenInt)nc;
+                continue;
+            }
         }
         if (fuseMax >= 5 && (i + 4u) < length && maxQPower >= 64u) {
             HipProfScope prof("qft_column5", stream);
