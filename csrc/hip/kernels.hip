@@ -1251,7 +1251,7 @@ void launchQftColumnGeneral(cplx<R>* sv, bitCapInt maxQPower, bitCapInt tPow, co
 // gates' four bit positions, applies gate A across its axis pair then
 // gate B across the other — one state read+write for two SU(4)s.
 template <typename R>
-__global__ void k_mtrx2q_pair2(cplx<R>* sv, Gate4x4Pair2Args<R> a)
+__global__ void __launch_bounds__(256, 3) k_mtrx2q_pair2(cplx<R>* sv, Gate4x4Pair2Args<R> a)
 {
     const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
     for (bitCapInt k = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; k < a.orbits;
@@ -2331,7 +2331,9 @@ void launchQftColumn(cplx<R>* sv, bitCapInt maxQPower, bitLenInt rampStart, bitL
 // k distinct-target 2x2s in ONE pass: each lane owns a 2^k-amplitude orbit in
 // registers (k <= 5 fp32 / 4 fp64), so k memory-bound passes collapse to one.
 
-template <typename R, int K> __global__ void k_mtrx_batch(cplx<R>* sv, Batch1qArgs<R> a)
+template <typename R, int K>
+__global__ void __launch_bounds__(256, K >= 5 ? 2 : (K >= 3 ? 3 : 4))
+    k_mtrx_batch(cplx<R>* sv, Batch1qArgs<R> a)
 {
     const bitCapInt stride = (bitCapInt)gridDim.x * blockDim.x;
     for (bitCapInt j = (bitCapInt)blockIdx.x * blockDim.x + threadIdx.x; j < a.maxI; j += stride) {
@@ -2373,7 +2375,9 @@ template <typename R, int K> __global__ void k_mtrx_batch(cplx<R>* sv, Batch1qAr
 
 // fp32 float4 variant (requires tPow[0] >= 2): each lane handles TWO adjacent
 // orbits via 16 B loads/stores — full-line HBM streams like k_apply2x2_1v.
-template <int K> __global__ void k_mtrx_batch_v(cplx<float>* sv, Batch1qArgs<float> a)
+template <int K>
+__global__ void __launch_bounds__(256, K >= 4 ? 2 : (K >= 3 ? 3 : 4))
+    k_mtrx_batch_v(cplx<float>* sv, Batch1qArgs<float> a)
 {
     float4* sv4 = reinterpret_cast<float4*>(sv);
     const bitCapInt half = a.maxI >> 1u;
