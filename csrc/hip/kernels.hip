@@ -2491,34 +2491,73 @@ __global__ void k_apply2x2_1mfma(cplx<float>* sv, GateArgs<float> a)
 // then writes the tile back: ONE global RMW pass for up to 12 fused gates,
 // with bit-0 targets handled as cheaply as any other.
 
+// Gates are applied in GROUPS of qaLdsBatchK<R>() (4 fp32 / 3 fp64) as
+// 2^K-amplitude register orbits: one LDS read + butterfly chain in
+// registers + one LDS write per amp per group, instead of a pair-RMW
+// through LDS per GATE. The engine pads a.k to a multiple of K with
+// identity gates on spare tile bits and sorts each group's targets
+// ascending (1q gates on distinct targets commute, so order is free).
 template <typename R> __global__ void k_mtrx_batch_lds(cplx<R>* sv, BatchLdsArgs<R> a)
 {
     constexpr int TB = qaLdsTileBits<R>();
+    constexpr int K = qaLdsBatchK<R>();
+    constexpr int NS = 1 << K;
     __shared__ cplx<R> tile[1u << TB];
     constexpr unsigned TILE = 1u << TB;
     const bitCapInt nTiles = a.maxQPower >> TB;
+    const int nG = a.k / K;
     for (bitCapInt t = blockIdx.x; t < nTiles; t += gridDim.x) {
         const bitCapInt base = t << TB;
         for (unsigned i = threadIdx.x; i < TILE; i += blockDim.x) {
-            tile[i] = sv[base + i];
+            tile[qaSwz((int)i)] = sv[base + i];
         }
         __syncthreads();
-        for (int g = 0; g < a.k; ++g) {
-            const unsigned tp = (unsigned)a.tPow[g];
-            const cplx<R> m0 = a.m[4 * g], m1 = a.m[4 * g + 1], m2 = a.m[4 * g + 2],
-                          m3 = a.m[4 * g + 3];
-            for (unsigned j = threadIdx.x; j < (TILE >> 1); j += blockDim.x) {
-                const unsigned i0 = ((j & ~(tp - 1u)) << 1u) | (j & (tp - 1u));
-                const unsigned i1 = i0 | tp;
-                const cplx<R> x = tile[i0];
-                const cplx<R> y = tile[i1];
-                tile[i0] = m0 * x + m1 * y;
-                tile[i1] = m2 * x + m3 * y;
+        for (int gi = 0; gi < nG; ++gi) {
+            const int g0 = gi * K;
+            unsigned p[K];
+#pragma unroll
+            for (int g = 0; g < K; ++g) p[g] = (unsigned)a.tPow[g0 + g];
+            for (unsigned o = threadIdx.x; o < (TILE >> K); o += blockDim.x) {
+                unsigned r = o;
+#pragma unroll
+                for (int g = 0; g < K; ++g) r = ((r & ~(p[g] - 1u)) << 1u) | (r & (p[g] - 1u));
+                cplx<R> v[NS];
+#pragma unroll
+                for (int s = 0; s < NS; ++s) {
+                    unsigned off = r;
+#pragma unroll
+                    for (int g = 0; g < K; ++g) {
+                        if (s & (1 << g)) off |= p[g];
+                    }
+                    v[s] = tile[qaSwz((int)off)];
+                }
+#pragma unroll
+                for (int g = 0; g < K; ++g) {
+                    const cplx<R> m0 = a.m[4 * (g0 + g)], m1 = a.m[4 * (g0 + g) + 1],
+                                  m2 = a.m[4 * (g0 + g) + 2], m3 = a.m[4 * (g0 + g) + 3];
+#pragma unroll
+                    for (int s = 0; s < NS; ++s) {
+                        if (s & (1 << g)) continue;
+                        const int hb = s | (1 << g);
+                        const cplx<R> x = v[s], y = v[hb];
+                        v[s] = m0 * x + m1 * y;
+                        v[hb] = m2 * x + m3 * y;
+                    }
+                }
+#pragma unroll
+                for (int s = 0; s < NS; ++s) {
+                    unsigned off = r;
+#pragma unroll
+                    for (int g = 0; g < K; ++g) {
+                        if (s & (1 << g)) off |= p[g];
+                    }
+                    tile[qaSwz((int)off)] = v[s];
+                }
             }
             __syncthreads();
         }
         for (unsigned i = threadIdx.x; i < TILE; i += blockDim.x) {
-            sv[base + i] = tile[i];
+            sv[base + i] = tile[qaSwz((int)i)];
         }
         __syncthreads();
     }

@@ -266,6 +266,9 @@ void launchMtrx1qBatch(cplx<R>* sv, const Batch1qArgs<R>& a, hipStream_t stream)
 // 32 KB LDS tiles both ways: 4096 fp32 amps / 2048 fp64 amps (a 64 KB fp64
 // tile would sit exactly on the per-workgroup LDS limit)
 template <typename R> constexpr int qaLdsTileBits() { return sizeof(R) == 4 ? 12 : 11; }
+// register-orbit group width inside the LDS gate-batch kernel; the engine
+// pads batches to a multiple of this with identity gates
+template <typename R> constexpr int qaLdsBatchK() { return sizeof(R) == 4 ? 4 : 3; }
 constexpr int QA_LDS_TILE_BITS = 12; // fp32 value; use qaLdsTileBits<R>()
 constexpr int QA_MAX_BATCH_LDS = 12;
 

@@ -687,7 +687,36 @@ void QEngineHIP<R>::Mtrx1qBatch(
             a.tPow[g] = pow2(targets[low[i + g]]);
             for (int e = 0; e < 4; ++e) a.m[4u * g + e] = mtrxs[4u * low[i + g] + e];
         }
-        a.k = (int)k;
+        // pad to a multiple of the kernel's register-orbit group width with
+        // identity gates on spare tile bits (distinct within the last group)
+        const size_t KG = (size_t)qaLdsBatchK<R>();
+        size_t kp = k;
+        while (kp % KG) {
+            const size_t groupStart = (kp / KG) * KG;
+            bitCapInt used = 0;
+            for (size_t g = groupStart; g < kp; ++g) used |= a.tPow[g];
+            bitCapInt pad = 1u;
+            while (used & pad) pad <<= 1u;
+            a.tPow[kp] = pad;
+            a.m[4u * kp] = cplx<R>{ (R)1, (R)0 };
+            a.m[4u * kp + 1u] = cplx<R>{ (R)0, (R)0 };
+            a.m[4u * kp + 2u] = cplx<R>{ (R)0, (R)0 };
+            a.m[4u * kp + 3u] = cplx<R>{ (R)1, (R)0 };
+            ++kp;
+        }
+        // each group's targets sorted ascending (required by the kernel's
+        // zero-bit insertion; 1q gates on distinct targets commute)
+        for (size_t g0 = 0; g0 < kp; g0 += KG) {
+            for (size_t x = g0; x < g0 + KG; ++x) {
+                for (size_t y = x + 1u; y < g0 + KG; ++y) {
+                    if (a.tPow[y] < a.tPow[x]) {
+                        std::swap(a.tPow[x], a.tPow[y]);
+                        for (int e = 0; e < 4; ++e) std::swap(a.m[4u * x + e], a.m[4u * y + e]);
+                    }
+                }
+            }
+        }
+        a.k = (int)kp;
         a.maxQPower = maxQPower;
         HipProfScope prof("mtrx_1q_batch_lds", stream);
         launchMtrx1qBatchLds<R>(dState, a, stream);
