@@ -629,9 +629,12 @@ class DistQPager:
     def qft(self, start, length):
         i = length - 1
         while i >= 0:
-            if i < 12 and self._identity_low(start, i):
-                # the whole remaining low ladder rides the engine's
-                # LDS-tiled one-pass kernel (identity map, no meta bits)
+            if self._identity_low(start, i):
+                # ALL remaining columns are identity-mapped local slots —
+                # their ramps use only bits < col, which are the same local
+                # bits — so the engine's own fused multi-column/LDS ladder
+                # finishes the whole register in one call. After the meta
+                # columns of an N-rank QFT this is the entire local part.
                 self.q.qft(0, i + 1)
                 return
             if i >= 1 and self._try_pair_local(start, i, +1, False):
@@ -643,7 +646,7 @@ class DistQPager:
     def iqft(self, start, length):
         i = 0
         if length > 0:
-            k = min(length, 12)
+            k = min(length, self.qpp)
             while k > 0 and not self._identity_low(start, k - 1):
                 k -= 1
             if k >= 2:
