@@ -399,6 +399,14 @@ template <typename R> void QPager<R>::QFT(bitLenInt start, bitLenInt length, boo
     const R s = SQRT1_2_R<R>;
     const cplx<R> h[4] = { { s, 0 }, { s, 0 }, { s, 0 }, { -s, 0 } };
     for (bitLenInt i = length; i-- > 0;) {
+        if (start == 0u && (start + i) < qpp) {
+            // every remaining column is intra-page on an identity layout —
+            // each page runs the engine's fused multi-column/LDS ladder for
+            // the whole rest of the register in ONE call (the columns' ramp
+            // bits are all below qpp, so pages are independent)
+            for (auto& p : qPages) p->QFT(0u, i + 1u);
+            return;
+        }
         DispatchGate(h, start + i, {}, 0u);
         if (!i) continue;
         const bitLenInt t = start + i;
@@ -430,7 +438,16 @@ template <typename R> void QPager<R>::IQFT(bitLenInt start, bitLenInt length, bo
     if (!length) return;
     const R s = SQRT1_2_R<R>;
     const cplx<R> h[4] = { { s, 0 }, { s, 0 }, { s, 0 }, { -s, 0 } };
-    for (bitLenInt i = 0; i < length; ++i) {
+    bitLenInt i0 = 0;
+    if (start == 0u) {
+        // the intra prefix of the ladder in one fused engine call per page
+        const bitLenInt k = (length < qpp) ? length : qpp;
+        if (k) {
+            for (auto& p : qPages) p->IQFT(0u, k);
+            i0 = k;
+        }
+    }
+    for (bitLenInt i = i0; i < length; ++i) {
         if (i) {
             const bitLenInt t = start + i;
             const R scale = -PI_R<R> / (R)pow2(i);

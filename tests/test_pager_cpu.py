@@ -178,3 +178,27 @@ def test_anti_meta_controlled_invert():
     q2.macinvert([0], 1, 1, n - 1)  # anti-control intra (off) on meta target: no-op
     cp2.macinvert([0], 1, 1, n - 1)
     assert_states_close(q2.get_state_vector(), cp2.get_state_vector(), 1e-6)
+
+
+@pytest.mark.parametrize("n,pages", [(8, 2), (10, 4), (12, 8)])
+def test_qft_identity_prefix_shortcut_matches_single_engine(n, pages):
+    """QPager::QFT's identity-prefix shortcut (all intra columns as one
+    fused per-page engine call) must match the single engine for forward,
+    inverse, and offset registers (csrc/qpager.cpp QFT/IQFT)."""
+    p = make_paged(n, pages, seed=3)
+    r = make_cpu(n, seed=3)
+    init = 0x2D & ((1 << n) - 1)
+    p.set_permutation(init)
+    r.set_permutation(init)
+    for q in range(0, n, 4):
+        p.ry(0.3 + q, q)
+        r.ry(0.3 + q, q)
+    p.qft(0, n)
+    r.qft(0, n)
+    assert_states_close(p.get_state_vector(), r.get_state_vector(), 1e-5)
+    p.iqft(0, n)
+    r.iqft(0, n)
+    assert_states_close(p.get_state_vector(), r.get_state_vector(), 1e-5)
+    p.qft(2, n - 3)  # offset register: shortcut must NOT fire mid-ladder
+    r.qft(2, n - 3)
+    assert_states_close(p.get_state_vector(), r.get_state_vector(), 1e-5)
