@@ -574,6 +574,26 @@ def test_qft_fused2_numerics_vs_cpu():
     assert_states_close(q.get_state_vector(), cp.get_state_vector(), 2e-4)
 
 
+def test_pager_qft_identity_prefix_on_hip():
+    """QPager over HIP pages: the identity-prefix QFT shortcut (all intra
+    columns as one fused per-page engine ladder, csrc/qpager.cpp QFT/IQFT)
+    must match the plain HIP engine, forward and inverse."""
+    n = 16
+    p = qa.create_simulator(n, layers=["pager", "hip"], seed=9, pages_per_device=4)
+    r = make(n, seed=9)
+    p.set_permutation(0x9A3C)
+    r.set_permutation(0x9A3C)
+    for i in range(0, n, 5):
+        p.ry(0.4 + i, i)
+        r.ry(0.4 + i, i)
+    p.qft(0, n)
+    r.qft(0, n)
+    assert_states_close(p.get_state_vector(), r.get_state_vector(), 2e-4)
+    p.iqft(0, n)
+    r.iqft(0, n)
+    assert_states_close(p.get_state_vector(), r.get_state_vector(), 2e-4)
+
+
 def test_qft_mid_lds_numerics_vs_cpu():
     """The 2D-tile mid-column LDS QFT kernel (k_qft_mid_lds: fires for
     start-0 registers wider than the low-ladder tile — n > 12 fp32,
