@@ -496,10 +496,24 @@ double QInterface<R>::ExpectationFloatsFactorized(
 {
     if (weights.size() < 2u * bits.size())
         throw QrackError("ExpectationFloatsFactorized: need (w0, w1) per bit");
+    if (bits.empty()) return 1.0;
+    if (bits.size() == 1u) {
+        const double p1 = (double)Prob(bits[0]);
+        return weights[0] * (1.0 - p1) + weights[1] * p1;
+    }
+    // expectation of the PRODUCT of per-qubit weights over the joint
+    // distribution (reference semantics: qinterface.cpp:771-803 multiplies
+    // the chosen weight per bit into each basis state's value)
+    const bitCapInt outLen = pow2((bitLenInt)bits.size());
+    std::vector<double> joint(outLen);
+    ProbBitsAll(bits, joint.data());
     double e = 0;
-    for (size_t b = 0; b < bits.size(); ++b) {
-        const double p1 = (double)Prob(bits[b]);
-        e += weights[2u * b] * (1.0 - p1) + weights[2u * b + 1u] * p1;
+    for (bitCapInt p = 0; p < outLen; ++p) {
+        double w = 1.0;
+        for (size_t b = 0; b < bits.size(); ++b) {
+            w *= ((p >> b) & 1u) ? weights[2u * b + 1u] : weights[2u * b];
+        }
+        e += w * joint[p];
     }
     return e;
 }
@@ -510,18 +524,19 @@ double QInterface<R>::VarianceFloatsFactorized(
 {
     if (weights.size() < 2u * bits.size())
         throw QrackError("VarianceFloatsFactorized: need (w0, w1) per bit");
-    // the variance of a sum needs the joint distribution (cross terms)
+    if (bits.empty()) return 0.0;
+    // variance of the PRODUCT of per-qubit weights (reference semantics)
     const bitCapInt outLen = pow2((bitLenInt)bits.size());
     std::vector<double> joint(outLen);
     ProbBitsAll(bits, joint.data());
     double mean = 0, e2 = 0;
     for (bitCapInt p = 0; p < outLen; ++p) {
-        double val = 0;
+        double w = 1.0;
         for (size_t b = 0; b < bits.size(); ++b) {
-            val += ((p >> b) & 1u) ? weights[2u * b + 1u] : weights[2u * b];
+            w *= ((p >> b) & 1u) ? weights[2u * b + 1u] : weights[2u * b];
         }
-        mean += val * joint[p];
-        e2 += val * val * joint[p];
+        mean += w * joint[p];
+        e2 += w * w * joint[p];
     }
     return e2 - mean * mean;
 }
@@ -532,13 +547,24 @@ double QInterface<R>::VarianceBitsFactorized(
 {
     if (perms.size() < bits.size())
         throw QrackError("VarianceBitsFactorized: need one perm per bit");
-    std::vector<double> w(2u * bits.size());
-    for (size_t b = 0; b < bits.size(); ++b) {
-        w[2u * b] = 0.0;
-        w[2u * b + 1u] = (double)perms[b];
+    // the Bits family is ADDITIVE (a register value: reference
+    // qinterface.cpp ExpectationBitsFactorized sums the chosen perm per
+    // bit), unlike the multiplicative Floats family. The variance of the
+    // sum needs the joint distribution for the cross terms; the uniform
+    // offset shifts the value and cancels in the variance.
+    const bitCapInt outLen = pow2((bitLenInt)bits.size());
+    std::vector<double> joint(outLen);
+    ProbBitsAll(bits, joint.data());
+    double mean = 0, e2 = 0;
+    for (bitCapInt p = 0; p < outLen; ++p) {
+        double val = 0;
+        for (size_t b = 0; b < bits.size(); ++b) {
+            if ((p >> b) & 1u) val += (double)perms[b];
+        }
+        mean += val * joint[p];
+        e2 += val * val * joint[p];
     }
-    // offset shifts the value uniformly: variance is unaffected by it
-    return VarianceFloatsFactorized(bits, w);
+    return e2 - mean * mean;
 }
 
 template <typename R>
@@ -573,8 +599,8 @@ double QInterface<R>::VarianceUnitaryAll(const std::vector<bitLenInt>& bits,
     return c->VarianceFloatsFactorized(bits, w);
 }
 
-// reference-semantics SUM of single-qubit Paulis (basis-rotate a clone, then
-// factorized +1/-1 expectation; qinterface.cpp:715-769)
+// reference-semantics tensor-PRODUCT of single-qubit Paulis (basis-rotate a
+// clone, then factorized +/-1 product expectation; qinterface.cpp:715-769)
 template <typename R>
 static QInterfacePtr<R> pauliBasisClone(QInterface<R>* self, std::vector<bitLenInt>& bits,
     std::vector<Pauli>& paulis, std::vector<double>& eig)

@@ -698,11 +698,13 @@ public:
         const std::vector<bitLenInt>& bits, const std::vector<bitCapInt>& perms, bitCapInt offset = 0);
     virtual double VarianceBitsAll(const std::vector<bitLenInt>& bits, bitCapInt offset = 0);
     virtual double PauliExpectation(const std::vector<bitLenInt>& bits, const std::vector<Pauli>& paulis);
-    // NOTE: PauliExpectation above is the tensor-PRODUCT observable
-    // <P_0 x P_1 x ...> (an extension kept for product-parity use); the
-    // reference's ExpectationPauliAll / VariancePauliAll are the factorized
-    // SUM of single-qubit Paulis (qinterface.cpp:715-769) — implemented here
-    // with matching semantics.
+    // PauliExpectation is the tensor-PRODUCT observable <P_0 x P_1 x ...>
+    // computed in-place (rotate, parity, rotate back); the reference's
+    // ExpectationPauliAll / VariancePauliAll compute the SAME product
+    // observable via a basis-rotated clone + the factorized per-qubit
+    // (+1,-1) PRODUCT expectation (qinterface.cpp:715-803, where the
+    // per-basis-state value MULTIPLIES the chosen weight per bit) —
+    // implemented below with matching semantics.
     virtual double ExpectationPauliAll(
         const std::vector<bitLenInt>& bits, const std::vector<Pauli>& paulis);
     virtual double VariancePauliAll(
@@ -720,8 +722,10 @@ public:
     // same over the set bits of a mask, low to high (parity: ProbMaskAll)
     virtual void ProbMaskAll(bitCapInt mask, double* probsOut);
 
-    // per-qubit weighted observables: weights holds (w0, w1) per bit;
-    // E = sum_b w0_b P(b=0) + w1_b P(b=1) (parity: ExpectationFloatsFactorized)
+    // per-qubit weighted observables: weights holds (w0, w1) per bit; the
+    // value of basis state p is the PRODUCT over bits of the chosen weight,
+    // E = sum_p prob(p) * prod_b w_{p_b} (parity: ExpectationFloatsFactorized,
+    // reference qinterface.cpp:771-803)
     virtual double ExpectationFloatsFactorized(
         const std::vector<bitLenInt>& bits, const std::vector<double>& weights);
     // variance of the same sum — needs the JOINT distribution (cross terms)

@@ -8,7 +8,7 @@ import sys
 import pytest
 
 EXAMPLES = ["grover.py", "teleport.py", "shor.py", "qft_demo.py", "pauli_chain_evolve.py",
-            "approx_supremacy.py", "graph_state.py"]
+            "approx_supremacy.py", "graph_state.py", "vqe_h2.py"]
 EXDIR = os.path.join(os.path.dirname(os.path.dirname(os.path.abspath(__file__))), "examples")
 
 

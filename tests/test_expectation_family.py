@@ -52,7 +52,9 @@ def test_floats_factorized_expectation_and_variance():
     bits = [0, 1, 3]
     w = [0.5, -1.5, 2.0, 0.25, -3.0, 1.0]
     jp = joint(sv, bits)
-    vals = np.array([sum(w[2 * k + ((p >> k) & 1)] for k in range(3))
+    # reference semantics: each basis state's value is the PRODUCT of the
+    # chosen per-qubit weights (reference qinterface.cpp:771-803)
+    vals = np.array([np.prod([w[2 * k + ((p >> k) & 1)] for k in range(3)])
                      for p in range(8)])
     e_ref = float(np.dot(vals, jp))
     v_ref = float(np.dot(vals ** 2, jp) - e_ref ** 2)
@@ -79,7 +81,7 @@ def test_unitary_all_x_basis_matches_pauli():
     h = [s2, s2, s2, -s2]
     for bits in ([1], [0, 3]):
         e_u = q.expectation_unitary_all(bits, h * len(bits))
-        # reference semantics: SUM of single-qubit Paulis
+        # both are the product observable <X x X x ...>
         e_p = q.expectation_pauli_all(bits, [1] * len(bits))  # PauliX
         assert e_u == pytest.approx(e_p, abs=1e-9)
         v_u = q.variance_unitary_all(bits, h * len(bits))
@@ -90,15 +92,18 @@ def test_unitary_all_x_basis_matches_pauli():
         q.pauli_expectation([2], [1]), abs=1e-9)
 
 
-def test_pauli_sum_vs_product_on_bell():
+def test_pauli_product_on_bell():
     q = qa.create_simulator(2, engine="cpu", precision="fp64", seed=1)
     q.h(0)
     q.cnot(0, 1)
-    # product <Z x Z> = 1 on Bell; sum <Z_0> + <Z_1> = 0
+    # product <Z x Z> = 1 on Bell, by both the in-place parity path and the
+    # reference's basis-rotated factorized-product path
     assert q.pauli_expectation([0, 1], [2, 2]) == pytest.approx(1.0, abs=1e-9)
-    assert q.expectation_pauli_all([0, 1], [2, 2]) == pytest.approx(0.0, abs=1e-9)
-    # variance of the sum: values (+2, 0, 0, -2) with joint (1/2, 0, 0, 1/2)
-    assert q.variance_pauli_all([0, 1], [2, 2]) == pytest.approx(4.0 * 0.5 + 4.0 * 0.5 - 0.0, abs=1e-9)
+    assert q.expectation_pauli_all([0, 1], [2, 2]) == pytest.approx(1.0, abs=1e-9)
+    assert q.expectation_pauli_all([0, 1], [1, 1]) == pytest.approx(1.0, abs=1e-9)
+    assert q.expectation_pauli_all([0, 1], [3, 3]) == pytest.approx(-1.0, abs=1e-9)
+    # the product observable squares to identity: Var = 1 - E^2 = 0 here
+    assert q.variance_pauli_all([0, 1], [2, 2]) == pytest.approx(0.0, abs=1e-9)
 
 
 def test_unitary_all_custom_eigenvalues():
