@@ -24,6 +24,10 @@ struct SimSlot {
     // the slot itself alive; this mutex serializes ops on one handle, same
     // contract as the reference pinvoke per-simulator locks).
     std::mutex op;
+    // pinvoke-compat qubit-id indirection (reference shards map): identity
+    // until allocateQubit/release diverges logical ids from indices
+    bool useQidMap = false;
+    std::map<uint64_t, bitLenInt> qidMap;
     bitLenInt Qubits() const { return f ? f->GetQubitCount() : (d ? d->GetQubitCount() : 0); }
 };
 using SimSlotPtr = std::shared_ptr<SimSlot>;
@@ -56,7 +60,8 @@ template <typename F> void guarded(quid sid, F&& fn)
         fn(*s);
     } catch (const std::bad_alloc&) {
         s->error = 2;
-    } catch (const std::exception&) {
+    } catch (const std::exception& e) {
+        if (std::getenv("QRACK_CAPI_DEBUG")) fprintf(stderr, "capi error: %s\n", e.what());
         s->error = 1;
     }
 }
@@ -1170,3 +1175,5 @@ void qrack_qcircuit_out_to_string(quid cid, char* out, uint64_t cap)
 }
 
 } // extern "C"
+
+#include "pinvoke_compat.inc"
