@@ -22,7 +22,7 @@ OBJS       := $(patsubst csrc/%.cpp,$(BUILD)/%.o,$(CPP_SRCS)) \
 
 all: $(TARGET)
 
-$(BUILD)/%.o: csrc/%.cpp $(wildcard csrc/*.hpp) $(wildcard csrc/common/*.hpp) $(wildcard csrc/hip/*.hpp)
+$(BUILD)/%.o: csrc/%.cpp $(wildcard csrc/*.hpp) $(wildcard csrc/*.inc) $(wildcard csrc/common/*.hpp) $(wildcard csrc/hip/*.hpp)
 	@mkdir -p $(dir $@)
 	$(HIPCC) $(CXXFLAGS) -c $< -o $@
 

@@ -214,3 +214,47 @@ def test_stack_inits_and_setters():
     L.SetReactiveSeparate(sid, True)
     assert L.GetUnitaryFidelity(sid) > 0.99
     L.destroy(sid)
+
+
+def test_c_example_compiles_and_runs():
+    """examples/pinvoke_bell.c builds against include/qrack_pinvoke_compat.h
+    and links the extension .so directly — reference-style C code runs
+    unchanged."""
+    import subprocess
+    import sys
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    exe = "/tmp/qa_pinvoke_bell_test"
+    soname = os.path.basename(_SO)
+    r = subprocess.run(
+        ["gcc", os.path.join(root, "examples", "pinvoke_bell.c"),
+         "-I", os.path.join(root, "include"),
+         "-L", os.path.join(root, "qrack_amd"), f"-l:{soname}",
+         f"-Wl,-rpath,{os.path.join(root, 'qrack_amd')}",
+         f"-lpython{sys.version_info.major}.{sys.version_info.minor}",
+         "-o", exe], capture_output=True, text=True)
+    assert r.returncode == 0, r.stderr
+    out = subprocess.run([exe], capture_output=True, text=True,
+                         env={**os.environ, "LD_LIBRARY_PATH": os.path.join(root, "qrack_amd")})
+    assert out.returncode == 0, out.stdout + out.stderr
+    assert "OK" in out.stdout
+
+
+def test_time_evolve():
+    """TimeEvolve with a single uncontrolled X Hamiltonian term:
+    exp(-i*H*t)|0> with H = X gives P(1) = sin^2(t)."""
+
+    class TEOH(ctypes.Structure):
+        _fields_ = [("target", ctypes.c_uint), ("controlLen", ctypes.c_uint),
+                    ("controls", ctypes.c_uint * 32)]
+
+    sid = L.init_count(uintq(1), False, False)
+    teo = (TEOH * 1)()
+    teo[0].target = 0
+    teo[0].controlLen = 0
+    # X as 8 doubles (re, im pairs row-major)
+    m = (ctypes.c_double * 8)(0, 0, 1, 0, 1, 0, 0, 0)
+    t = 0.6
+    L.TimeEvolve(sid, ctypes.c_double(t), uintq(1), teo, uintq(8), m)
+    assert abs(L.Prob(sid, uintq(0)) - math.sin(t) ** 2) < 1e-4
+    assert L.get_error(sid) == 0
+    L.destroy(sid)
