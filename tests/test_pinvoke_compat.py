@@ -258,3 +258,21 @@ def test_time_evolve():
     assert abs(L.Prob(sid, uintq(0)) - math.sin(t) ** 2) < 1e-4
     assert L.get_error(sid) == 0
     L.destroy(sid)
+
+
+@pytest.mark.gpu
+def test_pinvoke_compat_on_hip_engine():
+    """On a GPU box init_count's canonical stack resolves to the HIP engine —
+    the compat surface must drive it end to end at state-vector width."""
+    sid = L.init_count(uintq(24), False, False)
+    L.H(sid, uintq(23))
+    L.MCX(sid, uintq(1), arr([23]), uintq(0))
+    assert abs(L.Prob(sid, uintq(0)) - 0.5) < 1e-5
+    qs = arr(list(range(24)))
+    L.QFT(sid, uintq(24), qs)
+    L.IQFT(sid, uintq(24), qs)
+    assert abs(L.Prob(sid, uintq(0)) - 0.5) < 1e-4
+    e = L.PauliExpectation(sid, uintq(2), arr([0, 23]), arr([2, 2]))
+    assert abs(e - 1.0) < 1e-4
+    assert L.get_error(sid) == 0
+    L.destroy(sid)
