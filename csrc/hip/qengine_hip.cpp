@@ -954,10 +954,13 @@ template <typename R> void QEngineHIP<R>::QFT(bitLenInt start, bitLenInt length,
         if (ldsLow && ldsMid && start == 0u && (bitLenInt)i > (bitLenInt)tb &&
             maxQPower >= (ONE_BCI << tb)) {
             // mid columns per pass through a 2D LDS tile (6 -> two K=3
-            // groups; trailing 4/3/2 -> one uniform group)
+            // groups; trailing 4/3/2 -> one uniform group). A trailing 5
+            // takes a full 6-column pass dipping one column below the
+            // low-tile boundary (the kernel only needs colLo >= 6); the
+            // ladder below then peels to 4-alignment as usual.
             const int rem = (int)(i - (bitLenInt)tb);
-            const int nc = rem >= 6 ? 6 : (rem == 5 ? 4 : rem);
-            if (nc >= 2) {
+            const int nc = rem >= 5 ? 6 : rem;
+            if (nc >= 2 && (int)i - nc >= 6) {
                 HipProfScope prof("qft_mid_lds", stream);
                 launchQftMidLds<R>(dState, maxQPower, (int)i - nc, nc, +1, false, stream);
                 i -= (bitLenInt)nc;
