@@ -660,3 +660,28 @@ def test_mtrx2q_pair2_high_bits():
         for u, (a, b) in zip(us, pairs):
             qs.mtrx_2q([complex(x) for x in u.flatten()], a, b)
         assert float(qb.sum_sqr_diff(qs)) < 1e-5
+
+
+def test_qhybrid_paged_promotion_on_gpu():
+    """QHybrid's third tier: with QRACK_MAX_PAGE_QB forced low, a 16-qubit
+    hybrid sim promotes to QPager over HIP pages and must match the plain
+    engine (csrc/qhybrid.hpp three-tier migration)."""
+    import os as _os
+    old = _os.environ.get("QRACK_MAX_PAGE_QB")
+    _os.environ["QRACK_MAX_PAGE_QB"] = "14"
+    try:
+        h = qa.create_simulator(16, layers=["hybrid"], seed=4)
+        r = make(16, seed=4)
+        h.set_permutation(0x5A5A)
+        r.set_permutation(0x5A5A)
+        for i in range(0, 16, 3):
+            h.ry(0.3 + i, i)
+            r.ry(0.3 + i, i)
+        h.qft(0, 16)
+        r.qft(0, 16)
+        assert_states_close(h.get_state_vector(), r.get_state_vector(), 2e-4)
+    finally:
+        if old is None:
+            _os.environ.pop("QRACK_MAX_PAGE_QB", None)
+        else:
+            _os.environ["QRACK_MAX_PAGE_QB"] = old
