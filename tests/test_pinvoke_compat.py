@@ -276,3 +276,19 @@ def test_pinvoke_compat_on_hip_engine():
     assert abs(e - 1.0) < 1e-4
     assert L.get_error(sid) == 0
     L.destroy(sid)
+
+
+def test_compose_with_named_ids():
+    """Compose names the appended qubits via the caller's id array; a
+    non-index id must switch the target to explicit qid mapping."""
+    a = L.init_count(uintq(1), False, False)
+    L.X(a, uintq(0))
+    b = L.init_count(uintq(1), False, False)
+    L.H(b, uintq(0))
+    L.Compose(a, b, arr([42]))
+    assert L.num_qubits(a) == 2
+    assert abs(L.Prob(a, uintq(42)) - 0.5) < 1e-6
+    assert abs(L.Prob(a, uintq(0)) - 1.0) < 1e-6
+    assert L.get_error(a) == 0
+    L.destroy(a)
+    L.destroy(b)
