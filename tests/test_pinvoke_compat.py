@@ -292,3 +292,29 @@ def test_compose_with_named_ids():
     assert L.get_error(a) == 0
     L.destroy(a)
     L.destroy(b)
+
+
+def test_qcircuit_file_roundtrip_into_handle():
+    """qcircuit_in_from_file loads INTO the caller's existing handle
+    (reference semantics)."""
+    import tempfile
+    L.init_qcircuit.restype = uintq
+    L.get_qcircuit_qubit_count.restype = uintq
+    cid = L.init_qcircuit(False, False)
+    h = (ctypes.c_double * 8)(0.7071067811865476, 0, 0.7071067811865476, 0,
+                              0.7071067811865476, 0, -0.7071067811865476, 0)
+    L.qcircuit_append_1qb(cid, h, uintq(1))
+    with tempfile.NamedTemporaryFile(suffix=".qc", delete=False) as tf:
+        path = tf.name
+    L.qcircuit_out_to_file(cid, path.encode())
+    cid2 = L.init_qcircuit(False, False)
+    L.qcircuit_in_from_file(cid2, path.encode())
+    assert L.get_qcircuit_qubit_count(cid2) == L.get_qcircuit_qubit_count(cid)
+    # run the loaded circuit: H on qubit 1
+    sid = L.init_count(uintq(2), False, False)
+    L.qcircuit_run(cid2, sid)
+    assert abs(L.Prob(sid, uintq(1)) - 0.5) < 1e-6
+    L.destroy(sid)
+    L.destroy_qcircuit(cid)
+    L.destroy_qcircuit(cid2)
+    os.unlink(path)
