@@ -47,6 +47,24 @@ asan:
 	    tools/asan_smoke.cpp -o build/asan_smoke -lpthread
 	./build/asan_smoke
 
+# full-suite ASAN lane: CPU-only extension instrumented with ASan+UBSan,
+# whole pytest CPU battery runs against it (catches heap bugs in every
+# layer incl. the C ABI / pinvoke compat surfaces)
+asan-suite:
+	@mkdir -p build/asan_ext/qrack_amd
+	g++ -O1 -g -shared -fPIC -std=c++17 -fsanitize=address -fno-omit-frame-pointer \
+	    -Icsrc $(PYBIND_INC) \
+	    csrc/bindings.cpp csrc/capi.cpp csrc/qinterface.cpp csrc/qengine_cpu.cpp \
+	    csrc/qengine_sparse.cpp csrc/qengine_turboquant.cpp csrc/qstabilizer.cpp \
+	    csrc/qstabilizerhybrid.cpp csrc/qunit.cpp csrc/qbdt.cpp csrc/qpager.cpp \
+	    csrc/qfactory.cpp csrc/common/parallel_for.cpp \
+	    -o build/asan_ext/qrack_amd/_qrack$(EXT_SUFFIX) -lpthread
+	cp qrack_amd/*.py build/asan_ext/qrack_amd/
+	cd build/asan_ext && \
+	  LD_PRELOAD="$$(g++ -print-file-name=libasan.so) $$(g++ -print-file-name=libstdc++.so.6)" \
+	  PYTHONMALLOC=malloc ASAN_OPTIONS=detect_leaks=0 PYTHONPATH=.:$(CURDIR) \
+	  timeout 2400 $(PYTHON) -m pytest $(CURDIR)/tests -x -q -m "not gpu" -p no:cacheprovider
+
 # TSAN lane: exercises the thread-pool ParallelFor under the race detector
 # (the reference has no sanitizer lanes; the thread-safety
 # contract — one engine instance per thread — is enforced here instead)
