@@ -13,6 +13,8 @@ false-negative, and the mod-ALU modN heap overflow.
 import sys
 import os
 
+_SEED_OFF = int(os.environ.get("QA_FUZZ_SEED", "0"))
+
 sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)), ".."))
 
 def fuzz_state():
@@ -32,7 +34,7 @@ def fuzz_state():
             ["pager", "cpu"], ["qunit", "stabilizer", ], ["hybrid"], ["tensor_network", "cpu"],
         ]
     N = 5
-    rng = np.random.default_rng(20260913)
+    rng = np.random.default_rng(20260913 + _SEED_OFF)
     fails = 0
     for trial in range(60):
         layers = STACKS[trial % len(STACKS)]
@@ -103,7 +105,7 @@ def fuzz_measure():
         STACKS = [["cpu"], ["sparse"], ["stabilizer_hybrid", "cpu"], ["qunit", "cpu"],
                   ["qunit", "stabilizer_hybrid", "cpu"], ["pager", "cpu"], ["bdt"], ["hybrid"]]
     N = 5
-    rng = np.random.default_rng(99)
+    rng = np.random.default_rng(99 + _SEED_OFF)
     fails = 0
     for trial in range(64):
         layers = STACKS[trial % len(STACKS)]
@@ -168,7 +170,7 @@ def fuzz_struct():
 
     STACKS = [["cpu"], ["sparse"], ["stabilizer_hybrid", "cpu"], ["qunit", "cpu"],
               ["qunit", "stabilizer_hybrid", "cpu"], ["pager", "cpu"]]
-    rng = np.random.default_rng(123)
+    rng = np.random.default_rng(123 + _SEED_OFF)
     fails = 0
     for trial in range(48):
         layers = STACKS[trial % len(STACKS)]
@@ -223,7 +225,7 @@ def fuzz_serial():
     import tempfile
     sys.path.insert(0, "/root/repo")
 
-    rng = np.random.default_rng(77)
+    rng = np.random.default_rng(77 + _SEED_OFF)
     fails = 0
     # 1) lossy roundtrip across layer stacks and precisions
     for trial in range(24):
@@ -279,7 +281,7 @@ def fuzz_alu():
     import numpy as np
     import qrack_amd as qa
 
-    rng = np.random.default_rng(31)
+    rng = np.random.default_rng(31 + _SEED_OFF)
     fails = 0
     STACKS = [["cpu"], ["qunit", "cpu"], ["pager", "cpu"], ["stabilizer_hybrid", "cpu"], ["hybrid"]]
     for trial in range(50):
@@ -320,7 +322,7 @@ def fuzz_indexed():
     import numpy as np
     import qrack_amd as qa
 
-    rng = np.random.default_rng(41)
+    rng = np.random.default_rng(41 + _SEED_OFF)
     fails = 0
     for trial in range(30):
         # IndexedLDA semantics: value register loaded from table[index]
