@@ -518,7 +518,7 @@ template <typename R> __global__ void k_reduce(const cplx<R>* sv, ReduceArgs a, 
         case ReduceOp::EXP_PERM_SQ: {
             double val = a.offset;
             for (int b = 0; b < a.nBits; ++b) {
-                if ((i >> a.bits[b]) & 1u) val += (double)a.perms[b];
+                if ((i >> a.bitsArr[b]) & 1u) val += (double)a.permsArr[b];
             }
             s += ((ReduceOp)op == ReduceOp::EXP_PERM) ? val * n : val * val * n;
             break;

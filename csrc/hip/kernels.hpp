@@ -77,14 +77,20 @@ enum class ReduceOp : int {
     NORM_FLOOR,     // sum |amp|^2 with amplitude floor (UpdateRunningNorm)
 };
 
+constexpr int QA_REDUCE_MAX_BITS = 32;
+
 struct ReduceArgs {
     bitCapInt maxI;
     bitCapInt mask;
     bitCapInt perm;
     double offset;       // EXP_PERM value offset
     double normThresh;   // NORM_FLOOR
-    const bitLenInt* bits;   // device ptrs for EXP_PERM
-    const bitCapInt* perms;
+    // EXP_PERM inputs ride IN the kernarg segment (<= 32 listed bits; the
+    // engine falls back to the generic host path beyond that) — no device
+    // staging buffers, so no stream-ordered-allocator exposure on the
+    // expectation/variance query path
+    bitLenInt bitsArr[QA_REDUCE_MAX_BITS];
+    bitCapInt permsArr[QA_REDUCE_MAX_BITS];
     int nBits;
 };
 

@@ -1191,51 +1191,38 @@ template <typename R>
 double QEngineHIP<R>::ExpectationBitsFactorized(
     const std::vector<bitLenInt>& bits, const std::vector<bitCapInt>& perms, bitCapInt offset)
 {
+    if (bits.size() > (size_t)QA_REDUCE_MAX_BITS) {
+        return QEngine<R>::ExpectationBitsFactorized(bits, perms, offset);
+    }
     QA_HIP_CHECK(hipSetDevice(deviceId));
-    bitLenInt* dBits = nullptr;
-    bitCapInt* dPerms = nullptr;
-    QA_HIP_CHECK(hipMallocAsync(&dBits, sizeof(bitLenInt) * bits.size(), stream));
-    QA_HIP_CHECK(hipMallocAsync(&dPerms, sizeof(bitCapInt) * perms.size(), stream));
-    QA_HIP_CHECK(hipMemcpyAsync(
-        dBits, bits.data(), sizeof(bitLenInt) * bits.size(), hipMemcpyHostToDevice, stream));
-    QA_HIP_CHECK(hipMemcpyAsync(
-        dPerms, perms.data(), sizeof(bitCapInt) * perms.size(), hipMemcpyHostToDevice, stream));
     ReduceArgs a{};
     a.maxI = maxQPower;
     a.offset = (double)offset;
-    a.bits = dBits;
-    a.perms = dPerms;
+    for (size_t b = 0; b < bits.size(); ++b) {
+        a.bitsArr[b] = bits[b];
+        a.permsArr[b] = perms[b];
+    }
     a.nBits = (int)bits.size();
-    const double e = reduceSum((int)ReduceOp::EXP_PERM, a);
-    QA_HIP_CHECK(hipFreeAsync(dBits, stream));
-    QA_HIP_CHECK(hipFreeAsync(dPerms, stream));
-    return e;
+    return reduceSum((int)ReduceOp::EXP_PERM, a);
 }
 
 template <typename R>
 double QEngineHIP<R>::VarianceBitsAll(const std::vector<bitLenInt>& bits, bitCapInt offset)
 {
+    if (bits.size() > (size_t)QA_REDUCE_MAX_BITS) {
+        return QEngine<R>::VarianceBitsAll(bits, offset);
+    }
     const double mean = this->ExpectationBitsAll(bits, offset);
     QA_HIP_CHECK(hipSetDevice(deviceId));
-    std::vector<bitCapInt> perms;
-    for (size_t b = 0; b < bits.size(); ++b) perms.push_back(pow2((bitLenInt)b));
-    bitLenInt* dBits = nullptr;
-    bitCapInt* dPerms = nullptr;
-    QA_HIP_CHECK(hipMallocAsync(&dBits, sizeof(bitLenInt) * bits.size(), stream));
-    QA_HIP_CHECK(hipMallocAsync(&dPerms, sizeof(bitCapInt) * perms.size(), stream));
-    QA_HIP_CHECK(hipMemcpyAsync(
-        dBits, bits.data(), sizeof(bitLenInt) * bits.size(), hipMemcpyHostToDevice, stream));
-    QA_HIP_CHECK(hipMemcpyAsync(
-        dPerms, perms.data(), sizeof(bitCapInt) * perms.size(), hipMemcpyHostToDevice, stream));
     ReduceArgs a{};
     a.maxI = maxQPower;
     a.offset = (double)offset;
-    a.bits = dBits;
-    a.perms = dPerms;
+    for (size_t b = 0; b < bits.size(); ++b) {
+        a.bitsArr[b] = bits[b];
+        a.permsArr[b] = pow2((bitLenInt)b);
+    }
     a.nBits = (int)bits.size();
     const double e2 = reduceSum((int)ReduceOp::EXP_PERM_SQ, a);
-    QA_HIP_CHECK(hipFreeAsync(dBits, stream));
-    QA_HIP_CHECK(hipFreeAsync(dPerms, stream));
     return e2 - mean * mean;
 }
 

@@ -685,3 +685,22 @@ def test_qhybrid_paged_promotion_on_gpu():
             _os.environ.pop("QRACK_MAX_PAGE_QB", None)
         else:
             _os.environ["QRACK_MAX_PAGE_QB"] = old
+
+
+def test_variance_bits_all_vs_cpu():
+    """Regression for a fuzz-caught intermittent: HIP variance_bits_all
+    must match the CPU engine (EXP_PERM args now ride in the kernarg
+    segment — no async staging buffers on the query path)."""
+    ops = [("t", 3), ("cnot", 2, 4), ("ry", 1.1374093532031537, 3),
+           ("ry", 0.24300806879584336, 3), ("ry", 2.280324914256294, 4),
+           ("t", 4), ("h", 3), ("ry", 2.439288787613517, 4), ("t", 2),
+           ("cnot", 1, 2), ("cnot", 4, 2)]
+    q = make(5, seed=3)
+    cp = qa.create_simulator(5, engine="cpu", seed=3)
+    for op in ops:
+        getattr(q, op[0])(*op[1:])
+        getattr(cp, op[0])(*op[1:])
+    for _ in range(10):
+        assert abs(q.variance_bits_all([0, 1, 2]) - cp.variance_bits_all([0, 1, 2])) < 1e-3
+        assert abs(q.expectation_bits_all([0, 1, 2, 3, 4])
+                   - cp.expectation_bits_all([0, 1, 2, 3, 4])) < 1e-3
