@@ -330,9 +330,14 @@ protected:
                         cplx<R>(1, 0), polar<R>(1, (R)th), shards[p.c].mapped);
                 }
             } else { // p.c == q
+                // X_c·CX·CP(θ) = CX·CP(θ)·{e^{iθ}P_c(−θ)X_c}{P_t(−θ)X_t};
+                // P_t(−θ)·X_t = [[0,1],[e^{−iθ},0]] — the phase rides the
+                // BOTTOM-LEFT slot (fuzz-caught: the top-right variant is
+                // only equivalent up to global phase at θ = π, so S-commutes
+                // masked it and T-commutes exposed it)
                 const double th = p.angle;
                 shards[p.t].unit->Invert(
-                    polar<R>(1, (R)(-th)), cplx<R>(1, 0), shards[p.t].mapped);
+                    cplx<R>(1, 0), polar<R>(1, (R)(-th)), shards[p.t].mapped);
                 tr = tr * polar<R>(1, (R)th);
             }
         }

@@ -401,3 +401,26 @@ def test_graph_state_stays_separable_until_measurement():
         # makes are_factorized False, but Z marginals stay exact & local
         assert abs(q.prob(a) - 0.5) < 1e-6
     assert q.get_unitary_fidelity() == pytest.approx(1.0)
+
+
+def test_commute_invert_control_slot_regression():
+    """Fuzz-caught: a non-Clifford diagonal commuted onto a pending CX's
+    TARGET, followed by an invert on its CONTROL, must place the partner
+    phase on the bottom-left antidiagonal slot (qunit.hpp CommuteInvert
+    p.c==q branch). S-commutes mask the wrong slot (global phase at
+    theta=pi); T-commutes expose it."""
+    import numpy as np
+    import qrack_amd as qa
+
+    for g1, g2 in (("t", "y"), ("t", "x"), ("s", "y"), ("t", "z")):
+        q = qa.create_simulator(5, layers=["qunit", "cpu"], seed=3)
+        cp = qa.create_simulator(5, engine="cpu", seed=3)
+        for s in (q, cp):
+            s.ry(0.7, 4)
+            s.ry(0.5, 0)
+            s.cnot(4, 0)
+            getattr(s, g1)(0)
+            getattr(s, g2)(4)
+        sv = np.asarray(q.get_state_vector()).astype(np.complex128)
+        rv = np.asarray(cp.get_state_vector()).astype(np.complex128)
+        assert abs(np.vdot(rv, sv)) > 1 - 1e-6, (g1, g2)
